@@ -1,0 +1,155 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: metric-updates/sec for a 16-metric classification
+MetricCollection (BASELINE.json config 3 shape: multiclass C=1000, batch 8192
+bf16 logits, synthetic data), one rank per GPU over RCCL.
+
+One timed step = one ``collection.update(preds, target)``; every
+``--compute-every`` steps the timed region also runs ``collection.compute()``
+(sync_on_compute=True => RCCL state sync in multi-GPU runs).
+
+Output: ONE JSON line on rank 0 with the whole-job aggregate
+metric_updates_per_sec (n_gpus * 16 metrics * steps / max-rank elapsed).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def build_collection(num_classes: int, device: torch.device, curve_thresholds: int = 200):
+    import metrics_amd as ma
+
+    kw = dict(num_classes=num_classes, validate_args=False)
+    metrics = {
+        "acc_micro": ma.MulticlassAccuracy(average="micro", **kw),
+        "acc_macro": ma.MulticlassAccuracy(average="macro", **kw),
+        "precision": ma.MulticlassPrecision(average="macro", **kw),
+        "recall": ma.MulticlassRecall(average="macro", **kw),
+        "f1": ma.MulticlassF1Score(average="macro", **kw),
+        "fbeta2": ma.MulticlassFBetaScore(beta=2.0, average="macro", **kw),
+        "specificity": ma.MulticlassSpecificity(average="macro", **kw),
+        "npv": ma.MulticlassNegativePredictiveValue(average="macro", **kw),
+        "hamming": ma.MulticlassHammingDistance(average="macro", **kw),
+        "jaccard": ma.MulticlassJaccardIndex(average="macro", **kw),
+        "exact_match": ma.MulticlassExactMatch(**kw),
+        "cohen_kappa": ma.MulticlassCohenKappa(**kw),
+        "mcc": ma.MulticlassMatthewsCorrCoef(**kw),
+        "confmat": ma.MulticlassConfusionMatrix(**kw),
+        "auroc": ma.MulticlassAUROC(average="macro", thresholds=curve_thresholds, **kw),
+        "avg_precision": ma.MulticlassAveragePrecision(average="macro", thresholds=curve_thresholds, **kw),
+    }
+    coll = ma.MetricCollection(metrics)
+    return coll.to(device)
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=64)
+    parser.add_argument("--warmup", type=int, default=8)
+    parser.add_argument("--batch", type=int, default=8192)
+    parser.add_argument("--classes", type=int, default=1000)
+    parser.add_argument("--compute-every", type=int, default=32)
+    parser.add_argument("--curve-thresholds", type=int, default=200)
+    args = parser.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    n_gpus = max(world, 1)
+
+    use_gpu = torch.cuda.is_available()
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    distributed = world > 1
+    if distributed:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        torch.distributed.init_process_group("nccl" if use_gpu else "gloo")
+
+    dtype = torch.bfloat16 if use_gpu else torch.float32
+    torch.manual_seed(1234 + rank)
+
+    # synthetic data of the BASELINE config shape; a few pre-generated batches
+    # cycled so RNG is outside the timed region
+    n_unique = 4
+    preds = [torch.randn(args.batch, args.classes, device=device, dtype=dtype) for _ in range(n_unique)]
+    target = [torch.randint(0, args.classes, (args.batch,), device=device) for _ in range(n_unique)]
+
+    coll = build_collection(args.classes, device, args.curve_thresholds)
+
+    def one_step(i: int) -> None:
+        coll.update(preds[i % n_unique], target[i % n_unique])
+        if args.compute_every and (i + 1) % args.compute_every == 0:
+            coll.compute()
+
+    # warmup (untimed)
+    for i in range(args.warmup):
+        one_step(i)
+    coll.reset()
+
+    if distributed:
+        torch.distributed.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(i)
+    if use_gpu:
+        torch.cuda.synchronize()
+    if distributed:
+        torch.distributed.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if distributed:
+        t = torch.tensor([elapsed], device=device if use_gpu else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = t.item()
+
+    n_metrics = 16
+    value = n_gpus * n_metrics * args.steps / elapsed
+    ms_per_step = 1000.0 * elapsed / args.steps
+
+    if rank == 0:
+        print(
+            json.dumps({
+                "metric": "metric_updates_per_sec",
+                "value": value,
+                "unit": "metric-updates/s",
+                "n_gpus": n_gpus,
+                "steps": args.steps,
+                "warmup": args.warmup,
+                "ms_per_step": ms_per_step,
+                "higher_is_better": True,
+                "scaling": "weak",
+                "vs_baseline": None,
+                "dtype": "bf16" if use_gpu else "fp32",
+                "data": "synthetic",
+                "config": {
+                    "model": "16-metric multiclass MetricCollection",
+                    "global_batch": args.batch * n_gpus,
+                    "seq_len": args.classes,
+                    "parallelism": f"dp{n_gpus}",
+                    "num_classes": args.classes,
+                    "batch_per_gpu": args.batch,
+                    "compute_every": args.compute_every,
+                    "curve_thresholds": args.curve_thresholds,
+                    "validate_args": False,
+                },
+            })
+        )
+
+    if distributed:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

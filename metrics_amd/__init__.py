@@ -1,0 +1,49 @@
+"""metrics_amd — an MI355X-native metrics engine with the torchmetrics API.
+
+Compute path: PyTorch-ROCm tensors + in-tree gfx950 HIP kernels for the hot
+reductions (see ``csrc/``); distributed state sync over RCCL/xGMI with the
+reductions mapped onto fused all-reduces (see ``utilities/distributed.py``).
+"""
+import logging as __logging
+import os
+
+_logger = __logging.getLogger("metrics_amd")
+_logger.addHandler(__logging.StreamHandler())
+_logger.setLevel(__logging.INFO)
+
+_PACKAGE_ROOT = os.path.dirname(__file__)
+
+__version__ = "0.1.0"
+
+from metrics_amd.aggregation import (  # noqa: E402
+    CatMetric,
+    MaxMetric,
+    MeanMetric,
+    MinMetric,
+    RunningMean,
+    RunningSum,
+    SumMetric,
+)
+from metrics_amd.collections import MetricCollection  # noqa: E402
+from metrics_amd.metric import CompositionalMetric, Metric  # noqa: E402
+from metrics_amd import classification, functional, ops, utilities, wrappers  # noqa: E402
+from metrics_amd.classification import *  # noqa: E402,F401,F403
+
+__all__ = [
+    "CatMetric",
+    "CompositionalMetric",
+    "MaxMetric",
+    "MeanMetric",
+    "Metric",
+    "MetricCollection",
+    "MinMetric",
+    "RunningMean",
+    "RunningSum",
+    "SumMetric",
+    "classification",
+    "functional",
+    "ops",
+    "utilities",
+    "wrappers",
+]
+__all__ += classification.__all__

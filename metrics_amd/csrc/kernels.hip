@@ -1,0 +1,615 @@
+// MI355X (gfx950 / CDNA4) metric-update kernels.
+//
+// Design notes (see /root/repo/SURVEY.md §2.9 for the reference call sites
+// each kernel replaces):
+//  - every kernel is memory-bound: the rules applied are 64-wide wavefronts,
+//    coalesced vectorized loads (float4 / ushort4-bf16), grid capped at
+//    ~2048 blocks with grid-stride loops, per-wave shuffle reductions, and
+//    integer atomics (order-independent => deterministic by construction).
+//  - K1/K4 (torchmetrics functional/classification/stat_scores.py:371-449,
+//    confusion_matrix.py): fused argmax + per-class tp/fp/fn (+ optional
+//    full confusion matrix) in ONE pass over the (B,C) logits.
+//  - K5/K2 (precision_recall_curve.py:30-251): threshold curves as a
+//    bucketized histogram (binary search into sorted thresholds) — the
+//    (T,2,2) confmat state is a suffix-sum of the histogram, done on device.
+//  - K13 (functional/regression/*): fused elementwise-error reductions with
+//    fp64 block partials reduced in a second fixed-order pass
+//    (deterministic, unlike float atomics).
+//  - K6/K7 (functional/detection/iou.py): all-pairs box IoU with
+//    GIoU/DIoU/CIoU epilogues, boxes1 tile staged in LDS.
+//
+// Build: hipcc --offload-arch=gfx950 -O3 (see csrc/build.py). No CUDA
+// compatibility paths; wave size is hard-coded 64.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <math.h>
+#include <stdint.h>
+
+#define WAVE 64
+#define CHECK(x)                                                                                   \
+    do {                                                                                           \
+        hipError_t _e = (x);                                                                       \
+        if (_e != hipSuccess) return (int)_e;                                                      \
+    } while (0)
+
+typedef long long ll;
+
+__device__ __forceinline__ float bf16_to_f32(unsigned short u) {
+    unsigned int v = ((unsigned int)u) << 16;
+    return __uint_as_float(v);
+}
+
+// ---------------------------------------------------------------------------
+// K1a: fused row-argmax + multiclass stat scores from logits (B, C)
+// one wave per row; tp/fp/fn per class via global atomics; optional confmat.
+// Tie-break matches torch.argmax: lowest index wins.
+// ---------------------------------------------------------------------------
+template <typename T, bool IS_BF16>
+__global__ void __launch_bounds__(256) k_mc_stat_logits(
+    const T* __restrict__ preds, const ll* __restrict__ target, ll B, ll C, ll ignore_index,
+    int has_ignore, unsigned long long* __restrict__ tp, unsigned long long* __restrict__ fp,
+    unsigned long long* __restrict__ fn, unsigned long long* __restrict__ confmat,
+    unsigned long long* __restrict__ valid_count, ll* __restrict__ argmax_out) {
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wave_in_block = threadIdx.x / WAVE;
+    const int waves_per_block = blockDim.x / WAVE;
+    const ll row0 = (ll)blockIdx.x * waves_per_block + wave_in_block;
+    const ll row_stride = (ll)gridDim.x * waves_per_block;
+
+    for (ll row = row0; row < B; row += row_stride) {
+        const T* prow = preds + row * C;
+        float best = -INFINITY;
+        ll best_idx = 0x7fffffffffffffffLL;
+        // vectorized 4-wide when alignment allows (row start 16B/8B aligned iff C%4==0)
+        if ((C & 3) == 0) {
+            const ll nvec = C / 4;
+            if (IS_BF16) {
+                const ushort4* pv = reinterpret_cast<const ushort4*>(prow);
+                for (ll v = lane; v < nvec; v += WAVE) {
+                    ushort4 u = pv[v];
+                    float f0 = bf16_to_f32(u.x), f1 = bf16_to_f32(u.y);
+                    float f2 = bf16_to_f32(u.z), f3 = bf16_to_f32(u.w);
+                    ll c = v * 4;
+                    if (f0 > best || (f0 == best && c < best_idx)) { best = f0; best_idx = c; }
+                    if (f1 > best || (f1 == best && c + 1 < best_idx)) { best = f1; best_idx = c + 1; }
+                    if (f2 > best || (f2 == best && c + 2 < best_idx)) { best = f2; best_idx = c + 2; }
+                    if (f3 > best || (f3 == best && c + 3 < best_idx)) { best = f3; best_idx = c + 3; }
+                }
+            } else {
+                const float4* pv = reinterpret_cast<const float4*>(prow);
+                for (ll v = lane; v < nvec; v += WAVE) {
+                    float4 u = pv[v];
+                    ll c = v * 4;
+                    if (u.x > best || (u.x == best && c < best_idx)) { best = u.x; best_idx = c; }
+                    if (u.y > best || (u.y == best && c + 1 < best_idx)) { best = u.y; best_idx = c + 1; }
+                    if (u.z > best || (u.z == best && c + 2 < best_idx)) { best = u.z; best_idx = c + 2; }
+                    if (u.w > best || (u.w == best && c + 3 < best_idx)) { best = u.w; best_idx = c + 3; }
+                }
+            }
+        } else {
+            for (ll c = lane; c < C; c += WAVE) {
+                float f = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(prow)[c])
+                                  : (float)prow[c];
+                if (f > best || (f == best && c < best_idx)) { best = f; best_idx = c; }
+            }
+        }
+        // wave shuffle reduce: max value, lowest index on tie
+        for (int off = WAVE / 2; off > 0; off >>= 1) {
+            float ov = __shfl_down(best, off);
+            ll oi = __shfl_down(best_idx, off);
+            if (ov > best || (ov == best && oi < best_idx)) { best = ov; best_idx = oi; }
+        }
+        if (lane == 0) {
+            ll t = target[row];
+            ll p = best_idx;
+            if (argmax_out) argmax_out[row] = p;
+            // bounds guard: bad labels with validate_args=False must not corrupt memory
+            if (!(has_ignore && t == ignore_index) && t >= 0 && t < C && p >= 0 && p < C) {
+                atomicAdd(valid_count, 1ULL);
+                if (p == t) {
+                    atomicAdd(&tp[t], 1ULL);
+                } else {
+                    atomicAdd(&fp[p], 1ULL);
+                    atomicAdd(&fn[t], 1ULL);
+                }
+                if (confmat) atomicAdd(&confmat[t * C + p], 1ULL);
+            }
+        }
+    }
+}
+
+// K1b: multiclass stat scores from integer label preds (element-wise)
+__global__ void __launch_bounds__(256) k_mc_stat_labels(
+    const ll* __restrict__ preds, const ll* __restrict__ target, ll N, ll C, ll ignore_index,
+    int has_ignore, unsigned long long* __restrict__ tp, unsigned long long* __restrict__ fp,
+    unsigned long long* __restrict__ fn, unsigned long long* __restrict__ confmat,
+    unsigned long long* __restrict__ valid_count) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    ll stride = (ll)gridDim.x * blockDim.x;
+    unsigned long long local_valid = 0;
+    for (; i < N; i += stride) {
+        ll t = target[i];
+        if (has_ignore && t == ignore_index) continue;
+        ll p = preds[i];
+        if (t < 0 || t >= C || p < 0 || p >= C) continue;
+        local_valid++;
+        if (p == t) {
+            atomicAdd(&tp[t], 1ULL);
+        } else {
+            atomicAdd(&fp[p], 1ULL);
+            atomicAdd(&fn[t], 1ULL);
+        }
+        if (confmat) atomicAdd(&confmat[t * C + p], 1ULL);
+    }
+    if (local_valid) atomicAdd(valid_count, local_valid);
+}
+
+// ---------------------------------------------------------------------------
+// K1c: LDS-privatized bincount (deterministic: integer atomics)
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256) k_bincount_lds(
+    const ll* __restrict__ x, ll N, ll bins, unsigned long long* __restrict__ out) {
+    extern __shared__ unsigned int lbins[];
+    for (ll b = threadIdx.x; b < bins; b += blockDim.x) lbins[b] = 0;
+    __syncthreads();
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    ll stride = (ll)gridDim.x * blockDim.x;
+    for (; i < N; i += stride) {
+        ll v = x[i];
+        if (v >= 0 && v < bins) atomicAdd(&lbins[v], 1u);
+    }
+    __syncthreads();
+    for (ll b = threadIdx.x; b < bins; b += blockDim.x) {
+        unsigned int c = lbins[b];
+        if (c) atomicAdd(&out[b], (unsigned long long)c);
+    }
+}
+
+__global__ void __launch_bounds__(256) k_bincount_global(
+    const ll* __restrict__ x, ll N, ll bins, unsigned long long* __restrict__ out) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    ll stride = (ll)gridDim.x * blockDim.x;
+    for (; i < N; i += stride) {
+        ll v = x[i];
+        if (v >= 0 && v < bins) atomicAdd(&out[v], 1ULL);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// K4: fused binary stat scores. Counts BOTH the raw-threshold and the
+// sigmoid-threshold interpretation in one pass plus an "outside [0,1]" flag,
+// so the host picks the right one without a device sync (the reference's
+// normalize_logits_if_needed contortion, done properly in one kernel).
+// out layout: [2][4] = {raw,sig} x {tp,fp,tn,fn}; flag: 1 if any value
+// outside [0,1].
+// ---------------------------------------------------------------------------
+template <typename T, bool IS_BF16>
+__global__ void __launch_bounds__(256) k_binary_stat(
+    const T* __restrict__ preds, const ll* __restrict__ target, ll N, float threshold,
+    ll ignore_index, int has_ignore, unsigned long long* __restrict__ out,
+    unsigned int* __restrict__ outside_flag) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    ll stride = (ll)gridDim.x * blockDim.x;
+    // thresholding sigmoid(x) > thr  <=>  x > logit(thr)
+    const float logit_thr = logf(threshold / (1.0f - threshold));
+    unsigned long long cnt[2][4] = {{0, 0, 0, 0}, {0, 0, 0, 0}};
+    unsigned int outside = 0;
+    for (; i < N; i += stride) {
+        ll t = target[i];
+        if (has_ignore && t == ignore_index) continue;
+        float p = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(preds)[i])
+                          : (float)preds[i];
+        outside |= (p < 0.0f || p > 1.0f) ? 1u : 0u;
+        int pr_raw = p > threshold;
+        int pr_sig = p > logit_thr;
+        int tt = (int)t;
+        // idx: tp=0 fp=1 tn=2 fn=3
+        cnt[0][pr_raw == 1 ? (tt == 1 ? 0 : 1) : (tt == 0 ? 2 : 3)]++;
+        cnt[1][pr_sig == 1 ? (tt == 1 ? 0 : 1) : (tt == 0 ? 2 : 3)]++;
+    }
+    // wave reduce then one atomic per wave per counter
+    for (int v = 0; v < 2; v++)
+        for (int k = 0; k < 4; k++) {
+            unsigned long long c = cnt[v][k];
+            for (int off = WAVE / 2; off > 0; off >>= 1) c += __shfl_down(c, off);
+            if ((threadIdx.x & (WAVE - 1)) == 0 && c) atomicAdd(&out[v * 4 + k], c);
+        }
+    for (int off = WAVE / 2; off > 0; off >>= 1) outside |= __shfl_down(outside, off);
+    if ((threadIdx.x & (WAVE - 1)) == 0 && outside) atomicOr(outside_flag, 1u);
+}
+
+// K4b: fused multilabel stat scores: (N, L) -> per-label [2][L][4] counters.
+template <typename T, bool IS_BF16>
+__global__ void __launch_bounds__(256) k_multilabel_stat(
+    const T* __restrict__ preds, const ll* __restrict__ target, ll N, ll L, float threshold,
+    ll ignore_index, int has_ignore, unsigned long long* __restrict__ out,
+    unsigned int* __restrict__ outside_flag) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    ll stride = (ll)gridDim.x * blockDim.x;
+    const float logit_thr = logf(threshold / (1.0f - threshold));
+    unsigned int outside = 0;
+    for (; i < N * L; i += stride) {
+        ll l = i % L;
+        ll t = target[i];
+        if (has_ignore && t == ignore_index) continue;
+        float p = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(preds)[i])
+                          : (float)preds[i];
+        outside |= (p < 0.0f || p > 1.0f) ? 1u : 0u;
+        int pr_raw = p > threshold;
+        int pr_sig = p > logit_thr;
+        int tt = (int)t;
+        atomicAdd(&out[(0 * L + l) * 4 + (pr_raw == 1 ? (tt == 1 ? 0 : 1) : (tt == 0 ? 2 : 3))], 1ULL);
+        atomicAdd(&out[(1 * L + l) * 4 + (pr_sig == 1 ? (tt == 1 ? 0 : 1) : (tt == 0 ? 2 : 3))], 1ULL);
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) outside |= __shfl_down(outside, off);
+    if ((threadIdx.x & (WAVE - 1)) == 0 && outside) atomicOr(outside_flag, 1u);
+}
+
+// ---------------------------------------------------------------------------
+// K5/K2: threshold-curve histogram. preds are NORMALIZED probabilities.
+// bucket j = #thresholds <= p  (searchsorted right) in [0, T];
+// hist layout (T+1, 2): [bucket][target]. The (T,2,2) confmat state is the
+// suffix-sum, computed by k_curve_suffix below.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ int bucket_of(float p, const float* __restrict__ thr, int T) {
+    int lo = 0, hi = T;  // count of thresholds <= p
+    while (lo < hi) {
+        int mid = (lo + hi) >> 1;
+        if (thr[mid] <= p) lo = mid + 1;
+        else hi = mid;
+    }
+    return lo;
+}
+
+template <typename T_, bool IS_BF16>
+__global__ void __launch_bounds__(256) k_binary_curve_hist(
+    const T_* __restrict__ preds, const ll* __restrict__ target, ll N,
+    const float* __restrict__ thresholds, int T, ll ignore_index, int has_ignore,
+    unsigned long long* __restrict__ hist /* (T+1,2) */) {
+    extern __shared__ unsigned int lhist[];  // (T+1)*2
+    float* sthr = (float*)&lhist[(T + 1) * 2];
+    for (int b = threadIdx.x; b < (T + 1) * 2; b += blockDim.x) lhist[b] = 0;
+    for (int b = threadIdx.x; b < T; b += blockDim.x) sthr[b] = thresholds[b];
+    __syncthreads();
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    ll stride = (ll)gridDim.x * blockDim.x;
+    for (; i < N; i += stride) {
+        ll t = target[i];
+        if (has_ignore && t == ignore_index) continue;
+        float p = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(preds)[i])
+                          : (float)preds[i];
+        int j = bucket_of(p, sthr, T);
+        atomicAdd(&lhist[j * 2 + (t == 1 ? 1 : 0)], 1u);
+    }
+    __syncthreads();
+    for (int b = threadIdx.x; b < (T + 1) * 2; b += blockDim.x) {
+        unsigned int c = lhist[b];
+        if (c) atomicAdd(&hist[b], (unsigned long long)c);
+    }
+}
+
+// multiclass one-vs-rest: probs (B, C) -> hist (C, T+1, 2), global atomics,
+// coalesced row reads.
+// mode 0: multiclass one-vs-rest (target (B,), label = target[row]==c)
+// mode 1: multilabel (target (B,C), label = target[i])
+template <typename T_, bool IS_BF16>
+__global__ void __launch_bounds__(256) k_multiclass_curve_hist(
+    const T_* __restrict__ probs, const ll* __restrict__ target, ll B, ll C,
+    const float* __restrict__ thresholds, int T, ll ignore_index, int has_ignore, int mode,
+    unsigned long long* __restrict__ hist /* (C, T+1, 2) */) {
+    extern __shared__ float sthr2[];
+    for (int b = threadIdx.x; b < T; b += blockDim.x) sthr2[b] = thresholds[b];
+    __syncthreads();
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    ll stride = (ll)gridDim.x * blockDim.x;
+    for (; i < B * C; i += stride) {
+        ll row = i / C, c = i % C;
+        int label;
+        if (mode == 0) {
+            ll t = target[row];
+            if (has_ignore && t == ignore_index) continue;
+            label = (t == c) ? 1 : 0;
+        } else {
+            ll t = target[i];
+            if (has_ignore && t == ignore_index) continue;
+            label = (t == 1) ? 1 : 0;
+        }
+        float p = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(probs)[i])
+                          : (float)probs[i];
+        int j = bucket_of(p, sthr2, T);
+        atomicAdd(&hist[(c * (T + 1) + j) * 2 + label], 1ULL);
+    }
+}
+
+// suffix-sum the histogram into (/onto) the running confmat state:
+// state (..., T, 2, 2) += suffix(hist)(..., T, 2) expanded:
+//   confmat[t][1][1] += sum_{j>t} hist[j][1]   (tp)
+//   confmat[t][1][0] += sum_{j>t} hist[j][0]   (fp)
+//   confmat[t][0][1] += pos_total - tp         (fn)
+//   confmat[t][0][0] += neg_total - fp         (tn)
+// one block per outer index (class or 1), thread t in [0,T)
+__global__ void k_curve_suffix(
+    const unsigned long long* __restrict__ hist /* (O, T+1, 2) */, int T,
+    ll* __restrict__ confmat /* (O, T, 2, 2) */) {
+    const ll o = blockIdx.x;
+    const unsigned long long* h = hist + o * (ll)(T + 1) * 2;
+    ll* cm = confmat + o * (ll)T * 4;
+    __shared__ unsigned long long pos_total, neg_total;
+    if (threadIdx.x == 0) {
+        unsigned long long pt = 0, nt = 0;
+        for (int j = 0; j <= T; j++) { nt += h[j * 2 + 0]; pt += h[j * 2 + 1]; }
+        pos_total = pt; neg_total = nt;
+    }
+    __syncthreads();
+    for (int t = threadIdx.x; t < T; t += blockDim.x) {
+        unsigned long long tp = 0, fp = 0;
+        for (int j = t + 1; j <= T; j++) { fp += h[j * 2 + 0]; tp += h[j * 2 + 1]; }
+        // layout [t][target][pred] (reference: bins = 2*target + pred)
+        cm[t * 4 + 3] += (ll)tp;                    // [1][1] tp
+        cm[t * 4 + 1] += (ll)fp;                    // [0][1] fp
+        cm[t * 4 + 2] += (ll)(pos_total - tp);      // [1][0] fn
+        cm[t * 4 + 0] += (ll)(neg_total - fp);      // [0][0] tn
+    }
+}
+
+// ---------------------------------------------------------------------------
+// K13: fused elementwise-error reductions, deterministic fp64 two-pass.
+// op: 0 = squared error, 1 = abs error, 2 = abs percentage |d|/max(|t|,eps),
+//     3 = squared log error (log1p(x)-log1p(y))^2, 4 = moments
+//     (n. sum_x, sum_x2, sum_y, sum_y2, sum_xy -> 6 outputs), 5 = logcosh
+// partials: (num_blocks, n_out) f64; second kernel reduces in fixed order.
+// ---------------------------------------------------------------------------
+template <typename T, bool IS_BF16>
+__global__ void __launch_bounds__(256) k_err_reduce_partial(
+    const T* __restrict__ x, const T* __restrict__ y, ll N, int op, double eps,
+    double* __restrict__ partials, int n_out) {
+    __shared__ double sdata[256 * 2];  // up to 2 accumulators reduced at once per pass
+    double acc[6] = {0, 0, 0, 0, 0, 0};
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    ll stride = (ll)gridDim.x * blockDim.x;
+    for (; i < N; i += stride) {
+        float xf = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(x)[i]) : (float)x[i];
+        float yf = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(y)[i]) : (float)y[i];
+        double xd = xf, yd = yf;
+        switch (op) {
+            case 0: { double d = xd - yd; acc[0] += d * d; } break;
+            case 1: acc[0] += fabs(xd - yd); break;
+            case 2: acc[0] += fabs(xd - yd) / fmax(fabs(yd), eps); break;
+            case 3: { double d = log1p(xd) - log1p(yd); acc[0] += d * d; } break;
+            case 4:
+                acc[0] += xd; acc[1] += xd * xd; acc[2] += yd; acc[3] += yd * yd; acc[4] += xd * yd;
+                acc[5] += 1.0;
+                break;
+            case 5: { double d = xd - yd; acc[0] += d + log1p(exp(-2.0 * d)) - 0.6931471805599453; } break;
+        }
+    }
+    // block tree-reduce each accumulator
+    for (int a = 0; a < n_out; a++) {
+        sdata[threadIdx.x] = acc[a];
+        __syncthreads();
+        for (int s = blockDim.x / 2; s > 0; s >>= 1) {
+            if (threadIdx.x < s) sdata[threadIdx.x] += sdata[threadIdx.x + s];
+            __syncthreads();
+        }
+        if (threadIdx.x == 0) partials[(ll)blockIdx.x * n_out + a] = sdata[0];
+        __syncthreads();
+    }
+}
+
+__global__ void k_err_reduce_final(const double* __restrict__ partials, int num_blocks, int n_out,
+                                   double* __restrict__ out) {
+    // single block; fixed-order accumulation per output => deterministic
+    for (int a = threadIdx.x; a < n_out; a += blockDim.x) {
+        double s = 0;
+        for (int b = 0; b < num_blocks; b++) s += partials[(ll)b * n_out + a];
+        out[a] += s;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// K6/K7: all-pairs box IoU (xyxy) with GIoU/DIoU/CIoU epilogues.
+// variant: 0=iou 1=giou 2=diou 3=ciou. boxes1 tile staged in LDS.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256) k_box_iou(
+    const float* __restrict__ b1, ll N, const float* __restrict__ b2, ll M, int variant,
+    float* __restrict__ out /* (N, M) */) {
+    __shared__ float tile[64][4];
+    // grid: (ceil(M/256), ceil(N/64))
+    ll j = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    ll i0 = (ll)blockIdx.y * 64;
+    ll ilim = min((ll)64, N - i0);
+    for (int k = threadIdx.x; k < 64 * 4; k += blockDim.x) {
+        int r = k / 4;
+        if (i0 + r < N) tile[r][k % 4] = b1[(i0 + r) * 4 + (k % 4)];
+    }
+    __syncthreads();
+    if (j >= M) return;
+    float x1b = b2[j * 4 + 0], y1b = b2[j * 4 + 1], x2b = b2[j * 4 + 2], y2b = b2[j * 4 + 3];
+    float area_b = (x2b - x1b) * (y2b - y1b);
+    for (ll r = 0; r < ilim; r++) {
+        float x1a = tile[r][0], y1a = tile[r][1], x2a = tile[r][2], y2a = tile[r][3];
+        float area_a = (x2a - x1a) * (y2a - y1a);
+        float ix1 = fmaxf(x1a, x1b), iy1 = fmaxf(y1a, y1b);
+        float ix2 = fminf(x2a, x2b), iy2 = fminf(y2a, y2b);
+        float iw = fmaxf(ix2 - ix1, 0.0f), ih = fmaxf(iy2 - iy1, 0.0f);
+        float inter = iw * ih;
+        float uni = area_a + area_b - inter;
+        float iou = uni > 0.f ? inter / uni : 0.0f;
+        float res = iou;
+        if (variant >= 1) {
+            float cx1 = fminf(x1a, x1b), cy1 = fminf(y1a, y1b);
+            float cx2 = fmaxf(x2a, x2b), cy2 = fmaxf(y2a, y2b);
+            if (variant == 1) {  // GIoU
+                float carea = (cx2 - cx1) * (cy2 - cy1);
+                res = carea > 0.f ? iou - (carea - uni) / carea : iou;
+            } else {
+                float cw = cx2 - cx1, ch = cy2 - cy1;
+                float cdiag = cw * cw + ch * ch + 1e-7f;
+                float dx = (x1a + x2a - x1b - x2b) * 0.5f, dy = (y1a + y2a - y1b - y2b) * 0.5f;
+                float dist = dx * dx + dy * dy;
+                if (variant == 2) {  // DIoU
+                    res = iou - dist / cdiag;
+                } else {  // CIoU
+                    float wa = x2a - x1a, ha = y2a - y1a, wb = x2b - x1b, hb = y2b - y1b;
+                    float v = (4.0f / (M_PI * M_PI)) *
+                              powf(atanf(wb / (hb + 1e-7f)) - atanf(wa / (ha + 1e-7f)), 2.0f);
+                    float alpha = v / (1.0f - iou + v + 1e-7f);
+                    res = iou - dist / cdiag - alpha * v;
+                }
+            }
+        }
+        out[(i0 + r) * M + j] = res;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// extern "C" launchers
+// ---------------------------------------------------------------------------
+static inline int grid_for(ll n, int block) {
+    ll g = (n + block - 1) / block;
+    if (g > 2048) g = 2048;
+    if (g < 1) g = 1;
+    return (int)g;
+}
+
+extern "C" {
+
+int ma_mc_stat_logits(uintptr_t stream, uintptr_t preds, int dtype /*0=f32 1=bf16*/,
+                      uintptr_t target, ll B, ll C, ll ignore_index, int has_ignore, uintptr_t tp,
+                      uintptr_t fp, uintptr_t fn, uintptr_t confmat, uintptr_t valid_count,
+                      uintptr_t argmax_out) {
+    hipStream_t s = (hipStream_t)stream;
+    int waves_per_block = 4;
+    int grid = grid_for(B, waves_per_block);
+    if (dtype == 0)
+        k_mc_stat_logits<float, false><<<grid, 256, 0, s>>>(
+            (const float*)preds, (const ll*)target, B, C, ignore_index, has_ignore,
+            (unsigned long long*)tp, (unsigned long long*)fp, (unsigned long long*)fn,
+            (unsigned long long*)confmat, (unsigned long long*)valid_count, (ll*)argmax_out);
+    else
+        k_mc_stat_logits<unsigned short, true><<<grid, 256, 0, s>>>(
+            (const unsigned short*)preds, (const ll*)target, B, C, ignore_index, has_ignore,
+            (unsigned long long*)tp, (unsigned long long*)fp, (unsigned long long*)fn,
+            (unsigned long long*)confmat, (unsigned long long*)valid_count, (ll*)argmax_out);
+    return (int)hipGetLastError();
+}
+
+int ma_mc_stat_labels(uintptr_t stream, uintptr_t preds, uintptr_t target, ll N, ll C,
+                      ll ignore_index, int has_ignore, uintptr_t tp, uintptr_t fp, uintptr_t fn,
+                      uintptr_t confmat, uintptr_t valid_count) {
+    hipStream_t s = (hipStream_t)stream;
+    k_mc_stat_labels<<<grid_for(N, 256), 256, 0, s>>>(
+        (const ll*)preds, (const ll*)target, N, C, ignore_index, has_ignore,
+        (unsigned long long*)tp, (unsigned long long*)fp, (unsigned long long*)fn,
+        (unsigned long long*)confmat, (unsigned long long*)valid_count);
+    return (int)hipGetLastError();
+}
+
+int ma_bincount(uintptr_t stream, uintptr_t x, ll N, ll bins, uintptr_t out) {
+    hipStream_t s = (hipStream_t)stream;
+    if (bins * (ll)sizeof(unsigned int) <= 64 * 1024) {
+        k_bincount_lds<<<grid_for(N, 256), 256, bins * sizeof(unsigned int), s>>>(
+            (const ll*)x, N, bins, (unsigned long long*)out);
+    } else {
+        k_bincount_global<<<grid_for(N, 256), 256, 0, s>>>((const ll*)x, N, bins,
+                                                           (unsigned long long*)out);
+    }
+    return (int)hipGetLastError();
+}
+
+int ma_binary_stat(uintptr_t stream, uintptr_t preds, int dtype, uintptr_t target, ll N,
+                   float threshold, ll ignore_index, int has_ignore, uintptr_t out,
+                   uintptr_t outside_flag) {
+    hipStream_t s = (hipStream_t)stream;
+    if (dtype == 0)
+        k_binary_stat<float, false><<<grid_for(N, 256), 256, 0, s>>>(
+            (const float*)preds, (const ll*)target, N, threshold, ignore_index, has_ignore,
+            (unsigned long long*)out, (unsigned int*)outside_flag);
+    else
+        k_binary_stat<unsigned short, true><<<grid_for(N, 256), 256, 0, s>>>(
+            (const unsigned short*)preds, (const ll*)target, N, threshold, ignore_index, has_ignore,
+            (unsigned long long*)out, (unsigned int*)outside_flag);
+    return (int)hipGetLastError();
+}
+
+int ma_multilabel_stat(uintptr_t stream, uintptr_t preds, int dtype, uintptr_t target, ll N, ll L,
+                       float threshold, ll ignore_index, int has_ignore, uintptr_t out,
+                       uintptr_t outside_flag) {
+    hipStream_t s = (hipStream_t)stream;
+    if (dtype == 0)
+        k_multilabel_stat<float, false><<<grid_for(N * L, 256), 256, 0, s>>>(
+            (const float*)preds, (const ll*)target, N, L, threshold, ignore_index, has_ignore,
+            (unsigned long long*)out, (unsigned int*)outside_flag);
+    else
+        k_multilabel_stat<unsigned short, true><<<grid_for(N * L, 256), 256, 0, s>>>(
+            (const unsigned short*)preds, (const ll*)target, N, L, threshold, ignore_index,
+            has_ignore, (unsigned long long*)out, (unsigned int*)outside_flag);
+    return (int)hipGetLastError();
+}
+
+int ma_binary_curve_hist(uintptr_t stream, uintptr_t preds, int dtype, uintptr_t target, ll N,
+                         uintptr_t thresholds, int T, ll ignore_index, int has_ignore,
+                         uintptr_t hist) {
+    hipStream_t s = (hipStream_t)stream;
+    size_t shmem = (size_t)(T + 1) * 2 * sizeof(unsigned int) + (size_t)T * sizeof(float);
+    if (shmem > 160 * 1024) return -100;  // thresholds too large for LDS path
+    if (dtype == 0)
+        k_binary_curve_hist<float, false><<<grid_for(N, 256), 256, shmem, s>>>(
+            (const float*)preds, (const ll*)target, N, (const float*)thresholds, T, ignore_index,
+            has_ignore, (unsigned long long*)hist);
+    else
+        k_binary_curve_hist<unsigned short, true><<<grid_for(N, 256), 256, shmem, s>>>(
+            (const unsigned short*)preds, (const ll*)target, N, (const float*)thresholds, T,
+            ignore_index, has_ignore, (unsigned long long*)hist);
+    return (int)hipGetLastError();
+}
+
+int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintptr_t target, ll B,
+                             ll C, uintptr_t thresholds, int T, ll ignore_index, int has_ignore,
+                             int mode, uintptr_t hist) {
+    hipStream_t s = (hipStream_t)stream;
+    size_t shmem = (size_t)T * sizeof(float);
+    if (shmem > 160 * 1024) return -100;
+    if (dtype == 0)
+        k_multiclass_curve_hist<float, false><<<grid_for(B * C, 256), 256, shmem, s>>>(
+            (const float*)probs, (const ll*)target, B, C, (const float*)thresholds, T, ignore_index,
+            has_ignore, mode, (unsigned long long*)hist);
+    else
+        k_multiclass_curve_hist<unsigned short, true><<<grid_for(B * C, 256), 256, shmem, s>>>(
+            (const unsigned short*)probs, (const ll*)target, B, C, (const float*)thresholds, T,
+            ignore_index, has_ignore, mode, (unsigned long long*)hist);
+    return (int)hipGetLastError();
+}
+
+int ma_curve_suffix(uintptr_t stream, uintptr_t hist, ll outer, int T, uintptr_t confmat) {
+    hipStream_t s = (hipStream_t)stream;
+    k_curve_suffix<<<(int)outer, 256, 0, s>>>((const unsigned long long*)hist, T, (ll*)confmat);
+    return (int)hipGetLastError();
+}
+
+int ma_err_reduce(uintptr_t stream, uintptr_t x, uintptr_t y, int dtype, ll N, int op, double eps,
+                  uintptr_t partials, int num_blocks, int n_out, uintptr_t out) {
+    hipStream_t s = (hipStream_t)stream;
+    if (dtype == 0)
+        k_err_reduce_partial<float, false><<<num_blocks, 256, 0, s>>>(
+            (const float*)x, (const float*)y, N, op, eps, (double*)partials, n_out);
+    else
+        k_err_reduce_partial<unsigned short, true><<<num_blocks, 256, 0, s>>>(
+            (const unsigned short*)x, (const unsigned short*)y, N, op, eps, (double*)partials,
+            n_out);
+    k_err_reduce_final<<<1, 64, 0, s>>>((const double*)partials, num_blocks, n_out, (double*)out);
+    return (int)hipGetLastError();
+}
+
+int ma_box_iou(uintptr_t stream, uintptr_t b1, ll N, uintptr_t b2, ll M, int variant,
+               uintptr_t out) {
+    hipStream_t s = (hipStream_t)stream;
+    dim3 grid((unsigned)((M + 255) / 256), (unsigned)((N + 63) / 64));
+    k_box_iou<<<grid, 256, 0, s>>>((const float*)b1, N, (const float*)b2, M, variant, (float*)out);
+    return (int)hipGetLastError();
+}
+
+int ma_device_sync() { return (int)hipDeviceSynchronize(); }
+
+}  // extern "C"

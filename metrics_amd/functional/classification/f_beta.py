@@ -1,0 +1,231 @@
+"""F-beta / F1. Parity: torchmetrics ``functional/classification/f_beta.py``."""
+from __future__ import annotations
+
+from typing import Optional
+
+from torch import Tensor
+
+from metrics_amd.utilities.compute import _adjust_weights_safe_divide, _safe_divide
+from metrics_amd.utilities.enums import ClassificationTask
+from metrics_amd.functional.classification.stat_scores import (
+    _binary_stat_scores_arg_validation,
+    _binary_stat_scores_pipeline,
+    _binary_stat_scores_tensor_validation,
+    _multiclass_stat_scores_arg_validation,
+    _multiclass_stat_scores_pipeline,
+    _multiclass_stat_scores_tensor_validation,
+    _multilabel_stat_scores_arg_validation,
+    _multilabel_stat_scores_pipeline,
+    _multilabel_stat_scores_tensor_validation,
+)
+
+
+def _fbeta_reduce(
+    tp: Tensor,
+    fp: Tensor,
+    tn: Tensor,
+    fn: Tensor,
+    beta: float,
+    average: Optional[str],
+    multidim_average: str = "global",
+    multilabel: bool = False,
+    zero_division: float = 0,
+) -> Tensor:
+    beta2 = beta**2
+    if average == "binary":
+        return _safe_divide((1 + beta2) * tp, (1 + beta2) * tp + beta2 * fn + fp, zero_division)
+    if average == "micro":
+        tp = tp.sum(dim=0 if multidim_average == "global" else 1)
+        fn = fn.sum(dim=0 if multidim_average == "global" else 1)
+        fp = fp.sum(dim=0 if multidim_average == "global" else 1)
+        return _safe_divide((1 + beta2) * tp, (1 + beta2) * tp + beta2 * fn + fp, zero_division)
+
+    fbeta_score = _safe_divide((1 + beta2) * tp, (1 + beta2) * tp + beta2 * fn + fp, zero_division)
+    return _adjust_weights_safe_divide(fbeta_score, average, multilabel, tp, fp, fn)
+
+
+def _fbeta_arg_check(beta: float) -> None:
+    if not (isinstance(beta, float) and beta > 0):
+        raise ValueError(f"Expected argument `beta` to be a float larger than 0, but got {beta}.")
+
+
+def binary_fbeta_score(
+    preds: Tensor,
+    target: Tensor,
+    beta: float,
+    threshold: float = 0.5,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+    zero_division: float = 0,
+) -> Tensor:
+    """F-beta for binary tasks."""
+    if validate_args:
+        _fbeta_arg_check(beta)
+        _binary_stat_scores_arg_validation(threshold, multidim_average, ignore_index, zero_division)
+        _binary_stat_scores_tensor_validation(preds, target, multidim_average, ignore_index)
+    tp, fp, tn, fn = _binary_stat_scores_pipeline(preds, target, threshold, multidim_average, ignore_index)
+    return _fbeta_reduce(
+        tp, fp, tn, fn, beta, average="binary", multidim_average=multidim_average, zero_division=zero_division
+    )
+
+
+def multiclass_fbeta_score(
+    preds: Tensor,
+    target: Tensor,
+    beta: float,
+    num_classes: int,
+    average: Optional[str] = "macro",
+    top_k: int = 1,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+    zero_division: float = 0,
+) -> Tensor:
+    """F-beta for multiclass tasks."""
+    if validate_args:
+        _fbeta_arg_check(beta)
+        _multiclass_stat_scores_arg_validation(num_classes, top_k, average, multidim_average, ignore_index, zero_division)
+        _multiclass_stat_scores_tensor_validation(preds, target, num_classes, multidim_average, ignore_index)
+    tp, fp, tn, fn = _multiclass_stat_scores_pipeline(
+        preds, target, num_classes, top_k, average, multidim_average, ignore_index
+    )
+    return _fbeta_reduce(
+        tp, fp, tn, fn, beta, average=average, multidim_average=multidim_average, zero_division=zero_division
+    )
+
+
+def multilabel_fbeta_score(
+    preds: Tensor,
+    target: Tensor,
+    beta: float,
+    num_labels: int,
+    threshold: float = 0.5,
+    average: Optional[str] = "macro",
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+    zero_division: float = 0,
+) -> Tensor:
+    """F-beta for multilabel tasks."""
+    if validate_args:
+        _fbeta_arg_check(beta)
+        _multilabel_stat_scores_arg_validation(num_labels, threshold, average, multidim_average, ignore_index, zero_division)
+        _multilabel_stat_scores_tensor_validation(preds, target, num_labels, multidim_average, ignore_index)
+    tp, fp, tn, fn = _multilabel_stat_scores_pipeline(
+        preds, target, num_labels, threshold, multidim_average, ignore_index
+    )
+    return _fbeta_reduce(
+        tp, fp, tn, fn, beta, average=average, multidim_average=multidim_average, multilabel=True,
+        zero_division=zero_division,
+    )
+
+
+def binary_f1_score(
+    preds: Tensor,
+    target: Tensor,
+    threshold: float = 0.5,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+    zero_division: float = 0,
+) -> Tensor:
+    """F1 for binary tasks."""
+    return binary_fbeta_score(
+        preds, target, 1.0, threshold, multidim_average, ignore_index, validate_args, zero_division
+    )
+
+
+def multiclass_f1_score(
+    preds: Tensor,
+    target: Tensor,
+    num_classes: int,
+    average: Optional[str] = "macro",
+    top_k: int = 1,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+    zero_division: float = 0,
+) -> Tensor:
+    """F1 for multiclass tasks."""
+    return multiclass_fbeta_score(
+        preds, target, 1.0, num_classes, average, top_k, multidim_average, ignore_index, validate_args, zero_division
+    )
+
+
+def multilabel_f1_score(
+    preds: Tensor,
+    target: Tensor,
+    num_labels: int,
+    threshold: float = 0.5,
+    average: Optional[str] = "macro",
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+    zero_division: float = 0,
+) -> Tensor:
+    """F1 for multilabel tasks."""
+    return multilabel_fbeta_score(
+        preds, target, 1.0, num_labels, threshold, average, multidim_average, ignore_index, validate_args, zero_division
+    )
+
+
+def fbeta_score(
+    preds: Tensor,
+    target: Tensor,
+    task: str,
+    beta: float = 1.0,
+    threshold: float = 0.5,
+    num_classes: Optional[int] = None,
+    num_labels: Optional[int] = None,
+    average: Optional[str] = "micro",
+    multidim_average: str = "global",
+    top_k: int = 1,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+    zero_division: float = 0,
+) -> Tensor:
+    """Task-dispatching F-beta."""
+    task = ClassificationTask.from_str(task)
+    if task == ClassificationTask.BINARY:
+        return binary_fbeta_score(
+            preds, target, beta, threshold, multidim_average, ignore_index, validate_args, zero_division
+        )
+    if task == ClassificationTask.MULTICLASS:
+        if not isinstance(num_classes, int):
+            raise ValueError(f"`num_classes` is expected to be `int` but `{type(num_classes)} was passed.`")
+        if not isinstance(top_k, int):
+            raise ValueError(f"`top_k` is expected to be `int` but `{type(top_k)} was passed.`")
+        return multiclass_fbeta_score(
+            preds, target, beta, num_classes, average, top_k, multidim_average, ignore_index, validate_args,
+            zero_division,
+        )
+    if task == ClassificationTask.MULTILABEL:
+        if not isinstance(num_labels, int):
+            raise ValueError(f"`num_labels` is expected to be `int` but `{type(num_labels)} was passed.`")
+        return multilabel_fbeta_score(
+            preds, target, beta, num_labels, threshold, average, multidim_average, ignore_index, validate_args,
+            zero_division,
+        )
+    raise ValueError(f"Not handled value: {task}")
+
+
+def f1_score(
+    preds: Tensor,
+    target: Tensor,
+    task: str,
+    threshold: float = 0.5,
+    num_classes: Optional[int] = None,
+    num_labels: Optional[int] = None,
+    average: Optional[str] = "micro",
+    multidim_average: str = "global",
+    top_k: int = 1,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+    zero_division: float = 0,
+) -> Tensor:
+    """Task-dispatching F1."""
+    return fbeta_score(
+        preds, target, task, 1.0, threshold, num_classes, num_labels, average, multidim_average, top_k, ignore_index,
+        validate_args, zero_division,
+    )

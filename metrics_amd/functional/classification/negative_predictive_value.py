@@ -1,0 +1,131 @@
+"""Negative predictive value. Parity: torchmetrics ``functional/classification/negative_predictive_value.py``."""
+from __future__ import annotations
+
+from typing import Optional
+
+from torch import Tensor
+
+from metrics_amd.utilities.compute import _adjust_weights_safe_divide, _safe_divide
+from metrics_amd.utilities.enums import ClassificationTask
+from metrics_amd.functional.classification.stat_scores import (
+    _binary_stat_scores_arg_validation,
+    _binary_stat_scores_pipeline,
+    _binary_stat_scores_tensor_validation,
+    _multiclass_stat_scores_arg_validation,
+    _multiclass_stat_scores_pipeline,
+    _multiclass_stat_scores_tensor_validation,
+    _multilabel_stat_scores_arg_validation,
+    _multilabel_stat_scores_pipeline,
+    _multilabel_stat_scores_tensor_validation,
+)
+
+
+def _negative_predictive_value_reduce(
+    tp: Tensor,
+    fp: Tensor,
+    tn: Tensor,
+    fn: Tensor,
+    average: Optional[str],
+    multidim_average: str = "global",
+    multilabel: bool = False,
+) -> Tensor:
+    if average == "binary":
+        return _safe_divide(tn, tn + fn)
+    if average == "micro":
+        tn = tn.sum(dim=0 if multidim_average == "global" else 1)
+        fn = fn.sum(dim=0 if multidim_average == "global" else 1)
+        return _safe_divide(tn, tn + fn)
+    score = _safe_divide(tn, tn + fn)
+    return _adjust_weights_safe_divide(score, average, multilabel, tp, fp, fn)
+
+
+def binary_negative_predictive_value(
+    preds: Tensor,
+    target: Tensor,
+    threshold: float = 0.5,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tensor:
+    """NPV for binary tasks."""
+    if validate_args:
+        _binary_stat_scores_arg_validation(threshold, multidim_average, ignore_index)
+        _binary_stat_scores_tensor_validation(preds, target, multidim_average, ignore_index)
+    tp, fp, tn, fn = _binary_stat_scores_pipeline(preds, target, threshold, multidim_average, ignore_index)
+    return _negative_predictive_value_reduce(tp, fp, tn, fn, average="binary", multidim_average=multidim_average)
+
+
+def multiclass_negative_predictive_value(
+    preds: Tensor,
+    target: Tensor,
+    num_classes: int,
+    average: Optional[str] = "macro",
+    top_k: int = 1,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tensor:
+    """NPV for multiclass tasks."""
+    if validate_args:
+        _multiclass_stat_scores_arg_validation(num_classes, top_k, average, multidim_average, ignore_index)
+        _multiclass_stat_scores_tensor_validation(preds, target, num_classes, multidim_average, ignore_index)
+    tp, fp, tn, fn = _multiclass_stat_scores_pipeline(
+        preds, target, num_classes, top_k, average, multidim_average, ignore_index
+    )
+    return _negative_predictive_value_reduce(tp, fp, tn, fn, average=average, multidim_average=multidim_average)
+
+
+def multilabel_negative_predictive_value(
+    preds: Tensor,
+    target: Tensor,
+    num_labels: int,
+    threshold: float = 0.5,
+    average: Optional[str] = "macro",
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tensor:
+    """NPV for multilabel tasks."""
+    if validate_args:
+        _multilabel_stat_scores_arg_validation(num_labels, threshold, average, multidim_average, ignore_index)
+        _multilabel_stat_scores_tensor_validation(preds, target, num_labels, multidim_average, ignore_index)
+    tp, fp, tn, fn = _multilabel_stat_scores_pipeline(
+        preds, target, num_labels, threshold, multidim_average, ignore_index
+    )
+    return _negative_predictive_value_reduce(
+        tp, fp, tn, fn, average=average, multidim_average=multidim_average, multilabel=True
+    )
+
+
+def negative_predictive_value(
+    preds: Tensor,
+    target: Tensor,
+    task: str,
+    threshold: float = 0.5,
+    num_classes: Optional[int] = None,
+    num_labels: Optional[int] = None,
+    average: Optional[str] = "micro",
+    multidim_average: str = "global",
+    top_k: int = 1,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tensor:
+    """Task-dispatching negative predictive value."""
+    task = ClassificationTask.from_str(task)
+    if task == ClassificationTask.BINARY:
+        return binary_negative_predictive_value(
+            preds, target, threshold, multidim_average, ignore_index, validate_args
+        )
+    if task == ClassificationTask.MULTICLASS:
+        if not isinstance(num_classes, int):
+            raise ValueError(f"`num_classes` is expected to be `int` but `{type(num_classes)} was passed.`")
+        return multiclass_negative_predictive_value(
+            preds, target, num_classes, average, top_k, multidim_average, ignore_index, validate_args
+        )
+    if task == ClassificationTask.MULTILABEL:
+        if not isinstance(num_labels, int):
+            raise ValueError(f"`num_labels` is expected to be `int` but `{type(num_labels)} was passed.`")
+        return multilabel_negative_predictive_value(
+            preds, target, num_labels, threshold, average, multidim_average, ignore_index, validate_args
+        )
+    raise ValueError(f"Not handled value: {task}")

@@ -1,0 +1,645 @@
+"""Stat scores (tp/fp/tn/fn) — the foundation of the classification domain.
+
+Parity: torchmetrics ``functional/classification/stat_scores.py`` (the
+``_<task>_stat_scores_{arg_validation,tensor_validation,format,update,compute}``
+decomposition is kept — the modular layer reuses each stage).
+
+MI355X path: on GPU tensors the update stage routes to the fused HIP kernels
+(csrc/kernels.hip): one pass computing argmax + per-class counts for
+multiclass, and single-pass fused counters for binary/multilabel — replacing
+the reference's argmax -> bincount -> reshape -> diag/rowsum chain (5+ kernel
+launches and 3 materialized intermediates).
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+from torch import Tensor
+
+from metrics_amd import ops
+from metrics_amd.utilities.checks import _check_same_shape
+from metrics_amd.utilities.compute import normalize_logits_if_needed
+from metrics_amd.utilities.data import _bincount, select_topk
+from metrics_amd.utilities.enums import ClassificationTask
+
+
+def _binary_stat_scores_arg_validation(
+    threshold: float = 0.5,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    zero_division: float = 0,
+) -> None:
+    if not (isinstance(threshold, float) and (0 <= threshold <= 1)):
+        raise ValueError(f"Expected argument `threshold` to be a float in the [0,1] range, but got {threshold}.")
+    allowed_multidim_average = ("global", "samplewise")
+    if multidim_average not in allowed_multidim_average:
+        raise ValueError(
+            f"Expected argument `multidim_average` to be one of {allowed_multidim_average}, but got {multidim_average}"
+        )
+    if ignore_index is not None and not isinstance(ignore_index, int):
+        raise ValueError(f"Expected argument `ignore_index` to either be `None` or an integer, but got {ignore_index}")
+    if zero_division not in (0, 1):
+        raise ValueError(f"Expected argument `zero_division` to be 0 or 1, but got {zero_division}")
+
+
+def _binary_stat_scores_tensor_validation(
+    preds: Tensor,
+    target: Tensor,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+) -> None:
+    _check_same_shape(preds, target)
+    unique_values = torch.unique(target, dim=None)
+    if ignore_index is None:
+        check = torch.any((unique_values != 0) & (unique_values != 1))
+    else:
+        check = torch.any((unique_values != 0) & (unique_values != 1) & (unique_values != ignore_index))
+    if check:
+        raise RuntimeError(
+            f"Detected the following values in `target`: {unique_values} but expected only"
+            f" the following values {[0, 1] if ignore_index is None else [ignore_index]}."
+        )
+    if not preds.is_floating_point():
+        unique_values = torch.unique(preds, dim=None)
+        if torch.any((unique_values != 0) & (unique_values != 1)):
+            raise RuntimeError(
+                f"Detected the following values in `preds`: {unique_values} but expected only"
+                " the following values [0,1] since `preds` is a label tensor."
+            )
+    if multidim_average != "global" and preds.ndim < 2:
+        raise ValueError("Expected input to be at least 2D when multidim_average is set to `samplewise`")
+
+
+def _binary_stat_scores_format(
+    preds: Tensor,
+    target: Tensor,
+    threshold: float = 0.5,
+    ignore_index: Optional[int] = None,
+) -> Tuple[Tensor, Tensor]:
+    """Normalize, threshold and reshape to (N, -1); ignored positions get target -1."""
+    if preds.is_floating_point():
+        preds = normalize_logits_if_needed(preds, "sigmoid")
+        preds = preds > threshold
+    preds = preds.reshape(preds.shape[0], -1)
+    target = target.reshape(target.shape[0], -1)
+
+    if ignore_index is not None:
+        idx = target == ignore_index
+        target = target.clone()
+        target[idx] = -1
+    return preds, target
+
+
+def _binary_stat_scores_update(
+    preds: Tensor,
+    target: Tensor,
+    multidim_average: str = "global",
+) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
+    sum_dim = [0, 1] if multidim_average == "global" else [1]
+    tp = ((target == preds) & (target == 1)).sum(sum_dim).squeeze()
+    fn = ((target != preds) & (target == 1)).sum(sum_dim).squeeze()
+    fp = ((target != preds) & (target == 0)).sum(sum_dim).squeeze()
+    tn = ((target == preds) & (target == 0)).sum(sum_dim).squeeze()
+    return tp, fp, tn, fn
+
+
+def _binary_stat_scores_update_fused(
+    preds: Tensor,
+    target: Tensor,
+    threshold: float,
+    multidim_average: str,
+    ignore_index: Optional[int],
+) -> Optional[Tuple[Tensor, Tensor, Tensor, Tensor]]:
+    """Single-pass HIP path for the global-average float-preds case; None if not applicable."""
+    if (
+        preds.is_cuda
+        and preds.is_floating_point()
+        and preds.dtype in (torch.float32, torch.bfloat16)
+        and multidim_average == "global"
+    ):
+        return ops.binary_stat_scores_fused(preds, target, threshold, ignore_index)
+    return None
+
+
+def _binary_stat_scores_compute(
+    tp: Tensor, fp: Tensor, tn: Tensor, fn: Tensor, multidim_average: str = "global"
+) -> Tensor:
+    return torch.stack([tp, fp, tn, fn, tp + fn], dim=0 if multidim_average == "global" else 1).squeeze()
+
+
+def binary_stat_scores(
+    preds: Tensor,
+    target: Tensor,
+    threshold: float = 0.5,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tensor:
+    """Number of tp/fp/tn/fn + support for binary tasks. Returns (5,) or (N,5)."""
+    if validate_args:
+        _binary_stat_scores_arg_validation(threshold, multidim_average, ignore_index)
+        _binary_stat_scores_tensor_validation(preds, target, multidim_average, ignore_index)
+    fused = _binary_stat_scores_update_fused(preds, target, threshold, multidim_average, ignore_index)
+    if fused is not None:
+        tp, fp, tn, fn = fused
+    else:
+        preds, target = _binary_stat_scores_format(preds, target, threshold, ignore_index)
+        tp, fp, tn, fn = _binary_stat_scores_update(preds, target, multidim_average)
+    return _binary_stat_scores_compute(tp, fp, tn, fn, multidim_average)
+
+
+def _multiclass_stat_scores_arg_validation(
+    num_classes: int,
+    top_k: int = 1,
+    average: Optional[str] = "macro",
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    zero_division: float = 0,
+) -> None:
+    if not isinstance(num_classes, int) or num_classes < 2:
+        raise ValueError(f"Expected argument `num_classes` to be an integer larger than 1, but got {num_classes}")
+    if not isinstance(top_k, int) and top_k < 1:
+        raise ValueError(f"Expected argument `top_k` to be an integer larger than or equal to 1, but got {top_k}")
+    if top_k > num_classes:
+        raise ValueError(
+            f"Expected argument `top_k` to be smaller or equal to `num_classes` but got {top_k} and {num_classes}"
+        )
+    allowed_average = ("micro", "macro", "weighted", "none", None)
+    if average not in allowed_average:
+        raise ValueError(f"Expected argument `average` to be one of {allowed_average}, but got {average}")
+    allowed_multidim_average = ("global", "samplewise")
+    if multidim_average not in allowed_multidim_average:
+        raise ValueError(
+            f"Expected argument `multidim_average` to be one of {allowed_multidim_average}, but got {multidim_average}"
+        )
+    if ignore_index is not None and not isinstance(ignore_index, int):
+        raise ValueError(f"Expected argument `ignore_index` to either be `None` or an integer, but got {ignore_index}")
+    if zero_division not in (0, 1):
+        raise ValueError(f"Expected argument `zero_division` to be 0 or 1, but got {zero_division}")
+
+
+def _multiclass_stat_scores_tensor_validation(
+    preds: Tensor,
+    target: Tensor,
+    num_classes: Optional[int],
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+) -> None:
+    if preds.ndim == target.ndim + 1:
+        if not preds.is_floating_point():
+            raise ValueError("If `preds` have one dimension more than `target`, `preds` should be a float tensor.")
+        if num_classes is not None and preds.shape[1] != num_classes:
+            raise ValueError(
+                "If `preds` have one dimension more than `target`, `preds.shape[1]` should be"
+                " equal to number of classes."
+            )
+        if preds.shape[2:] != target.shape[1:]:
+            raise ValueError(
+                "If `preds` have one dimension more than `target`, the shape of `preds` should be"
+                " (N, C, ...), and the shape of `target` should be (N, ...)."
+            )
+        if multidim_average != "global" and preds.ndim < 3:
+            raise ValueError(
+                "If `preds` have one dimension more than `target`, the shape of `preds` should be"
+                " at least 3D when multidim_average is set to `samplewise`"
+            )
+    elif preds.ndim == target.ndim:
+        if preds.shape != target.shape:
+            raise ValueError(
+                "The `preds` and `target` should have the same shape,"
+                f" got `preds` with shape={preds.shape} and `target` with shape={target.shape}."
+            )
+        if multidim_average != "global" and preds.ndim < 2:
+            raise ValueError(
+                "When `preds` and `target` have the same shape, the shape of `preds` should be"
+                " at least 2D when multidim_average is set to `samplewise`"
+            )
+    else:
+        raise ValueError(
+            "Either `preds` and `target` both should have the (same) shape (N, ...), or `target` should be (N, ...)"
+            " and `preds` should be (N, C, ...)."
+        )
+
+    if num_classes is not None:
+        check_value = num_classes if ignore_index is None else num_classes + 1
+        to_check = [(target, "target")]
+        if not preds.is_floating_point():
+            to_check.append((preds, "preds"))
+        for t, name in to_check:
+            unique_values = torch.unique(t, dim=None)
+            if len(unique_values) > check_value:
+                raise RuntimeError(
+                    f"Detected more unique values in `{name}` than expected. Expected only {check_value} but found"
+                    f" {len(unique_values)} in `{name}`. Found values: {unique_values}."
+                )
+
+
+def _mc_fused_eligible(preds: Tensor, target: Tensor, top_k: int, multidim_average: str) -> bool:
+    """True when the single-pass HIP kernel (in-kernel argmax) will handle the update."""
+    return (
+        preds.is_cuda
+        and top_k == 1
+        and multidim_average == "global"
+        and preds.is_floating_point()
+        and preds.ndim == target.ndim + 1
+        and preds.dtype in (torch.float32, torch.bfloat16)
+    )
+
+
+def _multiclass_stat_scores_format(
+    preds: Tensor,
+    target: Tensor,
+    top_k: int = 1,
+    keep_logits: bool = False,
+) -> Tuple[Tensor, Tensor]:
+    """Argmax float preds (top_k == 1) and flatten trailing dims.
+
+    ``keep_logits`` (GPU fused path): leave preds as (N, C, -1) float so the
+    HIP kernel performs the argmax in the same pass as the counting.
+    """
+    if preds.ndim == target.ndim + 1 and top_k == 1 and not keep_logits:
+        preds = preds.argmax(dim=1)
+    if top_k != 1 or (keep_logits and preds.ndim == target.ndim + 1):
+        preds = preds.reshape(*preds.shape[:2], -1)
+    else:
+        preds = preds.reshape(preds.shape[0], -1)
+    target = target.reshape(target.shape[0], -1)
+    return preds, target
+
+
+def _refine_preds_oh(preds: Tensor, preds_oh: Tensor, target: Tensor, top_k: int) -> Tensor:
+    """If the target is inside the top-k predictions, credit the target; else the top-1."""
+    preds = preds.squeeze()
+    target = target.squeeze()
+    top_k_indices = torch.topk(preds, k=top_k, dim=1).indices
+    top_1_indices = top_k_indices[:, 0]
+    target_in_topk = torch.any(top_k_indices == target.unsqueeze(1), dim=1)
+    result = torch.where(target_in_topk, target, top_1_indices)
+    return torch.zeros_like(preds_oh, dtype=torch.int32).scatter_(-1, result.unsqueeze(1).unsqueeze(1), 1)
+
+
+def _multiclass_stat_scores_update(
+    preds: Tensor,
+    target: Tensor,
+    num_classes: int,
+    top_k: int = 1,
+    average: Optional[str] = "macro",
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
+    """Compute tp/fp/tn/fn.
+
+    GPU + global + top_k==1: single fused HIP kernel (argmax in-kernel when
+    preds are (N,C[,...]) float). Otherwise the one-hot / bincount torch paths.
+    """
+    if multidim_average == "samplewise" or top_k != 1:
+        ignore_in = 0 <= ignore_index <= num_classes - 1 if ignore_index is not None else None
+        if ignore_index is not None and not ignore_in:
+            preds = preds.clone()
+            target = target.clone()
+            idx = target == ignore_index
+            target[idx] = num_classes
+            idx = idx.unsqueeze(1).repeat(1, num_classes, 1) if preds.ndim > target.ndim else idx
+            preds[idx] = num_classes
+
+        if top_k > 1:
+            preds_oh = torch.movedim(select_topk(preds, topk=top_k, dim=1), 1, -1)
+            preds_oh = _refine_preds_oh(preds, preds_oh, target, top_k)
+        else:
+            preds_oh = torch.nn.functional.one_hot(
+                preds.long(), num_classes + 1 if ignore_index is not None and not ignore_in else num_classes
+            )
+        target_oh = torch.nn.functional.one_hot(
+            target.long(), num_classes + 1 if ignore_index is not None and not ignore_in else num_classes
+        )
+        if ignore_index is not None:
+            if 0 <= ignore_index <= num_classes - 1:
+                target_oh[target == ignore_index, :] = -1
+            else:
+                preds_oh = preds_oh[..., :-1] if top_k == 1 else preds_oh
+                target_oh = target_oh[..., :-1]
+                target_oh[target == num_classes, :] = -1
+        sum_dim = [0, 1] if multidim_average == "global" else [1]
+        tp = ((target_oh == preds_oh) & (target_oh == 1)).sum(sum_dim)
+        fn = ((target_oh != preds_oh) & (target_oh == 1)).sum(sum_dim)
+        fp = ((target_oh != preds_oh) & (target_oh == 0)).sum(sum_dim)
+        tn = ((target_oh == preds_oh) & (target_oh == 0)).sum(sum_dim)
+        return tp, fp, tn, fn
+
+    # ---- global & top_k == 1 ----
+    if preds.is_cuda and (not preds.is_floating_point() or preds.dtype in (torch.float32, torch.bfloat16)):
+        # fused HIP kernel: handles both (N, C, X) float (in-kernel argmax) and label preds
+        if preds.ndim == target.ndim + 1 and preds.is_floating_point():
+            # collapse trailing dims into batch: (N, C, X) -> (N*X, C)
+            p2 = preds.reshape(preds.shape[0], preds.shape[1], -1).movedim(1, -1).reshape(-1, preds.shape[1])
+            t2 = target.reshape(-1)
+        else:
+            p2 = preds.reshape(-1)
+            t2 = target.reshape(-1)
+        tp, fp, tn, fn, _ = ops.multiclass_stat_scores_fused(p2, t2, num_classes, ignore_index, want_confmat=False)
+        if average == "micro":
+            return tp.sum(), fp.sum(), tn.sum(), fn.sum()
+        return tp, fp, tn, fn
+
+    preds = preds.flatten()
+    target = target.flatten()
+    if ignore_index is not None:
+        idx = target != ignore_index
+        preds = preds[idx]
+        target = target[idx]
+    if average == "micro":
+        tp = (preds == target).sum()
+        fp = (preds != target).sum()
+        fn = (preds != target).sum()
+        tn = num_classes * preds.numel() - (fp + fn + tp)
+        return tp, fp, tn, fn
+    unique_mapping = target.to(torch.long) * num_classes + preds.to(torch.long)
+    bins = _bincount(unique_mapping, minlength=num_classes**2)
+    confmat = bins.reshape(num_classes, num_classes)
+    tp = confmat.diag()
+    fp = confmat.sum(0) - tp
+    fn = confmat.sum(1) - tp
+    tn = confmat.sum() - (fp + fn + tp)
+    return tp, fp, tn, fn
+
+
+def _multiclass_stat_scores_compute(
+    tp: Tensor,
+    fp: Tensor,
+    tn: Tensor,
+    fn: Tensor,
+    average: Optional[str] = "macro",
+    multidim_average: str = "global",
+) -> Tensor:
+    res = torch.stack([tp, fp, tn, fn, tp + fn], dim=-1)
+    sum_dim = 0 if multidim_average == "global" else 1
+    if average == "micro":
+        return res.sum(sum_dim) if res.ndim > 1 else res
+    if average == "macro":
+        return res.float().mean(sum_dim)
+    if average == "weighted":
+        weight = tp + fn
+        if multidim_average == "global":
+            return (res * (weight / weight.sum()).reshape(*weight.shape, 1)).sum(sum_dim)
+        return (res * (weight / weight.sum(-1, keepdim=True)).reshape(*weight.shape, 1)).sum(sum_dim)
+    if average is None or average == "none":
+        return res
+    return None
+
+
+def multiclass_stat_scores(
+    preds: Tensor,
+    target: Tensor,
+    num_classes: int,
+    average: Optional[str] = "macro",
+    top_k: int = 1,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tensor:
+    """Number of tp/fp/tn/fn + support for multiclass tasks."""
+    if validate_args:
+        _multiclass_stat_scores_arg_validation(num_classes, top_k, average, multidim_average, ignore_index)
+        _multiclass_stat_scores_tensor_validation(preds, target, num_classes, multidim_average, ignore_index)
+    preds, target = _multiclass_stat_scores_format(
+        preds, target, top_k, keep_logits=_mc_fused_eligible(preds, target, top_k, multidim_average)
+    )
+    tp, fp, tn, fn = _multiclass_stat_scores_update(
+        preds, target, num_classes, top_k, average, multidim_average, ignore_index
+    )
+    return _multiclass_stat_scores_compute(tp, fp, tn, fn, average, multidim_average)
+
+
+def _multilabel_stat_scores_arg_validation(
+    num_labels: int,
+    threshold: float = 0.5,
+    average: Optional[str] = "macro",
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    zero_division: float = 0,
+) -> None:
+    if not isinstance(num_labels, int) or num_labels < 2:
+        raise ValueError(f"Expected argument `num_labels` to be an integer larger than 1, but got {num_labels}")
+    if not (isinstance(threshold, float) and (0 <= threshold <= 1)):
+        raise ValueError(f"Expected argument `threshold` to be a float, but got {threshold}.")
+    allowed_average = ("micro", "macro", "weighted", "none", None)
+    if average not in allowed_average:
+        raise ValueError(f"Expected argument `average` to be one of {allowed_average}, but got {average}")
+    allowed_multidim_average = ("global", "samplewise")
+    if multidim_average not in allowed_multidim_average:
+        raise ValueError(
+            f"Expected argument `multidim_average` to be one of {allowed_multidim_average}, but got {multidim_average}"
+        )
+    if ignore_index is not None and not isinstance(ignore_index, int):
+        raise ValueError(f"Expected argument `ignore_index` to either be `None` or an integer, but got {ignore_index}")
+    if zero_division not in (0, 1):
+        raise ValueError(f"Expected argument `zero_division` to be 0 or 1, but got {zero_division}")
+
+
+def _multilabel_stat_scores_tensor_validation(
+    preds: Tensor,
+    target: Tensor,
+    num_labels: int,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+) -> None:
+    _check_same_shape(preds, target)
+    if preds.shape[1] != num_labels:
+        raise ValueError(
+            "Expected both `target.shape[1]` and `preds.shape[1]` to be equal to the number of labels, but got"
+            f" {preds.shape[1]} and expected {num_labels}"
+        )
+    unique_values = torch.unique(target, dim=None)
+    if ignore_index is None:
+        check = torch.any((unique_values != 0) & (unique_values != 1))
+    else:
+        check = torch.any((unique_values != 0) & (unique_values != 1) & (unique_values != ignore_index))
+    if check:
+        raise RuntimeError(
+            f"Detected the following values in `target`: {unique_values} but expected only"
+            f" the following values {[0, 1] if ignore_index is None else [ignore_index]}."
+        )
+    if not preds.is_floating_point():
+        unique_values = torch.unique(preds, dim=None)
+        if torch.any((unique_values != 0) & (unique_values != 1)):
+            raise RuntimeError(
+                f"Detected the following values in `preds`: {unique_values} but expected only"
+                " the following values [0,1] since preds is a label tensor."
+            )
+    if multidim_average != "global" and preds.ndim < 3:
+        raise ValueError("Expected input to be at least 3D when multidim_average is set to `samplewise`")
+
+
+def _multilabel_stat_scores_format(
+    preds: Tensor,
+    target: Tensor,
+    num_labels: int,
+    threshold: float = 0.5,
+    ignore_index: Optional[int] = None,
+) -> Tuple[Tensor, Tensor]:
+    if preds.is_floating_point():
+        preds = normalize_logits_if_needed(preds, "sigmoid")
+        preds = preds > threshold
+    preds = preds.reshape(*preds.shape[:2], -1)
+    target = target.reshape(*target.shape[:2], -1)
+
+    if ignore_index is not None:
+        idx = target == ignore_index
+        target = target.clone()
+        target[idx] = -1
+    return preds, target
+
+
+def _multilabel_stat_scores_update(
+    preds: Tensor,
+    target: Tensor,
+    multidim_average: str = "global",
+) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
+    sum_dim = [0, -1] if multidim_average == "global" else [-1]
+    tp = ((target == preds) & (target == 1)).sum(sum_dim).squeeze()
+    fn = ((target != preds) & (target == 1)).sum(sum_dim).squeeze()
+    fp = ((target != preds) & (target == 0)).sum(sum_dim).squeeze()
+    tn = ((target == preds) & (target == 0)).sum(sum_dim).squeeze()
+    return tp, fp, tn, fn
+
+
+def _multilabel_stat_scores_update_fused(
+    preds: Tensor,
+    target: Tensor,
+    num_labels: int,
+    threshold: float,
+    multidim_average: str,
+    ignore_index: Optional[int],
+) -> Optional[Tuple[Tensor, Tensor, Tensor, Tensor]]:
+    """Single-pass HIP path: (N, L) float preds, global averaging."""
+    if (
+        preds.is_cuda
+        and preds.is_floating_point()
+        and preds.dtype in (torch.float32, torch.bfloat16)
+        and multidim_average == "global"
+        and preds.ndim >= 2
+    ):
+        p2 = preds.reshape(*preds.shape[:2], -1).movedim(-1, 0).reshape(-1, num_labels)
+        t2 = target.reshape(*target.shape[:2], -1).movedim(-1, 0).reshape(-1, num_labels)
+        return ops.multilabel_stat_scores_fused(p2, t2, threshold, ignore_index)
+    return None
+
+
+def _multilabel_stat_scores_compute(
+    tp: Tensor, fp: Tensor, tn: Tensor, fn: Tensor, average: Optional[str] = "macro", multidim_average: str = "global"
+) -> Tensor:
+    res = torch.stack([tp, fp, tn, fn, tp + fn], dim=-1)
+    sum_dim = 0 if multidim_average == "global" else 1
+    if average == "micro":
+        return res.sum(sum_dim)
+    if average == "macro":
+        return res.float().mean(sum_dim)
+    if average == "weighted":
+        w = tp + fn
+        return (res * (w / w.sum()).reshape(*w.shape, 1)).sum(sum_dim)
+    if average is None or average == "none":
+        return res
+    return None
+
+
+def multilabel_stat_scores(
+    preds: Tensor,
+    target: Tensor,
+    num_labels: int,
+    threshold: float = 0.5,
+    average: Optional[str] = "macro",
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tensor:
+    """Number of tp/fp/tn/fn + support for multilabel tasks."""
+    if validate_args:
+        _multilabel_stat_scores_arg_validation(num_labels, threshold, average, multidim_average, ignore_index)
+        _multilabel_stat_scores_tensor_validation(preds, target, num_labels, multidim_average, ignore_index)
+    fused = _multilabel_stat_scores_update_fused(preds, target, num_labels, threshold, multidim_average, ignore_index)
+    if fused is not None:
+        tp, fp, tn, fn = fused
+    else:
+        preds, target = _multilabel_stat_scores_format(preds, target, num_labels, threshold, ignore_index)
+        tp, fp, tn, fn = _multilabel_stat_scores_update(preds, target, multidim_average)
+    return _multilabel_stat_scores_compute(tp, fp, tn, fn, average, multidim_average)
+
+
+def _binary_stat_scores_pipeline(
+    preds: Tensor,
+    target: Tensor,
+    threshold: float = 0.5,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
+    """format+update with the fused-GPU fast path — shared by all derived metrics."""
+    fused = _binary_stat_scores_update_fused(preds, target, threshold, multidim_average, ignore_index)
+    if fused is not None:
+        return fused
+    preds, target = _binary_stat_scores_format(preds, target, threshold, ignore_index)
+    return _binary_stat_scores_update(preds, target, multidim_average)
+
+
+def _multiclass_stat_scores_pipeline(
+    preds: Tensor,
+    target: Tensor,
+    num_classes: int,
+    top_k: int = 1,
+    average: Optional[str] = "macro",
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
+    preds, target = _multiclass_stat_scores_format(
+        preds, target, top_k, keep_logits=_mc_fused_eligible(preds, target, top_k, multidim_average)
+    )
+    return _multiclass_stat_scores_update(preds, target, num_classes, top_k, average, multidim_average, ignore_index)
+
+
+def _multilabel_stat_scores_pipeline(
+    preds: Tensor,
+    target: Tensor,
+    num_labels: int,
+    threshold: float = 0.5,
+    multidim_average: str = "global",
+    ignore_index: Optional[int] = None,
+) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
+    fused = _multilabel_stat_scores_update_fused(preds, target, num_labels, threshold, multidim_average, ignore_index)
+    if fused is not None:
+        return fused
+    preds, target = _multilabel_stat_scores_format(preds, target, num_labels, threshold, ignore_index)
+    return _multilabel_stat_scores_update(preds, target, multidim_average)
+
+
+def stat_scores(
+    preds: Tensor,
+    target: Tensor,
+    task: str,
+    threshold: float = 0.5,
+    num_classes: Optional[int] = None,
+    num_labels: Optional[int] = None,
+    average: Optional[str] = "micro",
+    multidim_average: str = "global",
+    top_k: int = 1,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tensor:
+    """Task-dispatching stat scores."""
+    task = ClassificationTask.from_str(task)
+    if task == ClassificationTask.BINARY:
+        return binary_stat_scores(preds, target, threshold, multidim_average, ignore_index, validate_args)
+    if task == ClassificationTask.MULTICLASS:
+        if not isinstance(num_classes, int):
+            raise ValueError(f"`num_classes` is expected to be `int` but `{type(num_classes)} was passed.`")
+        if not isinstance(top_k, int):
+            raise ValueError(f"`top_k` is expected to be `int` but `{type(top_k)} was passed.`")
+        return multiclass_stat_scores(
+            preds, target, num_classes, average, top_k, multidim_average, ignore_index, validate_args
+        )
+    if task == ClassificationTask.MULTILABEL:
+        if not isinstance(num_labels, int):
+            raise ValueError(f"`num_labels` is expected to be `int` but `{type(num_labels)} was passed.`")
+        return multilabel_stat_scores(
+            preds, target, num_labels, threshold, average, multidim_average, ignore_index, validate_args
+        )
+    raise ValueError(f"Not handled value: {task}")

@@ -1,0 +1,351 @@
+"""ctypes bindings for the in-tree gfx950 HIP kernel library.
+
+Every wrapper takes torch CUDA (= HIP on ROCm) tensors, validates layout,
+and launches on the *current* torch stream — so kernels order correctly with
+surrounding torch ops without any global synchronization.
+
+If the library is missing on a GPU machine, these raise loudly: metric
+updates must never silently fall back to stock torch kernels on MI355X.
+"""
+from __future__ import annotations
+
+import ctypes
+from typing import Optional, Tuple
+
+import torch
+from torch import Tensor
+
+from metrics_amd.csrc.build import lib_path
+
+_LIB: Optional[ctypes.CDLL] = None
+_LOAD_ERR: Optional[str] = None
+
+
+def _load() -> Optional[ctypes.CDLL]:
+    global _LIB, _LOAD_ERR
+    if _LIB is not None:
+        return _LIB
+    p = lib_path()
+    if not p.exists():
+        _LOAD_ERR = f"HIP kernel library not built at {p}. Run `python -m metrics_amd.csrc.build`."
+        return None
+    try:
+        _LIB = ctypes.CDLL(str(p))
+    except OSError as err:
+        _LOAD_ERR = f"Failed to load {p}: {err}"
+        return None
+    return _LIB
+
+
+def hip_available() -> bool:
+    """True if the kernel library is built and loadable."""
+    return _load() is not None
+
+
+def _lib() -> ctypes.CDLL:
+    lib = _load()
+    if lib is None:
+        raise RuntimeError(
+            f"metrics_amd HIP extension required for GPU tensors but unavailable: {_LOAD_ERR}"
+        )
+    return lib
+
+
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _check(rc: int, name: str) -> None:
+    if rc != 0:
+        raise RuntimeError(f"HIP kernel {name} failed with error code {rc}")
+
+
+def _dtype_code(t: Tensor) -> int:
+    if t.dtype == torch.float32:
+        return 0
+    if t.dtype == torch.bfloat16:
+        return 1
+    raise TypeError(f"HIP metric kernels support float32/bfloat16 inputs, got {t.dtype}")
+
+
+def mc_stat_logits(
+    preds: Tensor, target: Tensor, ignore_index: Optional[int], want_confmat: bool,
+    want_argmax: bool = False,
+) -> Tuple[Tensor, Tensor, Tensor, Tensor, Optional[Tensor], Optional[Tensor]]:
+    """Fused row-argmax + per-class tp/fp/fn (+confmat) over (B, C) logits.
+
+    Returns (tp, fp, fn, valid_count, confmat|None, argmax|None), all int64 on device.
+    """
+    lib = _lib()
+    assert preds.ndim == 2 and target.ndim == 1 and preds.shape[0] == target.shape[0]
+    preds = preds.contiguous()
+    target = target.contiguous().long()
+    B, C = preds.shape
+    dev = preds.device
+    tp = torch.zeros(C, dtype=torch.long, device=dev)
+    fp = torch.zeros(C, dtype=torch.long, device=dev)
+    fn = torch.zeros(C, dtype=torch.long, device=dev)
+    valid = torch.zeros(1, dtype=torch.long, device=dev)
+    confmat = torch.zeros(C, C, dtype=torch.long, device=dev) if want_confmat else None
+    argmax = torch.empty(B, dtype=torch.long, device=dev) if want_argmax else None
+    rc = lib.ma_mc_stat_logits(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(preds.data_ptr()),
+        ctypes.c_int(_dtype_code(preds)),
+        ctypes.c_uint64(target.data_ptr()),
+        ctypes.c_longlong(B),
+        ctypes.c_longlong(C),
+        ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+        ctypes.c_int(1 if ignore_index is not None else 0),
+        ctypes.c_uint64(tp.data_ptr()),
+        ctypes.c_uint64(fp.data_ptr()),
+        ctypes.c_uint64(fn.data_ptr()),
+        ctypes.c_uint64(confmat.data_ptr() if confmat is not None else 0),
+        ctypes.c_uint64(valid.data_ptr()),
+        ctypes.c_uint64(argmax.data_ptr() if argmax is not None else 0),
+    )
+    _check(rc, "ma_mc_stat_logits")
+    return tp, fp, fn, valid, confmat, argmax
+
+
+def mc_stat_labels(
+    preds: Tensor, target: Tensor, num_classes: int, ignore_index: Optional[int], want_confmat: bool
+) -> Tuple[Tensor, Tensor, Tensor, Tensor, Optional[Tensor]]:
+    """Per-class tp/fp/fn (+confmat) from integer label predictions."""
+    lib = _lib()
+    preds = preds.contiguous().long().flatten()
+    target = target.contiguous().long().flatten()
+    N = preds.numel()
+    dev = preds.device
+    tp = torch.zeros(num_classes, dtype=torch.long, device=dev)
+    fp = torch.zeros(num_classes, dtype=torch.long, device=dev)
+    fn = torch.zeros(num_classes, dtype=torch.long, device=dev)
+    valid = torch.zeros(1, dtype=torch.long, device=dev)
+    confmat = torch.zeros(num_classes, num_classes, dtype=torch.long, device=dev) if want_confmat else None
+    rc = lib.ma_mc_stat_labels(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(preds.data_ptr()),
+        ctypes.c_uint64(target.data_ptr()),
+        ctypes.c_longlong(N),
+        ctypes.c_longlong(num_classes),
+        ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+        ctypes.c_int(1 if ignore_index is not None else 0),
+        ctypes.c_uint64(tp.data_ptr()),
+        ctypes.c_uint64(fp.data_ptr()),
+        ctypes.c_uint64(fn.data_ptr()),
+        ctypes.c_uint64(confmat.data_ptr() if confmat is not None else 0),
+        ctypes.c_uint64(valid.data_ptr()),
+    )
+    _check(rc, "ma_mc_stat_labels")
+    return tp, fp, fn, valid, confmat
+
+
+def bincount(x: Tensor, minlength: int) -> Tensor:
+    """Deterministic LDS-privatized histogram of a non-negative int64 tensor."""
+    lib = _lib()
+    x = x.contiguous().long().flatten()
+    out = torch.zeros(minlength, dtype=torch.long, device=x.device)
+    rc = lib.ma_bincount(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(x.data_ptr()),
+        ctypes.c_longlong(x.numel()),
+        ctypes.c_longlong(minlength),
+        ctypes.c_uint64(out.data_ptr()),
+    )
+    _check(rc, "ma_bincount")
+    return out
+
+
+def binary_stat(
+    preds: Tensor, target: Tensor, threshold: float, ignore_index: Optional[int]
+) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
+    """Fused binary tp/fp/tn/fn in one pass; picks raw vs sigmoid thresholding on-device."""
+    lib = _lib()
+    preds = preds.contiguous()
+    target = target.contiguous().long()
+    N = preds.numel()
+    dev = preds.device
+    out = torch.zeros(2, 4, dtype=torch.long, device=dev)  # {raw,sig} x {tp,fp,tn,fn}
+    flag = torch.zeros(1, dtype=torch.int32, device=dev)
+    thr = min(max(threshold, 1e-7), 1 - 1e-7)
+    rc = lib.ma_binary_stat(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(preds.data_ptr()),
+        ctypes.c_int(_dtype_code(preds)),
+        ctypes.c_uint64(target.data_ptr()),
+        ctypes.c_longlong(N),
+        ctypes.c_float(thr),
+        ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+        ctypes.c_int(1 if ignore_index is not None else 0),
+        ctypes.c_uint64(out.data_ptr()),
+        ctypes.c_uint64(flag.data_ptr()),
+    )
+    _check(rc, "ma_binary_stat")
+    sel = flag.long().squeeze(0)  # 0 -> raw counts, 1 -> sigmoid counts (no host sync)
+    counts = out[0] * (1 - sel) + out[1] * sel
+    return counts[0], counts[1], counts[2], counts[3]
+
+
+def multilabel_stat(
+    preds: Tensor, target: Tensor, threshold: float, ignore_index: Optional[int]
+) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
+    """Fused per-label tp/fp/tn/fn over (N, L)."""
+    lib = _lib()
+    assert preds.ndim == 2
+    preds = preds.contiguous()
+    target = target.contiguous().long()
+    N, L = preds.shape
+    dev = preds.device
+    out = torch.zeros(2, L, 4, dtype=torch.long, device=dev)
+    flag = torch.zeros(1, dtype=torch.int32, device=dev)
+    thr = min(max(threshold, 1e-7), 1 - 1e-7)
+    rc = lib.ma_multilabel_stat(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(preds.data_ptr()),
+        ctypes.c_int(_dtype_code(preds)),
+        ctypes.c_uint64(target.data_ptr()),
+        ctypes.c_longlong(N),
+        ctypes.c_longlong(L),
+        ctypes.c_float(thr),
+        ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+        ctypes.c_int(1 if ignore_index is not None else 0),
+        ctypes.c_uint64(out.data_ptr()),
+        ctypes.c_uint64(flag.data_ptr()),
+    )
+    _check(rc, "ma_multilabel_stat")
+    sel = flag.long().squeeze(0)
+    counts = out[0] * (1 - sel) + out[1] * sel  # (L, 4)
+    return counts[:, 0], counts[:, 1], counts[:, 2], counts[:, 3]
+
+
+def binary_curve_confmat(
+    preds: Tensor, target: Tensor, thresholds: Tensor, ignore_index: Optional[int]
+) -> Tensor:
+    """(T,2,2) threshold confmat via bucketized histogram + on-device suffix-sum.
+
+    ``preds`` must already be probabilities in [0,1].
+    """
+    lib = _lib()
+    preds = preds.contiguous()
+    target = target.contiguous().long()
+    thr = thresholds.contiguous().float()
+    T = thr.numel()
+    dev = preds.device
+    hist = torch.zeros(T + 1, 2, dtype=torch.long, device=dev)
+    rc = lib.ma_binary_curve_hist(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(preds.data_ptr()),
+        ctypes.c_int(_dtype_code(preds)),
+        ctypes.c_uint64(target.data_ptr()),
+        ctypes.c_longlong(preds.numel()),
+        ctypes.c_uint64(thr.data_ptr()),
+        ctypes.c_int(T),
+        ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+        ctypes.c_int(1 if ignore_index is not None else 0),
+        ctypes.c_uint64(hist.data_ptr()),
+    )
+    _check(rc, "ma_binary_curve_hist")
+    confmat = torch.zeros(T, 2, 2, dtype=torch.long, device=dev)
+    rc = lib.ma_curve_suffix(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(hist.data_ptr()),
+        ctypes.c_longlong(1),
+        ctypes.c_int(T),
+        ctypes.c_uint64(confmat.data_ptr()),
+    )
+    _check(rc, "ma_curve_suffix")
+    return confmat
+
+
+def multiclass_curve_confmat(
+    probs: Tensor, target: Tensor, thresholds: Tensor, ignore_index: Optional[int], mode: int = 0
+) -> Tensor:
+    """(T,C,2,2) threshold confmats. mode 0: multiclass one-vs-rest (target (B,));
+    mode 1: multilabel (target (B,C)). ``probs`` already normalized (B,C)."""
+    lib = _lib()
+    probs = probs.contiguous()
+    target = target.contiguous().long()
+    thr = thresholds.contiguous().float()
+    B, C = probs.shape
+    T = thr.numel()
+    dev = probs.device
+    hist = torch.zeros(C, T + 1, 2, dtype=torch.long, device=dev)
+    rc = lib.ma_multiclass_curve_hist(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(probs.data_ptr()),
+        ctypes.c_int(_dtype_code(probs)),
+        ctypes.c_uint64(target.data_ptr()),
+        ctypes.c_longlong(B),
+        ctypes.c_longlong(C),
+        ctypes.c_uint64(thr.data_ptr()),
+        ctypes.c_int(T),
+        ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+        ctypes.c_int(1 if ignore_index is not None else 0),
+        ctypes.c_int(mode),
+        ctypes.c_uint64(hist.data_ptr()),
+    )
+    _check(rc, "ma_multiclass_curve_hist")
+    confmat_cls = torch.zeros(C, T, 2, 2, dtype=torch.long, device=dev)
+    rc = lib.ma_curve_suffix(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(hist.data_ptr()),
+        ctypes.c_longlong(C),
+        ctypes.c_int(T),
+        ctypes.c_uint64(confmat_cls.data_ptr()),
+    )
+    _check(rc, "ma_curve_suffix")
+    return confmat_cls.permute(1, 0, 2, 3).contiguous()  # (T, C, 2, 2)
+
+
+_ERR_OPS = {"sq_err": (0, 1), "abs_err": (1, 1), "ape": (2, 1), "sq_log_err": (3, 1), "moments": (4, 6), "logcosh": (5, 1)}
+
+
+def err_reduce(x: Tensor, y: Tensor, op: str, eps: float = 1.17e-6) -> Tensor:
+    """Deterministic fused elementwise-error reduction; returns fp64 sums (n_out,)."""
+    lib = _lib()
+    op_id, n_out = _ERR_OPS[op]
+    x = x.contiguous()
+    y = y.contiguous()
+    N = x.numel()
+    num_blocks = min(max((N + 255) // 256, 1), 2048)
+    dev = x.device
+    partials = torch.zeros(num_blocks, n_out, dtype=torch.float64, device=dev)
+    out = torch.zeros(n_out, dtype=torch.float64, device=dev)
+    rc = lib.ma_err_reduce(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(x.data_ptr()),
+        ctypes.c_uint64(y.data_ptr()),
+        ctypes.c_int(_dtype_code(x)),
+        ctypes.c_longlong(N),
+        ctypes.c_int(op_id),
+        ctypes.c_double(eps),
+        ctypes.c_uint64(partials.data_ptr()),
+        ctypes.c_int(num_blocks),
+        ctypes.c_int(n_out),
+        ctypes.c_uint64(out.data_ptr()),
+    )
+    _check(rc, "ma_err_reduce")
+    return out
+
+
+def box_iou(boxes1: Tensor, boxes2: Tensor, variant: str = "iou") -> Tensor:
+    """All-pairs (N,M) box IoU / GIoU / DIoU / CIoU on xyxy boxes."""
+    lib = _lib()
+    v = {"iou": 0, "giou": 1, "diou": 2, "ciou": 3}[variant]
+    boxes1 = boxes1.contiguous().float()
+    boxes2 = boxes2.contiguous().float()
+    N, M = boxes1.shape[0], boxes2.shape[0]
+    out = torch.empty(N, M, dtype=torch.float32, device=boxes1.device)
+    if N == 0 or M == 0:
+        return out
+    rc = lib.ma_box_iou(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(boxes1.data_ptr()),
+        ctypes.c_longlong(N),
+        ctypes.c_uint64(boxes2.data_ptr()),
+        ctypes.c_longlong(M),
+        ctypes.c_int(v),
+        ctypes.c_uint64(out.data_ptr()),
+    )
+    _check(rc, "ma_box_iou")
+    return out

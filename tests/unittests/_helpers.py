@@ -1,0 +1,124 @@
+"""Test helpers: oracle-comparison runner + 2-process gloo cluster emulation.
+
+Modeled on the reference's MetricTester strategy (torchmetrics
+tests/unittests/_helpers/testers.py): batch-by-batch forward vs reference,
+final compute vs reference on all data, clone/pickle/reset checks; DDP tests
+run the same check over CPU gloo processes with interleaved batches.
+"""
+from __future__ import annotations
+
+import os
+import pickle
+from functools import partial
+from typing import Any, Callable, Dict, Optional, Sequence
+
+import numpy as np
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+from torch import Tensor
+
+from metrics_amd.metric import Metric
+
+NUM_BATCHES = 4
+BATCH_SIZE = 32
+NUM_CLASSES = 5
+
+
+def seed_all(seed: int = 42) -> None:
+    np.random.seed(seed)
+    torch.manual_seed(seed)
+
+
+def _assert_allclose(res: Any, ref: Any, atol: float = 1e-5) -> None:
+    if isinstance(res, Tensor):
+        ref_t = torch.as_tensor(np.asarray(ref), dtype=torch.float64)
+        assert torch.allclose(res.double().cpu(), ref_t, atol=atol), f"{res} vs {ref}"
+    elif isinstance(res, dict):
+        for k in res:
+            _assert_allclose(res[k], ref[k], atol)
+    elif isinstance(res, (list, tuple)):
+        for r, rf in zip(res, ref):
+            _assert_allclose(r, rf, atol)
+    else:
+        assert abs(float(res) - float(ref)) < atol
+
+
+def run_class_metric_test(
+    metric_class: Callable,
+    ref_fn: Callable,
+    preds: Tensor,
+    target: Tensor,
+    metric_args: Optional[Dict[str, Any]] = None,
+    check_batch: bool = True,
+    atol: float = 1e-5,
+) -> None:
+    """Instantiate, run batch-by-batch forward + final compute vs the oracle.
+
+    ``preds``/``target`` are (NUM_BATCHES, B, ...); ``ref_fn(preds_np, target_np)``
+    computes the expected value over concatenated data.
+    """
+    metric_args = metric_args or {}
+    metric = metric_class(**metric_args)
+
+    # pickle roundtrip must work
+    metric = pickle.loads(pickle.dumps(metric))
+    cloned = metric.clone()
+    assert cloned is not metric
+
+    for i in range(preds.shape[0]):
+        batch_val = metric(preds[i], target[i])
+        if check_batch:
+            ref_b = ref_fn(preds[i], target[i])
+            _assert_allclose(batch_val, ref_b, atol)
+
+    total_val = metric.compute()
+    flat_p = preds.reshape(-1, *preds.shape[2:])
+    flat_t = target.reshape(-1, *target.shape[2:])
+    ref_total = ref_fn(flat_p, flat_t)
+    _assert_allclose(total_val, ref_total, atol)
+
+    # reset restores defaults
+    metric.reset()
+    for name, default in metric._defaults.items():
+        cur = getattr(metric, name)
+        if isinstance(default, Tensor):
+            assert torch.allclose(cur, default.to(cur.device))
+        else:
+            assert cur == []
+
+
+def run_functional_metric_test(
+    metric_fn: Callable,
+    ref_fn: Callable,
+    preds: Tensor,
+    target: Tensor,
+    metric_args: Optional[Dict[str, Any]] = None,
+    atol: float = 1e-5,
+) -> None:
+    metric_args = metric_args or {}
+    for i in range(preds.shape[0]):
+        res = metric_fn(preds[i], target[i], **metric_args)
+        ref = ref_fn(preds[i], target[i])
+        _assert_allclose(res, ref, atol)
+
+
+# ---------------------------------------------------------------- distributed
+def _dist_worker(rank: int, world_size: int, port: int, fn: Callable, args: tuple) -> None:
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        fn(rank, world_size, *args)
+    finally:
+        dist.destroy_process_group()
+
+
+def run_distributed(fn: Callable, world_size: int = 2, args: tuple = ()) -> None:
+    """Run ``fn(rank, world_size, *args)`` in ``world_size`` gloo processes on localhost."""
+    port = np.random.randint(20000, 65000)
+    mp.start_processes(
+        partial(_dist_worker, world_size=world_size, port=port, fn=fn, args=args),
+        nprocs=world_size,
+        start_method="spawn" if os.environ.get("MA_DIST_SPAWN") else "fork",
+    )

@@ -1,0 +1,215 @@
+"""GPU numerics tests: every HIP kernel vs the plain-torch fp32 CPU reference.
+
+All tests marked ``gpu`` — run on an MI355X via gpurun / the driver.
+"""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+import metrics_amd as ma
+from metrics_amd import ops
+
+
+@pytest.fixture(autouse=True)
+def _require_hip():
+    assert torch.cuda.is_available()
+    assert ops.hip_available(), "HIP kernel library must be built and loadable on a GPU box"
+
+
+def test_bincount_matches_torch():
+    x = torch.randint(0, 1000, (1_000_000,), device="cuda")
+    out = ops.hip_bincount(x, 1000)
+    ref = torch.bincount(x.cpu(), minlength=1000)
+    assert torch.equal(out.cpu(), ref)
+
+
+def test_bincount_large_bins():
+    x = torch.randint(0, 100_000, (2_000_000,), device="cuda")
+    out = ops.hip_bincount(x, 100_000)
+    ref = torch.bincount(x.cpu(), minlength=100_000)
+    assert torch.equal(out.cpu(), ref)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("C", [10, 1000, 1001])
+def test_mc_stat_logits_kernel(dtype, C):
+    B = 4096
+    preds = torch.randn(B, C, device="cuda", dtype=dtype)
+    target = torch.randint(0, C, (B,), device="cuda")
+    tp, fp, tn, fn, confmat = ops.multiclass_stat_scores_fused(preds, target, C, None, want_confmat=True)
+    # CPU fp32 reference
+    plabels = preds.float().cpu().argmax(-1)
+    tcpu = target.cpu()
+    ref_cm = torch.zeros(C, C, dtype=torch.long)
+    for t, p in zip(tcpu.tolist(), plabels.tolist()):
+        ref_cm[t, p] += 1
+    assert torch.equal(confmat.cpu(), ref_cm)
+    assert torch.equal(tp.cpu(), ref_cm.diag())
+    assert torch.equal(fp.cpu(), ref_cm.sum(0) - ref_cm.diag())
+    assert torch.equal(fn.cpu(), ref_cm.sum(1) - ref_cm.diag())
+    assert torch.equal(tn.cpu(), ref_cm.sum() - ref_cm.diag() - (ref_cm.sum(0) - ref_cm.diag()) - (ref_cm.sum(1) - ref_cm.diag()))
+
+
+def test_mc_stat_logits_argmax_tiebreak():
+    """torch.argmax returns the first max index; the kernel must match."""
+    B, C = 512, 37
+    preds = torch.randint(0, 3, (B, C), device="cuda").float()  # many ties
+    target = torch.randint(0, C, (B,), device="cuda")
+    tp, fp, tn, fn, cm = ops.multiclass_stat_scores_fused(preds, target, C, None, want_confmat=True)
+    plabels = preds.cpu().argmax(-1)
+    ref_cm = torch.zeros(C, C, dtype=torch.long)
+    for t, p in zip(target.cpu().tolist(), plabels.tolist()):
+        ref_cm[t, p] += 1
+    assert torch.equal(cm.cpu(), ref_cm)
+
+
+def test_mc_stat_ignore_index():
+    B, C = 2048, 100
+    preds = torch.randn(B, C, device="cuda")
+    target = torch.randint(0, C, (B,), device="cuda")
+    target[::5] = -1
+    tp, fp, tn, fn, _ = ops.multiclass_stat_scores_fused(preds, target, C, -1, want_confmat=False)
+    keep = target != -1
+    plabels = preds[keep].cpu().argmax(-1)
+    t = target[keep].cpu()
+    assert tp.sum().item() == (plabels == t).sum().item()
+    assert (tp + fn).sum().item() == keep.sum().item()
+
+
+def test_binary_stat_kernel_probs_and_logits():
+    N = 1_000_000
+    target = torch.randint(0, 2, (N,), device="cuda")
+    # probabilities in [0,1]: raw thresholding
+    probs = torch.rand(N, device="cuda")
+    tp, fp, tn, fn = ops.binary_stat_scores_fused(probs, target, 0.5, None)
+    ref_p = probs > 0.5
+    assert tp.item() == ((ref_p == 1) & (target == 1)).sum().item()
+    assert tn.item() == ((ref_p == 0) & (target == 0)).sum().item()
+    # logits outside [0,1]: sigmoid thresholding must be picked on-device
+    logits = torch.randn(N, device="cuda") * 3
+    tp2, fp2, tn2, fn2 = ops.binary_stat_scores_fused(logits, target, 0.5, None)
+    ref_p2 = torch.sigmoid(logits) > 0.5
+    assert tp2.item() == ((ref_p2 == 1) & (target == 1)).sum().item()
+    assert fn2.item() == ((ref_p2 == 0) & (target == 1)).sum().item()
+
+
+def test_multilabel_stat_kernel():
+    N, L = 65536, 20
+    preds = torch.rand(N, L, device="cuda")
+    target = torch.randint(0, 2, (N, L), device="cuda")
+    tp, fp, tn, fn = ops.multilabel_stat_scores_fused(preds, target, 0.5, None)
+    p = preds > 0.5
+    assert torch.equal(tp.cpu(), ((p == 1) & (target == 1)).sum(0).cpu())
+    assert torch.equal(fp.cpu(), ((p == 1) & (target == 0)).sum(0).cpu())
+
+
+def test_binary_curve_confmat_kernel():
+    N, T = 1_000_000, 1000
+    preds = torch.rand(N, device="cuda")
+    target = torch.randint(0, 2, (N,), device="cuda")
+    thr = torch.linspace(0, 1, T, device="cuda")
+    cm = ops.binary_curve_confmat(preds, target, thr, None)
+    # CPU reference via the bincount formulation on a subsample of thresholds
+    cm_ref = ops.binary_curve_confmat(preds.cpu(), target.cpu(), thr.cpu(), None)
+    assert torch.equal(cm.cpu(), cm_ref)
+
+
+def test_multiclass_curve_confmat_kernel():
+    B, C, T = 8192, 50, 200
+    probs = torch.randn(B, C, device="cuda").softmax(-1)
+    target = torch.randint(0, C, (B,), device="cuda")
+    thr = torch.linspace(0, 1, T, device="cuda")
+    cm = ops.multiclass_curve_confmat(probs, target, thr, None)
+    cm_ref = ops.multiclass_curve_confmat(probs.cpu(), target.cpu(), thr.cpu(), None)
+    assert torch.equal(cm.cpu(), cm_ref)
+
+
+def test_multilabel_curve_confmat_kernel():
+    B, L, T = 8192, 10, 100
+    probs = torch.rand(B, L, device="cuda")
+    target = torch.randint(0, 2, (B, L), device="cuda")
+    thr = torch.linspace(0, 1, T, device="cuda")
+    cm = ops.multilabel_curve_confmat(probs, target, thr, None)
+    cm_ref = ops.multilabel_curve_confmat(probs.cpu(), target.cpu(), thr.cpu(), None)
+    assert torch.equal(cm.cpu(), cm_ref)
+
+
+def test_err_reduce_kernel():
+    N = 1_000_000
+    x = torch.randn(N, device="cuda")
+    y = torch.randn(N, device="cuda")
+    for op in ("sq_err", "abs_err", "sq_log_err"):
+        if op == "sq_log_err":
+            x_, y_ = x.abs(), y.abs()
+        else:
+            x_, y_ = x, y
+        out = ops.err_reduce_sum(x_, y_, op)
+        ref = ops.err_reduce_sum(x_.cpu(), y_.cpu(), op)
+        assert torch.allclose(out.cpu(), ref, rtol=1e-10, atol=1e-6), (op, out, ref)
+    mom = ops.err_reduce_sum(x, y, "moments")
+    ref = ops.err_reduce_sum(x.cpu(), y.cpu(), "moments")
+    assert torch.allclose(mom.cpu(), ref, rtol=1e-10, atol=1e-5)
+
+
+def test_err_reduce_deterministic():
+    x = torch.randn(3_000_000, device="cuda")
+    y = torch.randn(3_000_000, device="cuda")
+    a = ops.err_reduce_sum(x, y, "sq_err")
+    b = ops.err_reduce_sum(x, y, "sq_err")
+    assert torch.equal(a, b)
+
+
+def test_box_iou_kernel():
+    N, M = 500, 700
+    def rand_boxes(n):
+        xy = torch.rand(n, 2, device="cuda") * 100
+        wh = torch.rand(n, 2, device="cuda") * 50 + 1
+        return torch.cat([xy, xy + wh], dim=1)
+    b1, b2 = rand_boxes(N), rand_boxes(M)
+    for variant in ("iou", "giou", "diou", "ciou"):
+        out = ops.box_iou_pairwise(b1, b2, variant)
+        ref = ops.box_iou_pairwise(b1.cpu(), b2.cpu(), variant)
+        assert torch.allclose(out.cpu(), ref, atol=1e-4), variant
+
+
+def test_metric_end_to_end_gpu_vs_cpu():
+    """Full modular metrics on GPU (HIP path) must match the CPU torch path."""
+    torch.manual_seed(0)
+    preds = torch.randn(4096, 100)
+    target = torch.randint(0, 100, (4096,))
+    for make in (
+        lambda: ma.MulticlassAccuracy(num_classes=100, average="micro"),
+        lambda: ma.MulticlassAccuracy(num_classes=100, average="macro"),
+        lambda: ma.MulticlassF1Score(num_classes=100, average="weighted"),
+        lambda: ma.MulticlassConfusionMatrix(num_classes=100),
+    ):
+        m_cpu, m_gpu = make(), make().to("cuda")
+        m_cpu.update(preds, target)
+        m_gpu.update(preds.cuda(), target.cuda())
+        assert torch.allclose(m_cpu.compute().float(), m_gpu.compute().float().cpu(), atol=1e-6)
+
+    bp, bt = torch.rand(100_000), torch.randint(0, 2, (100_000,))
+    for make in (
+        lambda: ma.BinaryAccuracy(),
+        lambda: ma.BinaryF1Score(),
+        lambda: ma.BinaryAUROC(thresholds=1000),
+        lambda: ma.BinaryPrecisionRecallCurve(thresholds=500),
+    ):
+        m_cpu, m_gpu = make(), make().to("cuda")
+        r_cpu = m_cpu(bp, bt)
+        r_gpu = m_gpu(bp.cuda(), bt.cuda())
+        if isinstance(r_cpu, tuple):
+            for a, b in zip(r_cpu, r_gpu):
+                assert torch.allclose(a, b.cpu(), atol=1e-6)
+        else:
+            assert torch.allclose(r_cpu, r_gpu.cpu(), atol=1e-6)
+
+
+def test_bf16_inputs_gpu():
+    preds = torch.randn(8192, 1000, device="cuda", dtype=torch.bfloat16)
+    target = torch.randint(0, 1000, (8192,), device="cuda")
+    m = ma.MulticlassAccuracy(num_classes=1000, average="micro").to("cuda")
+    v = m(preds, target)
+    ref = (preds.float().argmax(-1) == target).float().mean()
+    assert torch.allclose(v, ref, atol=1e-6)
