@@ -90,9 +90,11 @@ def main() -> None:
         if args.compute_every and (i + 1) % args.compute_every == 0:
             coll.compute()
 
-    # warmup (untimed)
+    # warmup (untimed) — includes one compute() so one-time lazy allocation /
+    # caching costs land outside the timed region
     for i in range(args.warmup):
         one_step(i)
+    coll.compute()
     coll.reset()
 
     if distributed:

@@ -250,7 +250,10 @@ def _multiclass_precision_recall_curve_format(
     average: Optional[str] = None,
     remove_ignored: bool = True,
 ) -> Tuple[Tensor, Tensor, Optional[Tensor]]:
-    preds = preds.transpose(0, 1).reshape(num_classes, -1).T
+    if preds.ndim == 2:
+        pass  # already (N, C); avoid the transpose->reshape round-trip copy
+    else:
+        preds = preds.transpose(0, 1).reshape(num_classes, -1).T
     target = target.flatten()
 
     if ignore_index is not None and remove_ignored:
@@ -389,8 +392,11 @@ def _multilabel_precision_recall_curve_format(
     ignore_index: Optional[int] = None,
     remove_ignored: bool = True,
 ) -> Tuple[Tensor, Tensor, Optional[Tensor]]:
-    preds = preds.transpose(0, 1).reshape(num_labels, -1).T
-    target = target.transpose(0, 1).reshape(num_labels, -1).T
+    if preds.ndim == 2:
+        pass  # already (N, L)
+    else:
+        preds = preds.transpose(0, 1).reshape(num_labels, -1).T
+        target = target.transpose(0, 1).reshape(num_labels, -1).T
 
     preds = normalize_logits_if_needed(preds, "sigmoid")
 

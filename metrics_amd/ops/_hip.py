@@ -218,6 +218,40 @@ def multilabel_stat(
     return counts[:, 0], counts[:, 1], counts[:, 2], counts[:, 3]
 
 
+_UNIFORM_CACHE: dict = {}
+
+
+def _uniform_params(thr: Tensor):
+    """Detect uniform threshold spacing for the O(1) bucket guess.
+
+    The kernel's +-1 fixup keeps results exact even if the guess is off.
+    Result is cached per (data_ptr, numel): thresholds are fixed per metric,
+    and the detection needs device->host syncs we don't want per update.
+    """
+    key = (thr.data_ptr(), thr.numel(), thr.device.index)
+    hit = _UNIFORM_CACHE.get(key)
+    if hit is not None:
+        return hit
+    T = thr.numel()
+    if T < 2:
+        out = (0, 0.0, 1.0)
+    else:
+        tc = thr.detach().float().cpu()
+        t0, t_last = float(tc[0]), float(tc[-1])
+        if t_last <= t0:
+            out = (0, 0.0, 1.0)
+        else:
+            step = (t_last - t0) / (T - 1)
+            diffs = tc[1:] - tc[:-1]
+            # strictly uniform within small tolerance -> O(1) guess stays O(1)
+            uniform = bool(((diffs - step).abs() <= 1e-3 * step + 1e-9).all())
+            out = (1 if uniform else 0, t0, (T - 1) / (t_last - t0))
+    if len(_UNIFORM_CACHE) > 256:
+        _UNIFORM_CACHE.clear()
+    _UNIFORM_CACHE[key] = out
+    return out
+
+
 def binary_curve_confmat(
     preds: Tensor, target: Tensor, thresholds: Tensor, ignore_index: Optional[int]
 ) -> Tensor:
@@ -231,6 +265,7 @@ def binary_curve_confmat(
     thr = thresholds.contiguous().float()
     T = thr.numel()
     dev = preds.device
+    uni, t0, inv_step = _uniform_params(thr)
     hist = torch.zeros(T + 1, 2, dtype=torch.long, device=dev)
     rc = lib.ma_binary_curve_hist(
         ctypes.c_uint64(_stream()),
@@ -242,6 +277,9 @@ def binary_curve_confmat(
         ctypes.c_int(T),
         ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
         ctypes.c_int(1 if ignore_index is not None else 0),
+        ctypes.c_int(uni),
+        ctypes.c_float(t0),
+        ctypes.c_float(inv_step),
         ctypes.c_uint64(hist.data_ptr()),
     )
     _check(rc, "ma_binary_curve_hist")
@@ -267,8 +305,10 @@ def multiclass_curve_confmat(
     target = target.contiguous().long()
     thr = thresholds.contiguous().float()
     B, C = probs.shape
+    assert B * C < 2**31, "curve histogram fast path requires B*C < 2^31; chunk the update"
     T = thr.numel()
     dev = probs.device
+    uni, t0, inv_step = _uniform_params(thr)
     hist = torch.zeros(C, T + 1, 2, dtype=torch.long, device=dev)
     rc = lib.ma_multiclass_curve_hist(
         ctypes.c_uint64(_stream()),
@@ -282,6 +322,9 @@ def multiclass_curve_confmat(
         ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
         ctypes.c_int(1 if ignore_index is not None else 0),
         ctypes.c_int(mode),
+        ctypes.c_int(uni),
+        ctypes.c_float(t0),
+        ctypes.c_float(inv_step),
         ctypes.c_uint64(hist.data_ptr()),
     )
     _check(rc, "ma_multiclass_curve_hist")
