@@ -305,7 +305,6 @@ def multiclass_curve_confmat(
     target = target.contiguous().long()
     thr = thresholds.contiguous().float()
     B, C = probs.shape
-    assert B * C < 2**31, "curve histogram fast path requires B*C < 2^31; chunk the update"
     T = thr.numel()
     dev = probs.device
     uni, t0, inv_step = _uniform_params(thr)
