@@ -26,8 +26,9 @@ from metrics_amd.aggregation import (  # noqa: E402
 )
 from metrics_amd.collections import MetricCollection  # noqa: E402
 from metrics_amd.metric import CompositionalMetric, Metric  # noqa: E402
-from metrics_amd import classification, functional, ops, utilities, wrappers  # noqa: E402
+from metrics_amd import classification, functional, ops, regression, utilities, wrappers  # noqa: E402
 from metrics_amd.classification import *  # noqa: E402,F401,F403
+from metrics_amd.regression import *  # noqa: E402,F401,F403
 
 __all__ = [
     "CatMetric",
@@ -41,9 +42,11 @@ __all__ = [
     "RunningSum",
     "SumMetric",
     "classification",
+    "regression",
     "functional",
     "ops",
     "utilities",
     "wrappers",
 ]
 __all__ += classification.__all__
+__all__ += regression.__all__

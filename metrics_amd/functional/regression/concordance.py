@@ -1,0 +1,29 @@
+"""Concordance correlation coefficient. Parity: torchmetrics ``functional/regression/concordance.py``."""
+from __future__ import annotations
+
+import torch
+from torch import Tensor
+
+from metrics_amd.functional.regression.pearson import _pearson_corrcoef_update
+
+
+def _concordance_corrcoef_compute(
+    mean_x: Tensor, mean_y: Tensor, var_x: Tensor, var_y: Tensor, corr_xy: Tensor, nb: Tensor
+) -> Tensor:
+    """CCC from the running pearson moments."""
+    var_x = var_x / nb
+    var_y = var_y / nb
+    corr_xy = corr_xy / nb
+    return (2.0 * corr_xy / (var_x + var_y + (mean_x - mean_y) ** 2)).squeeze()
+
+
+def concordance_corrcoef(preds: Tensor, target: Tensor) -> Tensor:
+    """Concordance correlation coefficient."""
+    d = preds.shape[1] if preds.ndim == 2 else 1
+    _temp = torch.zeros(d, dtype=preds.dtype, device=preds.device)
+    mean_x, mean_y, var_x = _temp.clone(), _temp.clone(), _temp.clone()
+    var_y, corr_xy, nb = _temp.clone(), _temp.clone(), _temp.clone()
+    mean_x, mean_y, var_x, var_y, corr_xy, nb = _pearson_corrcoef_update(
+        preds, target, mean_x, mean_y, var_x, var_y, corr_xy, nb, num_outputs=d
+    )
+    return _concordance_corrcoef_compute(mean_x, mean_y, var_x, var_y, corr_xy, nb)

@@ -1,0 +1,130 @@
+"""Pearson correlation — streaming Welford moments with parallel merge.
+
+Parity: torchmetrics ``functional/regression/pearson.py`` (the streaming
+mean/var/cov update and the cross-device ``_final_aggregation`` merge).
+"""
+from __future__ import annotations
+
+import math
+from typing import Tuple
+
+import torch
+from torch import Tensor
+
+from metrics_amd.utilities.checks import _check_same_shape
+
+
+def _pearson_corrcoef_update(
+    preds: Tensor,
+    target: Tensor,
+    mean_x: Tensor,
+    mean_y: Tensor,
+    var_x: Tensor,
+    var_y: Tensor,
+    corr_xy: Tensor,
+    num_prior: Tensor,
+    num_outputs: int,
+) -> Tuple[Tensor, Tensor, Tensor, Tensor, Tensor, Tensor]:
+    """Chan-style batch merge of mean/var/cov running statistics."""
+    _check_same_shape(preds, target)
+    _cond = num_prior.mean() > 0 or num_outputs > 1
+    n_obs = preds.shape[0]
+    if _cond:
+        mx_new = (num_prior * mean_x + preds.sum(0)) / (num_prior + n_obs)
+        my_new = (num_prior * mean_y + target.sum(0)) / (num_prior + n_obs)
+    else:
+        mx_new = preds.mean(0).to(mean_x.dtype)
+        my_new = target.mean(0).to(mean_y.dtype)
+
+    num_prior = num_prior + n_obs
+
+    if _cond:
+        var_x = var_x + ((preds - mx_new) * (preds - mean_x)).sum(0)
+        var_y = var_y + ((target - my_new) * (target - mean_y)).sum(0)
+    else:
+        var_x = var_x + preds.var(0) * (n_obs - 1)
+        var_y = var_y + target.var(0) * (n_obs - 1)
+    corr_xy = corr_xy + ((preds - mx_new) * (target - mean_y)).sum(0)
+    mean_x = mx_new
+    mean_y = my_new
+
+    return mean_x, mean_y, var_x, var_y, corr_xy, num_prior
+
+
+def _pearson_corrcoef_compute(var_x: Tensor, var_y: Tensor, corr_xy: Tensor, nb: Tensor) -> Tensor:
+    """Correlation from the accumulated second moments."""
+    var_x /= nb - 1
+    var_y /= nb - 1
+    corr_xy /= nb - 1
+    # compute in double for numerical stability on near-constant inputs
+    if var_x.dtype == torch.float32:
+        var_x = var_x.double()
+        var_y = var_y.double()
+        corr_xy = corr_xy.double()
+    bound = math.sqrt(torch.finfo(var_x.dtype).eps)
+    if (var_x < bound).any() or (var_y < bound).any():
+        import warnings
+
+        warnings.warn(
+            "The variance of predictions or target is close to zero. This can cause instability in Pearson correlation"
+            "coefficient, leading to wrong results. Consider re-scaling the input if possible or computing using a"
+            f"larger dtype (currently using {var_x.dtype}).",
+            UserWarning,
+            stacklevel=2,
+        )
+    corrcoef = (corr_xy / (var_x * var_y).sqrt()).squeeze()
+    return torch.clamp(corrcoef, -1.0, 1.0).to(corr_xy.dtype)
+
+
+def _final_aggregation(
+    means_x: Tensor,
+    means_y: Tensor,
+    vars_x: Tensor,
+    vars_y: Tensor,
+    corrs_xy: Tensor,
+    nbs: Tensor,
+) -> Tuple[Tensor, Tensor, Tensor, Tensor, Tensor, Tensor]:
+    """Pairwise merge of per-device (mean, var, cov, n) statistics."""
+    if len(means_x) == 1:
+        return means_x[0], means_y[0], vars_x[0], vars_y[0], corrs_xy[0], nbs[0]
+    mx1, my1, vx1, vy1, cxy1, n1 = means_x[0], means_y[0], vars_x[0], vars_y[0], corrs_xy[0], nbs[0]
+    for i in range(1, len(means_x)):
+        mx2, my2, vx2, vy2, cxy2, n2 = means_x[i], means_y[i], vars_x[i], vars_y[i], corrs_xy[i], nbs[i]
+        # count could be 0 for empty-rank corner case
+        nb = torch.where(n1 + n2 == 0, torch.ones_like(n1), n1 + n2)
+        mean_x = (n1 * mx1 + n2 * mx2) / nb
+        mean_y = (n1 * my1 + n2 * my2) / nb
+
+        # var_x
+        element_x1 = (n1 + 1) * mean_x - n1 * mx1
+        vx1 += (element_x1 - mx1) * (element_x1 - mean_x) - (element_x1 - mean_x) ** 2
+        element_x2 = (n2 + 1) * mean_x - n2 * mx2
+        vx2 += (element_x2 - mx2) * (element_x2 - mean_x) - (element_x2 - mean_x) ** 2
+        var_x = vx1 + vx2
+
+        # var_y
+        element_y1 = (n1 + 1) * mean_y - n1 * my1
+        vy1 += (element_y1 - my1) * (element_y1 - mean_y) - (element_y1 - mean_y) ** 2
+        element_y2 = (n2 + 1) * mean_y - n2 * my2
+        vy2 += (element_y2 - my2) * (element_y2 - mean_y) - (element_y2 - mean_y) ** 2
+        var_y = vy1 + vy2
+
+        # corr
+        cxy1 += (element_x1 - mx1) * (element_y1 - mean_y) - (element_x1 - mean_x) * (element_y1 - mean_y)
+        cxy2 += (element_x2 - mx2) * (element_y2 - mean_y) - (element_x2 - mean_x) * (element_y2 - mean_y)
+        corr_xy = cxy1 + cxy2
+
+        mx1, my1, vx1, vy1, cxy1, n1 = mean_x, mean_y, var_x, var_y, corr_xy, nb
+    return mean_x, mean_y, var_x, var_y, corr_xy, nb
+
+
+def pearson_corrcoef(preds: Tensor, target: Tensor) -> Tensor:
+    """Pearson correlation coefficient."""
+    d = preds.shape[1] if preds.ndim == 2 else 1
+    _temp = torch.zeros(d, dtype=preds.dtype, device=preds.device)
+    mean_x, mean_y, var_x = _temp.clone(), _temp.clone(), _temp.clone()
+    var_y, corr_xy, nb = _temp.clone(), _temp.clone(), _temp.clone()
+    _, _, var_x, var_y, corr_xy, nb = _pearson_corrcoef_update(
+        preds, target, mean_x, mean_y, var_x, var_y, corr_xy, nb, num_outputs=d
+    )
+    return _pearson_corrcoef_compute(var_x, var_y, corr_xy, nb)

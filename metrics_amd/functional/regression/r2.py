@@ -1,0 +1,87 @@
+"""R² score. Parity: torchmetrics ``functional/regression/r2.py``."""
+from __future__ import annotations
+
+from typing import Tuple, Union
+
+import torch
+from torch import Tensor
+
+from metrics_amd.utilities.checks import _check_same_shape
+from metrics_amd.utilities.prints import rank_zero_warn
+
+
+def _r2_score_update(preds: Tensor, target: Tensor) -> Tuple[Tensor, Tensor, Tensor, int]:
+    """Return (sum_y, sum_y^2, residual sum of squares, n)."""
+    _check_same_shape(preds, target)
+    if preds.ndim > 2:
+        raise ValueError(
+            f"Expected both prediction and target to be 1D or 2D tensors, but received tensors with dimension {preds.shape}"
+        )
+    sum_obs = torch.sum(target, dim=0)
+    sum_squared_obs = torch.sum(target * target, dim=0)
+    residual = target - preds
+    rss = torch.sum(residual * residual, dim=0)
+    return sum_squared_obs, sum_obs, rss, target.size(0)
+
+
+def _r2_score_compute(
+    sum_squared_obs: Tensor,
+    sum_obs: Tensor,
+    rss: Tensor,
+    num_obs: Union[int, Tensor],
+    adjusted: int = 0,
+    multioutput: str = "uniform_average",
+) -> Tensor:
+    if num_obs < 2:
+        raise ValueError("Needs at least two samples to calculate r2 score.")
+
+    mean_obs = sum_obs / num_obs
+    tss = sum_squared_obs - sum_obs * mean_obs
+
+    # Account for near constant targets
+    cond_rss = ~torch.isclose(rss, torch.zeros_like(rss), atol=1e-4)
+    cond_tss = ~torch.isclose(tss, torch.zeros_like(tss), atol=1e-4)
+    cond = cond_rss & cond_tss
+
+    raw_scores = torch.ones_like(rss)
+    raw_scores[cond] = 1 - (rss[cond] / tss[cond])
+    raw_scores[cond_rss & ~cond_tss] = 0.0
+
+    if multioutput == "raw_values":
+        r2 = raw_scores
+    elif multioutput == "uniform_average":
+        r2 = torch.mean(raw_scores)
+    elif multioutput == "variance_weighted":
+        tss_sum = torch.sum(tss)
+        r2 = torch.sum(tss / tss_sum * raw_scores)
+    else:
+        raise ValueError(
+            "Argument `multioutput` must be either `raw_values`,"
+            f" `uniform_average` or `variance_weighted`. Received {multioutput}."
+        )
+
+    if adjusted < 0 or not isinstance(adjusted, int):
+        raise ValueError("`adjusted` parameter should be an integer larger or equal to 0.")
+
+    if adjusted != 0:
+        if adjusted > num_obs - 1:
+            rank_zero_warn(
+                "More independent regressions than data points in adjusted r2 score. Falls back to standard r2 score.",
+                UserWarning,
+            )
+        elif adjusted == num_obs - 1:
+            rank_zero_warn("Division by zero in adjusted r2 score. Falls back to standard r2 score.", UserWarning)
+        else:
+            return 1 - (1 - r2) * (num_obs - 1) / (num_obs - adjusted - 1)
+    return r2
+
+
+def r2_score(
+    preds: Tensor,
+    target: Tensor,
+    adjusted: int = 0,
+    multioutput: str = "uniform_average",
+) -> Tensor:
+    """R² score."""
+    sum_squared_obs, sum_obs, rss, num_obs = _r2_score_update(preds, target)
+    return _r2_score_compute(sum_squared_obs, sum_obs, rss, num_obs, adjusted, multioutput)

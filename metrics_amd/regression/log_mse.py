@@ -1,0 +1,39 @@
+"""Modular Mean squared log error. Parity: torchmetrics ``regression/log_mse.py``."""
+from __future__ import annotations
+
+from typing import Any, Optional
+
+import torch
+from torch import Tensor
+
+from metrics_amd.metric import Metric
+from metrics_amd.functional.regression.log_mse import _mean_squared_log_error_compute, _mean_squared_log_error_update
+
+
+class MeanSquaredLogError(Metric):
+    """Mean squared log error (stateful)."""
+
+    is_differentiable = True
+    higher_is_better = False
+    full_state_update = False
+    plot_lower_bound: float = 0.0
+
+    sum_squared_log_error: Tensor
+    total: Tensor
+
+    def __init__(self, **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        self.add_state("sum_squared_log_error", default=torch.tensor(0.0), dist_reduce_fx="sum")
+        self.add_state("total", default=torch.tensor(0.0), dist_reduce_fx="sum")
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate squared log errors."""
+        sum_squared_log_error, num_obs = _mean_squared_log_error_update(preds, target)
+        self.sum_squared_log_error = self.sum_squared_log_error + sum_squared_log_error
+        self.total = self.total + num_obs
+
+    def compute(self) -> Tensor:
+        return _mean_squared_log_error_compute(self.sum_squared_log_error, self.total)
+
+    def plot(self, val=None, ax=None):
+        return self._plot(val, ax)

@@ -1,0 +1,42 @@
+"""Modular Mean absolute error. Parity: torchmetrics ``regression/mae.py``."""
+from __future__ import annotations
+
+from typing import Any, Optional
+
+import torch
+from torch import Tensor
+
+from metrics_amd.metric import Metric
+from metrics_amd.functional.regression.mae import _mean_absolute_error_compute, _mean_absolute_error_update
+
+
+class MeanAbsoluteError(Metric):
+    """Mean absolute error (stateful)."""
+
+    is_differentiable = True
+    higher_is_better = False
+    full_state_update = False
+    plot_lower_bound: float = 0.0
+
+    sum_abs_error: Tensor
+    total: Tensor
+
+    def __init__(self, num_outputs: int = 1, **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        if not (isinstance(num_outputs, int) and num_outputs > 0):
+            raise ValueError(f"Expected num_outputs to be a positive integer but got {num_outputs}")
+        self.num_outputs = num_outputs
+        self.add_state("sum_abs_error", default=torch.zeros(num_outputs), dist_reduce_fx="sum")
+        self.add_state("total", default=torch.tensor(0), dist_reduce_fx="sum")
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate absolute errors."""
+        sum_abs_error, num_obs = _mean_absolute_error_update(preds, target, self.num_outputs)
+        self.sum_abs_error = self.sum_abs_error + sum_abs_error
+        self.total = self.total + num_obs
+
+    def compute(self) -> Tensor:
+        return _mean_absolute_error_compute(self.sum_abs_error, self.total)
+
+    def plot(self, val=None, ax=None):
+        return self._plot(val, ax)
