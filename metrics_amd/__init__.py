@@ -26,9 +26,12 @@ from metrics_amd.aggregation import (  # noqa: E402
 )
 from metrics_amd.collections import MetricCollection  # noqa: E402
 from metrics_amd.metric import CompositionalMetric, Metric  # noqa: E402
-from metrics_amd import classification, functional, ops, regression, utilities, wrappers  # noqa: E402
+from metrics_amd import classification, clustering, functional, nominal, ops, regression, retrieval, utilities, wrappers  # noqa: E402
 from metrics_amd.classification import *  # noqa: E402,F401,F403
 from metrics_amd.regression import *  # noqa: E402,F401,F403
+from metrics_amd.retrieval import *  # noqa: E402,F401,F403
+from metrics_amd.clustering import *  # noqa: E402,F401,F403
+from metrics_amd.nominal import *  # noqa: E402,F401,F403
 
 __all__ = [
     "CatMetric",
@@ -50,3 +53,6 @@ __all__ = [
 ]
 __all__ += classification.__all__
 __all__ += regression.__all__
+__all__ += retrieval.__all__
+__all__ += clustering.__all__
+__all__ += nominal.__all__

@@ -1,4 +1,14 @@
 """Functional metric API (L3). Parity: torchmetrics ``functional/__init__.py``."""
 from metrics_amd.functional.classification import *  # noqa: F401,F403
 from metrics_amd.functional.regression import *  # noqa: F401,F403
-from metrics_amd.functional import classification, regression  # noqa: F401
+from metrics_amd.functional.retrieval import *  # noqa: F401,F403
+from metrics_amd.functional.clustering import *  # noqa: F401,F403
+from metrics_amd.functional.nominal import *  # noqa: F401,F403
+from metrics_amd.functional.pairwise import (  # noqa: F401
+    pairwise_cosine_similarity,
+    pairwise_euclidean_distance,
+    pairwise_linear_similarity,
+    pairwise_manhattan_distance,
+    pairwise_minkowski_distance,
+)
+from metrics_amd.functional import classification, clustering, nominal, pairwise, regression, retrieval  # noqa: F401

@@ -1,0 +1,107 @@
+"""Pairwise similarity/distance functions.
+
+Parity: torchmetrics ``functional/pairwise/*`` — matmul-shaped, so these ride
+hipBLASLt via torch.matmul (fp32 accumulate through _safe_matmul).
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+from torch import Tensor
+
+from metrics_amd.utilities.compute import _safe_matmul
+
+
+def _check_input(x: Tensor, y: Optional[Tensor], zero_diagonal: Optional[bool]) -> tuple:
+    if x.ndim != 2:
+        raise ValueError(f"Expected argument `x` to be a 2D tensor of shape `[N, d]` but got {x.shape}")
+    if y is not None:
+        if y.ndim != 2 or y.shape[1] != x.shape[1]:
+            raise ValueError(
+                "Expected argument `y` to be a 2D tensor of shape `[M, d]` where"
+                " `d` should be same as the last dimension of `x`"
+            )
+        zero_diagonal = False if zero_diagonal is None else zero_diagonal
+    else:
+        y = x.clone()
+        zero_diagonal = True if zero_diagonal is None else zero_diagonal
+    return x, y, zero_diagonal
+
+
+def _reduce_distance_matrix(distmat: Tensor, reduction: Optional[str] = None) -> Tensor:
+    if reduction == "mean":
+        return distmat.mean(dim=-1)
+    if reduction == "sum":
+        return distmat.sum(dim=-1)
+    if reduction is None or reduction == "none":
+        return distmat
+    raise ValueError(f"Expected reduction to be one of `['mean', 'sum', None]` but got {reduction}")
+
+
+def pairwise_cosine_similarity(
+    x: Tensor, y: Optional[Tensor] = None, reduction: Optional[str] = None, zero_diagonal: Optional[bool] = None
+) -> Tensor:
+    """Pairwise cosine similarity (N, M)."""
+    x, y, zero_diagonal = _check_input(x, y, zero_diagonal)
+    norm = torch.norm(x, p=2, dim=1)
+    x = x / norm.unsqueeze(1)
+    norm = torch.norm(y, p=2, dim=1)
+    y = y / norm.unsqueeze(1)
+    distance = _safe_matmul(x, y.T)
+    if zero_diagonal:
+        distance.fill_diagonal_(0)
+    return _reduce_distance_matrix(distance, reduction)
+
+
+def pairwise_euclidean_distance(
+    x: Tensor, y: Optional[Tensor] = None, reduction: Optional[str] = None, zero_diagonal: Optional[bool] = None
+) -> Tensor:
+    """Pairwise euclidean distance (N, M)."""
+    x, y, zero_diagonal = _check_input(x, y, zero_diagonal)
+    # upcast to float64 to prevent precision issues
+    _orig_dtype = x.dtype
+    x = x.to(torch.float64)
+    y = y.to(torch.float64)
+    x_norm = (x * x).sum(dim=1, keepdim=True)
+    y_norm = (y * y).sum(dim=1)
+    distance = (x_norm + y_norm - 2 * x.mm(y.T)).to(_orig_dtype)
+    if zero_diagonal:
+        distance.fill_diagonal_(0)
+    return _reduce_distance_matrix(distance.sqrt(), reduction)
+
+
+def pairwise_linear_similarity(
+    x: Tensor, y: Optional[Tensor] = None, reduction: Optional[str] = None, zero_diagonal: Optional[bool] = None
+) -> Tensor:
+    """Pairwise linear similarity x·yᵀ (N, M)."""
+    x, y, zero_diagonal = _check_input(x, y, zero_diagonal)
+    distance = _safe_matmul(x, y.T)
+    if zero_diagonal:
+        distance.fill_diagonal_(0)
+    return _reduce_distance_matrix(distance, reduction)
+
+
+def pairwise_manhattan_distance(
+    x: Tensor, y: Optional[Tensor] = None, reduction: Optional[str] = None, zero_diagonal: Optional[bool] = None
+) -> Tensor:
+    """Pairwise manhattan distance (N, M)."""
+    x, y, zero_diagonal = _check_input(x, y, zero_diagonal)
+    distance = (x.unsqueeze(1) - y.unsqueeze(0).repeat(x.shape[0], 1, 1)).abs().sum(dim=-1)
+    if zero_diagonal:
+        distance.fill_diagonal_(0)
+    return _reduce_distance_matrix(distance, reduction)
+
+
+def pairwise_minkowski_distance(
+    x: Tensor, y: Optional[Tensor] = None, exponent: float = 2, reduction: Optional[str] = None,
+    zero_diagonal: Optional[bool] = None,
+) -> Tensor:
+    """Pairwise minkowski distance of order ``exponent`` (N, M)."""
+    x, y, zero_diagonal = _check_input(x, y, zero_diagonal)
+    if not (isinstance(exponent, (float, int)) and exponent > 0):
+        raise ValueError(f"Argument `exponent` must be a positive int or float but got {exponent}")
+    distance = (x.unsqueeze(1) - y.unsqueeze(0)).abs().pow(exponent).sum(-1).pow(1.0 / exponent)
+    if zero_diagonal:
+        distance.fill_diagonal_(0)
+    return _reduce_distance_matrix(distance, reduction)

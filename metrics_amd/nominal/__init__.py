@@ -1,0 +1,140 @@
+"""Modular nominal metrics. Parity: torchmetrics ``nominal/*``."""
+from __future__ import annotations
+
+from typing import Any, List, Optional
+
+import torch
+from torch import Tensor
+
+from metrics_amd.metric import Metric
+from metrics_amd.utilities.data import dim_zero_cat
+from metrics_amd.functional.nominal.metrics import (
+    cramers_v,
+    fleiss_kappa,
+    pearsons_contingency_coefficient,
+    theils_u,
+    tschuprows_t,
+)
+
+
+class _NominalBase(Metric):
+    """Accumulate (preds, target) label tensors; score at compute."""
+
+    is_differentiable = False
+    higher_is_better = True
+    full_state_update: bool = False
+    plot_lower_bound: float = 0.0
+    plot_upper_bound: float = 1.0
+
+    preds: List[Tensor]
+    target: List[Tensor]
+
+    def __init__(self, nan_strategy: str = "replace", nan_replace_value: Optional[float] = 0.0, **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        from metrics_amd.functional.nominal.metrics import _nominal_input_validation
+
+        _nominal_input_validation(nan_strategy, nan_replace_value)
+        self.nan_strategy = nan_strategy
+        self.nan_replace_value = nan_replace_value
+        self.add_state("preds", default=[], dist_reduce_fx="cat")
+        self.add_state("target", default=[], dist_reduce_fx="cat")
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Append label tensors (2D inputs are argmaxed)."""
+        self.preds.append(preds.argmax(1) if preds.ndim == 2 else preds)
+        self.target.append(target.argmax(1) if target.ndim == 2 else target)
+
+    def plot(self, val=None, ax=None):
+        return self._plot(val, ax)
+
+
+class CramersV(_NominalBase):
+    """Cramer's V (stateful)."""
+
+    def __init__(self, num_classes: Optional[int] = None, bias_correction: bool = True,
+                 nan_strategy: str = "replace", nan_replace_value: Optional[float] = 0.0, **kwargs: Any) -> None:
+        super().__init__(nan_strategy, nan_replace_value, **kwargs)
+        self.num_classes = num_classes
+        self.bias_correction = bias_correction
+
+    def compute(self) -> Tensor:
+        return cramers_v(
+            dim_zero_cat(self.preds), dim_zero_cat(self.target), self.bias_correction,
+            self.nan_strategy, self.nan_replace_value,
+        )
+
+
+class PearsonsContingencyCoefficient(_NominalBase):
+    """Pearson's contingency coefficient (stateful)."""
+
+    def __init__(self, num_classes: Optional[int] = None, nan_strategy: str = "replace",
+                 nan_replace_value: Optional[float] = 0.0, **kwargs: Any) -> None:
+        super().__init__(nan_strategy, nan_replace_value, **kwargs)
+        self.num_classes = num_classes
+
+    def compute(self) -> Tensor:
+        return pearsons_contingency_coefficient(
+            dim_zero_cat(self.preds), dim_zero_cat(self.target), self.nan_strategy, self.nan_replace_value
+        )
+
+
+class TschuprowsT(_NominalBase):
+    """Tschuprow's T (stateful)."""
+
+    def __init__(self, num_classes: Optional[int] = None, bias_correction: bool = True,
+                 nan_strategy: str = "replace", nan_replace_value: Optional[float] = 0.0, **kwargs: Any) -> None:
+        super().__init__(nan_strategy, nan_replace_value, **kwargs)
+        self.num_classes = num_classes
+        self.bias_correction = bias_correction
+
+    def compute(self) -> Tensor:
+        return tschuprows_t(
+            dim_zero_cat(self.preds), dim_zero_cat(self.target), self.bias_correction,
+            self.nan_strategy, self.nan_replace_value,
+        )
+
+
+class TheilsU(_NominalBase):
+    """Theil's U (stateful)."""
+
+    def __init__(self, num_classes: Optional[int] = None, nan_strategy: str = "replace",
+                 nan_replace_value: Optional[float] = 0.0, **kwargs: Any) -> None:
+        super().__init__(nan_strategy, nan_replace_value, **kwargs)
+        self.num_classes = num_classes
+
+    def compute(self) -> Tensor:
+        return theils_u(
+            dim_zero_cat(self.preds), dim_zero_cat(self.target), self.nan_strategy, self.nan_replace_value
+        )
+
+
+class FleissKappa(Metric):
+    """Fleiss' kappa (stateful)."""
+
+    is_differentiable = False
+    higher_is_better = True
+    full_state_update: bool = False
+    plot_lower_bound: float = 0.0
+    plot_upper_bound: float = 1.0
+
+    ratings: List[Tensor]
+
+    def __init__(self, mode: str = "counts", **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        if mode not in ("counts", "probs"):
+            raise ValueError("Argument `mode` should be one of 'counts' or 'probs'")
+        self.mode = mode
+        self.add_state("ratings", default=[], dist_reduce_fx="cat")
+
+    def update(self, ratings: Tensor) -> None:
+        """Append a ratings matrix."""
+        self.ratings.append(ratings)
+
+    def compute(self) -> Tensor:
+        return fleiss_kappa(dim_zero_cat(self.ratings), self.mode)
+
+    def plot(self, val=None, ax=None):
+        return self._plot(val, ax)
+
+
+__all__ = ["CramersV", "FleissKappa", "PearsonsContingencyCoefficient", "TheilsU", "TschuprowsT"]

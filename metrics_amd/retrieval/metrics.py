@@ -1,0 +1,218 @@
+"""Modular retrieval metrics. Parity: torchmetrics ``retrieval/*``."""
+from __future__ import annotations
+
+from typing import Any, Optional, Tuple
+
+import torch
+from torch import Tensor
+
+from metrics_amd.retrieval.base import RetrievalMetric
+from metrics_amd.utilities.data import _flexible_bincount, dim_zero_cat
+from metrics_amd.functional.retrieval.metrics import (
+    retrieval_auroc,
+    retrieval_average_precision,
+    retrieval_fall_out,
+    retrieval_hit_rate,
+    retrieval_normalized_dcg,
+    retrieval_precision,
+    retrieval_precision_recall_curve,
+    retrieval_r_precision,
+    retrieval_recall,
+    retrieval_reciprocal_rank,
+)
+
+
+class _TopKRetrievalMetric(RetrievalMetric):
+    """Shared top_k handling."""
+
+    def __init__(self, empty_target_action: str = "neg", ignore_index: Optional[int] = None,
+                 top_k: Optional[int] = None, **kwargs: Any) -> None:
+        super().__init__(empty_target_action=empty_target_action, ignore_index=ignore_index, **kwargs)
+        if top_k is not None and not (isinstance(top_k, int) and top_k > 0):
+            raise ValueError("`top_k` has to be a positive integer or None")
+        self.top_k = top_k
+
+
+class RetrievalMAP(_TopKRetrievalMetric):
+    """Mean average precision."""
+
+    def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
+        return retrieval_average_precision(preds, target, top_k=self.top_k)
+
+
+class RetrievalMRR(_TopKRetrievalMetric):
+    """Mean reciprocal rank."""
+
+    def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
+        return retrieval_reciprocal_rank(preds, target, top_k=self.top_k)
+
+
+class RetrievalPrecision(_TopKRetrievalMetric):
+    """Precision@k."""
+
+    def __init__(self, empty_target_action: str = "neg", ignore_index: Optional[int] = None,
+                 top_k: Optional[int] = None, adaptive_k: bool = False, **kwargs: Any) -> None:
+        super().__init__(empty_target_action, ignore_index, top_k, **kwargs)
+        if not isinstance(adaptive_k, bool):
+            raise ValueError("`adaptive_k` has to be a boolean")
+        self.adaptive_k = adaptive_k
+
+    def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
+        return retrieval_precision(preds, target, top_k=self.top_k, adaptive_k=self.adaptive_k)
+
+
+class RetrievalRecall(_TopKRetrievalMetric):
+    """Recall@k."""
+
+    def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
+        return retrieval_recall(preds, target, top_k=self.top_k)
+
+
+class RetrievalHitRate(_TopKRetrievalMetric):
+    """Hit rate@k."""
+
+    def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
+        return retrieval_hit_rate(preds, target, top_k=self.top_k)
+
+
+class RetrievalFallOut(_TopKRetrievalMetric):
+    """Fall-out@k (lower is better; empty_target_action applies to queries with no NEGATIVES)."""
+
+    higher_is_better = False
+
+    def compute(self) -> Tensor:
+        indexes = dim_zero_cat(self.indexes)
+        preds = dim_zero_cat(self.preds)
+        target = dim_zero_cat(self.target)
+
+        indexes, indices = torch.sort(indexes)
+        preds = preds[indices]
+        target = target[indices]
+        split_sizes = _flexible_bincount(indexes).detach().cpu().tolist()
+
+        res = []
+        for mini_preds, mini_target in zip(
+            torch.split(preds, split_sizes, dim=0), torch.split(target, split_sizes, dim=0)
+        ):
+            if (1 - mini_target).sum() == 0:
+                if self.empty_target_action == "error":
+                    raise ValueError("`compute` method was provided with a query with no negative target.")
+                if self.empty_target_action == "pos":
+                    res.append(torch.tensor(1.0, device=preds.device))
+                elif self.empty_target_action == "neg":
+                    res.append(torch.tensor(0.0, device=preds.device))
+            else:
+                res.append(self._metric(mini_preds, mini_target))
+        return torch.stack([x.to(preds) for x in res]).mean() if res else torch.tensor(0.0).to(preds)
+
+    def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
+        return retrieval_fall_out(preds, target, top_k=self.top_k)
+
+
+class RetrievalNormalizedDCG(_TopKRetrievalMetric):
+    """Normalized DCG (graded relevance allowed)."""
+
+    def __init__(self, empty_target_action: str = "neg", ignore_index: Optional[int] = None,
+                 top_k: Optional[int] = None, **kwargs: Any) -> None:
+        super().__init__(empty_target_action, ignore_index, top_k, **kwargs)
+        self.allow_non_binary_target = True
+
+    def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
+        return retrieval_normalized_dcg(preds, target, top_k=self.top_k)
+
+
+class RetrievalRPrecision(RetrievalMetric):
+    """R-precision."""
+
+    def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
+        return retrieval_r_precision(preds, target)
+
+
+class RetrievalAUROC(_TopKRetrievalMetric):
+    """Per-query AUROC, averaged."""
+
+    def __init__(self, empty_target_action: str = "neg", ignore_index: Optional[int] = None,
+                 top_k: Optional[int] = None, max_fpr: Optional[float] = None, **kwargs: Any) -> None:
+        super().__init__(empty_target_action, ignore_index, top_k, **kwargs)
+        self.max_fpr = max_fpr
+
+    def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
+        return retrieval_auroc(preds, target, top_k=self.top_k, max_fpr=self.max_fpr)
+
+
+class RetrievalPrecisionRecallCurve(RetrievalMetric):
+    """Averaged precision/recall at k = 1..max_k."""
+
+    def __init__(self, max_k: Optional[int] = None, adaptive_k: bool = False,
+                 empty_target_action: str = "neg", ignore_index: Optional[int] = None, **kwargs: Any) -> None:
+        super().__init__(empty_target_action=empty_target_action, ignore_index=ignore_index, **kwargs)
+        if max_k is not None and not (isinstance(max_k, int) and max_k > 0):
+            raise ValueError("`max_k` has to be a positive integer or None")
+        self.max_k = max_k
+        if not isinstance(adaptive_k, bool):
+            raise ValueError("`adaptive_k` has to be a boolean")
+        self.adaptive_k = adaptive_k
+
+    def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
+        raise NotImplementedError
+
+    def compute(self) -> Tuple[Tensor, Tensor, Tensor]:
+        indexes = dim_zero_cat(self.indexes)
+        preds = dim_zero_cat(self.preds)
+        target = dim_zero_cat(self.target)
+
+        indexes, indices = torch.sort(indexes)
+        preds = preds[indices]
+        target = target[indices]
+        split_sizes = _flexible_bincount(indexes).detach().cpu().tolist()
+
+        max_k = self.max_k or max(split_sizes)
+        precisions, recalls = [], []
+        for mini_preds, mini_target in zip(
+            torch.split(preds, split_sizes, dim=0), torch.split(target, split_sizes, dim=0)
+        ):
+            if not mini_target.sum():
+                if self.empty_target_action == "error":
+                    raise ValueError("`compute` method was provided with a query with no positive target.")
+                if self.empty_target_action == "skip":
+                    continue
+                fill = 1.0 if self.empty_target_action == "pos" else 0.0
+                precisions.append(torch.full((max_k,), fill, device=preds.device))
+                recalls.append(torch.full((max_k,), fill, device=preds.device))
+            else:
+                p, r, _ = retrieval_precision_recall_curve(mini_preds, mini_target, max_k, self.adaptive_k)
+                precisions.append(p)
+                recalls.append(r)
+        top_k = torch.arange(1, max_k + 1, device=preds.device)
+        if not precisions:
+            return torch.zeros(max_k, device=preds.device), torch.zeros(max_k, device=preds.device), top_k
+        return torch.stack(precisions).mean(0), torch.stack(recalls).mean(0), top_k
+
+
+class RetrievalRecallAtFixedPrecision(RetrievalMetric):
+    """Max recall@k such that precision@k >= min_precision; returns (recall, k)."""
+
+    higher_is_better = True
+
+    def __init__(self, min_precision: float = 0.0, max_k: Optional[int] = None, adaptive_k: bool = False,
+                 empty_target_action: str = "neg", ignore_index: Optional[int] = None, **kwargs: Any) -> None:
+        super().__init__(empty_target_action=empty_target_action, ignore_index=ignore_index, **kwargs)
+        if not (isinstance(min_precision, float) and 0.0 <= min_precision <= 1.0):
+            raise ValueError("`min_precision` has to be a positive float between 0 and 1")
+        self.min_precision = min_precision
+        self.max_k = max_k
+        self.adaptive_k = adaptive_k
+
+    def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
+        raise NotImplementedError
+
+    def compute(self) -> Tuple[Tensor, Tensor]:
+        curve = RetrievalPrecisionRecallCurve.compute(self)  # reuse grouping logic
+        precisions, recalls, top_k = curve
+        condition = precisions >= self.min_precision
+        if condition.any():
+            recalls_at = recalls[condition]
+            best = recalls_at.max()
+            k = top_k[condition][recalls_at.argmax()]
+            return best, k
+        return torch.tensor(0.0, device=precisions.device), top_k[-1]
