@@ -26,7 +26,7 @@ from metrics_amd.aggregation import (  # noqa: E402
 )
 from metrics_amd.collections import MetricCollection  # noqa: E402
 from metrics_amd.metric import CompositionalMetric, Metric  # noqa: E402
-from metrics_amd import classification, clustering, detection, functional, nominal, ops, regression, retrieval, segmentation, utilities, wrappers  # noqa: E402
+from metrics_amd import classification, clustering, detection, functional, image, nominal, ops, regression, retrieval, segmentation, utilities, wrappers  # noqa: E402
 from metrics_amd.classification import *  # noqa: E402,F401,F403
 from metrics_amd.regression import *  # noqa: E402,F401,F403
 from metrics_amd.retrieval import *  # noqa: E402,F401,F403
@@ -34,6 +34,7 @@ from metrics_amd.clustering import *  # noqa: E402,F401,F403
 from metrics_amd.nominal import *  # noqa: E402,F401,F403
 from metrics_amd.detection import *  # noqa: E402,F401,F403
 from metrics_amd.segmentation import *  # noqa: E402,F401,F403
+from metrics_amd.image import *  # noqa: E402,F401,F403
 
 __all__ = [
     "CatMetric",
@@ -60,3 +61,4 @@ __all__ += clustering.__all__
 __all__ += nominal.__all__
 __all__ += detection.__all__
 __all__ += segmentation.__all__
+__all__ += image.__all__

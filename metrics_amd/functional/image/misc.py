@@ -1,0 +1,235 @@
+"""Image metrics: UQI, SAM, ERGAS, total variation, RMSE-SW, SCC, RASE, VIF.
+
+Parity: torchmetrics ``functional/image/{uqi,sam,ergas,tv,rmse_sw,scc,rase,vif}.py``.
+"""
+from __future__ import annotations
+
+from typing import Optional, Sequence, Tuple, Union
+
+import torch
+import torch.nn.functional as F
+from torch import Tensor
+
+from metrics_amd.utilities.checks import _check_same_shape
+from metrics_amd.utilities.distributed import reduce
+from metrics_amd.functional.image.ssim import _gaussian_kernel_2d
+
+
+def _image_check(preds: Tensor, target: Tensor) -> Tuple[Tensor, Tensor]:
+    if preds.dtype != target.dtype:
+        target = target.to(preds.dtype)
+    _check_same_shape(preds, target)
+    if preds.ndim != 4:
+        raise ValueError(f"Expected `preds` and `target` to have BxCxHxW shape. Got preds: {preds.shape}.")
+    return preds, target
+
+
+def universal_image_quality_index(
+    preds: Tensor,
+    target: Tensor,
+    kernel_size: Sequence[int] = (11, 11),
+    sigma: Sequence[float] = (1.5, 1.5),
+    reduction: Optional[str] = "elementwise_mean",
+) -> Tensor:
+    """Universal image quality index (SSIM with C1 = C2 = 0)."""
+    preds, target = _image_check(preds, target)
+    channel = preds.size(1)
+    kernel = _gaussian_kernel_2d(channel, kernel_size, sigma, preds.dtype, preds.device)
+    pad_h = (kernel_size[0] - 1) // 2
+    pad_w = (kernel_size[1] - 1) // 2
+    preds_p = F.pad(preds, (pad_w, pad_w, pad_h, pad_h), mode="reflect")
+    target_p = F.pad(target, (pad_w, pad_w, pad_h, pad_h), mode="reflect")
+
+    input_list = torch.cat((preds_p, target_p, preds_p * preds_p, target_p * target_p, preds_p * target_p))
+    outputs = F.conv2d(input_list, kernel, groups=channel)
+    output_list = outputs.split(preds.shape[0])
+
+    mu_pred_sq = output_list[0].pow(2)
+    mu_target_sq = output_list[1].pow(2)
+    mu_pred_target = output_list[0] * output_list[1]
+
+    sigma_pred_sq = output_list[2] - mu_pred_sq
+    sigma_target_sq = output_list[3] - mu_target_sq
+    sigma_pred_target = output_list[4] - mu_pred_target
+
+    upper = 2 * sigma_pred_target
+    lower = sigma_pred_sq + sigma_target_sq
+    eps = torch.finfo(preds.dtype).eps
+    uqi_idx = ((2 * mu_pred_target) * upper) / ((mu_pred_sq + mu_target_sq) * lower + eps)
+    uqi_idx = uqi_idx[..., pad_h:-pad_h, pad_w:-pad_w]
+    return reduce(uqi_idx.reshape(uqi_idx.shape[0], -1).mean(-1), reduction or "none")
+
+
+def spectral_angle_mapper(
+    preds: Tensor,
+    target: Tensor,
+    reduction: Optional[str] = "elementwise_mean",
+) -> Tensor:
+    """Spectral angle (radians) between pred/target spectra at each pixel."""
+    preds, target = _image_check(preds, target)
+    if preds.shape[1] <= 1:
+        raise ValueError(f"Expected channel dimension of `preds` and `target` to be larger than 1. Got preds: {preds.shape[1]}.")
+    dot_product = (preds * target).sum(dim=1)
+    preds_norm = preds.norm(dim=1)
+    target_norm = target.norm(dim=1)
+    sam_score = torch.clamp(dot_product / (preds_norm * target_norm), -1, 1).acos()
+    return reduce(sam_score, reduction or "none")
+
+
+def error_relative_global_dimensionless_synthesis(
+    preds: Tensor,
+    target: Tensor,
+    ratio: float = 4,
+    reduction: Optional[str] = "elementwise_mean",
+) -> Tensor:
+    """ERGAS for pan-sharpening quality."""
+    preds, target = _image_check(preds, target)
+    b, c, h, w = preds.shape
+    preds = preds.reshape(b, c, h * w)
+    target = target.reshape(b, c, h * w)
+
+    diff = preds - target
+    sum_squared_error = torch.sum(diff * diff, dim=2)
+    rmse_per_band = torch.sqrt(sum_squared_error / (h * w))
+    mean_target = torch.mean(target, dim=2)
+
+    ergas_score = 100 / ratio * torch.sqrt(torch.sum(rmse_per_band**2 / mean_target**2, dim=1) / c)
+    return reduce(ergas_score, reduction or "none")
+
+
+def relative_average_spectral_error(preds: Tensor, target: Tensor, window_size: int = 8) -> Tensor:
+    """RASE: relative average spectral error using sliding-window RMSE."""
+    preds, target = _image_check(preds, target)
+    _, rmse_map = _rmse_sw_maps(preds, target, window_size)
+    target_mean = target.mean(dim=(0, 2, 3), keepdim=True)  # mean over batch and spatial
+    rase_map = 100.0 / target_mean.squeeze() * torch.sqrt((rmse_map**2).mean(dim=1))
+    return rase_map.mean() if rase_map.numel() else rase_map
+
+
+def _rmse_sw_maps(preds: Tensor, target: Tensor, window_size: int) -> Tuple[Tensor, Tensor]:
+    kernel = torch.ones(preds.shape[1], 1, window_size, window_size, dtype=preds.dtype, device=preds.device) / (
+        window_size**2
+    )
+    diff_sq = (preds - target) ** 2
+    mse_map = F.conv2d(diff_sq, kernel, groups=preds.shape[1])
+    rmse_map = torch.sqrt(mse_map)
+    rmse_mean = rmse_map.mean()
+    return rmse_mean, rmse_map
+
+
+def root_mean_squared_error_using_sliding_window(
+    preds: Tensor, target: Tensor, window_size: int = 8, return_rmse_map: bool = False
+):
+    """RMSE averaged over sliding windows."""
+    preds, target = _image_check(preds, target)
+    if not isinstance(window_size, int) or window_size < 1:
+        raise ValueError(f"Argument `window_size` is expected to be a positive integer, but got {window_size}")
+    rmse_mean, rmse_map = _rmse_sw_maps(preds, target, window_size)
+    if return_rmse_map:
+        return rmse_mean, rmse_map
+    return rmse_mean
+
+
+def total_variation(img: Tensor, reduction: Optional[str] = "sum") -> Tensor:
+    """Total variation: sum of absolute spatial gradients."""
+    if img.ndim != 4:
+        raise RuntimeError(f"Expected input `img` to be an 4D tensor, but got {img.shape}")
+    diff1 = img[..., 1:, :] - img[..., :-1, :]
+    diff2 = img[..., :, 1:] - img[..., :, :-1]
+    res1 = diff1.abs().sum([1, 2, 3])
+    res2 = diff2.abs().sum([1, 2, 3])
+    score = res1 + res2
+    if reduction == "mean":
+        return score.mean()
+    if reduction == "sum":
+        return score.sum()
+    if reduction is None or reduction == "none":
+        return score
+    raise ValueError("Expected argument `reduction` to either be 'sum', 'mean', 'none' or None")
+
+
+def spatial_correlation_coefficient(
+    preds: Tensor,
+    target: Tensor,
+    hp_filter: Optional[Tensor] = None,
+    window_size: int = 8,
+    reduction: Optional[str] = "elementwise_mean",
+) -> Tensor:
+    """Spatial correlation coefficient after high-pass filtering."""
+    if preds.ndim == 3:
+        preds = preds.unsqueeze(1)
+        target = target.unsqueeze(1)
+    preds, target = _image_check(preds.float(), target.float())
+    if hp_filter is None:
+        hp_filter = torch.tensor([[-1.0, -1.0, -1.0], [-1.0, 8.0, -1.0], [-1.0, -1.0, -1.0]], device=preds.device)
+    c = preds.shape[1]
+    kern = hp_filter.to(preds).expand(c, 1, *hp_filter.shape)
+    pad = hp_filter.shape[-1] // 2
+    hp_preds = F.conv2d(F.pad(preds, [pad] * 4, mode="reflect"), kern, groups=c)
+    hp_target = F.conv2d(F.pad(target, [pad] * 4, mode="reflect"), kern, groups=c)
+
+    # windowed means
+    win = torch.ones(c, 1, window_size, window_size, device=preds.device, dtype=preds.dtype) / window_size**2
+    mu_x = F.conv2d(hp_preds, win, groups=c, stride=window_size)
+    mu_y = F.conv2d(hp_target, win, groups=c, stride=window_size)
+    mu_xx = F.conv2d(hp_preds * hp_preds, win, groups=c, stride=window_size)
+    mu_yy = F.conv2d(hp_target * hp_target, win, groups=c, stride=window_size)
+    mu_xy = F.conv2d(hp_preds * hp_target, win, groups=c, stride=window_size)
+
+    var_x = mu_xx - mu_x**2
+    var_y = mu_yy - mu_y**2
+    cov = mu_xy - mu_x * mu_y
+    denom = torch.sqrt(var_x.clamp(min=0)) * torch.sqrt(var_y.clamp(min=0))
+    scc = torch.where(denom > 0, cov / denom, torch.zeros_like(cov))
+    per_image = scc.mean(dim=(1, 2, 3))
+    return reduce(per_image, reduction or "none")
+
+
+def visual_information_fidelity(preds: Tensor, target: Tensor, sigma_n_sq: float = 2.0) -> Tensor:
+    """VIF-P (pixel-domain visual information fidelity), averaged over batch."""
+    preds, target = _image_check(preds.float(), target.float())
+    if preds.shape[-1] < 41 or preds.shape[-2] < 41:
+        raise ValueError("Invalid size of preds. Expected at least 41x41")
+    eps = torch.finfo(preds.dtype).eps
+    b, c = preds.shape[:2]
+    preds = preds.reshape(b * c, 1, *preds.shape[2:])
+    target = target.reshape(b * c, 1, *target.shape[2:])
+
+    num = torch.zeros(b * c, device=preds.device)
+    den = torch.zeros(b * c, device=preds.device)
+    for scale in range(1, 5):
+        n = 2 ** (4 - scale + 1) + 1
+        sd = n / 5.0
+        kernel = _gaussian_kernel_2d(1, (n, n), (sd, sd), preds.dtype, preds.device)
+        if scale > 1:
+            target = F.conv2d(target, kernel)[:, :, ::2, ::2]
+            preds = F.conv2d(preds, kernel)[:, :, ::2, ::2]
+
+        mu1 = F.conv2d(target, kernel)
+        mu2 = F.conv2d(preds, kernel)
+        mu1_sq, mu2_sq, mu1_mu2 = mu1 * mu1, mu2 * mu2, mu1 * mu2
+        sigma1_sq = F.conv2d(target * target, kernel) - mu1_sq
+        sigma2_sq = F.conv2d(preds * preds, kernel) - mu2_sq
+        sigma12 = F.conv2d(target * preds, kernel) - mu1_mu2
+
+        sigma1_sq = sigma1_sq.clamp(min=0)
+        sigma2_sq = sigma2_sq.clamp(min=0)
+
+        g = sigma12 / (sigma1_sq + eps)
+        sv_sq = sigma2_sq - g * sigma12
+
+        g = torch.where(sigma1_sq >= eps, g, torch.zeros_like(g))
+        sv_sq = torch.where(sigma1_sq >= eps, sv_sq, sigma2_sq)
+        sigma1_sq = torch.where(sigma1_sq >= eps, sigma1_sq, torch.zeros_like(sigma1_sq))
+
+        g = torch.where(sigma2_sq >= eps, g, torch.zeros_like(g))
+        sv_sq = torch.where(sigma2_sq >= eps, sv_sq, torch.zeros_like(sv_sq))
+
+        sv_sq = torch.where(g >= 0, sv_sq, sigma2_sq)
+        g = g.clamp(min=0)
+        sv_sq = sv_sq.clamp(min=eps)
+
+        num += torch.sum(torch.log10(1.0 + (g**2.0) * sigma1_sq / (sv_sq + sigma_n_sq)), dim=(1, 2, 3))
+        den += torch.sum(torch.log10(1.0 + sigma1_sq / sigma_n_sq), dim=(1, 2, 3))
+
+    return (num / den).reshape(b, c).mean()

@@ -1,0 +1,220 @@
+"""SSIM / MS-SSIM. Parity: torchmetrics ``functional/image/ssim.py``.
+
+Gaussian/uniform windows as separable grouped convs (MIOpen on ROCm).
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence, Tuple, Union
+
+import torch
+import torch.nn.functional as F
+from torch import Tensor
+
+from metrics_amd.utilities.checks import _check_same_shape
+
+
+def _gaussian(kernel_size: int, sigma: float, dtype: torch.dtype, device: torch.device) -> Tensor:
+    dist = torch.arange((1 - kernel_size) / 2, (1 + kernel_size) / 2, step=1, dtype=dtype, device=device)
+    gauss = torch.exp(-torch.pow(dist / sigma, 2) / 2)
+    return (gauss / gauss.sum()).unsqueeze(dim=0)
+
+
+def _gaussian_kernel_2d(
+    channel: int, kernel_size: Sequence[int], sigma: Sequence[float], dtype: torch.dtype, device: torch.device
+) -> Tensor:
+    gaussian_kernel_x = _gaussian(kernel_size[0], sigma[0], dtype, device)
+    gaussian_kernel_y = _gaussian(kernel_size[1], sigma[1], dtype, device)
+    kernel = torch.matmul(gaussian_kernel_x.t(), gaussian_kernel_y)
+    return kernel.expand(channel, 1, kernel_size[0], kernel_size[1])
+
+
+def _uniform_kernel_2d(
+    channel: int, kernel_size: Sequence[int], dtype: torch.dtype, device: torch.device
+) -> Tensor:
+    kernel = torch.ones(kernel_size[0], kernel_size[1], dtype=dtype, device=device) / (
+        kernel_size[0] * kernel_size[1]
+    )
+    return kernel.expand(channel, 1, kernel_size[0], kernel_size[1])
+
+
+def _ssim_check_inputs(preds: Tensor, target: Tensor) -> Tuple[Tensor, Tensor]:
+    if preds.dtype != target.dtype:
+        target = target.to(preds.dtype)
+    _check_same_shape(preds, target)
+    if len(preds.shape) not in (4, 5):
+        raise ValueError(f"Expected `preds` and `target` to have BxCxHxW or BxCxDxHxW shape. Got preds: {preds.shape}.")
+    return preds, target
+
+
+def _ssim_compute(
+    preds: Tensor,
+    target: Tensor,
+    gaussian_kernel: bool = True,
+    sigma: Union[float, Sequence[float]] = 1.5,
+    kernel_size: Union[int, Sequence[int]] = 11,
+    data_range: Optional[Union[float, Tuple[float, float]]] = None,
+    k1: float = 0.01,
+    k2: float = 0.03,
+    return_full_image: bool = False,
+    return_contrast_sensitivity: bool = False,
+):
+    is_3d = preds.ndim == 5
+    if is_3d:
+        raise NotImplementedError("3D SSIM not yet supported in this build (2D images only)")
+
+    if not isinstance(kernel_size, Sequence):
+        kernel_size = 2 * [kernel_size]
+    if not isinstance(sigma, Sequence):
+        sigma = 2 * [sigma]
+
+    if any(x % 2 == 0 or x <= 0 for x in kernel_size):
+        raise ValueError(f"Expected `kernel_size` to have odd positive number. Got {kernel_size}.")
+    if any(y <= 0 for y in sigma):
+        raise ValueError(f"Expected `sigma` to have positive number. Got {sigma}.")
+
+    if data_range is None:
+        data_range_t = max(preds.max() - preds.min(), target.max() - target.min())
+    elif isinstance(data_range, tuple):
+        preds = torch.clamp(preds, min=data_range[0], max=data_range[1])
+        target = torch.clamp(target, min=data_range[0], max=data_range[1])
+        data_range_t = data_range[1] - data_range[0]
+    else:
+        data_range_t = data_range
+
+    c1 = pow(k1 * data_range_t, 2)
+    c2 = pow(k2 * data_range_t, 2)
+
+    channel = preds.size(1)
+    dtype = preds.dtype
+    device = preds.device
+    if gaussian_kernel:
+        kernel = _gaussian_kernel_2d(channel, kernel_size, sigma, dtype, device)
+    else:
+        kernel = _uniform_kernel_2d(channel, kernel_size, dtype, device)
+
+    pad_h = (kernel_size[0] - 1) // 2
+    pad_w = (kernel_size[1] - 1) // 2
+    preds_p = F.pad(preds, (pad_w, pad_w, pad_h, pad_h), mode="reflect")
+    target_p = F.pad(target, (pad_w, pad_w, pad_h, pad_h), mode="reflect")
+
+    input_list = torch.cat((preds_p, target_p, preds_p * preds_p, target_p * target_p, preds_p * target_p))
+    outputs = F.conv2d(input_list, kernel, groups=channel)
+    output_list = outputs.split(preds.shape[0])
+
+    mu_pred_sq = output_list[0].pow(2)
+    mu_target_sq = output_list[1].pow(2)
+    mu_pred_target = output_list[0] * output_list[1]
+
+    sigma_pred_sq = output_list[2] - mu_pred_sq
+    sigma_target_sq = output_list[3] - mu_target_sq
+    sigma_pred_target = output_list[4] - mu_pred_target
+
+    upper = 2 * sigma_pred_target.to(dtype) + c2
+    lower = (sigma_pred_sq + sigma_target_sq).to(dtype) + c2
+
+    ssim_idx_full_image = ((2 * mu_pred_target + c1) * upper) / ((mu_pred_sq + mu_target_sq + c1) * lower)
+    # crop the padded border back off
+    ssim_idx = ssim_idx_full_image[..., pad_h:-pad_h, pad_w:-pad_w]
+
+    if return_contrast_sensitivity:
+        contrast_sensitivity = upper / lower
+        contrast_sensitivity = contrast_sensitivity[..., pad_h:-pad_h, pad_w:-pad_w]
+        return ssim_idx.reshape(ssim_idx.shape[0], -1).mean(-1), contrast_sensitivity.reshape(
+            contrast_sensitivity.shape[0], -1
+        ).mean(-1)
+    if return_full_image:
+        return ssim_idx.reshape(ssim_idx.shape[0], -1).mean(-1), ssim_idx_full_image
+    return ssim_idx.reshape(ssim_idx.shape[0], -1).mean(-1)
+
+
+def structural_similarity_index_measure(
+    preds: Tensor,
+    target: Tensor,
+    gaussian_kernel: bool = True,
+    sigma: Union[float, Sequence[float]] = 1.5,
+    kernel_size: Union[int, Sequence[int]] = 11,
+    reduction: Optional[str] = "elementwise_mean",
+    data_range: Optional[Union[float, Tuple[float, float]]] = None,
+    k1: float = 0.01,
+    k2: float = 0.03,
+    return_full_image: bool = False,
+    return_contrast_sensitivity: bool = False,
+):
+    """Structural similarity index measure."""
+    preds, target = _ssim_check_inputs(preds, target)
+    out = _ssim_compute(
+        preds, target, gaussian_kernel, sigma, kernel_size, data_range, k1, k2,
+        return_full_image, return_contrast_sensitivity,
+    )
+    if isinstance(out, tuple):
+        similarity, extra = out
+    else:
+        similarity, extra = out, None
+
+    from metrics_amd.utilities.distributed import reduce
+
+    similarity = reduce(similarity, reduction or "none")
+    if extra is not None:
+        return similarity, extra
+    return similarity
+
+
+_MS_SSIM_BETAS = (0.0448, 0.2856, 0.3001, 0.2363, 0.1333)
+
+
+def multiscale_structural_similarity_index_measure(
+    preds: Tensor,
+    target: Tensor,
+    gaussian_kernel: bool = True,
+    sigma: Union[float, Sequence[float]] = 1.5,
+    kernel_size: Union[int, Sequence[int]] = 11,
+    reduction: Optional[str] = "elementwise_mean",
+    data_range: Optional[Union[float, Tuple[float, float]]] = None,
+    k1: float = 0.01,
+    k2: float = 0.03,
+    betas: Tuple[float, ...] = _MS_SSIM_BETAS,
+    normalize: Optional[str] = "relu",
+) -> Tensor:
+    """Multi-scale SSIM."""
+    preds, target = _ssim_check_inputs(preds, target)
+    if not isinstance(betas, tuple) or not all(isinstance(beta, float) for beta in betas):
+        raise ValueError("Argument `betas` is expected to be of a tuple of floats")
+    if normalize not in ("relu", "simple", None):
+        raise ValueError("Argument `normalize` to be expected either `None` or one of 'relu' or 'simple'")
+
+    sizes = preds.shape[-2:]
+    if (sizes[0] // (2 ** (len(betas) - 1))) <= (kernel_size if isinstance(kernel_size, int) else kernel_size[0]) or (
+        sizes[1] // (2 ** (len(betas) - 1))
+    ) <= (kernel_size if isinstance(kernel_size, int) else kernel_size[-1]):
+        raise ValueError(
+            f"For a given number of `betas` parameters {len(betas)}, the image height and width should be larger than"
+            f" {(kernel_size if isinstance(kernel_size, int) else kernel_size[0]) * (2 ** (len(betas) - 1))}"
+        )
+
+    mcs_list: List[Tensor] = []
+    sim = None
+    for i in range(len(betas)):
+        sim, contrast_sensitivity = _ssim_compute(
+            preds, target, gaussian_kernel, sigma, kernel_size, data_range, k1, k2,
+            return_contrast_sensitivity=True,
+        )
+        mcs_list.append(contrast_sensitivity)
+        if i < len(betas) - 1:
+            preds = F.avg_pool2d(preds, (2, 2))
+            target = F.avg_pool2d(target, (2, 2))
+
+    mcs_list[-1] = sim
+    mcs_stack = torch.stack(mcs_list)
+
+    if normalize == "relu":
+        mcs_stack = torch.relu(mcs_stack)
+
+    betas_t = torch.tensor(betas, device=mcs_stack.device).view(-1, 1)
+    if normalize == "simple":
+        mcs_stack = (mcs_stack + 1) / 2
+    mcs_weighted = mcs_stack**betas_t
+    out = torch.prod(mcs_weighted, axis=0)
+
+    from metrics_amd.utilities.distributed import reduce
+
+    return reduce(out, reduction or "none")

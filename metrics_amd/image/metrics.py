@@ -1,0 +1,491 @@
+"""Modular image metrics (pure-math family). Parity: torchmetrics ``image/*``."""
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional, Sequence, Tuple, Union
+
+import torch
+from torch import Tensor
+
+from metrics_amd.metric import Metric
+from metrics_amd.utilities.data import dim_zero_cat
+from metrics_amd.functional.image.psnr import (
+    _psnr_compute,
+    _psnr_update,
+    peak_signal_noise_ratio_with_blocked_effect,
+)
+from metrics_amd.functional.image.ssim import (
+    multiscale_structural_similarity_index_measure,
+    structural_similarity_index_measure,
+)
+from metrics_amd.functional.image.misc import (
+    error_relative_global_dimensionless_synthesis,
+    relative_average_spectral_error,
+    root_mean_squared_error_using_sliding_window,
+    spatial_correlation_coefficient,
+    spectral_angle_mapper,
+    total_variation,
+    universal_image_quality_index,
+    visual_information_fidelity,
+)
+from metrics_amd.functional.image.pansharpening import (
+    quality_with_no_reference,
+    spatial_distortion_index,
+    spectral_distortion_index,
+)
+
+
+class PeakSignalNoiseRatio(Metric):
+    """PSNR (stateful)."""
+
+    is_differentiable = True
+    higher_is_better = True
+    full_state_update = False
+    plot_lower_bound: float = 0.0
+
+    def __init__(
+        self,
+        data_range: Optional[Union[float, Tuple[float, float]]] = None,
+        base: float = 10.0,
+        reduction: str = "elementwise_mean",
+        dim: Optional[Union[int, Tuple[int, ...]]] = None,
+        **kwargs: Any,
+    ) -> None:
+        super().__init__(**kwargs)
+        if dim is None and reduction != "elementwise_mean":
+            import warnings
+
+            warnings.warn(f"The `reduction={reduction}` will not have any effect when `dim` is None.", stacklevel=2)
+
+        if dim is None:
+            self.add_state("sum_squared_error", default=torch.tensor(0.0), dist_reduce_fx="sum")
+            self.add_state("total", default=torch.tensor(0), dist_reduce_fx="sum")
+        else:
+            self.add_state("sum_squared_error", default=[], dist_reduce_fx="cat")
+            self.add_state("total", default=[], dist_reduce_fx="cat")
+
+        if data_range is None:
+            if dim is not None:
+                raise ValueError("The `data_range` must be given when `dim` is not None.")
+            self.data_range = None
+            self.add_state("min_target", default=torch.tensor(float("inf")), dist_reduce_fx=torch.min)
+            self.add_state("max_target", default=torch.tensor(-float("inf")), dist_reduce_fx=torch.max)
+            self.clamping_fn = None
+        elif isinstance(data_range, tuple):
+            self.add_state("data_range", default=torch.tensor(data_range[1] - data_range[0]), dist_reduce_fx="mean")
+            self.clamping_fn = lambda x: torch.clamp(x, min=data_range[0], max=data_range[1])
+        else:
+            self.add_state("data_range", default=torch.tensor(float(data_range)), dist_reduce_fx="mean")
+            self.clamping_fn = None
+        self.base = base
+        self.reduction = reduction
+        self.dim = tuple(dim) if isinstance(dim, Sequence) else dim
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate squared error sums."""
+        if self.clamping_fn is not None:
+            preds = self.clamping_fn(preds)
+            target = self.clamping_fn(target)
+        sum_squared_error, num_obs = _psnr_update(preds, target, dim=self.dim)
+        if self.dim is None:
+            if self.data_range is None:
+                # keep track of min and max target values
+                self.min_target = torch.minimum(target.min(), self.min_target)
+                self.max_target = torch.maximum(target.max(), self.max_target)
+            self.sum_squared_error += sum_squared_error
+            self.total += num_obs
+        else:
+            self.sum_squared_error.append(sum_squared_error)
+            self.total.append(num_obs)
+
+    def compute(self) -> Tensor:
+        """PSNR over all data."""
+        data_range = self.data_range if self.data_range is not None else (self.max_target - self.min_target)
+        if self.dim is None:
+            sum_squared_error = self.sum_squared_error
+            total = self.total
+        else:
+            sum_squared_error = torch.cat([v.flatten() for v in self.sum_squared_error])
+            total = torch.cat([v.flatten() for v in self.total])
+        return _psnr_compute(sum_squared_error, total, data_range, base=self.base, reduction=self.reduction)
+
+    def plot(self, val=None, ax=None):
+        return self._plot(val, ax)
+
+
+class _ScoreAverageMetric(Metric):
+    """Base: accumulate per-image scores as (sum, count)."""
+
+    is_differentiable = True
+    higher_is_better = True
+    full_state_update = False
+
+    score_sum: Tensor
+    total: Tensor
+
+    def __init__(self, **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        self.add_state("score_sum", torch.tensor(0.0), dist_reduce_fx="sum")
+        self.add_state("total", torch.tensor(0), dist_reduce_fx="sum")
+
+    def _accumulate(self, scores: Tensor, n: int) -> None:
+        self.score_sum += scores.sum() if scores.ndim else scores * n
+        self.total += n
+
+    def compute(self) -> Tensor:
+        return self.score_sum / self.total
+
+    def plot(self, val=None, ax=None):
+        return self._plot(val, ax)
+
+
+class StructuralSimilarityIndexMeasure(_ScoreAverageMetric):
+    """SSIM (stateful)."""
+
+    plot_lower_bound: float = 0.0
+    plot_upper_bound: float = 1.0
+
+    def __init__(
+        self,
+        gaussian_kernel: bool = True,
+        sigma: Union[float, Sequence[float]] = 1.5,
+        kernel_size: Union[int, Sequence[int]] = 11,
+        reduction: str = "elementwise_mean",
+        data_range: Optional[Union[float, Tuple[float, float]]] = None,
+        k1: float = 0.01,
+        k2: float = 0.03,
+        **kwargs: Any,
+    ) -> None:
+        super().__init__(**kwargs)
+        self.gaussian_kernel = gaussian_kernel
+        self.sigma = sigma
+        self.kernel_size = kernel_size
+        self.reduction = reduction
+        self.data_range = data_range
+        self.k1 = k1
+        self.k2 = k2
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate per-image SSIM."""
+        sim = structural_similarity_index_measure(
+            preds, target, self.gaussian_kernel, self.sigma, self.kernel_size, None,
+            self.data_range, self.k1, self.k2,
+        )
+        self._accumulate(sim, preds.shape[0])
+
+
+class MultiScaleStructuralSimilarityIndexMeasure(_ScoreAverageMetric):
+    """MS-SSIM (stateful)."""
+
+    plot_lower_bound: float = 0.0
+    plot_upper_bound: float = 1.0
+
+    def __init__(
+        self,
+        gaussian_kernel: bool = True,
+        kernel_size: Union[int, Sequence[int]] = 11,
+        sigma: Union[float, Sequence[float]] = 1.5,
+        reduction: str = "elementwise_mean",
+        data_range: Optional[Union[float, Tuple[float, float]]] = None,
+        k1: float = 0.01,
+        k2: float = 0.03,
+        betas: Tuple[float, ...] = (0.0448, 0.2856, 0.3001, 0.2363, 0.1333),
+        normalize: Optional[str] = "relu",
+        **kwargs: Any,
+    ) -> None:
+        super().__init__(**kwargs)
+        self.gaussian_kernel = gaussian_kernel
+        self.sigma = sigma
+        self.kernel_size = kernel_size
+        self.reduction = reduction
+        self.data_range = data_range
+        self.k1 = k1
+        self.k2 = k2
+        self.betas = betas
+        self.normalize = normalize
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate per-image MS-SSIM."""
+        sim = multiscale_structural_similarity_index_measure(
+            preds, target, self.gaussian_kernel, self.sigma, self.kernel_size, None,
+            self.data_range, self.k1, self.k2, self.betas, self.normalize,
+        )
+        self._accumulate(sim, preds.shape[0])
+
+
+class UniversalImageQualityIndex(_ScoreAverageMetric):
+    """UQI (stateful)."""
+
+    plot_lower_bound: float = 0.0
+    plot_upper_bound: float = 1.0
+
+    def __init__(
+        self, kernel_size: Sequence[int] = (11, 11), sigma: Sequence[float] = (1.5, 1.5),
+        reduction: str = "elementwise_mean", **kwargs: Any,
+    ) -> None:
+        super().__init__(**kwargs)
+        self.kernel_size = kernel_size
+        self.sigma = sigma
+        self.reduction = reduction
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate per-image UQI."""
+        score = universal_image_quality_index(preds, target, self.kernel_size, self.sigma, None)
+        self._accumulate(score, preds.shape[0])
+
+
+class SpectralAngleMapper(_ScoreAverageMetric):
+    """SAM (stateful, radians; lower is better)."""
+
+    higher_is_better = False
+    plot_lower_bound: float = 0.0
+
+    def __init__(self, reduction: str = "elementwise_mean", **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        self.reduction = reduction
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate per-pixel spectral angles."""
+        score = spectral_angle_mapper(preds, target, None)
+        self._accumulate(score.mean(dim=(1, 2)) if score.ndim == 3 else score, preds.shape[0])
+
+
+class ErrorRelativeGlobalDimensionlessSynthesis(_ScoreAverageMetric):
+    """ERGAS (stateful; lower is better)."""
+
+    higher_is_better = False
+    plot_lower_bound: float = 0.0
+
+    def __init__(self, ratio: float = 4, reduction: str = "elementwise_mean", **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        self.ratio = ratio
+        self.reduction = reduction
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate per-image ERGAS."""
+        score = error_relative_global_dimensionless_synthesis(preds, target, self.ratio, None)
+        self._accumulate(score, preds.shape[0])
+
+
+class TotalVariation(Metric):
+    """Total variation (stateful)."""
+
+    is_differentiable = True
+    higher_is_better = False
+    full_state_update = False
+    plot_lower_bound: float = 0.0
+
+    score_sum: Tensor
+    num_elements: Tensor
+
+    def __init__(self, reduction: str = "sum", **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        if reduction is not None and reduction not in ("sum", "mean", "none"):
+            raise ValueError("Expected argument `reduction` to either be 'sum', 'mean', 'none' or None")
+        self.reduction = reduction
+        if self.reduction is None or self.reduction == "none":
+            self.add_state("score_list", default=[], dist_reduce_fx="cat")
+        else:
+            self.add_state("score_sum", default=torch.tensor(0.0), dist_reduce_fx="sum")
+        self.add_state("num_elements", default=torch.tensor(0, dtype=torch.int), dist_reduce_fx="sum")
+
+    def update(self, img: Tensor) -> None:
+        """Accumulate per-image total variation."""
+        score = total_variation(img, reduction="none")
+        if self.reduction is None or self.reduction == "none":
+            self.score_list.append(score)
+        else:
+            self.score_sum += score.sum()
+        self.num_elements += score.numel()
+
+    def compute(self) -> Tensor:
+        if self.reduction is None or self.reduction == "none":
+            return dim_zero_cat(self.score_list)
+        if self.reduction == "mean":
+            return self.score_sum / self.num_elements
+        return self.score_sum
+
+    def plot(self, val=None, ax=None):
+        return self._plot(val, ax)
+
+
+class RootMeanSquaredErrorUsingSlidingWindow(_ScoreAverageMetric):
+    """RMSE over sliding windows (stateful; lower is better)."""
+
+    higher_is_better = False
+    plot_lower_bound: float = 0.0
+
+    def __init__(self, window_size: int = 8, **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        if not isinstance(window_size, int) or window_size < 1:
+            raise ValueError("Argument `window_size` is expected to be a positive integer.")
+        self.window_size = window_size
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate windowed RMSE."""
+        score = root_mean_squared_error_using_sliding_window(preds, target, self.window_size)
+        self._accumulate(score, 1)
+
+
+class SpatialCorrelationCoefficient(_ScoreAverageMetric):
+    """SCC (stateful)."""
+
+    plot_lower_bound: float = -1.0
+    plot_upper_bound: float = 1.0
+
+    def __init__(self, hp_filter: Optional[Tensor] = None, window_size: int = 8, **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        self.hp_filter = hp_filter
+        self.window_size = window_size
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate per-image SCC."""
+        score = spatial_correlation_coefficient(preds, target, self.hp_filter, self.window_size, None)
+        self._accumulate(score, preds.shape[0])
+
+
+class VisualInformationFidelity(_ScoreAverageMetric):
+    """VIF-P (stateful)."""
+
+    plot_lower_bound: float = 0.0
+
+    def __init__(self, sigma_n_sq: float = 2.0, **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        self.sigma_n_sq = sigma_n_sq
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate per-batch VIF."""
+        score = visual_information_fidelity(preds, target, self.sigma_n_sq)
+        self._accumulate(score, 1)
+
+
+class RelativeAverageSpectralError(_ScoreAverageMetric):
+    """RASE (stateful; lower is better)."""
+
+    higher_is_better = False
+    plot_lower_bound: float = 0.0
+
+    def __init__(self, window_size: int = 8, **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        self.window_size = window_size
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate per-batch RASE."""
+        score = relative_average_spectral_error(preds, target, self.window_size)
+        self._accumulate(score, 1)
+
+
+class PeakSignalNoiseRatioWithBlockedEffect(_ScoreAverageMetric):
+    """PSNR-B (stateful)."""
+
+    plot_lower_bound: float = 0.0
+
+    def __init__(self, block_size: int = 8, **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        self.block_size = block_size
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate per-batch PSNR-B."""
+        score = peak_signal_noise_ratio_with_blocked_effect(preds, target, self.block_size)
+        self._accumulate(score, 1)
+
+
+class SpectralDistortionIndex(Metric):
+    """D_lambda (stateful; cat states)."""
+
+    is_differentiable = True
+    higher_is_better = False
+    full_state_update = False
+    plot_lower_bound: float = 0.0
+    plot_upper_bound: float = 1.0
+
+    preds: List[Tensor]
+    target: List[Tensor]
+
+    def __init__(self, p: int = 1, reduction: str = "elementwise_mean", **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        self.p = p
+        self.reduction = reduction
+        self.add_state("preds", default=[], dist_reduce_fx="cat")
+        self.add_state("target", default=[], dist_reduce_fx="cat")
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Append batches."""
+        self.preds.append(preds)
+        self.target.append(target)
+
+    def compute(self) -> Tensor:
+        return spectral_distortion_index(dim_zero_cat(self.preds), dim_zero_cat(self.target), self.p, self.reduction)
+
+    def plot(self, val=None, ax=None):
+        return self._plot(val, ax)
+
+
+class SpatialDistortionIndex(Metric):
+    """D_s (stateful; cat states)."""
+
+    is_differentiable = True
+    higher_is_better = False
+    full_state_update = False
+    plot_lower_bound: float = 0.0
+    plot_upper_bound: float = 1.0
+
+    def __init__(self, norm_order: int = 1, window_size: int = 7, reduction: str = "elementwise_mean", **kwargs: Any) -> None:
+        super().__init__(**kwargs)
+        self.norm_order = norm_order
+        self.window_size = window_size
+        self.reduction = reduction
+        self.add_state("preds", default=[], dist_reduce_fx="cat")
+        self.add_state("ms", default=[], dist_reduce_fx="cat")
+        self.add_state("pan", default=[], dist_reduce_fx="cat")
+
+    def update(self, preds: Tensor, target: Dict[str, Tensor]) -> None:
+        """Append fused prediction + {'ms','pan'} targets."""
+        self.preds.append(preds)
+        self.ms.append(target["ms"])
+        self.pan.append(target["pan"])
+
+    def compute(self) -> Tensor:
+        target = {"ms": dim_zero_cat(self.ms), "pan": dim_zero_cat(self.pan)}
+        return spatial_distortion_index(dim_zero_cat(self.preds), target, self.norm_order, self.window_size, self.reduction)
+
+    def plot(self, val=None, ax=None):
+        return self._plot(val, ax)
+
+
+class QualityWithNoReference(Metric):
+    """QNR (stateful; cat states)."""
+
+    is_differentiable = True
+    higher_is_better = True
+    full_state_update = False
+    plot_lower_bound: float = 0.0
+    plot_upper_bound: float = 1.0
+
+    def __init__(
+        self, alpha: float = 1.0, beta: float = 1.0, norm_order: int = 1, window_size: int = 7,
+        reduction: str = "elementwise_mean", **kwargs: Any,
+    ) -> None:
+        super().__init__(**kwargs)
+        self.alpha = alpha
+        self.beta = beta
+        self.norm_order = norm_order
+        self.window_size = window_size
+        self.reduction = reduction
+        self.add_state("preds", default=[], dist_reduce_fx="cat")
+        self.add_state("ms", default=[], dist_reduce_fx="cat")
+        self.add_state("pan", default=[], dist_reduce_fx="cat")
+
+    def update(self, preds: Tensor, target: Dict[str, Tensor]) -> None:
+        """Append fused prediction + {'ms','pan'} targets."""
+        self.preds.append(preds)
+        self.ms.append(target["ms"])
+        self.pan.append(target["pan"])
+
+    def compute(self) -> Tensor:
+        target = {"ms": dim_zero_cat(self.ms), "pan": dim_zero_cat(self.pan)}
+        return quality_with_no_reference(
+            dim_zero_cat(self.preds), target, self.alpha, self.beta, self.norm_order, self.window_size, self.reduction
+        )
+
+    def plot(self, val=None, ax=None):
+        return self._plot(val, ax)
