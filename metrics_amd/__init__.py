@@ -26,7 +26,7 @@ from metrics_amd.aggregation import (  # noqa: E402
 )
 from metrics_amd.collections import MetricCollection  # noqa: E402
 from metrics_amd.metric import CompositionalMetric, Metric  # noqa: E402
-from metrics_amd import classification, clustering, detection, functional, image, nominal, ops, regression, retrieval, segmentation, utilities, wrappers  # noqa: E402
+from metrics_amd import audio, classification, clustering, detection, functional, image, multimodal, nominal, ops, regression, retrieval, segmentation, shape, text, utilities, wrappers  # noqa: E402
 from metrics_amd.classification import *  # noqa: E402,F401,F403
 from metrics_amd.regression import *  # noqa: E402,F401,F403
 from metrics_amd.retrieval import *  # noqa: E402,F401,F403
@@ -35,6 +35,9 @@ from metrics_amd.nominal import *  # noqa: E402,F401,F403
 from metrics_amd.detection import *  # noqa: E402,F401,F403
 from metrics_amd.segmentation import *  # noqa: E402,F401,F403
 from metrics_amd.image import *  # noqa: E402,F401,F403
+from metrics_amd.audio import *  # noqa: E402,F401,F403
+from metrics_amd.text import *  # noqa: E402,F401,F403
+from metrics_amd.shape import ProcrustesDisparity  # noqa: E402,F401
 
 __all__ = [
     "CatMetric",
@@ -62,3 +65,6 @@ __all__ += nominal.__all__
 __all__ += detection.__all__
 __all__ += segmentation.__all__
 __all__ += image.__all__
+__all__ += audio.__all__
+__all__ += text.__all__
+__all__ += ["ProcrustesDisparity"]

@@ -7,6 +7,9 @@ from metrics_amd.functional.nominal import *  # noqa: F401,F403
 from metrics_amd.functional.detection import *  # noqa: F401,F403
 from metrics_amd.functional.segmentation import *  # noqa: F401,F403
 from metrics_amd.functional.image import *  # noqa: F401,F403
+from metrics_amd.functional.audio import *  # noqa: F401,F403
+from metrics_amd.functional.text import *  # noqa: F401,F403
+from metrics_amd.shape import procrustes_disparity  # noqa: F401
 from metrics_amd.functional.pairwise import (  # noqa: F401
     pairwise_cosine_similarity,
     pairwise_euclidean_distance,
@@ -14,4 +17,4 @@ from metrics_amd.functional.pairwise import (  # noqa: F401
     pairwise_manhattan_distance,
     pairwise_minkowski_distance,
 )
-from metrics_amd.functional import classification, clustering, detection, image, nominal, pairwise, regression, retrieval, segmentation  # noqa: F401
+from metrics_amd.functional import audio, classification, clustering, detection, image, nominal, pairwise, regression, retrieval, segmentation, text  # noqa: F401

@@ -1,0 +1,102 @@
+"""Text helpers: Levenshtein DP + tokenizers.
+
+Parity: torchmetrics ``functional/text/helper.py``.
+"""
+from __future__ import annotations
+
+import re
+from typing import List, Sequence, Tuple, Union
+
+
+def _edit_distance(prediction_tokens: Sequence, reference_tokens: Sequence) -> int:
+    """Levenshtein distance between two token sequences (O(nm) DP, two rows)."""
+    n, m = len(prediction_tokens), len(reference_tokens)
+    if n == 0:
+        return m
+    if m == 0:
+        return n
+    prev = list(range(m + 1))
+    cur = [0] * (m + 1)
+    for i in range(1, n + 1):
+        cur[0] = i
+        p = prediction_tokens[i - 1]
+        for j in range(1, m + 1):
+            cost = 0 if p == reference_tokens[j - 1] else 1
+            cur[j] = min(prev[j] + 1, cur[j - 1] + 1, prev[j - 1] + cost)
+        prev, cur = cur, prev
+    return prev[m]
+
+
+def _edit_distance_counts(prediction_tokens: Sequence, reference_tokens: Sequence) -> Tuple[int, int, int, int]:
+    """(substitutions, insertions, deletions, matches) via full DP backtrace."""
+    n, m = len(prediction_tokens), len(reference_tokens)
+    dp = [[0] * (m + 1) for _ in range(n + 1)]
+    for i in range(n + 1):
+        dp[i][0] = i
+    for j in range(m + 1):
+        dp[0][j] = j
+    for i in range(1, n + 1):
+        for j in range(1, m + 1):
+            cost = 0 if prediction_tokens[i - 1] == reference_tokens[j - 1] else 1
+            dp[i][j] = min(dp[i - 1][j] + 1, dp[i][j - 1] + 1, dp[i - 1][j - 1] + cost)
+    # backtrace
+    i, j = n, m
+    subs = ins = dels = hits = 0
+    while i > 0 or j > 0:
+        if i > 0 and j > 0 and dp[i][j] == dp[i - 1][j - 1] + (0 if prediction_tokens[i - 1] == reference_tokens[j - 1] else 1):
+            if prediction_tokens[i - 1] == reference_tokens[j - 1]:
+                hits += 1
+            else:
+                subs += 1
+            i -= 1
+            j -= 1
+        elif i > 0 and dp[i][j] == dp[i - 1][j] + 1:
+            ins += 1  # extra token in prediction
+            i -= 1
+        else:
+            dels += 1  # missing token from reference
+            j -= 1
+    return subs, ins, dels, hits
+
+
+_13A_RE1 = re.compile(r"([\{-\~\[-\` -\&\(-\+\:-\@\/])")
+_13A_RE_NUM = re.compile(r"([0-9])([\.,])")
+_13A_RE_NUM2 = re.compile(r"([\.,])([0-9])")
+_13A_RE_DASH = re.compile(r"([0-9])(-)")
+
+
+def _tokenize_13a(line: str, lowercase: bool = False) -> List[str]:
+    """mteval-v13a tokenizer (sacrebleu default)."""
+    line = line.replace("<skipped>", "")
+    line = line.replace("-\n", "")
+    line = line.replace("\n", " ")
+    line = line.replace("&quot;", '"').replace("&amp;", "&").replace("&lt;", "<").replace("&gt;", ">")
+    line = f" {line} "
+    line = _13A_RE_NUM.sub(r"\1 \2 ", line)
+    line = _13A_RE_NUM2.sub(r" \1 \2", line)
+    line = _13A_RE_DASH.sub(r"\1 \2 ", line)
+    line = _13A_RE1.sub(r" \1 ", line)
+    if lowercase:
+        line = line.lower()
+    return line.split()
+
+
+def _tokenize_char(line: str, lowercase: bool = False) -> List[str]:
+    if lowercase:
+        line = line.lower()
+    return list(line.strip())
+
+
+def _tokenize_none(line: str, lowercase: bool = False) -> List[str]:
+    if lowercase:
+        line = line.lower()
+    return line.strip().split()
+
+
+_TOKENIZERS = {"13a": _tokenize_13a, "char": _tokenize_char, "none": _tokenize_none, "intl": _tokenize_13a}
+
+
+def get_tokenizer(name: str):
+    if name not in _TOKENIZERS:
+        raise ValueError(f"Unsupported tokenizer {name}; expected one of {list(_TOKENIZERS)}")
+    return _TOKENIZERS[name]

@@ -1,0 +1,113 @@
+"""Text metric tests vs known values."""
+import torch
+
+import metrics_amd as ma
+
+
+def test_wer_cer():
+    assert abs(ma.WordErrorRate()(["there is an other sample"], ["there is another one sample"]).item() - 0.4) < 1e-6
+    assert abs(ma.CharErrorRate()(["abcd"], ["abce"]).item() - 0.25) < 1e-6
+
+
+def test_mer_wil_wip():
+    preds = ["hello world"]
+    target = ["hello duck"]
+    # 1 substitution, 1 hit
+    assert abs(ma.MatchErrorRate()(preds, target).item() - 0.5) < 1e-6
+    wip = ma.WordInfoPreserved()(preds, target).item()
+    # hits/len_target * hits/len_preds = (1/2) * (1/2)
+    assert abs(wip - 0.25) < 1e-6
+    assert abs(ma.WordInfoLost()(preds, target).item() - 0.75) < 1e-6
+
+
+def test_bleu_known():
+    preds = ["the cat is on the mat"]
+    target = [["there is a cat on the mat", "a cat is on the mat"]]
+    assert abs(ma.BLEUScore()(preds, target).item() - 0.7598) < 1e-3
+    assert abs(ma.SacreBLEUScore()(preds, target).item() - 0.7598) < 1e-3
+    assert ma.BLEUScore()(["completely different text here"], [["no overlap at all"]]).item() == 0.0
+
+
+def test_bleu_accumulation():
+    m = ma.BLEUScore()
+    m.update(["the cat is on the mat"], [["there is a cat on the mat", "a cat is on the mat"]])
+    m.update(["the cat is on the mat"], [["the cat is on the mat"]])
+    v = m.compute().item()
+    assert 0.75 < v <= 1.0
+
+
+def test_rouge():
+    r = ma.ROUGEScore()(["the quick brown fox"], ["the quick brown fox"])
+    for k in ("rouge1_fmeasure", "rouge2_fmeasure", "rougeL_fmeasure", "rougeLsum_fmeasure"):
+        assert abs(r[k].item() - 1.0) < 1e-6
+    r = ma.ROUGEScore(rouge_keys="rouge1")(["cat dog"], ["cat bird"])
+    assert abs(r["rouge1_fmeasure"].item() - 0.5) < 1e-6
+
+
+def test_perplexity():
+    B, T, V = 2, 8, 10
+    logits = torch.zeros(B, T, V)
+    tgt = torch.randint(0, V, (B, T))
+    assert abs(ma.Perplexity()(logits, tgt).item() - V) < 1e-4
+    # ignore index
+    tgt2 = tgt.clone()
+    tgt2[0, :4] = -100
+    assert abs(ma.Perplexity(ignore_index=-100)(logits, tgt2).item() - V) < 1e-4
+
+
+def test_squad():
+    # articles are stripped by normalization: "a cat" == "the cat"
+    p = [{"prediction_text": "1976", "id": "1"}, {"prediction_text": "a cat", "id": "2"}]
+    t = [
+        {"answers": {"text": ["1976"]}, "id": "1"},
+        {"answers": {"text": ["the cat"]}, "id": "2"},
+    ]
+    r = ma.SQuAD()(p, t)
+    assert abs(r["exact_match"].item() - 100.0) < 1e-6
+    # a genuinely different answer drops EM to 50
+    p2 = [{"prediction_text": "1976", "id": "1"}, {"prediction_text": "a dog", "id": "2"}]
+    r2 = ma.SQuAD()(p2, t)
+    assert abs(r2["exact_match"].item() - 50.0) < 1e-6
+    assert 50.0 <= r2["f1"].item() <= 100.0
+
+
+def test_edit_distance():
+    assert ma.EditDistance()(["rain"], ["shine"]).item() == 3.0
+    assert ma.EditDistance(reduction="sum")(["rain", "abc"], ["shine", "abc"]).item() == 3.0
+
+
+def test_chrf_ter_eed():
+    assert abs(ma.CHRFScore()(["hello world"], [["hello world"]]).item() - 1.0) < 1e-4
+    assert ma.TranslationEditRate()(["the cat"], [["the cat"]]).item() == 0.0
+    assert ma.ExtendedEditDistance()(["the cat"], [["the cat"]]).item() < 1e-6
+    # TER with one substitution over 3 tokens
+    v = ma.TranslationEditRate()(["the big cat"], [["the small cat"]]).item()
+    assert abs(v - 1 / 3) < 1e-6
+
+
+def test_bert_score_with_toy_model():
+    """BERTScore machinery works with any HF-style encoder."""
+
+    class ToyTok:
+        def __call__(self, texts, **kw):
+            ids = [[hash(w) % 50 for w in t.split()] for t in texts]
+            maxlen = max(len(i) for i in ids)
+            input_ids = torch.zeros(len(ids), maxlen, dtype=torch.long)
+            mask = torch.zeros(len(ids), maxlen, dtype=torch.long)
+            for r, i in enumerate(ids):
+                input_ids[r, : len(i)] = torch.tensor(i)
+                mask[r, : len(i)] = 1
+            return {"input_ids": input_ids, "attention_mask": mask}
+
+    class ToyModel(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.emb = torch.nn.Embedding(50, 16)
+
+        def forward(self, input_ids=None, attention_mask=None):
+            return (self.emb(input_ids),)
+
+    m = ma.BERTScore(model=ToyModel(), user_tokenizer=ToyTok())
+    m.update(["the cat sat"], ["the cat sat"])
+    res = m.compute()
+    assert abs(res["f1"].item() - 1.0) < 1e-5
