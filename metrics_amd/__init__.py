@@ -38,6 +38,18 @@ from metrics_amd.image import *  # noqa: E402,F401,F403
 from metrics_amd.audio import *  # noqa: E402,F401,F403
 from metrics_amd.text import *  # noqa: E402,F401,F403
 from metrics_amd.shape import ProcrustesDisparity  # noqa: E402,F401
+from metrics_amd.wrappers import (  # noqa: E402,F401
+    BinaryTargetTransformer,
+    BootStrapper,
+    ClasswiseWrapper,
+    FeatureShare,
+    LambdaInputTransformer,
+    MetricTracker,
+    MinMaxMetric,
+    MultioutputWrapper,
+    MultitaskWrapper,
+    Running,
+)
 
 __all__ = [
     "CatMetric",
@@ -68,3 +80,4 @@ __all__ += image.__all__
 __all__ += audio.__all__
 __all__ += text.__all__
 __all__ += ["ProcrustesDisparity"]
+__all__ += wrappers.__all__

@@ -1,4 +1,24 @@
 from metrics_amd.wrappers.abstract import WrapperMetric
+from metrics_amd.wrappers.bootstrapping import BootStrapper
+from metrics_amd.wrappers.classwise import ClasswiseWrapper
+from metrics_amd.wrappers.feature_share import FeatureShare
+from metrics_amd.wrappers.minmax import MinMaxMetric
+from metrics_amd.wrappers.multioutput import MultioutputWrapper
+from metrics_amd.wrappers.multitask import MultitaskWrapper
 from metrics_amd.wrappers.running import Running
+from metrics_amd.wrappers.tracker import MetricTracker
+from metrics_amd.wrappers.transformations import BinaryTargetTransformer, LambdaInputTransformer
 
-__all__ = ["Running", "WrapperMetric"]
+__all__ = [
+    "BinaryTargetTransformer",
+    "BootStrapper",
+    "ClasswiseWrapper",
+    "FeatureShare",
+    "LambdaInputTransformer",
+    "MetricTracker",
+    "MinMaxMetric",
+    "MultioutputWrapper",
+    "MultitaskWrapper",
+    "Running",
+    "WrapperMetric",
+]

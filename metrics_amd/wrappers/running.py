@@ -53,14 +53,10 @@ class Running(WrapperMetric):
         self._num_vals_seen += 1
 
     def forward(self, *args: Any, **kwargs: Any) -> Any:
-        """Forward input to the base metric and update the rolling window."""
-        val = self._num_vals_seen % self.window
-        self._forward_cache = self.base_metric.forward(*args, **kwargs)
-        for key in self.base_metric._defaults:
-            setattr(self, key + f"_{val}", deepcopy(getattr(self.base_metric, key)))
-        self.base_metric.reset()
-        self._num_vals_seen += 1
+        """Update the window with this batch and return the RUNNING value."""
+        self.update(*args, **kwargs)
         self._computed = None
+        self._forward_cache = self.compute()
         return self._forward_cache
 
     def compute(self) -> Any:

@@ -1,0 +1,113 @@
+"""Wrapper metric tests."""
+import pytest
+import torch
+
+import metrics_amd as ma
+from metrics_amd.wrappers import (
+    BinaryTargetTransformer,
+    BootStrapper,
+    ClasswiseWrapper,
+    LambdaInputTransformer,
+    MetricTracker,
+    MinMaxMetric,
+    MultioutputWrapper,
+    MultitaskWrapper,
+)
+
+
+def test_bootstrapper():
+    torch.manual_seed(0)
+    m = BootStrapper(ma.MulticlassAccuracy(num_classes=5, average="micro"), num_bootstraps=8)
+    preds = torch.randn(256, 5)
+    target = torch.randint(0, 5, (256,))
+    m.update(preds, target)
+    out = m.compute()
+    assert set(out) == {"mean", "std"}
+    base = ma.MulticlassAccuracy(num_classes=5, average="micro")(preds, target)
+    assert abs(out["mean"].item() - base.item()) < 0.1
+    assert out["std"].item() < 0.1
+
+
+def test_classwise_wrapper():
+    m = ClasswiseWrapper(ma.MulticlassAccuracy(num_classes=3, average=None), labels=["a", "b", "c"])
+    preds = torch.randn(64, 3)
+    target = torch.randint(0, 3, (64,))
+    m.update(preds, target)
+    out = m.compute()
+    assert set(out) == {"multiclassaccuracy_a", "multiclassaccuracy_b", "multiclassaccuracy_c"}
+
+
+def test_minmax():
+    m = MinMaxMetric(ma.MulticlassAccuracy(num_classes=3, average="micro"))
+    torch.manual_seed(1)
+    for _ in range(3):
+        preds = torch.randn(32, 3)
+        target = torch.randint(0, 3, (32,))
+        m.update(preds, target)
+        out = m.compute()
+        m.wrapped = None  # noop
+    assert out["min"] <= out["raw"] <= out["max"]
+
+
+def test_multioutput_wrapper():
+    m = MultioutputWrapper(ma.MeanSquaredError(), num_outputs=3)
+    preds = torch.randn(32, 3)
+    target = torch.randn(32, 3)
+    m.update(preds, target)
+    out = m.compute()
+    assert out.shape == (3,)
+    ref = ((preds - target) ** 2).mean(0)
+    assert torch.allclose(out, ref, atol=1e-6)
+
+
+def test_multitask_wrapper():
+    mt = MultitaskWrapper({
+        "cls": ma.BinaryAccuracy(),
+        "reg": ma.MeanSquaredError(),
+    })
+    preds = {"cls": torch.rand(16), "reg": torch.randn(16)}
+    target = {"cls": torch.randint(0, 2, (16,)), "reg": torch.randn(16)}
+    mt.update(preds, target)
+    out = mt.compute()
+    assert set(out) == {"cls", "reg"}
+
+
+def test_tracker():
+    tracker = MetricTracker(ma.MulticlassAccuracy(num_classes=3, average="micro"), maximize=True)
+    torch.manual_seed(2)
+    target = torch.randint(0, 3, (64,))
+    for step in range(3):
+        tracker.increment()
+        # make predictions progressively better
+        preds = torch.nn.functional.one_hot(target, 3).float() + (2 - step) * torch.randn(64, 3)
+        tracker.update(preds, target)
+    allv = tracker.compute_all()
+    assert allv.shape == (3,)
+    best, which = tracker.best_metric(return_step=True)
+    assert best == allv.max().item()
+    with pytest.raises(ValueError, match="cannot be called before"):
+        MetricTracker(ma.BinaryAccuracy()).update(torch.rand(2), torch.randint(0, 2, (2,)))
+
+
+def test_lambda_input_transformer():
+    m = LambdaInputTransformer(ma.BinaryAccuracy(), transform_pred=lambda p: 1 - p)
+    preds = torch.tensor([0.1, 0.9, 0.2])
+    target = torch.tensor([1, 0, 1])
+    m.update(preds, target)
+    assert m.compute().item() == 1.0
+
+
+def test_binary_target_transformer():
+    m = BinaryTargetTransformer(ma.BinaryAccuracy(), threshold=2)
+    preds = torch.tensor([1.0, 0.0, 1.0])
+    target = torch.tensor([5, 1, 7])  # binarized -> 1, 0, 1
+    m.update(preds, target)
+    assert m.compute().item() == 1.0
+
+
+def test_running_in_collection_context():
+    from metrics_amd import RunningMean
+
+    r = RunningMean(window=2)
+    outs = [r(v).item() for v in (1.0, 5.0, 9.0)]
+    assert outs == [1.0, 3.0, 7.0]
