@@ -97,3 +97,21 @@ from metrics_amd.functional.classification.stat_scores import (
     multilabel_stat_scores,
     stat_scores,
 )
+from metrics_amd.functional.classification.at_fixed import (
+    binary_logauc,
+    binary_precision_at_fixed_recall,
+    binary_recall_at_fixed_precision,
+    binary_sensitivity_at_specificity,
+    binary_specificity_at_sensitivity,
+    multiclass_logauc,
+    multiclass_precision_at_fixed_recall,
+    multiclass_recall_at_fixed_precision,
+    multiclass_sensitivity_at_specificity,
+    multiclass_specificity_at_sensitivity,
+    multilabel_logauc,
+    multilabel_precision_at_fixed_recall,
+    multilabel_recall_at_fixed_precision,
+    multilabel_sensitivity_at_specificity,
+    multilabel_specificity_at_sensitivity,
+)
+from metrics_amd.functional.classification.group_fairness import binary_fairness, binary_groups_stat_rates

@@ -1,0 +1,467 @@
+"""At-fixed-X curve metrics: recall@precision, precision@recall,
+sensitivity@specificity, specificity@sensitivity.
+
+Parity: torchmetrics ``functional/classification/{recall_fixed_precision,
+precision_fixed_recall,sensitivity_specificity,specificity_sensitivity}.py``.
+All reuse the PR-curve / ROC state machinery (HIP bucketized histograms on GPU).
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Tuple, Union
+
+import torch
+from torch import Tensor
+
+from metrics_amd.utilities.enums import ClassificationTask
+from metrics_amd.functional.classification.precision_recall_curve import (
+    _binary_precision_recall_curve_arg_validation,
+    _binary_precision_recall_curve_compute,
+    _binary_precision_recall_curve_format,
+    _binary_precision_recall_curve_tensor_validation,
+    _binary_precision_recall_curve_update,
+    _multiclass_precision_recall_curve_arg_validation,
+    _multiclass_precision_recall_curve_compute,
+    _multiclass_precision_recall_curve_format,
+    _multiclass_precision_recall_curve_tensor_validation,
+    _multiclass_precision_recall_curve_update,
+    _multilabel_precision_recall_curve_arg_validation,
+    _multilabel_precision_recall_curve_compute,
+    _multilabel_precision_recall_curve_format,
+    _multilabel_precision_recall_curve_tensor_validation,
+    _multilabel_precision_recall_curve_update,
+)
+from metrics_amd.functional.classification.roc import (
+    _binary_roc_compute,
+    _multiclass_roc_compute,
+    _multilabel_roc_compute,
+)
+
+
+def _best_at_constraint(
+    y: Tensor, x: Tensor, thresholds: Tensor, min_x: float
+) -> Tuple[Tensor, Tensor]:
+    """max y such that x >= min_x, and the threshold achieving it."""
+    try:
+        max_y = torch.max(y[x >= min_x])
+        idx = torch.nonzero((x >= min_x) & (y == max_y))[0]
+        thr = thresholds[min(int(idx), len(thresholds) - 1)] if thresholds.numel() else torch.tensor(1e6)
+    except (ValueError, IndexError, RuntimeError):
+        max_y = torch.tensor(0.0, device=y.device, dtype=y.dtype)
+        thr = torch.tensor(1e6, device=y.device)
+    return max_y, thr
+
+
+def _recall_at_precision(precision: Tensor, recall: Tensor, thresholds: Tensor, min_precision: float):
+    recall_at, thr = _best_at_constraint(recall, precision, thresholds, min_precision)
+    return recall_at, thr
+
+
+def _precision_at_recall(precision: Tensor, recall: Tensor, thresholds: Tensor, min_recall: float):
+    return _best_at_constraint(precision, recall, thresholds, min_recall)
+
+
+def binary_recall_at_fixed_precision(
+    preds: Tensor,
+    target: Tensor,
+    min_precision: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tuple[Tensor, Tensor]:
+    """(max recall with precision >= min_precision, its threshold)."""
+    if validate_args:
+        _binary_precision_recall_curve_arg_validation(thresholds, ignore_index)
+        _binary_precision_recall_curve_tensor_validation(preds, target, ignore_index)
+        if not isinstance(min_precision, float) or not (0 <= min_precision <= 1):
+            raise ValueError(f"Expected argument `min_precision` to be a float in the [0,1] range, but got {min_precision}")
+    remove_ignored = not (preds.is_cuda and thresholds is not None)
+    p_f, t_f, thr = _binary_precision_recall_curve_format(preds, target, thresholds, ignore_index, remove_ignored=remove_ignored)
+    state = _binary_precision_recall_curve_update(p_f, t_f, thr, ignore_index if not remove_ignored else None)
+    precision, recall, thresholds_out = _binary_precision_recall_curve_compute(state, thr)
+    return _recall_at_precision(precision, recall, thresholds_out, min_precision)
+
+
+def multiclass_recall_at_fixed_precision(
+    preds: Tensor,
+    target: Tensor,
+    num_classes: int,
+    min_precision: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tuple[Tensor, Tensor]:
+    """Per-class (recall, threshold) at fixed precision."""
+    if validate_args:
+        _multiclass_precision_recall_curve_arg_validation(num_classes, thresholds, ignore_index)
+        _multiclass_precision_recall_curve_tensor_validation(preds, target, num_classes, ignore_index)
+    remove_ignored = not (preds.is_cuda and thresholds is not None)
+    p_f, t_f, thr = _multiclass_precision_recall_curve_format(
+        preds, target, num_classes, thresholds, ignore_index, remove_ignored=remove_ignored
+    )
+    state = _multiclass_precision_recall_curve_update(
+        p_f, t_f, num_classes, thr, None, ignore_index if not remove_ignored else None
+    )
+    precision, recall, thresholds_out = _multiclass_precision_recall_curve_compute(state, num_classes, thr, None)
+    if isinstance(precision, Tensor):
+        res = [
+            _recall_at_precision(precision[i], recall[i], thresholds_out, min_precision) for i in range(num_classes)
+        ]
+    else:
+        res = [
+            _recall_at_precision(precision[i], recall[i], thresholds_out[i], min_precision) for i in range(num_classes)
+        ]
+    return torch.stack([r[0] for r in res]), torch.stack([torch.as_tensor(r[1]) for r in res])
+
+
+def multilabel_recall_at_fixed_precision(
+    preds: Tensor,
+    target: Tensor,
+    num_labels: int,
+    min_precision: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tuple[Tensor, Tensor]:
+    """Per-label (recall, threshold) at fixed precision."""
+    if validate_args:
+        _multilabel_precision_recall_curve_arg_validation(num_labels, thresholds, ignore_index)
+        _multilabel_precision_recall_curve_tensor_validation(preds, target, num_labels, ignore_index)
+    remove_ignored = not preds.is_cuda
+    p_f, t_f, thr = _multilabel_precision_recall_curve_format(
+        preds, target, num_labels, thresholds, ignore_index, remove_ignored=remove_ignored
+    )
+    state = _multilabel_precision_recall_curve_update(p_f, t_f, num_labels, thr, ignore_index if not remove_ignored else None)
+    precision, recall, thresholds_out = _multilabel_precision_recall_curve_compute(state, num_labels, thr, ignore_index)
+    if isinstance(precision, Tensor):
+        res = [_recall_at_precision(precision[i], recall[i], thresholds_out, min_precision) for i in range(num_labels)]
+    else:
+        res = [_recall_at_precision(precision[i], recall[i], thresholds_out[i], min_precision) for i in range(num_labels)]
+    return torch.stack([r[0] for r in res]), torch.stack([torch.as_tensor(r[1]) for r in res])
+
+
+def binary_precision_at_fixed_recall(
+    preds: Tensor,
+    target: Tensor,
+    min_recall: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tuple[Tensor, Tensor]:
+    """(max precision with recall >= min_recall, its threshold)."""
+    if validate_args:
+        _binary_precision_recall_curve_arg_validation(thresholds, ignore_index)
+        _binary_precision_recall_curve_tensor_validation(preds, target, ignore_index)
+    remove_ignored = not (preds.is_cuda and thresholds is not None)
+    p_f, t_f, thr = _binary_precision_recall_curve_format(preds, target, thresholds, ignore_index, remove_ignored=remove_ignored)
+    state = _binary_precision_recall_curve_update(p_f, t_f, thr, ignore_index if not remove_ignored else None)
+    precision, recall, thresholds_out = _binary_precision_recall_curve_compute(state, thr)
+    return _precision_at_recall(precision, recall, thresholds_out, min_recall)
+
+
+def multiclass_precision_at_fixed_recall(
+    preds: Tensor,
+    target: Tensor,
+    num_classes: int,
+    min_recall: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tuple[Tensor, Tensor]:
+    """Per-class (precision, threshold) at fixed recall."""
+    if validate_args:
+        _multiclass_precision_recall_curve_arg_validation(num_classes, thresholds, ignore_index)
+        _multiclass_precision_recall_curve_tensor_validation(preds, target, num_classes, ignore_index)
+    remove_ignored = not (preds.is_cuda and thresholds is not None)
+    p_f, t_f, thr = _multiclass_precision_recall_curve_format(
+        preds, target, num_classes, thresholds, ignore_index, remove_ignored=remove_ignored
+    )
+    state = _multiclass_precision_recall_curve_update(
+        p_f, t_f, num_classes, thr, None, ignore_index if not remove_ignored else None
+    )
+    precision, recall, thresholds_out = _multiclass_precision_recall_curve_compute(state, num_classes, thr, None)
+    if isinstance(precision, Tensor):
+        res = [_precision_at_recall(precision[i], recall[i], thresholds_out, min_recall) for i in range(num_classes)]
+    else:
+        res = [_precision_at_recall(precision[i], recall[i], thresholds_out[i], min_recall) for i in range(num_classes)]
+    return torch.stack([r[0] for r in res]), torch.stack([torch.as_tensor(r[1]) for r in res])
+
+
+def multilabel_precision_at_fixed_recall(
+    preds: Tensor,
+    target: Tensor,
+    num_labels: int,
+    min_recall: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tuple[Tensor, Tensor]:
+    """Per-label (precision, threshold) at fixed recall."""
+    if validate_args:
+        _multilabel_precision_recall_curve_arg_validation(num_labels, thresholds, ignore_index)
+        _multilabel_precision_recall_curve_tensor_validation(preds, target, num_labels, ignore_index)
+    remove_ignored = not preds.is_cuda
+    p_f, t_f, thr = _multilabel_precision_recall_curve_format(
+        preds, target, num_labels, thresholds, ignore_index, remove_ignored=remove_ignored
+    )
+    state = _multilabel_precision_recall_curve_update(p_f, t_f, num_labels, thr, ignore_index if not remove_ignored else None)
+    precision, recall, thresholds_out = _multilabel_precision_recall_curve_compute(state, num_labels, thr, ignore_index)
+    if isinstance(precision, Tensor):
+        res = [_precision_at_recall(precision[i], recall[i], thresholds_out, min_recall) for i in range(num_labels)]
+    else:
+        res = [_precision_at_recall(precision[i], recall[i], thresholds_out[i], min_recall) for i in range(num_labels)]
+    return torch.stack([r[0] for r in res]), torch.stack([torch.as_tensor(r[1]) for r in res])
+
+
+# ---------------------------------------------------------------- ROC variants
+def _sens_at_spec(fpr: Tensor, tpr: Tensor, thresholds: Tensor, min_specificity: float):
+    specificity = 1 - fpr
+    return _best_at_constraint(tpr, specificity, thresholds, min_specificity)
+
+
+def _spec_at_sens(fpr: Tensor, tpr: Tensor, thresholds: Tensor, min_sensitivity: float):
+    specificity = 1 - fpr
+    return _best_at_constraint(specificity, tpr, thresholds, min_sensitivity)
+
+
+def binary_sensitivity_at_specificity(
+    preds: Tensor,
+    target: Tensor,
+    min_specificity: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tuple[Tensor, Tensor]:
+    """(max sensitivity with specificity >= min_specificity, threshold)."""
+    if validate_args:
+        _binary_precision_recall_curve_arg_validation(thresholds, ignore_index)
+        _binary_precision_recall_curve_tensor_validation(preds, target, ignore_index)
+    remove_ignored = not (preds.is_cuda and thresholds is not None)
+    p_f, t_f, thr = _binary_precision_recall_curve_format(preds, target, thresholds, ignore_index, remove_ignored=remove_ignored)
+    state = _binary_precision_recall_curve_update(p_f, t_f, thr, ignore_index if not remove_ignored else None)
+    fpr, tpr, thresholds_out = _binary_roc_compute(state, thr)
+    return _sens_at_spec(fpr, tpr, thresholds_out, min_specificity)
+
+
+def multiclass_sensitivity_at_specificity(
+    preds: Tensor,
+    target: Tensor,
+    num_classes: int,
+    min_specificity: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tuple[Tensor, Tensor]:
+    """Per-class (sensitivity, threshold) at fixed specificity."""
+    if validate_args:
+        _multiclass_precision_recall_curve_arg_validation(num_classes, thresholds, ignore_index)
+        _multiclass_precision_recall_curve_tensor_validation(preds, target, num_classes, ignore_index)
+    remove_ignored = not (preds.is_cuda and thresholds is not None)
+    p_f, t_f, thr = _multiclass_precision_recall_curve_format(
+        preds, target, num_classes, thresholds, ignore_index, remove_ignored=remove_ignored
+    )
+    state = _multiclass_precision_recall_curve_update(
+        p_f, t_f, num_classes, thr, None, ignore_index if not remove_ignored else None
+    )
+    fpr, tpr, thresholds_out = _multiclass_roc_compute(state, num_classes, thr)
+    if isinstance(fpr, Tensor):
+        res = [_sens_at_spec(fpr[i], tpr[i], thresholds_out, min_specificity) for i in range(num_classes)]
+    else:
+        res = [_sens_at_spec(fpr[i], tpr[i], thresholds_out[i], min_specificity) for i in range(num_classes)]
+    return torch.stack([r[0] for r in res]), torch.stack([torch.as_tensor(r[1]) for r in res])
+
+
+def multilabel_sensitivity_at_specificity(
+    preds: Tensor,
+    target: Tensor,
+    num_labels: int,
+    min_specificity: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tuple[Tensor, Tensor]:
+    """Per-label (sensitivity, threshold) at fixed specificity."""
+    if validate_args:
+        _multilabel_precision_recall_curve_arg_validation(num_labels, thresholds, ignore_index)
+        _multilabel_precision_recall_curve_tensor_validation(preds, target, num_labels, ignore_index)
+    remove_ignored = not preds.is_cuda
+    p_f, t_f, thr = _multilabel_precision_recall_curve_format(
+        preds, target, num_labels, thresholds, ignore_index, remove_ignored=remove_ignored
+    )
+    state = _multilabel_precision_recall_curve_update(p_f, t_f, num_labels, thr, ignore_index if not remove_ignored else None)
+    fpr, tpr, thresholds_out = _multilabel_roc_compute(state, num_labels, thr, ignore_index)
+    if isinstance(fpr, Tensor):
+        res = [_sens_at_spec(fpr[i], tpr[i], thresholds_out, min_specificity) for i in range(num_labels)]
+    else:
+        res = [_sens_at_spec(fpr[i], tpr[i], thresholds_out[i], min_specificity) for i in range(num_labels)]
+    return torch.stack([r[0] for r in res]), torch.stack([torch.as_tensor(r[1]) for r in res])
+
+
+def binary_specificity_at_sensitivity(
+    preds: Tensor,
+    target: Tensor,
+    min_sensitivity: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tuple[Tensor, Tensor]:
+    """(max specificity with sensitivity >= min_sensitivity, threshold)."""
+    if validate_args:
+        _binary_precision_recall_curve_arg_validation(thresholds, ignore_index)
+        _binary_precision_recall_curve_tensor_validation(preds, target, ignore_index)
+    remove_ignored = not (preds.is_cuda and thresholds is not None)
+    p_f, t_f, thr = _binary_precision_recall_curve_format(preds, target, thresholds, ignore_index, remove_ignored=remove_ignored)
+    state = _binary_precision_recall_curve_update(p_f, t_f, thr, ignore_index if not remove_ignored else None)
+    fpr, tpr, thresholds_out = _binary_roc_compute(state, thr)
+    return _spec_at_sens(fpr, tpr, thresholds_out, min_sensitivity)
+
+
+def multiclass_specificity_at_sensitivity(
+    preds: Tensor,
+    target: Tensor,
+    num_classes: int,
+    min_sensitivity: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tuple[Tensor, Tensor]:
+    """Per-class (specificity, threshold) at fixed sensitivity."""
+    if validate_args:
+        _multiclass_precision_recall_curve_arg_validation(num_classes, thresholds, ignore_index)
+        _multiclass_precision_recall_curve_tensor_validation(preds, target, num_classes, ignore_index)
+    remove_ignored = not (preds.is_cuda and thresholds is not None)
+    p_f, t_f, thr = _multiclass_precision_recall_curve_format(
+        preds, target, num_classes, thresholds, ignore_index, remove_ignored=remove_ignored
+    )
+    state = _multiclass_precision_recall_curve_update(
+        p_f, t_f, num_classes, thr, None, ignore_index if not remove_ignored else None
+    )
+    fpr, tpr, thresholds_out = _multiclass_roc_compute(state, num_classes, thr)
+    if isinstance(fpr, Tensor):
+        res = [_spec_at_sens(fpr[i], tpr[i], thresholds_out, min_sensitivity) for i in range(num_classes)]
+    else:
+        res = [_spec_at_sens(fpr[i], tpr[i], thresholds_out[i], min_sensitivity) for i in range(num_classes)]
+    return torch.stack([r[0] for r in res]), torch.stack([torch.as_tensor(r[1]) for r in res])
+
+
+def multilabel_specificity_at_sensitivity(
+    preds: Tensor,
+    target: Tensor,
+    num_labels: int,
+    min_sensitivity: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tuple[Tensor, Tensor]:
+    """Per-label (specificity, threshold) at fixed sensitivity."""
+    if validate_args:
+        _multilabel_precision_recall_curve_arg_validation(num_labels, thresholds, ignore_index)
+        _multilabel_precision_recall_curve_tensor_validation(preds, target, num_labels, ignore_index)
+    remove_ignored = not preds.is_cuda
+    p_f, t_f, thr = _multilabel_precision_recall_curve_format(
+        preds, target, num_labels, thresholds, ignore_index, remove_ignored=remove_ignored
+    )
+    state = _multilabel_precision_recall_curve_update(p_f, t_f, num_labels, thr, ignore_index if not remove_ignored else None)
+    fpr, tpr, thresholds_out = _multilabel_roc_compute(state, num_labels, thr, ignore_index)
+    if isinstance(fpr, Tensor):
+        res = [_spec_at_sens(fpr[i], tpr[i], thresholds_out, min_sensitivity) for i in range(num_labels)]
+    else:
+        res = [_spec_at_sens(fpr[i], tpr[i], thresholds_out[i], min_sensitivity) for i in range(num_labels)]
+    return torch.stack([r[0] for r in res]), torch.stack([torch.as_tensor(r[1]) for r in res])
+
+
+# -------------------------------------------------------------------- LogAUC
+def _logauc_compute(fpr: Tensor, tpr: Tensor, fpr_range: Tuple[float, float]) -> Tensor:
+    """Area under the ROC in log10(fpr) space over ``fpr_range``, normalized."""
+    lo, hi = fpr_range
+    device = fpr.device
+    # interpolate tpr at the range bounds
+    from metrics_amd.utilities.data import interp
+
+    bounds = torch.tensor([lo, hi], dtype=fpr.dtype, device=device)
+    # fpr is increasing; add eps for duplicate handling
+    uniq, idx = torch.unique(fpr, return_inverse=True)
+    # take max tpr per unique fpr (best case at that fpr)
+    tpr_u = torch.zeros_like(uniq)
+    tpr_u.scatter_reduce_(0, idx, tpr, reduce="amax")
+    tpr_bounds = interp(bounds, uniq, tpr_u)
+    mask = (uniq > lo) & (uniq < hi)
+    x = torch.cat([bounds[:1], uniq[mask], bounds[1:]])
+    y = torch.cat([tpr_bounds[:1], tpr_u[mask], tpr_bounds[1:]])
+    x = torch.clamp(x, min=1e-10)
+    area = torch.trapz(y, torch.log10(x))
+    return area / (torch.log10(torch.tensor(hi)) - torch.log10(torch.tensor(lo)))
+
+
+def binary_logauc(
+    preds: Tensor,
+    target: Tensor,
+    fpr_range: Tuple[float, float] = (0.001, 0.1),
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tensor:
+    """Log-scaled AUC over an fpr range (virtual-screening style)."""
+    if validate_args:
+        _binary_precision_recall_curve_arg_validation(thresholds, ignore_index)
+        _binary_precision_recall_curve_tensor_validation(preds, target, ignore_index)
+        if not isinstance(fpr_range, (tuple, list)) or len(fpr_range) != 2 or fpr_range[0] >= fpr_range[1]:
+            raise ValueError(f"The `fpr_range` should be a tuple of two floats (low, high), but got {fpr_range}")
+    remove_ignored = not (preds.is_cuda and thresholds is not None)
+    p_f, t_f, thr = _binary_precision_recall_curve_format(preds, target, thresholds, ignore_index, remove_ignored=remove_ignored)
+    state = _binary_precision_recall_curve_update(p_f, t_f, thr, ignore_index if not remove_ignored else None)
+    fpr, tpr, _ = _binary_roc_compute(state, thr)
+    return _logauc_compute(fpr, tpr, tuple(fpr_range))
+
+
+def multiclass_logauc(
+    preds: Tensor,
+    target: Tensor,
+    num_classes: int,
+    fpr_range: Tuple[float, float] = (0.001, 0.1),
+    average: Optional[str] = "macro",
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tensor:
+    """Per-class (or macro) log AUC."""
+    if validate_args:
+        _multiclass_precision_recall_curve_arg_validation(num_classes, thresholds, ignore_index)
+        _multiclass_precision_recall_curve_tensor_validation(preds, target, num_classes, ignore_index)
+    remove_ignored = not (preds.is_cuda and thresholds is not None)
+    p_f, t_f, thr = _multiclass_precision_recall_curve_format(
+        preds, target, num_classes, thresholds, ignore_index, remove_ignored=remove_ignored
+    )
+    state = _multiclass_precision_recall_curve_update(
+        p_f, t_f, num_classes, thr, None, ignore_index if not remove_ignored else None
+    )
+    fpr, tpr, _ = _multiclass_roc_compute(state, num_classes, thr)
+    scores = torch.stack([_logauc_compute(fpr[i], tpr[i], tuple(fpr_range)) for i in range(num_classes)])
+    if average == "macro":
+        return scores.mean()
+    return scores
+
+
+def multilabel_logauc(
+    preds: Tensor,
+    target: Tensor,
+    num_labels: int,
+    fpr_range: Tuple[float, float] = (0.001, 0.1),
+    average: Optional[str] = "macro",
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Tensor:
+    """Per-label (or macro) log AUC."""
+    if validate_args:
+        _multilabel_precision_recall_curve_arg_validation(num_labels, thresholds, ignore_index)
+        _multilabel_precision_recall_curve_tensor_validation(preds, target, num_labels, ignore_index)
+    remove_ignored = not preds.is_cuda
+    p_f, t_f, thr = _multilabel_precision_recall_curve_format(
+        preds, target, num_labels, thresholds, ignore_index, remove_ignored=remove_ignored
+    )
+    state = _multilabel_precision_recall_curve_update(p_f, t_f, num_labels, thr, ignore_index if not remove_ignored else None)
+    fpr, tpr, _ = _multilabel_roc_compute(state, num_labels, thr, ignore_index)
+    scores = torch.stack([_logauc_compute(fpr[i], tpr[i], tuple(fpr_range)) for i in range(num_labels)])
+    if average == "macro":
+        return scores.mean()
+    return scores
