@@ -54,7 +54,8 @@ def _binary_groups_stat_scores(
     split_sizes = torch.bincount(indexes, minlength=num_groups).tolist()
     group_preds = list(torch.split(preds, split_sizes, dim=0))
     group_target = list(torch.split(target, split_sizes, dim=0))
-    return [_binary_stat_scores_update(p, t) for p, t in zip(group_preds, group_target)]
+    # _binary_stat_scores_update expects the (N, flattened...) 2D layout
+    return [_binary_stat_scores_update(p.unsqueeze(-1), t.unsqueeze(-1)) for p, t in zip(group_preds, group_target)]
 
 
 def binary_groups_stat_rates(
