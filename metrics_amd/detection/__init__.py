@@ -5,6 +5,7 @@ from metrics_amd.detection.iou import (
     IntersectionOverUnion,
 )
 from metrics_amd.detection.mean_ap import MeanAveragePrecision
+from metrics_amd.detection.panoptic_quality import ModifiedPanopticQuality, PanopticQuality
 
 __all__ = [
     "CompleteIntersectionOverUnion",
@@ -12,4 +13,6 @@ __all__ = [
     "GeneralizedIntersectionOverUnion",
     "IntersectionOverUnion",
     "MeanAveragePrecision",
+    "ModifiedPanopticQuality",
+    "PanopticQuality",
 ]

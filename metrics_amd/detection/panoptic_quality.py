@@ -1,0 +1,259 @@
+"""Panoptic quality (+ modified PQ).
+
+Parity: torchmetrics ``detection/panoptic_qualities.py`` /
+``functional/detection/_panoptic_quality_common.py``: segment matching at
+IoU > 0.5 per category with void handling; PQ = sum_iou / (TP + FP/2 + FN/2).
+"""
+from __future__ import annotations
+
+from typing import Any, Collection, Dict, List, Optional, Set, Tuple
+
+import torch
+from torch import Tensor
+
+from metrics_amd.metric import Metric
+
+
+def _parse_categories(things: Collection[int], stuffs: Collection[int]) -> Tuple[Set[int], Set[int]]:
+    things = set(int(t) for t in things)
+    stuffs = set(int(s) for s in stuffs)
+    if things & stuffs:
+        raise ValueError(f"Expected arguments `things` and `stuffs` to have distinct keys, but got {things} and {stuffs}")
+    if not (things | stuffs):
+        raise ValueError("At least one of `things` and `stuffs` must be non-empty.")
+    return things, stuffs
+
+
+def _validate_inputs(preds: Tensor, target: Tensor) -> None:
+    if not isinstance(preds, Tensor) or not isinstance(target, Tensor):
+        raise TypeError("Expected argument `preds` and `target` to be of type `torch.Tensor`")
+    if preds.shape != target.shape:
+        raise ValueError(
+            f"Expected argument `preds` and `target` to have the same shape, but got {preds.shape} and {target.shape}"
+        )
+    if preds.dim() < 3 or preds.shape[-1] != 2:
+        raise ValueError(
+            "Expected argument `preds` to have at least 3 dimensions and the final dimension equal to 2"
+        )
+
+
+def _segments(pan: Tensor, things: Set[int], stuffs: Set[int]) -> Dict[Tuple[int, int], Tensor]:
+    """Map (category, instance) -> boolean mask. Stuff collapses instance ids to 0."""
+    cats = pan[..., 0]
+    insts = pan[..., 1]
+    segs: Dict[Tuple[int, int], Tensor] = {}
+    pairs = torch.stack([cats.flatten(), insts.flatten()], dim=1)
+    uniq = torch.unique(pairs, dim=0)
+    for cat, inst in uniq.tolist():
+        if cat in stuffs:
+            key = (cat, 0)
+            mask = cats == cat
+        elif cat in things:
+            key = (cat, inst)
+            mask = (cats == cat) & (insts == inst)
+        else:
+            continue  # void
+        if key not in segs:
+            segs[key] = mask
+    return segs
+
+
+def _panoptic_quality_update_sample(
+    pred: Tensor,
+    target: Tensor,
+    things: Set[int],
+    stuffs: Set[int],
+    modified_metric_stuffs: Optional[Set[int]] = None,
+) -> Tuple[Dict[int, float], Dict[int, int], Dict[int, int], Dict[int, int]]:
+    """One image: per-category (iou_sum, tp, fp, fn)."""
+    void_pred = ~torch.isin(pred[..., 0], torch.tensor(sorted(things | stuffs), device=pred.device))
+    void_target = ~torch.isin(target[..., 0], torch.tensor(sorted(things | stuffs), device=target.device))
+
+    pred_segs = _segments(pred, things, stuffs)
+    target_segs = _segments(target, things, stuffs)
+
+    iou_sum: Dict[int, float] = {}
+    tp: Dict[int, int] = {}
+    fp: Dict[int, int] = {}
+    fn: Dict[int, int] = {}
+    matched_pred: Set[Tuple[int, int]] = set()
+    matched_target: Set[Tuple[int, int]] = set()
+
+    for tkey, tmask in target_segs.items():
+        cat = tkey[0]
+        if modified_metric_stuffs is not None and cat in modified_metric_stuffs:
+            continue  # handled separately below
+        best_iou, best_pkey = 0.0, None
+        for pkey, pmask in pred_segs.items():
+            if pkey[0] != cat or pkey in matched_pred:
+                continue
+            inter = (tmask & pmask).sum().item()
+            if inter == 0:
+                continue
+            union = (tmask | pmask).sum().item() - (pmask & void_target).sum().item()
+            iou = inter / union if union > 0 else 0.0
+            if iou > best_iou:
+                best_iou, best_pkey = iou, pkey
+        if best_iou > 0.5:
+            matched_pred.add(best_pkey)
+            matched_target.add(tkey)
+            iou_sum[cat] = iou_sum.get(cat, 0.0) + best_iou
+            tp[cat] = tp.get(cat, 0) + 1
+
+    # modified PQ: stuff categories scored by pixel-level IoU without the 0.5 gate
+    if modified_metric_stuffs:
+        for tkey, tmask in target_segs.items():
+            cat = tkey[0]
+            if cat not in modified_metric_stuffs:
+                continue
+            pmask = None
+            for pkey, pm in pred_segs.items():
+                if pkey[0] == cat:
+                    pmask = pm if pmask is None else (pmask | pm)
+                    matched_pred.add(pkey)
+            if pmask is None:
+                fn[cat] = fn.get(cat, 0) + 1
+                continue
+            inter = (tmask & pmask).sum().item()
+            union = (tmask | pmask).sum().item()
+            iou = inter / union if union > 0 else 0.0
+            iou_sum[cat] = iou_sum.get(cat, 0.0) + iou
+            tp[cat] = tp.get(cat, 0) + 1
+            matched_target.add(tkey)
+
+    for tkey, tmask in target_segs.items():
+        if tkey not in matched_target and not (
+            modified_metric_stuffs and tkey[0] in modified_metric_stuffs
+        ):
+            fn[tkey[0]] = fn.get(tkey[0], 0) + 1
+    for pkey, pmask in pred_segs.items():
+        if pkey in matched_pred:
+            continue
+        # FPs with >50% void overlap in the target are ignored (COCO PQ rule)
+        void_frac = (pmask & void_target).sum().item() / max(pmask.sum().item(), 1)
+        if void_frac > 0.5:
+            continue
+        fp[pkey[0]] = fp.get(pkey[0], 0) + 1
+    return iou_sum, tp, fp, fn
+
+
+class PanopticQuality(Metric):
+    """Panoptic quality for (category, instance)-encoded panoptic maps."""
+
+    is_differentiable = False
+    higher_is_better = True
+    full_state_update = False
+    plot_lower_bound: float = 0.0
+    plot_upper_bound: float = 1.0
+
+    iou_sum: Tensor
+    true_positives: Tensor
+    false_positives: Tensor
+    false_negatives: Tensor
+
+    _modified: bool = False
+
+    def __init__(
+        self,
+        things: Collection[int],
+        stuffs: Collection[int],
+        allow_unknown_preds_category: bool = False,
+        return_sq_and_rq: bool = False,
+        return_per_class: bool = False,
+        **kwargs: Any,
+    ) -> None:
+        super().__init__(**kwargs)
+        self.things, self.stuffs = _parse_categories(things, stuffs)
+        self.allow_unknown_preds_category = allow_unknown_preds_category
+        self.return_sq_and_rq = return_sq_and_rq
+        self.return_per_class = return_per_class
+        cats = sorted(self.things | self.stuffs)
+        self._cat_to_idx = {c: i for i, c in enumerate(cats)}
+        self._cats = cats
+        n = len(cats)
+        self.add_state("iou_sum", default=torch.zeros(n, dtype=torch.double), dist_reduce_fx="sum")
+        self.add_state("true_positives", default=torch.zeros(n, dtype=torch.long), dist_reduce_fx="sum")
+        self.add_state("false_positives", default=torch.zeros(n, dtype=torch.long), dist_reduce_fx="sum")
+        self.add_state("false_negatives", default=torch.zeros(n, dtype=torch.long), dist_reduce_fx="sum")
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Match segments per image and accumulate per-category counts."""
+        _validate_inputs(preds, target)
+        if not self.allow_unknown_preds_category:
+            known = torch.tensor(self._cats, device=preds.device)
+            unknown = ~torch.isin(preds[..., 0], known)
+            if bool(unknown.any()):
+                raise ValueError(
+                    "Unknown categories found in `preds`. Set `allow_unknown_preds_category=True` to map them to void."
+                )
+        batch_preds = preds.reshape(-1, *preds.shape[-3:]) if preds.dim() > 3 else preds.unsqueeze(0)
+        batch_target = target.reshape(-1, *target.shape[-3:]) if target.dim() > 3 else target.unsqueeze(0)
+        mod_stuffs = self.stuffs if self._modified else None
+        for p, t in zip(batch_preds, batch_target):
+            iou_sum, tp, fp, fn = _panoptic_quality_update_sample(p, t, self.things, self.stuffs, mod_stuffs)
+            for cat, v in iou_sum.items():
+                self.iou_sum[self._cat_to_idx[cat]] += v
+            for cat, v in tp.items():
+                self.true_positives[self._cat_to_idx[cat]] += v
+            for cat, v in fp.items():
+                self.false_positives[self._cat_to_idx[cat]] += v
+            for cat, v in fn.items():
+                self.false_negatives[self._cat_to_idx[cat]] += v
+
+    def compute(self) -> Tensor:
+        """PQ (optionally SQ and RQ / per class)."""
+        denom = self.true_positives + 0.5 * self.false_positives + 0.5 * self.false_negatives
+        valid = denom > 0
+        pq_per = torch.where(valid, self.iou_sum / denom.clamp(min=1e-9), torch.zeros_like(self.iou_sum))
+        sq_per = torch.where(
+            self.true_positives > 0, self.iou_sum / self.true_positives.clamp(min=1), torch.zeros_like(self.iou_sum)
+        )
+        rq_per = torch.where(valid, self.true_positives / denom.clamp(min=1e-9), torch.zeros_like(self.iou_sum))
+
+        if self.return_per_class:
+            if self.return_sq_and_rq:
+                return torch.stack([pq_per, sq_per, rq_per], dim=-1).unsqueeze(0)
+            return pq_per.unsqueeze(0)
+        pq = pq_per[valid].mean() if valid.any() else torch.tensor(0.0, dtype=torch.double)
+        if self.return_sq_and_rq:
+            sq = sq_per[valid].mean() if valid.any() else torch.tensor(0.0, dtype=torch.double)
+            rq = rq_per[valid].mean() if valid.any() else torch.tensor(0.0, dtype=torch.double)
+            return torch.stack([pq, sq, rq])
+        return pq
+
+    def plot(self, val=None, ax=None):
+        return self._plot(val, ax)
+
+
+class ModifiedPanopticQuality(PanopticQuality):
+    """Modified PQ (Porzi et al. 2019): stuff scored by per-pixel IoU, no 0.5 gate."""
+
+    _modified = True
+
+
+def panoptic_quality(
+    preds: Tensor,
+    target: Tensor,
+    things: Collection[int],
+    stuffs: Collection[int],
+    allow_unknown_preds_category: bool = False,
+    return_sq_and_rq: bool = False,
+    return_per_class: bool = False,
+) -> Tensor:
+    """Functional PQ."""
+    m = PanopticQuality(things, stuffs, allow_unknown_preds_category, return_sq_and_rq, return_per_class)
+    m.update(preds, target)
+    return m.compute()
+
+
+def modified_panoptic_quality(
+    preds: Tensor,
+    target: Tensor,
+    things: Collection[int],
+    stuffs: Collection[int],
+    allow_unknown_preds_category: bool = False,
+) -> Tensor:
+    """Functional modified PQ."""
+    m = ModifiedPanopticQuality(things, stuffs, allow_unknown_preds_category)
+    m.update(preds, target)
+    return m.compute()

@@ -4,3 +4,4 @@ from metrics_amd.functional.detection.iou import (
     generalized_intersection_over_union,
     intersection_over_union,
 )
+from metrics_amd.detection.panoptic_quality import modified_panoptic_quality, panoptic_quality
