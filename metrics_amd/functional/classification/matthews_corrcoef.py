@@ -33,9 +33,9 @@ def _matthews_corrcoef_reduce(confmat: Tensor) -> Tensor:
     c = torch.trace(confmat).float()
     s = confmat.sum().float()
 
-    cov_ytyp = c * s - sum(tk * pk)
-    cov_ypyp = s**2 - sum(pk * pk)
-    cov_ytyt = s**2 - sum(tk * tk)
+    cov_ytyp = c * s - (tk * pk).sum()
+    cov_ypyp = s**2 - (pk * pk).sum()
+    cov_ytyt = s**2 - (tk * tk).sum()
 
     numerator = cov_ytyp
     denom = cov_ypyp * cov_ytyt
