@@ -120,6 +120,13 @@ def main() -> None:
     value = n_gpus * n_metrics * args.steps / elapsed
     ms_per_step = 1000.0 * elapsed / args.steps
 
+    # Baseline: the reference (torchmetrics 1.7.0dev) measured on the SAME
+    # config on 1x MI355X via _refbench/run_ref_bench.py (r1, 2026-09-12):
+    # 360.6 metric-updates/s (44.37 ms/step). BASELINE.json has no published
+    # numbers, so this measured same-hardware number is the comparison point.
+    REFERENCE_UPDATES_PER_SEC_1GPU = 360.6
+    vs_baseline = value / (REFERENCE_UPDATES_PER_SEC_1GPU * n_gpus) if use_gpu else None
+
     if rank == 0:
         print(
             json.dumps({
@@ -132,7 +139,7 @@ def main() -> None:
                 "ms_per_step": ms_per_step,
                 "higher_is_better": True,
                 "scaling": "weak",
-                "vs_baseline": None,
+                "vs_baseline": vs_baseline,
                 "dtype": "bf16" if use_gpu else "fp32",
                 "data": "synthetic",
                 "config": {
