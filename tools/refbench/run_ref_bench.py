@@ -1,0 +1,103 @@
+"""Run the REFERENCE (torchmetrics) on the identical bench config for baseline numbers."""
+import argparse
+import json
+import sys
+import time
+from pathlib import Path
+
+# staged copy of the reference source (gitignored; see stage_reference.sh)
+sys.path.insert(0, str(Path(__file__).parent / "_staged"))
+sys.path.insert(0, str(Path(__file__).parent))
+
+import torch  # noqa: E402
+import torchmetrics as tm  # noqa: E402
+from torchmetrics.classification import (  # noqa: E402
+    MulticlassAccuracy,
+    MulticlassAUROC,
+    MulticlassAveragePrecision,
+    MulticlassCohenKappa,
+    MulticlassConfusionMatrix,
+    MulticlassExactMatch,
+    MulticlassF1Score,
+    MulticlassFBetaScore,
+    MulticlassHammingDistance,
+    MulticlassJaccardIndex,
+    MulticlassMatthewsCorrCoef,
+    MulticlassNegativePredictiveValue,
+    MulticlassPrecision,
+    MulticlassRecall,
+    MulticlassSpecificity,
+)
+
+
+def build_collection(num_classes, device, curve_thresholds=200):
+    kw = dict(num_classes=num_classes, validate_args=False)
+    metrics = {
+        "acc_micro": MulticlassAccuracy(average="micro", **kw),
+        "acc_macro": MulticlassAccuracy(average="macro", **kw),
+        "precision": MulticlassPrecision(average="macro", **kw),
+        "recall": MulticlassRecall(average="macro", **kw),
+        "f1": MulticlassF1Score(average="macro", **kw),
+        "fbeta2": MulticlassFBetaScore(beta=2.0, average="macro", **kw),
+        "specificity": MulticlassSpecificity(average="macro", **kw),
+        "npv": MulticlassNegativePredictiveValue(average="macro", **kw),
+        "hamming": MulticlassHammingDistance(average="macro", **kw),
+        "jaccard": MulticlassJaccardIndex(average="macro", **kw),
+        "exact_match": MulticlassExactMatch(**kw),
+        "cohen_kappa": MulticlassCohenKappa(**kw),
+        "mcc": MulticlassMatthewsCorrCoef(**kw),
+        "confmat": MulticlassConfusionMatrix(**kw),
+        "auroc": MulticlassAUROC(average="macro", thresholds=curve_thresholds, **kw),
+        "avg_precision": MulticlassAveragePrecision(average="macro", thresholds=curve_thresholds, **kw),
+    }
+    return tm.MetricCollection(metrics).to(device)
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--steps", type=int, default=64)
+    parser.add_argument("--warmup", type=int, default=8)
+    parser.add_argument("--batch", type=int, default=8192)
+    parser.add_argument("--classes", type=int, default=1000)
+    parser.add_argument("--compute-every", type=int, default=32)
+    parser.add_argument("--dtype", default="bf16")
+    args = parser.parse_args()
+
+    device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
+    dtype = torch.bfloat16 if (args.dtype == "bf16" and device.type == "cuda") else torch.float32
+    torch.manual_seed(1234)
+    n_unique = 4
+    preds = [torch.randn(args.batch, args.classes, device=device, dtype=dtype) for _ in range(n_unique)]
+    target = [torch.randint(0, args.classes, (args.batch,), device=device) for _ in range(n_unique)]
+
+    coll = build_collection(args.classes, device)
+
+    def one_step(i):
+        coll.update(preds[i % n_unique], target[i % n_unique])
+        if args.compute_every and (i + 1) % args.compute_every == 0:
+            coll.compute()
+
+    for i in range(args.warmup):
+        one_step(i)
+    coll.compute()
+    coll.reset()
+
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(i)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    print(json.dumps({
+        "framework": "torchmetrics-reference",
+        "metric": "metric_updates_per_sec",
+        "value": 16 * args.steps / elapsed,
+        "ms_per_step": 1000 * elapsed / args.steps,
+        "config": vars(args),
+    }))
+
+
+if __name__ == "__main__":
+    main()
