@@ -22,7 +22,7 @@ ARCH = os.environ.get("METRICS_AMD_ARCH", "gfx950")
 
 
 def build(verbose: bool = True) -> Path:
-    """Compile csrc/*.hip into metrics_amd/_lib/libmetrics_hip.so; returns the path."""
+    """Compile csrc/*.hip (gfx950) and csrc/*.cpp (host) into metrics_amd/_lib/."""
     LIBDIR.mkdir(exist_ok=True)
     out = LIBDIR / LIBNAME
     srcs = sorted(str(p) for p in CSRC.glob("*.hip"))
@@ -40,7 +40,20 @@ def build(verbose: bool = True) -> Path:
     if verbose:
         print("+", " ".join(cmd), file=sys.stderr)
     subprocess.run(cmd, check=True)
+
+    # host-side native components (COCO matcher): plain g++, no GPU linkage
+    cpp_srcs = sorted(str(p) for p in CSRC.glob("*.cpp"))
+    if cpp_srcs:
+        cpu_out = LIBDIR / "libmetrics_cpu.so"
+        cmd = ["g++", "-O3", "-std=c++17", "-fPIC", "-shared", *cpp_srcs, "-o", str(cpu_out)]
+        if verbose:
+            print("+", " ".join(cmd), file=sys.stderr)
+        subprocess.run(cmd, check=True)
     return out
+
+
+def cpu_lib_path() -> Path:
+    return LIBDIR / "libmetrics_cpu.so"
 
 
 def lib_path() -> Path:

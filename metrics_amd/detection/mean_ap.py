@@ -163,184 +163,257 @@ class MeanAveragePrecision(Metric):
                 area = default_area
             self.groundtruth_area.append(area)
 
-    def _evaluate_image(
-        self, det_boxes: Tensor, det_scores: Tensor, gt_boxes: Tensor, gt_crowd: Tensor, gt_area: Tensor,
-        area_rng: Tuple[float, float], max_det: int, iou_thrs: Tensor,
-    ) -> Optional[Tuple[Tensor, Tensor, Tensor, int]]:
-        """COCO evaluateImg for one (image, class): returns (scores, det_matched, det_ignore, n_valid_gt)."""
-        n_gt, n_dt = gt_boxes.shape[0], det_boxes.shape[0]
-        if n_gt == 0 and n_dt == 0:
-            return None
-        device = det_boxes.device if n_dt else gt_boxes.device
+    @staticmethod
+    def _iou_np(det: "np.ndarray", gt: "np.ndarray", iscrowd: "np.ndarray") -> "np.ndarray":
+        """All-pairs IoU on xyxy numpy boxes; crowd gts use intersection/det-area."""
+        import numpy as np
+
+        if det.shape[0] == 0 or gt.shape[0] == 0:
+            return np.zeros((det.shape[0], gt.shape[0]))
+        ix1 = np.maximum(det[:, None, 0], gt[None, :, 0])
+        iy1 = np.maximum(det[:, None, 1], gt[None, :, 1])
+        ix2 = np.minimum(det[:, None, 2], gt[None, :, 2])
+        iy2 = np.minimum(det[:, None, 3], gt[None, :, 3])
+        inter = np.clip(ix2 - ix1, 0, None) * np.clip(iy2 - iy1, 0, None)
+        area_d = (det[:, 2] - det[:, 0]) * (det[:, 3] - det[:, 1])
+        area_g = (gt[:, 2] - gt[:, 0]) * (gt[:, 3] - gt[:, 1])
+        union = area_d[:, None] + area_g[None, :] - inter
+        union = np.where(iscrowd[None, :] == 1, area_d[:, None], union)
+        return np.where(union > 0, inter / np.maximum(union, 1e-9), 0.0)
+
+    @staticmethod
+    def _evaluate_img_np(
+        ious: "np.ndarray",
+        scores: "np.ndarray",
+        det_areas: "np.ndarray",
+        gt_ignore_base: "np.ndarray",
+        gt_crowd: "np.ndarray",
+        gt_area: "np.ndarray",
+        area_rng: Tuple[float, float],
+        max_det: int,
+        iou_thrs: "np.ndarray",
+    ):
+        """COCO evaluateImg: greedy matching per IoU threshold (numpy, no device syncs).
+
+        ``ious``/``scores`` are already sorted by score desc and capped at the
+        TOP-level max_det; gts are already sorted ignore-last for the 'all'
+        range — per-area ignore flags re-sort here.
+        """
+        import numpy as np
+
+        n_dt, n_gt = ious.shape
+        gt_ignore = gt_ignore_base | (gt_area < area_rng[0]) | (gt_area > area_rng[1])
+        order = np.argsort(gt_ignore, kind="stable")
+        gt_ignore = gt_ignore[order]
+        crowd = gt_crowd[order]
+        ious_o = ious[:, order]
+
         T = len(iou_thrs)
-
-        gt_ignore = (gt_crowd == 1) | (gt_area < area_rng[0]) | (gt_area > area_rng[1])
-        # sort gts: non-ignored first (stable)
-        gt_order = torch.argsort(gt_ignore.to(torch.uint8), stable=True)
-        gt_boxes = gt_boxes[gt_order]
-        gt_ignore = gt_ignore[gt_order]
-        gt_crowd = gt_crowd[gt_order]
-
-        # sort dets by score desc, cap at max_det
-        dt_order = torch.argsort(det_scores, descending=True, stable=True)[:max_det]
-        det_boxes = det_boxes[dt_order]
-        det_scores = det_scores[dt_order]
-        n_dt = det_boxes.shape[0]
-
-        if n_dt and n_gt:
-            ious = ops.box_iou_pairwise(det_boxes, gt_boxes, "iou")
-            # crowd gts use union-over-det IoU semantics (iscrowd => iou vs det area)
-            if bool(gt_crowd.any()):
-                crowd_cols = torch.nonzero(gt_crowd == 1).flatten()
-                if crowd_cols.numel():
-                    da = ((det_boxes[:, 2] - det_boxes[:, 0]) * (det_boxes[:, 3] - det_boxes[:, 1])).clamp(min=1e-9)
-                    for c in crowd_cols.tolist():
-                        g = gt_boxes[c]
-                        ix1 = torch.maximum(det_boxes[:, 0], g[0])
-                        iy1 = torch.maximum(det_boxes[:, 1], g[1])
-                        ix2 = torch.minimum(det_boxes[:, 2], g[2])
-                        iy2 = torch.minimum(det_boxes[:, 3], g[3])
-                        inter = (ix2 - ix1).clamp(min=0) * (iy2 - iy1).clamp(min=0)
-                        ious[:, c] = inter / da
-        else:
-            ious = torch.zeros(n_dt, n_gt, device=device)
-
-        gt_matched = torch.zeros(T, n_gt, dtype=torch.bool, device=device)
-        dt_matched = torch.zeros(T, n_dt, dtype=torch.bool, device=device)
-        dt_ignore = torch.zeros(T, n_dt, dtype=torch.bool, device=device)
-
-        # greedy matching (CPU lists — small per (img,cls) sizes)
-        ious_c = ious.cpu()
-        gt_ignore_c = gt_ignore.cpu().tolist()
-        gt_crowd_c = gt_crowd.cpu().tolist()
-        for ti, t in enumerate(iou_thrs.tolist()):
-            gtm = gt_matched[ti]
+        gtm = np.zeros((T, n_gt), dtype=bool)
+        dtm = np.zeros((T, n_dt), dtype=bool)
+        dti = np.zeros((T, n_dt), dtype=bool)
+        for ti in range(T):
+            t = iou_thrs[ti]
             for d in range(n_dt):
-                best_iou = min(t, 1 - 1e-10)
+                best = min(t, 1 - 1e-10)
                 m = -1
+                row = ious_o[d]
                 for g in range(n_gt):
-                    if gtm[g] and not gt_crowd_c[g]:
+                    if gtm[ti, g] and not crowd[g]:
                         continue
-                    if m > -1 and not gt_ignore_c[m] and gt_ignore_c[g]:
-                        break  # gts sorted: once past non-ignored with a match, stop
-                    if ious_c[d, g] < best_iou:
+                    if m > -1 and not gt_ignore[m] and gt_ignore[g]:
+                        break
+                    if row[g] < best:
                         continue
-                    best_iou = ious_c[d, g]
+                    best = row[g]
                     m = g
                 if m == -1:
                     continue
-                dt_ignore[ti, d] = gt_ignore_c[m]
-                dt_matched[ti, d] = True
-                gt_matched[ti, m] = True
-
-        # unmatched dets outside the area range are ignored
-        dt_areas = (det_boxes[:, 2] - det_boxes[:, 0]) * (det_boxes[:, 3] - det_boxes[:, 1])
-        dt_out_of_rng = (dt_areas < area_rng[0]) | (dt_areas > area_rng[1])
-        dt_ignore = dt_ignore | (~dt_matched & dt_out_of_rng.unsqueeze(0))
-
+                dti[ti, d] = gt_ignore[m]
+                dtm[ti, d] = True
+                gtm[ti, m] = True
+        out_of_rng = (det_areas < area_rng[0]) | (det_areas > area_rng[1])
+        dti |= ~dtm & out_of_rng[None, :]
         n_valid_gt = int((~gt_ignore).sum())
-        return det_scores, dt_matched, dt_ignore, n_valid_gt
+        return scores, dtm, dti, n_valid_gt
+
+    def _eval_class_numpy(self, db, gb, gc, ga, dt_off, gt_off, area_rngs, iou_thrs):
+        """Fallback matcher with the same packed interface as the native one."""
+        import numpy as np
+
+        n_imgs = len(dt_off) - 1
+        A = area_rngs.shape[0]
+        T = len(iou_thrs)
+        total_dt = int(dt_off[-1])
+        dtm = [np.zeros((T, total_dt), dtype=bool) for _ in range(A)]
+        dti = [np.zeros((T, total_dt), dtype=bool) for _ in range(A)]
+        npig = np.zeros(A, dtype=np.int64)
+        for img in range(n_imgs):
+            d0, d1 = int(dt_off[img]), int(dt_off[img + 1])
+            g0, g1 = int(gt_off[img]), int(gt_off[img + 1])
+            if d1 == d0 and g1 == g0:
+                continue
+            ious = self._iou_np(db[d0:d1], gb[g0:g1], gc[g0:g1])
+            dareas = (db[d0:d1, 2] - db[d0:d1, 0]) * (db[d0:d1, 3] - db[d0:d1, 1])
+            for ai in range(A):
+                _, m, ig, n_valid = self._evaluate_img_np(
+                    ious, np.zeros(d1 - d0), dareas, gc[g0:g1] == 1, gc[g0:g1], ga[g0:g1],
+                    (float(area_rngs[ai][0]), float(area_rngs[ai][1])), total_dt, iou_thrs,
+                )
+                dtm[ai][:, d0:d1] = m
+                dti[ai][:, d0:d1] = ig
+                npig[ai] += n_valid
+        return dtm, dti, npig
 
     def compute(self) -> Dict[str, Tensor]:
-        """COCO mAP/mAR summary over all accumulated images."""
+        """COCO mAP/mAR summary over all accumulated images.
+
+        Engine: one host transfer of all boxes, per-(image,class) IoU cached
+        across area ranges, numpy greedy matching, vectorized accumulation.
+        """
+        import numpy as np
+
         device = self.detection_boxes[0].device if self.detection_boxes else torch.device("cpu")
-        iou_thrs = torch.tensor(self.iou_thresholds, device=device)
-        rec_thrs = torch.tensor(self.rec_thresholds, device=device)
+        iou_thrs = np.array(self.iou_thresholds)
+        rec_thrs = np.array(self.rec_thresholds)
         T = len(self.iou_thresholds)
         n_imgs = len(self.detection_boxes)
         max_dets = self.max_detection_thresholds
         max_det_top = max_dets[-1]
 
-        all_labels = (
-            torch.cat(self.detection_labels + self.groundtruth_labels)
-            if n_imgs
-            else torch.zeros(0, dtype=torch.long)
-        )
-        classes = torch.unique(all_labels).tolist() if all_labels.numel() else []
+        # one transfer to host
+        det_boxes = [b.detach().cpu().numpy() for b in self.detection_boxes]
+        det_scores = [s_.detach().cpu().numpy() for s_ in self.detection_scores]
+        det_labels = [l.detach().cpu().numpy() for l in self.detection_labels]
+        gt_boxes = [b.detach().cpu().numpy() for b in self.groundtruth_boxes]
+        gt_labels = [l.detach().cpu().numpy() for l in self.groundtruth_labels]
+        gt_crowds = [c.detach().cpu().numpy() for c in self.groundtruth_crowds]
+        gt_areas = [a.detach().cpu().numpy() for a in self.groundtruth_area]
+
+        classes = sorted(set(np.concatenate(det_labels + gt_labels).tolist())) if n_imgs else []
 
         A = len(self._AREA_RANGES)
         M = len(max_dets)
         K = len(classes)
         R = len(self.rec_thresholds)
-        precision = -torch.ones(T, R, K, A, M, device=device)
-        recall = -torch.ones(T, K, A, M, device=device)
-        scores_out = -torch.ones(T, R, K, A, M, device=device)
+        precision = -np.ones((T, R, K, A, M))
+        recall = -np.ones((T, K, A, M))
+        scores_out = -np.ones((T, R, K, A, M))
 
-        for ki, cls in enumerate(classes):
-            # per image per-class slices
-            per_img = []
-            for i in range(n_imgs):
-                det_m = self.detection_labels[i] == cls
-                gt_m = self.groundtruth_labels[i] == cls
-                per_img.append((
-                    self.detection_boxes[i][det_m],
-                    self.detection_scores[i][det_m],
-                    self.groundtruth_boxes[i][gt_m],
-                    self.groundtruth_crowds[i][gt_m],
-                    self.groundtruth_area[i][gt_m],
-                ))
-            for ai, area_rng in enumerate(self._AREA_RANGES):
-                results = [
-                    self._evaluate_image(db, ds, gb, gc, ga, area_rng, max_det_top, iou_thrs)
-                    for db, ds, gb, gc, ga in per_img
-                ]
-                results = [r for r in results if r is not None]
-                if not results:
+        from metrics_amd.ops._coco import coco_eval_class_packed, native_matcher_available
+
+        use_native = native_matcher_available()
+        if K == 0:
+            n_imgs = 0  # no classes at all: skip packing, summaries return -1
+
+        # ---- global packing: one lexsort puts dets contiguous by (class, img, -score)
+        img_ids_d = np.concatenate([np.full(det_labels[i].shape[0], i, dtype=np.int64) for i in range(n_imgs)]) if n_imgs else np.zeros(0, dtype=np.int64)
+        all_db = np.concatenate(det_boxes).reshape(-1, 4).astype(np.float32) if n_imgs else np.zeros((0, 4), np.float32)
+        all_ds = np.concatenate(det_scores).astype(np.float32) if n_imgs else np.zeros(0, np.float32)
+        all_dl = np.concatenate(det_labels) if n_imgs else np.zeros(0, np.int64)
+        order = np.lexsort((-all_ds, img_ids_d, all_dl))
+        all_db, all_ds, all_dl, img_ids_d = all_db[order], all_ds[order], all_dl[order], img_ids_d[order]
+
+        img_ids_g = np.concatenate([np.full(gt_labels[i].shape[0], i, dtype=np.int64) for i in range(n_imgs)]) if n_imgs else np.zeros(0, dtype=np.int64)
+        all_gb = np.concatenate(gt_boxes).reshape(-1, 4).astype(np.float32) if n_imgs else np.zeros((0, 4), np.float32)
+        all_gl = np.concatenate(gt_labels) if n_imgs else np.zeros(0, np.int64)
+        all_gc = np.concatenate(gt_crowds).astype(np.uint8) if n_imgs else np.zeros(0, np.uint8)
+        all_ga = np.concatenate(gt_areas).astype(np.float32) if n_imgs else np.zeros(0, np.float32)
+        orderg = np.lexsort((img_ids_g, all_gl))
+        all_gb, all_gl, all_gc, all_ga, img_ids_g = all_gb[orderg], all_gl[orderg], all_gc[orderg], all_ga[orderg], img_ids_g[orderg]
+
+        cls_index = {c: i for i, c in enumerate(classes)}
+        dl_idx = np.array([cls_index[c] for c in all_dl.tolist()], dtype=np.int64) if all_dl.size else np.zeros(0, np.int64)
+        gl_idx = np.array([cls_index[c] for c in all_gl.tolist()], dtype=np.int64) if all_gl.size else np.zeros(0, np.int64)
+
+        # per (class, img) counts -> offsets; within-group rank for max_det caps
+        d_counts = np.bincount(dl_idx * n_imgs + img_ids_d, minlength=K * n_imgs).reshape(K, n_imgs)
+        g_counts = np.bincount(gl_idx * n_imgs + img_ids_g, minlength=K * n_imgs).reshape(K, n_imgs)
+        d_flat = d_counts.reshape(-1)
+        group_start = np.repeat(np.concatenate([[0], np.cumsum(d_flat)[:-1]]), d_flat)
+        d_rank = np.arange(all_ds.shape[0]) - group_start  # score rank within (class, img)
+
+        # cap at the TOP max_det once (native matcher sees <= max_det_top dets)
+        keep_top = d_rank < max_det_top
+        if not keep_top.all():
+            all_db, all_ds, dl_idx, img_ids_d, d_rank = (
+                all_db[keep_top], all_ds[keep_top], dl_idx[keep_top], img_ids_d[keep_top], d_rank[keep_top]
+            )
+            d_counts = np.minimum(d_counts, max_det_top)
+
+        d_cls_off = np.concatenate([[0], np.cumsum(d_counts.sum(1))])
+        g_cls_off = np.concatenate([[0], np.cumsum(g_counts.sum(1))])
+
+        area_rngs = np.array(self._AREA_RANGES, dtype=np.float32)
+
+        for ki in range(K):
+            dlo, dhi = int(d_cls_off[ki]), int(d_cls_off[ki + 1])
+            glo, ghi = int(g_cls_off[ki]), int(g_cls_off[ki + 1])
+            if dhi == dlo and ghi == glo:
+                continue
+            dt_off = np.concatenate([[0], np.cumsum(d_counts[ki])]).astype(np.int64)
+            gt_off = np.concatenate([[0], np.cumsum(g_counts[ki])]).astype(np.int64)
+            scores_k = all_ds[dlo:dhi]
+            rank_k = d_rank[dlo:dhi]
+
+            if use_native:
+                dtm_a, dti_a, npig_a = coco_eval_class_packed(
+                    all_db[dlo:dhi], all_gb[glo:ghi], all_gc[glo:ghi], all_ga[glo:ghi],
+                    dt_off, gt_off, area_rngs, iou_thrs.astype(np.float32),
+                )
+            else:
+                dtm_a, dti_a, npig_a = self._eval_class_numpy(
+                    all_db[dlo:dhi], all_gb[glo:ghi], all_gc[glo:ghi], all_ga[glo:ghi],
+                    dt_off, gt_off, area_rngs, iou_thrs,
+                )
+
+            for ai in range(A):
+                npig = int(npig_a[ai])
+                if npig == 0:
                     continue
                 for mi, max_det in enumerate(max_dets):
-                    scores = torch.cat([r[0][:max_det] for r in results])
-                    matched = torch.cat([r[1][:, :max_det] for r in results], dim=1)
-                    ignored = torch.cat([r[2][:, :max_det] for r in results], dim=1)
-                    npig = sum(r[3] for r in results)
-                    if npig == 0:
-                        continue
-                    order = torch.argsort(scores, descending=True, stable=True)
+                    keep = rank_k < max_det
+                    scores = scores_k[keep]
+                    m_k = dtm_a[ai][:, keep]
+                    i_k = dti_a[ai][:, keep]
+                    order = np.argsort(-scores, kind="stable")
                     scores_sorted = scores[order]
-                    matched = matched[:, order]
-                    ignored = ignored[:, order]
-
-                    tps = (matched & ~ignored).float().cumsum(dim=1)
-                    fps = (~matched & ~ignored).float().cumsum(dim=1)
-
+                    m_s = m_k[:, order]
+                    i_s = i_k[:, order]
+                    tps = np.cumsum(m_s & ~i_s, axis=1, dtype=np.float64)
+                    fps = np.cumsum(~m_s & ~i_s, axis=1, dtype=np.float64)
                     rc = tps / npig
-                    pr = tps / (tps + fps + torch.finfo(torch.float32).eps)
-
+                    pr = tps / (tps + fps + np.finfo(np.float64).eps)
                     if rc.shape[1]:
                         recall[:, ki, ai, mi] = rc[:, -1]
                     else:
                         recall[:, ki, ai, mi] = 0.0
-
-                    # precision envelope (monotone non-increasing from the right)
-                    pr_env = pr.flip(1).cummax(dim=1).values.flip(1)
-                    # 101-point interpolation: first index where rc >= rec_thr
-                    inds = torch.searchsorted(rc.contiguous(), rec_thrs.unsqueeze(0).expand(T, -1).contiguous())
+                    pr_env = np.maximum.accumulate(pr[:, ::-1], axis=1)[:, ::-1]
                     for ti in range(T):
-                        row = pr_env[ti]
-                        srow = scores_sorted
-                        idx = inds[ti]
-                        valid = idx < row.shape[0]
-                        q = torch.zeros(R, device=device)
-                        s = torch.zeros(R, device=device)
-                        q[valid] = row[idx[valid]]
-                        s[valid] = srow[idx[valid]]
+                        inds = np.searchsorted(rc[ti], rec_thrs, side="left")
+                        valid = inds < pr_env.shape[1]
+                        q = np.zeros(R)
+                        ss = np.zeros(R)
+                        q[valid] = pr_env[ti][inds[valid]]
+                        ss[valid] = scores_sorted[inds[valid]]
                         precision[ti, :, ki, ai, mi] = q
-                        scores_out[ti, :, ki, ai, mi] = s
+                        scores_out[ti, :, ki, ai, mi] = ss
 
         def _summarize(ap: bool, iou_thr: Optional[float] = None, area: int = 0, max_det_idx: int = -1) -> Tensor:
             if ap:
-                s = precision[:, :, :, area, max_det_idx]
+                s_ = precision[:, :, :, area, max_det_idx]
                 if iou_thr is not None:
                     ti = self.iou_thresholds.index(iou_thr)
-                    s = s[ti : ti + 1]
+                    s_ = s_[ti : ti + 1]
             else:
-                s = recall[:, :, area, max_det_idx]
+                s_ = recall[:, :, area, max_det_idx]
                 if iou_thr is not None:
                     ti = self.iou_thresholds.index(iou_thr)
-                    s = s[ti : ti + 1]
-            valid = s > -1
+                    s_ = s_[ti : ti + 1]
+            valid = s_ > -1
             if valid.sum() == 0:
-                return torch.tensor(-1.0, device=device)
-            return s[valid].mean()
+                return torch.tensor(-1.0)
+            return torch.tensor(float(s_[valid].mean()))
 
         result: Dict[str, Tensor] = {}
         result["map"] = _summarize(True)
@@ -359,24 +432,24 @@ class MeanAveragePrecision(Metric):
             map_per_class = []
             mar_per_class = []
             for ki in range(K):
-                p = precision[:, :, ki, 0, -1]
-                v = p[p > -1]
-                map_per_class.append(v.mean() if v.numel() else torch.tensor(-1.0, device=device))
-                r = recall[:, ki, 0, -1]
-                v = r[r > -1]
-                mar_per_class.append(v.mean() if v.numel() else torch.tensor(-1.0, device=device))
+                p_ = precision[:, :, ki, 0, -1]
+                v = p_[p_ > -1]
+                map_per_class.append(torch.tensor(float(v.mean()) if v.size else -1.0))
+                r_ = recall[:, ki, 0, -1]
+                v = r_[r_ > -1]
+                mar_per_class.append(torch.tensor(float(v.mean()) if v.size else -1.0))
             result["map_per_class"] = torch.stack(map_per_class)
             result[f"mar_{max_dets[-1]}_per_class"] = torch.stack(mar_per_class)
         else:
-            result["map_per_class"] = torch.tensor(-1.0, device=device)
-            result[f"mar_{max_dets[-1]}_per_class"] = torch.tensor(-1.0, device=device)
-        result["classes"] = torch.tensor(classes, dtype=torch.int, device=device)
+            result["map_per_class"] = torch.tensor(-1.0)
+            result[f"mar_{max_dets[-1]}_per_class"] = torch.tensor(-1.0)
+        result["classes"] = torch.tensor(classes, dtype=torch.int)
 
         if self.extended_summary:
-            result["precision"] = precision
-            result["recall"] = recall
-            result["scores"] = scores_out
-            result["ious"] = torch.tensor([], device=device)  # per-pair ious not retained
+            result["precision"] = torch.from_numpy(precision)
+            result["recall"] = torch.from_numpy(recall)
+            result["scores"] = torch.from_numpy(scores_out)
+        result = {k: (v.to(device) if isinstance(v, Tensor) else v) for k, v in result.items()}
         return result
 
     def plot(self, val=None, ax=None):
