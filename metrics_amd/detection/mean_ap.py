@@ -330,8 +330,11 @@ class MeanAveragePrecision(Metric):
         d_counts = np.bincount(dl_idx * n_imgs + img_ids_d, minlength=K * n_imgs).reshape(K, n_imgs)
         g_counts = np.bincount(gl_idx * n_imgs + img_ids_g, minlength=K * n_imgs).reshape(K, n_imgs)
         d_flat = d_counts.reshape(-1)
-        group_start = np.repeat(np.concatenate([[0], np.cumsum(d_flat)[:-1]]), d_flat)
-        d_rank = np.arange(all_ds.shape[0]) - group_start  # score rank within (class, img)
+        if d_flat.size:
+            group_start = np.repeat(np.concatenate([[0], np.cumsum(d_flat)[:-1]]), d_flat)
+            d_rank = np.arange(all_ds.shape[0]) - group_start  # score rank within (class, img)
+        else:
+            d_rank = np.zeros(0, dtype=np.int64)
 
         # cap at the TOP max_det once (native matcher sees <= max_det_top dets)
         keep_top = d_rank < max_det_top
