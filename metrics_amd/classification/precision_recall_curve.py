@@ -77,6 +77,11 @@ class BinaryPrecisionRecallCurve(Metric):
         preds_f, target_f, _ = _binary_precision_recall_curve_format(
             preds, target, self.thresholds, self.ignore_index, remove_ignored=remove_ignored
         )
+        if preds_f.is_cuda and self.thresholds is not None and preds_f.dtype in (torch.float32, torch.bfloat16):
+            from metrics_amd.ops import _hip
+
+            _hip.curve_hist_into_confmat(preds_f, target_f, self.thresholds, self.ignore_index, self.confmat, mode=0)
+            return
         state = _binary_precision_recall_curve_update(
             preds_f, target_f, self.thresholds, self.ignore_index if not remove_ignored else None
         )
@@ -157,6 +162,19 @@ class MulticlassPrecisionRecallCurve(Metric):
             preds, target, self.num_classes, self.thresholds, self.ignore_index, self.average,
             remove_ignored=remove_ignored,
         )
+        if (
+            preds_f.is_cuda
+            and self.thresholds is not None
+            and self.average != "micro"
+            and preds_f.dtype in (torch.float32, torch.bfloat16)
+        ):
+            from metrics_amd.ops import _hip
+
+            _hip.curve_hist_into_confmat(
+                preds_f, target_f, self.thresholds, self.ignore_index if not remove_ignored else None,
+                self.confmat, mode=0,
+            )
+            return
         state = _multiclass_precision_recall_curve_update(
             preds_f, target_f, self.num_classes, self.thresholds, self.average,
             self.ignore_index if not remove_ignored else None,
@@ -228,6 +246,14 @@ class MultilabelPrecisionRecallCurve(Metric):
         preds_f, target_f, _ = _multilabel_precision_recall_curve_format(
             preds, target, self.num_labels, self.thresholds, self.ignore_index, remove_ignored=remove_ignored
         )
+        if preds_f.is_cuda and self.thresholds is not None and preds_f.dtype in (torch.float32, torch.bfloat16):
+            from metrics_amd.ops import _hip
+
+            _hip.curve_hist_into_confmat(
+                preds_f, target_f, self.thresholds, self.ignore_index if not remove_ignored else None,
+                self.confmat, mode=1,
+            )
+            return
         state = _multilabel_precision_recall_curve_update(
             preds_f, target_f, self.num_labels, self.thresholds, self.ignore_index if not remove_ignored else None
         )

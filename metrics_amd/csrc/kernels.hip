@@ -65,8 +65,22 @@ __global__ void __launch_bounds__(256) k_mc_stat_logits(
         const T* prow = preds + row * C;
         float best = -INFINITY;
         ll best_idx = 0x7fffffffffffffffLL;
-        // vectorized 4-wide when alignment allows (row start 16B/8B aligned iff C%4==0)
-        if ((C & 3) == 0) {
+        // vectorized loads: 16B/lane for bf16 when C%8==0, else 8B/4B paths
+        if (IS_BF16 && (C & 7) == 0) {
+            struct U8 { ushort4 a; ushort4 b; };
+            const U8* pv = reinterpret_cast<const U8*>(prow);
+            const ll nvec = C / 8;
+            for (ll v = lane; v < nvec; v += WAVE) {
+                U8 u = pv[v];
+                float f[8] = {bf16_to_f32(u.a.x), bf16_to_f32(u.a.y), bf16_to_f32(u.a.z), bf16_to_f32(u.a.w),
+                              bf16_to_f32(u.b.x), bf16_to_f32(u.b.y), bf16_to_f32(u.b.z), bf16_to_f32(u.b.w)};
+                ll c = v * 8;
+#pragma unroll
+                for (int k = 0; k < 8; k++) {
+                    if (f[k] > best || (f[k] == best && c + k < best_idx)) { best = f[k]; best_idx = c + k; }
+                }
+            }
+        } else if ((C & 3) == 0) {
             const ll nvec = C / 4;
             if (IS_BF16) {
                 const ushort4* pv = reinterpret_cast<const ushort4*>(prow);
@@ -392,33 +406,62 @@ __global__ void __launch_bounds__(256) k_multiclass_curve_hist(
 }
 
 // suffix-sum the histogram into (/onto) the running confmat state:
-// state (..., T, 2, 2) += suffix(hist)(..., T, 2) expanded:
 //   confmat[t][1][1] += sum_{j>t} hist[j][1]   (tp)
-//   confmat[t][1][0] += sum_{j>t} hist[j][0]   (fp)
-//   confmat[t][0][1] += pos_total - tp         (fn)
+//   confmat[t][0][1] += sum_{j>t} hist[j][0]   (fp)
+//   confmat[t][1][0] += pos_total - tp         (fn)
 //   confmat[t][0][0] += neg_total - fp         (tn)
-// one block per outer index (class or 1), thread t in [0,T)
+// one block per outer index (class or 1). The histogram is staged in LDS and
+// suffix-summed there (global O(T^2) loads -> LDS): T <= 4096.
+// transposed=1 writes the (T, O, 2, 2) layout (the metric state layout for
+// multiclass/multilabel curves) so the accumulation happens in-place with no
+// permute+copy of the 6.4MB state.
 __global__ void k_curve_suffix(
-    const unsigned long long* __restrict__ hist /* (O, T+1, 2) */, int T,
-    ll* __restrict__ confmat /* (O, T, 2, 2) */) {
+    const unsigned long long* __restrict__ hist /* (O, T+1, 2) */, int T, ll outer, int transposed,
+    ll* __restrict__ confmat /* (O, T, 2, 2) or (T, O, 2, 2) */) {
+    extern __shared__ unsigned long long sh[];  // (T+1) * 2
     const ll o = blockIdx.x;
     const unsigned long long* h = hist + o * (ll)(T + 1) * 2;
-    ll* cm = confmat + o * (ll)T * 4;
+    ll* cm = confmat + (transposed ? o * 4 : o * (ll)T * 4);
+    const ll tstride = transposed ? outer * 4 : 4;
+    for (int j = threadIdx.x; j < (T + 1) * 2; j += blockDim.x) sh[j] = h[j];
+    __syncthreads();
     __shared__ unsigned long long pos_total, neg_total;
     if (threadIdx.x == 0) {
         unsigned long long pt = 0, nt = 0;
-        for (int j = 0; j <= T; j++) { nt += h[j * 2 + 0]; pt += h[j * 2 + 1]; }
+        for (int j = 0; j <= T; j++) { nt += sh[j * 2 + 0]; pt += sh[j * 2 + 1]; }
         pos_total = pt; neg_total = nt;
     }
     __syncthreads();
     for (int t = threadIdx.x; t < T; t += blockDim.x) {
         unsigned long long tp = 0, fp = 0;
-        for (int j = t + 1; j <= T; j++) { fp += h[j * 2 + 0]; tp += h[j * 2 + 1]; }
+        for (int j = t + 1; j <= T; j++) { fp += sh[j * 2 + 0]; tp += sh[j * 2 + 1]; }
         // layout [t][target][pred] (reference: bins = 2*target + pred)
-        cm[t * 4 + 3] += (ll)tp;                    // [1][1] tp
-        cm[t * 4 + 1] += (ll)fp;                    // [0][1] fp
-        cm[t * 4 + 2] += (ll)(pos_total - tp);      // [1][0] fn
-        cm[t * 4 + 0] += (ll)(neg_total - fp);      // [0][0] tn
+        ll* c = cm + (ll)t * tstride;
+        c[3] += (ll)tp;                    // [1][1] tp
+        c[1] += (ll)fp;                    // [0][1] fp
+        c[2] += (ll)(pos_total - tp);      // [1][0] fn
+        c[0] += (ll)(neg_total - fp);      // [0][0] tn
+    }
+}
+
+// ---------------------------------------------------------------------------
+// fused stat-delta apply: given the kernel scratch [tp|fp|fn|valid] produced
+// by k_mc_stat_*, add the deltas into the four metric state tensors
+// (tn += valid - tp - fp - fn). ONE launch replaces ~8 small torch kernels.
+__global__ void k_apply_stat_deltas(
+    const unsigned long long* __restrict__ scratch /* 3*C + 1 */, ll C,
+    ll* __restrict__ tp, ll* __restrict__ fp, ll* __restrict__ tn, ll* __restrict__ fn) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    ll stride = (ll)gridDim.x * blockDim.x;
+    const unsigned long long valid = scratch[3 * C];
+    for (; i < C; i += stride) {
+        const ll dtp = (ll)scratch[i];
+        const ll dfp = (ll)scratch[C + i];
+        const ll dfn = (ll)scratch[2 * C + i];
+        tp[i] += dtp;
+        fp[i] += dfp;
+        fn[i] += dfn;
+        tn[i] += (ll)valid - dtp - dfp - dfn;
     }
 }
 
@@ -659,9 +702,21 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
     return (int)hipGetLastError();
 }
 
-int ma_curve_suffix(uintptr_t stream, uintptr_t hist, ll outer, int T, uintptr_t confmat) {
+int ma_curve_suffix(uintptr_t stream, uintptr_t hist, ll outer, int T, int transposed,
+                    uintptr_t confmat) {
     hipStream_t s = (hipStream_t)stream;
-    k_curve_suffix<<<(int)outer, 256, 0, s>>>((const unsigned long long*)hist, T, (ll*)confmat);
+    size_t shmem = (size_t)(T + 1) * 2 * sizeof(unsigned long long);
+    if (shmem > 160 * 1024) return -100;
+    k_curve_suffix<<<(int)outer, 256, shmem, s>>>((const unsigned long long*)hist, T, outer,
+                                                  transposed, (ll*)confmat);
+    return (int)hipGetLastError();
+}
+
+int ma_apply_stat_deltas(uintptr_t stream, uintptr_t scratch, ll C, uintptr_t tp, uintptr_t fp,
+                         uintptr_t tn, uintptr_t fn) {
+    hipStream_t s = (hipStream_t)stream;
+    k_apply_stat_deltas<<<grid_for(C, 256), 256, 0, s>>>(
+        (const unsigned long long*)scratch, C, (ll*)tp, (ll*)fp, (ll*)tn, (ll*)fn);
     return (int)hipGetLastError();
 }
 

@@ -338,8 +338,6 @@ def _multiclass_stat_scores_update(
             p2 = preds.reshape(-1)
             t2 = target.reshape(-1)
         tp, fp, tn, fn, _ = ops.multiclass_stat_scores_fused(p2, t2, num_classes, ignore_index, want_confmat=False)
-        if average == "micro":
-            return tp.sum(), fp.sum(), tn.sum(), fn.sum()
         return tp, fp, tn, fn
 
     preds = preds.flatten()
@@ -348,12 +346,6 @@ def _multiclass_stat_scores_update(
         idx = target != ignore_index
         preds = preds[idx]
         target = target[idx]
-    if average == "micro":
-        tp = (preds == target).sum()
-        fp = (preds != target).sum()
-        fn = (preds != target).sum()
-        tn = num_classes * preds.numel() - (fp + fn + tp)
-        return tp, fp, tn, fn
     unique_mapping = target.to(torch.long) * num_classes + preds.to(torch.long)
     bins = _bincount(unique_mapping, minlength=num_classes**2)
     confmat = bins.reshape(num_classes, num_classes)

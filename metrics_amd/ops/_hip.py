@@ -289,6 +289,7 @@ def binary_curve_confmat(
         ctypes.c_uint64(hist.data_ptr()),
         ctypes.c_longlong(1),
         ctypes.c_int(T),
+        ctypes.c_int(0),
         ctypes.c_uint64(confmat.data_ptr()),
     )
     _check(rc, "ma_curve_suffix")
@@ -327,16 +328,18 @@ def multiclass_curve_confmat(
         ctypes.c_uint64(hist.data_ptr()),
     )
     _check(rc, "ma_multiclass_curve_hist")
-    confmat_cls = torch.zeros(C, T, 2, 2, dtype=torch.long, device=dev)
+    # write the (T, C, 2, 2) state layout directly (transposed suffix kernel)
+    confmat = torch.zeros(T, C, 2, 2, dtype=torch.long, device=dev)
     rc = lib.ma_curve_suffix(
         ctypes.c_uint64(_stream()),
         ctypes.c_uint64(hist.data_ptr()),
         ctypes.c_longlong(C),
         ctypes.c_int(T),
-        ctypes.c_uint64(confmat_cls.data_ptr()),
+        ctypes.c_int(1),
+        ctypes.c_uint64(confmat.data_ptr()),
     )
     _check(rc, "ma_curve_suffix")
-    return confmat_cls.permute(1, 0, 2, 3).contiguous()  # (T, C, 2, 2)
+    return confmat
 
 
 _ERR_OPS = {"sq_err": (0, 1), "abs_err": (1, 1), "ape": (2, 1), "sq_log_err": (3, 1), "moments": (4, 6), "logcosh": (5, 1)}
@@ -391,3 +394,139 @@ def box_iou(boxes1: Tensor, boxes2: Tensor, variant: str = "iou") -> Tensor:
     )
     _check(rc, "ma_box_iou")
     return out
+
+
+def curve_hist_into_confmat(
+    probs: Tensor, target: Tensor, thresholds: Tensor, ignore_index: Optional[int],
+    confmat_state: Tensor, mode: int,
+) -> None:
+    """Bucketized histogram + transposed suffix-sum accumulated DIRECTLY into the
+    metric's confmat state ((T,2,2) binary / (T,C,2,2) multiclass|multilabel) —
+    no intermediate confmat materialization or host-side add."""
+    lib = _lib()
+    thr = thresholds.contiguous().float()
+    T = thr.numel()
+    uni, t0, inv_step = _uniform_params(thr)
+    dev = probs.device
+    if confmat_state.ndim == 3:  # binary (T,2,2)
+        preds = probs.contiguous().flatten()
+        tgt = target.contiguous().long().flatten()
+        hist = torch.zeros(T + 1, 2, dtype=torch.long, device=dev)
+        rc = lib.ma_binary_curve_hist(
+            ctypes.c_uint64(_stream()),
+            ctypes.c_uint64(preds.data_ptr()),
+            ctypes.c_int(_dtype_code(preds)),
+            ctypes.c_uint64(tgt.data_ptr()),
+            ctypes.c_longlong(preds.numel()),
+            ctypes.c_uint64(thr.data_ptr()),
+            ctypes.c_int(T),
+            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+            ctypes.c_int(1 if ignore_index is not None else 0),
+            ctypes.c_int(uni),
+            ctypes.c_float(t0),
+            ctypes.c_float(inv_step),
+            ctypes.c_uint64(hist.data_ptr()),
+        )
+        _check(rc, "ma_binary_curve_hist")
+        outer, transposed = 1, 0
+    else:
+        probs = probs.contiguous()
+        tgt = target.contiguous().long()
+        B, C = probs.shape
+        hist = torch.zeros(C, T + 1, 2, dtype=torch.long, device=dev)
+        rc = lib.ma_multiclass_curve_hist(
+            ctypes.c_uint64(_stream()),
+            ctypes.c_uint64(probs.data_ptr()),
+            ctypes.c_int(_dtype_code(probs)),
+            ctypes.c_uint64(tgt.data_ptr()),
+            ctypes.c_longlong(B),
+            ctypes.c_longlong(C),
+            ctypes.c_uint64(thr.data_ptr()),
+            ctypes.c_int(T),
+            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+            ctypes.c_int(1 if ignore_index is not None else 0),
+            ctypes.c_int(mode),
+            ctypes.c_int(uni),
+            ctypes.c_float(t0),
+            ctypes.c_float(inv_step),
+            ctypes.c_uint64(hist.data_ptr()),
+        )
+        _check(rc, "ma_multiclass_curve_hist")
+        outer, transposed = C, 1
+    assert confmat_state.is_contiguous()
+    rc = lib.ma_curve_suffix(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(hist.data_ptr()),
+        ctypes.c_longlong(outer),
+        ctypes.c_int(T),
+        ctypes.c_int(transposed),
+        ctypes.c_uint64(confmat_state.data_ptr()),
+    )
+    _check(rc, "ma_curve_suffix")
+
+
+def mc_stat_into(
+    preds: Tensor, target: Tensor, num_classes: int, ignore_index: Optional[int],
+    scratch: Tensor, tp: Tensor, fp: Tensor, tn: Tensor, fn: Tensor,
+) -> None:
+    """Fused stat-scores update accumulated DIRECTLY into the metric states.
+
+    ``scratch`` is a per-metric reusable (3*C+1,) int64 buffer; 3 launches
+    total (zero + count + apply) instead of ~12 small torch kernels.
+    """
+    lib = _lib()
+    C = num_classes
+    scratch.zero_()
+    s_tp = scratch[:C]
+    s_fp = scratch[C : 2 * C]
+    s_fn = scratch[2 * C : 3 * C]
+    s_valid = scratch[3 * C :]
+    if preds.ndim == 2 and preds.is_floating_point():
+        preds = preds.contiguous()
+        target = target.contiguous().long()
+        B, C2 = preds.shape
+        rc = lib.ma_mc_stat_logits(
+            ctypes.c_uint64(_stream()),
+            ctypes.c_uint64(preds.data_ptr()),
+            ctypes.c_int(_dtype_code(preds)),
+            ctypes.c_uint64(target.data_ptr()),
+            ctypes.c_longlong(B),
+            ctypes.c_longlong(C2),
+            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+            ctypes.c_int(1 if ignore_index is not None else 0),
+            ctypes.c_uint64(s_tp.data_ptr()),
+            ctypes.c_uint64(s_fp.data_ptr()),
+            ctypes.c_uint64(s_fn.data_ptr()),
+            ctypes.c_uint64(0),
+            ctypes.c_uint64(s_valid.data_ptr()),
+            ctypes.c_uint64(0),
+        )
+        _check(rc, "ma_mc_stat_logits")
+    else:
+        p2 = preds.contiguous().long().flatten()
+        t2 = target.contiguous().long().flatten()
+        rc = lib.ma_mc_stat_labels(
+            ctypes.c_uint64(_stream()),
+            ctypes.c_uint64(p2.data_ptr()),
+            ctypes.c_uint64(t2.data_ptr()),
+            ctypes.c_longlong(p2.numel()),
+            ctypes.c_longlong(C),
+            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+            ctypes.c_int(1 if ignore_index is not None else 0),
+            ctypes.c_uint64(s_tp.data_ptr()),
+            ctypes.c_uint64(s_fp.data_ptr()),
+            ctypes.c_uint64(s_fn.data_ptr()),
+            ctypes.c_uint64(0),
+            ctypes.c_uint64(s_valid.data_ptr()),
+        )
+        _check(rc, "ma_mc_stat_labels")
+    rc = lib.ma_apply_stat_deltas(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(scratch.data_ptr()),
+        ctypes.c_longlong(C),
+        ctypes.c_uint64(tp.data_ptr()),
+        ctypes.c_uint64(fp.data_ptr()),
+        ctypes.c_uint64(tn.data_ptr()),
+        ctypes.c_uint64(fn.data_ptr()),
+    )
+    _check(rc, "ma_apply_stat_deltas")
