@@ -37,6 +37,7 @@ from metrics_amd.utilities.data import (
 from metrics_amd.utilities.distributed import gather_all_tensors, sync_states_fast
 from metrics_amd.utilities.exceptions import MetricsUserError
 from metrics_amd.utilities.prints import rank_zero_warn
+from metrics_amd.utilities import tracing
 
 
 def jit_distributed_available() -> bool:
@@ -362,7 +363,8 @@ class Metric(Module, ABC):
         self._cache = self._copy_state_dict()
 
         # sync
-        self._sync_dist(dist_sync_fn, process_group=process_group)
+        with tracing.range(f"{self.__class__.__name__}.sync"):
+            self._sync_dist(dist_sync_fn, process_group=process_group)
         self._is_synced = True
 
     def unsync(self, should_unsync: bool = True) -> None:
@@ -406,7 +408,7 @@ class Metric(Module, ABC):
         def wrapped_func(*args: Any, **kwargs: Any) -> None:
             self._computed = None
             self._update_count += 1
-            with torch.set_grad_enabled(self._enable_grad):
+            with torch.set_grad_enabled(self._enable_grad), tracing.range(f"{self.__class__.__name__}.update"):
                 try:
                     update(*args, **kwargs)
                 except RuntimeError as err:
@@ -450,7 +452,7 @@ class Metric(Module, ABC):
                 dist_sync_fn=self.dist_sync_fn,
                 should_sync=self._to_sync,
                 should_unsync=self._should_unsync,
-            ):
+            ), tracing.range(f"{self.__class__.__name__}.compute"):
                 value = _squeeze_if_scalar(compute(*args, **kwargs))
                 # clone so later in-place ops cannot alter the returned result
                 value = apply_to_collection(value, Tensor, lambda x: x.clone())
