@@ -213,3 +213,71 @@ def test_bf16_inputs_gpu():
     v = m(preds, target)
     ref = (preds.float().argmax(-1) == target).float().mean()
     assert torch.allclose(v, ref, atol=1e-6)
+
+
+def test_into_state_stat_paths_match_cpu():
+    """The accumulate-into-state GPU update must equal the CPU torch path."""
+    torch.manual_seed(5)
+    preds = torch.randn(2048, 64)
+    target = torch.randint(0, 64, (2048,))
+    target[::9] = -1
+    for avg in ("micro", "macro", None):
+        m_cpu = ma.MulticlassStatScores(num_classes=64, average=avg, ignore_index=-1)
+        m_gpu = ma.MulticlassStatScores(num_classes=64, average=avg, ignore_index=-1).to("cuda")
+        for chunk in range(4):
+            sl = slice(chunk * 512, (chunk + 1) * 512)
+            m_cpu.update(preds[sl], target[sl])
+            m_gpu.update(preds[sl].cuda(), target[sl].cuda())
+        assert torch.equal(m_cpu.tp, m_gpu.tp.cpu())
+        assert torch.equal(m_cpu.tn, m_gpu.tn.cpu())
+        assert torch.allclose(m_cpu.compute().float(), m_gpu.compute().float().cpu(), atol=1e-6)
+
+
+def test_into_state_curve_paths_match_cpu():
+    torch.manual_seed(6)
+    # binary with ignore index
+    bp = torch.rand(50_000)
+    bt = torch.randint(0, 2, (50_000,))
+    bt[::11] = -1
+    m_cpu = ma.BinaryPrecisionRecallCurve(thresholds=100, ignore_index=-1)
+    m_gpu = ma.BinaryPrecisionRecallCurve(thresholds=100, ignore_index=-1).to("cuda")
+    m_cpu.update(bp, bt)
+    m_gpu.update(bp.cuda(), bt.cuda())
+    assert torch.equal(m_cpu.confmat, m_gpu.confmat.cpu())
+
+    # multiclass AUROC thresholded
+    probs = torch.randn(4096, 37).softmax(-1)
+    tgt = torch.randint(0, 37, (4096,))
+    a_cpu = ma.MulticlassAUROC(num_classes=37, thresholds=200)
+    a_gpu = ma.MulticlassAUROC(num_classes=37, thresholds=200).to("cuda")
+    a_cpu.update(probs, tgt)
+    a_gpu.update(probs.cuda(), tgt.cuda())
+    assert torch.equal(a_cpu.confmat, a_gpu.confmat.cpu())
+    assert torch.allclose(a_cpu.compute(), a_gpu.compute().cpu(), atol=1e-6)
+
+    # multilabel AP thresholded
+    mlp = torch.rand(4096, 12)
+    mlt = torch.randint(0, 2, (4096, 12))
+    p_cpu = ma.MultilabelAveragePrecision(num_labels=12, thresholds=50)
+    p_gpu = ma.MultilabelAveragePrecision(num_labels=12, thresholds=50).to("cuda")
+    p_cpu.update(mlp, mlt)
+    p_gpu.update(mlp.cuda(), mlt.cuda())
+    assert torch.equal(p_cpu.confmat, p_gpu.confmat.cpu())
+
+
+def test_multilabel_fused_with_logits():
+    torch.manual_seed(7)
+    logits = torch.randn(8192, 30) * 2
+    tgt = torch.randint(0, 2, (8192, 30))
+    m_cpu = ma.MultilabelF1Score(num_labels=30, average="macro")
+    m_gpu = ma.MultilabelF1Score(num_labels=30, average="macro").to("cuda")
+    m_cpu.update(logits, tgt)
+    m_gpu.update(logits.cuda(), tgt.cuda())
+    assert torch.allclose(m_cpu.compute(), m_gpu.compute().cpu(), atol=1e-6)
+
+
+def test_bincount_deterministic_repeat():
+    x = torch.randint(0, 777, (4_000_000,), device="cuda")
+    a = ops.hip_bincount(x, 777)
+    b = ops.hip_bincount(x, 777)
+    assert torch.equal(a, b)

@@ -1,0 +1,190 @@
+"""Image metric tests (known values + invariants)."""
+import math
+
+import pytest
+import torch
+
+import metrics_amd as ma
+from tests.unittests._helpers import seed_all
+
+
+@pytest.fixture()
+def imgs():
+    seed_all(31)
+    p = torch.rand(2, 3, 64, 64)
+    t = torch.rand(2, 3, 64, 64)
+    return p, t
+
+
+def test_psnr_known(imgs):
+    p, t = imgs
+    v = ma.PeakSignalNoiseRatio(data_range=1.0)(p, t).item()
+    mse = ((p - t) ** 2).mean().item()
+    assert abs(v - 10 * math.log10(1.0 / mse)) < 1e-4
+
+
+def test_psnr_accumulation(imgs):
+    p, t = imgs
+    m = ma.PeakSignalNoiseRatio(data_range=1.0)
+    m.update(p[:1], t[:1])
+    m.update(p[1:], t[1:])
+    mse = ((p - t) ** 2).mean().item()
+    assert abs(m.compute().item() - 10 * math.log10(1.0 / mse)) < 1e-4
+
+
+def test_psnr_auto_range(imgs):
+    p, t = imgs
+    m = ma.PeakSignalNoiseRatio()
+    m.update(p, t)
+    rng = (t.max() - t.min()).item()
+    mse = ((p - t) ** 2).mean().item()
+    assert abs(m.compute().item() - 10 * math.log10(rng**2 / mse)) < 1e-3
+
+
+def test_ssim_bounds(imgs):
+    p, t = imgs
+    assert abs(ma.StructuralSimilarityIndexMeasure(data_range=1.0)(p, p).item() - 1.0) < 1e-6
+    v = ma.StructuralSimilarityIndexMeasure(data_range=1.0)(p, t).item()
+    assert -1.0 <= v < 0.9
+
+
+def test_ssim_shift_sensitivity():
+    seed_all(32)
+    base = torch.rand(1, 1, 64, 64)
+    near = (base + 0.01 * torch.randn_like(base)).clamp(0, 1)
+    far = (base + 0.3 * torch.randn_like(base)).clamp(0, 1)
+    s_near = ma.StructuralSimilarityIndexMeasure(data_range=1.0)(near, base).item()
+    s_far = ma.StructuralSimilarityIndexMeasure(data_range=1.0)(far, base).item()
+    assert s_near > s_far
+
+
+def test_ms_ssim(imgs):
+    seed_all(33)
+    p = torch.rand(1, 1, 256, 256)
+    t = (p + 0.05 * torch.randn_like(p)).clamp(0, 1)
+    v = ma.MultiScaleStructuralSimilarityIndexMeasure(data_range=1.0)(p, t).item()
+    assert 0.5 < v <= 1.0
+    assert abs(ma.MultiScaleStructuralSimilarityIndexMeasure(data_range=1.0)(p, p).item() - 1.0) < 1e-5
+
+
+def test_uqi_sam_ergas_identity(imgs):
+    p, _ = imgs
+    assert abs(ma.UniversalImageQualityIndex()(p, p).item() - 1.0) < 1e-4
+    assert ma.SpectralAngleMapper()(p, p).item() < 1e-3
+    assert ma.ErrorRelativeGlobalDimensionlessSynthesis()(p, p).item() < 1e-3
+    assert ma.RelativeAverageSpectralError()(p, p).item() < 1e-3
+
+
+def test_total_variation_known():
+    img = torch.zeros(1, 1, 3, 3)
+    img[0, 0, 1, 1] = 1.0
+    assert abs(ma.TotalVariation()(img).item() - 4.0) < 1e-6
+    assert abs(ma.TotalVariation(reduction="mean")(img).item() - 4.0) < 1e-6
+
+
+def test_rmse_sw_zero_and_positive(imgs):
+    p, t = imgs
+    assert ma.RootMeanSquaredErrorUsingSlidingWindow()(p, p).item() < 1e-6
+    assert ma.RootMeanSquaredErrorUsingSlidingWindow()(p, t).item() > 0.1
+
+
+def test_scc_identity(imgs):
+    p, _ = imgs
+    assert abs(ma.SpatialCorrelationCoefficient()(p, p).item() - 1.0) < 1e-4
+
+
+def test_vif_identity():
+    seed_all(34)
+    x = torch.rand(1, 1, 64, 64)
+    assert abs(ma.VisualInformationFidelity()(x, x).item() - 1.0) < 1e-4
+
+
+def test_psnrb(imgs):
+    p, t = imgs
+    v = ma.PeakSignalNoiseRatioWithBlockedEffect()(p[:, :1], t[:, :1]).item()
+    assert 0 < v < 20
+
+
+def test_pansharpening_metrics():
+    seed_all(35)
+    fused = torch.rand(1, 4, 32, 32)
+    ms = torch.rand(1, 4, 32, 32)
+    pan = torch.rand(1, 1, 64, 64)
+    d_lambda = ma.SpectralDistortionIndex()(fused, ms).item()
+    assert 0 <= d_lambda <= 1
+    # identical fused == ms -> D_lambda == 0
+    assert ma.SpectralDistortionIndex()(ms, ms).item() < 1e-6
+    d_s = ma.SpatialDistortionIndex()(fused, {"ms": ms, "pan": pan}).item()
+    assert 0 <= d_s <= 1
+    qnr = ma.QualityWithNoReference()(fused, {"ms": ms, "pan": pan}).item()
+    assert 0 <= qnr <= 1
+
+
+class _ToyFeat(torch.nn.Module):
+    def __init__(self, d=24):
+        super().__init__()
+        torch.manual_seed(0)
+        self.lin = torch.nn.Linear(3 * 16 * 16, d)
+
+    def forward(self, x):
+        return self.lin(x.float().flatten(1))
+
+
+def test_fid_behaviour():
+    seed_all(36)
+    real = torch.rand(96, 3, 16, 16)
+    close = (real + 0.01 * torch.randn_like(real)).clamp(0, 1)
+    far = torch.rand(96, 3, 16, 16) * 0.3
+
+    def fid_of(fake):
+        m = ma.FrechetInceptionDistance(feature_extractor=_ToyFeat(), input_img_size=(3, 16, 16), normalize=True)
+        m.update(real, real=True)
+        m.update(fake, real=False)
+        return m.compute().item()
+
+    assert fid_of(close) < fid_of(far)
+
+
+def test_fid_reset_real_features():
+    real = torch.rand(8, 3, 16, 16)
+    m = ma.FrechetInceptionDistance(
+        feature_extractor=_ToyFeat(), input_img_size=(3, 16, 16), normalize=True, reset_real_features=False
+    )
+    m.update(real, real=True)
+    n = m.real_features_num_samples.clone()
+    m.reset()
+    assert m.real_features_num_samples == n
+
+
+def test_kid_and_is():
+    seed_all(37)
+    real = torch.rand(64, 3, 16, 16)
+    fake = torch.rand(64, 3, 16, 16) * 0.5
+    kid = ma.KernelInceptionDistance(feature_extractor=_ToyFeat(), subsets=3, subset_size=32)
+    kid.update(real, real=True)
+    kid.update(fake, real=False)
+    km, ks = kid.compute()
+    assert km.item() > 0
+    isc = ma.InceptionScore(feature_extractor=_ToyFeat(), splits=2)
+    isc.update(real)
+    m, s = isc.compute()
+    assert m.item() >= 1.0
+
+
+def test_mifid():
+    seed_all(38)
+    real = torch.rand(64, 3, 16, 16)
+    fake = torch.rand(64, 3, 16, 16)
+    m = ma.MemorizationInformedFrechetInceptionDistance(feature_extractor=_ToyFeat())
+    m.update(real, real=True)
+    m.update(fake, real=False)
+    assert m.compute().item() >= 0
+
+
+def test_model_backed_raise_without_net():
+    with pytest.raises(ModuleNotFoundError):
+        ma.FrechetInceptionDistance()
+    with pytest.raises(ModuleNotFoundError):
+        ma.LearnedPerceptualImagePatchSimilarity()
+    with pytest.raises(ModuleNotFoundError):
+        ma.multimodal.CLIPScore()
