@@ -97,30 +97,54 @@ def error_relative_global_dimensionless_synthesis(
     return reduce(ergas_score, reduction or "none")
 
 
+def _pad1(t: Tensor, dim: int, pad: int, outer_pad: int) -> Tensor:
+    # scipy-style symmetric padding (edge pixel repeated), size-preserving for a
+    # valid w-wide box filter: left pad `pad`, right pad `pad + outer_pad - 1`.
+    n = t.shape[dim]
+    left = t.index_select(dim, torch.arange(pad - 1, -1, -1, device=t.device))
+    right = t.index_select(dim, torch.arange(n - 1, n - pad - outer_pad, -1, device=t.device))
+    return torch.cat((left, t, right), dim)
+
+
+def _scipy_uniform_filter(x: Tensor, window_size: int) -> Tensor:
+    """Size-preserving box filter matching scipy.ndimage.uniform_filter(mode='reflect').
+
+    Parity: reference functional/image/utils.py:113 (_uniform_filter).
+    """
+    pad, outer = window_size // 2, window_size % 2
+    x = _pad1(_pad1(x, 2, pad, outer), 3, pad, outer)
+    c = x.shape[1]
+    kernel = torch.ones(c, 1, window_size, window_size, dtype=x.dtype, device=x.device) / (window_size**2)
+    return F.conv2d(x, kernel, groups=c)
+
+
 def relative_average_spectral_error(preds: Tensor, target: Tensor, window_size: int = 8) -> Tensor:
-    """RASE: relative average spectral error using sliding-window RMSE."""
+    """RASE: relative average spectral error using sliding-window RMSE.
+
+    Parity: reference functional/image/rase.py (_rase_update/_rase_compute).
+    """
     preds, target = _image_check(preds, target)
     _, rmse_map = _rmse_sw_maps(preds, target, window_size)
-    target_mean = target.mean(dim=(0, 2, 3), keepdim=True)  # mean over batch and spatial
-    rase_map = 100.0 / target_mean.squeeze() * torch.sqrt((rmse_map**2).mean(dim=1))
-    return rase_map.mean() if rase_map.numel() else rase_map
+    rmse_map = rmse_map.sum(0) / preds.shape[0]  # mean over images -> (C,H,W)
+    target_mean = (_scipy_uniform_filter(target, window_size) / (window_size**2)).sum(0) / preds.shape[0]
+    target_mean = target_mean.mean(0)  # mean over channels -> (H,W)
+    rase_map = 100 / target_mean * torch.sqrt(torch.mean(rmse_map**2, 0))
+    crop = round(window_size / 2)
+    return torch.mean(rase_map[crop:-crop, crop:-crop])
 
 
 def _rmse_sw_maps(preds: Tensor, target: Tensor, window_size: int) -> Tuple[Tensor, Tensor]:
-    kernel = torch.ones(preds.shape[1], 1, window_size, window_size, dtype=preds.dtype, device=preds.device) / (
-        window_size**2
-    )
-    diff_sq = (preds - target) ** 2
-    mse_map = F.conv2d(diff_sq, kernel, groups=preds.shape[1])
-    rmse_map = torch.sqrt(mse_map)
-    rmse_mean = rmse_map.mean()
+    """Per-pixel sliding-window RMSE map (size-preserving) + edge-cropped mean."""
+    rmse_map = torch.sqrt(_scipy_uniform_filter((preds - target) ** 2, window_size))
+    crop = round(window_size / 2)
+    rmse_mean = rmse_map[..., crop:-crop, crop:-crop].mean()
     return rmse_mean, rmse_map
 
 
 def root_mean_squared_error_using_sliding_window(
     preds: Tensor, target: Tensor, window_size: int = 8, return_rmse_map: bool = False
 ):
-    """RMSE averaged over sliding windows."""
+    """RMSE averaged over sliding windows (edges cropped by round(w/2) like the reference)."""
     preds, target = _image_check(preds, target)
     if not isinstance(window_size, int) or window_size < 1:
         raise ValueError(f"Argument `window_size` is expected to be a positive integer, but got {window_size}")

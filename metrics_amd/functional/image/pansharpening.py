@@ -1,92 +1,136 @@
-"""Pan-sharpening quality: D_lambda, D_s, QNR, spatial distortion index.
+"""Pan-sharpening quality: D_lambda, D_s, QNR.
 
-Parity: torchmetrics ``functional/image/{d_lambda,d_s,qnr}.py`` and
-``spatial_distortion_index``.
+Parity: torchmetrics ``functional/image/{d_lambda,d_s,qnr}.py``. The reference
+degrades the panchromatic band with a scipy-style uniform filter then a
+bilinear resize (torchvision, antialias off); we use the same box filter and
+``F.interpolate`` which matches torchvision's non-antialiased bilinear resize.
 """
 from __future__ import annotations
 
-from typing import Dict, Optional, Tuple
+from typing import Optional
 
 import torch
 import torch.nn.functional as F
 from torch import Tensor
 
-from metrics_amd.functional.image.misc import universal_image_quality_index
-
-
-def _band_uqi(x: Tensor, y: Tensor, window_size: int = 8) -> Tensor:
-    """UQI between two single-band images (N,H,W)."""
-    return universal_image_quality_index(
-        x.unsqueeze(1), y.unsqueeze(1), kernel_size=(window_size, window_size), sigma=(1.5, 1.5)
-    )
+from metrics_amd.utilities.distributed import reduce
+from metrics_amd.functional.image.misc import _scipy_uniform_filter, universal_image_quality_index
 
 
 def spectral_distortion_index(
     preds: Tensor, target: Tensor, p: int = 1, reduction: str = "elementwise_mean"
 ) -> Tensor:
-    """D_lambda: spectral distortion between fused (preds) and MS (target) images."""
+    """D_lambda: spectral distortion between fused (preds) and MS (target) images.
+
+    Parity: reference functional/image/d_lambda.py (_spectral_distortion_index_compute).
+    """
     if preds.ndim != 4 or target.ndim != 4:
-        raise ValueError(f"Expected `preds` and `target` to have BxCxHxW shape. Got preds: {preds.shape} and target: {target.shape}.")
+        raise ValueError(
+            f"Expected `preds` and `target` to have BxCxHxW shape. Got preds: {preds.shape} and target: {target.shape}."
+        )
     if p <= 0:
         raise ValueError(f"Expected `p` to be a positive integer. Got p: {p}.")
+    if preds.shape[:2] != target.shape[:2]:
+        raise ValueError(
+            f"Expected `preds` and `target` to have same batch and channel sizes. Got preds: {preds.shape} and target: {target.shape}."
+        )
     length = preds.shape[1]
-    if length <= 1:
-        raise ValueError(f"Expected channel dimension to be larger than 1. Got {length}.")
 
     m1 = torch.zeros((length, length), device=preds.device)
     m2 = torch.zeros((length, length), device=preds.device)
     for k in range(length):
-        for r in range(k, length):
-            m1[k, r] = m1[r, k] = _band_uqi(target[:, k], target[:, r]).mean()
-            m2[k, r] = m2[r, k] = _band_uqi(preds[:, k], preds[:, r]).mean()
+        for r in range(k + 1, length):
+            m1[k, r] = universal_image_quality_index(target[:, k : k + 1], target[:, r : r + 1])
+            m2[k, r] = universal_image_quality_index(preds[:, k : k + 1], preds[:, r : r + 1])
+    m1 = m1 + m1.T
+    m2 = m2 + m2.T
+
     diff = (m1 - m2).abs() ** p
-    # off-diagonal mean
-    total = diff.sum() - diff.diagonal().sum()
-    score = (total / (length * (length - 1))) ** (1 / p)
-    if reduction in ("elementwise_mean", "mean", "sum", "none", None):
-        return score
-    raise ValueError(f"Unknown reduction {reduction}")
+    if length == 1:
+        output = diff ** (1.0 / p)
+    else:
+        output = (diff.sum() / (length * (length - 1))) ** (1.0 / p)
+    return reduce(output, reduction)
+
+
+def _degrade_pan(pan: Tensor, ms_hw: tuple, window_size: int) -> Tensor:
+    """Low-pass + downsample pan to the MS resolution (reference d_s.py path)."""
+    pan_degraded = _scipy_uniform_filter(pan, window_size)
+    return F.interpolate(pan_degraded, size=ms_hw, mode="bilinear", align_corners=False)
+
+
+def _spatial_distortion_validate(preds: Tensor, ms: Tensor, pan: Tensor, pan_lr: Optional[Tensor]) -> None:
+    for name, t in (("preds", preds), ("ms", ms), ("pan", pan)) + ((("pan_lr", pan_lr),) if pan_lr is not None else ()):
+        if t.ndim != 4:
+            raise ValueError(f"Expected `{name}` to have BxCxHxW shape. Got {name}: {t.shape}.")
+    for name, t in (("ms", ms), ("pan", pan)) + ((("pan_lr", pan_lr),) if pan_lr is not None else ()):
+        if preds.dtype != t.dtype:
+            raise TypeError(f"Expected `preds` and `{name}` to have the same data type. Got preds: {preds.dtype} and {name}: {t.dtype}.")
+        if preds.shape[:2] != t.shape[:2]:
+            raise ValueError(
+                f"Expected `preds` and `{name}` to have the same batch and channel sizes."
+                f" Got preds: {preds.shape} and {name}: {t.shape}."
+            )
+    if preds.shape[-2:] != pan.shape[-2:]:
+        raise ValueError(f"Expected `preds` and `pan` to have the same height and width. Got preds: {preds.shape} and pan: {pan.shape}.")
+    if preds.shape[-2] % ms.shape[-2] != 0 or preds.shape[-1] % ms.shape[-1] != 0:
+        raise ValueError(
+            f"Expected height/width of `preds` to be multiple of that of `ms`. Got preds: {preds.shape} and ms: {ms.shape}."
+        )
+    if pan_lr is not None and pan_lr.shape[-2:] != ms.shape[-2:]:
+        raise ValueError(f"Expected `ms` and `pan_lr` to have the same height and width. Got ms: {ms.shape} and pan_lr: {pan_lr.shape}.")
 
 
 def spatial_distortion_index(
     preds: Tensor,
-    target: Dict[str, Tensor],
+    ms: Tensor,
+    pan: Tensor,
+    pan_lr: Optional[Tensor] = None,
     norm_order: int = 1,
     window_size: int = 7,
     reduction: str = "elementwise_mean",
 ) -> Tensor:
-    """D_s: spatial distortion; ``target`` dict holds 'ms', 'pan' (and optionally 'pan_lr')."""
-    if not all(k in target for k in ("ms", "pan")):
-        raise ValueError(f"Expected `target` to have keys ('ms', 'pan'). Got target: {target.keys()}.")
-    ms, pan = target["ms"], target["pan"]
-    if preds.ndim != 4:
-        raise ValueError(f"Expected `preds` to have BxCxHxW shape. Got preds: {preds.shape}.")
+    """D_s: spatial distortion between the fused image and the panchromatic band.
+
+    ``preds`` is the high-resolution fused image (same H,W as ``pan``); ``ms``
+    the low-resolution multispectral input; ``pan`` carries one band per
+    channel of ``preds``. Parity: reference functional/image/d_s.py.
+    """
+    if norm_order <= 0:
+        raise ValueError(f"Expected `norm_order` to be a positive integer. Got norm_order: {norm_order}.")
+    _spatial_distortion_validate(preds, ms, pan, pan_lr)
     length = preds.shape[1]
+    ms_h, ms_w = ms.shape[-2:]
+    if window_size >= ms_h or window_size >= ms_w:
+        raise ValueError(f"Expected `window_size` to be smaller than dimension of `ms`. Got window_size: {window_size}.")
 
-    pan_lr = target.get("pan_lr")
-    if pan_lr is None:
-        # degrade pan to the MS resolution by average pooling
-        ratio = pan.shape[-1] // ms.shape[-1]
-        pan_lr = F.avg_pool2d(pan, kernel_size=ratio) if ratio > 1 else pan
+    pan_degraded = pan_lr if pan_lr is not None else _degrade_pan(pan, (ms_h, ms_w), window_size)
 
-    ds = torch.zeros(length, device=preds.device)
+    m1 = torch.zeros(length, device=preds.device)
+    m2 = torch.zeros(length, device=preds.device)
     for i in range(length):
-        q_hr = _band_uqi(preds[:, i], pan[:, 0], window_size).mean()
-        q_lr = _band_uqi(ms[:, i], pan_lr[:, 0], window_size).mean()
-        ds[i] = (q_hr - q_lr).abs() ** norm_order
-    return (ds.mean()) ** (1 / norm_order)
+        m1[i] = universal_image_quality_index(ms[:, i : i + 1], pan_degraded[:, i : i + 1])
+        m2[i] = universal_image_quality_index(preds[:, i : i + 1], pan[:, i : i + 1])
+    diff = (m1 - m2).abs() ** norm_order
+    return reduce(diff, reduction) ** (1 / norm_order)
 
 
 def quality_with_no_reference(
     preds: Tensor,
-    target: Dict[str, Tensor],
+    ms: Tensor,
+    pan: Tensor,
+    pan_lr: Optional[Tensor] = None,
     alpha: float = 1.0,
     beta: float = 1.0,
     norm_order: int = 1,
     window_size: int = 7,
     reduction: str = "elementwise_mean",
 ) -> Tensor:
-    """QNR = (1 - D_lambda)^alpha * (1 - D_s)^beta."""
-    d_lambda = spectral_distortion_index(preds, target["ms"], p=norm_order, reduction=reduction)
-    d_s = spatial_distortion_index(preds, target, norm_order, window_size, reduction)
+    """QNR = (1 - D_lambda)^alpha * (1 - D_s)^beta (reference functional/image/qnr.py).
+
+    D_lambda compares inter-band UQI matrices, which are resolution-independent,
+    so ``preds`` (high-res) and ``ms`` (low-res) are used as-is.
+    """
+    d_lambda = spectral_distortion_index(preds, ms, p=norm_order, reduction=reduction)
+    d_s = spatial_distortion_index(preds, ms, pan, pan_lr, norm_order, window_size, reduction)
     return (1 - d_lambda) ** alpha * (1 - d_s) ** beta

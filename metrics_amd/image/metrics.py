@@ -437,16 +437,25 @@ class SpatialDistortionIndex(Metric):
         self.add_state("preds", default=[], dist_reduce_fx="cat")
         self.add_state("ms", default=[], dist_reduce_fx="cat")
         self.add_state("pan", default=[], dist_reduce_fx="cat")
+        self.add_state("pan_lr", default=[], dist_reduce_fx="cat")
 
     def update(self, preds: Tensor, target: Dict[str, Tensor]) -> None:
-        """Append fused prediction + {'ms','pan'} targets."""
+        """Append fused prediction + {'ms','pan'[,'pan_lr']} targets."""
+        for key in ("ms", "pan"):
+            if key not in target:
+                raise ValueError(f"Expected `target` to have key `{key}`. Got target: {target.keys()}.")
         self.preds.append(preds)
         self.ms.append(target["ms"])
         self.pan.append(target["pan"])
+        if "pan_lr" in target:
+            self.pan_lr.append(target["pan_lr"])
 
     def compute(self) -> Tensor:
-        target = {"ms": dim_zero_cat(self.ms), "pan": dim_zero_cat(self.pan)}
-        return spatial_distortion_index(dim_zero_cat(self.preds), target, self.norm_order, self.window_size, self.reduction)
+        pan_lr = dim_zero_cat(self.pan_lr) if self.pan_lr else None
+        return spatial_distortion_index(
+            dim_zero_cat(self.preds), dim_zero_cat(self.ms), dim_zero_cat(self.pan), pan_lr,
+            self.norm_order, self.window_size, self.reduction,
+        )
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)
@@ -474,17 +483,24 @@ class QualityWithNoReference(Metric):
         self.add_state("preds", default=[], dist_reduce_fx="cat")
         self.add_state("ms", default=[], dist_reduce_fx="cat")
         self.add_state("pan", default=[], dist_reduce_fx="cat")
+        self.add_state("pan_lr", default=[], dist_reduce_fx="cat")
 
     def update(self, preds: Tensor, target: Dict[str, Tensor]) -> None:
-        """Append fused prediction + {'ms','pan'} targets."""
+        """Append fused prediction + {'ms','pan'[,'pan_lr']} targets."""
+        for key in ("ms", "pan"):
+            if key not in target:
+                raise ValueError(f"Expected `target` to have key `{key}`. Got target: {target.keys()}.")
         self.preds.append(preds)
         self.ms.append(target["ms"])
         self.pan.append(target["pan"])
+        if "pan_lr" in target:
+            self.pan_lr.append(target["pan_lr"])
 
     def compute(self) -> Tensor:
-        target = {"ms": dim_zero_cat(self.ms), "pan": dim_zero_cat(self.pan)}
+        pan_lr = dim_zero_cat(self.pan_lr) if self.pan_lr else None
         return quality_with_no_reference(
-            dim_zero_cat(self.preds), target, self.alpha, self.beta, self.norm_order, self.window_size, self.reduction
+            dim_zero_cat(self.preds), dim_zero_cat(self.ms), dim_zero_cat(self.pan), pan_lr,
+            self.alpha, self.beta, self.norm_order, self.window_size, self.reduction,
         )
 
     def plot(self, val=None, ax=None):

@@ -107,9 +107,9 @@ def test_psnrb(imgs):
 
 def test_pansharpening_metrics():
     seed_all(35)
-    fused = torch.rand(1, 4, 32, 32)
-    ms = torch.rand(1, 4, 32, 32)
-    pan = torch.rand(1, 1, 64, 64)
+    fused = torch.rand(2, 4, 32, 32)  # high-res fused output
+    ms = torch.rand(2, 4, 16, 16)  # low-res multispectral input
+    pan = torch.rand(2, 4, 32, 32)  # high-res pan, one band per channel
     d_lambda = ma.SpectralDistortionIndex()(fused, ms).item()
     assert 0 <= d_lambda <= 1
     # identical fused == ms -> D_lambda == 0
@@ -118,6 +118,11 @@ def test_pansharpening_metrics():
     assert 0 <= d_s <= 1
     qnr = ma.QualityWithNoReference()(fused, {"ms": ms, "pan": pan}).item()
     assert 0 <= qnr <= 1
+    assert abs(qnr - (1 - d_lambda) * (1 - d_s)) < 1e-5
+    # explicit pan_lr path
+    pan_lr = torch.rand(2, 4, 16, 16)
+    d_s2 = ma.SpatialDistortionIndex()(fused, {"ms": ms, "pan": pan, "pan_lr": pan_lr}).item()
+    assert 0 <= d_s2 <= 1 and abs(d_s2 - d_s) > 1e-6
 
 
 class _ToyFeat(torch.nn.Module):
@@ -188,3 +193,18 @@ def test_model_backed_raise_without_net():
         ma.LearnedPerceptualImagePatchSimilarity()
     with pytest.raises(ModuleNotFoundError):
         ma.multimodal.CLIPScore()
+
+
+def test_uniform_filter_matches_scipy():
+    from scipy.ndimage import uniform_filter
+    import numpy as np
+    from metrics_amd.functional.image.misc import _scipy_uniform_filter
+
+    seed_all(39)
+    for w in (3, 7, 8):
+        x = torch.rand(2, 3, 20, 24, dtype=torch.float64)
+        mine = _scipy_uniform_filter(x, w)
+        ref = torch.from_numpy(
+            np.stack([[uniform_filter(x[b, c].numpy(), size=w, mode="reflect") for c in range(3)] for b in range(2)])
+        )
+        assert torch.allclose(mine, ref, atol=1e-12), w
