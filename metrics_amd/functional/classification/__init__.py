@@ -113,5 +113,17 @@ from metrics_amd.functional.classification.at_fixed import (
     multilabel_recall_at_fixed_precision,
     multilabel_sensitivity_at_specificity,
     multilabel_specificity_at_sensitivity,
+    logauc,
+    precision_at_fixed_recall,
+    recall_at_fixed_precision,
+    sensitivity_at_specificity,
+    specificity_at_sensitivity,
 )
-from metrics_amd.functional.classification.group_fairness import binary_fairness, binary_groups_stat_rates
+from metrics_amd.functional.classification.group_fairness import (
+    binary_fairness,
+    binary_groups_stat_rates,
+    demographic_parity,
+    equal_opportunity,
+)
+from metrics_amd.functional.classification.dice import dice
+from metrics_amd.functional.segmentation import generalized_dice_score  # noqa: F401  (reference re-export)

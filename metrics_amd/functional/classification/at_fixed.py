@@ -465,3 +465,150 @@ def multilabel_logauc(
     if average == "macro":
         return scores.mean()
     return scores
+
+
+def recall_at_fixed_precision(
+    preds: Tensor,
+    target: Tensor,
+    task: str,
+    min_precision: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    num_classes: Optional[int] = None,
+    num_labels: Optional[int] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Optional[Tuple[Tensor, Tensor]]:
+    """Task-dispatching recall-at-fixed-precision (reference functional/classification/recall_fixed_precision.py:401)."""
+    task = ClassificationTask.from_str(task)
+    if task == ClassificationTask.BINARY:
+        return binary_recall_at_fixed_precision(preds, target, min_precision, thresholds, ignore_index, validate_args)
+    if task == ClassificationTask.MULTICLASS:
+        if not isinstance(num_classes, int):
+            raise ValueError(f"`num_classes` is expected to be `int` but `{type(num_classes)} was passed.`")
+        return multiclass_recall_at_fixed_precision(
+            preds, target, num_classes, min_precision, thresholds, ignore_index, validate_args
+        )
+    if task == ClassificationTask.MULTILABEL:
+        if not isinstance(num_labels, int):
+            raise ValueError(f"`num_labels` is expected to be `int` but `{type(num_labels)} was passed.`")
+        return multilabel_recall_at_fixed_precision(
+            preds, target, num_labels, min_precision, thresholds, ignore_index, validate_args
+        )
+    raise ValueError(f"Not handled value: {task}")
+
+
+def precision_at_fixed_recall(
+    preds: Tensor,
+    target: Tensor,
+    task: str,
+    min_recall: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    num_classes: Optional[int] = None,
+    num_labels: Optional[int] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Optional[Tuple[Tensor, Tensor]]:
+    """Task-dispatching precision-at-fixed-recall (reference functional/classification/precision_fixed_recall.py)."""
+    task = ClassificationTask.from_str(task)
+    if task == ClassificationTask.BINARY:
+        return binary_precision_at_fixed_recall(preds, target, min_recall, thresholds, ignore_index, validate_args)
+    if task == ClassificationTask.MULTICLASS:
+        if not isinstance(num_classes, int):
+            raise ValueError(f"`num_classes` is expected to be `int` but `{type(num_classes)} was passed.`")
+        return multiclass_precision_at_fixed_recall(
+            preds, target, num_classes, min_recall, thresholds, ignore_index, validate_args
+        )
+    if task == ClassificationTask.MULTILABEL:
+        if not isinstance(num_labels, int):
+            raise ValueError(f"`num_labels` is expected to be `int` but `{type(num_labels)} was passed.`")
+        return multilabel_precision_at_fixed_recall(
+            preds, target, num_labels, min_recall, thresholds, ignore_index, validate_args
+        )
+    raise ValueError(f"Not handled value: {task}")
+
+
+def sensitivity_at_specificity(
+    preds: Tensor,
+    target: Tensor,
+    task: str,
+    min_specificity: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    num_classes: Optional[int] = None,
+    num_labels: Optional[int] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Optional[Tuple[Tensor, Tensor]]:
+    """Task-dispatching sensitivity-at-specificity (reference functional/classification/sensitivity_specificity.py)."""
+    task = ClassificationTask.from_str(task)
+    if task == ClassificationTask.BINARY:
+        return binary_sensitivity_at_specificity(preds, target, min_specificity, thresholds, ignore_index, validate_args)
+    if task == ClassificationTask.MULTICLASS:
+        if not isinstance(num_classes, int):
+            raise ValueError(f"`num_classes` is expected to be `int` but `{type(num_classes)} was passed.`")
+        return multiclass_sensitivity_at_specificity(
+            preds, target, num_classes, min_specificity, thresholds, ignore_index, validate_args
+        )
+    if task == ClassificationTask.MULTILABEL:
+        if not isinstance(num_labels, int):
+            raise ValueError(f"`num_labels` is expected to be `int` but `{type(num_labels)} was passed.`")
+        return multilabel_sensitivity_at_specificity(
+            preds, target, num_labels, min_specificity, thresholds, ignore_index, validate_args
+        )
+    raise ValueError(f"Not handled value: {task}")
+
+
+def specificity_at_sensitivity(
+    preds: Tensor,
+    target: Tensor,
+    task: str,
+    min_sensitivity: float,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    num_classes: Optional[int] = None,
+    num_labels: Optional[int] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Optional[Tuple[Tensor, Tensor]]:
+    """Task-dispatching specificity-at-sensitivity (reference functional/classification/specificity_sensitivity.py)."""
+    task = ClassificationTask.from_str(task)
+    if task == ClassificationTask.BINARY:
+        return binary_specificity_at_sensitivity(preds, target, min_sensitivity, thresholds, ignore_index, validate_args)
+    if task == ClassificationTask.MULTICLASS:
+        if not isinstance(num_classes, int):
+            raise ValueError(f"`num_classes` is expected to be `int` but `{type(num_classes)} was passed.`")
+        return multiclass_specificity_at_sensitivity(
+            preds, target, num_classes, min_sensitivity, thresholds, ignore_index, validate_args
+        )
+    if task == ClassificationTask.MULTILABEL:
+        if not isinstance(num_labels, int):
+            raise ValueError(f"`num_labels` is expected to be `int` but `{type(num_labels)} was passed.`")
+        return multilabel_specificity_at_sensitivity(
+            preds, target, num_labels, min_sensitivity, thresholds, ignore_index, validate_args
+        )
+    raise ValueError(f"Not handled value: {task}")
+
+
+def logauc(
+    preds: Tensor,
+    target: Tensor,
+    task: str,
+    thresholds: Optional[Union[int, List[float], Tensor]] = None,
+    num_classes: Optional[int] = None,
+    num_labels: Optional[int] = None,
+    fpr_range: Tuple[float, float] = (0.001, 0.1),
+    average: Optional[str] = None,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Optional[Tensor]:
+    """Task-dispatching LogAUC (reference functional/classification/logauc.py:323)."""
+    task = ClassificationTask.from_str(task)
+    if task == ClassificationTask.BINARY:
+        return binary_logauc(preds, target, fpr_range, thresholds, ignore_index, validate_args)
+    if task == ClassificationTask.MULTICLASS:
+        if not isinstance(num_classes, int):
+            raise ValueError(f"`num_classes` is expected to be `int` but `{type(num_classes)} was passed.`")
+        return multiclass_logauc(preds, target, num_classes, fpr_range, average or "macro", thresholds, ignore_index, validate_args)
+    if task == ClassificationTask.MULTILABEL:
+        if not isinstance(num_labels, int):
+            raise ValueError(f"`num_labels` is expected to be `int` but `{type(num_labels)} was passed.`")
+        return multilabel_logauc(preds, target, num_labels, fpr_range, average or "macro", thresholds, ignore_index, validate_args)
+    raise ValueError(f"Not handled value: {task}")

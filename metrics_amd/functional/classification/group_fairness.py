@@ -123,3 +123,33 @@ def binary_fairness(
         max_g = int(torch.argmax(rates))
         out[f"EO_{min_g}_{max_g}"] = _safe_divide(rates[min_g], rates[max_g])
     return out
+
+
+def demographic_parity(
+    preds: Tensor,
+    groups: Tensor,
+    threshold: float = 0.5,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Dict[str, Tensor]:
+    """Demographic parity: min/max positivity-rate ratio between groups.
+
+    Parity: reference functional/classification/group_fairness.py:177.
+    """
+    return binary_fairness(preds, torch.zeros_like(preds, dtype=torch.long), groups,
+                           "demographic_parity", threshold, ignore_index, validate_args)
+
+
+def equal_opportunity(
+    preds: Tensor,
+    target: Tensor,
+    groups: Tensor,
+    threshold: float = 0.5,
+    ignore_index: Optional[int] = None,
+    validate_args: bool = True,
+) -> Dict[str, Tensor]:
+    """Equal opportunity: min/max true-positive-rate ratio between groups.
+
+    Parity: reference functional/classification/group_fairness.py:258.
+    """
+    return binary_fairness(preds, target, groups, "equal_opportunity", threshold, ignore_index, validate_args)

@@ -21,3 +21,8 @@ from metrics_amd.functional.image.ssim import (
     multiscale_structural_similarity_index_measure,
     structural_similarity_index_measure,
 )
+from metrics_amd.functional.image.gradients import image_gradients
+from metrics_amd.functional.image.perceptual import (
+    learned_perceptual_image_patch_similarity,
+    perceptual_path_length,
+)

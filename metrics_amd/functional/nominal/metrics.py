@@ -210,3 +210,70 @@ def fleiss_kappa(ratings: Tensor, mode: str = "counts") -> Tensor:
     p_bar = p_subject.mean()
     p_e = (p_cat**2).sum()
     return (p_bar - p_e) / (1 - p_e)
+
+
+def _matrix_over_columns(matrix: Tensor, pair_fn, symmetric: bool) -> Tensor:
+    """Pairwise nominal statistic over the columns of a (N, V) data matrix."""
+    if matrix.ndim != 2:
+        raise ValueError(f"Expected `matrix` to be a 2D tensor of shape (observations, variables), got {matrix.shape}")
+    num_variables = matrix.shape[1]
+    out = torch.ones(num_variables, num_variables, device=matrix.device)
+    for i in range(num_variables):
+        for j in range(i + 1, num_variables):
+            x, y = matrix[:, i], matrix[:, j]
+            out[i, j] = pair_fn(x, y)
+            out[j, i] = out[i, j] if symmetric else pair_fn(y, x)
+    return out
+
+
+def cramers_v_matrix(
+    matrix: Tensor,
+    bias_correction: bool = True,
+    nan_strategy: str = "replace",
+    nan_replace_value: Optional[float] = 0.0,
+) -> Tensor:
+    """Pairwise Cramer's V over the columns of a categorical data matrix.
+
+    Parity: reference functional/nominal/cramers.py:141.
+    """
+    _nominal_input_validation(nan_strategy, nan_replace_value)
+    return _matrix_over_columns(
+        matrix, lambda x, y: cramers_v(x, y, bias_correction, nan_strategy, nan_replace_value), symmetric=True
+    )
+
+
+def pearsons_contingency_coefficient_matrix(
+    matrix: Tensor,
+    nan_strategy: str = "replace",
+    nan_replace_value: Optional[float] = 0.0,
+) -> Tensor:
+    """Pairwise Pearson's contingency coefficient over data-matrix columns (reference functional/nominal/pearson.py)."""
+    _nominal_input_validation(nan_strategy, nan_replace_value)
+    return _matrix_over_columns(
+        matrix, lambda x, y: pearsons_contingency_coefficient(x, y, nan_strategy, nan_replace_value), symmetric=True
+    )
+
+
+def tschuprows_t_matrix(
+    matrix: Tensor,
+    bias_correction: bool = True,
+    nan_strategy: str = "replace",
+    nan_replace_value: Optional[float] = 0.0,
+) -> Tensor:
+    """Pairwise Tschuprow's T over data-matrix columns (reference functional/nominal/tschuprows.py)."""
+    _nominal_input_validation(nan_strategy, nan_replace_value)
+    return _matrix_over_columns(
+        matrix, lambda x, y: tschuprows_t(x, y, bias_correction, nan_strategy, nan_replace_value), symmetric=True
+    )
+
+
+def theils_u_matrix(
+    matrix: Tensor,
+    nan_strategy: str = "replace",
+    nan_replace_value: Optional[float] = 0.0,
+) -> Tensor:
+    """Pairwise (asymmetric) Theil's U over data-matrix columns (reference functional/nominal/theils_u.py)."""
+    _nominal_input_validation(nan_strategy, nan_replace_value)
+    return _matrix_over_columns(
+        matrix, lambda x, y: theils_u(x, y, nan_strategy, nan_replace_value), symmetric=False
+    )

@@ -7,7 +7,7 @@ from metrics_amd.wrappers.multioutput import MultioutputWrapper
 from metrics_amd.wrappers.multitask import MultitaskWrapper
 from metrics_amd.wrappers.running import Running
 from metrics_amd.wrappers.tracker import MetricTracker
-from metrics_amd.wrappers.transformations import BinaryTargetTransformer, LambdaInputTransformer
+from metrics_amd.wrappers.transformations import BinaryTargetTransformer, LambdaInputTransformer, MetricInputTransformer
 
 __all__ = [
     "BinaryTargetTransformer",
@@ -15,6 +15,7 @@ __all__ = [
     "ClasswiseWrapper",
     "FeatureShare",
     "LambdaInputTransformer",
+    "MetricInputTransformer",
     "MetricTracker",
     "MinMaxMetric",
     "MultioutputWrapper",
