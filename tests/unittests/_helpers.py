@@ -105,6 +105,10 @@ def run_functional_metric_test(
 
 # ---------------------------------------------------------------- distributed
 def _dist_worker(rank: int, world_size: int, port: int, fn: Callable, args: tuple) -> None:
+    # Forked children inherit the parent's (possibly mid-operation) OpenMP pool
+    # state; entering a parallel region then deadlocks on a stale futex. Keep
+    # the child single-threaded so no parallel region is ever entered.
+    torch.set_num_threads(1)
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world_size)
@@ -114,9 +118,19 @@ def _dist_worker(rank: int, world_size: int, port: int, fn: Callable, args: tupl
         dist.destroy_process_group()
 
 
+def _free_port() -> int:
+    """Ask the OS for a free TCP port (avoids RNG-driven collisions across tests)."""
+    import socket
+
+    with socket.socket(socket.AF_INET, socket.SOCK_STREAM) as s:
+        s.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def run_distributed(fn: Callable, world_size: int = 2, args: tuple = ()) -> None:
     """Run ``fn(rank, world_size, *args)`` in ``world_size`` gloo processes on localhost."""
-    port = np.random.randint(20000, 65000)
+    port = _free_port()
     mp.start_processes(
         partial(_dist_worker, world_size=world_size, port=port, fn=fn, args=args),
         nprocs=world_size,

@@ -216,3 +216,105 @@ def _test_metric_collection_ddp(rank, world_size):
 
 def test_metric_collection_ddp():
     run_distributed(_test_metric_collection_ddp, world_size=2)
+
+
+def _test_curve_metrics_ddp(rank, world_size):
+    """Thresholded curve metrics (confmat sum states) must equal a single-process run."""
+    import metrics_amd as ma
+
+    torch.manual_seed(7)
+    preds = torch.rand(4, 64)
+    target = torch.randint(0, 2, (4, 64))
+    makes = [
+        lambda: ma.BinaryPrecisionRecallCurve(thresholds=25),
+        lambda: ma.BinaryAUROC(thresholds=25),
+        lambda: ma.BinaryAveragePrecision(thresholds=25),
+        lambda: ma.BinaryROC(thresholds=25),
+    ]
+    for make in makes:
+        m = make()
+        for i in range(rank, 4, world_size):
+            m.update(preds[i], target[i])
+        res = m.compute()
+        ref = make()
+        ref.sync_on_compute = False
+        for i in range(4):
+            ref.update(preds[i], target[i])
+        expected = ref.compute()
+        if isinstance(res, tuple):
+            for a, b in zip(res, expected):
+                assert torch.allclose(a, b, atol=1e-6)
+        else:
+            assert torch.allclose(res, expected, atol=1e-6)
+
+
+def _test_mc_curve_ddp(rank, world_size):
+    import metrics_amd as ma
+
+    torch.manual_seed(8)
+    preds = torch.randn(4, 32, 7).softmax(-1)
+    target = torch.randint(0, 7, (4, 32))
+    m = ma.MulticlassAUROC(num_classes=7, thresholds=50)
+    for i in range(rank, 4, world_size):
+        m.update(preds[i], target[i])
+    res = m.compute()
+    ref = ma.MulticlassAUROC(num_classes=7, thresholds=50)
+    ref.sync_on_compute = False
+    for i in range(4):
+        ref.update(preds[i], target[i])
+    assert torch.allclose(res, ref.compute(), atol=1e-6)
+
+
+def _test_pearson_welford_merge_ddp(rank, world_size):
+    """Welford moment states merge across ranks via the custom parallel-merge path."""
+    import metrics_amd as ma
+
+    torch.manual_seed(9)
+    x = torch.randn(4, 50)
+    y = 0.5 * x + 0.3 * torch.randn(4, 50)
+    m = ma.PearsonCorrCoef()
+    for i in range(rank, 4, world_size):
+        m.update(x[i], y[i])
+    res = m.compute()
+    ref = ma.PearsonCorrCoef()
+    ref.sync_on_compute = False
+    for i in range(4):
+        ref.update(x[i], y[i])
+    assert torch.allclose(res, ref.compute(), atol=1e-5)
+
+
+def _test_mean_ap_ddp(rank, world_size):
+    """List states (cat) in MeanAveragePrecision gather correctly."""
+    import metrics_amd as ma
+
+    torch.manual_seed(10)
+    def boxes(n):
+        xy = torch.rand(n, 2) * 50
+        wh = torch.rand(n, 2) * 20 + 2
+        return torch.cat([xy, xy + wh], 1)
+
+    all_preds, all_tgts = [], []
+    for i in range(4):
+        n = 3 + i
+        all_preds.append({"boxes": boxes(n), "scores": torch.rand(n), "labels": torch.randint(0, 2, (n,))})
+        all_tgts.append({"boxes": boxes(3), "labels": torch.randint(0, 2, (3,))})
+
+    m = ma.detection.MeanAveragePrecision()
+    for i in range(rank, 4, world_size):
+        m.update([all_preds[i]], [all_tgts[i]])
+    res = m.compute()
+    ref = ma.detection.MeanAveragePrecision()
+    ref.sync_on_compute = False
+    for i in range(4):
+        ref.update([all_preds[i]], [all_tgts[i]])
+    expected = ref.compute()
+    assert torch.allclose(res["map"], expected["map"], atol=1e-6), (res["map"], expected["map"])
+    assert torch.allclose(res["mar_100"], expected["mar_100"], atol=1e-6)
+
+
+@pytest.mark.parametrize(
+    "fn",
+    [_test_curve_metrics_ddp, _test_mc_curve_ddp, _test_pearson_welford_merge_ddp, _test_mean_ap_ddp],
+)
+def test_ddp_metric_parity(fn):
+    run_distributed(fn, world_size=2)
