@@ -57,8 +57,18 @@ def box_convert(boxes: Tensor, in_fmt: str, out_fmt: str) -> Tensor:
     raise ValueError(f"Unknown box format {out_fmt}")
 
 
-def _input_validator(preds, targets, iou_type="bbox") -> None:
-    item = "boxes" if iou_type == "bbox" else "masks"
+def _validate_iou_type_arg(iou_type) -> Tuple[str, ...]:
+    """Normalize ``iou_type`` to a tuple; reference detection/helpers.py (_validate_iou_type_arg)."""
+    allowed = ("bbox", "segm")
+    if isinstance(iou_type, str):
+        iou_type = (iou_type,)
+    if not all(t in allowed for t in iou_type):
+        raise ValueError(f"Expected argument `iou_type` to be one of {allowed} or a tuple of them, but got {iou_type}")
+    return tuple(iou_type)
+
+
+def _input_validator(preds, targets, iou_type=("bbox",)) -> None:
+    items = ["boxes" if t == "bbox" else "masks" for t in iou_type]
     if not isinstance(preds, Sequence):
         raise ValueError(f"Expected argument `preds` to be of type Sequence, but got {preds}")
     if not isinstance(targets, Sequence):
@@ -67,12 +77,61 @@ def _input_validator(preds, targets, iou_type="bbox") -> None:
         raise ValueError(
             f"Expected argument `preds` and `target` to have the same length, but got {len(preds)} and {len(targets)}"
         )
-    for k in (item, "scores", "labels"):
+    for k in items + ["scores", "labels"]:
         if any(k not in p for p in preds):
             raise ValueError(f"Expected all dicts in `preds` to contain the `{k}` key")
-    for k in (item, "labels"):
+    for k in items + ["labels"]:
         if any(k not in p for p in targets):
             raise ValueError(f"Expected all dicts in `target` to contain the `{k}` key")
+
+
+def _encode_masks_rle(masks: Tensor) -> Tensor:
+    """Pack (N, H, W) boolean masks into one 1-D int64 RLE tensor.
+
+    Layout: ``[H, W, N, len_0..len_{N-1}, counts_0 ... counts_{N-1}]`` with
+    COCO column-major run lengths (alternating background/foreground, starting
+    with background). Keeps mask states compact and single-tensor per image so
+    the list-state gather path can sync them.
+    """
+    if masks.numel() == 0:
+        shape = masks.shape
+        h = shape[-2] if masks.ndim >= 2 else 0
+        w = shape[-1] if masks.ndim >= 2 else 0
+        return torch.tensor([h, w, 0], dtype=torch.int64)
+    if masks.ndim != 3:
+        raise ValueError(f"Expected `masks` to have (N, H, W) shape, got {tuple(masks.shape)}")
+    n, h, w = masks.shape
+    flat = masks.detach().to(torch.bool).cpu().transpose(1, 2).reshape(n, -1)  # column-major per mask
+    counts_per_mask = []
+    for i in range(n):
+        f = flat[i]
+        change = torch.nonzero(f[1:] != f[:-1]).flatten() + 1
+        idx = torch.cat([torch.zeros(1, dtype=torch.long), change, torch.tensor([f.numel()])])
+        counts = idx[1:] - idx[:-1]
+        if bool(f[0]):
+            counts = torch.cat([torch.zeros(1, dtype=torch.long), counts])
+        counts_per_mask.append(counts)
+    header = torch.tensor([h, w, n] + [c.numel() for c in counts_per_mask], dtype=torch.int64)
+    return torch.cat([header] + counts_per_mask)
+
+
+def _decode_masks_rle(pack: "np.ndarray") -> "np.ndarray":
+    """Inverse of :func:`_encode_masks_rle`: returns (N, H*W) uint8, column-major pixels."""
+    import numpy as np
+
+    h, w, n = int(pack[0]), int(pack[1]), int(pack[2])
+    if n == 0:
+        return np.zeros((0, h * w), dtype=np.uint8)
+    lens = pack[3 : 3 + n].astype(np.int64)
+    out = np.zeros((n, h * w), dtype=np.uint8)
+    off = 3 + n
+    for i in range(n):
+        counts = pack[off : off + lens[i]]
+        off += lens[i]
+        vals = np.zeros(len(counts), dtype=np.uint8)
+        vals[1::2] = 1
+        out[i] = np.repeat(vals, counts)
+    return out
 
 
 class MeanAveragePrecision(Metric):
@@ -108,12 +167,14 @@ class MeanAveragePrecision(Metric):
         super().__init__(**kwargs)
         if box_format not in ("xyxy", "xywh", "cxcywh"):
             raise ValueError(f"Expected argument `box_format` to be one of ('xyxy', 'xywh', 'cxcywh') but got {box_format}")
-        if iou_type not in ("bbox",):
-            raise ValueError(f"This MI355X-native evaluator supports iou_type='bbox'; got {iou_type} (segm RLE planned)")
         self.box_format = box_format
-        self.iou_type = iou_type
-        self.iou_thresholds = iou_thresholds or torch.linspace(0.5, 0.95, 10).tolist()
-        self.rec_thresholds = rec_thresholds or torch.linspace(0.0, 1.0, 101).tolist()
+        self.iou_type = _validate_iou_type_arg(iou_type)
+        # the reference builds this grid with float32 torch.linspace
+        # (mean_ap.py:411), so thr[2] = 0.6000000238… — and the backends
+        # compare float64 IoU against it, which decides exact-ratio IoU cases
+        # like 3/5; replicate bit-for-bit
+        self.iou_thresholds = iou_thresholds or torch.linspace(0.5, 0.95, round((0.95 - 0.5) / 0.05) + 1).tolist()
+        self.rec_thresholds = rec_thresholds or torch.linspace(0.0, 1.0, round(1.0 / 0.01) + 1).tolist()
         self.max_detection_thresholds = sorted(max_detection_thresholds or [1, 10, 100])
         if not isinstance(class_metrics, bool):
             raise ValueError("Expected argument `class_metrics` to be a boolean")
@@ -125,9 +186,11 @@ class MeanAveragePrecision(Metric):
         self.backend = backend
 
         self.add_state("detection_boxes", default=[], dist_reduce_fx=None)
+        self.add_state("detection_masks", default=[], dist_reduce_fx=None)
         self.add_state("detection_scores", default=[], dist_reduce_fx=None)
         self.add_state("detection_labels", default=[], dist_reduce_fx=None)
         self.add_state("groundtruth_boxes", default=[], dist_reduce_fx=None)
+        self.add_state("groundtruth_masks", default=[], dist_reduce_fx=None)
         self.add_state("groundtruth_labels", default=[], dist_reduce_fx=None)
         self.add_state("groundtruth_crowds", default=[], dist_reduce_fx=None)
         self.add_state("groundtruth_area", default=[], dist_reduce_fx=None)
@@ -143,24 +206,39 @@ class MeanAveragePrecision(Metric):
     def update(self, preds: List[Dict[str, Tensor]], target: List[Dict[str, Tensor]]) -> None:
         """Append per-image detections and ground truths."""
         _input_validator(preds, target, self.iou_type)
+        use_boxes = "bbox" in self.iou_type
+        use_masks = "segm" in self.iou_type
         for item in preds:
-            boxes = _fix_empty_tensors(item["boxes"]).float()
-            boxes = box_convert(boxes, in_fmt=self.box_format, out_fmt="xyxy")
-            self.detection_boxes.append(boxes)
+            n = item["labels"].shape[0]
+            if use_boxes:
+                boxes = _fix_empty_tensors(item["boxes"]).float()
+                boxes = box_convert(boxes, in_fmt=self.box_format, out_fmt="xyxy")
+                self.detection_boxes.append(boxes)
+            else:
+                self.detection_boxes.append(torch.zeros(n, 4, device=item["labels"].device))
+            if use_masks:
+                self.detection_masks.append(_encode_masks_rle(item["masks"]))
             self.detection_scores.append(item["scores"].float())
             self.detection_labels.append(item["labels"].long())
         for item in target:
-            boxes = _fix_empty_tensors(item["boxes"]).float()
-            boxes = box_convert(boxes, in_fmt=self.box_format, out_fmt="xyxy")
-            self.groundtruth_boxes.append(boxes)
+            n = item["labels"].shape[0]
+            if use_boxes:
+                boxes = _fix_empty_tensors(item["boxes"]).float()
+                boxes = box_convert(boxes, in_fmt=self.box_format, out_fmt="xyxy")
+                self.groundtruth_boxes.append(boxes)
+            else:
+                self.groundtruth_boxes.append(torch.zeros(n, 4, device=item["labels"].device))
+            if use_masks:
+                self.groundtruth_masks.append(_encode_masks_rle(item["masks"]))
             self.groundtruth_labels.append(item["labels"].long())
-            n = boxes.shape[0]
-            crowds = item.get("iscrowd", torch.zeros(n, dtype=torch.long, device=boxes.device)).long()
+            device = item["labels"].device
+            crowds = item.get("iscrowd", torch.zeros(n, dtype=torch.long, device=device)).long()
             self.groundtruth_crowds.append(crowds)
-            default_area = (boxes[:, 2] - boxes[:, 0]) * (boxes[:, 3] - boxes[:, 1]) if n else torch.zeros(0, device=boxes.device)
-            area = item.get("area", default_area).float()
+            # -1 sentinel: the per-iou-type default (box area vs mask area) is
+            # substituted at compute time (the two differ for tuple iou_type)
+            area = item.get("area", torch.full((n,), -1.0, device=device)).float()
             if area.numel() == 0 and n:
-                area = default_area
+                area = torch.full((n,), -1.0, device=device)
             self.groundtruth_area.append(area)
 
     @staticmethod
@@ -237,8 +315,13 @@ class MeanAveragePrecision(Metric):
         n_valid_gt = int((~gt_ignore).sum())
         return scores, dtm, dti, n_valid_gt
 
-    def _eval_class_numpy(self, db, gb, gc, ga, dt_off, gt_off, area_rngs, iou_thrs):
-        """Fallback matcher with the same packed interface as the native one."""
+    def _eval_class_numpy(self, db, gb, gc, ga, dt_off, gt_off, area_rngs, iou_thrs, iou_fn=None, det_areas=None):
+        """Fallback matcher with the same packed interface as the native one.
+
+        ``iou_fn(img, d0, d1, g0, g1) -> (n_dt, n_gt) IoU array`` overrides the
+        box IoU (used for mask IoU); ``det_areas`` overrides box-derived
+        detection areas for the area-range ignore flags.
+        """
         import numpy as np
 
         n_imgs = len(dt_off) - 1
@@ -253,8 +336,14 @@ class MeanAveragePrecision(Metric):
             g0, g1 = int(gt_off[img]), int(gt_off[img + 1])
             if d1 == d0 and g1 == g0:
                 continue
-            ious = self._iou_np(db[d0:d1], gb[g0:g1], gc[g0:g1])
-            dareas = (db[d0:d1, 2] - db[d0:d1, 0]) * (db[d0:d1, 3] - db[d0:d1, 1])
+            if iou_fn is not None:
+                ious = iou_fn(img, d0, d1, g0, g1)
+            else:
+                ious = self._iou_np(db[d0:d1], gb[g0:g1], gc[g0:g1])
+            if det_areas is not None:
+                dareas = det_areas[d0:d1]
+            else:
+                dareas = (db[d0:d1, 2] - db[d0:d1, 0]) * (db[d0:d1, 3] - db[d0:d1, 1])
             for ai in range(A):
                 _, m, ig, n_valid = self._evaluate_img_np(
                     ious, np.zeros(d1 - d0), dareas, gc[g0:g1] == 1, gc[g0:g1], ga[g0:g1],
@@ -265,11 +354,39 @@ class MeanAveragePrecision(Metric):
                 npig[ai] += n_valid
         return dtm, dti, npig
 
+    @staticmethod
+    def _rle_areas(pack: "np.ndarray") -> "np.ndarray":
+        """Foreground pixel count per mask straight from the RLE pack (no decode)."""
+        import numpy as np
+
+        n = int(pack[2])
+        areas = np.zeros(n, dtype=np.float32)
+        if n == 0:
+            return areas
+        lens = pack[3 : 3 + n].astype(np.int64)
+        off = 3 + n
+        for i in range(n):
+            areas[i] = pack[off + 1 : off + lens[i] : 2].sum()
+            off += lens[i]
+        return areas
+
     def compute(self) -> Dict[str, Tensor]:
-        """COCO mAP/mAR summary over all accumulated images.
+        """COCO mAP/mAR summary; one result set per iou_type (prefixed when >1)."""
+        result: Dict[str, Tensor] = {}
+        for i_type in self.iou_type:
+            prefix = "" if len(self.iou_type) == 1 else f"{i_type}_"
+            one = self._compute_one_type(i_type)
+            for k, v in one.items():
+                result["classes" if k == "classes" else f"{prefix}{k}"] = v
+        return result
+
+    def _compute_one_type(self, i_type: str) -> Dict[str, Tensor]:
+        """COCO mAP/mAR summary over all accumulated images for one iou_type.
 
         Engine: one host transfer of all boxes, per-(image,class) IoU cached
         across area ranges, numpy greedy matching, vectorized accumulation.
+        Mask IoU (segm) decodes each image's RLE pack only for classes present
+        in that image and takes the intersection via a uint8 matmul.
         """
         import numpy as np
 
@@ -277,9 +394,10 @@ class MeanAveragePrecision(Metric):
         iou_thrs = np.array(self.iou_thresholds)
         rec_thrs = np.array(self.rec_thresholds)
         T = len(self.iou_thresholds)
-        n_imgs = len(self.detection_boxes)
+        n_imgs = len(self.detection_labels)
         max_dets = self.max_detection_thresholds
         max_det_top = max_dets[-1]
+        use_masks = i_type == "segm"
 
         # one transfer to host
         det_boxes = [b.detach().cpu().numpy() for b in self.detection_boxes]
@@ -288,7 +406,23 @@ class MeanAveragePrecision(Metric):
         gt_boxes = [b.detach().cpu().numpy() for b in self.groundtruth_boxes]
         gt_labels = [l.detach().cpu().numpy() for l in self.groundtruth_labels]
         gt_crowds = [c.detach().cpu().numpy() for c in self.groundtruth_crowds]
-        gt_areas = [a.detach().cpu().numpy() for a in self.groundtruth_area]
+        user_areas = [a.detach().cpu().numpy() for a in self.groundtruth_area]
+        det_packs = [m.detach().cpu().numpy() for m in self.detection_masks] if use_masks else None
+        gt_packs = [m.detach().cpu().numpy() for m in self.groundtruth_masks] if use_masks else None
+
+        # per-type default gt area: box area for bbox, mask pixel count for segm
+        gt_areas = []
+        det_rle_areas = []
+        for i in range(n_imgs):
+            a = user_areas[i].astype(np.float32)
+            if use_masks:
+                default = self._rle_areas(gt_packs[i])
+            else:
+                b = gt_boxes[i]
+                default = ((b[:, 2] - b[:, 0]) * (b[:, 3] - b[:, 1])).astype(np.float32) if b.size else np.zeros(0, np.float32)
+            gt_areas.append(np.where(a >= 0, a, default) if a.size else default)
+            if use_masks:
+                det_rle_areas.append(self._rle_areas(det_packs[i]))
 
         classes = sorted(set(np.concatenate(det_labels + gt_labels).tolist())) if n_imgs else []
 
@@ -308,19 +442,26 @@ class MeanAveragePrecision(Metric):
 
         # ---- global packing: one lexsort puts dets contiguous by (class, img, -score)
         img_ids_d = np.concatenate([np.full(det_labels[i].shape[0], i, dtype=np.int64) for i in range(n_imgs)]) if n_imgs else np.zeros(0, dtype=np.int64)
+        orig_d = np.concatenate([np.arange(det_labels[i].shape[0], dtype=np.int64) for i in range(n_imgs)]) if n_imgs else np.zeros(0, dtype=np.int64)
         all_db = np.concatenate(det_boxes).reshape(-1, 4).astype(np.float32) if n_imgs else np.zeros((0, 4), np.float32)
         all_ds = np.concatenate(det_scores).astype(np.float32) if n_imgs else np.zeros(0, np.float32)
         all_dl = np.concatenate(det_labels) if n_imgs else np.zeros(0, np.int64)
+        all_da = (np.concatenate(det_rle_areas).astype(np.float32) if n_imgs else np.zeros(0, np.float32)) if use_masks else None
         order = np.lexsort((-all_ds, img_ids_d, all_dl))
-        all_db, all_ds, all_dl, img_ids_d = all_db[order], all_ds[order], all_dl[order], img_ids_d[order]
+        all_db, all_ds, all_dl, img_ids_d, orig_d = all_db[order], all_ds[order], all_dl[order], img_ids_d[order], orig_d[order]
+        if use_masks:
+            all_da = all_da[order]
 
         img_ids_g = np.concatenate([np.full(gt_labels[i].shape[0], i, dtype=np.int64) for i in range(n_imgs)]) if n_imgs else np.zeros(0, dtype=np.int64)
+        orig_g = np.concatenate([np.arange(gt_labels[i].shape[0], dtype=np.int64) for i in range(n_imgs)]) if n_imgs else np.zeros(0, dtype=np.int64)
         all_gb = np.concatenate(gt_boxes).reshape(-1, 4).astype(np.float32) if n_imgs else np.zeros((0, 4), np.float32)
         all_gl = np.concatenate(gt_labels) if n_imgs else np.zeros(0, np.int64)
         all_gc = np.concatenate(gt_crowds).astype(np.uint8) if n_imgs else np.zeros(0, np.uint8)
         all_ga = np.concatenate(gt_areas).astype(np.float32) if n_imgs else np.zeros(0, np.float32)
         orderg = np.lexsort((img_ids_g, all_gl))
-        all_gb, all_gl, all_gc, all_ga, img_ids_g = all_gb[orderg], all_gl[orderg], all_gc[orderg], all_ga[orderg], img_ids_g[orderg]
+        all_gb, all_gl, all_gc, all_ga, img_ids_g, orig_g = (
+            all_gb[orderg], all_gl[orderg], all_gc[orderg], all_ga[orderg], img_ids_g[orderg], orig_g[orderg]
+        )
 
         cls_index = {c: i for i, c in enumerate(classes)}
         dl_idx = np.array([cls_index[c] for c in all_dl.tolist()], dtype=np.int64) if all_dl.size else np.zeros(0, np.int64)
@@ -339,9 +480,11 @@ class MeanAveragePrecision(Metric):
         # cap at the TOP max_det once (native matcher sees <= max_det_top dets)
         keep_top = d_rank < max_det_top
         if not keep_top.all():
-            all_db, all_ds, dl_idx, img_ids_d, d_rank = (
-                all_db[keep_top], all_ds[keep_top], dl_idx[keep_top], img_ids_d[keep_top], d_rank[keep_top]
+            all_db, all_ds, dl_idx, img_ids_d, d_rank, orig_d = (
+                all_db[keep_top], all_ds[keep_top], dl_idx[keep_top], img_ids_d[keep_top], d_rank[keep_top], orig_d[keep_top]
             )
+            if use_masks:
+                all_da = all_da[keep_top]
             d_counts = np.minimum(d_counts, max_det_top)
 
         d_cls_off = np.concatenate([[0], np.cumsum(d_counts.sum(1))])
@@ -359,7 +502,31 @@ class MeanAveragePrecision(Metric):
             scores_k = all_ds[dlo:dhi]
             rank_k = d_rank[dlo:dhi]
 
-            if use_native:
+            if use_masks:
+                d_orig_k = orig_d[dlo:dhi]
+                g_orig_k = orig_g[glo:ghi]
+                gc_k = all_gc[glo:ghi]
+                da_k = all_da[dlo:dhi]
+
+                def _mask_iou(img, d0, d1, g0, g1):
+                    # float64 division: pycocotools computes IoU in double, and
+                    # exact ratios like 3/5 sit right at a threshold gridpoint
+                    dm = _decode_masks_rle(det_packs[img])[d_orig_k[d0:d1]].astype(np.float64)
+                    gm = _decode_masks_rle(gt_packs[img])[g_orig_k[g0:g1]].astype(np.float64)
+                    if dm.shape[0] == 0 or gm.shape[0] == 0:
+                        return np.zeros((dm.shape[0], gm.shape[0]))
+                    inter = dm @ gm.T
+                    area_d = dm.sum(1)
+                    area_g = gm.sum(1)
+                    union = area_d[:, None] + area_g[None, :] - inter
+                    union = np.where(gc_k[g0:g1][None, :] == 1, area_d[:, None], union)
+                    return np.where(union > 0, inter / np.maximum(union, 1e-9), 0.0)
+
+                dtm_a, dti_a, npig_a = self._eval_class_numpy(
+                    all_db[dlo:dhi], all_gb[glo:ghi], gc_k, all_ga[glo:ghi],
+                    dt_off, gt_off, area_rngs, iou_thrs, iou_fn=_mask_iou, det_areas=da_k,
+                )
+            elif use_native:
                 dtm_a, dti_a, npig_a = coco_eval_class_packed(
                     all_db[dlo:dhi], all_gb[glo:ghi], all_gc[glo:ghi], all_ga[glo:ghi],
                     dt_off, gt_off, area_rngs, iou_thrs.astype(np.float32),
