@@ -1,46 +1,79 @@
 """Kendall rank correlation (tau-a / tau-b / tau-c).
 
-Parity: torchmetrics ``functional/regression/kendall.py``. Concordant/
-discordant pair counting is done with a vectorized O(n^2/chunk) loop —
-fine for the list-state sizes metrics see; a merge-sort O(n log n) kernel is
-a future optimization.
+Parity: torchmetrics ``functional/regression/kendall.py`` (which counts pairs
+with a per-element O(n^2) python loop). Here pair counting uses Knight's
+O(n log n) algorithm: lexsort by (x, y), count y-inversions with a
+divide-and-conquer merge (vectorized ``searchsorted`` per level), and derive
+concordant/tie counts from group sizes.
 """
 from __future__ import annotations
 
 from typing import Optional, Tuple
 
+import numpy as np
 import torch
 from torch import Tensor
 
 from metrics_amd.utilities.checks import _check_same_shape
 
-_CHUNK = 2048
+_BASE = 64
+
+
+def _inversions(a: np.ndarray) -> Tuple[int, np.ndarray]:
+    """Count strict inversions (a[i] > a[j] for i < j); returns (count, sorted)."""
+    n = a.size
+    if n <= _BASE:
+        if n < 2:
+            return 0, a
+        inv = int(np.sum(np.triu(a[:, None] > a[None, :], k=1)))
+        return inv, np.sort(a, kind="stable")
+    mid = n // 2
+    inv_l, left = _inversions(a[:mid])
+    inv_r, right = _inversions(a[mid:])
+    # cross inversions: for each element of the right half, how many left
+    # elements are strictly greater
+    cross = left.size * right.size - int(np.searchsorted(left, right, side="right").sum())
+    return inv_l + inv_r + cross, np.sort(np.concatenate([left, right]), kind="stable")
+
+
+def _tie_pairs(sorted_keys: np.ndarray) -> int:
+    """Sum of t*(t-1)/2 over runs of equal values in a sorted array."""
+    if sorted_keys.size < 2:
+        return 0
+    change = np.nonzero(sorted_keys[1:] != sorted_keys[:-1])[0] + 1
+    idx = np.concatenate([[0], change, [sorted_keys.size]])
+    t = np.diff(idx).astype(np.int64)
+    return int((t * (t - 1) // 2).sum())
+
+
+def _tie_pairs2(x: np.ndarray, y: np.ndarray) -> int:
+    """Pairs tied in BOTH keys; x must be the primary sort key, y secondary."""
+    if x.size < 2:
+        return 0
+    change = np.nonzero((x[1:] != x[:-1]) | (y[1:] != y[:-1]))[0] + 1
+    idx = np.concatenate([[0], change, [x.size]])
+    t = np.diff(idx).astype(np.int64)
+    return int((t * (t - 1) // 2).sum())
 
 
 def _count_pairs(x: Tensor, y: Tensor) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
-    """Return (concordant, discordant, ties_x_only, ties_y_only) pair counts."""
-    n = x.numel()
-    con = torch.zeros((), dtype=torch.float64, device=x.device)
-    dis = torch.zeros((), dtype=torch.float64, device=x.device)
-    tie_x = torch.zeros((), dtype=torch.float64, device=x.device)
-    tie_y = torch.zeros((), dtype=torch.float64, device=x.device)
-    for i0 in range(0, n, _CHUNK):
-        xi = x[i0 : i0 + _CHUNK].unsqueeze(1)
-        yi = y[i0 : i0 + _CHUNK].unsqueeze(1)
-        # only pairs (i, j) with j > i
-        xj = x.unsqueeze(0)
-        yj = y.unsqueeze(0)
-        mask = torch.arange(n, device=x.device).unsqueeze(0) > (
-            torch.arange(i0, min(i0 + _CHUNK, n), device=x.device).unsqueeze(1)
-        )
-        sx = torch.sign(xj - xi)
-        sy = torch.sign(yj - yi)
-        prod = sx * sy
-        con += ((prod > 0) & mask).sum()
-        dis += ((prod < 0) & mask).sum()
-        tie_x += ((sx == 0) & (sy != 0) & mask).sum()
-        tie_y += ((sy == 0) & (sx != 0) & mask).sum()
-    return con, dis, tie_x, tie_y
+    """Return (concordant, discordant, ties_x, ties_y) pair counts (Knight)."""
+    xn = x.detach().cpu().numpy().astype(np.float64)
+    yn = y.detach().cpu().numpy().astype(np.float64)
+    n = xn.size
+    order = np.lexsort((yn, xn))  # sort by x, ties broken by y ascending
+    xs, ys = xn[order], yn[order]
+    tot = n * (n - 1) // 2
+    xtie = _tie_pairs(xs)
+    ytie = _tie_pairs(np.sort(yn, kind="stable"))
+    xytie = _tie_pairs2(xs, ys)
+    dis, _ = _inversions(ys)
+    both_untied = tot - xtie - ytie + xytie
+    con = both_untied - dis
+
+    dev = x.device
+    as_t = lambda v: torch.tensor(float(v), dtype=torch.float64, device=dev)
+    return as_t(con), as_t(dis), as_t(xtie), as_t(ytie)
 
 
 def _kendall_corrcoef_compute(
@@ -59,13 +92,7 @@ def _kendall_corrcoef_compute(
         if variant == "a":
             tau = (con - dis) / n0
         elif variant == "b":
-            # total ties (pairs tied in x, in y — incl. both)
-            sx = x.unsqueeze(0) - x.unsqueeze(1)
-            sy = y.unsqueeze(0) - y.unsqueeze(1)
-            iu = torch.triu_indices(n, n, offset=1, device=x.device)
-            tx = (sx[iu[0], iu[1]] == 0).sum().double()
-            ty = (sy[iu[0], iu[1]] == 0).sum().double()
-            tau = (con - dis) / torch.sqrt((n0 - tx) * (n0 - ty))
+            tau = (con - dis) / torch.sqrt((n0 - tie_x) * (n0 - tie_y))
         elif variant == "c":
             m = min(len(torch.unique(x)), len(torch.unique(y)))
             tau = 2 * (con - dis) / (n**2 * (m - 1) / m)
