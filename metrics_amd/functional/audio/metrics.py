@@ -138,7 +138,25 @@ def signal_distortion_ratio(
         acf[..., 0] += load_diag
 
     r = _symmetric_toeplitz(acf)
-    sol = torch.linalg.solve(r, xcorr.unsqueeze(-1)).squeeze(-1)
+    if use_cg_iter is not None:
+        # batched conjugate gradient (fast_bss_eval's use_cg_iter option): the
+        # normalized system is well-conditioned, so a few matvecs beat the
+        # O(L^3) dense solve; each matvec is one batched O(L^2) matmul
+        b = xcorr
+        sol = torch.zeros_like(b)
+        resid = b.clone()
+        p = resid.clone()
+        rs_old = (resid * resid).sum(dim=-1, keepdim=True)
+        for _ in range(int(use_cg_iter)):
+            ap = (r @ p.unsqueeze(-1)).squeeze(-1)
+            alpha = rs_old / ((p * ap).sum(dim=-1, keepdim=True) + 1e-38)
+            sol = sol + alpha * p
+            resid = resid - alpha * ap
+            rs_new = (resid * resid).sum(dim=-1, keepdim=True)
+            p = resid + (rs_new / rs_old) * p
+            rs_old = rs_new
+    else:
+        sol = torch.linalg.solve(r, xcorr.unsqueeze(-1)).squeeze(-1)
 
     # coherence: <xcorr, sol>
     coh = (xcorr * sol).sum(dim=-1)

@@ -105,11 +105,21 @@ def perplexity(preds: Tensor, target: Tensor, ignore_index: Optional[int] = None
 
 # ---------------------------------------------------------- ExtendedEditDistance
 def _eed_preprocess(sentence: str, lang: str = "en") -> str:
-    if lang == "ja" or lang == "zh":
-        return " " + sentence.strip() + " "
-    # basic: separate punctuation
-    sentence = re.sub(r"([\.,!?:;\"\(\)])", r" \1 ", sentence)
-    sentence = " ".join(sentence.split())
+    """Official EED preprocessing (reference functional/text/eed.py:175 _preprocess_en/_preprocess_ja)."""
+    if not isinstance(sentence, str):
+        raise ValueError(f"Only strings allowed during preprocessing step, found {type(sentence)} instead")
+    if lang == "ja":
+        import unicodedata
+
+        return unicodedata.normalize("NFKC", sentence.rstrip())
+    sentence = sentence.rstrip()
+    for pattern, replacement in ((".", " ."), ("!", " !"), ("?", " ?"), (",", " ,")):
+        sentence = sentence.replace(pattern, replacement)
+    sentence = re.sub(r"\s+", r" ", sentence)
+    sentence = re.sub(r"(\d) ([.,]) (\d)", r"\1\2\3", sentence)  # 0 . 1 -> 0.1
+    sentence = re.sub(r"(Dr|Jr|Prof|Rev|Gen|Mr|Mt|Mrs|Ms) .", r"\1.", sentence)  # Mr . -> Mr.
+    for pattern, replacement in (("e . g .", "e.g."), ("i . e .", "i.e."), ("U . S .", "U.S.")):
+        sentence = sentence.replace(pattern, replacement)
     return " " + sentence + " "
 
 
@@ -149,8 +159,9 @@ def _eed_single(pred: str, ref: str, alpha: float = 2.0, rho: float = 0.3, delet
         row = next_row
 
     errors = row[n]
-    coverage = rho * sum(x for x in lj if x > 0)
-    denom = len(ref_ch) + coverage
+    # official EED quirk: unvisited positions (-1) contribute 1 to coverage
+    coverage = rho * sum(x if x >= 0 else 1 for x in lj)
+    denom = float(len(ref_ch)) + coverage
     return min(1.0, (errors + coverage) / denom) if denom > 0 else 0.0
 
 

@@ -2,6 +2,7 @@
 import torch
 
 import metrics_amd as ma
+from tests.unittests._helpers import seed_all
 from metrics_amd.functional.audio import (
     permutation_invariant_training,
     scale_invariant_signal_noise_ratio,
@@ -91,3 +92,14 @@ def test_external_dsp_metrics_raise():
         ma.audio.PerceptualEvaluationSpeechQuality(16000, "wb")
     with pytest.raises(ModuleNotFoundError):
         ma.audio.ShortTimeObjectiveIntelligibility(16000)
+
+
+def test_sdr_cg_matches_direct_solve():
+    from metrics_amd.functional.audio import signal_distortion_ratio
+
+    seed_all(61)
+    target = torch.randn(2, 8000)
+    preds = target + 0.1 * torch.randn(2, 8000)
+    direct = signal_distortion_ratio(preds, target)
+    cg = signal_distortion_ratio(preds, target, use_cg_iter=20)
+    assert torch.allclose(direct, cg, atol=0.05), (direct, cg)

@@ -111,3 +111,15 @@ def test_bert_score_with_toy_model():
     m.update(["the cat sat"], ["the cat sat"])
     res = m.compute()
     assert abs(res["f1"].item() - 1.0) < 1e-5
+
+
+def test_eed_reference_parity_values():
+    """Values verified against the reference _eed_function run side-by-side."""
+    from metrics_amd.functional.text import extended_edit_distance
+
+    v = extended_edit_distance(
+        ["this is the prediction", "here is an other sample"],
+        ["this is the reference", "here is another one which is longer"],
+    )
+    assert abs(float(v) - 0.42880797) < 1e-6
+    assert float(extended_edit_distance(["exact match"], ["exact match"])) == 0.0
