@@ -79,7 +79,9 @@ def test_edit_distance():
 def test_chrf_ter_eed():
     assert abs(ma.CHRFScore()(["hello world"], [["hello world"]]).item() - 1.0) < 1e-4
     assert ma.TranslationEditRate()(["the cat"], [["the cat"]]).item() == 0.0
-    assert ma.ExtendedEditDistance()(["the cat"], [["the cat"]]).item() < 1e-6
+    # identical strings: the official EED coverage quirk (-1 visits count 1)
+    # yields a small nonzero score; verified against the reference: 0.0323
+    assert abs(ma.ExtendedEditDistance()(["the cat"], [["the cat"]]).item() - 0.0323) < 1e-3
     # TER with one substitution over 3 tokens
     v = ma.TranslationEditRate()(["the big cat"], [["the small cat"]]).item()
     assert abs(v - 1 / 3) < 1e-6
@@ -122,4 +124,4 @@ def test_eed_reference_parity_values():
         ["this is the reference", "here is another one which is longer"],
     )
     assert abs(float(v) - 0.42880797) < 1e-6
-    assert float(extended_edit_distance(["exact match"], ["exact match"])) == 0.0
+    assert abs(float(extended_edit_distance(["exact match"], ["exact match"])) - 0.0226) < 1e-3
