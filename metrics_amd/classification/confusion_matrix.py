@@ -115,10 +115,15 @@ class MulticlassConfusionMatrix(Metric):
                 p2 = preds.reshape(preds.shape[0], preds.shape[1], -1).movedim(1, -1).reshape(-1, preds.shape[1])
             else:
                 p2 = preds.reshape(-1)
-            _, _, _, _, confmat = ops.multiclass_stat_scores_fused(
-                p2, target.reshape(-1), self.num_classes, self.ignore_index, want_confmat=True
-            )
-            self.confmat += confmat
+            from metrics_amd.ops import _hip
+
+            dummy = getattr(self, "_hip_dummy", None)
+            if dummy is None or dummy.device != preds.device:
+                # write-only sink for the per-class counters the fused kernel
+                # also produces; never read, never zeroed
+                dummy = torch.zeros(3 * self.num_classes + 1, dtype=torch.long, device=preds.device)
+                self._hip_dummy = dummy
+            _hip.mc_confmat_into(p2, target.reshape(-1), self.num_classes, self.ignore_index, self.confmat, dummy)
             return
         preds, target = _multiclass_confusion_matrix_format(preds, target, self.ignore_index)
         confmat = _multiclass_confusion_matrix_update(preds, target, self.num_classes)

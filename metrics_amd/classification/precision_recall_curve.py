@@ -73,14 +73,18 @@ class BinaryPrecisionRecallCurve(Metric):
         """Accumulate either (preds, target) lists or the (T,2,2) confmat."""
         if self.validate_args:
             _binary_precision_recall_curve_tensor_validation(preds, target, self.ignore_index)
-        remove_ignored = not (preds.is_cuda and self.thresholds is not None)
+        gpu_fast = preds.is_cuda and self.thresholds is not None and preds.dtype in (torch.float32, torch.bfloat16)
+        remove_ignored = not gpu_fast
         preds_f, target_f, _ = _binary_precision_recall_curve_format(
-            preds, target, self.thresholds, self.ignore_index, remove_ignored=remove_ignored
+            preds, target, self.thresholds, self.ignore_index, remove_ignored=remove_ignored,
+            normalize=not gpu_fast,  # kernel applies sigmoid-iff-logits in-flight
         )
-        if preds_f.is_cuda and self.thresholds is not None and preds_f.dtype in (torch.float32, torch.bfloat16):
+        if gpu_fast:
             from metrics_amd.ops import _hip
 
-            _hip.curve_hist_into_confmat(preds_f, target_f, self.thresholds, self.ignore_index, self.confmat, mode=0)
+            _hip.curve_hist_into_confmat(
+                preds_f, target_f, self.thresholds, self.ignore_index, self.confmat, mode=0, norm="sigmoid"
+            )
             return
         state = _binary_precision_recall_curve_update(
             preds_f, target_f, self.thresholds, self.ignore_index if not remove_ignored else None
@@ -157,22 +161,24 @@ class MulticlassPrecisionRecallCurve(Metric):
         """Accumulate either (preds, target) lists or the (T,C,2,2) confmat."""
         if self.validate_args:
             _multiclass_precision_recall_curve_tensor_validation(preds, target, self.num_classes, self.ignore_index)
-        remove_ignored = not (preds.is_cuda and self.thresholds is not None and self.average != "micro")
+        gpu_fast = (
+            preds.is_cuda
+            and self.thresholds is not None
+            and self.average != "micro"
+            and preds.dtype in (torch.float32, torch.bfloat16)
+        )
+        remove_ignored = not gpu_fast
         preds_f, target_f, _ = _multiclass_precision_recall_curve_format(
             preds, target, self.num_classes, self.thresholds, self.ignore_index, self.average,
             remove_ignored=remove_ignored,
+            normalize=not gpu_fast,  # kernel applies softmax-iff-logits in-flight
         )
-        if (
-            preds_f.is_cuda
-            and self.thresholds is not None
-            and self.average != "micro"
-            and preds_f.dtype in (torch.float32, torch.bfloat16)
-        ):
+        if gpu_fast:
             from metrics_amd.ops import _hip
 
             _hip.curve_hist_into_confmat(
                 preds_f, target_f, self.thresholds, self.ignore_index if not remove_ignored else None,
-                self.confmat, mode=0,
+                self.confmat, mode=0, norm="softmax",
             )
             return
         state = _multiclass_precision_recall_curve_update(
@@ -242,16 +248,18 @@ class MultilabelPrecisionRecallCurve(Metric):
         """Accumulate either (preds, target) lists or the (T,L,2,2) confmat."""
         if self.validate_args:
             _multilabel_precision_recall_curve_tensor_validation(preds, target, self.num_labels, self.ignore_index)
+        gpu_fast = preds.is_cuda and self.thresholds is not None and preds.dtype in (torch.float32, torch.bfloat16)
         remove_ignored = not preds.is_cuda
         preds_f, target_f, _ = _multilabel_precision_recall_curve_format(
-            preds, target, self.num_labels, self.thresholds, self.ignore_index, remove_ignored=remove_ignored
+            preds, target, self.num_labels, self.thresholds, self.ignore_index, remove_ignored=remove_ignored,
+            normalize=not gpu_fast,  # kernel applies sigmoid-iff-logits in-flight
         )
-        if preds_f.is_cuda and self.thresholds is not None and preds_f.dtype in (torch.float32, torch.bfloat16):
+        if gpu_fast:
             from metrics_amd.ops import _hip
 
             _hip.curve_hist_into_confmat(
                 preds_f, target_f, self.thresholds, self.ignore_index if not remove_ignored else None,
-                self.confmat, mode=1,
+                self.confmat, mode=1, norm="sigmoid",
             )
             return
         state = _multilabel_precision_recall_curve_update(

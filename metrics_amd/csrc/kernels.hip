@@ -309,17 +309,70 @@ __device__ __forceinline__ int bucket_of_uniform(float p, const float* __restric
     return j;
 }
 
+// detect values outside [0,1] (=> inputs are logits, normalize in-kernel).
+// Epoch protocol: the flag buffer is never reset; a value of `epoch` means
+// "this call saw out-of-range values" (epochs increase monotonically per call).
+template <typename T_, bool IS_BF16>
+__global__ void __launch_bounds__(256) k_range_flag(
+    const T_* __restrict__ x, ll N, int epoch, int* __restrict__ flag) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    ll stride = (ll)gridDim.x * blockDim.x;
+    unsigned int outside = 0;
+    for (; i < N; i += stride) {
+        float p = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(x)[i])
+                          : (float)x[i];
+        outside |= (p < 0.0f || p > 1.0f) ? 1u : 0u;
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) outside |= __shfl_down(outside, off);
+    if ((threadIdx.x & (WAVE - 1)) == 0 && outside) atomicMax(flag, epoch);
+}
+
+// per-row max + 1/sum(exp(x - max)) for in-kernel softmax (online, one pass),
+// plus the outside-[0,1] flag. Wave per row, lanes stride classes (coalesced).
+template <typename T_, bool IS_BF16>
+__global__ void __launch_bounds__(256) k_mc_rowstats(
+    const T_* __restrict__ probs, ll B, ll C, int epoch, int* __restrict__ flag,
+    float* __restrict__ rowmax, float* __restrict__ rowinv) {
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wave = threadIdx.x / WAVE;
+    const int waves_per_block = blockDim.x / WAVE;
+    ll row = (ll)blockIdx.x * waves_per_block + wave;
+    const ll row_stride = (ll)gridDim.x * waves_per_block;
+    unsigned int outside = 0;
+    for (; row < B; row += row_stride) {
+        const T_* prow = probs + row * C;
+        float m = -3.4e38f, s = 0.0f;
+        for (ll c = lane; c < C; c += WAVE) {
+            float x = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(prow)[c])
+                              : (float)prow[c];
+            outside |= (x < 0.0f || x > 1.0f) ? 1u : 0u;
+            if (x > m) { s = s * expf(m - x) + 1.0f; m = x; }
+            else { s += expf(x - m); }
+        }
+        for (int off = WAVE / 2; off > 0; off >>= 1) {
+            float om = __shfl_down(m, off);
+            float os = __shfl_down(s, off);
+            if (om > m) { s = s * expf(m - om) + os; m = om; }
+            else { s += os * expf(om - m); }
+        }
+        if (lane == 0) { rowmax[row] = m; rowinv[row] = 1.0f / s; }
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) outside |= __shfl_down(outside, off);
+    if (lane == 0 && outside) atomicMax(flag, epoch);
+}
+
 template <typename T_, bool IS_BF16>
 __global__ void __launch_bounds__(256) k_binary_curve_hist(
     const T_* __restrict__ preds, const ll* __restrict__ target, ll N,
     const float* __restrict__ thresholds, int T, ll ignore_index, int has_ignore,
-    int uniform, float t0, float inv_step,
-    unsigned long long* __restrict__ hist /* (T+1,2) */) {
+    int uniform, float t0, float inv_step, int norm_sigmoid, const int* __restrict__ flag,
+    int epoch, unsigned long long* __restrict__ hist /* (T+1,2) */) {
     extern __shared__ unsigned int lhist[];  // (T+1)*2
     float* sthr = (float*)&lhist[(T + 1) * 2];
     for (int b = threadIdx.x; b < (T + 1) * 2; b += blockDim.x) lhist[b] = 0;
     for (int b = threadIdx.x; b < T; b += blockDim.x) sthr[b] = thresholds[b];
     __syncthreads();
+    const bool do_sigmoid = norm_sigmoid && flag && *flag == epoch;
     ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
     ll stride = (ll)gridDim.x * blockDim.x;
     for (; i < N; i += stride) {
@@ -327,6 +380,7 @@ __global__ void __launch_bounds__(256) k_binary_curve_hist(
         if (has_ignore && t == ignore_index) continue;
         float p = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(preds)[i])
                           : (float)preds[i];
+        if (do_sigmoid) p = 1.0f / (1.0f + expf(-p));
         int j = uniform ? bucket_of_uniform(p, sthr, T, t0, inv_step) : bucket_of(p, sthr, T);
         atomicAdd(&lhist[j * 2 + (t == 1 ? 1 : 0)], 1u);
     }
@@ -355,6 +409,8 @@ __global__ void __launch_bounds__(256) k_multiclass_curve_hist(
     const T_* __restrict__ probs, const ll* __restrict__ target, ll B, ll C,
     const float* __restrict__ thresholds, int T, ll ignore_index, int has_ignore, int mode,
     int uniform, float t0, float inv_step, int c_chunk,
+    int norm_kind /*0 none, 1 softmax (rowstats), 2 sigmoid*/, const int* __restrict__ flag,
+    int epoch, const float* __restrict__ rowmax, const float* __restrict__ rowinv,
     unsigned long long* __restrict__ hist /* (C, T+1, 2) */) {
     extern __shared__ float sthr2[];
     for (int b = threadIdx.x; b < T; b += blockDim.x) sthr2[b] = thresholds[b];
@@ -370,6 +426,9 @@ __global__ void __launch_bounds__(256) k_multiclass_curve_hist(
         trow = target[row];
         if (has_ignore && trow == ignore_index) valid = false;
     }
+    const int norm = (norm_kind && flag && *flag == epoch) ? norm_kind : 0;
+    const float rmax = (norm == 1 && valid) ? rowmax[row] : 0.0f;
+    const float rinv = (norm == 1 && valid) ? rowinv[row] : 0.0f;
     const T_* prow = probs + (valid ? row * C : 0);
     const ll* tgt_row = target + (valid ? row * C : 0);
     for (ll c = c_lo; c < c_hi; c++) {
@@ -386,6 +445,8 @@ __global__ void __launch_bounds__(256) k_multiclass_curve_hist(
             if (label >= 0) {
                 float p = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(prow)[c])
                                   : (float)prow[c];
+                if (norm == 1) p = expf(p - rmax) * rinv;
+                else if (norm == 2) p = 1.0f / (1.0f + expf(-p));
                 int j = uniform ? bucket_of_uniform(p, sthr2, T, t0, inv_step)
                                 : bucket_of(p, sthr2, T);
                 key = j * 2 + label;
@@ -416,15 +477,19 @@ __global__ void __launch_bounds__(256) k_multiclass_curve_hist(
 // multiclass/multilabel curves) so the accumulation happens in-place with no
 // permute+copy of the 6.4MB state.
 __global__ void k_curve_suffix(
-    const unsigned long long* __restrict__ hist /* (O, T+1, 2) */, int T, ll outer, int transposed,
-    ll* __restrict__ confmat /* (O, T, 2, 2) or (T, O, 2, 2) */) {
+    unsigned long long* __restrict__ hist /* (O, T+1, 2) */, int T, ll outer, int transposed,
+    int zero_hist, ll* __restrict__ confmat /* (O, T, 2, 2) or (T, O, 2, 2) */) {
     extern __shared__ unsigned long long sh[];  // (T+1) * 2
     const ll o = blockIdx.x;
-    const unsigned long long* h = hist + o * (ll)(T + 1) * 2;
+    unsigned long long* h = hist + o * (ll)(T + 1) * 2;
     ll* cm = confmat + (transposed ? o * 4 : o * (ll)T * 4);
     const ll tstride = transposed ? outer * 4 : 4;
     for (int j = threadIdx.x; j < (T + 1) * 2; j += blockDim.x) sh[j] = h[j];
     __syncthreads();
+    // the histogram scratch is consumed here: zero it in-flight so the next
+    // update skips a separate fill kernel (buffer is reused, stream-ordered)
+    if (zero_hist)
+        for (int j = threadIdx.x; j < (T + 1) * 2; j += blockDim.x) h[j] = 0;
     __shared__ unsigned long long pos_total, neg_total;
     if (threadIdx.x == 0) {
         unsigned long long pt = 0, nt = 0;
@@ -449,15 +514,21 @@ __global__ void k_curve_suffix(
 // by k_mc_stat_*, add the deltas into the four metric state tensors
 // (tn += valid - tp - fp - fn). ONE launch replaces ~8 small torch kernels.
 __global__ void k_apply_stat_deltas(
-    const unsigned long long* __restrict__ scratch /* 3*C + 1 */, ll C,
+    unsigned long long* __restrict__ scratch /* 3*C + 2 (two valid slots) */, ll C, int epoch,
     ll* __restrict__ tp, ll* __restrict__ fp, ll* __restrict__ tn, ll* __restrict__ fn) {
     ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
     ll stride = (ll)gridDim.x * blockDim.x;
-    const unsigned long long valid = scratch[3 * C];
+    // ping-pong valid slots: read this epoch's slot, zero the OTHER (consumed
+    // last epoch, to be reused next) — avoids a cross-block read/zero race
+    const unsigned long long valid = scratch[3 * C + (epoch & 1)];
+    if (i == 0) scratch[3 * C + ((epoch + 1) & 1)] = 0;
     for (; i < C; i += stride) {
         const ll dtp = (ll)scratch[i];
         const ll dfp = (ll)scratch[C + i];
         const ll dfn = (ll)scratch[2 * C + i];
+        scratch[i] = 0;  // same thread read it: safe in-flight zeroing
+        scratch[C + i] = 0;
+        scratch[2 * C + i] = 0;
         tp[i] += dtp;
         fp[i] += dfp;
         fn[i] += dfn;
@@ -662,27 +733,58 @@ int ma_multilabel_stat(uintptr_t stream, uintptr_t preds, int dtype, uintptr_t t
 
 int ma_binary_curve_hist(uintptr_t stream, uintptr_t preds, int dtype, uintptr_t target, ll N,
                          uintptr_t thresholds, int T, ll ignore_index, int has_ignore,
-                         int uniform, float t0, float inv_step, uintptr_t hist) {
+                         int uniform, float t0, float inv_step, int norm_sigmoid, uintptr_t flag,
+                         int epoch, uintptr_t hist) {
     hipStream_t s = (hipStream_t)stream;
     size_t shmem = (size_t)(T + 1) * 2 * sizeof(unsigned int) + (size_t)T * sizeof(float);
     if (shmem > 160 * 1024) return -100;  // thresholds too large for LDS path
+    if (norm_sigmoid && flag) {
+        if (dtype == 0)
+            k_range_flag<float, false><<<grid_for(N, 256), 256, 0, s>>>(
+                (const float*)preds, N, epoch, (int*)flag);
+        else
+            k_range_flag<unsigned short, true><<<grid_for(N, 256), 256, 0, s>>>(
+                (const unsigned short*)preds, N, epoch, (int*)flag);
+    }
     if (dtype == 0)
         k_binary_curve_hist<float, false><<<grid_for(N, 256), 256, shmem, s>>>(
             (const float*)preds, (const ll*)target, N, (const float*)thresholds, T, ignore_index,
-            has_ignore, uniform, t0, inv_step, (unsigned long long*)hist);
+            has_ignore, uniform, t0, inv_step, norm_sigmoid, (const int*)flag, epoch,
+            (unsigned long long*)hist);
     else
         k_binary_curve_hist<unsigned short, true><<<grid_for(N, 256), 256, shmem, s>>>(
             (const unsigned short*)preds, (const ll*)target, N, (const float*)thresholds, T,
-            ignore_index, has_ignore, uniform, t0, inv_step, (unsigned long long*)hist);
+            ignore_index, has_ignore, uniform, t0, inv_step, norm_sigmoid, (const int*)flag, epoch,
+            (unsigned long long*)hist);
     return (int)hipGetLastError();
 }
 
 int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintptr_t target, ll B,
                              ll C, uintptr_t thresholds, int T, ll ignore_index, int has_ignore,
-                             int mode, int uniform, float t0, float inv_step, uintptr_t hist) {
+                             int mode, int uniform, float t0, float inv_step, int norm_kind,
+                             uintptr_t flag, int epoch, uintptr_t rowmax, uintptr_t rowinv,
+                             uintptr_t hist) {
     hipStream_t s = (hipStream_t)stream;
     size_t shmem = (size_t)T * sizeof(float);
     if (shmem > 160 * 1024) return -100;
+    if (norm_kind == 1 && flag) {
+        // softmax stats (one fused pass also sets the out-of-range flag)
+        int grid = grid_for(B, 4);
+        if (dtype == 0)
+            k_mc_rowstats<float, false><<<grid, 256, 0, s>>>(
+                (const float*)probs, B, C, epoch, (int*)flag, (float*)rowmax, (float*)rowinv);
+        else
+            k_mc_rowstats<unsigned short, true><<<grid, 256, 0, s>>>(
+                (const unsigned short*)probs, B, C, epoch, (int*)flag, (float*)rowmax,
+                (float*)rowinv);
+    } else if (norm_kind == 2 && flag) {
+        if (dtype == 0)
+            k_range_flag<float, false><<<grid_for(B * C, 256), 256, 0, s>>>(
+                (const float*)probs, B * C, epoch, (int*)flag);
+        else
+            k_range_flag<unsigned short, true><<<grid_for(B * C, 256), 256, 0, s>>>(
+                (const unsigned short*)probs, B * C, epoch, (int*)flag);
+    }
     ll row_chunks = (B + 255) / 256;
     // pick the class chunk so the grid comfortably overfills 256 CUs
     int c_chunk = 32;
@@ -693,30 +795,32 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
     if (dtype == 0)
         k_multiclass_curve_hist<float, false><<<grid, 256, shmem, s>>>(
             (const float*)probs, (const ll*)target, B, C, (const float*)thresholds, T, ignore_index,
-            has_ignore, mode, uniform, t0, inv_step, c_chunk, (unsigned long long*)hist);
+            has_ignore, mode, uniform, t0, inv_step, c_chunk, norm_kind, (const int*)flag, epoch,
+            (const float*)rowmax, (const float*)rowinv, (unsigned long long*)hist);
     else
         k_multiclass_curve_hist<unsigned short, true><<<grid, 256, shmem, s>>>(
             (const unsigned short*)probs, (const ll*)target, B, C, (const float*)thresholds, T,
-            ignore_index, has_ignore, mode, uniform, t0, inv_step, c_chunk,
+            ignore_index, has_ignore, mode, uniform, t0, inv_step, c_chunk, norm_kind,
+            (const int*)flag, epoch, (const float*)rowmax, (const float*)rowinv,
             (unsigned long long*)hist);
     return (int)hipGetLastError();
 }
 
 int ma_curve_suffix(uintptr_t stream, uintptr_t hist, ll outer, int T, int transposed,
-                    uintptr_t confmat) {
+                    int zero_hist, uintptr_t confmat) {
     hipStream_t s = (hipStream_t)stream;
     size_t shmem = (size_t)(T + 1) * 2 * sizeof(unsigned long long);
     if (shmem > 160 * 1024) return -100;
-    k_curve_suffix<<<(int)outer, 256, shmem, s>>>((const unsigned long long*)hist, T, outer,
-                                                  transposed, (ll*)confmat);
+    k_curve_suffix<<<(int)outer, 256, shmem, s>>>((unsigned long long*)hist, T, outer,
+                                                  transposed, zero_hist, (ll*)confmat);
     return (int)hipGetLastError();
 }
 
-int ma_apply_stat_deltas(uintptr_t stream, uintptr_t scratch, ll C, uintptr_t tp, uintptr_t fp,
-                         uintptr_t tn, uintptr_t fn) {
+int ma_apply_stat_deltas(uintptr_t stream, uintptr_t scratch, ll C, int epoch, uintptr_t tp,
+                         uintptr_t fp, uintptr_t tn, uintptr_t fn) {
     hipStream_t s = (hipStream_t)stream;
     k_apply_stat_deltas<<<grid_for(C, 256), 256, 0, s>>>(
-        (const unsigned long long*)scratch, C, (ll*)tp, (ll*)fp, (ll*)tn, (ll*)fn);
+        (unsigned long long*)scratch, C, epoch, (ll*)tp, (ll*)fp, (ll*)tn, (ll*)fn);
     return (int)hipGetLastError();
 }
 

@@ -127,6 +127,7 @@ def _binary_precision_recall_curve_format(
     thresholds: Optional[Union[int, List[float], Tensor]] = None,
     ignore_index: Optional[int] = None,
     remove_ignored: bool = True,
+    normalize: bool = True,
 ) -> Tuple[Tensor, Tensor, Optional[Tensor]]:
     preds = preds.flatten()
     target = target.flatten()
@@ -135,7 +136,8 @@ def _binary_precision_recall_curve_format(
         preds = preds[idx]
         target = target[idx]
 
-    preds = normalize_logits_if_needed(preds, "sigmoid")
+    if normalize:
+        preds = normalize_logits_if_needed(preds, "sigmoid")
     thresholds = _adjust_threshold_arg(thresholds, preds.device)
     return preds, target, thresholds
 
@@ -249,6 +251,7 @@ def _multiclass_precision_recall_curve_format(
     ignore_index: Optional[int] = None,
     average: Optional[str] = None,
     remove_ignored: bool = True,
+    normalize: bool = True,
 ) -> Tuple[Tensor, Tensor, Optional[Tensor]]:
     if preds.ndim == 2:
         pass  # already (N, C); avoid the transpose->reshape round-trip copy
@@ -261,7 +264,8 @@ def _multiclass_precision_recall_curve_format(
         preds = preds[idx]
         target = target[idx]
 
-    preds = normalize_logits_if_needed(preds, "softmax")
+    if normalize:
+        preds = normalize_logits_if_needed(preds, "softmax")
 
     if average == "micro":
         preds = preds.flatten()
@@ -391,6 +395,7 @@ def _multilabel_precision_recall_curve_format(
     thresholds: Optional[Union[int, List[float], Tensor]] = None,
     ignore_index: Optional[int] = None,
     remove_ignored: bool = True,
+    normalize: bool = True,
 ) -> Tuple[Tensor, Tensor, Optional[Tensor]]:
     if preds.ndim == 2:
         pass  # already (N, L)
@@ -398,7 +403,8 @@ def _multilabel_precision_recall_curve_format(
         preds = preds.transpose(0, 1).reshape(num_labels, -1).T
         target = target.transpose(0, 1).reshape(num_labels, -1).T
 
-    preds = normalize_logits_if_needed(preds, "sigmoid")
+    if normalize:
+        preds = normalize_logits_if_needed(preds, "sigmoid")
 
     thresholds = _adjust_threshold_arg(thresholds, preds.device)
     if ignore_index is not None and thresholds is not None and remove_ignored:

@@ -220,6 +220,40 @@ def multilabel_stat(
 
 _UNIFORM_CACHE: dict = {}
 
+# Per-device out-of-range flag buffer + monotonically increasing epoch. The
+# flag is NEVER reset: kernels write the current epoch into it when they see
+# values outside [0,1], and consumers compare against that exact epoch — so no
+# per-update fill is needed. Stream-safety: all launches go to the caller's
+# current stream; buffers are shared per device, which is safe because each
+# (detect -> consume) pair is issued back-to-back on one stream.
+_FLAG_BUFS: dict = {}
+_EPOCHS: dict = {}
+_HIST_POOL: dict = {}
+
+
+def _flag_and_epoch(device) -> Tuple[Tensor, int]:
+    key = device.index
+    buf = _FLAG_BUFS.get(key)
+    if buf is None:
+        buf = torch.zeros(1, dtype=torch.int32, device=device)
+        _FLAG_BUFS[key] = buf
+        _EPOCHS[key] = 0
+    _EPOCHS[key] += 1
+    return buf, _EPOCHS[key]
+
+
+def _pooled_hist(outer: int, T: int, device) -> Tensor:
+    """Persistent histogram scratch, zeroed once; the suffix kernel re-zeroes
+    it in-flight after consuming it (zero_hist=1), so reuse needs no fill."""
+    key = (outer, T, device.index)
+    buf = _HIST_POOL.get(key)
+    if buf is None:
+        buf = torch.zeros(outer, T + 1, 2, dtype=torch.long, device=device)
+        if len(_HIST_POOL) > 64:
+            _HIST_POOL.clear()
+        _HIST_POOL[key] = buf
+    return buf
+
 
 def _uniform_params(thr: Tensor):
     """Detect uniform threshold spacing for the O(1) bucket guess.
@@ -280,6 +314,9 @@ def binary_curve_confmat(
         ctypes.c_int(uni),
         ctypes.c_float(t0),
         ctypes.c_float(inv_step),
+        ctypes.c_int(0),
+        ctypes.c_uint64(0),
+        ctypes.c_int(0),
         ctypes.c_uint64(hist.data_ptr()),
     )
     _check(rc, "ma_binary_curve_hist")
@@ -289,6 +326,7 @@ def binary_curve_confmat(
         ctypes.c_uint64(hist.data_ptr()),
         ctypes.c_longlong(1),
         ctypes.c_int(T),
+        ctypes.c_int(0),
         ctypes.c_int(0),
         ctypes.c_uint64(confmat.data_ptr()),
     )
@@ -325,6 +363,11 @@ def multiclass_curve_confmat(
         ctypes.c_int(uni),
         ctypes.c_float(t0),
         ctypes.c_float(inv_step),
+        ctypes.c_int(0),
+        ctypes.c_uint64(0),
+        ctypes.c_int(0),
+        ctypes.c_uint64(0),
+        ctypes.c_uint64(0),
         ctypes.c_uint64(hist.data_ptr()),
     )
     _check(rc, "ma_multiclass_curve_hist")
@@ -336,6 +379,7 @@ def multiclass_curve_confmat(
         ctypes.c_longlong(C),
         ctypes.c_int(T),
         ctypes.c_int(1),
+        ctypes.c_int(0),
         ctypes.c_uint64(confmat.data_ptr()),
     )
     _check(rc, "ma_curve_suffix")
@@ -398,11 +442,20 @@ def box_iou(boxes1: Tensor, boxes2: Tensor, variant: str = "iou") -> Tensor:
 
 def curve_hist_into_confmat(
     probs: Tensor, target: Tensor, thresholds: Tensor, ignore_index: Optional[int],
-    confmat_state: Tensor, mode: int,
+    confmat_state: Tensor, mode: int, norm: Optional[str] = None,
 ) -> None:
     """Bucketized histogram + transposed suffix-sum accumulated DIRECTLY into the
     metric's confmat state ((T,2,2) binary / (T,C,2,2) multiclass|multilabel) —
-    no intermediate confmat materialization or host-side add."""
+    no intermediate confmat materialization or host-side add.
+
+    ``norm`` ("softmax" / "sigmoid" / None) moves the reference's
+    normalize-iff-outside-[0,1] semantics INTO the kernels: a fused row-stats
+    (or range-detect) pass sets a per-device epoch flag and the histogram
+    kernel normalizes inline — replacing the 4-kernel torch chain
+    (compare / any-reduce / softmax / where) and its 3 extra full-tensor
+    round-trips through HBM. The histogram scratch is pooled and re-zeroed by
+    the suffix kernel in-flight, so steady-state updates launch no fills.
+    """
     lib = _lib()
     thr = thresholds.contiguous().float()
     T = thr.numel()
@@ -411,7 +464,12 @@ def curve_hist_into_confmat(
     if confmat_state.ndim == 3:  # binary (T,2,2)
         preds = probs.contiguous().flatten()
         tgt = target.contiguous().long().flatten()
-        hist = torch.zeros(T + 1, 2, dtype=torch.long, device=dev)
+        hist = _pooled_hist(1, T, dev)
+        if norm == "sigmoid":
+            flag, epoch = _flag_and_epoch(dev)
+            norm_i, flag_ptr = 1, flag.data_ptr()
+        else:
+            norm_i, flag_ptr, epoch = 0, 0, 0
         rc = lib.ma_binary_curve_hist(
             ctypes.c_uint64(_stream()),
             ctypes.c_uint64(preds.data_ptr()),
@@ -425,6 +483,9 @@ def curve_hist_into_confmat(
             ctypes.c_int(uni),
             ctypes.c_float(t0),
             ctypes.c_float(inv_step),
+            ctypes.c_int(norm_i),
+            ctypes.c_uint64(flag_ptr),
+            ctypes.c_int(epoch),
             ctypes.c_uint64(hist.data_ptr()),
         )
         _check(rc, "ma_binary_curve_hist")
@@ -433,7 +494,18 @@ def curve_hist_into_confmat(
         probs = probs.contiguous()
         tgt = target.contiguous().long()
         B, C = probs.shape
-        hist = torch.zeros(C, T + 1, 2, dtype=torch.long, device=dev)
+        hist = _pooled_hist(C, T, dev)
+        if norm == "softmax":
+            flag, epoch = _flag_and_epoch(dev)
+            rowmax = torch.empty(B, dtype=torch.float32, device=dev)
+            rowinv = torch.empty(B, dtype=torch.float32, device=dev)
+            norm_i, flag_ptr = 1, flag.data_ptr()
+            rm_ptr, ri_ptr = rowmax.data_ptr(), rowinv.data_ptr()
+        elif norm == "sigmoid":
+            flag, epoch = _flag_and_epoch(dev)
+            norm_i, flag_ptr, rm_ptr, ri_ptr = 2, flag.data_ptr(), 0, 0
+        else:
+            norm_i, flag_ptr, epoch, rm_ptr, ri_ptr = 0, 0, 0, 0, 0
         rc = lib.ma_multiclass_curve_hist(
             ctypes.c_uint64(_stream()),
             ctypes.c_uint64(probs.data_ptr()),
@@ -449,6 +521,11 @@ def curve_hist_into_confmat(
             ctypes.c_int(uni),
             ctypes.c_float(t0),
             ctypes.c_float(inv_step),
+            ctypes.c_int(norm_i),
+            ctypes.c_uint64(flag_ptr),
+            ctypes.c_int(epoch),
+            ctypes.c_uint64(rm_ptr),
+            ctypes.c_uint64(ri_ptr),
             ctypes.c_uint64(hist.data_ptr()),
         )
         _check(rc, "ma_multiclass_curve_hist")
@@ -460,9 +537,13 @@ def curve_hist_into_confmat(
         ctypes.c_longlong(outer),
         ctypes.c_int(T),
         ctypes.c_int(transposed),
+        ctypes.c_int(1),  # re-zero the pooled hist in-flight
         ctypes.c_uint64(confmat_state.data_ptr()),
     )
     _check(rc, "ma_curve_suffix")
+
+
+_SCRATCH_EPOCHS: dict = {}
 
 
 def mc_stat_into(
@@ -471,16 +552,23 @@ def mc_stat_into(
 ) -> None:
     """Fused stat-scores update accumulated DIRECTLY into the metric states.
 
-    ``scratch`` is a per-metric reusable (3*C+1,) int64 buffer; 3 launches
-    total (zero + count + apply) instead of ~12 small torch kernels.
+    ``scratch`` is a per-metric reusable (3*C+2,) int64 buffer (tp|fp|fn counts
+    plus TWO ping-pong valid slots); 2 launches total (count + apply): the
+    apply kernel zeroes the scratch in-flight for the next update, so no fill
+    kernel runs in steady state.
     """
     lib = _lib()
     C = num_classes
-    scratch.zero_()
+    key = scratch.data_ptr()
+    epoch = _SCRATCH_EPOCHS.get(key, 0)
+    _SCRATCH_EPOCHS[key] = epoch + 1
+    if len(_SCRATCH_EPOCHS) > 4096:
+        _SCRATCH_EPOCHS.clear()
+        _SCRATCH_EPOCHS[key] = epoch + 1
     s_tp = scratch[:C]
     s_fp = scratch[C : 2 * C]
     s_fn = scratch[2 * C : 3 * C]
-    s_valid = scratch[3 * C :]
+    s_valid_ptr = scratch.data_ptr() + (3 * C + (epoch & 1)) * 8
     if preds.ndim == 2 and preds.is_floating_point():
         preds = preds.contiguous()
         target = target.contiguous().long()
@@ -498,7 +586,7 @@ def mc_stat_into(
             ctypes.c_uint64(s_fp.data_ptr()),
             ctypes.c_uint64(s_fn.data_ptr()),
             ctypes.c_uint64(0),
-            ctypes.c_uint64(s_valid.data_ptr()),
+            ctypes.c_uint64(s_valid_ptr),
             ctypes.c_uint64(0),
         )
         _check(rc, "ma_mc_stat_logits")
@@ -517,16 +605,72 @@ def mc_stat_into(
             ctypes.c_uint64(s_fp.data_ptr()),
             ctypes.c_uint64(s_fn.data_ptr()),
             ctypes.c_uint64(0),
-            ctypes.c_uint64(s_valid.data_ptr()),
+            ctypes.c_uint64(s_valid_ptr),
         )
         _check(rc, "ma_mc_stat_labels")
     rc = lib.ma_apply_stat_deltas(
         ctypes.c_uint64(_stream()),
         ctypes.c_uint64(scratch.data_ptr()),
         ctypes.c_longlong(C),
+        ctypes.c_int(epoch),
         ctypes.c_uint64(tp.data_ptr()),
         ctypes.c_uint64(fp.data_ptr()),
         ctypes.c_uint64(tn.data_ptr()),
         ctypes.c_uint64(fn.data_ptr()),
     )
     _check(rc, "ma_apply_stat_deltas")
+
+
+def mc_confmat_into(
+    preds: Tensor, target: Tensor, num_classes: int, ignore_index: Optional[int],
+    confmat_state: Tensor, dummy: Tensor,
+) -> None:
+    """Confusion-matrix update atomically accumulated DIRECTLY into the (C,C)
+    state — no per-batch confmat materialization, fill, or host-side add.
+
+    ``dummy`` is a per-metric (3*C+1,) int64 sink for the per-class counters
+    the kernel also produces (never read, never zeroed: garbage accumulates
+    harmlessly in int64).
+    """
+    lib = _lib()
+    C = num_classes
+    assert confmat_state.is_contiguous()
+    if preds.ndim == 2 and preds.is_floating_point():
+        preds = preds.contiguous()
+        target = target.contiguous().long()
+        B, C2 = preds.shape
+        rc = lib.ma_mc_stat_logits(
+            ctypes.c_uint64(_stream()),
+            ctypes.c_uint64(preds.data_ptr()),
+            ctypes.c_int(_dtype_code(preds)),
+            ctypes.c_uint64(target.data_ptr()),
+            ctypes.c_longlong(B),
+            ctypes.c_longlong(C2),
+            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+            ctypes.c_int(1 if ignore_index is not None else 0),
+            ctypes.c_uint64(dummy[:C].data_ptr()),
+            ctypes.c_uint64(dummy[C : 2 * C].data_ptr()),
+            ctypes.c_uint64(dummy[2 * C : 3 * C].data_ptr()),
+            ctypes.c_uint64(confmat_state.data_ptr()),
+            ctypes.c_uint64(dummy[3 * C :].data_ptr()),
+            ctypes.c_uint64(0),
+        )
+        _check(rc, "ma_mc_stat_logits")
+    else:
+        p2 = preds.contiguous().long().flatten()
+        t2 = target.contiguous().long().flatten()
+        rc = lib.ma_mc_stat_labels(
+            ctypes.c_uint64(_stream()),
+            ctypes.c_uint64(p2.data_ptr()),
+            ctypes.c_uint64(t2.data_ptr()),
+            ctypes.c_longlong(p2.numel()),
+            ctypes.c_longlong(C),
+            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+            ctypes.c_int(1 if ignore_index is not None else 0),
+            ctypes.c_uint64(dummy[:C].data_ptr()),
+            ctypes.c_uint64(dummy[C : 2 * C].data_ptr()),
+            ctypes.c_uint64(dummy[2 * C : 3 * C].data_ptr()),
+            ctypes.c_uint64(confmat_state.data_ptr()),
+            ctypes.c_uint64(dummy[3 * C :].data_ptr()),
+        )
+        _check(rc, "ma_mc_stat_labels")

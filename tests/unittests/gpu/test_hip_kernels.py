@@ -281,3 +281,93 @@ def test_bincount_deterministic_repeat():
     a = ops.hip_bincount(x, 777)
     b = ops.hip_bincount(x, 777)
     assert torch.equal(a, b)
+
+
+def test_curve_in_kernel_softmax_logits():
+    """Raw logits to curve metrics: the kernel softmaxes in-flight; values must
+    match the CPU torch path to float tolerance (sum-order differs by design),
+    and be deterministic across runs."""
+    torch.manual_seed(12)
+    logits = torch.randn(4096, 50, device="cuda") * 3
+    tgt = torch.randint(0, 50, (4096,), device="cuda")
+    a1 = ma.MulticlassAUROC(num_classes=50, thresholds=100).to("cuda")
+    a2 = ma.MulticlassAUROC(num_classes=50, thresholds=100).to("cuda")
+    a1.update(logits, tgt)
+    a2.update(logits, tgt)
+    assert torch.equal(a1.confmat, a2.confmat)  # deterministic
+    cpu = ma.MulticlassAUROC(num_classes=50, thresholds=100)
+    cpu.update(logits.cpu(), tgt.cpu())
+    # confmat counts can differ by ulp-at-threshold cases; the metric must agree
+    assert torch.allclose(a1.compute().cpu(), cpu.compute(), atol=1e-3)
+    diff = (a1.confmat.cpu() - cpu.confmat).abs().sum().item()
+    assert diff <= 8, f"confmat count drift too large: {diff}"
+    # pre-normalized probabilities keep exact equality (flag never set)
+    probs = logits.softmax(-1)
+    g = ma.MulticlassAUROC(num_classes=50, thresholds=100).to("cuda")
+    c = ma.MulticlassAUROC(num_classes=50, thresholds=100)
+    g.update(probs, tgt)
+    c.update(probs.cpu(), tgt.cpu())
+    assert torch.equal(g.confmat.cpu(), c.confmat)
+
+
+def test_curve_in_kernel_sigmoid_logits():
+    torch.manual_seed(13)
+    logits = torch.randn(100_000, device="cuda") * 4
+    tgt = torch.randint(0, 2, (100_000,), device="cuda")
+    g = ma.BinaryROC(thresholds=64).to("cuda")
+    c = ma.BinaryROC(thresholds=64)
+    g.update(logits, tgt)
+    c.update(logits.cpu(), tgt.cpu())
+    diff = (g.confmat.cpu() - c.confmat).abs().sum().item()
+    assert diff <= 4, diff
+    # multilabel sigmoid
+    ml_logits = torch.randn(8192, 12, device="cuda") * 4
+    ml_tgt = torch.randint(0, 2, (8192, 12), device="cuda")
+    gm = ma.MultilabelAveragePrecision(num_labels=12, thresholds=50).to("cuda")
+    cm = ma.MultilabelAveragePrecision(num_labels=12, thresholds=50)
+    gm.update(ml_logits, ml_tgt)
+    cm.update(ml_logits.cpu(), ml_tgt.cpu())
+    assert torch.allclose(gm.compute().cpu(), cm.compute(), atol=1e-3)
+
+
+def test_pooled_hist_interleaved_metrics():
+    """Two metrics with the same (C,T) share the pooled hist scratch; the
+    suffix kernel re-zeroes it in-flight, so interleaved updates must not leak."""
+    torch.manual_seed(14)
+    p1 = torch.rand(2048, 20, device="cuda")
+    p1 = p1 / p1.sum(-1, keepdim=True)
+    t1 = torch.randint(0, 20, (2048,), device="cuda")
+    p2 = torch.rand(2048, 20, device="cuda")
+    p2 = p2 / p2.sum(-1, keepdim=True)
+    t2 = torch.randint(0, 20, (2048,), device="cuda")
+    m1 = ma.MulticlassPrecisionRecallCurve(num_classes=20, thresholds=30).to("cuda")
+    m2 = ma.MulticlassPrecisionRecallCurve(num_classes=20, thresholds=30).to("cuda")
+    for _ in range(3):
+        m1.update(p1, t1)
+        m2.update(p2, t2)
+    r1 = ma.MulticlassPrecisionRecallCurve(num_classes=20, thresholds=30)
+    r2 = ma.MulticlassPrecisionRecallCurve(num_classes=20, thresholds=30)
+    for _ in range(3):
+        r1.update(p1.cpu(), t1.cpu())
+        r2.update(p2.cpu(), t2.cpu())
+    assert torch.equal(m1.confmat.cpu(), r1.confmat)
+    assert torch.equal(m2.confmat.cpu(), r2.confmat)
+
+
+def test_confmat_into_state_repeated():
+    torch.manual_seed(15)
+    m_gpu = ma.MulticlassConfusionMatrix(num_classes=33).to("cuda")
+    m_cpu = ma.MulticlassConfusionMatrix(num_classes=33)
+    for _ in range(5):
+        preds = torch.randn(1024, 33)
+        tgt = torch.randint(0, 33, (1024,))
+        m_gpu.update(preds.cuda(), tgt.cuda())
+        m_cpu.update(preds, tgt)
+    assert torch.equal(m_gpu.confmat.cpu(), m_cpu.confmat)
+    m_gpu.reset()
+    m_cpu.reset()
+    preds = torch.randn(512, 33)
+    tgt = torch.randint(0, 33, (512,))
+    m_gpu.update(preds.cuda(), tgt.cuda())
+    m_cpu.update(preds, tgt)
+    assert torch.equal(m_gpu.confmat.cpu(), m_cpu.confmat)
