@@ -323,8 +323,15 @@ __global__ void __launch_bounds__(256) k_range_flag(
                           : (float)x[i];
         outside |= (p < 0.0f || p > 1.0f) ? 1u : 0u;
     }
+    __shared__ unsigned int blk_outside;
+    if (threadIdx.x == 0) blk_outside = 0;
+    __syncthreads();
     for (int off = WAVE / 2; off > 0; off >>= 1) outside |= __shfl_down(outside, off);
-    if ((threadIdx.x & (WAVE - 1)) == 0 && outside) atomicMax(flag, epoch);
+    if ((threadIdx.x & (WAVE - 1)) == 0 && outside) atomicOr(&blk_outside, 1u);
+    __syncthreads();
+    // poll before the global atomic: once any block set the epoch, the rest
+    // skip — a per-wave atomicMax on one address serializes ~100ns each
+    if (threadIdx.x == 0 && blk_outside && *flag != epoch) atomicMax(flag, epoch);
 }
 
 // per-row max + 1/sum(exp(x - max)) for in-kernel softmax (online, one pass),
@@ -389,8 +396,13 @@ __global__ void __launch_bounds__(256) k_mc_rowstats(
         }
         if (lane == 0) { rowmax[row] = m; rowinv[row] = 1.0f / s; }
     }
+    __shared__ unsigned int blk_outside2;
+    if (threadIdx.x == 0) blk_outside2 = 0;
+    __syncthreads();
     for (int off = WAVE / 2; off > 0; off >>= 1) outside |= __shfl_down(outside, off);
-    if (lane == 0 && outside) atomicMax(flag, epoch);
+    if (lane == 0 && outside) atomicOr(&blk_outside2, 1u);
+    __syncthreads();
+    if (threadIdx.x == 0 && blk_outside2 && *flag != epoch) atomicMax(flag, epoch);
 }
 
 template <typename T_, bool IS_BF16>
