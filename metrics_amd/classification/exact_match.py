@@ -68,6 +68,22 @@ class MulticlassExactMatch(Metric):
             _multiclass_stat_scores_tensor_validation(
                 preds, target, self.num_classes, self.multidim_average, self.ignore_index
             )
+        if (
+            preds.is_cuda
+            and self.multidim_average == "global"
+            and target.ndim == 1
+            and (not preds.is_floating_point() or (preds.ndim == 2 and preds.dtype in (torch.float32, torch.bfloat16)))
+        ):
+            # 1-D-sample exact match == argmax-equals-target: reuse the fused
+            # stat kernel, correct = sum(tp), total = valid (2 launches)
+            from metrics_amd.ops import _hip
+
+            scratch = getattr(self, "_hip_scratch", None)
+            if scratch is None or scratch.device != preds.device:
+                scratch = torch.zeros(3 * self.num_classes + 2, dtype=torch.long, device=preds.device)
+                self._hip_scratch = scratch
+            _hip.mc_exact_into(preds, target, self.num_classes, self.ignore_index, scratch, self.correct, self.total)
+            return
         preds, target = _multiclass_stat_scores_format(preds, target, 1)
         correct, total = _multiclass_exact_match_update(preds, target, self.multidim_average, self.ignore_index)
         if self.multidim_average == "samplewise":

@@ -580,6 +580,31 @@ __global__ void k_apply_stat_deltas(
     }
 }
 
+// exact-match epilogue: correct += sum(tp), total += valid — one block,
+// zeroes the scratch in-flight (same ping-pong valid protocol as apply_deltas).
+__global__ void k_exact_apply(unsigned long long* __restrict__ scratch, ll C, int epoch,
+                              ll* __restrict__ correct, ll* __restrict__ total) {
+    __shared__ unsigned long long part[256];
+    unsigned long long acc = 0;
+    for (ll i = threadIdx.x; i < C; i += blockDim.x) {
+        acc += scratch[i];
+        scratch[i] = 0;
+        scratch[C + i] = 0;
+        scratch[2 * C + i] = 0;
+    }
+    part[threadIdx.x] = acc;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if (threadIdx.x < off) part[threadIdx.x] += part[threadIdx.x + off];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        correct[0] += (ll)part[0];
+        total[0] += (ll)scratch[3 * C + (epoch & 1)];
+        scratch[3 * C + ((epoch + 1) & 1)] = 0;
+    }
+}
+
 // ---------------------------------------------------------------------------
 // K13: fused elementwise-error reductions, deterministic fp64 two-pass.
 // op: 0 = squared error, 1 = abs error, 2 = abs percentage |d|/max(|t|,eps),
@@ -865,6 +890,14 @@ int ma_apply_stat_deltas(uintptr_t stream, uintptr_t scratch, ll C, int epoch, u
     hipStream_t s = (hipStream_t)stream;
     k_apply_stat_deltas<<<grid_for(C, 256), 256, 0, s>>>(
         (unsigned long long*)scratch, C, epoch, (ll*)tp, (ll*)fp, (ll*)tn, (ll*)fn);
+    return (int)hipGetLastError();
+}
+
+int ma_exact_apply(uintptr_t stream, uintptr_t scratch, ll C, int epoch, uintptr_t correct,
+                   uintptr_t total) {
+    hipStream_t s = (hipStream_t)stream;
+    k_exact_apply<<<1, 256, 0, s>>>((unsigned long long*)scratch, C, epoch, (ll*)correct,
+                                    (ll*)total);
     return (int)hipGetLastError();
 }
 

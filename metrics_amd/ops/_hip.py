@@ -674,3 +674,69 @@ def mc_confmat_into(
             ctypes.c_uint64(dummy[3 * C :].data_ptr()),
         )
         _check(rc, "ma_mc_stat_labels")
+
+
+def mc_exact_into(
+    preds: Tensor, target: Tensor, num_classes: int, ignore_index: Optional[int],
+    scratch: Tensor, correct: Tensor, total: Tensor,
+) -> None:
+    """Exact-match (== row argmax equals target) accumulated into correct/total.
+
+    Reuses the fused stat kernel's per-class tp counts: correct = sum(tp),
+    total = valid rows. Same (3*C+2,) self-zeroing scratch protocol as
+    :func:`mc_stat_into`; 2 launches, zero fills in steady state.
+    """
+    lib = _lib()
+    C = num_classes
+    key = scratch.data_ptr()
+    epoch = _SCRATCH_EPOCHS.get(key, 0)
+    _SCRATCH_EPOCHS[key] = epoch + 1
+    s_valid_ptr = scratch.data_ptr() + (3 * C + (epoch & 1)) * 8
+    if preds.ndim == 2 and preds.is_floating_point():
+        preds = preds.contiguous()
+        target = target.contiguous().long()
+        B, C2 = preds.shape
+        rc = lib.ma_mc_stat_logits(
+            ctypes.c_uint64(_stream()),
+            ctypes.c_uint64(preds.data_ptr()),
+            ctypes.c_int(_dtype_code(preds)),
+            ctypes.c_uint64(target.data_ptr()),
+            ctypes.c_longlong(B),
+            ctypes.c_longlong(C2),
+            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+            ctypes.c_int(1 if ignore_index is not None else 0),
+            ctypes.c_uint64(scratch[:C].data_ptr()),
+            ctypes.c_uint64(scratch[C : 2 * C].data_ptr()),
+            ctypes.c_uint64(scratch[2 * C : 3 * C].data_ptr()),
+            ctypes.c_uint64(0),
+            ctypes.c_uint64(s_valid_ptr),
+            ctypes.c_uint64(0),
+        )
+        _check(rc, "ma_mc_stat_logits")
+    else:
+        p2 = preds.contiguous().long().flatten()
+        t2 = target.contiguous().long().flatten()
+        rc = lib.ma_mc_stat_labels(
+            ctypes.c_uint64(_stream()),
+            ctypes.c_uint64(p2.data_ptr()),
+            ctypes.c_uint64(t2.data_ptr()),
+            ctypes.c_longlong(p2.numel()),
+            ctypes.c_longlong(C),
+            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
+            ctypes.c_int(1 if ignore_index is not None else 0),
+            ctypes.c_uint64(scratch[:C].data_ptr()),
+            ctypes.c_uint64(scratch[C : 2 * C].data_ptr()),
+            ctypes.c_uint64(scratch[2 * C : 3 * C].data_ptr()),
+            ctypes.c_uint64(0),
+            ctypes.c_uint64(s_valid_ptr),
+        )
+        _check(rc, "ma_mc_stat_labels")
+    rc = lib.ma_exact_apply(
+        ctypes.c_uint64(_stream()),
+        ctypes.c_uint64(scratch.data_ptr()),
+        ctypes.c_longlong(C),
+        ctypes.c_int(epoch),
+        ctypes.c_uint64(correct.data_ptr()),
+        ctypes.c_uint64(total.data_ptr()),
+    )
+    _check(rc, "ma_exact_apply")

@@ -371,3 +371,26 @@ def test_confmat_into_state_repeated():
     m_gpu.update(preds.cuda(), tgt.cuda())
     m_cpu.update(preds, tgt)
     assert torch.equal(m_gpu.confmat.cpu(), m_cpu.confmat)
+
+
+def test_exact_match_fused_gpu():
+    torch.manual_seed(16)
+    m_gpu = ma.MulticlassExactMatch(num_classes=40).to("cuda")
+    m_cpu = ma.MulticlassExactMatch(num_classes=40)
+    for _ in range(4):
+        preds = torch.randn(1024, 40)
+        tgt = torch.randint(0, 40, (1024,))
+        m_gpu.update(preds.cuda(), tgt.cuda())
+        m_cpu.update(preds, tgt)
+    assert torch.equal(m_gpu.correct.cpu(), m_cpu.correct)
+    assert torch.equal(m_gpu.total.cpu(), m_cpu.total)
+    # with ignore_index
+    gi = ma.MulticlassExactMatch(num_classes=40, ignore_index=-1).to("cuda")
+    ci = ma.MulticlassExactMatch(num_classes=40, ignore_index=-1)
+    preds = torch.randn(2048, 40)
+    tgt = torch.randint(0, 40, (2048,))
+    tgt[::5] = -1
+    gi.update(preds.cuda(), tgt.cuda())
+    ci.update(preds, tgt)
+    assert torch.equal(gi.correct.cpu(), ci.correct) and torch.equal(gi.total.cpu(), ci.total)
+    assert torch.allclose(gi.compute().cpu(), ci.compute())
