@@ -582,7 +582,7 @@ __global__ void k_apply_stat_deltas(
 
 // exact-match epilogue: correct += sum(tp), total += valid — one block,
 // zeroes the scratch in-flight (same ping-pong valid protocol as apply_deltas).
-__global__ void k_exact_apply(unsigned long long* __restrict__ scratch, ll C, int epoch,
+__global__ void k_exact_apply(unsigned long long* __restrict__ scratch, ll C, ll B, int epoch,
                               ll* __restrict__ correct, ll* __restrict__ total) {
     __shared__ unsigned long long part[256];
     unsigned long long acc = 0;
@@ -599,8 +599,11 @@ __global__ void k_exact_apply(unsigned long long* __restrict__ scratch, ll C, in
         __syncthreads();
     }
     if (threadIdx.x == 0) {
-        correct[0] += (ll)part[0];
-        total[0] += (ll)scratch[3 * C + (epoch & 1)];
+        // reference semantics (_multiclass_exact_match_update): ignored
+        // samples are forced equal, so they count as correct AND in total
+        const ll valid = (ll)scratch[3 * C + (epoch & 1)];
+        correct[0] += (ll)part[0] + (B - valid);
+        total[0] += B;
         scratch[3 * C + ((epoch + 1) & 1)] = 0;
     }
 }
@@ -893,10 +896,10 @@ int ma_apply_stat_deltas(uintptr_t stream, uintptr_t scratch, ll C, int epoch, u
     return (int)hipGetLastError();
 }
 
-int ma_exact_apply(uintptr_t stream, uintptr_t scratch, ll C, int epoch, uintptr_t correct,
+int ma_exact_apply(uintptr_t stream, uintptr_t scratch, ll C, ll B, int epoch, uintptr_t correct,
                    uintptr_t total) {
     hipStream_t s = (hipStream_t)stream;
-    k_exact_apply<<<1, 256, 0, s>>>((unsigned long long*)scratch, C, epoch, (ll*)correct,
+    k_exact_apply<<<1, 256, 0, s>>>((unsigned long long*)scratch, C, B, epoch, (ll*)correct,
                                     (ll*)total);
     return (int)hipGetLastError();
 }

@@ -735,6 +735,7 @@ def mc_exact_into(
         ctypes.c_uint64(_stream()),
         ctypes.c_uint64(scratch.data_ptr()),
         ctypes.c_longlong(C),
+        ctypes.c_longlong(target.numel()),
         ctypes.c_int(epoch),
         ctypes.c_uint64(correct.data_ptr()),
         ctypes.c_uint64(total.data_ptr()),
