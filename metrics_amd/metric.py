@@ -522,9 +522,25 @@ class Metric(Module, ABC):
         self.update = self._wrap_update(self.update)  # type: ignore[method-assign]
         self.compute = self._wrap_compute(self.compute)  # type: ignore[method-assign]
 
+    # hot internal scalars set on every update/compute: nn.Module.__setattr__
+    # walks parameter/buffer/module dicts per call (~1us each, dozens per
+    # collection update) — bypass it for names that can never be any of those
+    _FAST_ATTRS = frozenset(
+        {"_computed", "_update_count", "_forward_cache", "_to_sync", "_should_unsync", "_enable_grad", "_is_synced"}
+    )
+
     def __setattr__(self, name: str, value: Any) -> None:
+        if name in Metric._FAST_ATTRS:
+            object.__setattr__(self, name, value)
+            return
         if name in ("higher_is_better", "is_differentiable", "full_state_update", "plot_lower_bound", "plot_upper_bound", "plot_legend_name"):
             raise RuntimeError(f"Can't change const `{name}`.")
+        # metric STATES are plain tensor/list attributes (never
+        # parameters/buffers/modules): skip the nn.Module bookkeeping walk
+        defaults = self.__dict__.get("_defaults")
+        if defaults is not None and name in defaults and isinstance(value, (Tensor, list)):
+            object.__setattr__(self, name, value)
+            return
         super().__setattr__(name, value)
 
     def type(self, dst_type: Union[str, torch.dtype]) -> "Metric":  # noqa: A003

@@ -34,7 +34,37 @@ def _load() -> Optional[ctypes.CDLL]:
     except OSError as err:
         _LOAD_ERR = f"Failed to load {p}: {err}"
         return None
+    _declare_argtypes(_LIB)
     return _LIB
+
+
+_U64 = ctypes.c_uint64
+_LL = ctypes.c_longlong
+_I = ctypes.c_int
+_F = ctypes.c_float
+_D = ctypes.c_double
+
+_SIGNATURES = {
+    "ma_mc_stat_logits": [_U64, _U64, _I, _U64, _LL, _LL, _LL, _I, _U64, _U64, _U64, _U64, _U64, _U64],
+    "ma_mc_stat_labels": [_U64, _U64, _U64, _LL, _LL, _LL, _I, _U64, _U64, _U64, _U64, _U64],
+    "ma_bincount": [_U64, _U64, _LL, _LL, _U64],
+    "ma_binary_stat": [_U64, _U64, _I, _U64, _LL, _F, _LL, _I, _U64, _U64],
+    "ma_multilabel_stat": [_U64, _U64, _I, _U64, _LL, _LL, _F, _LL, _I, _U64, _U64],
+    "ma_binary_curve_hist": [_U64, _U64, _I, _U64, _LL, _U64, _I, _LL, _I, _I, _F, _F, _I, _U64, _I, _U64],
+    "ma_multiclass_curve_hist": [_U64, _U64, _I, _U64, _LL, _LL, _U64, _I, _LL, _I, _I, _I, _F, _F, _I, _U64, _I, _U64, _U64, _U64],
+    "ma_curve_suffix": [_U64, _U64, _LL, _I, _I, _I, _U64],
+    "ma_apply_stat_deltas": [_U64, _U64, _LL, _I, _U64, _U64, _U64, _U64],
+    "ma_exact_apply": [_U64, _U64, _LL, _LL, _I, _U64, _U64],
+    "ma_err_reduce": [_U64, _U64, _U64, _I, _LL, _I, _D, _U64, _I, _I, _U64],
+    "ma_box_iou": [_U64, _U64, _LL, _U64, _LL, _I, _U64],
+}
+
+
+def _declare_argtypes(lib: ctypes.CDLL) -> None:
+    for name, argtypes in _SIGNATURES.items():
+        fn = getattr(lib, name)
+        fn.argtypes = argtypes
+        fn.restype = ctypes.c_int
 
 
 def hip_available() -> bool:
@@ -89,20 +119,20 @@ def mc_stat_logits(
     confmat = torch.zeros(C, C, dtype=torch.long, device=dev) if want_confmat else None
     argmax = torch.empty(B, dtype=torch.long, device=dev) if want_argmax else None
     rc = lib.ma_mc_stat_logits(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(preds.data_ptr()),
-        ctypes.c_int(_dtype_code(preds)),
-        ctypes.c_uint64(target.data_ptr()),
-        ctypes.c_longlong(B),
-        ctypes.c_longlong(C),
-        ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-        ctypes.c_int(1 if ignore_index is not None else 0),
-        ctypes.c_uint64(tp.data_ptr()),
-        ctypes.c_uint64(fp.data_ptr()),
-        ctypes.c_uint64(fn.data_ptr()),
-        ctypes.c_uint64(confmat.data_ptr() if confmat is not None else 0),
-        ctypes.c_uint64(valid.data_ptr()),
-        ctypes.c_uint64(argmax.data_ptr() if argmax is not None else 0),
+        _stream(),
+        preds.data_ptr(),
+        _dtype_code(preds),
+        target.data_ptr(),
+        B,
+        C,
+        ignore_index if ignore_index is not None else 0,
+        1 if ignore_index is not None else 0,
+        tp.data_ptr(),
+        fp.data_ptr(),
+        fn.data_ptr(),
+        confmat.data_ptr() if confmat is not None else 0,
+        valid.data_ptr(),
+        argmax.data_ptr() if argmax is not None else 0,
     )
     _check(rc, "ma_mc_stat_logits")
     return tp, fp, fn, valid, confmat, argmax
@@ -123,18 +153,18 @@ def mc_stat_labels(
     valid = torch.zeros(1, dtype=torch.long, device=dev)
     confmat = torch.zeros(num_classes, num_classes, dtype=torch.long, device=dev) if want_confmat else None
     rc = lib.ma_mc_stat_labels(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(preds.data_ptr()),
-        ctypes.c_uint64(target.data_ptr()),
-        ctypes.c_longlong(N),
-        ctypes.c_longlong(num_classes),
-        ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-        ctypes.c_int(1 if ignore_index is not None else 0),
-        ctypes.c_uint64(tp.data_ptr()),
-        ctypes.c_uint64(fp.data_ptr()),
-        ctypes.c_uint64(fn.data_ptr()),
-        ctypes.c_uint64(confmat.data_ptr() if confmat is not None else 0),
-        ctypes.c_uint64(valid.data_ptr()),
+        _stream(),
+        preds.data_ptr(),
+        target.data_ptr(),
+        N,
+        num_classes,
+        ignore_index if ignore_index is not None else 0,
+        1 if ignore_index is not None else 0,
+        tp.data_ptr(),
+        fp.data_ptr(),
+        fn.data_ptr(),
+        confmat.data_ptr() if confmat is not None else 0,
+        valid.data_ptr(),
     )
     _check(rc, "ma_mc_stat_labels")
     return tp, fp, fn, valid, confmat
@@ -146,11 +176,11 @@ def bincount(x: Tensor, minlength: int) -> Tensor:
     x = x.contiguous().long().flatten()
     out = torch.zeros(minlength, dtype=torch.long, device=x.device)
     rc = lib.ma_bincount(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(x.data_ptr()),
-        ctypes.c_longlong(x.numel()),
-        ctypes.c_longlong(minlength),
-        ctypes.c_uint64(out.data_ptr()),
+        _stream(),
+        x.data_ptr(),
+        x.numel(),
+        minlength,
+        out.data_ptr(),
     )
     _check(rc, "ma_bincount")
     return out
@@ -169,16 +199,16 @@ def binary_stat(
     flag = torch.zeros(1, dtype=torch.int32, device=dev)
     thr = min(max(threshold, 1e-7), 1 - 1e-7)
     rc = lib.ma_binary_stat(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(preds.data_ptr()),
-        ctypes.c_int(_dtype_code(preds)),
-        ctypes.c_uint64(target.data_ptr()),
-        ctypes.c_longlong(N),
-        ctypes.c_float(thr),
-        ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-        ctypes.c_int(1 if ignore_index is not None else 0),
-        ctypes.c_uint64(out.data_ptr()),
-        ctypes.c_uint64(flag.data_ptr()),
+        _stream(),
+        preds.data_ptr(),
+        _dtype_code(preds),
+        target.data_ptr(),
+        N,
+        thr,
+        ignore_index if ignore_index is not None else 0,
+        1 if ignore_index is not None else 0,
+        out.data_ptr(),
+        flag.data_ptr(),
     )
     _check(rc, "ma_binary_stat")
     sel = flag.long().squeeze(0)  # 0 -> raw counts, 1 -> sigmoid counts (no host sync)
@@ -200,17 +230,17 @@ def multilabel_stat(
     flag = torch.zeros(1, dtype=torch.int32, device=dev)
     thr = min(max(threshold, 1e-7), 1 - 1e-7)
     rc = lib.ma_multilabel_stat(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(preds.data_ptr()),
-        ctypes.c_int(_dtype_code(preds)),
-        ctypes.c_uint64(target.data_ptr()),
-        ctypes.c_longlong(N),
-        ctypes.c_longlong(L),
-        ctypes.c_float(thr),
-        ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-        ctypes.c_int(1 if ignore_index is not None else 0),
-        ctypes.c_uint64(out.data_ptr()),
-        ctypes.c_uint64(flag.data_ptr()),
+        _stream(),
+        preds.data_ptr(),
+        _dtype_code(preds),
+        target.data_ptr(),
+        N,
+        L,
+        thr,
+        ignore_index if ignore_index is not None else 0,
+        1 if ignore_index is not None else 0,
+        out.data_ptr(),
+        flag.data_ptr(),
     )
     _check(rc, "ma_multilabel_stat")
     sel = flag.long().squeeze(0)
@@ -229,6 +259,7 @@ _UNIFORM_CACHE: dict = {}
 _FLAG_BUFS: dict = {}
 _EPOCHS: dict = {}
 _HIST_POOL: dict = {}
+_ROWSTATS_POOL: dict = {}
 
 
 def _flag_and_epoch(device) -> Tuple[Tensor, int]:
@@ -302,33 +333,33 @@ def binary_curve_confmat(
     uni, t0, inv_step = _uniform_params(thr)
     hist = torch.zeros(T + 1, 2, dtype=torch.long, device=dev)
     rc = lib.ma_binary_curve_hist(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(preds.data_ptr()),
-        ctypes.c_int(_dtype_code(preds)),
-        ctypes.c_uint64(target.data_ptr()),
-        ctypes.c_longlong(preds.numel()),
-        ctypes.c_uint64(thr.data_ptr()),
-        ctypes.c_int(T),
-        ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-        ctypes.c_int(1 if ignore_index is not None else 0),
-        ctypes.c_int(uni),
-        ctypes.c_float(t0),
-        ctypes.c_float(inv_step),
-        ctypes.c_int(0),
-        ctypes.c_uint64(0),
-        ctypes.c_int(0),
-        ctypes.c_uint64(hist.data_ptr()),
+        _stream(),
+        preds.data_ptr(),
+        _dtype_code(preds),
+        target.data_ptr(),
+        preds.numel(),
+        thr.data_ptr(),
+        T,
+        ignore_index if ignore_index is not None else 0,
+        1 if ignore_index is not None else 0,
+        uni,
+        t0,
+        inv_step,
+        0,
+        0,
+        0,
+        hist.data_ptr(),
     )
     _check(rc, "ma_binary_curve_hist")
     confmat = torch.zeros(T, 2, 2, dtype=torch.long, device=dev)
     rc = lib.ma_curve_suffix(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(hist.data_ptr()),
-        ctypes.c_longlong(1),
-        ctypes.c_int(T),
-        ctypes.c_int(0),
-        ctypes.c_int(0),
-        ctypes.c_uint64(confmat.data_ptr()),
+        _stream(),
+        hist.data_ptr(),
+        1,
+        T,
+        0,
+        0,
+        confmat.data_ptr(),
     )
     _check(rc, "ma_curve_suffix")
     return confmat
@@ -349,38 +380,38 @@ def multiclass_curve_confmat(
     uni, t0, inv_step = _uniform_params(thr)
     hist = torch.zeros(C, T + 1, 2, dtype=torch.long, device=dev)
     rc = lib.ma_multiclass_curve_hist(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(probs.data_ptr()),
-        ctypes.c_int(_dtype_code(probs)),
-        ctypes.c_uint64(target.data_ptr()),
-        ctypes.c_longlong(B),
-        ctypes.c_longlong(C),
-        ctypes.c_uint64(thr.data_ptr()),
-        ctypes.c_int(T),
-        ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-        ctypes.c_int(1 if ignore_index is not None else 0),
-        ctypes.c_int(mode),
-        ctypes.c_int(uni),
-        ctypes.c_float(t0),
-        ctypes.c_float(inv_step),
-        ctypes.c_int(0),
-        ctypes.c_uint64(0),
-        ctypes.c_int(0),
-        ctypes.c_uint64(0),
-        ctypes.c_uint64(0),
-        ctypes.c_uint64(hist.data_ptr()),
+        _stream(),
+        probs.data_ptr(),
+        _dtype_code(probs),
+        target.data_ptr(),
+        B,
+        C,
+        thr.data_ptr(),
+        T,
+        ignore_index if ignore_index is not None else 0,
+        1 if ignore_index is not None else 0,
+        mode,
+        uni,
+        t0,
+        inv_step,
+        0,
+        0,
+        0,
+        0,
+        0,
+        hist.data_ptr(),
     )
     _check(rc, "ma_multiclass_curve_hist")
     # write the (T, C, 2, 2) state layout directly (transposed suffix kernel)
     confmat = torch.zeros(T, C, 2, 2, dtype=torch.long, device=dev)
     rc = lib.ma_curve_suffix(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(hist.data_ptr()),
-        ctypes.c_longlong(C),
-        ctypes.c_int(T),
-        ctypes.c_int(1),
-        ctypes.c_int(0),
-        ctypes.c_uint64(confmat.data_ptr()),
+        _stream(),
+        hist.data_ptr(),
+        C,
+        T,
+        1,
+        0,
+        confmat.data_ptr(),
     )
     _check(rc, "ma_curve_suffix")
     return confmat
@@ -401,17 +432,17 @@ def err_reduce(x: Tensor, y: Tensor, op: str, eps: float = 1.17e-6) -> Tensor:
     partials = torch.zeros(num_blocks, n_out, dtype=torch.float64, device=dev)
     out = torch.zeros(n_out, dtype=torch.float64, device=dev)
     rc = lib.ma_err_reduce(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(x.data_ptr()),
-        ctypes.c_uint64(y.data_ptr()),
-        ctypes.c_int(_dtype_code(x)),
-        ctypes.c_longlong(N),
-        ctypes.c_int(op_id),
-        ctypes.c_double(eps),
-        ctypes.c_uint64(partials.data_ptr()),
-        ctypes.c_int(num_blocks),
-        ctypes.c_int(n_out),
-        ctypes.c_uint64(out.data_ptr()),
+        _stream(),
+        x.data_ptr(),
+        y.data_ptr(),
+        _dtype_code(x),
+        N,
+        op_id,
+        eps,
+        partials.data_ptr(),
+        num_blocks,
+        n_out,
+        out.data_ptr(),
     )
     _check(rc, "ma_err_reduce")
     return out
@@ -428,13 +459,13 @@ def box_iou(boxes1: Tensor, boxes2: Tensor, variant: str = "iou") -> Tensor:
     if N == 0 or M == 0:
         return out
     rc = lib.ma_box_iou(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(boxes1.data_ptr()),
-        ctypes.c_longlong(N),
-        ctypes.c_uint64(boxes2.data_ptr()),
-        ctypes.c_longlong(M),
-        ctypes.c_int(v),
-        ctypes.c_uint64(out.data_ptr()),
+        _stream(),
+        boxes1.data_ptr(),
+        N,
+        boxes2.data_ptr(),
+        M,
+        v,
+        out.data_ptr(),
     )
     _check(rc, "ma_box_iou")
     return out
@@ -471,22 +502,22 @@ def curve_hist_into_confmat(
         else:
             norm_i, flag_ptr, epoch = 0, 0, 0
         rc = lib.ma_binary_curve_hist(
-            ctypes.c_uint64(_stream()),
-            ctypes.c_uint64(preds.data_ptr()),
-            ctypes.c_int(_dtype_code(preds)),
-            ctypes.c_uint64(tgt.data_ptr()),
-            ctypes.c_longlong(preds.numel()),
-            ctypes.c_uint64(thr.data_ptr()),
-            ctypes.c_int(T),
-            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-            ctypes.c_int(1 if ignore_index is not None else 0),
-            ctypes.c_int(uni),
-            ctypes.c_float(t0),
-            ctypes.c_float(inv_step),
-            ctypes.c_int(norm_i),
-            ctypes.c_uint64(flag_ptr),
-            ctypes.c_int(epoch),
-            ctypes.c_uint64(hist.data_ptr()),
+            _stream(),
+            preds.data_ptr(),
+            _dtype_code(preds),
+            tgt.data_ptr(),
+            preds.numel(),
+            thr.data_ptr(),
+            T,
+            ignore_index if ignore_index is not None else 0,
+            1 if ignore_index is not None else 0,
+            uni,
+            t0,
+            inv_step,
+            norm_i,
+            flag_ptr,
+            epoch,
+            hist.data_ptr(),
         )
         _check(rc, "ma_binary_curve_hist")
         outer, transposed = 1, 0
@@ -497,48 +528,53 @@ def curve_hist_into_confmat(
         hist = _pooled_hist(C, T, dev)
         if norm == "softmax":
             flag, epoch = _flag_and_epoch(dev)
-            rowmax = torch.empty(B, dtype=torch.float32, device=dev)
-            rowinv = torch.empty(B, dtype=torch.float32, device=dev)
+            rkey = (B, dev.index)
+            rbuf = _ROWSTATS_POOL.get(rkey)
+            if rbuf is None:
+                rbuf = torch.empty(2, B, dtype=torch.float32, device=dev)
+                if len(_ROWSTATS_POOL) > 64:
+                    _ROWSTATS_POOL.clear()
+                _ROWSTATS_POOL[rkey] = rbuf
             norm_i, flag_ptr = 1, flag.data_ptr()
-            rm_ptr, ri_ptr = rowmax.data_ptr(), rowinv.data_ptr()
+            rm_ptr, ri_ptr = rbuf[0].data_ptr(), rbuf[1].data_ptr()
         elif norm == "sigmoid":
             flag, epoch = _flag_and_epoch(dev)
             norm_i, flag_ptr, rm_ptr, ri_ptr = 2, flag.data_ptr(), 0, 0
         else:
             norm_i, flag_ptr, epoch, rm_ptr, ri_ptr = 0, 0, 0, 0, 0
         rc = lib.ma_multiclass_curve_hist(
-            ctypes.c_uint64(_stream()),
-            ctypes.c_uint64(probs.data_ptr()),
-            ctypes.c_int(_dtype_code(probs)),
-            ctypes.c_uint64(tgt.data_ptr()),
-            ctypes.c_longlong(B),
-            ctypes.c_longlong(C),
-            ctypes.c_uint64(thr.data_ptr()),
-            ctypes.c_int(T),
-            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-            ctypes.c_int(1 if ignore_index is not None else 0),
-            ctypes.c_int(mode),
-            ctypes.c_int(uni),
-            ctypes.c_float(t0),
-            ctypes.c_float(inv_step),
-            ctypes.c_int(norm_i),
-            ctypes.c_uint64(flag_ptr),
-            ctypes.c_int(epoch),
-            ctypes.c_uint64(rm_ptr),
-            ctypes.c_uint64(ri_ptr),
-            ctypes.c_uint64(hist.data_ptr()),
+            _stream(),
+            probs.data_ptr(),
+            _dtype_code(probs),
+            tgt.data_ptr(),
+            B,
+            C,
+            thr.data_ptr(),
+            T,
+            ignore_index if ignore_index is not None else 0,
+            1 if ignore_index is not None else 0,
+            mode,
+            uni,
+            t0,
+            inv_step,
+            norm_i,
+            flag_ptr,
+            epoch,
+            rm_ptr,
+            ri_ptr,
+            hist.data_ptr(),
         )
         _check(rc, "ma_multiclass_curve_hist")
         outer, transposed = C, 1
     assert confmat_state.is_contiguous()
     rc = lib.ma_curve_suffix(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(hist.data_ptr()),
-        ctypes.c_longlong(outer),
-        ctypes.c_int(T),
-        ctypes.c_int(transposed),
-        ctypes.c_int(1),  # re-zero the pooled hist in-flight
-        ctypes.c_uint64(confmat_state.data_ptr()),
+        _stream(),
+        hist.data_ptr(),
+        outer,
+        T,
+        transposed,
+        1,  # re-zero the pooled hist in-flight
+        confmat_state.data_ptr(),
     )
     _check(rc, "ma_curve_suffix")
 
@@ -574,49 +610,49 @@ def mc_stat_into(
         target = target.contiguous().long()
         B, C2 = preds.shape
         rc = lib.ma_mc_stat_logits(
-            ctypes.c_uint64(_stream()),
-            ctypes.c_uint64(preds.data_ptr()),
-            ctypes.c_int(_dtype_code(preds)),
-            ctypes.c_uint64(target.data_ptr()),
-            ctypes.c_longlong(B),
-            ctypes.c_longlong(C2),
-            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-            ctypes.c_int(1 if ignore_index is not None else 0),
-            ctypes.c_uint64(s_tp.data_ptr()),
-            ctypes.c_uint64(s_fp.data_ptr()),
-            ctypes.c_uint64(s_fn.data_ptr()),
-            ctypes.c_uint64(0),
-            ctypes.c_uint64(s_valid_ptr),
-            ctypes.c_uint64(0),
+            _stream(),
+            preds.data_ptr(),
+            _dtype_code(preds),
+            target.data_ptr(),
+            B,
+            C2,
+            ignore_index if ignore_index is not None else 0,
+            1 if ignore_index is not None else 0,
+            s_tp.data_ptr(),
+            s_fp.data_ptr(),
+            s_fn.data_ptr(),
+            0,
+            s_valid_ptr,
+            0,
         )
         _check(rc, "ma_mc_stat_logits")
     else:
         p2 = preds.contiguous().long().flatten()
         t2 = target.contiguous().long().flatten()
         rc = lib.ma_mc_stat_labels(
-            ctypes.c_uint64(_stream()),
-            ctypes.c_uint64(p2.data_ptr()),
-            ctypes.c_uint64(t2.data_ptr()),
-            ctypes.c_longlong(p2.numel()),
-            ctypes.c_longlong(C),
-            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-            ctypes.c_int(1 if ignore_index is not None else 0),
-            ctypes.c_uint64(s_tp.data_ptr()),
-            ctypes.c_uint64(s_fp.data_ptr()),
-            ctypes.c_uint64(s_fn.data_ptr()),
-            ctypes.c_uint64(0),
-            ctypes.c_uint64(s_valid_ptr),
+            _stream(),
+            p2.data_ptr(),
+            t2.data_ptr(),
+            p2.numel(),
+            C,
+            ignore_index if ignore_index is not None else 0,
+            1 if ignore_index is not None else 0,
+            s_tp.data_ptr(),
+            s_fp.data_ptr(),
+            s_fn.data_ptr(),
+            0,
+            s_valid_ptr,
         )
         _check(rc, "ma_mc_stat_labels")
     rc = lib.ma_apply_stat_deltas(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(scratch.data_ptr()),
-        ctypes.c_longlong(C),
-        ctypes.c_int(epoch),
-        ctypes.c_uint64(tp.data_ptr()),
-        ctypes.c_uint64(fp.data_ptr()),
-        ctypes.c_uint64(tn.data_ptr()),
-        ctypes.c_uint64(fn.data_ptr()),
+        _stream(),
+        scratch.data_ptr(),
+        C,
+        epoch,
+        tp.data_ptr(),
+        fp.data_ptr(),
+        tn.data_ptr(),
+        fn.data_ptr(),
     )
     _check(rc, "ma_apply_stat_deltas")
 
@@ -640,38 +676,38 @@ def mc_confmat_into(
         target = target.contiguous().long()
         B, C2 = preds.shape
         rc = lib.ma_mc_stat_logits(
-            ctypes.c_uint64(_stream()),
-            ctypes.c_uint64(preds.data_ptr()),
-            ctypes.c_int(_dtype_code(preds)),
-            ctypes.c_uint64(target.data_ptr()),
-            ctypes.c_longlong(B),
-            ctypes.c_longlong(C2),
-            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-            ctypes.c_int(1 if ignore_index is not None else 0),
-            ctypes.c_uint64(dummy[:C].data_ptr()),
-            ctypes.c_uint64(dummy[C : 2 * C].data_ptr()),
-            ctypes.c_uint64(dummy[2 * C : 3 * C].data_ptr()),
-            ctypes.c_uint64(confmat_state.data_ptr()),
-            ctypes.c_uint64(dummy[3 * C :].data_ptr()),
-            ctypes.c_uint64(0),
+            _stream(),
+            preds.data_ptr(),
+            _dtype_code(preds),
+            target.data_ptr(),
+            B,
+            C2,
+            ignore_index if ignore_index is not None else 0,
+            1 if ignore_index is not None else 0,
+            dummy[:C].data_ptr(),
+            dummy[C : 2 * C].data_ptr(),
+            dummy[2 * C : 3 * C].data_ptr(),
+            confmat_state.data_ptr(),
+            dummy[3 * C :].data_ptr(),
+            0,
         )
         _check(rc, "ma_mc_stat_logits")
     else:
         p2 = preds.contiguous().long().flatten()
         t2 = target.contiguous().long().flatten()
         rc = lib.ma_mc_stat_labels(
-            ctypes.c_uint64(_stream()),
-            ctypes.c_uint64(p2.data_ptr()),
-            ctypes.c_uint64(t2.data_ptr()),
-            ctypes.c_longlong(p2.numel()),
-            ctypes.c_longlong(C),
-            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-            ctypes.c_int(1 if ignore_index is not None else 0),
-            ctypes.c_uint64(dummy[:C].data_ptr()),
-            ctypes.c_uint64(dummy[C : 2 * C].data_ptr()),
-            ctypes.c_uint64(dummy[2 * C : 3 * C].data_ptr()),
-            ctypes.c_uint64(confmat_state.data_ptr()),
-            ctypes.c_uint64(dummy[3 * C :].data_ptr()),
+            _stream(),
+            p2.data_ptr(),
+            t2.data_ptr(),
+            p2.numel(),
+            C,
+            ignore_index if ignore_index is not None else 0,
+            1 if ignore_index is not None else 0,
+            dummy[:C].data_ptr(),
+            dummy[C : 2 * C].data_ptr(),
+            dummy[2 * C : 3 * C].data_ptr(),
+            confmat_state.data_ptr(),
+            dummy[3 * C :].data_ptr(),
         )
         _check(rc, "ma_mc_stat_labels")
 
@@ -697,47 +733,47 @@ def mc_exact_into(
         target = target.contiguous().long()
         B, C2 = preds.shape
         rc = lib.ma_mc_stat_logits(
-            ctypes.c_uint64(_stream()),
-            ctypes.c_uint64(preds.data_ptr()),
-            ctypes.c_int(_dtype_code(preds)),
-            ctypes.c_uint64(target.data_ptr()),
-            ctypes.c_longlong(B),
-            ctypes.c_longlong(C2),
-            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-            ctypes.c_int(1 if ignore_index is not None else 0),
-            ctypes.c_uint64(scratch[:C].data_ptr()),
-            ctypes.c_uint64(scratch[C : 2 * C].data_ptr()),
-            ctypes.c_uint64(scratch[2 * C : 3 * C].data_ptr()),
-            ctypes.c_uint64(0),
-            ctypes.c_uint64(s_valid_ptr),
-            ctypes.c_uint64(0),
+            _stream(),
+            preds.data_ptr(),
+            _dtype_code(preds),
+            target.data_ptr(),
+            B,
+            C2,
+            ignore_index if ignore_index is not None else 0,
+            1 if ignore_index is not None else 0,
+            scratch[:C].data_ptr(),
+            scratch[C : 2 * C].data_ptr(),
+            scratch[2 * C : 3 * C].data_ptr(),
+            0,
+            s_valid_ptr,
+            0,
         )
         _check(rc, "ma_mc_stat_logits")
     else:
         p2 = preds.contiguous().long().flatten()
         t2 = target.contiguous().long().flatten()
         rc = lib.ma_mc_stat_labels(
-            ctypes.c_uint64(_stream()),
-            ctypes.c_uint64(p2.data_ptr()),
-            ctypes.c_uint64(t2.data_ptr()),
-            ctypes.c_longlong(p2.numel()),
-            ctypes.c_longlong(C),
-            ctypes.c_longlong(ignore_index if ignore_index is not None else 0),
-            ctypes.c_int(1 if ignore_index is not None else 0),
-            ctypes.c_uint64(scratch[:C].data_ptr()),
-            ctypes.c_uint64(scratch[C : 2 * C].data_ptr()),
-            ctypes.c_uint64(scratch[2 * C : 3 * C].data_ptr()),
-            ctypes.c_uint64(0),
-            ctypes.c_uint64(s_valid_ptr),
+            _stream(),
+            p2.data_ptr(),
+            t2.data_ptr(),
+            p2.numel(),
+            C,
+            ignore_index if ignore_index is not None else 0,
+            1 if ignore_index is not None else 0,
+            scratch[:C].data_ptr(),
+            scratch[C : 2 * C].data_ptr(),
+            scratch[2 * C : 3 * C].data_ptr(),
+            0,
+            s_valid_ptr,
         )
         _check(rc, "ma_mc_stat_labels")
     rc = lib.ma_exact_apply(
-        ctypes.c_uint64(_stream()),
-        ctypes.c_uint64(scratch.data_ptr()),
-        ctypes.c_longlong(C),
-        ctypes.c_longlong(target.numel()),
-        ctypes.c_int(epoch),
-        ctypes.c_uint64(correct.data_ptr()),
-        ctypes.c_uint64(total.data_ptr()),
+        _stream(),
+        scratch.data_ptr(),
+        C,
+        target.numel(),
+        epoch,
+        correct.data_ptr(),
+        total.data_ptr(),
     )
     _check(rc, "ma_exact_apply")
