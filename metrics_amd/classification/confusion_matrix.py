@@ -80,6 +80,27 @@ class BinaryConfusionMatrix(Metric):
         return plot_confusion_matrix(val, ax=ax, add_text=add_text, labels=labels, cmap=cmap)
 
 
+def _fused_confmat_update(metric, preds, target) -> bool:
+    """GPU fast path shared by every confmat-state metric (ConfusionMatrix,
+    CohenKappa, MatthewsCorrCoef, JaccardIndex): accumulate the batch counts
+    atomically into ``metric.confmat`` — no per-batch confmat, fill, or add.
+    Returns False when the caller must take the torch path."""
+    if not (preds.is_cuda and (not preds.is_floating_point() or preds.dtype in (torch.float32, torch.bfloat16))):
+        return False
+    if preds.ndim == target.ndim + 1 and preds.is_floating_point():
+        p2 = preds.reshape(preds.shape[0], preds.shape[1], -1).movedim(1, -1).reshape(-1, preds.shape[1])
+    else:
+        p2 = preds.reshape(-1)
+    from metrics_amd.ops import _hip
+
+    dummy = getattr(metric, "_hip_dummy", None)
+    if dummy is None or dummy.device != preds.device:
+        dummy = torch.zeros(3 * metric.num_classes + 1, dtype=torch.long, device=preds.device)
+        metric._hip_dummy = dummy
+    _hip.mc_confmat_into(p2, target.reshape(-1), metric.num_classes, metric.ignore_index, metric.confmat, dummy)
+    return True
+
+
 class MulticlassConfusionMatrix(Metric):
     """(C,C) confusion matrix for multiclass tasks (stateful)."""
 
@@ -110,20 +131,7 @@ class MulticlassConfusionMatrix(Metric):
         """Accumulate the batch confusion matrix (GPU: fused HIP kernel)."""
         if self.validate_args:
             _multiclass_confusion_matrix_tensor_validation(preds, target, self.num_classes, self.ignore_index)
-        if preds.is_cuda and (not preds.is_floating_point() or preds.dtype in (torch.float32, torch.bfloat16)):
-            if preds.ndim == target.ndim + 1 and preds.is_floating_point():
-                p2 = preds.reshape(preds.shape[0], preds.shape[1], -1).movedim(1, -1).reshape(-1, preds.shape[1])
-            else:
-                p2 = preds.reshape(-1)
-            from metrics_amd.ops import _hip
-
-            dummy = getattr(self, "_hip_dummy", None)
-            if dummy is None or dummy.device != preds.device:
-                # write-only sink for the per-class counters the fused kernel
-                # also produces; never read, never zeroed
-                dummy = torch.zeros(3 * self.num_classes + 1, dtype=torch.long, device=preds.device)
-                self._hip_dummy = dummy
-            _hip.mc_confmat_into(p2, target.reshape(-1), self.num_classes, self.ignore_index, self.confmat, dummy)
+        if _fused_confmat_update(self, preds, target):
             return
         preds, target = _multiclass_confusion_matrix_format(preds, target, self.ignore_index)
         confmat = _multiclass_confusion_matrix_update(preds, target, self.num_classes)

@@ -96,17 +96,9 @@ class MulticlassMatthewsCorrCoef(Metric):
         """Accumulate the batch confusion matrix (GPU: fused HIP kernel)."""
         if self.validate_args:
             _multiclass_confusion_matrix_tensor_validation(preds, target, self.num_classes, self.ignore_index)
-        if preds.is_cuda and (not preds.is_floating_point() or preds.dtype in (torch.float32, torch.bfloat16)):
-            from metrics_amd import ops
+        from metrics_amd.classification.confusion_matrix import _fused_confmat_update
 
-            if preds.ndim == target.ndim + 1 and preds.is_floating_point():
-                p2 = preds.reshape(preds.shape[0], preds.shape[1], -1).movedim(1, -1).reshape(-1, preds.shape[1])
-            else:
-                p2 = preds.reshape(-1)
-            _, _, _, _, confmat = ops.multiclass_stat_scores_fused(
-                p2, target.reshape(-1), self.num_classes, self.ignore_index, want_confmat=True
-            )
-            self.confmat += confmat
+        if _fused_confmat_update(self, preds, target):
             return
         preds, target = _multiclass_confusion_matrix_format(preds, target, self.ignore_index)
         self.confmat += _multiclass_confusion_matrix_update(preds, target, self.num_classes)
