@@ -15,6 +15,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 import torch
@@ -55,6 +56,7 @@ def main() -> None:
     parser.add_argument("--classes", type=int, default=1000)
     parser.add_argument("--compute-every", type=int, default=32)
     parser.add_argument("--curve-thresholds", type=int, default=200)
+    parser.add_argument("--no-graphs", action="store_true", help="disable hipGraph capture of the update")
     args = parser.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -85,8 +87,23 @@ def main() -> None:
 
     coll = build_collection(args.classes, device, args.curve_thresholds)
 
+    # hipGraph capture: one graph replay replaces the whole per-step launch
+    # train (the update is capture-safe by design — device-side epoch flags,
+    # persistent kernel scratch, fixed-shape states)
+    graphed = None
+    if use_gpu and not args.no_graphs:
+        try:
+            from metrics_amd.graphs import GraphedUpdate
+
+            graphed = GraphedUpdate(coll, preds[0], target[0])
+        except Exception as err:  # fall back to eager updates
+            print(f"hipGraph capture unavailable ({err}); running eager updates", file=sys.stderr)
+
     def one_step(i: int) -> None:
-        coll.update(preds[i % n_unique], target[i % n_unique])
+        if graphed is not None:
+            graphed.update(preds[i % n_unique], target[i % n_unique])
+        else:
+            coll.update(preds[i % n_unique], target[i % n_unique])
         if args.compute_every and (i + 1) % args.compute_every == 0:
             coll.compute()
 
@@ -95,7 +112,10 @@ def main() -> None:
     for i in range(args.warmup):
         one_step(i)
     coll.compute()
-    coll.reset()
+    if graphed is not None:
+        graphed.reset_states()
+    else:
+        coll.reset()
 
     if distributed:
         torch.distributed.barrier()

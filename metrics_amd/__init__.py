@@ -81,3 +81,5 @@ __all__ += audio.__all__
 __all__ += text.__all__
 __all__ += ["ProcrustesDisparity"]
 __all__ += wrappers.__all__
+
+from metrics_amd import graphs  # noqa: F401,E402  (hipGraph-captured updates)

@@ -80,7 +80,7 @@ class MulticlassExactMatch(Metric):
 
             scratch = getattr(self, "_hip_scratch", None)
             if scratch is None or scratch.device != preds.device:
-                scratch = torch.zeros(3 * self.num_classes + 2, dtype=torch.long, device=preds.device)
+                scratch = torch.zeros(3 * self.num_classes + 1, dtype=torch.long, device=preds.device)
                 self._hip_scratch = scratch
             _hip.mc_exact_into(preds, target, self.num_classes, self.ignore_index, scratch, self.correct, self.total)
             return

@@ -181,7 +181,7 @@ class MulticlassStatScores(_AbstractStatScores):
                 p2 = preds.reshape(-1)
             scratch = getattr(self, "_hip_scratch", None)
             if scratch is None or scratch.device != preds.device:
-                scratch = torch.zeros(3 * self.num_classes + 2, dtype=torch.long, device=preds.device)
+                scratch = torch.zeros(3 * self.num_classes + 1, dtype=torch.long, device=preds.device)
                 self._hip_scratch = scratch
             _hip.mc_stat_into(
                 p2, target.reshape(-1), self.num_classes, self.ignore_index, scratch,

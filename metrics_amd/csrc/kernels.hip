@@ -310,11 +310,13 @@ __device__ __forceinline__ int bucket_of_uniform(float p, const float* __restric
 }
 
 // detect values outside [0,1] (=> inputs are logits, normalize in-kernel).
-// Epoch protocol: the flag buffer is never reset; a value of `epoch` means
-// "this call saw out-of-range values" (epochs increase monotonically per call).
+// DEVICE-epoch protocol (hipGraph-capturable, no host state): E[1] counts
+// completed curve updates (bumped by k_curve_suffix), E[0] records the epoch
+// whose inputs were out-of-range. "Normalize this update" <=> E[0] == E[1]+1.
 template <typename T_, bool IS_BF16>
 __global__ void __launch_bounds__(256) k_range_flag(
-    const T_* __restrict__ x, ll N, int epoch, int* __restrict__ flag) {
+    const T_* __restrict__ x, ll N, unsigned int* __restrict__ E) {
+    const unsigned int cur = E[1] + 1u;  // stable: only k_curve_suffix writes E[1]
     ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
     ll stride = (ll)gridDim.x * blockDim.x;
     unsigned int outside = 0;
@@ -331,15 +333,16 @@ __global__ void __launch_bounds__(256) k_range_flag(
     __syncthreads();
     // poll before the global atomic: once any block set the epoch, the rest
     // skip — a per-wave atomicMax on one address serializes ~100ns each
-    if (threadIdx.x == 0 && blk_outside && *flag != epoch) atomicMax(flag, epoch);
+    if (threadIdx.x == 0 && blk_outside && E[0] != cur) atomicMax(&E[0], cur);
 }
 
 // per-row max + 1/sum(exp(x - max)) for in-kernel softmax (online, one pass),
 // plus the outside-[0,1] flag. Wave per row, lanes stride classes (coalesced).
 template <typename T_, bool IS_BF16>
 __global__ void __launch_bounds__(256) k_mc_rowstats(
-    const T_* __restrict__ probs, ll B, ll C, int epoch, int* __restrict__ flag,
+    const T_* __restrict__ probs, ll B, ll C, unsigned int* __restrict__ E,
     float* __restrict__ rowmax, float* __restrict__ rowinv) {
+    const unsigned int cur = E[1] + 1u;
     const int lane = threadIdx.x & (WAVE - 1);
     const int wave = threadIdx.x / WAVE;
     const int waves_per_block = blockDim.x / WAVE;
@@ -402,21 +405,21 @@ __global__ void __launch_bounds__(256) k_mc_rowstats(
     for (int off = WAVE / 2; off > 0; off >>= 1) outside |= __shfl_down(outside, off);
     if (lane == 0 && outside) atomicOr(&blk_outside2, 1u);
     __syncthreads();
-    if (threadIdx.x == 0 && blk_outside2 && *flag != epoch) atomicMax(flag, epoch);
+    if (threadIdx.x == 0 && blk_outside2 && E[0] != cur) atomicMax(&E[0], cur);
 }
 
 template <typename T_, bool IS_BF16>
 __global__ void __launch_bounds__(256) k_binary_curve_hist(
     const T_* __restrict__ preds, const ll* __restrict__ target, ll N,
     const float* __restrict__ thresholds, int T, ll ignore_index, int has_ignore,
-    int uniform, float t0, float inv_step, int norm_sigmoid, const int* __restrict__ flag,
-    int epoch, unsigned long long* __restrict__ hist /* (T+1,2) */) {
+    int uniform, float t0, float inv_step, int norm_sigmoid, const unsigned int* __restrict__ E,
+    unsigned long long* __restrict__ hist /* (T+1,2) */) {
     extern __shared__ unsigned int lhist[];  // (T+1)*2
     float* sthr = (float*)&lhist[(T + 1) * 2];
     for (int b = threadIdx.x; b < (T + 1) * 2; b += blockDim.x) lhist[b] = 0;
     for (int b = threadIdx.x; b < T; b += blockDim.x) sthr[b] = thresholds[b];
     __syncthreads();
-    const bool do_sigmoid = norm_sigmoid && flag && *flag == epoch;
+    const bool do_sigmoid = norm_sigmoid && E && E[0] == E[1] + 1u;
     ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
     ll stride = (ll)gridDim.x * blockDim.x;
     for (; i < N; i += stride) {
@@ -453,8 +456,8 @@ __global__ void __launch_bounds__(256) k_multiclass_curve_hist(
     const T_* __restrict__ probs, const ll* __restrict__ target, ll B, ll C,
     const float* __restrict__ thresholds, int T, ll ignore_index, int has_ignore, int mode,
     int uniform, float t0, float inv_step, int c_chunk,
-    int norm_kind /*0 none, 1 softmax (rowstats), 2 sigmoid*/, const int* __restrict__ flag,
-    int epoch, const float* __restrict__ rowmax, const float* __restrict__ rowinv,
+    int norm_kind /*0 none, 1 softmax (rowstats), 2 sigmoid*/, const unsigned int* __restrict__ E,
+    const float* __restrict__ rowmax, const float* __restrict__ rowinv,
     unsigned long long* __restrict__ hist /* (C, T+1, 2) */) {
     extern __shared__ float sthr2[];
     for (int b = threadIdx.x; b < T; b += blockDim.x) sthr2[b] = thresholds[b];
@@ -470,7 +473,7 @@ __global__ void __launch_bounds__(256) k_multiclass_curve_hist(
         trow = target[row];
         if (has_ignore && trow == ignore_index) valid = false;
     }
-    const int norm = (norm_kind && flag && *flag == epoch) ? norm_kind : 0;
+    const int norm = (norm_kind && E && E[0] == E[1] + 1u) ? norm_kind : 0;
     const float rmax = (norm == 1 && valid) ? rowmax[row] : 0.0f;
     const float rinv = (norm == 1 && valid) ? rowinv[row] : 0.0f;
     const T_* prow = probs + (valid ? row * C : 0);
@@ -522,7 +525,8 @@ __global__ void __launch_bounds__(256) k_multiclass_curve_hist(
 // permute+copy of the 6.4MB state.
 __global__ void k_curve_suffix(
     unsigned long long* __restrict__ hist /* (O, T+1, 2) */, int T, ll outer, int transposed,
-    int zero_hist, ll* __restrict__ confmat /* (O, T, 2, 2) or (T, O, 2, 2) */) {
+    int zero_hist, unsigned int* __restrict__ E /* nullable device-epoch */,
+    ll* __restrict__ confmat /* (O, T, 2, 2) or (T, O, 2, 2) */) {
     extern __shared__ unsigned long long sh[];  // (T+1) * 2
     const ll o = blockIdx.x;
     unsigned long long* h = hist + o * (ll)(T + 1) * 2;
@@ -551,6 +555,8 @@ __global__ void k_curve_suffix(
         c[2] += (ll)(pos_total - tp);      // [1][0] fn
         c[0] += (ll)(neg_total - fp);      // [0][0] tn
     }
+    // close this update's epoch: next curve update compares against E[1]+1
+    if (E && blockIdx.x == 0 && threadIdx.x == 0) atomicAdd(&E[1], 1u);
 }
 
 // ---------------------------------------------------------------------------
@@ -558,15 +564,18 @@ __global__ void k_curve_suffix(
 // by k_mc_stat_*, add the deltas into the four metric state tensors
 // (tn += valid - tp - fp - fn). ONE launch replaces ~8 small torch kernels.
 __global__ void k_apply_stat_deltas(
-    unsigned long long* __restrict__ scratch /* 3*C + 2 (two valid slots) */, ll C, int epoch,
+    unsigned long long* __restrict__ scratch /* 3*C + 1 */, ll C,
     ll* __restrict__ tp, ll* __restrict__ fp, ll* __restrict__ tn, ll* __restrict__ fn) {
-    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
-    ll stride = (ll)gridDim.x * blockDim.x;
-    // ping-pong valid slots: read this epoch's slot, zero the OTHER (consumed
-    // last epoch, to be reused next) — avoids a cross-block read/zero race
-    const unsigned long long valid = scratch[3 * C + (epoch & 1)];
-    if (i == 0) scratch[3 * C + ((epoch + 1) & 1)] = 0;
-    for (; i < C; i += stride) {
+    // SINGLE block: every thread can read the valid slot before thread 0
+    // zeroes it (post-sync), so the whole scratch is consumed and re-zeroed
+    // in one launch with no cross-block race and no host-side epochs —
+    // which also makes the launch hipGraph-capturable.
+    __shared__ unsigned long long valid_s;
+    if (threadIdx.x == 0) valid_s = scratch[3 * C];
+    __syncthreads();
+    const ll valid = (ll)valid_s;
+    if (threadIdx.x == 0) scratch[3 * C] = 0;
+    for (ll i = threadIdx.x; i < C; i += blockDim.x) {
         const ll dtp = (ll)scratch[i];
         const ll dfp = (ll)scratch[C + i];
         const ll dfn = (ll)scratch[2 * C + i];
@@ -576,13 +585,13 @@ __global__ void k_apply_stat_deltas(
         tp[i] += dtp;
         fp[i] += dfp;
         fn[i] += dfn;
-        tn[i] += (ll)valid - dtp - dfp - dfn;
+        tn[i] += valid - dtp - dfp - dfn;
     }
 }
 
 // exact-match epilogue: correct += sum(tp), total += valid — one block,
 // zeroes the scratch in-flight (same ping-pong valid protocol as apply_deltas).
-__global__ void k_exact_apply(unsigned long long* __restrict__ scratch, ll C, ll B, int epoch,
+__global__ void k_exact_apply(unsigned long long* __restrict__ scratch, ll C, ll B,
                               ll* __restrict__ correct, ll* __restrict__ total) {
     __shared__ unsigned long long part[256];
     unsigned long long acc = 0;
@@ -601,10 +610,10 @@ __global__ void k_exact_apply(unsigned long long* __restrict__ scratch, ll C, ll
     if (threadIdx.x == 0) {
         // reference semantics (_multiclass_exact_match_update): ignored
         // samples are forced equal, so they count as correct AND in total
-        const ll valid = (ll)scratch[3 * C + (epoch & 1)];
+        const ll valid = (ll)scratch[3 * C];
         correct[0] += (ll)part[0] + (B - valid);
         total[0] += B;
-        scratch[3 * C + ((epoch + 1) & 1)] = 0;
+        scratch[3 * C] = 0;
     }
 }
 
@@ -806,27 +815,27 @@ int ma_multilabel_stat(uintptr_t stream, uintptr_t preds, int dtype, uintptr_t t
 int ma_binary_curve_hist(uintptr_t stream, uintptr_t preds, int dtype, uintptr_t target, ll N,
                          uintptr_t thresholds, int T, ll ignore_index, int has_ignore,
                          int uniform, float t0, float inv_step, int norm_sigmoid, uintptr_t flag,
-                         int epoch, uintptr_t hist) {
+                         uintptr_t hist) {
     hipStream_t s = (hipStream_t)stream;
     size_t shmem = (size_t)(T + 1) * 2 * sizeof(unsigned int) + (size_t)T * sizeof(float);
     if (shmem > 160 * 1024) return -100;  // thresholds too large for LDS path
     if (norm_sigmoid && flag) {
         if (dtype == 0)
             k_range_flag<float, false><<<grid_for(N, 256), 256, 0, s>>>(
-                (const float*)preds, N, epoch, (int*)flag);
+                (const float*)preds, N, (unsigned int*)flag);
         else
             k_range_flag<unsigned short, true><<<grid_for(N, 256), 256, 0, s>>>(
-                (const unsigned short*)preds, N, epoch, (int*)flag);
+                (const unsigned short*)preds, N, (unsigned int*)flag);
     }
     if (dtype == 0)
         k_binary_curve_hist<float, false><<<grid_for(N, 256), 256, shmem, s>>>(
             (const float*)preds, (const ll*)target, N, (const float*)thresholds, T, ignore_index,
-            has_ignore, uniform, t0, inv_step, norm_sigmoid, (const int*)flag, epoch,
+            has_ignore, uniform, t0, inv_step, norm_sigmoid, (const unsigned int*)flag,
             (unsigned long long*)hist);
     else
         k_binary_curve_hist<unsigned short, true><<<grid_for(N, 256), 256, shmem, s>>>(
             (const unsigned short*)preds, (const ll*)target, N, (const float*)thresholds, T,
-            ignore_index, has_ignore, uniform, t0, inv_step, norm_sigmoid, (const int*)flag, epoch,
+            ignore_index, has_ignore, uniform, t0, inv_step, norm_sigmoid, (const unsigned int*)flag,
             (unsigned long long*)hist);
     return (int)hipGetLastError();
 }
@@ -834,7 +843,7 @@ int ma_binary_curve_hist(uintptr_t stream, uintptr_t preds, int dtype, uintptr_t
 int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintptr_t target, ll B,
                              ll C, uintptr_t thresholds, int T, ll ignore_index, int has_ignore,
                              int mode, int uniform, float t0, float inv_step, int norm_kind,
-                             uintptr_t flag, int epoch, uintptr_t rowmax, uintptr_t rowinv,
+                             uintptr_t flag, uintptr_t rowmax, uintptr_t rowinv,
                              uintptr_t hist) {
     hipStream_t s = (hipStream_t)stream;
     size_t shmem = (size_t)T * sizeof(float);
@@ -844,18 +853,18 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
         int grid = grid_for(B, 4);
         if (dtype == 0)
             k_mc_rowstats<float, false><<<grid, 256, 0, s>>>(
-                (const float*)probs, B, C, epoch, (int*)flag, (float*)rowmax, (float*)rowinv);
+                (const float*)probs, B, C, (unsigned int*)flag, (float*)rowmax, (float*)rowinv);
         else
             k_mc_rowstats<unsigned short, true><<<grid, 256, 0, s>>>(
-                (const unsigned short*)probs, B, C, epoch, (int*)flag, (float*)rowmax,
+                (const unsigned short*)probs, B, C, (unsigned int*)flag, (float*)rowmax,
                 (float*)rowinv);
     } else if (norm_kind == 2 && flag) {
         if (dtype == 0)
             k_range_flag<float, false><<<grid_for(B * C, 256), 256, 0, s>>>(
-                (const float*)probs, B * C, epoch, (int*)flag);
+                (const float*)probs, B * C, (unsigned int*)flag);
         else
             k_range_flag<unsigned short, true><<<grid_for(B * C, 256), 256, 0, s>>>(
-                (const unsigned short*)probs, B * C, epoch, (int*)flag);
+                (const unsigned short*)probs, B * C, (unsigned int*)flag);
     }
     ll row_chunks = (B + 255) / 256;
     // pick the class chunk so the grid comfortably overfills 256 CUs
@@ -867,40 +876,41 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
     if (dtype == 0)
         k_multiclass_curve_hist<float, false><<<grid, 256, shmem, s>>>(
             (const float*)probs, (const ll*)target, B, C, (const float*)thresholds, T, ignore_index,
-            has_ignore, mode, uniform, t0, inv_step, c_chunk, norm_kind, (const int*)flag, epoch,
-            (const float*)rowmax, (const float*)rowinv, (unsigned long long*)hist);
+            has_ignore, mode, uniform, t0, inv_step, c_chunk, norm_kind,
+            (const unsigned int*)flag, (const float*)rowmax, (const float*)rowinv,
+            (unsigned long long*)hist);
     else
         k_multiclass_curve_hist<unsigned short, true><<<grid, 256, shmem, s>>>(
             (const unsigned short*)probs, (const ll*)target, B, C, (const float*)thresholds, T,
             ignore_index, has_ignore, mode, uniform, t0, inv_step, c_chunk, norm_kind,
-            (const int*)flag, epoch, (const float*)rowmax, (const float*)rowinv,
+            (const unsigned int*)flag, (const float*)rowmax, (const float*)rowinv,
             (unsigned long long*)hist);
     return (int)hipGetLastError();
 }
 
 int ma_curve_suffix(uintptr_t stream, uintptr_t hist, ll outer, int T, int transposed,
-                    int zero_hist, uintptr_t confmat) {
+                    int zero_hist, uintptr_t epoch_buf, uintptr_t confmat) {
     hipStream_t s = (hipStream_t)stream;
     size_t shmem = (size_t)(T + 1) * 2 * sizeof(unsigned long long);
     if (shmem > 160 * 1024) return -100;
-    k_curve_suffix<<<(int)outer, 256, shmem, s>>>((unsigned long long*)hist, T, outer,
-                                                  transposed, zero_hist, (ll*)confmat);
+    k_curve_suffix<<<(int)outer, 256, shmem, s>>>((unsigned long long*)hist, T, outer, transposed,
+                                                  zero_hist, (unsigned int*)epoch_buf,
+                                                  (ll*)confmat);
     return (int)hipGetLastError();
 }
 
-int ma_apply_stat_deltas(uintptr_t stream, uintptr_t scratch, ll C, int epoch, uintptr_t tp,
+int ma_apply_stat_deltas(uintptr_t stream, uintptr_t scratch, ll C, uintptr_t tp,
                          uintptr_t fp, uintptr_t tn, uintptr_t fn) {
     hipStream_t s = (hipStream_t)stream;
-    k_apply_stat_deltas<<<grid_for(C, 256), 256, 0, s>>>(
-        (unsigned long long*)scratch, C, epoch, (ll*)tp, (ll*)fp, (ll*)tn, (ll*)fn);
+    k_apply_stat_deltas<<<1, 256, 0, s>>>(
+        (unsigned long long*)scratch, C, (ll*)tp, (ll*)fp, (ll*)tn, (ll*)fn);
     return (int)hipGetLastError();
 }
 
-int ma_exact_apply(uintptr_t stream, uintptr_t scratch, ll C, ll B, int epoch, uintptr_t correct,
+int ma_exact_apply(uintptr_t stream, uintptr_t scratch, ll C, ll B, uintptr_t correct,
                    uintptr_t total) {
     hipStream_t s = (hipStream_t)stream;
-    k_exact_apply<<<1, 256, 0, s>>>((unsigned long long*)scratch, C, B, epoch, (ll*)correct,
-                                    (ll*)total);
+    k_exact_apply<<<1, 256, 0, s>>>((unsigned long long*)scratch, C, B, (ll*)correct, (ll*)total);
     return (int)hipGetLastError();
 }
 

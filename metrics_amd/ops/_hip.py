@@ -50,11 +50,11 @@ _SIGNATURES = {
     "ma_bincount": [_U64, _U64, _LL, _LL, _U64],
     "ma_binary_stat": [_U64, _U64, _I, _U64, _LL, _F, _LL, _I, _U64, _U64],
     "ma_multilabel_stat": [_U64, _U64, _I, _U64, _LL, _LL, _F, _LL, _I, _U64, _U64],
-    "ma_binary_curve_hist": [_U64, _U64, _I, _U64, _LL, _U64, _I, _LL, _I, _I, _F, _F, _I, _U64, _I, _U64],
-    "ma_multiclass_curve_hist": [_U64, _U64, _I, _U64, _LL, _LL, _U64, _I, _LL, _I, _I, _I, _F, _F, _I, _U64, _I, _U64, _U64, _U64],
-    "ma_curve_suffix": [_U64, _U64, _LL, _I, _I, _I, _U64],
-    "ma_apply_stat_deltas": [_U64, _U64, _LL, _I, _U64, _U64, _U64, _U64],
-    "ma_exact_apply": [_U64, _U64, _LL, _LL, _I, _U64, _U64],
+    "ma_binary_curve_hist": [_U64, _U64, _I, _U64, _LL, _U64, _I, _LL, _I, _I, _F, _F, _I, _U64, _U64],
+    "ma_multiclass_curve_hist": [_U64, _U64, _I, _U64, _LL, _LL, _U64, _I, _LL, _I, _I, _I, _F, _F, _I, _U64, _U64, _U64, _U64],
+    "ma_curve_suffix": [_U64, _U64, _LL, _I, _I, _I, _U64, _U64],
+    "ma_apply_stat_deltas": [_U64, _U64, _LL, _U64, _U64, _U64, _U64],
+    "ma_exact_apply": [_U64, _U64, _LL, _LL, _U64, _U64],
     "ma_err_reduce": [_U64, _U64, _U64, _I, _LL, _I, _D, _U64, _I, _I, _U64],
     "ma_box_iou": [_U64, _U64, _LL, _U64, _LL, _I, _U64],
 }
@@ -250,27 +250,24 @@ def multilabel_stat(
 
 _UNIFORM_CACHE: dict = {}
 
-# Per-device out-of-range flag buffer + monotonically increasing epoch. The
-# flag is NEVER reset: kernels write the current epoch into it when they see
-# values outside [0,1], and consumers compare against that exact epoch — so no
-# per-update fill is needed. Stream-safety: all launches go to the caller's
-# current stream; buffers are shared per device, which is safe because each
-# (detect -> consume) pair is issued back-to-back on one stream.
+# Per-device DEVICE-side epoch buffer E (2 x uint32): E[1] counts completed
+# curve updates (bumped by the suffix kernel), E[0] records the epoch whose
+# inputs were out-of-range. "Normalize this update" <=> E[0] == E[1]+1. All
+# state lives on device, so the whole protocol is hipGraph-capturable.
+# Stream-safety: launches go to the caller's current stream; each
+# (detect -> histogram -> suffix) triple is issued back-to-back on one stream.
 _FLAG_BUFS: dict = {}
-_EPOCHS: dict = {}
 _HIST_POOL: dict = {}
 _ROWSTATS_POOL: dict = {}
 
 
-def _flag_and_epoch(device) -> Tuple[Tensor, int]:
+def _epoch_buf(device) -> Tensor:
     key = device.index
     buf = _FLAG_BUFS.get(key)
     if buf is None:
-        buf = torch.zeros(1, dtype=torch.int32, device=device)
+        buf = torch.zeros(2, dtype=torch.int32, device=device)
         _FLAG_BUFS[key] = buf
-        _EPOCHS[key] = 0
-    _EPOCHS[key] += 1
-    return buf, _EPOCHS[key]
+    return buf
 
 
 def _pooled_hist(outer: int, T: int, device) -> Tensor:
@@ -347,7 +344,6 @@ def binary_curve_confmat(
         inv_step,
         0,
         0,
-        0,
         hist.data_ptr(),
     )
     _check(rc, "ma_binary_curve_hist")
@@ -357,6 +353,7 @@ def binary_curve_confmat(
         hist.data_ptr(),
         1,
         T,
+        0,
         0,
         0,
         confmat.data_ptr(),
@@ -398,7 +395,6 @@ def multiclass_curve_confmat(
         0,
         0,
         0,
-        0,
         hist.data_ptr(),
     )
     _check(rc, "ma_multiclass_curve_hist")
@@ -410,6 +406,7 @@ def multiclass_curve_confmat(
         C,
         T,
         1,
+        0,
         0,
         confmat.data_ptr(),
     )
@@ -497,10 +494,9 @@ def curve_hist_into_confmat(
         tgt = target.contiguous().long().flatten()
         hist = _pooled_hist(1, T, dev)
         if norm == "sigmoid":
-            flag, epoch = _flag_and_epoch(dev)
-            norm_i, flag_ptr = 1, flag.data_ptr()
+            norm_i, flag_ptr = 1, _epoch_buf(dev).data_ptr()
         else:
-            norm_i, flag_ptr, epoch = 0, 0, 0
+            norm_i, flag_ptr = 0, 0
         rc = lib.ma_binary_curve_hist(
             _stream(),
             preds.data_ptr(),
@@ -516,7 +512,6 @@ def curve_hist_into_confmat(
             inv_step,
             norm_i,
             flag_ptr,
-            epoch,
             hist.data_ptr(),
         )
         _check(rc, "ma_binary_curve_hist")
@@ -527,7 +522,6 @@ def curve_hist_into_confmat(
         B, C = probs.shape
         hist = _pooled_hist(C, T, dev)
         if norm == "softmax":
-            flag, epoch = _flag_and_epoch(dev)
             rkey = (B, dev.index)
             rbuf = _ROWSTATS_POOL.get(rkey)
             if rbuf is None:
@@ -535,13 +529,12 @@ def curve_hist_into_confmat(
                 if len(_ROWSTATS_POOL) > 64:
                     _ROWSTATS_POOL.clear()
                 _ROWSTATS_POOL[rkey] = rbuf
-            norm_i, flag_ptr = 1, flag.data_ptr()
+            norm_i, flag_ptr = 1, _epoch_buf(dev).data_ptr()
             rm_ptr, ri_ptr = rbuf[0].data_ptr(), rbuf[1].data_ptr()
         elif norm == "sigmoid":
-            flag, epoch = _flag_and_epoch(dev)
-            norm_i, flag_ptr, rm_ptr, ri_ptr = 2, flag.data_ptr(), 0, 0
+            norm_i, flag_ptr, rm_ptr, ri_ptr = 2, _epoch_buf(dev).data_ptr(), 0, 0
         else:
-            norm_i, flag_ptr, epoch, rm_ptr, ri_ptr = 0, 0, 0, 0, 0
+            norm_i, flag_ptr, rm_ptr, ri_ptr = 0, 0, 0, 0
         rc = lib.ma_multiclass_curve_hist(
             _stream(),
             probs.data_ptr(),
@@ -559,7 +552,6 @@ def curve_hist_into_confmat(
             inv_step,
             norm_i,
             flag_ptr,
-            epoch,
             rm_ptr,
             ri_ptr,
             hist.data_ptr(),
@@ -574,12 +566,10 @@ def curve_hist_into_confmat(
         T,
         transposed,
         1,  # re-zero the pooled hist in-flight
+        _epoch_buf(dev).data_ptr() if norm is not None else 0,
         confmat_state.data_ptr(),
     )
     _check(rc, "ma_curve_suffix")
-
-
-_SCRATCH_EPOCHS: dict = {}
 
 
 def mc_stat_into(
@@ -588,23 +578,17 @@ def mc_stat_into(
 ) -> None:
     """Fused stat-scores update accumulated DIRECTLY into the metric states.
 
-    ``scratch`` is a per-metric reusable (3*C+2,) int64 buffer (tp|fp|fn counts
-    plus TWO ping-pong valid slots); 2 launches total (count + apply): the
-    apply kernel zeroes the scratch in-flight for the next update, so no fill
-    kernel runs in steady state.
+    ``scratch`` is a per-metric reusable (3*C+1,) int64 buffer; 2 launches
+    total (count + single-block apply): the apply kernel consumes AND zeroes
+    the whole scratch in one pass, so no fill kernel runs in steady state and
+    no host-side epoch is needed (hipGraph-capturable).
     """
     lib = _lib()
     C = num_classes
-    key = scratch.data_ptr()
-    epoch = _SCRATCH_EPOCHS.get(key, 0)
-    _SCRATCH_EPOCHS[key] = epoch + 1
-    if len(_SCRATCH_EPOCHS) > 4096:
-        _SCRATCH_EPOCHS.clear()
-        _SCRATCH_EPOCHS[key] = epoch + 1
     s_tp = scratch[:C]
     s_fp = scratch[C : 2 * C]
     s_fn = scratch[2 * C : 3 * C]
-    s_valid_ptr = scratch.data_ptr() + (3 * C + (epoch & 1)) * 8
+    s_valid_ptr = scratch.data_ptr() + 3 * C * 8
     if preds.ndim == 2 and preds.is_floating_point():
         preds = preds.contiguous()
         target = target.contiguous().long()
@@ -648,7 +632,6 @@ def mc_stat_into(
         _stream(),
         scratch.data_ptr(),
         C,
-        epoch,
         tp.data_ptr(),
         fp.data_ptr(),
         tn.data_ptr(),
@@ -719,15 +702,12 @@ def mc_exact_into(
     """Exact-match (== row argmax equals target) accumulated into correct/total.
 
     Reuses the fused stat kernel's per-class tp counts: correct = sum(tp),
-    total = valid rows. Same (3*C+2,) self-zeroing scratch protocol as
+    total = valid rows. Same (3*C+1,) self-zeroing scratch protocol as
     :func:`mc_stat_into`; 2 launches, zero fills in steady state.
     """
     lib = _lib()
     C = num_classes
-    key = scratch.data_ptr()
-    epoch = _SCRATCH_EPOCHS.get(key, 0)
-    _SCRATCH_EPOCHS[key] = epoch + 1
-    s_valid_ptr = scratch.data_ptr() + (3 * C + (epoch & 1)) * 8
+    s_valid_ptr = scratch.data_ptr() + 3 * C * 8
     if preds.ndim == 2 and preds.is_floating_point():
         preds = preds.contiguous()
         target = target.contiguous().long()
@@ -772,7 +752,6 @@ def mc_exact_into(
         scratch.data_ptr(),
         C,
         target.numel(),
-        epoch,
         correct.data_ptr(),
         total.data_ptr(),
     )

@@ -394,3 +394,51 @@ def test_exact_match_fused_gpu():
     ci.update(preds, tgt)
     assert torch.equal(gi.correct.cpu(), ci.correct) and torch.equal(gi.total.cpu(), ci.total)
     assert torch.allclose(gi.compute().cpu(), ci.compute())
+
+
+def test_graphed_update_matches_eager():
+    """hipGraph-captured collection update == eager updates, incl. the
+    device-side epoch protocol (logits-normalization) inside the graph."""
+    torch.manual_seed(17)
+    def make():
+        return ma.MetricCollection({
+            "acc": ma.MulticlassAccuracy(num_classes=50, average="macro", validate_args=False),
+            "f1": ma.MulticlassF1Score(num_classes=50, average="weighted", validate_args=False),
+            "confmat": ma.MulticlassConfusionMatrix(num_classes=50, validate_args=False),
+            "exact": ma.MulticlassExactMatch(num_classes=50, validate_args=False),
+            "auroc": ma.MulticlassAUROC(num_classes=50, thresholds=64, validate_args=False),
+        }).to("cuda")
+
+    batches = [(torch.randn(2048, 50, device="cuda", dtype=torch.bfloat16),
+                torch.randint(0, 50, (2048,), device="cuda")) for _ in range(3)]
+    eager = make()
+    for p, t in batches:
+        eager.update(p, t)
+    res_eager = eager.compute()
+
+    from metrics_amd.graphs import GraphedUpdate
+
+    graphed_coll = make()
+    graphed = GraphedUpdate(graphed_coll, batches[0][0], batches[0][1])
+    for p, t in batches:
+        graphed.update(p, t)
+    res_graph = graphed_coll.compute()
+    for k in res_eager:
+        assert torch.allclose(res_eager[k].float(), res_graph[k].float(), atol=1e-6), k
+
+    # reset_states + reuse
+    graphed.reset_states()
+    graphed.update(*batches[0])
+    single = make()
+    single.update(*batches[0])
+    r1, r2 = graphed_coll.compute(), single.compute()
+    for k in r1:
+        assert torch.allclose(r1[k].float(), r2[k].float(), atol=1e-6), k
+
+
+def test_graphed_update_rejects_list_states():
+    from metrics_amd.graphs import GraphedUpdate
+
+    m = ma.MulticlassPrecisionRecallCurve(num_classes=5, thresholds=None).to("cuda")
+    with pytest.raises(RuntimeError, match="list state"):
+        GraphedUpdate(m, torch.rand(8, 5, device="cuda"), torch.randint(0, 5, (8,), device="cuda"))
