@@ -83,7 +83,7 @@ class BinaryPrecisionRecallCurve(Metric):
             from metrics_amd.ops import _hip
 
             _hip.curve_hist_into_confmat(
-                preds_f, target_f, self.thresholds, self.ignore_index, self.confmat, mode=0, norm="sigmoid"
+                preds_f, target_f, self.thresholds, self.ignore_index, self.confmat, mode=0, norm="sigmoid", owner=self
             )
             return
         state = _binary_precision_recall_curve_update(
@@ -178,7 +178,7 @@ class MulticlassPrecisionRecallCurve(Metric):
 
             _hip.curve_hist_into_confmat(
                 preds_f, target_f, self.thresholds, self.ignore_index if not remove_ignored else None,
-                self.confmat, mode=0, norm="softmax",
+                self.confmat, mode=0, norm="softmax", owner=self,
             )
             return
         state = _multiclass_precision_recall_curve_update(
@@ -259,7 +259,7 @@ class MultilabelPrecisionRecallCurve(Metric):
 
             _hip.curve_hist_into_confmat(
                 preds_f, target_f, self.thresholds, self.ignore_index if not remove_ignored else None,
-                self.confmat, mode=1, norm="sigmoid",
+                self.confmat, mode=1, norm="sigmoid", owner=self,
             )
             return
         state = _multilabel_precision_recall_curve_update(

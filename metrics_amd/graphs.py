@@ -49,7 +49,7 @@ class GraphedUpdate:
     copies the batch into the static input buffers and replays the graph.
     """
 
-    def __init__(self, target: Any, *example_args: Tensor, warmup: int = 3) -> None:
+    def __init__(self, target: Any, *example_args: Tensor, warmup: int = 3, parallel: bool = True) -> None:
         if not torch.cuda.is_available():
             raise RuntimeError("GraphedUpdate requires a GPU (hipGraph capture)")
         self.target = target
@@ -71,9 +71,26 @@ class GraphedUpdate:
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
 
+        # compute-group leaders are independent (disjoint states, read-only
+        # inputs, per-metric kernel scratch): record each on its own stream so
+        # replay overlaps their kernels — the graph becomes a parallel DAG
+        leaders = None
+        if parallel and not isinstance(target, Metric) and getattr(target, "_groups_checked", False):
+            leaders = [getattr(target, members[0]) for members in target._groups.values()]
+        self._streams = [torch.cuda.Stream() for _ in (leaders or [])]
+
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
-            target.update(*self._static)
+            if leaders and len(leaders) > 1:
+                main = torch.cuda.current_stream()
+                for m, st in zip(leaders, self._streams):
+                    st.wait_stream(main)
+                    with torch.cuda.stream(st):
+                        m.update(*self._static)
+                for st in self._streams:
+                    main.wait_stream(st)
+            else:
+                target.update(*self._static)
 
         # capture records but does not execute: only the warmup polluted the
         # states — restore defaults IN PLACE (the graph holds state pointers)
@@ -105,5 +122,11 @@ class GraphedUpdate:
         for m in self._metrics:
             m._update_count += 1
             m._computed = None
+        # copy-on-access (items()/values()) de-aliases compute-group members;
+        # re-establish the state refs so compute() sees the leaders' states
+        t = self.target
+        if getattr(t, "_state_is_copy", False):
+            t._compute_groups_create_state_ref()
+            t._state_is_copy = False
 
     __call__ = update

@@ -261,7 +261,16 @@ _HIST_POOL: dict = {}
 _ROWSTATS_POOL: dict = {}
 
 
-def _epoch_buf(device) -> Tensor:
+def _epoch_buf(device, owner=None) -> Tensor:
+    """Device-epoch buffer E (2 x uint32). Owned per-METRIC when ``owner`` is
+    given so independent metrics can update on parallel streams without racing
+    on the protocol state; falls back to a per-device buffer otherwise."""
+    if owner is not None:
+        buf = owner.__dict__.get("_hip_epoch_buf")
+        if buf is None or buf.device != device:
+            buf = torch.zeros(2, dtype=torch.int32, device=device)
+            owner.__dict__["_hip_epoch_buf"] = buf
+        return buf
     key = device.index
     buf = _FLAG_BUFS.get(key)
     if buf is None:
@@ -270,9 +279,16 @@ def _epoch_buf(device) -> Tensor:
     return buf
 
 
-def _pooled_hist(outer: int, T: int, device) -> Tensor:
+def _pooled_hist(outer: int, T: int, device, owner=None) -> Tensor:
     """Persistent histogram scratch, zeroed once; the suffix kernel re-zeroes
-    it in-flight after consuming it (zero_hist=1), so reuse needs no fill."""
+    it in-flight after consuming it (zero_hist=1), so reuse needs no fill.
+    Owned per-METRIC when ``owner`` is given (parallel-stream safety)."""
+    if owner is not None:
+        buf = owner.__dict__.get("_hip_hist_buf")
+        if buf is None or buf.device != device or buf.shape != (outer, T + 1, 2):
+            buf = torch.zeros(outer, T + 1, 2, dtype=torch.long, device=device)
+            owner.__dict__["_hip_hist_buf"] = buf
+        return buf
     key = (outer, T, device.index)
     buf = _HIST_POOL.get(key)
     if buf is None:
@@ -470,7 +486,7 @@ def box_iou(boxes1: Tensor, boxes2: Tensor, variant: str = "iou") -> Tensor:
 
 def curve_hist_into_confmat(
     probs: Tensor, target: Tensor, thresholds: Tensor, ignore_index: Optional[int],
-    confmat_state: Tensor, mode: int, norm: Optional[str] = None,
+    confmat_state: Tensor, mode: int, norm: Optional[str] = None, owner=None,
 ) -> None:
     """Bucketized histogram + transposed suffix-sum accumulated DIRECTLY into the
     metric's confmat state ((T,2,2) binary / (T,C,2,2) multiclass|multilabel) —
@@ -492,9 +508,9 @@ def curve_hist_into_confmat(
     if confmat_state.ndim == 3:  # binary (T,2,2)
         preds = probs.contiguous().flatten()
         tgt = target.contiguous().long().flatten()
-        hist = _pooled_hist(1, T, dev)
+        hist = _pooled_hist(1, T, dev, owner)
         if norm == "sigmoid":
-            norm_i, flag_ptr = 1, _epoch_buf(dev).data_ptr()
+            norm_i, flag_ptr = 1, _epoch_buf(dev, owner).data_ptr()
         else:
             norm_i, flag_ptr = 0, 0
         rc = lib.ma_binary_curve_hist(
@@ -520,19 +536,25 @@ def curve_hist_into_confmat(
         probs = probs.contiguous()
         tgt = target.contiguous().long()
         B, C = probs.shape
-        hist = _pooled_hist(C, T, dev)
+        hist = _pooled_hist(C, T, dev, owner)
         if norm == "softmax":
-            rkey = (B, dev.index)
-            rbuf = _ROWSTATS_POOL.get(rkey)
-            if rbuf is None:
-                rbuf = torch.empty(2, B, dtype=torch.float32, device=dev)
-                if len(_ROWSTATS_POOL) > 64:
-                    _ROWSTATS_POOL.clear()
-                _ROWSTATS_POOL[rkey] = rbuf
-            norm_i, flag_ptr = 1, _epoch_buf(dev).data_ptr()
+            if owner is not None:
+                rbuf = owner.__dict__.get("_hip_rowstats_buf")
+                if rbuf is None or rbuf.device != dev or rbuf.shape[1] < B:
+                    rbuf = torch.empty(2, B, dtype=torch.float32, device=dev)
+                    owner.__dict__["_hip_rowstats_buf"] = rbuf
+            else:
+                rkey = (B, dev.index)
+                rbuf = _ROWSTATS_POOL.get(rkey)
+                if rbuf is None:
+                    rbuf = torch.empty(2, B, dtype=torch.float32, device=dev)
+                    if len(_ROWSTATS_POOL) > 64:
+                        _ROWSTATS_POOL.clear()
+                    _ROWSTATS_POOL[rkey] = rbuf
+            norm_i, flag_ptr = 1, _epoch_buf(dev, owner).data_ptr()
             rm_ptr, ri_ptr = rbuf[0].data_ptr(), rbuf[1].data_ptr()
         elif norm == "sigmoid":
-            norm_i, flag_ptr, rm_ptr, ri_ptr = 2, _epoch_buf(dev).data_ptr(), 0, 0
+            norm_i, flag_ptr, rm_ptr, ri_ptr = 2, _epoch_buf(dev, owner).data_ptr(), 0, 0
         else:
             norm_i, flag_ptr, rm_ptr, ri_ptr = 0, 0, 0, 0
         rc = lib.ma_multiclass_curve_hist(
@@ -566,7 +588,7 @@ def curve_hist_into_confmat(
         T,
         transposed,
         1,  # re-zero the pooled hist in-flight
-        _epoch_buf(dev).data_ptr() if norm is not None else 0,
+        _epoch_buf(dev, owner).data_ptr() if norm is not None else 0,
         confmat_state.data_ptr(),
     )
     _check(rc, "ma_curve_suffix")
