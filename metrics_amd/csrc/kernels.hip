@@ -617,6 +617,85 @@ __global__ void k_exact_apply(unsigned long long* __restrict__ scratch, ll C, ll
     }
 }
 
+// one-launch stat-score compute: every precision/recall/accuracy/F-beta/
+// specificity/NPV/hamming reduction is post(safe_div(n.s, d.s)) with linear
+// coefficients over (tp,fp,tn,fn), then micro / macro / weighted averaging —
+// ONE kernel replaces the ~15-launch torch chain per metric at compute time.
+// avg_mode: 0 macro, 1 weighted, 2 micro. Matches _adjust_weights_safe_divide
+// bit-for-bit: per-class (w*s)/W summed (not sum(w*s)/W).
+__global__ void k_linear_stat_compute(
+    const ll* __restrict__ tp, const ll* __restrict__ fp, const ll* __restrict__ tn,
+    const ll* __restrict__ fn, ll C, float n0, float n1, float n2, float n3, float d0, float d1,
+    float d2, float d3, int avg_mode, int zero_w_topk, float zero_division, float post_a,
+    float post_b, float* __restrict__ out) {
+    __shared__ double red[256];
+    // pass 1: total weight W (macro/weighted) or nothing (micro)
+    double wsum = 0.0;
+    if (avg_mode != 2) {
+        for (ll i = threadIdx.x; i < C; i += blockDim.x) {
+            const float vtp = (float)tp[i], vfp = (float)fp[i], vfn = (float)fn[i];
+            if (avg_mode == 1) {
+                wsum += (double)(vtp + vfn);
+            } else {
+                const float empty = zero_w_topk ? (vtp + vfn) : (vtp + vfp + vfn);
+                wsum += (empty == 0.0f) ? 0.0 : 1.0;
+            }
+        }
+    }
+    red[threadIdx.x] = wsum;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+        __syncthreads();
+    }
+    const float W = (float)red[0];
+    __syncthreads();
+
+    // pass 2: sum of per-class contributions, matching torch's
+    // _safe_divide(w*s, W).sum() element order: each term is (w*s)/W in f32
+    double a = 0.0, b = 0.0;
+    for (ll i = threadIdx.x; i < C; i += blockDim.x) {
+        const float vtp = (float)tp[i], vfp = (float)fp[i], vtn = (float)tn[i], vfn = (float)fn[i];
+        const float num = n0 * vtp + n1 * vfp + n2 * vtn + n3 * vfn;
+        const float den = d0 * vtp + d1 * vfp + d2 * vtn + d3 * vfn;
+        if (avg_mode == 2) {
+            a += (double)num;
+            b += (double)den;
+        } else {
+            const float s = post_a * (den != 0.0f ? num / den : zero_division) + post_b;
+            float w;
+            if (avg_mode == 1) {
+                w = vtp + vfn;
+            } else {
+                const float empty = zero_w_topk ? (vtp + vfn) : (vtp + vfp + vfn);
+                w = (empty == 0.0f) ? 0.0f : 1.0f;
+            }
+            a += (double)(W != 0.0f ? (w * s) / W : 0.0f);
+        }
+    }
+    red[threadIdx.x] = a;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+        __syncthreads();
+    }
+    if (avg_mode == 2) {
+        __shared__ double redb[256];
+        redb[threadIdx.x] = b;
+        __syncthreads();
+        for (int off = 128; off > 0; off >>= 1) {
+            if (threadIdx.x < off) redb[threadIdx.x] += redb[threadIdx.x + off];
+            __syncthreads();
+        }
+        if (threadIdx.x == 0) {
+            const float fa = (float)red[0], fb = (float)redb[0];
+            out[0] = post_a * (fb != 0.0f ? fa / fb : zero_division) + post_b;
+        }
+    } else if (threadIdx.x == 0) {
+        out[0] = (float)red[0];
+    }
+}
+
 // ---------------------------------------------------------------------------
 // K13: fused elementwise-error reductions, deterministic fp64 two-pass.
 // op: 0 = squared error, 1 = abs error, 2 = abs percentage |d|/max(|t|,eps),
@@ -904,6 +983,18 @@ int ma_apply_stat_deltas(uintptr_t stream, uintptr_t scratch, ll C, uintptr_t tp
     hipStream_t s = (hipStream_t)stream;
     k_apply_stat_deltas<<<1, 256, 0, s>>>(
         (unsigned long long*)scratch, C, (ll*)tp, (ll*)fp, (ll*)tn, (ll*)fn);
+    return (int)hipGetLastError();
+}
+
+int ma_linear_stat_compute(uintptr_t stream, uintptr_t tp, uintptr_t fp, uintptr_t tn,
+                           uintptr_t fn, ll C, float n0, float n1, float n2, float n3, float d0,
+                           float d1, float d2, float d3, int avg_mode, int zero_w_topk,
+                           float zero_division, float post_a, float post_b, uintptr_t out) {
+    hipStream_t s = (hipStream_t)stream;
+    k_linear_stat_compute<<<1, 256, 0, s>>>((const ll*)tp, (const ll*)fp, (const ll*)tn,
+                                            (const ll*)fn, C, n0, n1, n2, n3, d0, d1, d2, d3,
+                                            avg_mode, zero_w_topk, zero_division, post_a, post_b,
+                                            (float*)out);
     return (int)hipGetLastError();
 }
 

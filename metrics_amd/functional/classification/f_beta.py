@@ -32,6 +32,17 @@ def _fbeta_reduce(
     zero_division: float = 0,
 ) -> Tensor:
     beta2 = beta**2
+    if (
+        tp.is_cuda and tp.ndim == 1 and multidim_average == "global" and not multilabel
+        and average in ("micro", "macro", "weighted")
+    ):
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            return _hip.linear_stat_compute(
+                tp, fp, tn, fn, (1 + beta2, 0, 0, 0), (1 + beta2, 1, 0, beta2), average,
+                False, zero_division,
+            )
     if average == "binary":
         return _safe_divide((1 + beta2) * tp, (1 + beta2) * tp + beta2 * fn + fp, zero_division)
     if average == "micro":

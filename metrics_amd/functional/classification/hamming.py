@@ -29,6 +29,16 @@ def _hamming_distance_reduce(
     multidim_average: str = "global",
     multilabel: bool = False,
 ) -> Tensor:
+    if (
+        tp.is_cuda and tp.ndim == 1 and multidim_average == "global" and not multilabel
+        and average in ("micro", "macro", "weighted")
+    ):
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            return _hip.linear_stat_compute(
+                tp, fp, tn, fn, (1, 0, 0, 0), (1, 0, 0, 1), average, post_a=-1.0, post_b=1.0
+            )
     if average == "binary":
         return 1 - _safe_divide(tp + tn, tp + fp + tn + fn)
     if average == "micro":

@@ -29,6 +29,14 @@ def _negative_predictive_value_reduce(
     multidim_average: str = "global",
     multilabel: bool = False,
 ) -> Tensor:
+    if (
+        tp.is_cuda and tp.ndim == 1 and multidim_average == "global" and not multilabel
+        and average in ("micro", "macro", "weighted")
+    ):
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            return _hip.linear_stat_compute(tp, fp, tn, fn, (0, 0, 1, 0), (0, 0, 1, 1), average)
     if average == "binary":
         return _safe_divide(tn, tn + fn)
     if average == "micro":

@@ -33,6 +33,17 @@ def _precision_recall_reduce(
     zero_division: float = 0,
 ) -> Tensor:
     different_stat = fp if stat == "precision" else fn  # this is what differs between the two scores
+    if (
+        tp.is_cuda and tp.ndim == 1 and multidim_average == "global" and not multilabel
+        and average in ("micro", "macro", "weighted")
+    ):
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            den = (1, 1, 0, 0) if stat == "precision" else (1, 0, 0, 1)
+            return _hip.linear_stat_compute(
+                tp, fp, tn, fn, (1, 0, 0, 0), den, average, top_k != 1, zero_division
+            )
     if average == "binary":
         return _safe_divide(tp, tp + different_stat, zero_division)
     if average == "micro":

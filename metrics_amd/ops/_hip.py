@@ -55,6 +55,7 @@ _SIGNATURES = {
     "ma_curve_suffix": [_U64, _U64, _LL, _I, _I, _I, _U64, _U64],
     "ma_apply_stat_deltas": [_U64, _U64, _LL, _U64, _U64, _U64, _U64],
     "ma_exact_apply": [_U64, _U64, _LL, _LL, _U64, _U64],
+    "ma_linear_stat_compute": [_U64, _U64, _U64, _U64, _U64, _LL, _F, _F, _F, _F, _F, _F, _F, _F, _I, _I, _F, _F, _F, _U64],
     "ma_err_reduce": [_U64, _U64, _U64, _I, _LL, _I, _D, _U64, _I, _I, _U64],
     "ma_box_iou": [_U64, _U64, _LL, _U64, _LL, _I, _U64],
 }
@@ -778,3 +779,35 @@ def mc_exact_into(
         total.data_ptr(),
     )
     _check(rc, "ma_exact_apply")
+
+
+def linear_stat_compute(
+    tp: Tensor, fp: Tensor, tn: Tensor, fn: Tensor,
+    num_coefs, den_coefs, average: str, zero_w_topk: bool = False,
+    zero_division: float = 0.0, post_a: float = 1.0, post_b: float = 0.0,
+) -> Tensor:
+    """One-launch linear-ratio stat reduction (precision/recall/accuracy/...).
+
+    score_c = post(safe_div(num_coefs . stats_c, den_coefs . stats_c)), then
+    micro / macro / weighted averaging exactly like
+    ``_adjust_weights_safe_divide``. Returns a 0-dim float32 tensor.
+    """
+    lib = _lib()
+    C = tp.numel()
+    out = torch.empty((), dtype=torch.float32, device=tp.device)
+    avg_mode = {"macro": 0, "weighted": 1, "micro": 2}[average]
+    rc = lib.ma_linear_stat_compute(
+        _stream(),
+        tp.data_ptr(), fp.data_ptr(), tn.data_ptr(), fn.data_ptr(),
+        C,
+        float(num_coefs[0]), float(num_coefs[1]), float(num_coefs[2]), float(num_coefs[3]),
+        float(den_coefs[0]), float(den_coefs[1]), float(den_coefs[2]), float(den_coefs[3]),
+        avg_mode,
+        1 if zero_w_topk else 0,
+        float(zero_division),
+        float(post_a),
+        float(post_b),
+        out.data_ptr(),
+    )
+    _check(rc, "ma_linear_stat_compute")
+    return out

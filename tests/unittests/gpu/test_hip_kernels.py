@@ -442,3 +442,40 @@ def test_graphed_update_rejects_list_states():
     m = ma.MulticlassPrecisionRecallCurve(num_classes=5, thresholds=None).to("cuda")
     with pytest.raises(RuntimeError, match="list state"):
         GraphedUpdate(m, torch.rand(8, 5, device="cuda"), torch.randint(0, 5, (8,), device="cuda"))
+
+
+@pytest.mark.parametrize("average", ["micro", "macro", "weighted"])
+def test_linear_stat_compute_matches_cpu(average):
+    """The one-launch compute kernel must match the torch reduce chain."""
+    torch.manual_seed(18)
+    preds = torch.randn(4096, 37)
+    target = torch.randint(0, 37, (4096,))
+    target[1000:] = torch.randint(0, 20, (3096,))  # leave some classes empty
+    makes = [
+        lambda: ma.MulticlassAccuracy(num_classes=37, average=average),
+        lambda: ma.MulticlassPrecision(num_classes=37, average=average),
+        lambda: ma.MulticlassRecall(num_classes=37, average=average),
+        lambda: ma.MulticlassF1Score(num_classes=37, average=average),
+        lambda: ma.MulticlassFBetaScore(num_classes=37, beta=2.0, average=average),
+        lambda: ma.MulticlassSpecificity(num_classes=37, average=average),
+        lambda: ma.MulticlassNegativePredictiveValue(num_classes=37, average=average),
+        lambda: ma.MulticlassHammingDistance(num_classes=37, average=average),
+    ]
+    for make in makes:
+        g, c = make().to("cuda"), make()
+        g.update(preds.cuda(), target.cuda())
+        c.update(preds, target)
+        rg, rc_ = g.compute(), c.compute()
+        assert rg.ndim == rc_.ndim
+        assert torch.allclose(rg.cpu(), rc_, atol=1e-6), (make().__class__.__name__, rg, rc_)
+
+
+def test_linear_stat_compute_functional_gpu():
+    from metrics_amd.functional import multiclass_precision
+
+    torch.manual_seed(19)
+    preds = torch.randn(1024, 10, device="cuda")
+    target = torch.randint(0, 10, (1024,), device="cuda")
+    v_gpu = multiclass_precision(preds, target, num_classes=10, average="macro")
+    v_cpu = multiclass_precision(preds.cpu(), target.cpu(), num_classes=10, average="macro")
+    assert torch.allclose(v_gpu.cpu(), v_cpu, atol=1e-6)
