@@ -696,6 +696,61 @@ __global__ void k_linear_stat_compute(
     }
 }
 
+// per-class AUROC / AveragePrecision straight from the thresholded curve
+// confmat state (T, C, 2, 2) — one block per class, threads stride the
+// threshold axis; replaces the ~25-launch torch chain (safe-divides, flips,
+// cats, trapz) at compute time. mode 0: AUROC (trapz of tpr over fpr, flipped
+// t), weights = support at the last threshold. mode 1: AP
+// (-sum((r[t+1]-r[t]) * p[t]) with p_T=1, r_T=0), weights = support at t=0.
+__global__ void k_curve_auc_from_confmat(
+    const ll* __restrict__ confmat /* (T, C, 2, 2) */, int T, ll C, int mode,
+    float* __restrict__ out /* (C,) */, float* __restrict__ weights /* (C,) nullable */) {
+    const ll c = blockIdx.x;
+    __shared__ double red[256];
+    auto cm = [&](int t, int i, int j) -> float {
+        return (float)confmat[(((ll)t * C + c) * 2 + i) * 2 + j];
+    };
+    auto sdiv = [](float n, float d) -> float { return d != 0.0f ? n / d : 0.0f; };
+    double acc = 0.0;
+    if (mode == 0) {
+        for (int t = threadIdx.x; t < T - 1; t += blockDim.x) {
+            const float tp0 = cm(t, 1, 1), fn0 = cm(t, 1, 0), fp0 = cm(t, 0, 1), tn0 = cm(t, 0, 0);
+            const float tp1 = cm(t + 1, 1, 1), fn1 = cm(t + 1, 1, 0), fp1 = cm(t + 1, 0, 1),
+                        tn1 = cm(t + 1, 0, 0);
+            const float tpr0 = sdiv(tp0, tp0 + fn0), tpr1 = sdiv(tp1, tp1 + fn1);
+            const float fpr0 = sdiv(fp0, fp0 + tn0), fpr1 = sdiv(fp1, fp1 + tn1);
+            acc += (double)((fpr0 - fpr1) * 0.5f * (tpr0 + tpr1));
+        }
+    } else {
+        for (int t = threadIdx.x; t < T; t += blockDim.x) {
+            const float tp0 = cm(t, 1, 1), fn0 = cm(t, 1, 0), fp0 = cm(t, 0, 1);
+            const float p0 = sdiv(tp0, tp0 + fp0);
+            const float r0 = sdiv(tp0, tp0 + fn0);
+            float r1;
+            if (t + 1 < T) {
+                const float tp1 = cm(t + 1, 1, 1), fn1 = cm(t + 1, 1, 0);
+                r1 = sdiv(tp1, tp1 + fn1);
+            } else {
+                r1 = 0.0f;  // appended endpoint (recall_T = 0)
+            }
+            acc += (double)(-(r1 - r0) * p0);
+        }
+    }
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        out[c] = (float)red[0];
+        if (weights) {
+            const int tw = mode == 0 ? T - 1 : 0;
+            weights[c] = cm(tw, 1, 1) + cm(tw, 1, 0);
+        }
+    }
+}
+
 // ---------------------------------------------------------------------------
 // K13: fused elementwise-error reductions, deterministic fp64 two-pass.
 // op: 0 = squared error, 1 = abs error, 2 = abs percentage |d|/max(|t|,eps),
@@ -995,6 +1050,14 @@ int ma_linear_stat_compute(uintptr_t stream, uintptr_t tp, uintptr_t fp, uintptr
                                             (const ll*)fn, C, n0, n1, n2, n3, d0, d1, d2, d3,
                                             avg_mode, zero_w_topk, zero_division, post_a, post_b,
                                             (float*)out);
+    return (int)hipGetLastError();
+}
+
+int ma_curve_auc_from_confmat(uintptr_t stream, uintptr_t confmat, int T, ll C, int mode,
+                               uintptr_t out, uintptr_t weights) {
+    hipStream_t s = (hipStream_t)stream;
+    k_curve_auc_from_confmat<<<(unsigned)C, 256, 0, s>>>((const ll*)confmat, T, C, mode,
+                                                         (float*)out, (float*)weights);
     return (int)hipGetLastError();
 }
 

@@ -136,6 +136,19 @@ def _multiclass_auroc_compute(
     average: Optional[str] = "macro",
     thresholds: Optional[Tensor] = None,
 ) -> Tensor:
+    if isinstance(state, Tensor) and thresholds is not None and state.is_cuda and state.ndim == 4:
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            # one kernel: per-class trapz over the thresholded ROC + supports
+            res, weights = _hip.curve_auc_from_confmat(state, mode=0)
+            if average in (None, "none"):
+                return res
+            if average == "macro":
+                return res.mean()
+            if average == "weighted":
+                w = _safe_divide(weights, weights.sum())
+                return (res * w).sum()
     fpr, tpr, _ = _multiclass_roc_compute(state, num_classes, thresholds)
     return _reduce_auroc(
         fpr,

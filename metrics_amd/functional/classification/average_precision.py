@@ -104,6 +104,18 @@ def _multiclass_average_precision_compute(
     average: Optional[str] = "macro",
     thresholds: Optional[Tensor] = None,
 ) -> Tensor:
+    if isinstance(state, Tensor) and thresholds is not None and state.is_cuda and state.ndim == 4:
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            res, weights = _hip.curve_auc_from_confmat(state, mode=1)
+            if average in (None, "none"):
+                return res
+            if average == "macro":
+                return res.mean()
+            if average == "weighted":
+                w = _safe_divide(weights, weights.sum())
+                return (res * w).sum()
     precision, recall, _ = _multiclass_precision_recall_curve_compute(state, num_classes, thresholds, average=None)
     return _reduce_average_precision(
         precision,

@@ -55,6 +55,7 @@ _SIGNATURES = {
     "ma_curve_suffix": [_U64, _U64, _LL, _I, _I, _I, _U64, _U64],
     "ma_apply_stat_deltas": [_U64, _U64, _LL, _U64, _U64, _U64, _U64],
     "ma_exact_apply": [_U64, _U64, _LL, _LL, _U64, _U64],
+    "ma_curve_auc_from_confmat": [_U64, _U64, _I, _LL, _I, _U64, _U64],
     "ma_linear_stat_compute": [_U64, _U64, _U64, _U64, _U64, _LL, _F, _F, _F, _F, _F, _F, _F, _F, _I, _I, _F, _F, _F, _U64],
     "ma_err_reduce": [_U64, _U64, _U64, _I, _LL, _I, _D, _U64, _I, _I, _U64],
     "ma_box_iou": [_U64, _U64, _LL, _U64, _LL, _I, _U64],
@@ -811,3 +812,18 @@ def linear_stat_compute(
     )
     _check(rc, "ma_linear_stat_compute")
     return out
+
+
+def curve_auc_from_confmat(confmat_state: Tensor, mode: int) -> Tuple[Tensor, Tensor]:
+    """Per-class AUROC (mode 0) / AveragePrecision (mode 1) + support weights
+    straight from the (T, C, 2, 2) thresholded confmat state — one launch."""
+    lib = _lib()
+    assert confmat_state.ndim == 4 and confmat_state.is_contiguous()
+    T, C = confmat_state.shape[0], confmat_state.shape[1]
+    out = torch.empty(C, dtype=torch.float32, device=confmat_state.device)
+    weights = torch.empty(C, dtype=torch.float32, device=confmat_state.device)
+    rc = lib.ma_curve_auc_from_confmat(
+        _stream(), confmat_state.data_ptr(), T, C, mode, out.data_ptr(), weights.data_ptr()
+    )
+    _check(rc, "ma_curve_auc_from_confmat")
+    return out, weights

@@ -479,3 +479,18 @@ def test_linear_stat_compute_functional_gpu():
     v_gpu = multiclass_precision(preds, target, num_classes=10, average="macro")
     v_cpu = multiclass_precision(preds.cpu(), target.cpu(), num_classes=10, average="macro")
     assert torch.allclose(v_gpu.cpu(), v_cpu, atol=1e-6)
+
+
+@pytest.mark.parametrize("average", ["macro", "weighted", "none"])
+def test_auroc_ap_compute_kernel(average):
+    torch.manual_seed(20)
+    preds = torch.randn(4096, 29).softmax(-1)
+    target = torch.randint(0, 29, (4096,))
+    for make in (
+        lambda: ma.MulticlassAUROC(num_classes=29, thresholds=75, average=average),
+        lambda: ma.MulticlassAveragePrecision(num_classes=29, thresholds=75, average=average),
+    ):
+        g, c = make().to("cuda"), make()
+        g.update(preds.cuda(), target.cuda())
+        c.update(preds, target)
+        assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-5), make().__class__.__name__
