@@ -513,6 +513,76 @@ __global__ void __launch_bounds__(256) k_multiclass_curve_hist(
     }
 }
 
+// LDS-privatized variant: each block owns a (class-chunk x T+1 x 2) uint32
+// histogram in LDS, loops a row range with (8 rows x 32 classes) thread tiles
+// (coalesced loads), and flushes nonzero LDS bins to global once. Wins when
+// the per-wave ballot-merge variant is atomic-bound (softmax skew piles most
+// of a class's mass into few buckets -> LDS same-address atomics are ~4x
+// cheaper than L2, and the flush is one atomic per NONZERO bin per block).
+template <typename T_, bool IS_BF16>
+__global__ void __launch_bounds__(256) k_multiclass_curve_hist_lds(
+    const T_* __restrict__ probs, const ll* __restrict__ target, ll B, ll C,
+    const float* __restrict__ thresholds, int T, ll ignore_index, int has_ignore, int mode,
+    int uniform, float t0, float inv_step, int c_chunk, int rows_per_block,
+    int norm_kind, const unsigned int* __restrict__ E, const float* __restrict__ rowmax,
+    const float* __restrict__ rowinv, unsigned long long* __restrict__ hist /* (C, T+1, 2) */) {
+    extern __shared__ unsigned int lds[];  // [c_chunk][(T+1)][2] then thresholds
+    const int bins_per_class = (T + 1) * 2;
+    const int lds_bins = c_chunk * bins_per_class;
+    float* sthr = (float*)&lds[lds_bins];
+    for (int b = threadIdx.x; b < lds_bins; b += blockDim.x) lds[b] = 0;
+    for (int b = threadIdx.x; b < T; b += blockDim.x) sthr[b] = thresholds[b];
+    __syncthreads();
+
+    const ll c_lo = (ll)blockIdx.x * c_chunk;
+    const ll c_hi = min(c_lo + (ll)c_chunk, C);
+    const ll r_lo = (ll)blockIdx.y * rows_per_block;
+    const ll r_hi = min(r_lo + (ll)rows_per_block, B);
+    const int cc = threadIdx.x & 31;           // class within chunk (32-wide)
+    const int rr = threadIdx.x >> 5;           // row within 8-row tile
+    const ll c = c_lo + cc;
+    const int norm = (norm_kind && E && E[0] == E[1] + 1u) ? norm_kind : 0;
+
+    for (ll row = r_lo + rr; row < r_hi; row += 8) {
+        ll trow = 0;
+        bool rvalid = true;
+        if (mode == 0) {
+            trow = target[row];
+            if (has_ignore && trow == ignore_index) rvalid = false;
+        }
+        if (rvalid && c < c_hi) {
+            int label;
+            if (mode == 0) {
+                label = (trow == c) ? 1 : 0;
+            } else {
+                ll t = target[row * C + c];
+                label = (t == 1) ? 1 : 0;
+                if (has_ignore && t == ignore_index) label = -1;
+            }
+            if (label >= 0) {
+                float p = IS_BF16
+                              ? bf16_to_f32(reinterpret_cast<const unsigned short*>(probs)[row * C + c])
+                              : (float)probs[row * C + c];
+                if (norm == 1) p = expf(p - rowmax[row]) * rowinv[row];
+                else if (norm == 2) p = 1.0f / (1.0f + expf(-p));
+                const int j = uniform ? bucket_of_uniform(p, sthr, T, t0, inv_step)
+                                      : bucket_of(p, sthr, T);
+                atomicAdd(&lds[cc * bins_per_class + j * 2 + label], 1u);
+            }
+        }
+    }
+    __syncthreads();
+    for (int b = threadIdx.x; b < lds_bins; b += blockDim.x) {
+        const unsigned int v = lds[b];
+        if (v) {
+            const int lc = b / bins_per_class;
+            const int bin = b - lc * bins_per_class;
+            const ll gc = c_lo + lc;
+            if (gc < C) atomicAdd(&hist[gc * (ll)(T + 1) * 2 + bin], (unsigned long long)v);
+        }
+    }
+}
+
 // suffix-sum the histogram into (/onto) the running confmat state:
 //   confmat[t][1][1] += sum_{j>t} hist[j][1]   (tp)
 //   confmat[t][0][1] += sum_{j>t} hist[j][0]   (fp)
@@ -977,7 +1047,7 @@ int ma_binary_curve_hist(uintptr_t stream, uintptr_t preds, int dtype, uintptr_t
 int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintptr_t target, ll B,
                              ll C, uintptr_t thresholds, int T, ll ignore_index, int has_ignore,
                              int mode, int uniform, float t0, float inv_step, int norm_kind,
-                             uintptr_t flag, uintptr_t rowmax, uintptr_t rowinv,
+                             uintptr_t flag, uintptr_t rowmax, uintptr_t rowinv, int variant,
                              uintptr_t hist) {
     hipStream_t s = (hipStream_t)stream;
     size_t shmem = (size_t)T * sizeof(float);
@@ -999,6 +1069,33 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
         else
             k_range_flag<unsigned short, true><<<grid_for(B * C, 256), 256, 0, s>>>(
                 (const unsigned short*)probs, B * C, (unsigned int*)flag);
+    }
+    // LDS-privatized variant when the per-block histogram fits comfortably
+    const int lds_c_chunk = 32;
+    const size_t lds_bytes = (size_t)lds_c_chunk * (T + 1) * 2 * sizeof(unsigned int)
+                             + (size_t)T * sizeof(float);
+    const int want_lds = variant == 1 || (variant < 0 && lds_bytes <= 96 * 1024 && B >= 1024);
+    if (want_lds && lds_bytes <= 160 * 1024) {
+        ll c_chunks_l = (C + lds_c_chunk - 1) / lds_c_chunk;
+        ll row_chunks_l = 1536 / (c_chunks_l > 0 ? c_chunks_l : 1);
+        if (row_chunks_l < 1) row_chunks_l = 1;
+        ll max_rc = (B + 255) / 256;
+        if (row_chunks_l > max_rc) row_chunks_l = max_rc;
+        int rows_per_block = (int)((B + row_chunks_l - 1) / row_chunks_l);
+        dim3 gridl((unsigned)c_chunks_l, (unsigned)row_chunks_l);
+        if (dtype == 0)
+            k_multiclass_curve_hist_lds<float, false><<<gridl, 256, lds_bytes, s>>>(
+                (const float*)probs, (const ll*)target, B, C, (const float*)thresholds, T,
+                ignore_index, has_ignore, mode, uniform, t0, inv_step, lds_c_chunk,
+                rows_per_block, norm_kind, (const unsigned int*)flag, (const float*)rowmax,
+                (const float*)rowinv, (unsigned long long*)hist);
+        else
+            k_multiclass_curve_hist_lds<unsigned short, true><<<gridl, 256, lds_bytes, s>>>(
+                (const unsigned short*)probs, (const ll*)target, B, C, (const float*)thresholds,
+                T, ignore_index, has_ignore, mode, uniform, t0, inv_step, lds_c_chunk,
+                rows_per_block, norm_kind, (const unsigned int*)flag, (const float*)rowmax,
+                (const float*)rowinv, (unsigned long long*)hist);
+        return (int)hipGetLastError();
     }
     ll row_chunks = (B + 255) / 256;
     // pick the class chunk so the grid comfortably overfills 256 CUs
