@@ -212,6 +212,18 @@ def _multilabel_auroc_compute(
             target = target[idx]
         return _binary_auroc_compute((preds, target), thresholds, max_fpr=None)
 
+    if isinstance(state, Tensor) and thresholds is not None and state.is_cuda and state.ndim == 4:
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            res, weights = _hip.curve_auc_from_confmat(state, mode=0)
+            if average in (None, "none"):
+                return res
+            if average == "macro":
+                return res.mean()
+            if average == "weighted":
+                w = _safe_divide(weights, weights.sum())
+                return (res * w).sum()
     fpr, tpr, _ = _multilabel_roc_compute(state, num_labels, thresholds, ignore_index)
     return _reduce_auroc(
         fpr,

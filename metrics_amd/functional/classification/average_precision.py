@@ -180,6 +180,18 @@ def _multilabel_average_precision_compute(
             target = target[idx]
         return _binary_average_precision_compute((preds, target), thresholds)
 
+    if isinstance(state, Tensor) and thresholds is not None and state.is_cuda and state.ndim == 4:
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            res, weights = _hip.curve_auc_from_confmat(state, mode=1)
+            if average in (None, "none"):
+                return res
+            if average == "macro":
+                return res.mean()
+            if average == "weighted":
+                w = _safe_divide(weights, weights.sum())
+                return (res * w).sum()
     precision, recall, _ = _multilabel_precision_recall_curve_compute(state, num_labels, thresholds, ignore_index)
     return _reduce_average_precision(
         precision,

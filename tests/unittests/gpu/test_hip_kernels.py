@@ -494,3 +494,18 @@ def test_auroc_ap_compute_kernel(average):
         g.update(preds.cuda(), target.cuda())
         c.update(preds, target)
         assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-5), make().__class__.__name__
+
+
+@pytest.mark.parametrize("average", ["macro", "weighted"])
+def test_multilabel_auroc_ap_compute_kernel(average):
+    torch.manual_seed(21)
+    preds = torch.rand(4096, 14)
+    target = torch.randint(0, 2, (4096, 14))
+    for make in (
+        lambda: ma.MultilabelAUROC(num_labels=14, thresholds=60, average=average),
+        lambda: ma.MultilabelAveragePrecision(num_labels=14, thresholds=60, average=average),
+    ):
+        g, c = make().to("cuda"), make()
+        g.update(preds.cuda(), target.cuda())
+        c.update(preds, target)
+        assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-5), make().__class__.__name__
