@@ -25,6 +25,28 @@ def _cohen_kappa_reduce(confmat: Tensor, weights: Optional[str] = None) -> Tenso
     num_classes = confmat.shape[0]
     sum0 = confmat.sum(dim=0, keepdim=True)
     sum1 = confmat.sum(dim=1, keepdim=True)
+
+    if weights in (None, "quadratic") and confmat.is_cuda:
+        # separable identities avoid materializing the (C,C) weight and
+        # expected matrices (the (C,C) outer product dominated compute time):
+        #   None:      sum(w*cm) = s - trace;  sum(w*E) = s - sum(tk*pk)/s
+        #   quadratic: w_ij=(i-j)^2 = i^2 - 2ij + j^2 (separable in i, j)
+        tk = sum1.flatten()
+        pk = sum0.flatten()
+        s_total = pk.sum()
+        if weights is None:
+            obs = s_total - torch.diagonal(confmat).sum()
+            exp = s_total - (tk * pk).sum() / s_total
+        else:
+            idx = torch.arange(num_classes, dtype=confmat.dtype, device=confmat.device)
+            i2cm_rows = (confmat * (idx**2).unsqueeze(1)).sum()  # sum i^2 cm_ij
+            j2cm_cols = (confmat * (idx**2).unsqueeze(0)).sum()
+            ijcm = (idx.unsqueeze(1) * idx.unsqueeze(0) * confmat).sum()
+            obs = i2cm_rows + j2cm_cols - 2 * ijcm
+            # sum(w*E) with E = outer(tk,pk)/s and w_ij = i^2 - 2ij + j^2:
+            exp = (idx**2 * tk).sum() + (idx**2 * pk).sum() - 2 * (idx * tk).sum() * (idx * pk).sum() / s_total
+        return 1 - obs / exp
+
     expected = sum1 @ sum0 / sum0.sum()  # outer product
 
     if weights is None:

@@ -40,6 +40,15 @@ def _matthews_corrcoef_reduce(confmat: Tensor) -> Tensor:
     numerator = cov_ytyp
     denom = cov_ypyp * cov_ytyt
 
+    if confmat.is_cuda and confmat.numel() != 4:
+        # branchless: the `denom == 0` python bool forces a device->host sync
+        # (~0.1 ms); the degenerate-binary special case below can't apply here
+        return torch.where(
+            denom == 0,
+            torch.zeros((), dtype=torch.float32, device=confmat.device),
+            numerator / denom.clamp(min=torch.finfo(torch.float32).tiny).sqrt(),
+        )
+
     if denom == 0 and confmat.numel() == 4:
         # degenerate binary cases: perfect or perfectly-wrong single-class preds
         if c == 0 or c == s:

@@ -509,3 +509,20 @@ def test_multilabel_auroc_ap_compute_kernel(average):
         g.update(preds.cuda(), target.cuda())
         c.update(preds, target)
         assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-5), make().__class__.__name__
+
+
+def test_mcc_kappa_compute_fast_paths_gpu():
+    torch.manual_seed(22)
+    preds = torch.randn(4096, 100)
+    target = torch.randint(0, 100, (4096,))
+    for make in (
+        lambda: ma.MulticlassMatthewsCorrCoef(num_classes=100),
+        lambda: ma.MulticlassCohenKappa(num_classes=100),
+        lambda: ma.MulticlassCohenKappa(num_classes=100, weights="quadratic"),
+        lambda: ma.MulticlassCohenKappa(num_classes=100, weights="linear"),
+        lambda: ma.MulticlassJaccardIndex(num_classes=100),
+    ):
+        g, c = make().to("cuda"), make()
+        g.update(preds.cuda(), target.cuda())
+        c.update(preds, target)
+        assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-5), make().__class__.__name__
