@@ -526,3 +526,64 @@ def test_mcc_kappa_compute_fast_paths_gpu():
         g.update(preds.cuda(), target.cuda())
         c.update(preds, target)
         assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-5), make().__class__.__name__
+
+
+def test_device_sweep_all_domains():
+    """Representative metrics from every domain run update+compute on device."""
+    torch.manual_seed(23)
+    dev = "cuda"
+    n = 256
+    # classification
+    ma.BinaryF1Score().to(dev)(torch.rand(n, device=dev), torch.randint(0, 2, (n,), device=dev))
+    ma.MulticlassCalibrationError(num_classes=5).to(dev)(
+        torch.randn(n, 5, device=dev).softmax(-1), torch.randint(0, 5, (n,), device=dev))
+    ma.MultilabelRankingAveragePrecision(num_labels=6).to(dev)(
+        torch.rand(n, 6, device=dev), torch.randint(0, 2, (n, 6), device=dev))
+    # regression
+    ma.PearsonCorrCoef().to(dev)(torch.randn(n, device=dev), torch.randn(n, device=dev))
+    ma.SpearmanCorrCoef().to(dev)(torch.randn(n, device=dev), torch.randn(n, device=dev))
+    ma.KendallRankCorrCoef().to(dev)(torch.randn(n, device=dev), torch.randn(n, device=dev))
+    ma.R2Score().to(dev)(torch.randn(n, device=dev), torch.randn(n, device=dev))
+    ma.MeanSquaredLogError().to(dev)(torch.rand(n, device=dev), torch.rand(n, device=dev))
+    # retrieval
+    idx = torch.randint(0, 8, (n,), device=dev)
+    ma.RetrievalMRR().to(dev)(torch.rand(n, device=dev), torch.randint(0, 2, (n,), device=dev), indexes=idx)
+    ma.RetrievalNormalizedDCG().to(dev)(torch.rand(n, device=dev), torch.randint(0, 2, (n,), device=dev), indexes=idx)
+    # clustering
+    ma.clustering.NormalizedMutualInfoScore().to(dev)(
+        torch.randint(0, 5, (n,), device=dev), torch.randint(0, 5, (n,), device=dev))
+    ma.clustering.CalinskiHarabaszScore().to(dev)(torch.randn(n, 4, device=dev), torch.randint(0, 3, (n,), device=dev))
+    # nominal
+    ma.CramersV(num_classes=5).to(dev)(torch.randint(0, 5, (n,), device=dev), torch.randint(0, 5, (n,), device=dev))
+    # segmentation
+    ma.segmentation.MeanIoU(num_classes=4, input_format="index").to(dev)(
+        torch.randint(0, 4, (2, 16, 16), device=dev), torch.randint(0, 4, (2, 16, 16), device=dev))
+    ma.segmentation.GeneralizedDiceScore(num_classes=4, input_format="index").to(dev)(
+        torch.randint(0, 4, (2, 16, 16), device=dev), torch.randint(0, 4, (2, 16, 16), device=dev))
+    # detection
+    ma.detection.IntersectionOverUnion().to(dev)(
+        [dict(boxes=torch.tensor([[0.0, 0, 10, 10]], device=dev), scores=torch.tensor([0.9], device=dev),
+              labels=torch.tensor([0], device=dev))],
+        [dict(boxes=torch.tensor([[2.0, 2, 12, 12]], device=dev), labels=torch.tensor([0], device=dev))])
+    m = ma.detection.MeanAveragePrecision().to(dev)
+    m.update(
+        [dict(boxes=torch.tensor([[0.0, 0, 10, 10]], device=dev), scores=torch.tensor([0.9], device=dev),
+              labels=torch.tensor([0], device=dev))],
+        [dict(boxes=torch.tensor([[0.0, 0, 10, 10]], device=dev), labels=torch.tensor([0], device=dev))])
+    assert abs(float(m.compute()["map"]) - 1.0) < 1e-6
+    # image
+    ma.StructuralSimilarityIndexMeasure(data_range=1.0).to(dev)(
+        torch.rand(1, 3, 32, 32, device=dev), torch.rand(1, 3, 32, 32, device=dev))
+    ma.PeakSignalNoiseRatio(data_range=1.0).to(dev)(
+        torch.rand(1, 3, 32, 32, device=dev), torch.rand(1, 3, 32, 32, device=dev))
+    # audio
+    ma.audio.ScaleInvariantSignalDistortionRatio().to(dev)(
+        torch.randn(2, 8000, device=dev), torch.randn(2, 8000, device=dev))
+    ma.audio.SignalDistortionRatio().to(dev)(torch.randn(1, 4000, device=dev), torch.randn(1, 4000, device=dev))
+    # aggregation + wrappers
+    agg = ma.MeanMetric().to(dev)
+    agg.update(torch.rand(n, device=dev))
+    assert agg.compute().is_cuda
+    boot = ma.wrappers.BootStrapper(ma.BinaryAccuracy(), num_bootstraps=4).to(dev)
+    boot.update(torch.rand(n, device=dev), torch.randint(0, 2, (n,), device=dev))
+    boot.compute()
