@@ -108,20 +108,21 @@ def plot_confusion_matrix(
 ):
     """Heatmap plot of a (C,C) or (N,2,2) confusion matrix."""
     _error_on_missing_matplotlib()
-    if confmat.ndim == 3:  # multilabel
+    multilabel = confmat.ndim == 3
+    if multilabel:
         nb, n_classes = confmat.shape[0], 2
         rows, cols = _get_col_row_split(nb)
     else:
         nb, n_classes, rows, cols = 1, confmat.shape[0], 1, 1
         confmat = confmat[None]
 
-    if labels is not None and confmat.ndim != 3 and len(labels) != n_classes:
+    if labels is not None and not multilabel and len(labels) != n_classes:
         raise ValueError(
             "Expected number of elements in arg `labels` to match number of labels in confmat but "
             f"got {len(labels)} and {n_classes}"
         )
 
-    if confmat.ndim == 3:
+    if multilabel:
         fig_label = labels or np_arange(nb)
         labels = [0, 1]
     else:
@@ -138,7 +139,7 @@ def plot_confusion_matrix(
         ax_ = axs[i] if nb > 1 else axs[0]
         if fig_label is not None:
             ax_.set_title(f"Label {fig_label[i]}", fontsize=15)
-        ax_.imshow(confmat[i].cpu().detach() if confmat.ndim == 3 else confmat[0].cpu().detach(), cmap=cmap)
+        ax_.imshow(confmat[i].cpu().detach(), cmap=cmap)
         ax_.set_xlabel("Predicted class", fontsize=15)
         ax_.set_ylabel("True class", fontsize=15)
         ax_.set_xticks(list(range(n_classes)))
@@ -148,7 +149,7 @@ def plot_confusion_matrix(
 
         if add_text:
             for ii, jj in product(range(n_classes), range(n_classes)):
-                val = confmat[i, ii, jj] if confmat.ndim == 3 else confmat[0, ii, jj]
+                val = confmat[i, ii, jj]
                 ax_.text(jj, ii, str(round(val.item(), 2)), ha="center", va="center", fontsize=15)
 
     return fig, axs

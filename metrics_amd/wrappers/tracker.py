@@ -90,6 +90,22 @@ class MetricTracker(ModuleList):
         for metric in self:
             metric.reset()
 
+    def plot(self, val=None, ax=None):
+        """Plot the tracked metric values over increments (reference wrappers/tracker.py:305)."""
+        from metrics_amd.utilities.plot import plot_single_or_multi_val
+
+        val = val if val is not None else self.compute_all()
+        if isinstance(val, dict):
+            import matplotlib.pyplot as plt
+
+            fig, ax_ = (plt.subplots() if ax is None else (None, ax))
+            for k, v in val.items():
+                plot_single_or_multi_val(list(v) if v.ndim else v, ax=ax_)
+            return fig, ax_
+        if isinstance(val, (list, tuple)):
+            return plot_single_or_multi_val(list(val), ax=ax)
+        return plot_single_or_multi_val(list(val.flatten()) if val.ndim else val, ax=ax)
+
     def best_metric(
         self, return_step: bool = False
     ) -> Union[
