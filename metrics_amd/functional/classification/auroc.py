@@ -75,6 +75,15 @@ def _binary_auroc_compute(
     max_fpr: Optional[float] = None,
     pos_label: int = 1,
 ) -> Tensor:
+    if (
+        max_fpr is None and isinstance(state, Tensor) and thresholds is not None
+        and state.is_cuda and state.ndim == 3
+    ):
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            res, _ = _hip.curve_auc_from_confmat(state.unsqueeze(1), mode=0)
+            return res.reshape(())
     fpr, tpr, _ = _binary_roc_compute(state, thresholds, pos_label)
     if max_fpr is None or max_fpr == 1 or fpr.sum() == 0 or tpr.sum() == 0:
         return _auc_compute_without_check(fpr, tpr, 1.0)

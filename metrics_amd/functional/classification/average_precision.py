@@ -61,6 +61,12 @@ def _binary_average_precision_compute(
     thresholds: Optional[Tensor],
     pos_label: int = 1,
 ) -> Tensor:
+    if isinstance(state, Tensor) and thresholds is not None and state.is_cuda and state.ndim == 3:
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            res, _ = _hip.curve_auc_from_confmat(state.unsqueeze(1), mode=1)
+            return res.reshape(())
     precision, recall, _ = _binary_precision_recall_curve_compute(state, thresholds, pos_label)
     return -torch.sum((recall[1:] - recall[:-1]) * precision[:-1])
 

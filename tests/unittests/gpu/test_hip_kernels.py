@@ -587,3 +587,25 @@ def test_device_sweep_all_domains():
     boot = ma.wrappers.BootStrapper(ma.BinaryAccuracy(), num_bootstraps=4).to(dev)
     boot.update(torch.rand(n, device=dev), torch.randint(0, 2, (n,), device=dev))
     boot.compute()
+
+
+def test_binary_auroc_ap_compute_kernel():
+    torch.manual_seed(24)
+    preds = torch.rand(20_000)
+    target = torch.randint(0, 2, (20_000,))
+    for make in (
+        lambda: ma.BinaryAUROC(thresholds=120),
+        lambda: ma.BinaryAveragePrecision(thresholds=120),
+    ):
+        g, c = make().to("cuda"), make()
+        g.update(preds.cuda(), target.cuda())
+        c.update(preds, target)
+        rg, rc_ = g.compute(), c.compute()
+        assert rg.ndim == 0
+        assert torch.allclose(rg.cpu(), rc_, atol=1e-6), make().__class__.__name__
+    # max_fpr path must still use the torch chain and agree with CPU
+    g = ma.BinaryAUROC(thresholds=120, max_fpr=0.5).to("cuda")
+    c = ma.BinaryAUROC(thresholds=120, max_fpr=0.5)
+    g.update(preds.cuda(), target.cuda())
+    c.update(preds, target)
+    assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-6)
