@@ -163,7 +163,9 @@ def chrf_score(
 
     for p, refs in zip(preds_, target_):
         best_f = tensor(0.0)
-        best_stats = None
+        # reference quirk (chrf.py:358): strictly-better wins; if every ref
+        # scores 0 the zero-init stats are kept (no target counts accumulate)
+        best_stats = (torch.zeros(total_orders), torch.zeros(total_orders), torch.zeros(total_orders))
         for ref in refs:
             tp = torch.zeros(total_orders)
             fp = torch.zeros(total_orders)
@@ -186,7 +188,7 @@ def chrf_score(
                 fp[i] = sum(png.values()) - overlap
                 fn[i] = sum(rng.values()) - overlap
             f = _chrf_f_score(tp, fp, fn, beta)
-            if best_stats is None or f >= best_f:
+            if f > best_f:
                 best_f = f
                 best_stats = (tp, fp, fn)
         total_tp += best_stats[0]
@@ -201,18 +203,15 @@ def chrf_score(
 
 
 def _chrf_f_score(tp: Tensor, fp: Tensor, fn: Tensor, beta: float) -> Tensor:
-    eps = 1e-16
-    precision = tp / (tp + fp + eps)
-    recall = tp / (tp + fn + eps)
-    # average over orders with any n-grams
-    valid = (tp + fp + fn) > 0
-    if valid.sum() == 0:
-        return tensor(0.0)
-    p = precision[valid].mean()
-    r = recall[valid].mean()
-    if p == 0 and r == 0:
-        return tensor(0.0)
-    return (1 + beta**2) * p * r / (beta**2 * p + r)
+    """Mean over ALL n-gram orders of per-order F_beta (reference chrf.py:264-285)."""
+    hyp = tp + fp
+    ref = tp + fn
+    zero = torch.zeros_like(tp)
+    precision = torch.where(hyp > 0, tp / torch.clamp(hyp, min=1.0), zero)
+    recall = torch.where(ref > 0, tp / torch.clamp(ref, min=1.0), zero)
+    den = torch.clamp(beta**2 * precision + recall, min=1e-16)
+    f = (1 + beta**2) * precision * recall / den
+    return f.mean()
 
 
 # ------------------------------------------------------------------------- TER

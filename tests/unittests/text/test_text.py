@@ -125,3 +125,25 @@ def test_eed_reference_parity_values():
     )
     assert abs(float(v) - 0.42880797) < 1e-6
     assert abs(float(extended_edit_distance(["exact match"], ["exact match"])) - 0.0226) < 1e-3
+
+
+def test_text_reference_doctest_values():
+    """Values pinned to the reference's doctests (same inputs)."""
+    from metrics_amd.functional.text import (
+        bleu_score, sacre_bleu_score, chrf_score, word_error_rate, match_error_rate,
+        word_information_lost, word_information_preserved, rouge_score, translation_edit_rate,
+    )
+
+    preds = ["the cat is on the mat"]
+    target = [["there is a cat on the mat", "a cat is on the mat"]]
+    assert abs(float(bleu_score(preds, target)) - 0.7598) < 5e-4
+    assert abs(float(sacre_bleu_score(preds, target)) - 0.7598) < 5e-4
+    assert abs(float(chrf_score(preds, target)) - 0.8640) < 5e-4
+    p2 = ["this is the prediction", "there is an other sample"]
+    t2 = ["this is the reference", "there is another one"]
+    assert abs(float(word_error_rate(p2, t2)) - 0.5) < 1e-6
+    assert abs(float(match_error_rate(p2, t2)) - 0.4444) < 5e-4
+    assert abs(float(word_information_lost(p2, t2)) - 0.6528) < 5e-4
+    assert abs(float(word_information_preserved(p2, t2)) - 0.3472) < 5e-4
+    r = rouge_score(["My name is John"], ["Is your name John"])
+    assert abs(float(r["rouge1_fmeasure"]) - 0.75) < 1e-4
