@@ -245,3 +245,36 @@ def test_pearson_ddp_merge():
         assert abs(v.item() - ref) < 1e-4, (v.item(), ref)
 
     run_distributed(_worker, world_size=2)
+
+
+def test_reference_doctest_values_cross_domain():
+    """Values pinned to reference doctests (same operand order)."""
+    import metrics_amd as ma
+    from metrics_amd import functional as F
+
+    idx = torch.tensor([0, 0, 0, 1, 1, 1, 1])
+    preds = torch.tensor([0.2, 0.3, 0.5, 0.1, 0.3, 0.5, 0.2])
+    target = torch.tensor([False, False, True, False, True, False, True])
+    m = ma.RetrievalMRR(); m.update(preds, target, indexes=idx)
+    assert abs(float(m.compute()) - 0.75) < 1e-4
+    m = ma.RetrievalMAP(); m.update(preds, target, indexes=idx)
+    assert abs(float(m.compute()) - 0.7917) < 1e-4
+
+    assert abs(float(F.cosine_similarity(
+        torch.tensor([[0., 1.], [1., 1.]]), torch.tensor([[0., 1.], [0., 1.]]), reduction="mean")) - 0.8536) < 1e-4
+    # reference doc: preds=[2.5,0,2,8], target=[3,-0.5,2,7] -> 0.9572
+    assert abs(float(F.explained_variance(
+        torch.tensor([2.5, 0.0, 2, 8]), torch.tensor([3., -0.5, 2, 7]))) - 0.9572) < 1e-4
+    assert abs(float(F.kendall_rank_corrcoef(
+        torch.tensor([2., 7, 20, 200]), torch.tensor([0.3, 0.2, 0.6, 0.1]))) + 0.3333) < 1e-4
+    assert abs(float(F.tweedie_deviance_score(
+        torch.tensor([4.0, 3.0, 2.0, 1.0]), torch.tensor([1.0, 2.0, 3.0, 4.0]), power=2)) - 1.2083) < 1e-4
+    assert abs(float(F.mean_squared_log_error(
+        torch.tensor([0., 1, 2, 3]), torch.tensor([0., 1, 2, 2]))) - 0.0207) < 1e-4
+    assert abs(float(F.symmetric_mean_absolute_percentage_error(
+        torch.tensor([1., 10, 1e6]), torch.tensor([0.9, 15, 1.2e6]))) - 0.2290) < 1e-4
+
+    m = ma.RunningMean(window=3)
+    for v in [1.0, 2.0, 3.0, 4.0]:
+        m.update(torch.tensor(v))
+    assert abs(float(m.compute()) - 3.0) < 1e-6
