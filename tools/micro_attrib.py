@@ -7,12 +7,6 @@ coll = build_collection(1000, torch.device("cuda"))
 preds = torch.randn(8192, 1000, device="cuda", dtype=torch.bfloat16)
 tgt = torch.randint(0, 1000, (8192,), device="cuda")
 
-# leaders only (what actually runs)
-import metrics_amd.collections as mc
-for name, m in coll.items(keep_base=True, copy_state=False):
-    if not getattr(m, "_computed_as_group_member", False):
-        pass
-
 for name, m in list(coll.items(keep_base=True, copy_state=False)):
     m.update(preds, tgt)  # warm
 torch.cuda.synchronize()
