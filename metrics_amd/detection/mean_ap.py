@@ -537,37 +537,68 @@ class MeanAveragePrecision(Metric):
                     dt_off, gt_off, area_rngs, iou_thrs,
                 )
 
-            for ai in range(A):
-                npig = int(npig_a[ai])
-                if npig == 0:
+            # fully batched accumulation over (area-range, max-det): one score
+            # sort per class; smaller max_det caps are masks on the sorted
+            # arrays — a masked det freezes the cumsums, which plateaus rc/pr
+            # and leaves the interpolated precision unchanged (COCO-equivalent)
+            order = np.argsort(-scores_k, kind="stable")
+            scores_sorted = scores_k[order]
+            rank_sorted = rank_k[order]
+            eps = np.finfo(np.float64).eps
+            n_cols = order.shape[0]
+            valid_ai = npig_a > 0
+            if n_cols == 0 or not valid_ai.any():
+                for ai in range(A):
+                    if npig_a[ai] > 0:
+                        recall[:, ki, ai, :] = 0.0
+                        precision[:, :, ki, ai, :] = 0.0
+                continue
+            m4_all = np.stack([dtm_a[ai][:, order] for ai in range(A)])  # (A,T,n)
+            i4_all = np.stack([dti_a[ai][:, order] for ai in range(A)])
+            npig_safe = np.where(valid_ai, npig_a, 1).astype(np.float64)
+            for mi, max_det in enumerate(max_dets):
+                # compress to the capped dets (maxdet=1 keeps ~n_imgs entries)
+                if max_det >= max_det_top:
+                    m4, i4, sc = m4_all, i4_all, scores_sorted
+                else:
+                    keep = rank_sorted < max_det
+                    m4, i4, sc = m4_all[..., keep], i4_all[..., keep], scores_sorted[keep]
+                nc = m4.shape[-1]
+                if nc == 0:
+                    for ai in range(A):
+                        if valid_ai[ai]:
+                            recall[:, ki, ai, mi] = 0.0
+                            precision[:, :, ki, ai, mi] = 0.0
                     continue
-                for mi, max_det in enumerate(max_dets):
-                    keep = rank_k < max_det
-                    scores = scores_k[keep]
-                    m_k = dtm_a[ai][:, keep]
-                    i_k = dti_a[ai][:, keep]
-                    order = np.argsort(-scores, kind="stable")
-                    scores_sorted = scores[order]
-                    m_s = m_k[:, order]
-                    i_s = i_k[:, order]
-                    tps = np.cumsum(m_s & ~i_s, axis=1, dtype=np.float64)
-                    fps = np.cumsum(~m_s & ~i_s, axis=1, dtype=np.float64)
-                    rc = tps / npig
-                    pr = tps / (tps + fps + np.finfo(np.float64).eps)
-                    if rc.shape[1]:
-                        recall[:, ki, ai, mi] = rc[:, -1]
-                    else:
-                        recall[:, ki, ai, mi] = 0.0
-                    pr_env = np.maximum.accumulate(pr[:, ::-1], axis=1)[:, ::-1]
-                    for ti in range(T):
-                        inds = np.searchsorted(rc[ti], rec_thrs, side="left")
-                        valid = inds < pr_env.shape[1]
-                        q = np.zeros(R)
-                        ss = np.zeros(R)
-                        q[valid] = pr_env[ti][inds[valid]]
-                        ss[valid] = scores_sorted[inds[valid]]
-                        precision[ti, :, ki, ai, mi] = q
-                        scores_out[ti, :, ki, ai, mi] = ss
+                scored = ~i4
+                tps = np.cumsum(m4 & scored, axis=-1, dtype=np.float64)  # (A,T,nc)
+                fps = np.cumsum(~m4 & scored, axis=-1, dtype=np.float64)
+                rc = tps / npig_safe[:, None, None]
+                pr = tps / (tps + fps + eps)
+                pr_env = np.maximum.accumulate(pr[..., ::-1], axis=-1)[..., ::-1]
+                # one searchsorted for ALL (A,T) rows: offset each row by 2*row
+                # so the (ascending, in [0,1]) rows concatenate globally ascending
+                n_rows = A * T
+                row_off = 2.0 * np.arange(n_rows, dtype=np.float64)
+                big = (rc.reshape(n_rows, nc) + row_off[:, None]).ravel()
+                queries = (rec_thrs[None, :] + row_off[:, None]).ravel()
+                inds = np.searchsorted(big, queries, side="left")
+                local = inds - np.repeat(np.arange(n_rows) * nc, R)
+                valid = local < nc
+                q = np.where(valid, pr_env.reshape(-1)[np.minimum(inds, n_rows * nc - 1)], 0.0)
+                q3 = q.reshape(A, T, R)
+                rec3 = rc[..., -1]  # (A,T)
+                for ai in range(A):
+                    if not valid_ai[ai]:
+                        continue
+                    precision[:, :, ki, ai, mi] = q3[ai]
+                    recall[:, ki, ai, mi] = rec3[ai]
+                if self.extended_summary:
+                    ss = np.where(valid, sc[np.minimum(local, nc - 1)], 0.0)
+                    ss3 = ss.reshape(A, T, R)
+                    for ai in range(A):
+                        if valid_ai[ai]:
+                            scores_out[:, :, ki, ai, mi] = ss3[ai]
 
         def _summarize(ap: bool, iou_thr: Optional[float] = None, area: int = 0, max_det_idx: int = -1) -> Tensor:
             if ap:
