@@ -492,11 +492,11 @@ class MeanAveragePrecision(Metric):
 
         area_rngs = np.array(self._AREA_RANGES, dtype=np.float32)
 
-        for ki in range(K):
+        def _process_class(ki: int) -> None:
             dlo, dhi = int(d_cls_off[ki]), int(d_cls_off[ki + 1])
             glo, ghi = int(g_cls_off[ki]), int(g_cls_off[ki + 1])
             if dhi == dlo and ghi == glo:
-                continue
+                return
             dt_off = np.concatenate([[0], np.cumsum(d_counts[ki])]).astype(np.int64)
             gt_off = np.concatenate([[0], np.cumsum(g_counts[ki])]).astype(np.int64)
             scores_k = all_ds[dlo:dhi]
@@ -552,7 +552,7 @@ class MeanAveragePrecision(Metric):
                     if npig_a[ai] > 0:
                         recall[:, ki, ai, :] = 0.0
                         precision[:, :, ki, ai, :] = 0.0
-                continue
+                return
             m4_all = np.stack([dtm_a[ai][:, order] for ai in range(A)])  # (A,T,n)
             i4_all = np.stack([dti_a[ai][:, order] for ai in range(A)])
             npig_safe = np.where(valid_ai, npig_a, 1).astype(np.float64)
@@ -599,6 +599,9 @@ class MeanAveragePrecision(Metric):
                     for ai in range(A):
                         if valid_ai[ai]:
                             scores_out[:, :, ki, ai, mi] = ss3[ai]
+
+        for ki in range(K):
+            _process_class(ki)
 
         def _summarize(ap: bool, iou_thr: Optional[float] = None, area: int = 0, max_det_idx: int = -1) -> Tensor:
             if ap:
