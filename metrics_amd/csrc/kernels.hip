@@ -1048,7 +1048,7 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
                              ll C, uintptr_t thresholds, int T, ll ignore_index, int has_ignore,
                              int mode, int uniform, float t0, float inv_step, int norm_kind,
                              uintptr_t flag, uintptr_t rowmax, uintptr_t rowinv, int variant,
-                             uintptr_t hist) {
+                             int c_chunk_override, uintptr_t hist) {
     hipStream_t s = (hipStream_t)stream;
     size_t shmem = (size_t)T * sizeof(float);
     if (shmem > 160 * 1024) return -100;
@@ -1071,7 +1071,7 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
                 (const unsigned short*)probs, B * C, (unsigned int*)flag);
     }
     // LDS-privatized variant when the per-block histogram fits comfortably
-    const int lds_c_chunk = 32;
+    const int lds_c_chunk = c_chunk_override > 0 ? c_chunk_override : 32;
     const size_t lds_bytes = (size_t)lds_c_chunk * (T + 1) * 2 * sizeof(unsigned int)
                              + (size_t)T * sizeof(float);
     const int want_lds = variant == 1 || (variant < 0 && lds_bytes <= 96 * 1024 && B >= 1024);
@@ -1099,8 +1099,8 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
     }
     ll row_chunks = (B + 255) / 256;
     // pick the class chunk so the grid comfortably overfills 256 CUs
-    int c_chunk = 32;
-    while (c_chunk > 1 && ((C + c_chunk - 1) / c_chunk) * row_chunks < 1024) c_chunk /= 2;
+    int c_chunk = c_chunk_override > 0 ? c_chunk_override : 32;
+    while (c_chunk_override <= 0 && c_chunk > 1 && ((C + c_chunk - 1) / c_chunk) * row_chunks < 1024) c_chunk /= 2;
     ll c_chunks = (C + c_chunk - 1) / c_chunk;
     if (row_chunks > 65535 || c_chunks > 2147483647LL) return -101;
     dim3 grid((unsigned)c_chunks, (unsigned)row_chunks);

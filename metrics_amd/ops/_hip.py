@@ -52,7 +52,7 @@ _SIGNATURES = {
     "ma_binary_stat": [_U64, _U64, _I, _U64, _LL, _F, _LL, _I, _U64, _U64],
     "ma_multilabel_stat": [_U64, _U64, _I, _U64, _LL, _LL, _F, _LL, _I, _U64, _U64],
     "ma_binary_curve_hist": [_U64, _U64, _I, _U64, _LL, _U64, _I, _LL, _I, _I, _F, _F, _I, _U64, _U64],
-    "ma_multiclass_curve_hist": [_U64, _U64, _I, _U64, _LL, _LL, _U64, _I, _LL, _I, _I, _I, _F, _F, _I, _U64, _U64, _U64, _I, _U64],
+    "ma_multiclass_curve_hist": [_U64, _U64, _I, _U64, _LL, _LL, _U64, _I, _LL, _I, _I, _I, _F, _F, _I, _U64, _U64, _U64, _I, _I, _U64],
     "ma_curve_suffix": [_U64, _U64, _LL, _I, _I, _I, _U64, _U64],
     "ma_apply_stat_deltas": [_U64, _U64, _LL, _U64, _U64, _U64, _U64],
     "ma_exact_apply": [_U64, _U64, _LL, _LL, _U64, _U64],
@@ -264,6 +264,7 @@ _HIST_POOL: dict = {}
 _ROWSTATS_POOL: dict = {}
 # -1 auto, 0 wave-aggregated global atomics, 1 LDS-privatized (A/B probe knob)
 _CURVE_VARIANT = int(os.environ.get("METRICS_AMD_CURVE_VARIANT", "-1"))
+_CURVE_CCHUNK = int(os.environ.get("METRICS_AMD_CURVE_CCHUNK", "0"))
 
 
 def _epoch_buf(device, owner=None) -> Tensor:
@@ -417,6 +418,7 @@ def multiclass_curve_confmat(
         0,
         0,
         _CURVE_VARIANT,
+        _CURVE_CCHUNK,
         hist.data_ptr(),
     )
     _check(rc, "ma_multiclass_curve_hist")
@@ -583,6 +585,7 @@ def curve_hist_into_confmat(
             rm_ptr,
             ri_ptr,
             _CURVE_VARIANT,
+            _CURVE_CCHUNK,
             hist.data_ptr(),
         )
         _check(rc, "ma_multiclass_curve_hist")
