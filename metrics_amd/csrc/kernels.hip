@@ -1074,7 +1074,10 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
     const int lds_c_chunk = c_chunk_override > 0 ? c_chunk_override : 32;
     const size_t lds_bytes = (size_t)lds_c_chunk * (T + 1) * 2 * sizeof(unsigned int)
                              + (size_t)T * sizeof(float);
-    const int want_lds = variant == 1 || (variant < 0 && lds_bytes <= 96 * 1024 && B >= 1024);
+    // measured (tools/curve_sweep.py, B=8192 C=1000 T=200): ballot-merge with a
+    // small class chunk (oversubscribed grid) beats the LDS variant; LDS stays
+    // available behind the env knob for other shapes
+    const int want_lds = variant == 1;
     if (want_lds && lds_bytes <= 160 * 1024) {
         ll c_chunks_l = (C + lds_c_chunk - 1) / lds_c_chunk;
         ll row_chunks_l = 1536 / (c_chunks_l > 0 ? c_chunks_l : 1);
@@ -1098,9 +1101,11 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
         return (int)hipGetLastError();
     }
     ll row_chunks = (B + 255) / 256;
-    // pick the class chunk so the grid comfortably overfills 256 CUs
-    int c_chunk = c_chunk_override > 0 ? c_chunk_override : 32;
-    while (c_chunk_override <= 0 && c_chunk > 1 && ((C + c_chunk - 1) / c_chunk) * row_chunks < 1024) c_chunk /= 2;
+    // small class chunks oversubscribe the grid (latency hiding beats the
+    // per-wave reuse of staged thresholds): sweep plateaus at ~4 for the
+    // bench shape; halve further only if the grid is still tiny
+    int c_chunk = c_chunk_override > 0 ? c_chunk_override : 4;
+    while (c_chunk_override <= 0 && c_chunk > 1 && ((C + c_chunk - 1) / c_chunk) * row_chunks < 4096) c_chunk /= 2;
     ll c_chunks = (C + c_chunk - 1) / c_chunk;
     if (row_chunks > 65535 || c_chunks > 2147483647LL) return -101;
     dim3 grid((unsigned)c_chunks, (unsigned)row_chunks);
