@@ -25,6 +25,7 @@
 #include <hip/hip_bf16.h>
 #include <math.h>
 #include <stdint.h>
+#include <stdlib.h>
 
 #define WAVE 64
 #define CHECK(x)                                                                                   \
@@ -1053,8 +1054,16 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
     size_t shmem = (size_t)T * sizeof(float);
     if (shmem > 160 * 1024) return -100;
     if (norm_kind == 1 && flag) {
-        // softmax stats (one fused pass also sets the out-of-range flag)
-        int grid = grid_for(B, 4);
+        // softmax stats (one fused pass also sets the out-of-range flag).
+        // Fewer blocks -> each wave loops several rows -> more loads in
+        // flight per lane (the kernel is latency-bound at 2 vec-loads/lane).
+        static int rows_div = 0;
+        if (rows_div == 0) {
+            const char* e = getenv("MA_ROWSTATS_DIV");
+            rows_div = e ? atoi(e) : 16;
+            if (rows_div < 4) rows_div = 4;
+        }
+        int grid = grid_for(B, rows_div);
         if (dtype == 0)
             k_mc_rowstats<float, false><<<grid, 256, 0, s>>>(
                 (const float*)probs, B, C, (unsigned int*)flag, (float*)rowmax, (float*)rowinv);
