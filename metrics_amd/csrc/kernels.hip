@@ -949,8 +949,15 @@ int ma_mc_stat_logits(uintptr_t stream, uintptr_t preds, int dtype /*0=f32 1=bf1
                       uintptr_t fp, uintptr_t fn, uintptr_t confmat, uintptr_t valid_count,
                       uintptr_t argmax_out) {
     hipStream_t s = (hipStream_t)stream;
-    int waves_per_block = 4;
-    int grid = grid_for(B, waves_per_block);
+    // each wave loops several rows: more vector loads in flight per lane
+    // (latency-bound otherwise; sweep tools/curve_sweep.py analogue)
+    static int rows_div = 0;
+    if (rows_div == 0) {
+        const char* e = getenv("MA_STAT_DIV");
+        rows_div = e ? atoi(e) : 16;
+        if (rows_div < 4) rows_div = 4;
+    }
+    int grid = grid_for(B, rows_div);
     if (dtype == 0)
         k_mc_stat_logits<float, false><<<grid, 256, 0, s>>>(
             (const float*)preds, (const ll*)target, B, C, ignore_index, has_ignore,
