@@ -630,6 +630,78 @@ __global__ void k_curve_suffix(
     if (E && blockIdx.x == 0 && threadIdx.x == 0) atomicAdd(&E[1], 1u);
 }
 
+// tiled suffix kernel for the TRANSPOSED (T, C, 2, 2) state layout: one block
+// per tile of CTILE classes; per-class suffix sums are scanned once into LDS
+// and the state writes go out as contiguous (CTILE*4)-element runs per t —
+// the one-class-per-block version wrote 32 KB-strided 32 B chunks (every
+// write its own cache line RMW).
+template <int CTILE>
+__global__ void k_curve_suffix_tiled(
+    unsigned long long* __restrict__ hist /* (C, T+1, 2) */, int T, ll C, int zero_hist,
+    unsigned int* __restrict__ E, ll* __restrict__ confmat /* (T, C, 2, 2) */) {
+    extern __shared__ unsigned long long shs[];
+    unsigned long long* stage = shs;                       // CTILE * (T+1) * 2
+    unsigned long long* suf = shs + CTILE * (ll)(T + 1) * 2;  // CTILE * (T+1) * 2
+    const ll c0 = (ll)blockIdx.x * CTILE;
+    const int ctile = (int)min((ll)CTILE, C - c0);
+    for (int i = threadIdx.x; i < ctile * (T + 1) * 2; i += blockDim.x) {
+        const int lc = i / ((T + 1) * 2);
+        const int rem = i - lc * (T + 1) * 2;
+        stage[lc * (T + 1) * 2 + rem] = hist[(c0 + lc) * (ll)(T + 1) * 2 + rem];
+        if (zero_hist) hist[(c0 + lc) * (ll)(T + 1) * 2 + rem] = 0;
+    }
+    __syncthreads();
+    // per-class suffix scan (one thread per class, O(T))
+    if (threadIdx.x < ctile) {
+        const int lc = threadIdx.x;
+        unsigned long long* h = stage + lc * (T + 1) * 2;
+        unsigned long long* sf = suf + lc * (T + 1) * 2;
+        unsigned long long tp = 0, fp = 0;
+        for (int j = T; j >= 0; j--) {
+            sf[j * 2 + 0] = fp;  // suffix EXCLUDING j ... shifted below
+            sf[j * 2 + 1] = tp;
+            fp += h[j * 2 + 0];
+            tp += h[j * 2 + 1];
+        }
+        // store totals at slot 0's running end: reuse registers via last iter:
+        // after the loop, fp/tp hold the FULL totals; stash them in sf[...]
+        sf[0] = fp;  // slot (j=0, which is never read as suffix) = neg_total
+        sf[1] = tp;  //                                            pos_total
+    }
+    __syncthreads();
+    // wait: suffix for threshold t is sum over j > t, i.e. sf at j=t+1..T —
+    // the scan above stored, at position j, the sum over (j, T] EXCLUSIVE of
+    // j... no: at iteration j we stored the sum of elements j+1..T. So the
+    // confmat row t needs sf[t] as stored BEFORE adding h[t]? Position t
+    // holds sum over (t, T] = j > t. Correct as stored, except slot 0 was
+    // overwritten with totals — handle t=0 via totals minus h[0].
+    for (int i = threadIdx.x; i < T * ctile * 4; i += blockDim.x) {
+        const int t = i / (ctile * 4);
+        const int rem = i - t * (ctile * 4);
+        const int lc = rem >> 2;
+        const int q = rem & 3;
+        const unsigned long long* sf = suf + lc * (T + 1) * 2;
+        const unsigned long long* h = stage + lc * (T + 1) * 2;
+        const unsigned long long neg_total = sf[0];
+        const unsigned long long pos_total = sf[1];
+        unsigned long long fp_s, tp_s;
+        if (t == 0) {
+            fp_s = neg_total - h[0];
+            tp_s = pos_total - h[1];
+        } else {
+            fp_s = sf[t * 2 + 0];
+            tp_s = sf[t * 2 + 1];
+        }
+        ll v;
+        if (q == 0) v = (ll)(neg_total - fp_s);       // tn
+        else if (q == 1) v = (ll)fp_s;                // fp
+        else if (q == 2) v = (ll)(pos_total - tp_s);  // fn
+        else v = (ll)tp_s;                            // tp
+        confmat[((ll)t * C + (c0 + lc)) * 4 + q] += v;
+    }
+    if (E && blockIdx.x == 0 && threadIdx.x == 0) atomicAdd(&E[1], 1u);
+}
+
 // ---------------------------------------------------------------------------
 // fused stat-delta apply: given the kernel scratch [tp|fp|fn|valid] produced
 // by k_mc_stat_*, add the deltas into the four metric state tensors
@@ -1143,6 +1215,17 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
 int ma_curve_suffix(uintptr_t stream, uintptr_t hist, ll outer, int T, int transposed,
                     int zero_hist, uintptr_t epoch_buf, uintptr_t confmat) {
     hipStream_t s = (hipStream_t)stream;
+    if (transposed) {
+        constexpr int CTILE = 8;
+        size_t shmem_t = (size_t)2 * CTILE * (T + 1) * 2 * sizeof(unsigned long long);
+        if (shmem_t <= 160 * 1024) {
+            int grid = (int)((outer + CTILE - 1) / CTILE);
+            k_curve_suffix_tiled<CTILE><<<grid, 256, shmem_t, s>>>(
+                (unsigned long long*)hist, T, outer, zero_hist, (unsigned int*)epoch_buf,
+                (ll*)confmat);
+            return (int)hipGetLastError();
+        }
+    }
     size_t shmem = (size_t)(T + 1) * 2 * sizeof(unsigned long long);
     if (shmem > 160 * 1024) return -100;
     k_curve_suffix<<<(int)outer, 256, shmem, s>>>((unsigned long long*)hist, T, outer, transposed,
