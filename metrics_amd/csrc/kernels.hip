@@ -1216,13 +1216,24 @@ int ma_curve_suffix(uintptr_t stream, uintptr_t hist, ll outer, int T, int trans
                     int zero_hist, uintptr_t epoch_buf, uintptr_t confmat) {
     hipStream_t s = (hipStream_t)stream;
     if (transposed) {
-        constexpr int CTILE = 8;
-        size_t shmem_t = (size_t)2 * CTILE * (T + 1) * 2 * sizeof(unsigned long long);
-        if (shmem_t <= 160 * 1024) {
-            int grid = (int)((outer + CTILE - 1) / CTILE);
-            k_curve_suffix_tiled<CTILE><<<grid, 256, shmem_t, s>>>(
-                (unsigned long long*)hist, T, outer, zero_hist, (unsigned int*)epoch_buf,
-                (ll*)confmat);
+        static int ctile_sel = 0;
+        if (ctile_sel == 0) {
+            const char* e = getenv("MA_SUFFIX_CTILE");
+            ctile_sel = e ? atoi(e) : 2;
+        }
+        const int CT = ctile_sel;
+        size_t shmem_t = (size_t)2 * CT * (T + 1) * 2 * sizeof(unsigned long long);
+        if (CT >= 2 && shmem_t <= 160 * 1024) {
+            int grid = (int)((outer + CT - 1) / CT);
+            if (CT == 2)
+                k_curve_suffix_tiled<2><<<grid, 256, shmem_t, s>>>(
+                    (unsigned long long*)hist, T, outer, zero_hist, (unsigned int*)epoch_buf, (ll*)confmat);
+            else if (CT == 4)
+                k_curve_suffix_tiled<4><<<grid, 256, shmem_t, s>>>(
+                    (unsigned long long*)hist, T, outer, zero_hist, (unsigned int*)epoch_buf, (ll*)confmat);
+            else
+                k_curve_suffix_tiled<8><<<grid, 256, shmem_t, s>>>(
+                    (unsigned long long*)hist, T, outer, zero_hist, (unsigned int*)epoch_buf, (ll*)confmat);
             return (int)hipGetLastError();
         }
     }
