@@ -479,6 +479,20 @@ __global__ void __launch_bounds__(256) k_multiclass_curve_hist(
     const float rinv = (norm == 1 && valid) ? rowinv[row] : 0.0f;
     const T_* prow = probs + (valid ? row * C : 0);
     const ll* tgt_row = target + (valid ? row * C : 0);
+    // one vector load covers the whole class chunk (4 scalar 2B/4B loads were
+    // the latency bottleneck at small chunks)
+    float pv[4];
+    const bool vec4 = (c_hi - c_lo) == 4 && ((c_lo & 3) == 0) && ((C & 3) == 0);
+    if (valid && vec4) {
+        if (IS_BF16) {
+            ushort4 u = reinterpret_cast<const ushort4*>(prow)[c_lo >> 2];
+            pv[0] = bf16_to_f32(u.x); pv[1] = bf16_to_f32(u.y);
+            pv[2] = bf16_to_f32(u.z); pv[3] = bf16_to_f32(u.w);
+        } else {
+            float4 u = reinterpret_cast<const float4*>(prow)[c_lo >> 2];
+            pv[0] = u.x; pv[1] = u.y; pv[2] = u.z; pv[3] = u.w;
+        }
+    }
     for (ll c = c_lo; c < c_hi; c++) {
         int key = -1;
         if (valid) {
@@ -491,8 +505,9 @@ __global__ void __launch_bounds__(256) k_multiclass_curve_hist(
                 if (has_ignore && t == ignore_index) label = -1;
             }
             if (label >= 0) {
-                float p = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(prow)[c])
-                                  : (float)prow[c];
+                float p = vec4 ? pv[c - c_lo]
+                               : (IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(prow)[c])
+                                          : (float)prow[c]);
                 if (norm == 1) p = expf(p - rmax) * rinv;
                 else if (norm == 2) p = 1.0f / (1.0f + expf(-p));
                 int j = uniform ? bucket_of_uniform(p, sthr2, T, t0, inv_step)
