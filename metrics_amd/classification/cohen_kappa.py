@@ -70,6 +70,7 @@ class BinaryCohenKappa(Metric):
 
 
 class MulticlassCohenKappa(Metric):
+    _hip_fused_kind = "mc_confmat"
     """Cohen's kappa for multiclass tasks (stateful)."""
 
     is_differentiable = False

@@ -102,6 +102,7 @@ def _fused_confmat_update(metric, preds, target) -> bool:
 
 
 class MulticlassConfusionMatrix(Metric):
+    _hip_fused_kind = "mc_confmat"
     """(C,C) confusion matrix for multiclass tasks (stateful)."""
 
     is_differentiable = False

@@ -26,6 +26,7 @@ from metrics_amd.functional.classification.stat_scores import (
 
 
 class MulticlassExactMatch(Metric):
+    _hip_fused_kind = "mc_exact"
     """Exact match for multiclass tasks (stateful)."""
 
     is_differentiable = False

@@ -67,6 +67,7 @@ class BinaryJaccardIndex(Metric):
 
 
 class MulticlassJaccardIndex(Metric):
+    _hip_fused_kind = "mc_confmat"
     """Jaccard index for multiclass tasks (stateful)."""
 
     is_differentiable = False

@@ -67,6 +67,7 @@ class BinaryMatthewsCorrCoef(Metric):
 
 
 class MulticlassMatthewsCorrCoef(Metric):
+    _hip_fused_kind = "mc_confmat"
     """MCC for multiclass tasks (stateful)."""
 
     is_differentiable = False

@@ -124,6 +124,7 @@ class BinaryStatScores(_AbstractStatScores):
 
 
 class MulticlassStatScores(_AbstractStatScores):
+    _hip_fused_kind = "mc_stat"
     """Per-class tp/fp/tn/fn counts for multiclass tasks."""
 
     is_differentiable: bool = False

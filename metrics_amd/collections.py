@@ -36,6 +36,76 @@ def _strip_suffix(s: str, suffix: str) -> str:
     return s[: -len(suffix)] if s.endswith(suffix) else s
 
 
+class _FusedMulticlassUpdatePlan:
+    """Collection-level cross-metric fusion: the stat-scores, confusion-matrix
+    and exact-match group leaders all start with the same argmax pass over the
+    (B, C) logits — ONE fused kernel call feeds all of them (MI355X-first:
+    the logits are read once from HBM instead of three times)."""
+
+    def __init__(self, stat, confmat, exact):
+        self.stat = stat
+        self.confmat = confmat
+        self.exact = exact
+        self.leaders = tuple(m for m in (stat, confmat, exact) if m is not None)
+
+    @staticmethod
+    def build(collection) -> "Optional[_FusedMulticlassUpdatePlan]":
+        stat = confmat = exact = None
+        for members in collection._groups.values():
+            leader = getattr(collection, members[0])
+            kind = getattr(type(leader), "_hip_fused_kind", None)
+            if kind is None or getattr(leader, "validate_args", True):
+                continue  # validations are skipped on the fused path
+            if kind == "mc_stat" and stat is None:
+                if getattr(leader, "multidim_average", "global") == "global" and getattr(leader, "top_k", 1) == 1:
+                    stat = leader
+            elif kind == "mc_confmat" and confmat is None:
+                confmat = leader
+            elif kind == "mc_exact" and exact is None:
+                if getattr(leader, "multidim_average", "global") == "global":
+                    exact = leader
+        participants = [m for m in (stat, confmat, exact) if m is not None]
+        if stat is None or len(participants) < 2:
+            return None
+        ncs = {m.num_classes for m in participants}
+        igs = {m.ignore_index for m in participants}
+        if len(ncs) != 1 or len(igs) != 1:
+            return None
+        return _FusedMulticlassUpdatePlan(stat, confmat, exact)
+
+    def try_run(self, *args: Any, **kwargs: Any) -> tuple:
+        """Run the fused update if the inputs qualify; returns the handled
+        leaders (empty tuple -> caller falls back to per-leader updates)."""
+        if kwargs or len(args) != 2:
+            return ()
+        preds, target = args
+        if not (isinstance(preds, Tensor) and isinstance(target, Tensor) and preds.is_cuda):
+            return ()
+        if target.ndim != 1:
+            return ()
+        if preds.is_floating_point():
+            if preds.ndim != 2 or preds.dtype not in (torch.float32, torch.bfloat16):
+                return ()
+        elif preds.shape != target.shape:
+            return ()
+        from metrics_amd.ops import _hip
+
+        if not _hip.hip_available():
+            return ()
+        stat = self.stat
+        scratch = getattr(stat, "_hip_scratch", None)
+        if scratch is None or scratch.device != preds.device:
+            scratch = torch.zeros(3 * stat.num_classes + 1, dtype=torch.long, device=preds.device)
+            stat._hip_scratch = scratch
+        _hip.mc_fused_collection_update(
+            preds, target, stat.num_classes, stat.ignore_index,
+            stat=(scratch, stat.tp, stat.fp, stat.tn, stat.fn),
+            confmat=self.confmat.confmat if self.confmat is not None else None,
+            exact=(self.exact.correct, self.exact.total) if self.exact is not None else None,
+        )
+        return self.leaders
+
+
 class MetricCollection(ModuleDict):
     """A dict-like container of metrics sharing one ``update``/``compute`` call pattern.
 
@@ -60,6 +130,7 @@ class MetricCollection(ModuleDict):
         compute_groups: Union[bool, List[List[str]]] = True,
     ) -> None:
         super().__init__()
+        self._fused_plan: Any = False
         self.prefix = self._check_arg(prefix, "prefix")
         self.postfix = self._check_arg(postfix, "postfix")
         self._enable_compute_groups = compute_groups
@@ -85,9 +156,17 @@ class MetricCollection(ModuleDict):
             # invalidate cached compute results on ALL members
             for k in self.keys(keep_base=True):
                 getattr(self, str(k))._computed = None
+            if self._fused_plan is False:
+                self._fused_plan = _FusedMulticlassUpdatePlan.build(self)
+            fused_done: tuple = ()
+            if self._fused_plan is not None:
+                fused_done = self._fused_plan.try_run(*args, **kwargs)
             # run only each group's leader
             for members in self._groups.values():
                 leader = getattr(self, members[0])
+                if leader in fused_done:
+                    leader._update_count += 1
+                    continue
                 leader.update(*args, **leader._filter_kwargs(**kwargs))
             if self._state_is_copy:
                 # re-establish the alias links broken by a copy-on-access

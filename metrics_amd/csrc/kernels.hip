@@ -750,14 +750,16 @@ __global__ void k_apply_stat_deltas(
 // exact-match epilogue: correct += sum(tp), total += valid — one block,
 // zeroes the scratch in-flight (same ping-pong valid protocol as apply_deltas).
 __global__ void k_exact_apply(unsigned long long* __restrict__ scratch, ll C, ll B,
-                              ll* __restrict__ correct, ll* __restrict__ total) {
+                              int zero_scratch, ll* __restrict__ correct, ll* __restrict__ total) {
     __shared__ unsigned long long part[256];
     unsigned long long acc = 0;
     for (ll i = threadIdx.x; i < C; i += blockDim.x) {
         acc += scratch[i];
-        scratch[i] = 0;
-        scratch[C + i] = 0;
-        scratch[2 * C + i] = 0;
+        if (zero_scratch) {
+            scratch[i] = 0;
+            scratch[C + i] = 0;
+            scratch[2 * C + i] = 0;
+        }
     }
     part[threadIdx.x] = acc;
     __syncthreads();
@@ -771,7 +773,7 @@ __global__ void k_exact_apply(unsigned long long* __restrict__ scratch, ll C, ll
         const ll valid = (ll)scratch[3 * C];
         correct[0] += (ll)part[0] + (B - valid);
         total[0] += B;
-        scratch[3 * C] = 0;
+        if (zero_scratch) scratch[3 * C] = 0;
     }
 }
 
@@ -1288,10 +1290,11 @@ int ma_curve_auc_from_confmat(uintptr_t stream, uintptr_t confmat, int T, ll C, 
     return (int)hipGetLastError();
 }
 
-int ma_exact_apply(uintptr_t stream, uintptr_t scratch, ll C, ll B, uintptr_t correct,
-                   uintptr_t total) {
+int ma_exact_apply(uintptr_t stream, uintptr_t scratch, ll C, ll B, int zero_scratch,
+                   uintptr_t correct, uintptr_t total) {
     hipStream_t s = (hipStream_t)stream;
-    k_exact_apply<<<1, 256, 0, s>>>((unsigned long long*)scratch, C, B, (ll*)correct, (ll*)total);
+    k_exact_apply<<<1, 256, 0, s>>>((unsigned long long*)scratch, C, B, zero_scratch,
+                                    (ll*)correct, (ll*)total);
     return (int)hipGetLastError();
 }
 

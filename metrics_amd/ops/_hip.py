@@ -55,7 +55,7 @@ _SIGNATURES = {
     "ma_multiclass_curve_hist": [_U64, _U64, _I, _U64, _LL, _LL, _U64, _I, _LL, _I, _I, _I, _F, _F, _I, _U64, _U64, _U64, _I, _I, _U64],
     "ma_curve_suffix": [_U64, _U64, _LL, _I, _I, _I, _U64, _U64],
     "ma_apply_stat_deltas": [_U64, _U64, _LL, _U64, _U64, _U64, _U64],
-    "ma_exact_apply": [_U64, _U64, _LL, _LL, _U64, _U64],
+    "ma_exact_apply": [_U64, _U64, _LL, _LL, _I, _U64, _U64],
     "ma_curve_auc_from_confmat": [_U64, _U64, _I, _LL, _I, _U64, _U64],
     "ma_linear_stat_compute": [_U64, _U64, _U64, _U64, _U64, _LL, _F, _F, _F, _F, _F, _F, _F, _F, _I, _I, _F, _F, _F, _U64],
     "ma_err_reduce": [_U64, _U64, _U64, _I, _LL, _I, _D, _U64, _I, _I, _U64],
@@ -784,6 +784,7 @@ def mc_exact_into(
         scratch.data_ptr(),
         C,
         target.numel(),
+        1,
         correct.data_ptr(),
         total.data_ptr(),
     )
@@ -835,3 +836,59 @@ def curve_auc_from_confmat(confmat_state: Tensor, mode: int) -> Tuple[Tensor, Te
     )
     _check(rc, "ma_curve_auc_from_confmat")
     return out, weights
+
+
+def mc_fused_collection_update(
+    preds: Tensor, target: Tensor, num_classes: int, ignore_index: Optional[int],
+    stat=None, confmat=None, exact=None,
+) -> None:
+    """Collection-level fused update: ONE pass over the (B, C) logits feeds the
+    stat-scores, confusion-matrix and exact-match leaders at once.
+
+    ``stat``  = (scratch, tp, fp, tn, fn) — required (owns the count scratch)
+    ``confmat`` = confmat state tensor (atomics accumulate directly) or None
+    ``exact`` = (correct, total) or None (reads the shared scratch BEFORE the
+    apply kernel consumes and zeroes it).
+    """
+    lib = _lib()
+    C = num_classes
+    scratch, tp, fp, tn, fn = stat
+    s_valid_ptr = scratch.data_ptr() + 3 * C * 8
+    if preds.ndim == 2 and preds.is_floating_point():
+        preds = preds.contiguous()
+        target = target.contiguous().long()
+        B = preds.shape[0]
+        rc = lib.ma_mc_stat_logits(
+            _stream(), preds.data_ptr(), _dtype_code(preds), target.data_ptr(),
+            B, C,
+            ignore_index if ignore_index is not None else 0,
+            1 if ignore_index is not None else 0,
+            scratch[:C].data_ptr(), scratch[C : 2 * C].data_ptr(), scratch[2 * C : 3 * C].data_ptr(),
+            confmat.data_ptr() if confmat is not None else 0,
+            s_valid_ptr, 0,
+        )
+        _check(rc, "ma_mc_stat_logits")
+    else:
+        p2 = preds.contiguous().long().flatten()
+        t2 = target.contiguous().long().flatten()
+        B = t2.numel()
+        rc = lib.ma_mc_stat_labels(
+            _stream(), p2.data_ptr(), t2.data_ptr(), B, C,
+            ignore_index if ignore_index is not None else 0,
+            1 if ignore_index is not None else 0,
+            scratch[:C].data_ptr(), scratch[C : 2 * C].data_ptr(), scratch[2 * C : 3 * C].data_ptr(),
+            confmat.data_ptr() if confmat is not None else 0,
+            s_valid_ptr,
+        )
+        _check(rc, "ma_mc_stat_labels")
+    if exact is not None:
+        correct, total = exact
+        rc = lib.ma_exact_apply(
+            _stream(), scratch.data_ptr(), C, B, 0, correct.data_ptr(), total.data_ptr()
+        )
+        _check(rc, "ma_exact_apply")
+    rc = lib.ma_apply_stat_deltas(
+        _stream(), scratch.data_ptr(), C,
+        tp.data_ptr(), fp.data_ptr(), tn.data_ptr(), fn.data_ptr(),
+    )
+    _check(rc, "ma_apply_stat_deltas")

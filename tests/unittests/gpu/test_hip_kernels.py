@@ -609,3 +609,59 @@ def test_binary_auroc_ap_compute_kernel():
     g.update(preds.cuda(), target.cuda())
     c.update(preds, target)
     assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-6)
+
+
+def test_fused_collection_update_plan():
+    """Collection-level fusion: one kernel pass must equal separate updates."""
+    torch.manual_seed(25)
+
+    def make(device=None):
+        c = ma.MetricCollection({
+            "acc": ma.MulticlassAccuracy(num_classes=50, average="macro", validate_args=False),
+            "prec": ma.MulticlassPrecision(num_classes=50, average="micro", validate_args=False),
+            "kappa": ma.MulticlassCohenKappa(num_classes=50, validate_args=False),
+            "confmat": ma.MulticlassConfusionMatrix(num_classes=50, validate_args=False),
+            "exact": ma.MulticlassExactMatch(num_classes=50, validate_args=False),
+        })
+        return c.to(device) if device else c
+
+    batches = [(torch.randn(2048, 50, dtype=torch.bfloat16), torch.randint(0, 50, (2048,))) for _ in range(3)]
+    gpu, cpu = make("cuda"), make()
+    for p, t in batches:
+        gpu.update(p.cuda(), t.cuda())
+        cpu.update(p, t)
+    # the plan must actually engage
+    assert gpu._fused_plan is not None and gpu._fused_plan is not False
+    rg, rc_ = gpu.compute(), cpu.compute()
+    for k in rg:
+        assert torch.allclose(rg[k].float().cpu(), rc_[k].float(), atol=1e-6), k
+
+    # with ignore_index on every participant
+    gi = ma.MetricCollection({
+        "acc": ma.MulticlassAccuracy(num_classes=50, ignore_index=-1, validate_args=False),
+        "confmat": ma.MulticlassConfusionMatrix(num_classes=50, ignore_index=-1, validate_args=False),
+        "exact": ma.MulticlassExactMatch(num_classes=50, ignore_index=-1, validate_args=False),
+    }).to("cuda")
+    ci = ma.MetricCollection({
+        "acc": ma.MulticlassAccuracy(num_classes=50, ignore_index=-1, validate_args=False),
+        "confmat": ma.MulticlassConfusionMatrix(num_classes=50, ignore_index=-1, validate_args=False),
+        "exact": ma.MulticlassExactMatch(num_classes=50, ignore_index=-1, validate_args=False),
+    })
+    p = torch.randn(4096, 50)
+    t = torch.randint(0, 50, (4096,))
+    t[::7] = -1
+    gi.update(p.cuda(), t.cuda())
+    ci.update(p, t)
+    r1, r2 = gi.compute(), ci.compute()
+    for k in r1:
+        assert torch.allclose(r1[k].float().cpu(), r2[k].float(), atol=1e-6), k
+
+
+def test_fused_plan_not_built_with_validate_args():
+    coll = ma.MetricCollection({
+        "acc": ma.MulticlassAccuracy(num_classes=5),  # validate_args=True
+        "confmat": ma.MulticlassConfusionMatrix(num_classes=5),
+    }).to("cuda")
+    coll.update(torch.randn(64, 5, device="cuda"), torch.randint(0, 5, (64,), device="cuda"))
+    coll.update(torch.randn(64, 5, device="cuda"), torch.randint(0, 5, (64,), device="cuda"))
+    assert coll._fused_plan is None  # validations would be skipped -> no fusion
