@@ -164,7 +164,9 @@ class MetricCollection(ModuleDict):
             # run only each group's leader
             for members in self._groups.values():
                 leader = getattr(self, members[0])
-                if leader in fused_done:
+                # identity check: Metric.__eq__ is the compositional operator
+                # (returns a CompositionalMetric, which is always truthy)
+                if any(leader is m for m in fused_done):
                     leader._update_count += 1
                     continue
                 leader.update(*args, **leader._filter_kwargs(**kwargs))
