@@ -121,6 +121,7 @@ def _auc_score(x: Tensor, y: Tensor) -> Tensor:
 
 
 class MulticlassPrecisionRecallCurve(Metric):
+    _hip_fused_kind = "mc_curve"
     """PR curves for multiclass tasks (stateful)."""
 
     is_differentiable: bool = False

@@ -42,15 +42,16 @@ class _FusedMulticlassUpdatePlan:
     (B, C) logits — ONE fused kernel call feeds all of them (MI355X-first:
     the logits are read once from HBM instead of three times)."""
 
-    def __init__(self, stat, confmat, exact):
+    def __init__(self, stat, confmat, exact, curve):
         self.stat = stat
         self.confmat = confmat
         self.exact = exact
-        self.leaders = tuple(m for m in (stat, confmat, exact) if m is not None)
+        self.curve = curve
+        self.leaders = tuple(m for m in (stat, confmat, exact, curve) if m is not None)
 
     @staticmethod
     def build(collection) -> "Optional[_FusedMulticlassUpdatePlan]":
-        stat = confmat = exact = None
+        stat = confmat = exact = curve = None
         for members in collection._groups.values():
             leader = getattr(collection, members[0])
             kind = getattr(type(leader), "_hip_fused_kind", None)
@@ -64,14 +65,17 @@ class _FusedMulticlassUpdatePlan:
             elif kind == "mc_exact" and exact is None:
                 if getattr(leader, "multidim_average", "global") == "global":
                     exact = leader
-        participants = [m for m in (stat, confmat, exact) if m is not None]
+            elif kind == "mc_curve" and curve is None:
+                if leader.thresholds is not None and getattr(leader, "average", None) != "micro":
+                    curve = leader
+        participants = [m for m in (stat, confmat, exact, curve) if m is not None]
         if stat is None or len(participants) < 2:
             return None
         ncs = {m.num_classes for m in participants}
         igs = {m.ignore_index for m in participants}
         if len(ncs) != 1 or len(igs) != 1:
             return None
-        return _FusedMulticlassUpdatePlan(stat, confmat, exact)
+        return _FusedMulticlassUpdatePlan(stat, confmat, exact, curve)
 
     def try_run(self, *args: Any, **kwargs: Any) -> tuple:
         """Run the fused update if the inputs qualify; returns the handled
@@ -97,13 +101,35 @@ class _FusedMulticlassUpdatePlan:
         if scratch is None or scratch.device != preds.device:
             scratch = torch.zeros(3 * stat.num_classes + 1, dtype=torch.long, device=preds.device)
             stat._hip_scratch = scratch
+        curve = self.curve
+        rowstats = None
+        if curve is not None and preds.is_floating_point():
+            # the fused pass also emits softmax row-stats + the epoch flag;
+            # the curve leader's own row-stats kernel is skipped
+            rbuf = curve.__dict__.get("_hip_rowstats_buf")
+            if rbuf is None or rbuf.device != preds.device or rbuf.shape[1] < preds.shape[0]:
+                rbuf = torch.empty(2, preds.shape[0], dtype=torch.float32, device=preds.device)
+                curve.__dict__["_hip_rowstats_buf"] = rbuf
+            ebuf = _hip._epoch_buf(preds.device, curve)
+            rowstats = (rbuf[0], rbuf[1], ebuf)
+        elif curve is not None:
+            curve = None  # label preds: curve can't ride the fused pass
         _hip.mc_fused_collection_update(
             preds, target, stat.num_classes, stat.ignore_index,
             stat=(scratch, stat.tp, stat.fp, stat.tn, stat.fn),
             confmat=self.confmat.confmat if self.confmat is not None else None,
             exact=(self.exact.correct, self.exact.total) if self.exact is not None else None,
+            rowstats=rowstats,
         )
-        return self.leaders
+        if curve is not None:
+            _hip.curve_hist_into_confmat(
+                preds, target, curve.thresholds, curve.ignore_index, curve.confmat,
+                mode=0, norm="softmax", owner=curve, stats_ready=True,
+            )
+            handled = self.leaders
+        else:
+            handled = tuple(m for m in self.leaders if m is not self.curve)
+        return handled
 
 
 class MetricCollection(ModuleDict):

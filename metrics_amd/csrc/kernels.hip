@@ -51,14 +51,19 @@ __global__ void __launch_bounds__(256) k_mc_stat_logits(
     const T* __restrict__ preds, const ll* __restrict__ target, ll B, ll C, ll ignore_index,
     int has_ignore, unsigned long long* __restrict__ tp, unsigned long long* __restrict__ fp,
     unsigned long long* __restrict__ fn, unsigned long long* __restrict__ confmat,
-    unsigned long long* __restrict__ valid_count, ll* __restrict__ argmax_out) {
+    unsigned long long* __restrict__ valid_count, ll* __restrict__ argmax_out,
+    float* __restrict__ rowmax, float* __restrict__ rowinv, unsigned int* __restrict__ E) {
     const int lane = threadIdx.x & (WAVE - 1);
     const int wave_in_block = threadIdx.x / WAVE;
     const int waves_per_block = blockDim.x / WAVE;
     const ll row0 = (ll)blockIdx.x * waves_per_block + wave_in_block;
     const ll row_stride = (ll)gridDim.x * waves_per_block;
+    const bool want_stats = rowmax != nullptr;  // softmax row-stats piggyback
+    const unsigned int cur_epoch = E ? E[1] + 1u : 0u;
+    unsigned int outside = 0;
     __shared__ unsigned int block_valid;
-    if (threadIdx.x == 0) block_valid = 0;
+    __shared__ unsigned int blk_outside;
+    if (threadIdx.x == 0) { block_valid = 0; blk_outside = 0; }
     __syncthreads();
     unsigned int my_valid = 0;
 
@@ -66,6 +71,19 @@ __global__ void __launch_bounds__(256) k_mc_stat_logits(
         const T* prow = preds + row * C;
         float best = -INFINITY;
         ll best_idx = 0x7fffffffffffffffLL;
+        float sm_m = -3.4e38f, sm_s = 0.0f;  // online softmax (max, sumexp)
+        auto stats_fold = [&](const float* f, int k) {
+            if (!want_stats) return;
+            float m8 = f[0];
+            for (int i = 1; i < k; i++) m8 = fmaxf(m8, f[i]);
+            float s8 = 0.0f;
+            for (int i = 0; i < k; i++) {
+                outside |= (f[i] < 0.0f || f[i] > 1.0f) ? 1u : 0u;
+                s8 += __expf(f[i] - m8);
+            }
+            if (m8 > sm_m) { sm_s = sm_s * __expf(sm_m - m8) + s8; sm_m = m8; }
+            else { sm_s += s8 * __expf(m8 - sm_m); }
+        };
         // vectorized loads: 16B/lane for bf16 when C%8==0, else 8B/4B paths
         if (IS_BF16 && (C & 7) == 0) {
             struct U8 { ushort4 a; ushort4 b; };
@@ -76,6 +94,7 @@ __global__ void __launch_bounds__(256) k_mc_stat_logits(
                 float f[8] = {bf16_to_f32(u.a.x), bf16_to_f32(u.a.y), bf16_to_f32(u.a.z), bf16_to_f32(u.a.w),
                               bf16_to_f32(u.b.x), bf16_to_f32(u.b.y), bf16_to_f32(u.b.z), bf16_to_f32(u.b.w)};
                 ll c = v * 8;
+                stats_fold(f, 8);
 #pragma unroll
                 for (int k = 0; k < 8; k++) {
                     if (f[k] > best || (f[k] == best && c + k < best_idx)) { best = f[k]; best_idx = c + k; }
@@ -90,6 +109,8 @@ __global__ void __launch_bounds__(256) k_mc_stat_logits(
                     float f0 = bf16_to_f32(u.x), f1 = bf16_to_f32(u.y);
                     float f2 = bf16_to_f32(u.z), f3 = bf16_to_f32(u.w);
                     ll c = v * 4;
+                    float fv[4] = {f0, f1, f2, f3};
+                    stats_fold(fv, 4);
                     if (f0 > best || (f0 == best && c < best_idx)) { best = f0; best_idx = c; }
                     if (f1 > best || (f1 == best && c + 1 < best_idx)) { best = f1; best_idx = c + 1; }
                     if (f2 > best || (f2 == best && c + 2 < best_idx)) { best = f2; best_idx = c + 2; }
@@ -100,6 +121,8 @@ __global__ void __launch_bounds__(256) k_mc_stat_logits(
                 for (ll v = lane; v < nvec; v += WAVE) {
                     float4 u = pv[v];
                     ll c = v * 4;
+                    float fv[4] = {u.x, u.y, u.z, u.w};
+                    stats_fold(fv, 4);
                     if (u.x > best || (u.x == best && c < best_idx)) { best = u.x; best_idx = c; }
                     if (u.y > best || (u.y == best && c + 1 < best_idx)) { best = u.y; best_idx = c + 1; }
                     if (u.z > best || (u.z == best && c + 2 < best_idx)) { best = u.z; best_idx = c + 2; }
@@ -110,8 +133,18 @@ __global__ void __launch_bounds__(256) k_mc_stat_logits(
             for (ll c = lane; c < C; c += WAVE) {
                 float f = IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(prow)[c])
                                   : (float)prow[c];
+                stats_fold(&f, 1);
                 if (f > best || (f == best && c < best_idx)) { best = f; best_idx = c; }
             }
+        }
+        if (want_stats) {
+            for (int off = WAVE / 2; off > 0; off >>= 1) {
+                float om = __shfl_down(sm_m, off);
+                float os = __shfl_down(sm_s, off);
+                if (om > sm_m) { sm_s = sm_s * __expf(sm_m - om) + os; sm_m = om; }
+                else { sm_s += os * __expf(om - sm_m); }
+            }
+            if (lane == 0) { rowmax[row] = sm_m; rowinv[row] = 1.0f / sm_s; }
         }
         // wave shuffle reduce: max value, lowest index on tie
         for (int off = WAVE / 2; off > 0; off >>= 1) {
@@ -138,8 +171,13 @@ __global__ void __launch_bounds__(256) k_mc_stat_logits(
     }
     // one valid-count atomic per block, not per row
     if (my_valid) atomicAdd(&block_valid, my_valid);
+    for (int off = WAVE / 2; off > 0; off >>= 1) outside |= __shfl_down(outside, off);
+    if (lane == 0 && outside) atomicOr(&blk_outside, 1u);
     __syncthreads();
-    if (threadIdx.x == 0 && block_valid) atomicAdd(valid_count, (unsigned long long)block_valid);
+    if (threadIdx.x == 0) {
+        if (block_valid) atomicAdd(valid_count, (unsigned long long)block_valid);
+        if (E && blk_outside && E[0] != cur_epoch) atomicMax(&E[0], cur_epoch);
+    }
 }
 
 // K1b: multiclass stat scores from integer label preds (element-wise)
@@ -1036,7 +1074,7 @@ extern "C" {
 int ma_mc_stat_logits(uintptr_t stream, uintptr_t preds, int dtype /*0=f32 1=bf16*/,
                       uintptr_t target, ll B, ll C, ll ignore_index, int has_ignore, uintptr_t tp,
                       uintptr_t fp, uintptr_t fn, uintptr_t confmat, uintptr_t valid_count,
-                      uintptr_t argmax_out) {
+                      uintptr_t argmax_out, uintptr_t rowmax, uintptr_t rowinv, uintptr_t epoch_buf) {
     hipStream_t s = (hipStream_t)stream;
     // each wave loops several rows: more vector loads in flight per lane
     // (latency-bound otherwise; sweep tools/curve_sweep.py analogue)
@@ -1051,12 +1089,14 @@ int ma_mc_stat_logits(uintptr_t stream, uintptr_t preds, int dtype /*0=f32 1=bf1
         k_mc_stat_logits<float, false><<<grid, 256, 0, s>>>(
             (const float*)preds, (const ll*)target, B, C, ignore_index, has_ignore,
             (unsigned long long*)tp, (unsigned long long*)fp, (unsigned long long*)fn,
-            (unsigned long long*)confmat, (unsigned long long*)valid_count, (ll*)argmax_out);
+            (unsigned long long*)confmat, (unsigned long long*)valid_count, (ll*)argmax_out,
+            (float*)rowmax, (float*)rowinv, (unsigned int*)epoch_buf);
     else
         k_mc_stat_logits<unsigned short, true><<<grid, 256, 0, s>>>(
             (const unsigned short*)preds, (const ll*)target, B, C, ignore_index, has_ignore,
             (unsigned long long*)tp, (unsigned long long*)fp, (unsigned long long*)fn,
-            (unsigned long long*)confmat, (unsigned long long*)valid_count, (ll*)argmax_out);
+            (unsigned long long*)confmat, (unsigned long long*)valid_count, (ll*)argmax_out,
+            (float*)rowmax, (float*)rowinv, (unsigned int*)epoch_buf);
     return (int)hipGetLastError();
 }
 
@@ -1145,11 +1185,11 @@ int ma_multiclass_curve_hist(uintptr_t stream, uintptr_t probs, int dtype, uintp
                              ll C, uintptr_t thresholds, int T, ll ignore_index, int has_ignore,
                              int mode, int uniform, float t0, float inv_step, int norm_kind,
                              uintptr_t flag, uintptr_t rowmax, uintptr_t rowinv, int variant,
-                             int c_chunk_override, uintptr_t hist) {
+                             int c_chunk_override, int stats_ready, uintptr_t hist) {
     hipStream_t s = (hipStream_t)stream;
     size_t shmem = (size_t)T * sizeof(float);
     if (shmem > 160 * 1024) return -100;
-    if (norm_kind == 1 && flag) {
+    if (norm_kind == 1 && flag && !stats_ready) {
         // softmax stats (one fused pass also sets the out-of-range flag).
         // Fewer blocks -> each wave loops several rows -> more loads in
         // flight per lane (the kernel is latency-bound at 2 vec-loads/lane).

@@ -46,13 +46,13 @@ _F = ctypes.c_float
 _D = ctypes.c_double
 
 _SIGNATURES = {
-    "ma_mc_stat_logits": [_U64, _U64, _I, _U64, _LL, _LL, _LL, _I, _U64, _U64, _U64, _U64, _U64, _U64],
+    "ma_mc_stat_logits": [_U64, _U64, _I, _U64, _LL, _LL, _LL, _I, _U64, _U64, _U64, _U64, _U64, _U64, _U64, _U64, _U64],
     "ma_mc_stat_labels": [_U64, _U64, _U64, _LL, _LL, _LL, _I, _U64, _U64, _U64, _U64, _U64],
     "ma_bincount": [_U64, _U64, _LL, _LL, _U64],
     "ma_binary_stat": [_U64, _U64, _I, _U64, _LL, _F, _LL, _I, _U64, _U64],
     "ma_multilabel_stat": [_U64, _U64, _I, _U64, _LL, _LL, _F, _LL, _I, _U64, _U64],
     "ma_binary_curve_hist": [_U64, _U64, _I, _U64, _LL, _U64, _I, _LL, _I, _I, _F, _F, _I, _U64, _U64],
-    "ma_multiclass_curve_hist": [_U64, _U64, _I, _U64, _LL, _LL, _U64, _I, _LL, _I, _I, _I, _F, _F, _I, _U64, _U64, _U64, _I, _I, _U64],
+    "ma_multiclass_curve_hist": [_U64, _U64, _I, _U64, _LL, _LL, _U64, _I, _LL, _I, _I, _I, _F, _F, _I, _U64, _U64, _U64, _I, _I, _I, _U64],
     "ma_curve_suffix": [_U64, _U64, _LL, _I, _I, _I, _U64, _U64],
     "ma_apply_stat_deltas": [_U64, _U64, _LL, _U64, _U64, _U64, _U64],
     "ma_exact_apply": [_U64, _U64, _LL, _LL, _I, _U64, _U64],
@@ -136,6 +136,7 @@ def mc_stat_logits(
         confmat.data_ptr() if confmat is not None else 0,
         valid.data_ptr(),
         argmax.data_ptr() if argmax is not None else 0,
+        0, 0, 0,
     )
     _check(rc, "ma_mc_stat_logits")
     return tp, fp, fn, valid, confmat, argmax
@@ -419,6 +420,7 @@ def multiclass_curve_confmat(
         0,
         _CURVE_VARIANT,
         _CURVE_CCHUNK,
+        0,
         hist.data_ptr(),
     )
     _check(rc, "ma_multiclass_curve_hist")
@@ -495,6 +497,7 @@ def box_iou(boxes1: Tensor, boxes2: Tensor, variant: str = "iou") -> Tensor:
 def curve_hist_into_confmat(
     probs: Tensor, target: Tensor, thresholds: Tensor, ignore_index: Optional[int],
     confmat_state: Tensor, mode: int, norm: Optional[str] = None, owner=None,
+    stats_ready: bool = False,
 ) -> None:
     """Bucketized histogram + transposed suffix-sum accumulated DIRECTLY into the
     metric's confmat state ((T,2,2) binary / (T,C,2,2) multiclass|multilabel) —
@@ -586,6 +589,7 @@ def curve_hist_into_confmat(
             ri_ptr,
             _CURVE_VARIANT,
             _CURVE_CCHUNK,
+            1 if stats_ready else 0,
             hist.data_ptr(),
         )
         _check(rc, "ma_multiclass_curve_hist")
@@ -640,6 +644,7 @@ def mc_stat_into(
             0,
             s_valid_ptr,
             0,
+            0, 0, 0,
         )
         _check(rc, "ma_mc_stat_logits")
     else:
@@ -705,6 +710,7 @@ def mc_confmat_into(
             confmat_state.data_ptr(),
             dummy[3 * C :].data_ptr(),
             0,
+            0, 0, 0,
         )
         _check(rc, "ma_mc_stat_logits")
     else:
@@ -759,6 +765,7 @@ def mc_exact_into(
             0,
             s_valid_ptr,
             0,
+            0, 0, 0,
         )
         _check(rc, "ma_mc_stat_logits")
     else:
@@ -840,7 +847,7 @@ def curve_auc_from_confmat(confmat_state: Tensor, mode: int) -> Tuple[Tensor, Te
 
 def mc_fused_collection_update(
     preds: Tensor, target: Tensor, num_classes: int, ignore_index: Optional[int],
-    stat=None, confmat=None, exact=None,
+    stat=None, confmat=None, exact=None, rowstats=None,
 ) -> None:
     """Collection-level fused update: ONE pass over the (B, C) logits feeds the
     stat-scores, confusion-matrix and exact-match leaders at once.
@@ -849,6 +856,9 @@ def mc_fused_collection_update(
     ``confmat`` = confmat state tensor (atomics accumulate directly) or None
     ``exact`` = (correct, total) or None (reads the shared scratch BEFORE the
     apply kernel consumes and zeroes it).
+    ``rowstats`` = (rowmax, rowinv, epoch_buf) or None: the same pass also
+    emits the softmax row statistics + out-of-range epoch flag that the
+    threshold-curve leader consumes (its own row-stats kernel is skipped).
     """
     lib = _lib()
     C = num_classes
@@ -866,6 +876,9 @@ def mc_fused_collection_update(
             scratch[:C].data_ptr(), scratch[C : 2 * C].data_ptr(), scratch[2 * C : 3 * C].data_ptr(),
             confmat.data_ptr() if confmat is not None else 0,
             s_valid_ptr, 0,
+            rowstats[0].data_ptr() if rowstats is not None else 0,
+            rowstats[1].data_ptr() if rowstats is not None else 0,
+            rowstats[2].data_ptr() if rowstats is not None else 0,
         )
         _check(rc, "ma_mc_stat_logits")
     else:
