@@ -66,7 +66,13 @@ class _FusedMulticlassUpdatePlan:
                 if getattr(leader, "multidim_average", "global") == "global":
                     exact = leader
             elif kind == "mc_curve" and curve is None:
-                if leader.thresholds is not None and getattr(leader, "average", None) != "micro":
+                import os
+
+                if (
+                    leader.thresholds is not None
+                    and getattr(leader, "average", None) != "micro"
+                    and os.environ.get("METRICS_AMD_FUSE_CURVE", "1") != "0"
+                ):
                     curve = leader
         participants = [m for m in (stat, confmat, exact, curve) if m is not None]
         if stat is None or len(participants) < 2:
