@@ -103,3 +103,18 @@ def test_sdr_cg_matches_direct_solve():
     direct = signal_distortion_ratio(preds, target)
     cg = signal_distortion_ratio(preds, target, use_cg_iter=20)
     assert torch.allclose(direct, cg, atol=0.05), (direct, cg)
+
+
+def test_external_dsp_stubs_raise():
+    """Reference behavior when optional DSP deps are missing: constructor raises."""
+    import pytest as _pytest
+
+    for cls, kwargs in [
+        (ma.audio.DeepNoiseSuppressionMeanOpinionScore, {"fs": 16000, "personalized": False}),
+        (ma.audio.NonIntrusiveSpeechQualityAssessment, {"fs": 16000}),
+        (ma.audio.SpeechReverberationModulationEnergyRatio, {"fs": 16000}),
+        (ma.audio.PerceptualEvaluationSpeechQuality, {"fs": 16000, "mode": "wb"}),
+        (ma.audio.ShortTimeObjectiveIntelligibility, {"fs": 16000}),
+    ]:
+        with _pytest.raises(ModuleNotFoundError):
+            cls(**kwargs)

@@ -1,0 +1,166 @@
+"""Coverage for modular classes not exercised elsewhere: sklearn oracles where
+available, reference doctest values otherwise, smoke for the rest."""
+import numpy as np
+import pytest
+import torch
+from sklearn import metrics as skm
+
+import metrics_amd as ma
+from tests.unittests._helpers import seed_all
+
+seed_all(61)
+N, L = 300, 4
+MLP = torch.rand(N, L)
+MLT = torch.randint(0, 2, (N, L))
+MLB = (MLP > 0.5).int().numpy()
+BP = torch.rand(N)
+BT = torch.randint(0, 2, (N,))
+
+
+def test_binary_stat_scores_and_confmat():
+    m = ma.BinaryStatScores()
+    m.update(BP, BT)
+    tp, fp, tn, fn, sup = m.compute()
+    pb = (BP > 0.5).int().numpy()
+    cm = skm.confusion_matrix(BT.numpy(), pb)
+    assert (tp, fp, tn, fn) == (cm[1, 1], cm[0, 1], cm[0, 0], cm[1, 0])
+    c = ma.BinaryConfusionMatrix()
+    c.update(BP, BT)
+    assert np.array_equal(c.compute().numpy(), cm)
+
+
+def test_binary_fbeta_and_logauc():
+    m = ma.BinaryFBetaScore(beta=0.5)
+    m.update(BP, BT)
+    pb = (BP > 0.5).int().numpy()
+    assert abs(float(m.compute()) - skm.fbeta_score(BT.numpy(), pb, beta=0.5)) < 1e-6
+    la = ma.BinaryLogAUC(thresholds=None)
+    la.update(BP, BT)
+    from metrics_amd.functional.classification import binary_logauc
+
+    assert torch.allclose(la.compute(), binary_logauc(BP, BT))
+
+
+@pytest.mark.parametrize(
+    ("cls", "kwargs"),
+    [
+        (ma.BinaryRecallAtFixedPrecision, {"min_precision": 0.5}),
+        (ma.BinaryPrecisionAtFixedRecall, {"min_recall": 0.5}),
+        (ma.BinarySensitivityAtSpecificity, {"min_specificity": 0.5}),
+        (ma.BinarySpecificityAtSensitivity, {"min_sensitivity": 0.5}),
+    ],
+)
+def test_binary_at_fixed_classes(cls, kwargs):
+    m = cls(thresholds=50, **kwargs)
+    m.update(BP, BT)
+    v, thr = m.compute()
+    assert 0 <= float(v) <= 1 and 0 <= float(thr) <= 1
+
+
+def test_multiclass_at_fixed_and_logauc_classes():
+    p = torch.randn(256, 5).softmax(-1)
+    t = torch.randint(0, 5, (256,))
+    for cls, kw in [
+        (ma.MulticlassRecallAtFixedPrecision, {"min_precision": 0.3}),
+        (ma.MulticlassPrecisionAtFixedRecall, {"min_recall": 0.3}),
+        (ma.MulticlassSensitivityAtSpecificity, {"min_specificity": 0.3}),
+        (ma.MulticlassSpecificityAtSensitivity, {"min_sensitivity": 0.3}),
+    ]:
+        m = cls(num_classes=5, thresholds=25, **kw)
+        m.update(p, t)
+        v, thr = m.compute()
+        assert v.shape == (5,)
+    la = ma.MulticlassLogAUC(num_classes=5, thresholds=50)
+    la.update(p, t)
+    assert la.compute().ndim == 0
+    # task-dispatch wrappers construct the right subclass
+    assert isinstance(ma.LogAUC(task="binary"), ma.BinaryLogAUC)
+    assert isinstance(ma.PrecisionAtFixedRecall(task="binary", min_recall=0.5), ma.BinaryPrecisionAtFixedRecall)
+    assert isinstance(ma.SensitivityAtSpecificity(task="binary", min_specificity=0.5), ma.BinarySensitivityAtSpecificity)
+    assert isinstance(ma.SpecificityAtSensitivity(task="binary", min_sensitivity=0.5), ma.BinarySpecificityAtSensitivity)
+
+
+def test_multilabel_family_vs_sklearn():
+    cases = [
+        (ma.MultilabelAccuracy(num_labels=L, average="macro"),
+         np.mean([(MLB[:, i] == MLT[:, i].numpy()).mean() for i in range(L)])),
+        (ma.MultilabelFBetaScore(num_labels=L, beta=2.0, average="macro"),
+         skm.fbeta_score(MLT.numpy(), MLB, beta=2.0, average="macro")),
+        (ma.MultilabelHammingDistance(num_labels=L, average="macro"),
+         1 - np.mean([(MLB[:, i] == MLT[:, i].numpy()).mean() for i in range(L)])),
+        (ma.MultilabelJaccardIndex(num_labels=L, average="macro"),
+         skm.jaccard_score(MLT.numpy(), MLB, average="macro")),
+        (ma.MultilabelSpecificity(num_labels=L, average="macro"),
+         np.mean([skm.recall_score(1 - MLT[:, i].numpy(), 1 - MLB[:, i]) for i in range(L)])),
+        (ma.MultilabelNegativePredictiveValue(num_labels=L, average="macro"),
+         np.mean([skm.precision_score(1 - MLT[:, i].numpy(), 1 - MLB[:, i]) for i in range(L)])),
+    ]
+    for m, ref in cases:
+        m.update(MLP, MLT)
+        assert abs(float(m.compute()) - ref) < 1e-6, m.__class__.__name__
+    cm = ma.MultilabelConfusionMatrix(num_labels=L)
+    cm.update(MLP, MLT)
+    assert np.array_equal(cm.compute().numpy(), skm.multilabel_confusion_matrix(MLT.numpy(), MLB))
+    mcc = ma.MultilabelMatthewsCorrCoef(num_labels=L)
+    mcc.update(MLP, MLT)
+    summed = skm.multilabel_confusion_matrix(MLT.numpy(), MLB).sum(0)
+    tn, fp, fn_, tp = summed.ravel()
+    denom = np.sqrt(float((tp + fp) * (tp + fn_) * (tn + fp) * (tn + fn_)))
+    assert abs(float(mcc.compute()) - (tp * tn - fp * fn_) / denom) < 1e-5
+    ss = ma.MultilabelStatScores(num_labels=L, average=None)
+    ss.update(MLP, MLT)
+    out = ss.compute()
+    assert out.shape == (L, 5)
+
+
+def test_multilabel_curves():
+    prc = ma.MultilabelPrecisionRecallCurve(num_labels=L, thresholds=None)
+    prc.update(MLP, MLT)
+    precisions, recalls, thrs = prc.compute()
+    skp, skr, _ = skm.precision_recall_curve(MLT[:, 1].numpy(), MLP[:, 1].numpy())
+    assert np.allclose(sorted(precisions[1].numpy()), sorted(skp), atol=1e-6)
+    roc = ma.MultilabelROC(num_labels=L, thresholds=None)
+    roc.update(MLP, MLT)
+    fprs, tprs, _ = roc.compute()
+    sk_fpr, sk_tpr, _ = skm.roc_curve(MLT[:, 2].numpy(), MLP[:, 2].numpy())
+    grid = np.linspace(0, 1, 30)
+    assert np.allclose(np.interp(grid, sk_fpr, sk_tpr), np.interp(grid, fprs[2].numpy(), tprs[2].numpy()), atol=1e-6)
+
+
+def test_multiclass_roc_class_vs_sklearn():
+    p = torch.randn(200, 4).softmax(-1)
+    t = torch.randint(0, 4, (200,))
+    m = ma.MulticlassROC(num_classes=4, thresholds=None)
+    m.update(p, t)
+    fpr, tpr, _ = m.compute()
+    grid = np.linspace(0, 1, 30)
+    for c in range(4):
+        sk_fpr, sk_tpr, _ = skm.roc_curve((t == c).int().numpy(), p[:, c].numpy())
+        assert np.allclose(np.interp(grid, sk_fpr, sk_tpr), np.interp(grid, fpr[c].numpy(), tpr[c].numpy()), atol=1e-6)
+
+
+def test_hinge_loss_reference_values():
+    preds = torch.tensor([[0.25, 0.20, 0.55], [0.55, 0.05, 0.40], [0.10, 0.30, 0.60], [0.90, 0.05, 0.05]])
+    target = torch.tensor([0, 1, 2, 0])
+    m = ma.MulticlassHingeLoss(num_classes=3)
+    m.update(preds, target)
+    assert abs(float(m.compute()) - 0.9125) < 1e-4
+    m2 = ma.MulticlassHingeLoss(num_classes=3, squared=True)
+    m2.update(preds, target)
+    assert abs(float(m2.compute()) - 1.1131) < 1e-4
+    m3 = ma.MulticlassHingeLoss(num_classes=3, multiclass_mode="one-vs-all")
+    m3.update(preds, target)
+    assert torch.allclose(m3.compute(), torch.tensor([0.8750, 1.1250, 1.1000]), atol=1e-4)
+
+
+def test_fairness_classes():
+    groups = torch.randint(0, 2, (N,))
+    f = ma.BinaryFairness(num_groups=2)
+    f.update(BP, BT, groups)
+    out = f.compute()
+    assert any(k.startswith("DP_") for k in out) and any(k.startswith("EO_") for k in out)
+    r = ma.BinaryGroupStatRates(num_groups=2)
+    r.update(BP, BT, groups)
+    rates = r.compute()
+    assert set(rates) == {"group_0", "group_1"}
+    assert torch.allclose(rates["group_0"].sum(), torch.tensor(1.0))

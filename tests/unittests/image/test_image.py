@@ -208,3 +208,34 @@ def test_uniform_filter_matches_scipy():
             np.stack([[uniform_filter(x[b, c].numpy(), size=w, mode="reflect") for c in range(3)] for b in range(2)])
         )
         assert torch.allclose(mine, ref, atol=1e-12), w
+
+
+def test_perceptual_path_length_toy():
+    class ToyGen(torch.nn.Module):
+        num_samples = None
+
+        def __init__(self):
+            super().__init__()
+            torch.manual_seed(0)
+            self.lin = torch.nn.Linear(8, 3 * 8 * 8)
+
+        def sample(self, n):
+            return torch.randn(n, 8)
+
+        def forward(self, z):
+            return self.lin(z).reshape(-1, 3, 8, 8)
+
+    def sim(a, b):
+        return ((a - b) ** 2).flatten(1).mean(1)
+
+    m = ma.PerceptualPathLength(num_samples=64, batch_size=32, sim_net=sim, lower_discard=None, upper_discard=None)
+    m.update(ToyGen())
+    mean, std, med = m.compute()
+    assert mean.item() > 0 and std.item() >= 0
+
+
+def test_clip_iqa_and_infolm_raise_without_models():
+    with pytest.raises(ModuleNotFoundError):
+        ma.multimodal.CLIPImageQualityAssessment()
+    with pytest.raises(ModuleNotFoundError):
+        ma.text.InfoLM()

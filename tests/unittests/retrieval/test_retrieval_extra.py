@@ -1,0 +1,55 @@
+"""RetrievalAUROC + wrapper bits not covered elsewhere."""
+import torch
+from sklearn.metrics import roc_auc_score
+
+import metrics_amd as ma
+from tests.unittests._helpers import seed_all
+
+
+def test_retrieval_auroc_vs_sklearn():
+    seed_all(62)
+    idx = torch.repeat_interleave(torch.arange(6), 20)
+    preds = torch.rand(120)
+    target = torch.randint(0, 2, (120,))
+    # make sure every group has both classes
+    for g in range(6):
+        target[g * 20] = 1
+        target[g * 20 + 1] = 0
+    m = ma.RetrievalAUROC()
+    m.update(preds, target, indexes=idx)
+    ref = sum(
+        roc_auc_score(target[g * 20 : (g + 1) * 20].numpy(), preds[g * 20 : (g + 1) * 20].numpy()) for g in range(6)
+    ) / 6
+    assert abs(float(m.compute()) - ref) < 1e-6
+
+
+def test_fleiss_kappa_class():
+    seed_all(63)
+    ratings = torch.zeros(20, 3, dtype=torch.long)
+    for i in range(20):
+        picks = torch.randint(0, 3, (5,))
+        for p in picks.tolist():
+            ratings[i, p] += 1
+    m = ma.FleissKappa(mode="counts")
+    m.update(ratings)
+    from metrics_amd.functional.nominal import fleiss_kappa
+
+    assert torch.allclose(m.compute(), fleiss_kappa(ratings, mode="counts"))
+
+
+def test_procrustes_class_and_wrappers():
+    a = torch.randn(4, 12, 3)
+    m = ma.shape.ProcrustesDisparity()
+    m.update(a, a)
+    assert float(m.compute()) < 1e-10
+
+    # MetricInputTransformer base: subclass transforms inputs
+    from metrics_amd.wrappers import MetricInputTransformer
+
+    class Halve(MetricInputTransformer):
+        def transform_pred(self, pred):
+            return pred / 2
+
+    w = Halve(ma.MeanSquaredError())
+    w.update(torch.ones(4) * 2, torch.zeros(4))
+    assert abs(float(w.compute()) - 1.0) < 1e-6
