@@ -164,3 +164,25 @@ def test_fairness_classes():
     rates = r.compute()
     assert set(rates) == {"group_0", "group_1"}
     assert torch.allclose(rates["group_0"].sum(), torch.tensor(1.0))
+
+
+def test_multilabel_at_fixed_and_logauc_classes():
+    for cls, kw in [
+        (ma.MultilabelRecallAtFixedPrecision, {"min_precision": 0.3}),
+        (ma.MultilabelPrecisionAtFixedRecall, {"min_recall": 0.3}),
+        (ma.MultilabelSensitivityAtSpecificity, {"min_specificity": 0.3}),
+        (ma.MultilabelSpecificityAtSensitivity, {"min_sensitivity": 0.3}),
+    ]:
+        m = cls(num_labels=L, thresholds=25, **kw)
+        m.update(MLP, MLT)
+        v, thr = m.compute()
+        assert v.shape == (L,), cls.__name__
+    la = ma.MultilabelLogAUC(num_labels=L, thresholds=25)
+    la.update(MLP, MLT)
+    assert la.compute().ndim == 0
+    # abstract bases are exported for subclassing (reference parity)
+    from metrics_amd.retrieval import RetrievalMetric
+    from metrics_amd.wrappers import WrapperMetric
+
+    assert issubclass(ma.RetrievalMRR, RetrievalMetric)
+    assert issubclass(ma.wrappers.Running, WrapperMetric)
