@@ -665,3 +665,63 @@ def test_fused_plan_not_built_with_validate_args():
     coll.update(torch.randn(64, 5, device="cuda"), torch.randint(0, 5, (64,), device="cuda"))
     coll.update(torch.randn(64, 5, device="cuda"), torch.randint(0, 5, (64,), device="cuda"))
     assert coll._fused_plan is None  # validations would be skipped -> no fusion
+
+
+@pytest.mark.parametrize("C", [3, 7, 63, 65, 1001, 4096])
+def test_stat_kernel_boundary_class_counts(C):
+    """Odd/unaligned C exercises the scalar load path; large C the vector path."""
+    torch.manual_seed(26)
+    B = 511
+    preds = torch.randn(B, C)
+    target = torch.randint(0, C, (B,))
+    g = ma.MulticlassAccuracy(num_classes=C, average="micro").to("cuda")
+    c = ma.MulticlassAccuracy(num_classes=C, average="micro")
+    g.update(preds.cuda(), target.cuda())
+    c.update(preds, target)
+    assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-6), C
+
+
+@pytest.mark.parametrize("T", [1, 2, 999, 4000])
+def test_curve_kernel_boundary_thresholds(T):
+    torch.manual_seed(27)
+    preds = torch.rand(10_000)
+    target = torch.randint(0, 2, (10_000,))
+    g = ma.BinaryPrecisionRecallCurve(thresholds=T).to("cuda")
+    c = ma.BinaryPrecisionRecallCurve(thresholds=T)
+    g.update(preds.cuda(), target.cuda())
+    c.update(preds, target)
+    assert torch.equal(g.confmat.cpu(), c.confmat), T
+
+
+def test_curve_kernel_large_binary_input():
+    torch.manual_seed(28)
+    N = 4_000_000
+    preds = torch.rand(N, device="cuda")
+    target = torch.randint(0, 2, (N,), device="cuda")
+    m = ma.BinaryAUROC(thresholds=500).to("cuda")
+    m.update(preds, target)
+    v = float(m.compute())
+    assert 0.49 < v < 0.51  # random scores -> ~0.5
+
+
+def test_mc_curve_nonuniform_thresholds():
+    torch.manual_seed(29)
+    thr = torch.tensor([0.01, 0.2, 0.21, 0.5, 0.93])  # non-uniform grid: binary-search path
+    preds = torch.randn(2048, 9).softmax(-1)
+    target = torch.randint(0, 9, (2048,))
+    g = ma.MulticlassPrecisionRecallCurve(num_classes=9, thresholds=thr).to("cuda")
+    c = ma.MulticlassPrecisionRecallCurve(num_classes=9, thresholds=thr)
+    g.update(preds.cuda(), target.cuda())
+    c.update(preds, target)
+    assert torch.equal(g.confmat.cpu(), c.confmat)
+
+
+def test_tiny_batches_and_single_row():
+    for B in (1, 2, 63):
+        preds = torch.randn(B, 17)
+        target = torch.randint(0, 17, (B,))
+        g = ma.MulticlassF1Score(num_classes=17, average="macro").to("cuda")
+        c = ma.MulticlassF1Score(num_classes=17, average="macro")
+        g.update(preds.cuda(), target.cuda())
+        c.update(preds, target)
+        assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-6), B
