@@ -725,3 +725,25 @@ def test_tiny_batches_and_single_row():
         g.update(preds.cuda(), target.cuda())
         c.update(preds, target)
         assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-6), B
+
+
+def test_noncontiguous_inputs_gpu():
+    """Kernels must handle (or contiguous-ify) transposed/strided inputs."""
+    torch.manual_seed(30)
+    base = torch.randn(40, 2048, device="cuda")  # transpose -> (2048, 40) non-contig
+    preds = base.T
+    assert not preds.is_contiguous()
+    target = torch.randint(0, 40, (2048,), device="cuda")
+    g = ma.MulticlassAccuracy(num_classes=40, average="macro").to("cuda")
+    c = ma.MulticlassAccuracy(num_classes=40, average="macro")
+    g.update(preds, target)
+    c.update(preds.cpu(), target.cpu())
+    assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-6)
+    # strided binary preds
+    bp = torch.rand(20_000, 2, device="cuda")[:, 0]
+    bt = torch.randint(0, 2, (20_000,), device="cuda")
+    gb = ma.BinaryF1Score().to("cuda")
+    cb = ma.BinaryF1Score()
+    gb.update(bp, bt)
+    cb.update(bp.cpu(), bt.cpu())
+    assert torch.allclose(gb.compute().cpu(), cb.compute(), atol=1e-6)

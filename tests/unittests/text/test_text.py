@@ -147,3 +147,40 @@ def test_text_reference_doctest_values():
     assert abs(float(word_information_preserved(p2, t2)) - 0.3472) < 5e-4
     r = rouge_score(["My name is John"], ["Is your name John"])
     assert abs(float(r["rouge1_fmeasure"]) - 0.75) < 1e-4
+
+
+def test_bleu_weights_and_ngram_args():
+    from metrics_amd.functional.text import bleu_score
+
+    preds = ["the cat is on the mat"]
+    target = [["there is a cat on the mat", "a cat is on the mat"]]
+    v2 = bleu_score(preds, target, n_gram=2)
+    v4 = bleu_score(preds, target, n_gram=4)
+    assert v2 > v4 > 0
+    vw = bleu_score(preds, target, n_gram=2, weights=[0.75, 0.25])
+    assert 0 < float(vw) <= 1
+    # smoothing on a no-overlap high-order case must not be zero/ nan
+    vs = bleu_score(["a b"], [["c d"]], n_gram=2, smooth=True)
+    assert float(vs) >= 0
+
+
+def test_rouge_keys_and_aggregation():
+    from metrics_amd.functional.text import rouge_score
+
+    r = rouge_score(["the cat sat"], ["the cat sat on the mat"], rouge_keys=("rouge1", "rouge2", "rougeL"))
+    assert set(r) == {f"{k}_{s}" for k in ("rouge1", "rouge2", "rougeL") for s in ("fmeasure", "precision", "recall")}
+    assert float(r["rouge1_recall"]) == pytest.approx(3 / 6, abs=1e-6)
+    assert float(r["rouge1_precision"]) == pytest.approx(1.0, abs=1e-6)
+
+
+def test_wer_accumulation_matches_corpus():
+    m = ma.text.WordErrorRate()
+    m.update(["this is the prediction"], ["this is the reference"])
+    m.update(["there is an other sample"], ["there is another one"])
+    from metrics_amd.functional.text import word_error_rate
+
+    corpus = word_error_rate(
+        ["this is the prediction", "there is an other sample"],
+        ["this is the reference", "there is another one"],
+    )
+    assert torch.allclose(m.compute(), corpus)
