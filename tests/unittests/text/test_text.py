@@ -1,4 +1,5 @@
 """Text metric tests vs known values."""
+import pytest
 import torch
 
 import metrics_amd as ma
