@@ -512,10 +512,6 @@ class Metric(Module, ABC):
         """Return a deep copy of the metric."""
         return deepcopy(self)
 
-    def __iter__(self):
-        """Metrics are not iterable (reference metric.py parity)."""
-        raise NotImplementedError("Metrics does not support iteration.")
-
     def __getstate__(self) -> Dict[str, Any]:
         # ignore update and compute functions for pickling
         return {k: v for k, v in self.__dict__.items() if k not in ("update", "compute", "_update_signature")}
