@@ -53,3 +53,19 @@ def test_procrustes_class_and_wrappers():
     w = Halve(ma.MeanSquaredError())
     w.update(torch.ones(4) * 2, torch.zeros(4))
     assert abs(float(w.compute()) - 1.0) < 1e-6
+
+
+def test_empty_target_action_semantics():
+    idx = torch.tensor([0, 0, 1, 1])
+    preds = torch.tensor([0.9, 0.1, 0.8, 0.2])
+    target = torch.tensor([0, 0, 1, 0])  # group 0 has no positives
+    for action, expect in [("skip", 1.0), ("neg", 0.5), ("pos", 1.0)]:
+        m = ma.RetrievalMRR(empty_target_action=action)
+        m.update(preds, target, indexes=idx)
+        assert abs(float(m.compute()) - expect) < 1e-6, action
+    import pytest as _pytest
+
+    m = ma.RetrievalMRR(empty_target_action="error")
+    m.update(preds, target, indexes=idx)
+    with _pytest.raises(ValueError):
+        m.compute()
