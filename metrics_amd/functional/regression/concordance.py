@@ -4,17 +4,17 @@ from __future__ import annotations
 import torch
 from torch import Tensor
 
-from metrics_amd.functional.regression.pearson import _pearson_corrcoef_update
+from metrics_amd.functional.regression.pearson import _pearson_corrcoef_compute, _pearson_corrcoef_update
 
 
 def _concordance_corrcoef_compute(
     mean_x: Tensor, mean_y: Tensor, var_x: Tensor, var_y: Tensor, corr_xy: Tensor, nb: Tensor
 ) -> Tensor:
-    """CCC from the running pearson moments."""
-    var_x = var_x / nb
-    var_y = var_y / nb
-    corr_xy = corr_xy / nb
-    return (2.0 * corr_xy / (var_x + var_y + (mean_x - mean_y) ** 2)).squeeze()
+    """CCC from the running pearson moments (reference functional/regression/concordance.py:29)."""
+    pearson = _pearson_corrcoef_compute(var_x, var_y, corr_xy, nb)
+    var_x = var_x / (nb - 1)
+    var_y = var_y / (nb - 1)
+    return 2.0 * pearson * var_x.sqrt() * var_y.sqrt() / (var_x + var_y + (mean_x - mean_y) ** 2)
 
 
 def concordance_corrcoef(preds: Tensor, target: Tensor) -> Tensor:

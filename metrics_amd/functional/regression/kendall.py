@@ -10,6 +10,8 @@ from __future__ import annotations
 
 from typing import Optional, Tuple
 
+import math
+
 import numpy as np
 import torch
 from torch import Tensor
@@ -36,14 +38,28 @@ def _inversions(a: np.ndarray) -> Tuple[int, np.ndarray]:
     return inv_l + inv_r + cross, np.sort(np.concatenate([left, right]), kind="stable")
 
 
-def _tie_pairs(sorted_keys: np.ndarray) -> int:
-    """Sum of t*(t-1)/2 over runs of equal values in a sorted array."""
-    if sorted_keys.size < 2:
-        return 0
+def _tie_group_sizes(sorted_keys: np.ndarray) -> np.ndarray:
+    if sorted_keys.size == 0:
+        return np.zeros(0, dtype=np.int64)
     change = np.nonzero(sorted_keys[1:] != sorted_keys[:-1])[0] + 1
     idx = np.concatenate([[0], change, [sorted_keys.size]])
-    t = np.diff(idx).astype(np.int64)
+    return np.diff(idx).astype(np.int64)
+
+
+def _tie_pairs(sorted_keys: np.ndarray) -> int:
+    """Sum of t*(t-1)/2 over runs of equal values in a sorted array."""
+    t = _tie_group_sizes(sorted_keys)
     return int((t * (t - 1) // 2).sum())
+
+
+def _tie_stats(sorted_keys: np.ndarray):
+    """(pairs, p1, p2) tie statistics for the tau-b significance test:
+    pairs = sum t(t-1)/2, p1 = sum t(t-1)(t-2), p2 = sum t(t-1)(2t+5)."""
+    t = _tie_group_sizes(sorted_keys)
+    pairs = float((t * (t - 1) // 2).sum())
+    p1 = float((t * (t - 1) * (t - 2)).sum())
+    p2 = float((t * (t - 1) * (2 * t + 5)).sum())
+    return pairs, p1, p2
 
 
 def _tie_pairs2(x: np.ndarray, y: np.ndarray) -> int:
@@ -56,16 +72,18 @@ def _tie_pairs2(x: np.ndarray, y: np.ndarray) -> int:
     return int((t * (t - 1) // 2).sum())
 
 
-def _count_pairs(x: Tensor, y: Tensor) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
-    """Return (concordant, discordant, ties_x, ties_y) pair counts (Knight)."""
+def _count_pairs(x: Tensor, y: Tensor):
+    """(concordant, discordant, ties_x, ties_y, tie-stat tuples) via Knight's algorithm."""
     xn = x.detach().cpu().numpy().astype(np.float64)
     yn = y.detach().cpu().numpy().astype(np.float64)
     n = xn.size
     order = np.lexsort((yn, xn))  # sort by x, ties broken by y ascending
     xs, ys = xn[order], yn[order]
     tot = n * (n - 1) // 2
-    xtie = _tie_pairs(xs)
-    ytie = _tie_pairs(np.sort(yn, kind="stable"))
+    x_stats = _tie_stats(xs)
+    y_stats = _tie_stats(np.sort(yn, kind="stable"))
+    xtie = x_stats[0]
+    ytie = y_stats[0]
     xytie = _tie_pairs2(xs, ys)
     dis, _ = _inversions(ys)
     both_untied = tot - xtie - ytie + xytie
@@ -73,7 +91,7 @@ def _count_pairs(x: Tensor, y: Tensor) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
 
     dev = x.device
     as_t = lambda v: torch.tensor(float(v), dtype=torch.float64, device=dev)
-    return as_t(con), as_t(dis), as_t(xtie), as_t(ytie)
+    return as_t(con), as_t(dis), as_t(xtie), as_t(ytie), x_stats, y_stats
 
 
 def _kendall_corrcoef_compute(
@@ -83,14 +101,16 @@ def _kendall_corrcoef_compute(
         preds = preds.unsqueeze(1)
         target = target.unsqueeze(1)
     taus = []
+    p_values = []
     for d in range(preds.shape[1]):
         x = preds[:, d].double()
         y = target[:, d].double()
         n = x.numel()
-        con, dis, tie_x, tie_y = _count_pairs(x, y)
+        con, dis, tie_x, tie_y, x_stats, y_stats = _count_pairs(x, y)
         n0 = n * (n - 1) / 2
         if variant == "a":
-            tau = (con - dis) / n0
+            # reference tau-a: (con - dis) / (con + dis)  (ties excluded)
+            tau = (con - dis) / (con + dis)
         elif variant == "b":
             tau = (con - dis) / torch.sqrt((n0 - tie_x) * (n0 - tie_y))
         elif variant == "c":
@@ -99,23 +119,39 @@ def _kendall_corrcoef_compute(
         else:
             raise ValueError(f"Unknown variant {variant}")
         taus.append(tau.float())
+        if alternative is not None:
+            # reference tie-corrected significance test
+            # (reference functional/regression/kendall.py:192-222)
+            cmd = float(con - dis)
+            base = n * (n - 1) * (2 * n + 5)
+            if variant == "a":
+                t_value = 3 * cmd / math.sqrt(base / 2)
+            else:
+                mm = n * (n - 1)
+                denom = (base - x_stats[2] - y_stats[2]) / 18
+                denom += (2 * x_stats[0] * y_stats[0]) / mm
+                # n == 2 gives 0/0 here in the reference (always nan)
+                denom += x_stats[1] * y_stats[1] / (9 * mm * (n - 2)) if n != 2 else float("nan")
+                t_value = cmd / math.sqrt(denom) if denom == denom and denom > 0 else float("nan")
+            p_values.append(t_value)
     tau_t = torch.stack(taus).squeeze()
 
     p_value = None
     if alternative is not None:
-        # normal approximation for the two-sided test
-        n = preds.shape[0]
-        var = torch.tensor(2.0 * (2 * n + 5) / (9 * n * (n - 1)), device=tau_t.device)
-        z = tau_t / var.sqrt()
-        normal = torch.distributions.Normal(0.0, 1.0)
-        if alternative == "two-sided":
-            p_value = 2 * (1 - normal.cdf(z.abs()))
-        elif alternative == "greater":
-            p_value = 1 - normal.cdf(z)
-        elif alternative == "less":
-            p_value = normal.cdf(z)
-        else:
+        if alternative not in ("two-sided", "greater", "less"):
             raise ValueError(f"Unknown alternative {alternative}")
+        t = torch.tensor(p_values, dtype=torch.float32)
+        if alternative == "two-sided":
+            t = t.abs()
+        if alternative in ("two-sided", "greater"):
+            t = -t
+        normal = torch.distributions.Normal(0.0, 1.0)
+        is_nan = t.isnan()
+        p_value = normal.cdf(t.nan_to_num())
+        p_value = p_value.where(~is_nan, torch.tensor(float("nan")))
+        if alternative == "two-sided":
+            p_value = 2 * p_value
+        p_value = p_value.squeeze()
     return tau_t, p_value
 
 

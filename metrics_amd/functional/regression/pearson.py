@@ -53,9 +53,12 @@ def _pearson_corrcoef_update(
 
 def _pearson_corrcoef_compute(var_x: Tensor, var_y: Tensor, corr_xy: Tensor, nb: Tensor) -> Tensor:
     """Correlation from the accumulated second moments."""
-    var_x /= nb - 1
-    var_y /= nb - 1
-    corr_xy /= nb - 1
+    out_dtype = corr_xy.dtype
+    # NON-inplace: these are the metric's live state tensors (a second
+    # compute() call must not re-divide them)
+    var_x = var_x / (nb - 1)
+    var_y = var_y / (nb - 1)
+    corr_xy = corr_xy / (nb - 1)
     # compute in double for numerical stability on near-constant inputs
     if var_x.dtype == torch.float32:
         var_x = var_x.double()
@@ -73,7 +76,7 @@ def _pearson_corrcoef_compute(var_x: Tensor, var_y: Tensor, corr_xy: Tensor, nb:
             stacklevel=2,
         )
     corrcoef = (corr_xy / (var_x * var_y).sqrt()).squeeze()
-    return torch.clamp(corrcoef, -1.0, 1.0).to(corr_xy.dtype)
+    return torch.clamp(corrcoef, -1.0, 1.0).to(out_dtype)
 
 
 def _final_aggregation(
