@@ -10,6 +10,7 @@ from metrics_amd.classification.base import _ClassificationTaskWrapper
 from metrics_amd.utilities.data import dim_zero_cat
 from metrics_amd.utilities.enums import ClassificationTaskNoMultilabel
 from metrics_amd.functional.classification.calibration_error import (
+    _binary_calibration_error_update,
     _binary_calibration_error_arg_validation,
     _binary_calibration_error_tensor_validation,
     _ce_compute,
@@ -59,13 +60,11 @@ class BinaryCalibrationError(Metric):
         if self.validate_args:
             _binary_calibration_error_tensor_validation(preds, target, self.ignore_index)
         preds, target = _binary_confusion_matrix_format(
-            preds, target, threshold=0.5, ignore_index=self.ignore_index, convert_to_labels=False
+            preds, target, threshold=0.0, ignore_index=self.ignore_index, convert_to_labels=False
         )
-        pred_labels = (preds >= 0.5).long()
-        accuracies = (pred_labels == target).float()
-        confidences = torch.where(pred_labels == 1, preds, 1 - preds)
+        confidences, accuracies = _binary_calibration_error_update(preds, target)
         self.confidences.append(confidences.float())
-        self.accuracies.append(accuracies)
+        self.accuracies.append(accuracies.float())
 
     def compute(self) -> Tensor:
         """Calibration error."""

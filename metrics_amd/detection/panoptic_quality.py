@@ -211,9 +211,11 @@ class PanopticQuality(Metric):
         rq_per = torch.where(valid, self.true_positives / denom.clamp(min=1e-9), torch.zeros_like(self.iou_sum))
 
         if self.return_per_class:
+            # reference shapes (panoptic_qualities.py:223-226): (C, 3) with
+            # sq_and_rq, else (1, C)
             if self.return_sq_and_rq:
-                return torch.stack([pq_per, sq_per, rq_per], dim=-1).unsqueeze(0)
-            return pq_per.unsqueeze(0)
+                return torch.stack([pq_per, sq_per, rq_per], dim=-1)
+            return pq_per.view(1, -1)
         pq = pq_per[valid].mean() if valid.any() else torch.tensor(0.0, dtype=torch.double)
         if self.return_sq_and_rq:
             sq = sq_per[valid].mean() if valid.any() else torch.tensor(0.0, dtype=torch.double)

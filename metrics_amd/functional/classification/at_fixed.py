@@ -37,27 +37,48 @@ from metrics_amd.functional.classification.roc import (
 )
 
 
-def _best_at_constraint(
-    y: Tensor, x: Tensor, thresholds: Tensor, min_x: float
-) -> Tuple[Tensor, Tensor]:
-    """max y such that x >= min_x, and the threshold achieving it."""
-    try:
-        max_y = torch.max(y[x >= min_x])
-        idx = torch.nonzero((x >= min_x) & (y == max_y))[0]
-        thr = thresholds[min(int(idx), len(thresholds) - 1)] if thresholds.numel() else torch.tensor(1e6)
-    except (ValueError, IndexError, RuntimeError):
-        max_y = torch.tensor(0.0, device=y.device, dtype=y.dtype)
-        thr = torch.tensor(1e6, device=y.device)
-    return max_y, thr
+def _lexargmax(x: Tensor) -> Tensor:
+    """Index of the lexicographic maximum row (reference recall_fixed_precision.py:40)."""
+    idx = None
+    for k in range(x.shape[1]):
+        col = x[idx, k] if idx is not None else x[:, k]
+        z = torch.where(col == col.max())[0]
+        idx = z if idx is None else idx[z]
+        if len(idx) < 2:
+            break
+    if idx is None:
+        raise ValueError("Failed to extract index")
+    return idx
 
 
 def _recall_at_precision(precision: Tensor, recall: Tensor, thresholds: Tensor, min_precision: float):
-    recall_at, thr = _best_at_constraint(recall, precision, thresholds, min_precision)
-    return recall_at, thr
+    """Reference recall_fixed_precision.py:58 — lexicographic max over
+    (recall, precision, threshold) among rows with precision >= min."""
+    max_recall = torch.tensor(0.0, device=recall.device, dtype=recall.dtype)
+    best_threshold = torch.tensor(0)
+    zipped_len = min(t.shape[0] for t in (recall, precision, thresholds))
+    zipped = torch.vstack((recall[:zipped_len], precision[:zipped_len], thresholds[:zipped_len])).T
+    zipped_masked = zipped[zipped[:, 1] >= min_precision]
+    if zipped_masked.shape[0] > 0:
+        idx = _lexargmax(zipped_masked)[0]
+        max_recall, _, best_threshold = zipped_masked[idx]
+    if max_recall == 0.0:
+        best_threshold = torch.tensor(1e6, device=thresholds.device, dtype=thresholds.dtype)
+    return max_recall, best_threshold
 
 
 def _precision_at_recall(precision: Tensor, recall: Tensor, thresholds: Tensor, min_recall: float):
-    return _best_at_constraint(precision, recall, thresholds, min_recall)
+    """Reference precision_fixed_recall.py:42 — python tuple-max over (p, r, t)."""
+    try:
+        max_precision, _, best_threshold = max(
+            (p, r, t) for p, r, t in zip(precision, recall, thresholds) if r >= min_recall
+        )
+    except ValueError:
+        max_precision = torch.tensor(0.0, device=precision.device, dtype=precision.dtype)
+        best_threshold = torch.tensor(0)
+    if max_precision == 0.0:
+        best_threshold = torch.tensor(1e6, device=thresholds.device, dtype=thresholds.dtype)
+    return max_precision, best_threshold
 
 
 def binary_recall_at_fixed_precision(
@@ -214,13 +235,27 @@ def multilabel_precision_at_fixed_recall(
 
 # ---------------------------------------------------------------- ROC variants
 def _sens_at_spec(fpr: Tensor, tpr: Tensor, thresholds: Tensor, min_specificity: float):
+    """Reference sensitivity_specificity.py:47 — argmax sensitivity where spec >= min."""
     specificity = 1 - fpr
-    return _best_at_constraint(tpr, specificity, thresholds, min_specificity)
+    indices = specificity >= min_specificity
+    if not indices.any():
+        return (torch.tensor(0.0, device=tpr.device, dtype=tpr.dtype),
+                torch.tensor(1e6, device=thresholds.device, dtype=thresholds.dtype))
+    sens, thr = tpr[indices], thresholds[indices]
+    idx = torch.argmax(sens)
+    return sens[idx], thr[idx]
 
 
 def _spec_at_sens(fpr: Tensor, tpr: Tensor, thresholds: Tensor, min_sensitivity: float):
+    """Reference specificity_sensitivity.py:48 — argmax specificity where sens >= min."""
     specificity = 1 - fpr
-    return _best_at_constraint(specificity, tpr, thresholds, min_sensitivity)
+    indices = tpr >= min_sensitivity
+    if not indices.any():
+        return (torch.tensor(0.0, device=specificity.device, dtype=specificity.dtype),
+                torch.tensor(1e6, device=thresholds.device, dtype=thresholds.dtype))
+    spec, thr = specificity[indices], thresholds[indices]
+    idx = torch.argmax(spec)
+    return spec[idx], thr[idx]
 
 
 def binary_sensitivity_at_specificity(
@@ -371,25 +406,38 @@ def multilabel_specificity_at_sensitivity(
 
 # -------------------------------------------------------------------- LogAUC
 def _logauc_compute(fpr: Tensor, tpr: Tensor, fpr_range: Tuple[float, float]) -> Tensor:
-    """Area under the ROC in log10(fpr) space over ``fpr_range``, normalized."""
-    lo, hi = fpr_range
-    device = fpr.device
-    # interpolate tpr at the range bounds
-    from metrics_amd.utilities.data import interp
+    """Area under the ROC in log10(fpr) space over ``fpr_range``, normalized.
 
-    bounds = torch.tensor([lo, hi], dtype=fpr.dtype, device=device)
-    # fpr is increasing; add eps for duplicate handling
-    uniq, idx = torch.unique(fpr, return_inverse=True)
-    # take max tpr per unique fpr (best case at that fpr)
-    tpr_u = torch.zeros_like(uniq)
-    tpr_u.scatter_reduce_(0, idx, tpr, reduce="amax")
-    tpr_bounds = interp(bounds, uniq, tpr_u)
-    mask = (uniq > lo) & (uniq < hi)
-    x = torch.cat([bounds[:1], uniq[mask], bounds[1:]])
-    y = torch.cat([tpr_bounds[:1], tpr_u[mask], tpr_bounds[1:]])
-    x = torch.clamp(x, min=1e-10)
-    area = torch.trapz(y, torch.log10(x))
-    return area / (torch.log10(torch.tensor(hi)) - torch.log10(torch.tensor(lo)))
+    Reference functional/classification/logauc.py:35 (_binary_logauc_compute):
+    interpolated bounds are appended and fpr/tpr sorted INDEPENDENTLY (both
+    are monotone along the curve), then the log-space trapezoid is taken
+    between the last occurrences of the two bounds.
+    """
+    from metrics_amd.utilities.data import interp
+    from metrics_amd.utilities.prints import rank_zero_warn
+
+    fpr_range_t = torch.tensor(fpr_range).to(fpr.device)
+    if fpr.numel() < 2 or tpr.numel() < 2:
+        rank_zero_warn(
+            "At least two values on for the fpr and tpr are required to compute the log AUC. Returns 0 score."
+        )
+        return torch.tensor(0.0, device=fpr.device)
+
+    tpr = torch.cat([tpr, interp(fpr_range_t, fpr, tpr)]).sort().values
+    fpr = torch.cat([fpr, fpr_range_t]).sort().values
+
+    log_fpr = torch.log10(fpr)
+    bounds = torch.log10(torch.tensor(fpr_range))
+
+    lower_bound_idx = torch.where(log_fpr == bounds[0])[0][-1]
+    upper_bound_idx = torch.where(log_fpr == bounds[1])[0][-1]
+
+    trimmed_log_fpr = log_fpr[lower_bound_idx : upper_bound_idx + 1]
+    trimmed_tpr = tpr[lower_bound_idx : upper_bound_idx + 1]
+
+    from metrics_amd.utilities.compute import _auc_compute_without_check
+
+    return _auc_compute_without_check(trimmed_log_fpr, trimmed_tpr, 1.0) / (bounds[1] - bounds[0])
 
 
 def binary_logauc(

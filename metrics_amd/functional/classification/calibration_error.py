@@ -113,11 +113,7 @@ def binary_calibration_error(
         preds, target, threshold=0.5, ignore_index=ignore_index, convert_to_labels=False
     )
     confidences, accuracies = _binary_calibration_error_update(preds, target)
-    # fold p<0.5 onto the confident side: confidence = max(p, 1-p), accuracy = pred==target
-    pred_labels = (confidences >= 0.5).long()
-    accuracies = (pred_labels == accuracies).float()
-    confidences = torch.where(pred_labels == 1, confidences, 1 - confidences)
-    return _ce_compute(confidences.float(), accuracies, n_bins, norm)
+    return _ce_compute(confidences.float(), accuracies.float(), n_bins, norm)
 
 
 def _multiclass_calibration_error_arg_validation(

@@ -7,6 +7,7 @@ from __future__ import annotations
 
 from typing import Optional
 
+import torch
 from torch import Tensor
 
 from metrics_amd.utilities.compute import _safe_divide
@@ -32,7 +33,12 @@ def dice(
     if average not in ("micro", "macro", "weighted", "none", None):
         raise ValueError(f"The `average` has to be one of 'micro'/'macro'/'weighted'/'none', got {average}.")
     if num_classes is None:
-        raise ValueError("Argument `num_classes` must be provided for `dice`")
+        # legacy API: infer the class count from the inputs (reference dice.py
+        # routes through the legacy input-format classifier)
+        if preds.is_floating_point() and preds.ndim == target.ndim + 1:
+            num_classes = preds.shape[1]
+        else:
+            num_classes = int(torch.max(torch.stack([preds.max(), target.max()])).item()) + 1
     _multiclass_stat_scores_arg_validation(num_classes, top_k or 1, average or "micro", "global", ignore_index)
     _multiclass_stat_scores_tensor_validation(preds, target, num_classes, "global", ignore_index)
     preds_f, target_f = _multiclass_stat_scores_format(preds, target, top_k or 1)

@@ -211,6 +211,6 @@ def dunn_index(data: Tensor, labels: Tensor, p: float = 2) -> Tensor:
         [torch.linalg.vector_norm(a - b, ord=p) for i, a in enumerate(centroids) for j, b in enumerate(centroids) if i != j]
     )
     max_intracluster = torch.stack(
-        [torch.linalg.vector_norm(ci - mu, ord=p).max() for ci, mu in zip(clusters, centroids)]
+        [torch.linalg.vector_norm(ci - mu, ord=p, dim=1).max() for ci, mu in zip(clusters, centroids)]
     ).max()
     return intercluster.min() / max_intracluster

@@ -188,28 +188,25 @@ def fleiss_kappa(ratings: Tensor, mode: str = "counts") -> Tensor:
     if mode == "probs":
         if ratings.ndim != 3 or not ratings.is_floating_point():
             raise ValueError("If argument `mode` is 'probs', ratings must have 3 dimensions with the format [n_samples, n_categories, n_raters] and be floating point")
+        # verbatim reference transform (functional/nominal/fleiss_kappa.py:27-35)
         ratings = ratings.argmax(dim=1)
-        one_hot = torch.nn.functional.one_hot(ratings, num_classes=ratings.max().item() + 1)
-        ratings = one_hot.sum(dim=1).float() if one_hot.ndim == 3 else one_hot.float()
-        # ratings now (n_samples, n_categories)... recompute counts over raters
+        one_hot = torch.nn.functional.one_hot(ratings, num_classes=ratings.shape[1]).permute(0, 2, 1)
+        ratings = one_hot.sum(dim=-1)
     elif mode == "counts":
         if ratings.ndim != 2 or ratings.is_floating_point():
             raise ValueError("If argument `mode` is `counts`, ratings must have 2 dimensions with the format [n_samples, n_categories] and be none floating point")
-        ratings = ratings.float()
     else:
         raise ValueError("Argument `mode` should be one of 'counts' or 'probs'")
 
-    n, _ = ratings.shape
-    raters_per_subject = ratings.sum(1)
-    if not torch.all(raters_per_subject == raters_per_subject[0]):
-        raise ValueError("Expected all subjects to have the same number of ratings")
-    r = raters_per_subject[0]
-
-    p_cat = ratings.sum(0) / (n * r)
-    p_subject = ((ratings**2).sum(1) - r) / (r * (r - 1))
-    p_bar = p_subject.mean()
-    p_e = (p_cat**2).sum()
-    return (p_bar - p_e) / (1 - p_e)
+    counts = ratings.float()
+    total = counts.shape[0]
+    # reference semantics: unequal rater counts allowed, normalized by the MAX
+    num_raters = counts.sum(1).max()
+    p_i = counts.sum(dim=0) / (total * num_raters)
+    p_j = ((counts**2).sum(dim=1) - num_raters) / (num_raters * (num_raters - 1))
+    p_bar = p_j.mean()
+    pe_bar = (p_i**2).sum()
+    return (p_bar - pe_bar) / (1 - pe_bar + 1e-5)
 
 
 def _matrix_over_columns(matrix: Tensor, pair_fn, symmetric: bool) -> Tensor:

@@ -88,13 +88,14 @@ def test_thresholded_state_matches_batch_accumulation(bin_scores):
 def test_binary_calibration_error(bin_scores):
     preds, target = bin_scores
     v = ma.BinaryCalibrationError(n_bins=10, norm="l1")(preds, target).item()
-    # manual ECE reference
-    conf = torch.where(preds >= 0.5, preds, 1 - preds)
-    acc = ((preds >= 0.5).long() == target).float()
+    # manual ECE oracle (reference semantics: confidence = p(class 1),
+    # accuracy = target; see reference functional/classification/calibration_error.py)
+    conf = preds.float()
+    acc = target.float()
     bins = torch.linspace(0, 1, 11)
-    idx = torch.bucketize(conf, bins, right=True).clamp(1, 10) - 1
+    idx = (torch.bucketize(conf, bins, right=True) - 1).clamp(0, 10)
     ece = 0.0
-    for b in range(10):
+    for b in range(11):
         mask = idx == b
         if mask.any():
             ece += (mask.float().mean() * (acc[mask].mean() - conf[mask].mean()).abs()).item()

@@ -74,4 +74,4 @@ def test_fleiss_kappa_vs_manual():
     p_i = ((m**2).sum(1) - n) / (n * (n - 1))
     pj = m.sum(0) / m.sum()
     kappa_ref = (p_i.mean() - (pj**2).sum()) / (1 - (pj**2).sum())
-    assert abs(fleiss_kappa(ratings, mode="counts").item() - kappa_ref) < 1e-6
+    assert abs(fleiss_kappa(ratings, mode="counts").item() - kappa_ref) < 1e-4

@@ -45,7 +45,10 @@ def run(domains=None, verbose=True):
                     doc = ast.get_docstring(node)
                     if not doc or ">>>" not in doc:
                         continue
+                    import metrics_amd.functional as F
+
                     globs = {"torch": torch, "tensor": torch.tensor}
+                    globs.update({k: v for k, v in vars(F).items() if not k.startswith("_")})
                     name = f"{d}/{path.split('/')[-1]}::{node.name}"
                     test = parser.get_doctest(doc, globs, name, path, 0)
                     out = []
