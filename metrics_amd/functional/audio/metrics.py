@@ -66,7 +66,11 @@ def source_aggregated_signal_distortion_ratio(
         target = target - torch.mean(target, dim=-1, keepdim=True)
         preds = preds - torch.mean(preds, dim=-1, keepdim=True)
     if scale_invariant:
-        alpha = ((preds * target).sum(dim=-1, keepdim=True) + eps) / ((target**2).sum(dim=-1, keepdim=True) + eps)
+        # one alpha shared by all speakers (shape [..., 1, 1]) — that is the
+        # "source-aggregated" part of SA-SDR
+        alpha = ((preds * target).sum(dim=(-2, -1), keepdim=True) + eps) / (
+            (target**2).sum(dim=(-2, -1), keepdim=True) + eps
+        )
         target = alpha * target
     distortion = target - preds
     val = ((target**2).sum(dim=(-2, -1)) + eps) / ((distortion**2).sum(dim=(-2, -1)) + eps)
@@ -210,17 +214,22 @@ def permutation_invariant_training(
     # permutation-wise: enumerate all permutations
     import itertools
 
-    perms = list(itertools.permutations(range(spk_num)))
-    vals = []
-    for p in perms:
-        v = torch.stack([metric_func(preds[:, p[t], ...], target[:, t, ...], **kwargs) for t in range(spk_num)], dim=1)
-        vals.append(v.mean(dim=1))
-    vals_t = torch.stack(vals, dim=1)  # (batch, n_perms)
+    # the metric sees whole (batch, spk, ...) tensors with preds permuted, all
+    # permutations batched into one call (reference functional/audio/pit.py:173-188);
+    # per-speaker metric outputs are averaged over trailing dims
+    perms = torch.tensor(list(itertools.permutations(range(spk_num))), device=preds.device)
+    perm_num = perms.shape[0]
+    ppreds = torch.index_select(preds, dim=1, index=perms.reshape(-1)).reshape(
+        batch_size * perm_num, *preds.shape[1:]
+    )
+    ptarget = target.repeat_interleave(repeats=perm_num, dim=0)
+    vals = metric_func(ppreds, ptarget, **kwargs)
+    vals_t = torch.mean(vals.reshape(batch_size, perm_num, -1), dim=-1)  # (batch, n_perms)
     if eval_func == "max":
         best, idx = vals_t.max(dim=1)
     else:
         best, idx = vals_t.min(dim=1)
-    best_perm = torch.tensor(perms, device=target.device)[idx]
+    best_perm = perms[idx.detach(), :]
     return best, best_perm
 
 

@@ -172,41 +172,77 @@ def total_variation(img: Tensor, reduction: Optional[str] = "sum") -> Tensor:
     raise ValueError("Expected argument `reduction` to either be 'sum', 'mean', 'none' or None")
 
 
+def _symmetric_pad2d(x: Tensor, left: int, right: int, top: int, bottom: int) -> Tensor:
+    """Symmetric padding (``d c b a | a b c d | d c b a``) — edge pixel repeated.
+
+    torch's ``reflect`` mode excludes the edge pixel, so it cannot be used here
+    (reference functional/image/scc.py ``_symmetric_reflect_pad_2d``).
+    """
+    parts_w = [x[:, :, :, :left].flip(dims=[3]), x, x[:, :, :, x.shape[3] - right :].flip(dims=[3])]
+    x = torch.cat([p for p in parts_w if p.shape[3] > 0], dim=3)
+    parts_h = [x[:, :, :top, :].flip(dims=[2]), x, x[:, :, x.shape[2] - bottom :, :].flip(dims=[2])]
+    return torch.cat([p for p in parts_h if p.shape[2] > 0], dim=2)
+
+
 def spatial_correlation_coefficient(
     preds: Tensor,
     target: Tensor,
     hp_filter: Optional[Tensor] = None,
     window_size: int = 8,
-    reduction: Optional[str] = "elementwise_mean",
+    reduction: Optional[str] = "mean",
 ) -> Tensor:
-    """Spatial correlation coefficient after high-pass filtering."""
+    """Spatial correlation coefficient after high-pass (Laplacian) filtering.
+
+    Matches the reference/sewar pipeline (reference functional/image/scc.py):
+    signal-convolved doubled Laplacian with symmetric padding, then stride-1
+    zero-padded window statistics, correlation per pixel, mean per sample.
+    """
     if preds.ndim == 3:
         preds = preds.unsqueeze(1)
         target = target.unsqueeze(1)
+    if target.dtype != preds.dtype:
+        target = target.to(preds.dtype)
     preds, target = _image_check(preds.float(), target.float())
+    if not window_size > 0:
+        raise ValueError(f"Expected `window_size` to be a positive integer. Got {window_size}.")
+    if window_size > preds.size(2) or window_size > preds.size(3):
+        raise ValueError(
+            f"Expected `window_size` to be less than or equal to the size of the image."
+            f" Got window_size: {window_size} and image size: {preds.size(2)}x{preds.size(3)}."
+        )
+    if reduction is None:
+        reduction = "none"
+    if reduction not in ("mean", "none"):
+        raise ValueError(f"Expected reduction to be 'mean' or 'none', but got {reduction}")
     if hp_filter is None:
-        hp_filter = torch.tensor([[-1.0, -1.0, -1.0], [-1.0, 8.0, -1.0], [-1.0, -1.0, -1.0]], device=preds.device)
+        hp_filter = torch.tensor([[-1.0, -1.0, -1.0], [-1.0, 8.0, -1.0], [-1.0, -1.0, -1.0]])
     c = preds.shape[1]
-    kern = hp_filter.to(preds).expand(c, 1, *hp_filter.shape)
-    pad = hp_filter.shape[-1] // 2
-    hp_preds = F.conv2d(F.pad(preds, [pad] * 4, mode="reflect"), kern, groups=c)
-    hp_target = F.conv2d(F.pad(target, [pad] * 4, mode="reflect"), kern, groups=c)
+    kern = hp_filter.to(preds).flip([0, 1]).expand(c, 1, *hp_filter.shape)
 
-    # windowed means
+    # doubled Laplacian via signal convolution (flipped kernel, symmetric pad)
+    kh, kw = hp_filter.shape
+    lw, rw = (kw - 1) // 2, kw - 1 - (kw - 1) // 2
+    th, bh = (kh - 1) // 2, kh - 1 - (kh - 1) // 2
+    hp_preds = F.conv2d(_symmetric_pad2d(preds, lw, rw, th, bh), kern, groups=c) * 2.0
+    hp_target = F.conv2d(_symmetric_pad2d(target, lw, rw, th, bh), kern, groups=c) * 2.0
+
+    # stride-1 window statistics with zero padding (ceil left/top, floor right/bottom)
+    lp = (window_size - 1 + 1) // 2
+    rp = (window_size - 1) // 2
     win = torch.ones(c, 1, window_size, window_size, device=preds.device, dtype=preds.dtype) / window_size**2
-    mu_x = F.conv2d(hp_preds, win, groups=c, stride=window_size)
-    mu_y = F.conv2d(hp_target, win, groups=c, stride=window_size)
-    mu_xx = F.conv2d(hp_preds * hp_preds, win, groups=c, stride=window_size)
-    mu_yy = F.conv2d(hp_target * hp_target, win, groups=c, stride=window_size)
-    mu_xy = F.conv2d(hp_preds * hp_target, win, groups=c, stride=window_size)
+    xp = F.pad(hp_preds, (lp, rp, lp, rp))
+    yp = F.pad(hp_target, (lp, rp, lp, rp))
+    mu_x = F.conv2d(xp, win, groups=c)
+    mu_y = F.conv2d(yp, win, groups=c)
+    var_x = F.conv2d(xp**2, win, groups=c) - mu_x**2
+    var_y = F.conv2d(yp**2, win, groups=c) - mu_y**2
+    cov = F.conv2d(yp * xp, win, groups=c) - mu_y * mu_x
 
-    var_x = mu_xx - mu_x**2
-    var_y = mu_yy - mu_y**2
-    cov = mu_xy - mu_x * mu_y
-    denom = torch.sqrt(var_x.clamp(min=0)) * torch.sqrt(var_y.clamp(min=0))
-    scc = torch.where(denom > 0, cov / denom, torch.zeros_like(cov))
-    per_image = scc.mean(dim=(1, 2, 3))
-    return reduce(per_image, reduction or "none")
+    denom = torch.sqrt(var_y.clamp(min=0)) * torch.sqrt(var_x.clamp(min=0))
+    scc = torch.where(denom > 0, cov / torch.where(denom > 0, denom, torch.ones_like(denom)), torch.zeros_like(cov))
+    if reduction == "none":
+        return scc.mean(dim=(1, 2, 3))
+    return scc.mean()
 
 
 def visual_information_fidelity(preds: Tensor, target: Tensor, sigma_n_sq: float = 2.0) -> Tensor:

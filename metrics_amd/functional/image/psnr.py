@@ -1,6 +1,7 @@
 """PSNR (+ PSNRB). Parity: torchmetrics ``functional/image/{psnr,psnrb}.py``."""
 from __future__ import annotations
 
+import math
 from typing import Optional, Tuple, Union
 
 import torch
@@ -74,33 +75,37 @@ def peak_signal_noise_ratio(
 
 
 def _blocking_effect_factor(x: Tensor, block_size: int = 8) -> Tensor:
-    """Mean boundary-vs-interior squared difference factor used by PSNRB."""
-    _, _, h, w = x.shape
-    h_blocks, w_blocks = h // block_size, w // block_size
+    """Boundary-vs-interior squared-difference factor used by PSNRB.
 
-    h_bound = torch.arange(block_size - 1, block_size * h_blocks - 1, block_size, device=x.device)
-    w_bound = torch.arange(block_size - 1, block_size * w_blocks - 1, block_size, device=x.device)
+    Matches the reference's counting exactly (reference functional/image/psnrb.py
+    ``_compute_bef``): the "boundary" columns/rows are every ``block_size``-th
+    difference, normalizers use the reference's ``H * (W / block) - 1`` form, and
+    one scalar is produced for the whole (grayscale) batch.
+    """
+    _, channels, height, width = x.shape
+    if channels > 1:
+        raise ValueError(f"`psnrb` metric expects grayscale images, but got images with {channels} channels.")
 
-    d_b_h = (x[:, :, h_bound, :] - x[:, :, h_bound + 1, :]).pow(2).sum(dim=(1, 2, 3))
-    d_b_w = (x[:, :, :, w_bound] - x[:, :, :, w_bound + 1]).pow(2).sum(dim=(1, 2, 3))
+    h_bound = torch.arange(block_size - 1, width - 1, block_size, device=x.device)
+    all_w = torch.arange(width - 1, device=x.device)
+    h_nonb = all_w[~torch.isin(all_w, h_bound)]
+    v_bound = torch.arange(block_size - 1, height - 1, block_size, device=x.device)
+    all_h = torch.arange(height - 1, device=x.device)
+    v_nonb = all_h[~torch.isin(all_h, v_bound)]
 
-    all_h = torch.arange(0, h - 1, device=x.device)
-    all_w = torch.arange(0, w - 1, device=x.device)
-    nonb_h = all_h[~torch.isin(all_h, h_bound)]
-    nonb_w = all_w[~torch.isin(all_w, w_bound)]
+    d_b = (x[:, :, :, h_bound] - x[:, :, :, h_bound + 1]).pow(2).sum()
+    d_bc = (x[:, :, :, h_nonb] - x[:, :, :, h_nonb + 1]).pow(2).sum()
+    d_b = d_b + (x[:, :, v_bound, :] - x[:, :, v_bound + 1, :]).pow(2).sum()
+    d_bc = d_bc + (x[:, :, v_nonb, :] - x[:, :, v_nonb + 1, :]).pow(2).sum()
 
-    d_bc_h = (x[:, :, nonb_h, :] - x[:, :, nonb_h + 1, :]).pow(2).sum(dim=(1, 2, 3))
-    d_bc_w = (x[:, :, :, nonb_w] - x[:, :, :, nonb_w + 1]).pow(2).sum(dim=(1, 2, 3))
-
-    n_b = x.shape[1] * (w * len(h_bound) + h * len(w_bound))
-    n_bc = x.shape[1] * (w * len(nonb_h) + h * len(nonb_w))
-
-    d_b = (d_b_h + d_b_w) / n_b
-    d_bc = (d_bc_h + d_bc_w) / n_bc
-    t = torch.log2(torch.tensor(block_size, device=x.device).float()) / torch.log2(
-        torch.tensor(min(h, w), device=x.device).float()
-    )
-    return torch.where(d_b > d_bc, t * (d_b - d_bc), torch.zeros_like(d_b))
+    n_hb = height * (width / block_size) - 1
+    n_hbc = height * (width - 1) - n_hb
+    n_vb = width * (height / block_size) - 1
+    n_vbc = width * (height - 1) - n_vb
+    d_b = d_b / (n_hb + n_vb)
+    d_bc = d_bc / (n_hbc + n_vbc)
+    t = math.log2(block_size) / math.log2(min(height, width)) if d_b > d_bc else 0
+    return t * (d_b - d_bc)
 
 
 def peak_signal_noise_ratio_with_blocked_effect(
@@ -115,6 +120,7 @@ def peak_signal_noise_ratio_with_blocked_effect(
     data_range = target.max() - target.min()
     bef = _blocking_effect_factor(preds, block_size=block_size)
 
-    mse = ((preds - target) ** 2).mean(dim=(1, 2, 3))
-    mse_b = mse + bef
-    return (10 * torch.log10(data_range**2 / mse_b)).mean()
+    mse_b = ((preds - target) ** 2).sum() / target.numel() + bef
+    if data_range > 2:
+        return 10 * torch.log10(data_range**2 / mse_b)
+    return 10 * torch.log10(1.0 / mse_b)

@@ -87,13 +87,18 @@ def _ssim_compute(
     channel = preds.size(1)
     dtype = preds.dtype
     device = preds.device
+    # with a gaussian window the effective kernel (and padding) size follows
+    # sigma, not the kernel_size argument (reference functional/image/ssim.py:126)
+    gauss_kernel_size = [int(3.5 * s + 0.5) * 2 + 1 for s in sigma]
     if gaussian_kernel:
-        kernel = _gaussian_kernel_2d(channel, kernel_size, sigma, dtype, device)
+        pad_h = (gauss_kernel_size[0] - 1) // 2
+        pad_w = (gauss_kernel_size[1] - 1) // 2
+        kernel = _gaussian_kernel_2d(channel, gauss_kernel_size, sigma, dtype, device)
     else:
+        pad_h = (kernel_size[0] - 1) // 2
+        pad_w = (kernel_size[1] - 1) // 2
         kernel = _uniform_kernel_2d(channel, kernel_size, dtype, device)
 
-    pad_h = (kernel_size[0] - 1) // 2
-    pad_w = (kernel_size[1] - 1) // 2
     preds_p = F.pad(preds, (pad_w, pad_w, pad_h, pad_h), mode="reflect")
     target_p = F.pad(target, (pad_w, pad_w, pad_h, pad_h), mode="reflect")
 
@@ -105,26 +110,24 @@ def _ssim_compute(
     mu_target_sq = output_list[1].pow(2)
     mu_pred_target = output_list[0] * output_list[1]
 
-    sigma_pred_sq = output_list[2] - mu_pred_sq
-    sigma_target_sq = output_list[3] - mu_target_sq
+    sigma_pred_sq = torch.clamp(output_list[2] - mu_pred_sq, min=0.0)
+    sigma_target_sq = torch.clamp(output_list[3] - mu_target_sq, min=0.0)
     sigma_pred_target = output_list[4] - mu_pred_target
 
     upper = 2 * sigma_pred_target.to(dtype) + c2
     lower = (sigma_pred_sq + sigma_target_sq).to(dtype) + c2
 
+    # the similarity mean is taken over the FULL map (border ring included);
+    # only contrast sensitivity is cropped (reference functional/image/ssim.py:173-186)
     ssim_idx_full_image = ((2 * mu_pred_target + c1) * upper) / ((mu_pred_sq + mu_target_sq + c1) * lower)
-    # crop the padded border back off
-    ssim_idx = ssim_idx_full_image[..., pad_h:-pad_h, pad_w:-pad_w]
+    sim = ssim_idx_full_image.reshape(ssim_idx_full_image.shape[0], -1).mean(-1)
 
     if return_contrast_sensitivity:
-        contrast_sensitivity = upper / lower
-        contrast_sensitivity = contrast_sensitivity[..., pad_h:-pad_h, pad_w:-pad_w]
-        return ssim_idx.reshape(ssim_idx.shape[0], -1).mean(-1), contrast_sensitivity.reshape(
-            contrast_sensitivity.shape[0], -1
-        ).mean(-1)
+        contrast_sensitivity = (upper / lower)[..., pad_h:-pad_h, pad_w:-pad_w]
+        return sim, contrast_sensitivity.reshape(contrast_sensitivity.shape[0], -1).mean(-1)
     if return_full_image:
-        return ssim_idx.reshape(ssim_idx.shape[0], -1).mean(-1), ssim_idx_full_image
-    return ssim_idx.reshape(ssim_idx.shape[0], -1).mean(-1)
+        return sim, ssim_idx_full_image
+    return sim
 
 
 def structural_similarity_index_measure(

@@ -9,3 +9,4 @@ from metrics_amd.functional.text.error_rates import (
 )
 from metrics_amd.functional.text.misc import extended_edit_distance, perplexity, squad
 from metrics_amd.functional.text.rouge import rouge_score
+from metrics_amd.functional.text.bert_infolm import bert_score, infolm

@@ -121,9 +121,9 @@ def edit_distance(preds, target, substitution_cost: int = 1, reduction: str = "m
         vals = [dist(p, t) for p, t in zip(preds, target)]
     else:
         vals = [_edit_distance(list(p), list(t)) for p, t in zip(preds, target)]
-    res = torch.tensor(vals, dtype=torch.float)
+    res = torch.tensor(vals, dtype=torch.int32)
     if reduction == "mean":
-        return res.mean()
+        return res.float().mean()
     if reduction == "sum":
         return res.sum()
     if reduction is None or reduction == "none":

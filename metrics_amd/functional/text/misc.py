@@ -88,7 +88,8 @@ def _perplexity_update(preds: Tensor, target: Tensor, ignore_index: Optional[int
         mask = torch.ones_like(target_flat, dtype=torch.bool)
 
     nll = -probs.gather(1, target_flat.unsqueeze(1)).squeeze(1)
-    total_log_probs = (nll * mask).sum()
+    # accumulate in double for accuracy, report in the input dtype (reference parity)
+    total_log_probs = (nll * mask).sum().to(preds.dtype)
     count = mask.sum()
     return total_log_probs, count
 

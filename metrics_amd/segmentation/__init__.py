@@ -51,27 +51,23 @@ class MeanIoU(Metric):
         self.input_format = input_format
 
         num_stat_classes = (num_classes - (0 if include_background else 1)) if num_classes else 1
-        self.add_state("score", default=torch.zeros(num_stat_classes if per_class else 1), dist_reduce_fx="sum")
-        self.add_state("num_batches", default=torch.zeros(num_stat_classes if per_class else 1), dist_reduce_fx="sum")
+        self.add_state("score", default=torch.zeros(num_stat_classes if per_class else 1).squeeze(), dist_reduce_fx="sum")
+        self.add_state("num_batches", default=torch.tensor(0), dist_reduce_fx="sum")
 
     def update(self, preds: Tensor, target: Tensor) -> None:
         """Accumulate per-class IoU sums."""
-        intersection, union = _mean_iou_update(
-            preds, target, self.num_classes, self.include_background, self.input_format
+        from metrics_amd.functional.segmentation import mean_iou as _miou_fn
+
+        score = _miou_fn(
+            preds, target, self.num_classes, self.include_background, self.per_class, self.input_format
         )
-        score = _mean_iou_compute(intersection, union, zero_division="nan")
-        score = score.nanmean(0) if self.per_class else score.nanmean(-1)
-        valid = ~torch.isnan(score)
-        if self.per_class:
-            self.score += torch.where(valid, score, torch.zeros_like(score))
-            self.num_batches += valid.to(self.num_batches.dtype)
-        else:
-            self.score += score.nansum() if score.ndim else torch.nan_to_num(score)
-            self.num_batches += valid.sum() if score.ndim else valid.to(self.num_batches.dtype)
+        # reference segmentation/mean_iou.py:117-128: running mean over batches
+        self.score += score.mean(0) if self.per_class else score.mean()
+        self.num_batches += 1
 
     def compute(self) -> Tensor:
         """Average IoU over updates."""
-        return (self.score / self.num_batches).squeeze()
+        return self.score / self.num_batches
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)
@@ -222,11 +218,11 @@ class HausdorffDistance(Metric):
             preds, target, self.num_classes, self.include_background, self.distance_metric,
             self.spacing, self.directed, self.input_format,
         )
-        self.score += hd.mean()
-        self.total += 1
+        self.score += hd.sum()
+        self.total += hd.numel()
 
     def compute(self) -> Tensor:
-        """Average hausdorff distance over updates."""
+        """Average hausdorff distance over all samples and classes."""
         return self.score / self.total
 
     def plot(self, val=None, ax=None):

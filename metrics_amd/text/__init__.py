@@ -148,23 +148,30 @@ class EditDistance(Metric):
             raise ValueError(f"Expected argument `reduction` to be one of {allowed}, but got {reduction}")
         self.reduction = reduction
 
-        if self.reduction in ("mean", "sum"):
+        if self.reduction == "none" or self.reduction is None:
             self.add_state("edit_scores_list", default=[], dist_reduce_fx="cat")
         else:
-            self.add_state("edit_scores_list", default=[], dist_reduce_fx="cat")
+            self.add_state("edit_scores", default=torch.tensor(0), dist_reduce_fx="sum")
+            self.add_state("num_elements", default=torch.tensor(0), dist_reduce_fx="sum")
 
     def update(self, preds, target) -> None:
         """Accumulate per-pair edit distances."""
         scores = _edit_distance_fn(preds, target, self.substitution_cost, reduction="none")
-        self.edit_scores_list.append(scores)
+        if self.reduction == "none" or self.reduction is None:
+            self.edit_scores_list.append(scores)
+        else:
+            self.edit_scores += scores.sum()
+            self.num_elements += scores.shape[0]
 
     def compute(self) -> Tensor:
-        scores = dim_zero_cat(self.edit_scores_list)
+        if self.reduction == "none" or self.reduction is None:
+            scores = dim_zero_cat(self.edit_scores_list)
+            return scores if scores.numel() else torch.tensor(0, dtype=torch.int32)
+        if self.num_elements == 0:
+            return torch.tensor(0, dtype=torch.int32)
         if self.reduction == "mean":
-            return scores.mean()
-        if self.reduction == "sum":
-            return scores.sum()
-        return scores
+            return self.edit_scores / self.num_elements
+        return self.edit_scores
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)
@@ -588,8 +595,8 @@ class Perplexity(Metric):
         if ignore_index is not None and not isinstance(ignore_index, int):
             raise ValueError(f"Argument `ignore_index` expected to either be `None` or an `int` but got {ignore_index}")
         self.ignore_index = ignore_index
-        self.add_state("total_log_probs", default=tensor(0.0, dtype=torch.float64), dist_reduce_fx="sum")
-        self.add_state("count", default=tensor(0.0, dtype=torch.float64), dist_reduce_fx="sum")
+        self.add_state("total_log_probs", default=tensor(0.0), dist_reduce_fx="sum")
+        self.add_state("count", default=tensor(0.0), dist_reduce_fx="sum")
 
     def update(self, preds: Tensor, target: Tensor) -> None:
         """Accumulate total NLL + token count."""
