@@ -514,3 +514,28 @@ class LogAUC(_ClassificationTaskWrapper):
         if task == ClassificationTask.MULTILABEL:
             return MultilabelLogAUC(num_labels, fpr_range, average, thresholds, ignore_index, validate_args, **kwargs)
         raise ValueError(f"Not handled value: {task}")
+
+
+def _plot_first_value(self, val=None, ax=None):
+    """Plot the metric value; defaults to the value component of (value, threshold)."""
+    if val is None:
+        val = self.compute()[0]
+    return self._plot(val, ax)
+
+
+def _plot_scalar(self, val=None, ax=None):
+    """Plot the (scalar) metric value."""
+    return self._plot(val, ax)
+
+
+# the at-fixed metrics compute (value, threshold) tuples and LogAUC a scalar —
+# neither can use the inherited precision-recall-curve plot
+for _cls in (
+    BinaryRecallAtFixedPrecision, MulticlassRecallAtFixedPrecision, MultilabelRecallAtFixedPrecision,
+    BinaryPrecisionAtFixedRecall, MulticlassPrecisionAtFixedRecall, MultilabelPrecisionAtFixedRecall,
+    BinarySensitivityAtSpecificity, MulticlassSensitivityAtSpecificity, MultilabelSensitivityAtSpecificity,
+    BinarySpecificityAtSensitivity, MulticlassSpecificityAtSensitivity, MultilabelSpecificityAtSensitivity,
+):
+    _cls.plot = _plot_first_value
+for _cls in (BinaryLogAUC, MulticlassLogAUC, MultilabelLogAUC):
+    _cls.plot = _plot_scalar

@@ -7,6 +7,7 @@ from __future__ import annotations
 
 from typing import Any, Optional
 
+import torch
 from torch import Tensor
 
 from metrics_amd.classification.stat_scores import MulticlassStatScores
@@ -24,35 +25,86 @@ class Dice(MulticlassStatScores):
 
     def __init__(
         self,
+        zero_division: int = 0,
         num_classes: Optional[int] = None,
         threshold: float = 0.5,
         average: Optional[str] = "micro",
         ignore_index: Optional[int] = None,
-        top_k: int = 1,
+        top_k: Optional[int] = None,
         **kwargs: Any,
     ) -> None:
+        allowed_average = ("micro", "macro", "weighted", "samples", "none", None)
+        if average not in allowed_average:
+            raise ValueError(f"The `average` has to be one of {allowed_average}, got {average}.")
+        if average == "macro" and (not num_classes or num_classes < 1):
+            raise ValueError("When you set `average` as 'macro', you have to provide the number of classes.")
         if num_classes is None:
-            raise ValueError("Argument `num_classes` must be provided for the Dice metric")
-        super().__init__(
-            num_classes=num_classes,
-            top_k=top_k,
-            average=average or "micro",
-            multidim_average="global",
-            ignore_index=ignore_index,
-            **kwargs,
-        )
+            # micro averaging needs only global counts — the class count is
+            # inferred per batch (reference classification/dice.py allows
+            # Dice(average='micro') with no num_classes)
+            from metrics_amd.metric import Metric
+
+            Metric.__init__(self, **kwargs)
+            self._lazy_micro = True
+            self.average = average or "micro"
+            self.ignore_index = ignore_index
+            self.top_k = top_k or 1
+            self.add_state("tp", default=torch.tensor(0), dist_reduce_fx="sum")
+            self.add_state("fp", default=torch.tensor(0), dist_reduce_fx="sum")
+            self.add_state("tn", default=torch.tensor(0), dist_reduce_fx="sum")
+            self.add_state("fn", default=torch.tensor(0), dist_reduce_fx="sum")
+        else:
+            super().__init__(
+                num_classes=num_classes,
+                top_k=top_k or 1,
+                average=average or "micro",
+                multidim_average="global",
+                ignore_index=ignore_index,
+                **kwargs,
+            )
+            self._lazy_micro = False
         self.threshold = threshold
+        self.zero_division = zero_division
+
+    def update(self, preds: Tensor, target: Tensor) -> None:
+        """Accumulate stat scores (class count inferred when not given)."""
+        if not self._lazy_micro:
+            return super().update(preds, target)
+        from metrics_amd.functional.classification.stat_scores import (
+            _multiclass_stat_scores_format,
+            _multiclass_stat_scores_update,
+        )
+
+        if preds.is_floating_point() and preds.ndim == target.ndim + 1:
+            num_classes = preds.shape[1]
+        else:
+            num_classes = int(torch.max(torch.stack([preds.max(), target.max()])).item()) + 1
+        preds_f, target_f = _multiclass_stat_scores_format(preds, target, self.top_k)
+        tp, fp, tn, fn = _multiclass_stat_scores_update(
+            preds_f, target_f, num_classes, self.top_k, "micro", "global", self.ignore_index
+        )
+        self.tp += tp.sum()
+        self.fp += fp.sum()
+        self.tn += tn.sum()
+        self.fn += fn.sum()
+        return None
 
     def compute(self) -> Tensor:
         """Dice coefficient with the configured averaging."""
+        if self._lazy_micro:
+            return _safe_divide(2 * self.tp, 2 * self.tp + self.fp + self.fn, self.zero_division)
         tp, fp, tn, fn = self._final_state()
         if self.average == "micro":
             tp, fp, fn = tp.sum(), fp.sum(), fn.sum()
-            return _safe_divide(2 * tp, 2 * tp + fp + fn)
-        score = _safe_divide(2 * tp, 2 * tp + fp + fn)
+            return _safe_divide(2 * tp, 2 * tp + fp + fn, self.zero_division)
+        score = _safe_divide(2 * tp, 2 * tp + fp + fn, self.zero_division)
         if self.average == "macro":
             return score.float().mean()
         if self.average == "weighted":
             w = tp + fn
             return (score * _safe_divide(w, w.sum())).sum()
         return score
+
+    def plot(self, val=None, ax=None):
+        """Plot the dice value(s)."""
+        return self._plot(val, ax)

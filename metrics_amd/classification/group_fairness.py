@@ -137,3 +137,12 @@ class BinaryFairness(_AbstractGroupStatScores):
             max_g = int(torch.argmax(rates))
             out[f"EO_{min_g}_{max_g}"] = _safe_divide(rates[min_g], rates[max_g])
         return out
+
+
+def _plot_group_stats(self, val=None, ax=None):
+    """Plot the per-group rate dict."""
+    return self._plot(val, ax)
+
+
+BinaryGroupStatRates.plot = _plot_group_stats
+BinaryFairness.plot = _plot_group_stats

@@ -115,6 +115,48 @@ def _encode_masks_rle(masks: Tensor) -> Tensor:
     return torch.cat([header] + counts_per_mask)
 
 
+def _coco_rle_str_encode(counts) -> str:
+    """COCO compressed-RLE string from run counts (5-bit varint, delta every 2nd)."""
+    out = []
+    prev2 = [0, 0]
+    for i, c in enumerate(counts):
+        x = int(c)
+        if i > 2:
+            x -= prev2[i % 2]
+        prev2[i % 2] = int(c)
+        more = True
+        while more:
+            bits = x & 0x1F
+            x >>= 5
+            more = not (x == 0 and not (bits & 0x10)) and not (x == -1 and (bits & 0x10))
+            if more:
+                bits |= 0x20
+            out.append(chr(bits + 48))
+    return "".join(out)
+
+
+def _coco_rle_str_decode(s: str) -> list:
+    """Run counts from a COCO compressed-RLE string."""
+    counts = []
+    i = 0
+    while i < len(s):
+        x = 0
+        k = 0
+        more = True
+        while more:
+            c = ord(s[i]) - 48
+            x |= (c & 0x1F) << (5 * k)
+            more = bool(c & 0x20)
+            i += 1
+            k += 1
+            if not more and (c & 0x10):
+                x |= -1 << (5 * k)
+        if len(counts) > 2:
+            x += counts[-2]
+        counts.append(x)
+    return counts
+
+
 def _decode_masks_rle(pack: "np.ndarray") -> "np.ndarray":
     """Inverse of :func:`_encode_masks_rle`: returns (N, H*W) uint8, column-major pixels."""
     import numpy as np
@@ -658,6 +700,149 @@ class MeanAveragePrecision(Metric):
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)
+
+    def _get_coco_format(self, labels, boxes=None, mask_packs=None, scores=None, crowds=None, area=None) -> dict:
+        """Cached states -> COCO dataset dict (reference mean_ap.py:_get_coco_format)."""
+        import numpy as np
+
+        images, annotations = [], []
+        ann_id = 1
+        for image_id, image_labels in enumerate(labels):
+            image_labels = image_labels.cpu().tolist()
+            images.append({"id": image_id})
+            pack = None
+            if mask_packs is not None:
+                pack = mask_packs[image_id].cpu().numpy()
+                if int(pack[2]) > 0:
+                    images[-1]["height"], images[-1]["width"] = int(pack[0]), int(pack[1])
+            lens = pack[3 : 3 + int(pack[2])].astype(np.int64) if pack is not None else None
+            mask_off = 3 + int(pack[2]) if pack is not None else 0
+            for k, image_label in enumerate(image_labels):
+                ann = {
+                    "id": ann_id,
+                    "image_id": image_id,
+                    "category_id": int(image_label),
+                    "iscrowd": int(crowds[image_id][k].item()) if crowds is not None else 0,
+                }
+                if boxes is not None:
+                    x1, y1, x2, y2 = boxes[image_id][k].cpu().tolist()
+                    ann["bbox"] = [x1, y1, x2 - x1, y2 - y1]
+                    ann["area"] = (x2 - x1) * (y2 - y1)
+                if pack is not None and k < len(lens):
+                    counts = pack[mask_off : mask_off + lens[k]]
+                    mask_off += int(lens[k])
+                    ann["segmentation"] = {
+                        "size": [int(pack[0]), int(pack[1])],
+                        "counts": _coco_rle_str_encode(counts),
+                    }
+                    ann["area"] = float(counts[1::2].sum())
+                if area is not None and float(area[image_id][k].item()) >= 0:
+                    ann["area"] = float(area[image_id][k].item())
+                if scores is not None:
+                    ann["score"] = float(scores[image_id][k].item())
+                annotations.append(ann)
+                ann_id += 1
+        classes = [{"id": int(i), "name": str(i)} for i in self._classes_list()]
+        return {"images": images, "annotations": annotations, "categories": classes}
+
+    def _classes_list(self) -> list:
+        if len(self.detection_labels) > 0 or len(self.groundtruth_labels) > 0:
+            return torch.cat(self.detection_labels + self.groundtruth_labels).unique().cpu().tolist()
+        return []
+
+    def tm_to_coco(self, name: str = "tm_map_input") -> None:
+        """Write cached inputs as COCO-format json: ``{name}_preds.json`` / ``{name}_target.json``."""
+        import json
+
+        use_masks = "segm" in self.iou_type
+        target_dataset = self._get_coco_format(
+            labels=self.groundtruth_labels,
+            boxes=self.groundtruth_boxes if "bbox" in self.iou_type else None,
+            mask_packs=self.groundtruth_masks if use_masks else None,
+            crowds=self.groundtruth_crowds,
+            area=self.groundtruth_area,
+        )
+        preds_dataset = self._get_coco_format(
+            labels=self.detection_labels,
+            boxes=self.detection_boxes if "bbox" in self.iou_type else None,
+            mask_packs=self.detection_masks if use_masks else None,
+            scores=self.detection_scores,
+        )
+        with open(f"{name}_preds.json", "w") as f:
+            f.write(json.dumps(preds_dataset["annotations"], indent=4))
+        with open(f"{name}_target.json", "w") as f:
+            f.write(json.dumps(target_dataset, indent=4))
+
+    @staticmethod
+    def coco_to_tm(coco_preds: str, coco_target: str, iou_type="bbox", backend: str = "pycocotools"):
+        """Read COCO-format json files into this metric's (preds, target) input lists.
+
+        Pure-python json parsing (no pycocotools needed); compressed or
+        uncompressed RLE segmentations are decoded for ``iou_type='segm'``.
+        """
+        import json
+
+        import numpy as np
+
+        iou_type = _validate_iou_type_arg(iou_type)
+        with open(coco_target) as f:
+            gt_data = json.load(f)
+        with open(coco_preds) as f:
+            dt_data = json.load(f)
+        gt_anns = gt_data["annotations"] if isinstance(gt_data, dict) else gt_data
+        dt_anns = dt_data["annotations"] if isinstance(dt_data, dict) else dt_data
+        img_sizes = {}
+        if isinstance(gt_data, dict):
+            for im in gt_data.get("images", []):
+                if "height" in im:
+                    img_sizes[im["id"]] = (im["height"], im["width"])
+
+        def _ann_mask(ann):
+            seg = ann["segmentation"]
+            if isinstance(seg, dict):
+                h, w = seg["size"]
+                counts = seg["counts"]
+                if isinstance(counts, str):
+                    counts = _coco_rle_str_decode(counts)
+                vals = np.zeros(len(counts), dtype=np.uint8)
+                vals[1::2] = 1
+                flat = np.repeat(vals, np.asarray(counts, dtype=np.int64))
+                return flat.reshape(w, h).T  # column-major
+            raise ValueError("Polygon segmentations are not supported without pycocotools; use RLE.")
+
+        def _collect(anns, with_scores):
+            per_img: dict = {}
+            for ann in anns:
+                d = per_img.setdefault(ann["image_id"], {"labels": [], "iscrowd": [], "area": [], "boxes": [], "masks": [], "scores": []})
+                d["labels"].append(ann["category_id"])
+                d["iscrowd"].append(ann.get("iscrowd", 0))
+                d["area"].append(ann.get("area", 0))
+                if "bbox" in iou_type:
+                    x, y, w, h = ann["bbox"]
+                    d["boxes"].append([x, y, x + w, y + h])
+                if "segm" in iou_type:
+                    d["masks"].append(_ann_mask(ann))
+                if with_scores:
+                    d["scores"].append(ann.get("score", 0.0))
+            out = []
+            for key in sorted(per_img):
+                d = per_img[key]
+                item = {
+                    "labels": torch.tensor(d["labels"], dtype=torch.long),
+                }
+                if "bbox" in iou_type:
+                    item["boxes"] = torch.tensor(d["boxes"], dtype=torch.float32).reshape(-1, 4)
+                if "segm" in iou_type:
+                    item["masks"] = torch.tensor(np.array(d["masks"]), dtype=torch.uint8)
+                if with_scores:
+                    item["scores"] = torch.tensor(d["scores"], dtype=torch.float32)
+                else:
+                    item["iscrowd"] = torch.tensor(d["iscrowd"], dtype=torch.long)
+                    item["area"] = torch.tensor(d["area"], dtype=torch.float32)
+                out.append(item)
+            return out
+
+        return _collect(dt_anns, True), _collect(gt_anns, False)
 
     def _sync_dist(self, dist_sync_fn=None, process_group=None) -> None:
         """List-of-variable-shape states: gather each element (reference uses all_gather_object)."""

@@ -53,11 +53,21 @@ class Running(WrapperMetric):
         self._num_vals_seen += 1
 
     def forward(self, *args: Any, **kwargs: Any) -> Any:
-        """Update the window with this batch and return the RUNNING value."""
-        self.update(*args, **kwargs)
+        """Update the window with this batch and return THIS batch's value.
+
+        The running value is what ``compute()`` returns (reference
+        wrappers/running.py:116-125 forwards the base metric and snapshots its
+        states into the window slot).
+        """
+        val = self._num_vals_seen % self.window
+        res = self.base_metric.forward(*args, **kwargs)
+        for key in self.base_metric._defaults:
+            setattr(self, key + f"_{val}", deepcopy(getattr(self.base_metric, key)))
+        self.base_metric.reset()
+        self._num_vals_seen += 1
         self._computed = None
-        self._forward_cache = self.compute()
-        return self._forward_cache
+        self._forward_cache = res
+        return res
 
     def compute(self) -> Any:
         """Merge the window states into the base metric and compute."""
@@ -65,7 +75,7 @@ class Running(WrapperMetric):
             self.base_metric._reduce_states(
                 {key: getattr(self, key + f"_{i}") for key in self.base_metric._defaults}
             )
-        self.base_metric._update_count = min(self._num_vals_seen, self.window)
+        self.base_metric._update_count = self._num_vals_seen
         val = self.base_metric.compute()
         self.base_metric.reset()
         return val

@@ -27,16 +27,25 @@ class MetricInputTransformer(WrapperMetric):
         """Identity by default."""
         return target
 
-    def update(self, pred: Tensor, target: Tensor) -> None:
-        """Transform inputs then update the wrapped metric."""
-        self.wrapped_metric.update(self.transform_pred(pred), self.transform_target(target))
+    def _wrap_transform(self, *args: Tensor) -> tuple:
+        """Transform the positional (preds, target) args; pass any extras through."""
+        out = list(args)
+        if len(out) > 0:
+            out[0] = self.transform_pred(out[0])
+        if len(out) > 1:
+            out[1] = self.transform_target(out[1])
+        return tuple(out)
+
+    def update(self, *args: Tensor, **kwargs: Any) -> None:
+        """Transform inputs then update the wrapped metric (extra kwargs pass through)."""
+        self.wrapped_metric.update(*self._wrap_transform(*args), **kwargs)
 
     def compute(self) -> Any:
         """Delegate."""
         return self.wrapped_metric.compute()
 
-    def forward(self, pred: Tensor, target: Tensor) -> Any:
-        self._forward_cache = self.wrapped_metric(self.transform_pred(pred), self.transform_target(target))
+    def forward(self, *args: Tensor, **kwargs: Any) -> Any:
+        self._forward_cache = self.wrapped_metric(*self._wrap_transform(*args), **kwargs)
         return self._forward_cache
 
     def reset(self) -> None:

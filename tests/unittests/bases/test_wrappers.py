@@ -109,5 +109,11 @@ def test_running_in_collection_context():
     from metrics_amd import RunningMean
 
     r = RunningMean(window=2)
-    outs = [r(v).item() for v in (1.0, 5.0, 9.0)]
-    assert outs == [1.0, 3.0, 7.0]
+    # forward returns THIS batch's value; compute() the windowed running value
+    outs = []
+    running = []
+    for v in (1.0, 5.0, 9.0):
+        outs.append(r(v).item())
+        running.append(r.compute().item())
+    assert outs == [1.0, 5.0, 9.0]
+    assert running == [1.0, 3.0, 7.0]
