@@ -45,7 +45,7 @@ def build(verbose: bool = True) -> Path:
     cpp_srcs = sorted(str(p) for p in CSRC.glob("*.cpp"))
     if cpp_srcs:
         cpu_out = LIBDIR / "libmetrics_cpu.so"
-        cmd = ["g++", "-O3", "-std=c++17", "-fPIC", "-shared", *cpp_srcs, "-o", str(cpu_out)]
+        cmd = ["g++", "-O3", "-std=c++17", "-fPIC", "-shared", "-fopenmp", *cpp_srcs, "-o", str(cpu_out)]
         if verbose:
             print("+", " ".join(cmd), file=sys.stderr)
         subprocess.run(cmd, check=True)

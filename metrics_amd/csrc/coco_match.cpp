@@ -39,16 +39,16 @@ int coco_match_class(
     uint8_t* dtm, uint8_t* dti, int32_t* npig) {
     const int64_t total_dt = dt_off[n_imgs];
 
-    std::vector<int> gt_order;
-    std::vector<uint8_t> gt_ign;
-    std::vector<uint8_t> gtm;
-
+    // every (area, image) cell writes disjoint slices of dtm/dti/npig
+#pragma omp parallel for collapse(2) schedule(dynamic, 16)
     for (int64_t a = 0; a < n_areas; a++) {
-        const float lo = area_rngs[2 * a], hi = area_rngs[2 * a + 1];
-        uint8_t* dtm_a = dtm + a * n_thrs * total_dt;
-        uint8_t* dti_a = dti + a * n_thrs * total_dt;
-
         for (int64_t img = 0; img < n_imgs; img++) {
+            const float lo = area_rngs[2 * a], hi = area_rngs[2 * a + 1];
+            uint8_t* dtm_a = dtm + a * n_thrs * total_dt;
+            uint8_t* dti_a = dti + a * n_thrs * total_dt;
+            std::vector<int> gt_order;
+            std::vector<uint8_t> gt_ign;
+            std::vector<uint8_t> gtm;
             const int64_t d0 = dt_off[img], d1 = dt_off[img + 1];
             const int64_t g0 = gt_off[img], g1 = gt_off[img + 1];
             const int n_dt = (int)(d1 - d0);
@@ -124,6 +124,7 @@ int coco_iou_class(
     const float* det_boxes, const int64_t* dt_off,
     const float* gt_boxes, const int64_t* gt_off, int64_t n_imgs,
     const uint8_t* gt_crowd, float* ious, const int64_t* iou_off) {
+#pragma omp parallel for schedule(dynamic, 32)
     for (int64_t img = 0; img < n_imgs; img++) {
         const int64_t d0 = dt_off[img], d1 = dt_off[img + 1];
         const int64_t g0 = gt_off[img], g1 = gt_off[img + 1];

@@ -1,7 +1,11 @@
 """Regenerate docs/PARITY.md: reference __all__ -> providing metrics_amd module."""
 import ast
 import importlib
+import os
+import sys
 from collections import defaultdict
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 SECTIONS = [
     ("torchmetrics (top level)", "", "metrics_amd"),
