@@ -180,6 +180,107 @@ __global__ void __launch_bounds__(256) k_mc_stat_logits(
     }
 }
 
+// K1-tiny: thread-per-row variant for small C. A 64-lane wave wastes
+// (64-C)/64 of its lanes when C < 64 in the wave-per-row mapping; here each
+// THREAD walks one row serially (vector loads keep the per-wave footprint
+// contiguous: lane i streams row i, so a wave touches 64 consecutive rows =
+// full cachelines). No shuffles; per-row tail work is identical to K1.
+template <typename T, bool IS_BF16>
+__global__ void __launch_bounds__(256) k_mc_stat_logits_tiny(
+    const T* __restrict__ preds, const ll* __restrict__ target, ll B, ll C, ll ignore_index,
+    int has_ignore, unsigned long long* __restrict__ tp, unsigned long long* __restrict__ fp,
+    unsigned long long* __restrict__ fn, unsigned long long* __restrict__ confmat,
+    unsigned long long* __restrict__ valid_count, ll* __restrict__ argmax_out,
+    float* __restrict__ rowmax, float* __restrict__ rowinv, unsigned int* __restrict__ E) {
+    const bool want_stats = rowmax != nullptr;
+    const unsigned int cur_epoch = E ? E[1] + 1u : 0u;
+    unsigned int outside = 0;
+    __shared__ unsigned int block_valid;
+    __shared__ unsigned int blk_outside;
+    // LDS-privatized counters: with only C distinct addresses, global atomics
+    // serialize badly at small C — accumulate per block, flush once
+    extern __shared__ unsigned int s_cnt[];  // [tp C][fp C][fn C][confmat C*C?]
+    unsigned int* s_tp = s_cnt;
+    unsigned int* s_fp = s_cnt + C;
+    unsigned int* s_fn = s_cnt + 2 * C;
+    // confmat privatized only while C*C fits comfortably in LDS; beyond that
+    // its C*C global addresses are uncontended enough anyway
+    const bool priv_cm = confmat && C <= 64;
+    unsigned int* s_cm = priv_cm ? s_cnt + 3 * C : nullptr;
+    const ll lds_words = 3 * C + (priv_cm ? C * C : 0);
+    for (ll i = threadIdx.x; i < lds_words; i += blockDim.x) s_cnt[i] = 0;
+    if (threadIdx.x == 0) { block_valid = 0; blk_outside = 0; }
+    __syncthreads();
+    unsigned int my_valid = 0;
+
+    for (ll row = (ll)blockIdx.x * blockDim.x + threadIdx.x; row < B;
+         row += (ll)gridDim.x * blockDim.x) {
+        const T* prow = preds + row * C;
+        float best = -INFINITY;
+        ll best_idx = 0;
+        float sm_m = -3.4e38f, sm_s = 0.0f;
+        auto fold = [&](float f, ll c) {
+            if (f > best) { best = f; best_idx = c; }
+            if (want_stats) {
+                outside |= (f < 0.0f || f > 1.0f) ? 1u : 0u;
+                if (f > sm_m) { sm_s = sm_s * __expf(sm_m - f) + 1.0f; sm_m = f; }
+                else { sm_s += __expf(f - sm_m); }
+            }
+        };
+        if (IS_BF16 && (C & 3) == 0) {
+            const ushort4* pv = reinterpret_cast<const ushort4*>(prow);
+            for (ll v = 0; v < C / 4; v++) {
+                ushort4 u = pv[v];
+                fold(bf16_to_f32(u.x), v * 4);
+                fold(bf16_to_f32(u.y), v * 4 + 1);
+                fold(bf16_to_f32(u.z), v * 4 + 2);
+                fold(bf16_to_f32(u.w), v * 4 + 3);
+            }
+        } else if (!IS_BF16 && (C & 3) == 0) {
+            const float4* pv = reinterpret_cast<const float4*>(prow);
+            for (ll v = 0; v < C / 4; v++) {
+                float4 u = pv[v];
+                fold(u.x, v * 4); fold(u.y, v * 4 + 1); fold(u.z, v * 4 + 2); fold(u.w, v * 4 + 3);
+            }
+        } else {
+            for (ll c = 0; c < C; c++)
+                fold(IS_BF16 ? bf16_to_f32(reinterpret_cast<const unsigned short*>(prow)[c])
+                             : (float)prow[c], c);
+        }
+        if (want_stats) { rowmax[row] = sm_m; rowinv[row] = 1.0f / sm_s; }
+        ll t = target[row];
+        ll p = best_idx;
+        if (argmax_out) argmax_out[row] = p;
+        if (!(has_ignore && t == ignore_index) && t >= 0 && t < C && p >= 0 && p < C) {
+            my_valid++;
+            if (p == t) {
+                atomicAdd(&s_tp[t], 1u);
+            } else {
+                atomicAdd(&s_fp[p], 1u);
+                atomicAdd(&s_fn[t], 1u);
+            }
+            if (s_cm) atomicAdd(&s_cm[t * C + p], 1u);
+            else if (confmat) atomicAdd(&confmat[t * C + p], 1ULL);
+        }
+    }
+    if (my_valid) atomicAdd(&block_valid, my_valid);
+    if (outside) atomicOr(&blk_outside, 1u);
+    __syncthreads();
+    for (ll c = threadIdx.x; c < C; c += blockDim.x) {
+        if (s_tp[c]) atomicAdd(&tp[c], (unsigned long long)s_tp[c]);
+        if (s_fp[c]) atomicAdd(&fp[c], (unsigned long long)s_fp[c]);
+        if (s_fn[c]) atomicAdd(&fn[c], (unsigned long long)s_fn[c]);
+    }
+    if (s_cm) {
+        for (ll i = threadIdx.x; i < C * C; i += blockDim.x)
+            if (s_cm[i]) atomicAdd(&confmat[i], (unsigned long long)s_cm[i]);
+    }
+    if (threadIdx.x == 0) {
+        if (block_valid) atomicAdd(valid_count, (unsigned long long)block_valid);
+        if (E && blk_outside && E[0] != cur_epoch) atomicMax(&E[0], cur_epoch);
+    }
+}
+
 // K1b: multiclass stat scores from integer label preds (element-wise)
 __global__ void __launch_bounds__(256) k_mc_stat_labels(
     const ll* __restrict__ preds, const ll* __restrict__ target, ll N, ll C, ll ignore_index,
@@ -1083,6 +1184,30 @@ int ma_mc_stat_logits(uintptr_t stream, uintptr_t preds, int dtype /*0=f32 1=bf1
         const char* e = getenv("MA_STAT_DIV");
         rows_div = e ? atoi(e) : 16;
         if (rows_div < 4) rows_div = 4;
+    }
+    // small-C: thread-per-row variant (wave-per-row wastes (64-C)/64 lanes)
+    static int smallc = -1;
+    if (smallc < 0) {
+        const char* e = getenv("MA_STAT_SMALLC");
+        smallc = e ? atoi(e) : 128;
+    }
+    if (C <= smallc) {
+        int grid = grid_for(B, 256);
+        if (grid > 4096) grid = 4096;
+        const size_t shmem = (size_t)(3 * C + (confmat && C <= 64 ? C * C : 0)) * sizeof(unsigned int);
+        if (dtype == 0)
+            k_mc_stat_logits_tiny<float, false><<<grid, 256, shmem, s>>>(
+                (const float*)preds, (const ll*)target, B, C, ignore_index, has_ignore,
+                (unsigned long long*)tp, (unsigned long long*)fp, (unsigned long long*)fn,
+                (unsigned long long*)confmat, (unsigned long long*)valid_count, (ll*)argmax_out,
+                (float*)rowmax, (float*)rowinv, (unsigned int*)epoch_buf);
+        else
+            k_mc_stat_logits_tiny<unsigned short, true><<<grid, 256, shmem, s>>>(
+                (const unsigned short*)preds, (const ll*)target, B, C, ignore_index, has_ignore,
+                (unsigned long long*)tp, (unsigned long long*)fp, (unsigned long long*)fn,
+                (unsigned long long*)confmat, (unsigned long long*)valid_count, (ll*)argmax_out,
+                (float*)rowmax, (float*)rowinv, (unsigned int*)epoch_buf);
+        return (int)hipGetLastError();
     }
     int grid = grid_for(B, rows_div);
     if (dtype == 0)
