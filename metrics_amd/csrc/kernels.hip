@@ -281,12 +281,24 @@ __global__ void __launch_bounds__(256) k_mc_stat_logits_tiny(
     }
 }
 
-// K1b: multiclass stat scores from integer label preds (element-wise)
+// K1b: multiclass stat scores from integer label preds (element-wise).
+// use_lds=1: tp/fp/fn (and confmat when priv_cm) accumulate in LDS and flush
+// once per block — global atomics over only C addresses serialize at small C.
 __global__ void __launch_bounds__(256) k_mc_stat_labels(
     const ll* __restrict__ preds, const ll* __restrict__ target, ll N, ll C, ll ignore_index,
     int has_ignore, unsigned long long* __restrict__ tp, unsigned long long* __restrict__ fp,
     unsigned long long* __restrict__ fn, unsigned long long* __restrict__ confmat,
-    unsigned long long* __restrict__ valid_count) {
+    unsigned long long* __restrict__ valid_count, int use_lds, int priv_cm) {
+    extern __shared__ unsigned int s_cnt[];
+    unsigned int* s_tp = use_lds ? s_cnt : nullptr;
+    unsigned int* s_fp = use_lds ? s_cnt + C : nullptr;
+    unsigned int* s_fn = use_lds ? s_cnt + 2 * C : nullptr;
+    unsigned int* s_cm = (use_lds && priv_cm && confmat) ? s_cnt + 3 * C : nullptr;
+    if (use_lds) {
+        const ll lds_words = 3 * C + (s_cm ? C * C : 0);
+        for (ll b = threadIdx.x; b < lds_words; b += blockDim.x) s_cnt[b] = 0;
+        __syncthreads();
+    }
     ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
     ll stride = (ll)gridDim.x * blockDim.x;
     unsigned long long local_valid = 0;
@@ -296,19 +308,40 @@ __global__ void __launch_bounds__(256) k_mc_stat_labels(
         ll p = preds[i];
         if (t < 0 || t >= C || p < 0 || p >= C) continue;
         local_valid++;
-        if (p == t) {
-            atomicAdd(&tp[t], 1ULL);
+        if (use_lds) {
+            if (p == t) {
+                atomicAdd(&s_tp[t], 1u);
+            } else {
+                atomicAdd(&s_fp[p], 1u);
+                atomicAdd(&s_fn[t], 1u);
+            }
+            if (s_cm) atomicAdd(&s_cm[t * C + p], 1u);
+            else if (confmat) atomicAdd(&confmat[t * C + p], 1ULL);
         } else {
-            atomicAdd(&fp[p], 1ULL);
-            atomicAdd(&fn[t], 1ULL);
+            if (p == t) {
+                atomicAdd(&tp[t], 1ULL);
+            } else {
+                atomicAdd(&fp[p], 1ULL);
+                atomicAdd(&fn[t], 1ULL);
+            }
+            if (confmat) atomicAdd(&confmat[t * C + p], 1ULL);
         }
-        if (confmat) atomicAdd(&confmat[t * C + p], 1ULL);
     }
     __shared__ unsigned long long blk_valid;
     if (threadIdx.x == 0) blk_valid = 0;
     __syncthreads();
     if (local_valid) atomicAdd(&blk_valid, local_valid);
     __syncthreads();
+    if (use_lds) {
+        for (ll c = threadIdx.x; c < C; c += blockDim.x) {
+            if (s_tp[c]) atomicAdd(&tp[c], (unsigned long long)s_tp[c]);
+            if (s_fp[c]) atomicAdd(&fp[c], (unsigned long long)s_fp[c]);
+            if (s_fn[c]) atomicAdd(&fn[c], (unsigned long long)s_fn[c]);
+        }
+        if (s_cm)
+            for (ll b = threadIdx.x; b < C * C; b += blockDim.x)
+                if (s_cm[b]) atomicAdd(&confmat[b], (unsigned long long)s_cm[b]);
+    }
     if (threadIdx.x == 0 && blk_valid) atomicAdd(valid_count, blk_valid);
 }
 
@@ -399,7 +432,13 @@ template <typename T, bool IS_BF16>
 __global__ void __launch_bounds__(256) k_multilabel_stat(
     const T* __restrict__ preds, const ll* __restrict__ target, ll N, ll L, float threshold,
     ll ignore_index, int has_ignore, unsigned long long* __restrict__ out,
-    unsigned int* __restrict__ outside_flag) {
+    unsigned int* __restrict__ outside_flag, int use_lds) {
+    // use_lds: privatize the 8L counters per block (small-L atomic contention)
+    extern __shared__ unsigned int s_ml[];
+    if (use_lds) {
+        for (ll b = threadIdx.x; b < 8 * L; b += blockDim.x) s_ml[b] = 0;
+        __syncthreads();
+    }
     ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
     ll stride = (ll)gridDim.x * blockDim.x;
     const float logit_thr = logf(threshold / (1.0f - threshold));
@@ -414,8 +453,20 @@ __global__ void __launch_bounds__(256) k_multilabel_stat(
         int pr_raw = p > threshold;
         int pr_sig = p > logit_thr;
         int tt = (int)t;
-        atomicAdd(&out[(0 * L + l) * 4 + (pr_raw == 1 ? (tt == 1 ? 0 : 1) : (tt == 0 ? 2 : 3))], 1ULL);
-        atomicAdd(&out[(1 * L + l) * 4 + (pr_sig == 1 ? (tt == 1 ? 0 : 1) : (tt == 0 ? 2 : 3))], 1ULL);
+        const ll j0 = (0 * L + l) * 4 + (pr_raw == 1 ? (tt == 1 ? 0 : 1) : (tt == 0 ? 2 : 3));
+        const ll j1 = (1 * L + l) * 4 + (pr_sig == 1 ? (tt == 1 ? 0 : 1) : (tt == 0 ? 2 : 3));
+        if (use_lds) {
+            atomicAdd(&s_ml[j0], 1u);
+            atomicAdd(&s_ml[j1], 1u);
+        } else {
+            atomicAdd(&out[j0], 1ULL);
+            atomicAdd(&out[j1], 1ULL);
+        }
+    }
+    if (use_lds) {
+        __syncthreads();
+        for (ll b = threadIdx.x; b < 8 * L; b += blockDim.x)
+            if (s_ml[b]) atomicAdd(&out[b], (unsigned long long)s_ml[b]);
     }
     for (int off = WAVE / 2; off > 0; off >>= 1) outside |= __shfl_down(outside, off);
     if ((threadIdx.x & (WAVE - 1)) == 0 && outside) atomicOr(outside_flag, 1u);
@@ -1229,10 +1280,15 @@ int ma_mc_stat_labels(uintptr_t stream, uintptr_t preds, uintptr_t target, ll N,
                       ll ignore_index, int has_ignore, uintptr_t tp, uintptr_t fp, uintptr_t fn,
                       uintptr_t confmat, uintptr_t valid_count) {
     hipStream_t s = (hipStream_t)stream;
-    k_mc_stat_labels<<<grid_for(N, 256), 256, 0, s>>>(
+    const int use_lds = C <= 128;
+    const int priv_cm = confmat && C <= 64;
+    const size_t shmem = use_lds ? (size_t)(3 * C + (priv_cm ? C * C : 0)) * sizeof(unsigned int) : 0;
+    int grid = grid_for(N, 256);
+    if (grid > 4096) grid = 4096;
+    k_mc_stat_labels<<<grid, 256, shmem, s>>>(
         (const ll*)preds, (const ll*)target, N, C, ignore_index, has_ignore,
         (unsigned long long*)tp, (unsigned long long*)fp, (unsigned long long*)fn,
-        (unsigned long long*)confmat, (unsigned long long*)valid_count);
+        (unsigned long long*)confmat, (unsigned long long*)valid_count, use_lds, priv_cm);
     return (int)hipGetLastError();
 }
 
@@ -1267,14 +1323,18 @@ int ma_multilabel_stat(uintptr_t stream, uintptr_t preds, int dtype, uintptr_t t
                        float threshold, ll ignore_index, int has_ignore, uintptr_t out,
                        uintptr_t outside_flag) {
     hipStream_t s = (hipStream_t)stream;
+    const int use_lds = L <= 512;  // 8L uint counters, <=16 KB LDS
+    const size_t shmem = use_lds ? (size_t)(8 * L) * sizeof(unsigned int) : 0;
+    int grid = grid_for(N * L, 256);
+    if (grid > 4096) grid = 4096;
     if (dtype == 0)
-        k_multilabel_stat<float, false><<<grid_for(N * L, 256), 256, 0, s>>>(
+        k_multilabel_stat<float, false><<<grid, 256, shmem, s>>>(
             (const float*)preds, (const ll*)target, N, L, threshold, ignore_index, has_ignore,
-            (unsigned long long*)out, (unsigned int*)outside_flag);
+            (unsigned long long*)out, (unsigned int*)outside_flag, use_lds);
     else
-        k_multilabel_stat<unsigned short, true><<<grid_for(N * L, 256), 256, 0, s>>>(
+        k_multilabel_stat<unsigned short, true><<<grid, 256, shmem, s>>>(
             (const unsigned short*)preds, (const ll*)target, N, L, threshold, ignore_index,
-            has_ignore, (unsigned long long*)out, (unsigned int*)outside_flag);
+            has_ignore, (unsigned long long*)out, (unsigned int*)outside_flag, use_lds);
     return (int)hipGetLastError();
 }
 
