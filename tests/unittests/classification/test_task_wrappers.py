@@ -128,3 +128,33 @@ def test_functional_clip_raise_without_model():
 
     with pytest.raises(ModuleNotFoundError):
         clip_score(torch.rand(3, 8, 8), "a photo")
+
+
+def test_dice_without_num_classes_micro():
+    """Legacy Dice infers the class count for micro averaging (reference parity)."""
+    import metrics_amd as ma
+
+    preds = torch.tensor([2, 0, 2, 1])
+    target = torch.tensor([1, 1, 2, 0])
+    assert ma.Dice(average="micro")(preds, target).item() == pytest.approx(0.25)
+    assert ma.Dice()(preds, target).item() == pytest.approx(0.25)
+    # probs input path
+    probs = torch.nn.functional.one_hot(preds, 3).float()
+    assert ma.Dice()(probs, target).item() == pytest.approx(0.25)
+
+
+def test_input_transformer_passes_kwargs_through():
+    """BinaryTargetTransformer forwards retrieval-style kwargs untouched."""
+    import metrics_amd as ma  # noqa: F401
+    from metrics_amd.retrieval import RetrievalMRR
+    from metrics_amd.wrappers import BinaryTargetTransformer
+
+    torch.manual_seed(0)
+    preds = torch.rand(10)
+    topics = torch.randint(0, 2, (10,))
+    targets = torch.randint(0, 5, (10,))
+    m = BinaryTargetTransformer(RetrievalMRR(), threshold=2)
+    m.update(preds, targets, indexes=topics)
+    expected = RetrievalMRR()
+    expected.update(preds, (targets > 2).long(), indexes=topics)
+    assert torch.allclose(m.compute(), expected.compute())

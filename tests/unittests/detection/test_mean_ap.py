@@ -1,4 +1,5 @@
 """MeanAveragePrecision: reference doctest oracles + segm/RLE tests."""
+import pytest
 import torch
 
 from metrics_amd.detection import MeanAveragePrecision
@@ -104,3 +105,50 @@ def test_map_segm_rle_roundtrip():
     assert (dec == ref).all()
     empty = _encode_masks_rle(torch.zeros(0, 5, 5, dtype=torch.bool))
     assert _decode_masks_rle(empty.numpy()).shape == (0, 25)
+
+
+def test_tm_to_coco_round_trip(tmp_path):
+    """tm_to_coco -> coco_to_tm reproduces boxes/masks (compressed-RLE strings)."""
+    import torch
+    from torch import tensor
+
+    from metrics_amd.detection import MeanAveragePrecision
+
+    name = str(tmp_path / "rt")
+    m = MeanAveragePrecision(iou_type="bbox")
+    m.update(
+        [dict(boxes=tensor([[258.0, 41.0, 606.0, 285.0]]), scores=tensor([0.536]), labels=tensor([0]))],
+        [dict(boxes=tensor([[214.0, 41.0, 562.0, 285.0]]), labels=tensor([0]))],
+    )
+    m.tm_to_coco(name)
+    preds, target = MeanAveragePrecision.coco_to_tm(f"{name}_preds.json", f"{name}_target.json", iou_type="bbox")
+    assert torch.allclose(preds[0]["boxes"], tensor([[258.0, 41.0, 606.0, 285.0]]))
+    assert torch.allclose(target[0]["boxes"], tensor([[214.0, 41.0, 562.0, 285.0]]))
+    assert preds[0]["scores"].item() == pytest.approx(0.536)
+
+    masks = torch.zeros(2, 10, 12, dtype=torch.uint8)
+    masks[0, 2:6, 3:9] = 1
+    masks[1, 1:3, 1:3] = 1
+    ms = MeanAveragePrecision(iou_type="segm")
+    ms.update(
+        [dict(masks=masks, scores=tensor([0.7, 0.3]), labels=tensor([1, 2]))],
+        [dict(masks=masks, labels=tensor([1, 2]))],
+    )
+    ms.tm_to_coco(str(tmp_path / "seg"))
+    ps, ts = MeanAveragePrecision.coco_to_tm(
+        str(tmp_path / "seg_preds.json"), str(tmp_path / "seg_target.json"), iou_type="segm"
+    )
+    assert (ps[0]["masks"] == masks).all()
+    assert (ts[0]["masks"] == masks).all()
+
+
+def test_coco_rle_string_codec():
+    """Compressed-RLE string encode/decode round-trips arbitrary run lists."""
+    import numpy as np
+
+    from metrics_amd.detection.mean_ap import _coco_rle_str_decode, _coco_rle_str_encode
+
+    rng = np.random.default_rng(0)
+    for _ in range(20):
+        counts = rng.integers(0, 10_000, size=rng.integers(1, 40)).tolist()
+        assert _coco_rle_str_decode(_coco_rle_str_encode(counts)) == counts
