@@ -32,6 +32,14 @@ class _TopKRetrievalMetric(RetrievalMetric):
             raise ValueError("`top_k` has to be a positive integer or None")
         self.top_k = top_k
 
+    def _k_eff(self, g) -> Tensor:
+        """Per-query truncation length: min(top_k, query size)."""
+        return g.counts.clamp(max=self.top_k) if self.top_k is not None else g.counts
+
+    def _in_k(self, g) -> Tensor:
+        """Elementwise mask: element is within its query's top-k by pred."""
+        return g.ranks < self._k_eff(g)[g.gid]
+
 
 class RetrievalMAP(_TopKRetrievalMetric):
     """Mean average precision."""
@@ -39,12 +47,31 @@ class RetrievalMAP(_TopKRetrievalMetric):
     def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
         return retrieval_average_precision(preds, target, top_k=self.top_k)
 
+    def _batched_scores(self, g):
+        tf = (g.target.float() > 0).float()
+        in_k = self._in_k(g).float()
+        cum = torch.cumsum(tf, 0)
+        base = (cum - tf)[g.starts]  # exclusive prefix at group start
+        cum_in_group = cum - base[g.gid]  # inclusive positive count at each rank
+        prec_at_pos = cum_in_group / (g.ranks + 1).float()
+        hits_in_k = g.seg_sum(tf * in_k)
+        ap_num = g.seg_sum(prec_at_pos * tf * in_k)
+        return torch.where(hits_in_k > 0, ap_num / hits_in_k.clamp(min=1), torch.zeros_like(ap_num))
+
 
 class RetrievalMRR(_TopKRetrievalMetric):
     """Mean reciprocal rank."""
 
     def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
         return retrieval_reciprocal_rank(preds, target, top_k=self.top_k)
+
+    def _batched_scores(self, g):
+        pos = g.target.float() > 0
+        big = torch.iinfo(torch.long).max
+        r0 = torch.full((g.G,), big, device=g.ranks.device, dtype=torch.long)
+        r0.scatter_reduce_(0, g.gid[pos], g.ranks[pos], reduce="amin")
+        found = r0 < self._k_eff(g)
+        return torch.where(found, 1.0 / (r0.clamp(max=big - 1) + 1).float(), torch.zeros(g.G, device=g.ranks.device))
 
 
 class RetrievalPrecision(_TopKRetrievalMetric):
@@ -60,12 +87,27 @@ class RetrievalPrecision(_TopKRetrievalMetric):
     def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
         return retrieval_precision(preds, target, top_k=self.top_k, adaptive_k=self.adaptive_k)
 
+    def _batched_scores(self, g):
+        # divisor: top_k (even past the query size) unless adaptive/None -> query size
+        if self.top_k is None:
+            div = g.counts.float()
+        elif self.adaptive_k:
+            div = g.counts.clamp(max=self.top_k).float()
+        else:
+            div = torch.full((g.G,), float(self.top_k), device=g.counts.device)
+        hits = g.seg_sum((g.target.float() > 0).float() * self._in_k(g).float())
+        return hits / div
+
 
 class RetrievalRecall(_TopKRetrievalMetric):
     """Recall@k."""
 
     def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
         return retrieval_recall(preds, target, top_k=self.top_k)
+
+    def _batched_scores(self, g):
+        hits = g.seg_sum((g.target.float() > 0).float() * self._in_k(g).float())
+        return hits / g.npos.clamp(min=1)
 
 
 class RetrievalHitRate(_TopKRetrievalMetric):
@@ -74,39 +116,23 @@ class RetrievalHitRate(_TopKRetrievalMetric):
     def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
         return retrieval_hit_rate(preds, target, top_k=self.top_k)
 
+    def _batched_scores(self, g):
+        hits = g.seg_sum((g.target.float() > 0).float() * self._in_k(g).float())
+        return (hits > 0).float()
+
 
 class RetrievalFallOut(_TopKRetrievalMetric):
     """Fall-out@k (lower is better; empty_target_action applies to queries with no NEGATIVES)."""
 
     higher_is_better = False
-
-    def compute(self) -> Tensor:
-        indexes = dim_zero_cat(self.indexes)
-        preds = dim_zero_cat(self.preds)
-        target = dim_zero_cat(self.target)
-
-        indexes, indices = torch.sort(indexes)
-        preds = preds[indices]
-        target = target[indices]
-        split_sizes = _flexible_bincount(indexes).detach().cpu().tolist()
-
-        res = []
-        for mini_preds, mini_target in zip(
-            torch.split(preds, split_sizes, dim=0), torch.split(target, split_sizes, dim=0)
-        ):
-            if (1 - mini_target).sum() == 0:
-                if self.empty_target_action == "error":
-                    raise ValueError("`compute` method was provided with a query with no negative target.")
-                if self.empty_target_action == "pos":
-                    res.append(torch.tensor(1.0, device=preds.device))
-                elif self.empty_target_action == "neg":
-                    res.append(torch.tensor(0.0, device=preds.device))
-            else:
-                res.append(self._metric(mini_preds, mini_target))
-        return torch.stack([x.to(preds) for x in res]).mean() if res else torch.tensor(0.0).to(preds)
+    _empty_on_negatives = True
 
     def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
         return retrieval_fall_out(preds, target, top_k=self.top_k)
+
+    def _batched_scores(self, g):
+        neg_hits = g.seg_sum((g.target.float() == 0).float() * self._in_k(g).float())
+        return neg_hits / g.nneg.clamp(min=1)
 
 
 class RetrievalNormalizedDCG(_TopKRetrievalMetric):
@@ -120,12 +146,29 @@ class RetrievalNormalizedDCG(_TopKRetrievalMetric):
     def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
         return retrieval_normalized_dcg(preds, target, top_k=self.top_k)
 
+    def _batched_scores(self, g):
+        tf = g.target.float()
+        in_k = self._in_k(g).float()
+        disc = torch.log2(g.ranks.float() + 2.0)
+        dcg = g.seg_sum(tf / disc * in_k)
+        # ideal ordering: stable lexsort (gid asc, target desc) reusing g.ranks grid
+        ord1 = torch.argsort(tf, descending=True, stable=True)
+        ord2 = torch.argsort(g.gid[ord1], stable=True)
+        t_ideal = tf[ord1][ord2]
+        idcg = g.seg_sum(t_ideal / disc * in_k)
+        return torch.where(idcg > 0, dcg / idcg.clamp(min=1e-38), torch.zeros_like(dcg))
+
 
 class RetrievalRPrecision(RetrievalMetric):
     """R-precision."""
 
     def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
         return retrieval_r_precision(preds, target)
+
+    def _batched_scores(self, g):
+        in_r = (g.ranks < g.npos[g.gid].long()).float()
+        hits = g.seg_sum((g.target.float() > 0).float() * in_r)
+        return hits / g.npos.clamp(min=1)
 
 
 class RetrievalAUROC(_TopKRetrievalMetric):

@@ -1,4 +1,5 @@
 """RetrievalAUROC + wrapper bits not covered elsewhere."""
+import pytest
 import torch
 from sklearn.metrics import roc_auc_score
 
@@ -69,3 +70,60 @@ def test_empty_target_action_semantics():
     m.update(preds, target, indexes=idx)
     with _pytest.raises(ValueError):
         m.compute()
+
+
+@pytest.mark.parametrize(
+    ("cls", "kwargs", "graded"),
+    [
+        ("RetrievalMAP", {}, False),
+        ("RetrievalMAP", {"top_k": 5}, False),
+        ("RetrievalMRR", {}, False),
+        ("RetrievalMRR", {"top_k": 3}, False),
+        ("RetrievalPrecision", {"top_k": 4}, False),
+        ("RetrievalPrecision", {"top_k": 100}, False),
+        ("RetrievalPrecision", {"top_k": 100, "adaptive_k": True}, False),
+        ("RetrievalRecall", {"top_k": 4}, False),
+        ("RetrievalHitRate", {"top_k": 2}, False),
+        ("RetrievalFallOut", {"top_k": 4}, False),
+        ("RetrievalNormalizedDCG", {}, True),
+        ("RetrievalNormalizedDCG", {"top_k": 6}, True),
+        ("RetrievalRPrecision", {}, False),
+    ],
+)
+def test_batched_retrieval_matches_per_query_loop(cls, kwargs, graded):
+    """The vectorized all-queries path equals the per-query loop exactly."""
+    import metrics_amd.retrieval as R
+
+    torch.manual_seed(11)
+    n, q = 3000, 200
+    idx = torch.randint(0, q, (n,))
+    preds = torch.rand(n)
+    target = torch.randint(0, 4 if graded else 2, (n,))
+    klass = getattr(R, cls)
+    fast = klass(**kwargs)
+    fast.update(preds, target, indexes=idx)
+    loop = klass(**kwargs)
+    loop._batched_scores = lambda g: None  # force the fallback loop
+    loop.update(preds, target, indexes=idx)
+    assert torch.allclose(fast.compute(), loop.compute(), atol=1e-6)
+
+
+@pytest.mark.parametrize("action", ["neg", "pos", "skip"])
+def test_batched_retrieval_empty_target_actions(action):
+    """empty_target_action handling matches between batched and loop paths."""
+    import metrics_amd.retrieval as R
+
+    idx = torch.tensor([0, 0, 1, 1, 2, 2])
+    preds = torch.tensor([0.9, 0.2, 0.4, 0.3, 0.8, 0.1])
+    target = torch.tensor([1, 0, 0, 0, 1, 1])  # query 1 has no positives
+    fast = R.RetrievalMAP(empty_target_action=action)
+    fast.update(preds, target, indexes=idx)
+    loop = R.RetrievalMAP(empty_target_action=action)
+    loop._batched_scores = lambda g: None
+    loop.update(preds, target, indexes=idx)
+    assert torch.allclose(fast.compute(), loop.compute())
+
+    err = R.RetrievalMAP(empty_target_action="error")
+    err.update(preds, target, indexes=idx)
+    with pytest.raises(ValueError, match="no positive"):
+        err.compute()
