@@ -747,3 +747,45 @@ def test_noncontiguous_inputs_gpu():
     gb.update(bp, bt)
     cb.update(bp.cpu(), bt.cpu())
     assert torch.allclose(gb.compute().cpu(), cb.compute(), atol=1e-6)
+
+
+def test_retrieval_batched_gpu_matches_cpu():
+    """Vectorized retrieval path on device == CPU reference (and the loop path)."""
+    torch.manual_seed(31)
+    n, q = 20_000, 1000
+    idx = torch.randint(0, q, (n,))
+    preds = torch.rand(n)
+    target = torch.randint(0, 2, (n,))
+    for cls, kw in (
+        (ma.retrieval.RetrievalMAP, {}),
+        (ma.retrieval.RetrievalMRR, {"top_k": 3}),
+        (ma.retrieval.RetrievalNormalizedDCG, {"top_k": 10}),
+        (ma.retrieval.RetrievalPrecision, {"top_k": 5}),
+    ):
+        g = cls(**kw).to("cuda")
+        g.update(preds.cuda(), target.cuda(), indexes=idx.cuda())
+        c = cls(**kw)
+        c.update(preds, target, indexes=idx)
+        assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-5), cls.__name__
+
+
+def test_smallc_stat_kernel_gpu():
+    """Thread-per-row small-C variant: confusion matrix + accuracy parity, C in {3, 48, 64, 100}."""
+    torch.manual_seed(32)
+    for C in (3, 48, 64, 100):
+        preds = torch.randn(10_000, C, device="cuda", dtype=torch.bfloat16)
+        target = torch.randint(0, C, (10_000,), device="cuda")
+        g = ma.MulticlassConfusionMatrix(num_classes=C).to("cuda")
+        g.update(preds, target)
+        cm = g.compute().cpu()
+        am = preds.float().argmax(1).cpu()
+        ref = torch.zeros(C, C, dtype=torch.long)
+        for t, p in zip(target.cpu().tolist(), am.tolist()):
+            ref[t, p] += 1
+        assert (cm.long() == ref).all(), C
+        # label-path (no logits) small-C
+        gl = ma.MulticlassAccuracy(num_classes=C, average="micro").to("cuda")
+        gl.update(am.cuda(), target)
+        cl = ma.MulticlassAccuracy(num_classes=C, average="micro")
+        cl.update(am, target.cpu())
+        assert torch.allclose(gl.compute().cpu(), cl.compute(), atol=1e-6), C
