@@ -303,8 +303,7 @@ def _test_mean_ap_ddp(rank, world_size):
     for i in range(rank, 4, world_size):
         m.update([all_preds[i]], [all_tgts[i]])
     res = m.compute()
-    ref = ma.detection.MeanAveragePrecision()
-    ref.sync_on_compute = False
+    ref = ma.detection.MeanAveragePrecision(sync_on_compute=False)
     for i in range(4):
         ref.update([all_preds[i]], [all_tgts[i]])
     expected = ref.compute()
@@ -312,9 +311,33 @@ def _test_mean_ap_ddp(rank, world_size):
     assert torch.allclose(res["mar_100"], expected["mar_100"], atol=1e-6)
 
 
+def _test_retrieval_batched_ddp(rank, world_size):
+    """Retrieval list states gather; the batched compute path sees all ranks' queries."""
+    import metrics_amd as ma
+
+    torch.manual_seed(12)
+    idx = torch.randint(0, 40, (400,))
+    preds = torch.rand(400)
+    target = torch.randint(0, 2, (400,))
+    shard = slice(rank * 200, (rank + 1) * 200)
+    for cls, kw in (
+        (ma.retrieval.RetrievalMAP, {}),
+        (ma.retrieval.RetrievalMRR, {"top_k": 3}),
+        (ma.retrieval.RetrievalNormalizedDCG, {}),
+    ):
+        m = cls(**kw)
+        m.update(preds[shard], target[shard], indexes=idx[shard])
+        res = m.compute()
+        # sync_on_compute must go through the constructor: _to_sync snapshots it
+        ref = cls(**kw, sync_on_compute=False)
+        ref.update(preds, target, indexes=idx)
+        assert torch.allclose(res, ref.compute(), atol=1e-6), cls.__name__
+
+
 @pytest.mark.parametrize(
     "fn",
-    [_test_curve_metrics_ddp, _test_mc_curve_ddp, _test_pearson_welford_merge_ddp, _test_mean_ap_ddp],
+    [_test_curve_metrics_ddp, _test_mc_curve_ddp, _test_pearson_welford_merge_ddp, _test_mean_ap_ddp,
+     _test_retrieval_batched_ddp],
 )
 def test_ddp_metric_parity(fn):
     run_distributed(fn, world_size=2)
