@@ -67,16 +67,26 @@ def _flatten_dict(x: dict) -> tuple:
     return new_dict, duplicates
 
 
-def to_onehot(label_tensor: Tensor, num_classes: int) -> Tensor:
-    """Convert a dense label tensor ``(N, ...)`` to one-hot ``(N, C, ...)``."""
+def to_onehot(label_tensor: Tensor, num_classes: Optional[int] = None) -> Tensor:
+    """Convert a dense label tensor ``(N, ...)`` to one-hot ``(N, C, ...)``.
+
+    ``num_classes`` defaults to ``max(labels) + 1`` (reference utilities/data.py:81).
+    """
     if label_tensor.ndim == 0:
         label_tensor = label_tensor.unsqueeze(0)
+    if num_classes is None:
+        num_classes = int(label_tensor.max().detach().item() + 1)
     shape = label_tensor.shape
     out = torch.zeros(
         shape[0], num_classes, *shape[1:], dtype=label_tensor.dtype, device=label_tensor.device
     )
     index = label_tensor.long().unsqueeze(1).expand_as(out.narrow(1, 0, 1)).clamp_(0, num_classes - 1)
     return out.scatter_(1, index, 1.0)
+
+
+def to_categorical(x: Tensor, argmax_dim: int = 1) -> Tensor:
+    """Convert a probability tensor to dense labels via argmax (reference utilities/data.py:151)."""
+    return torch.argmax(x, dim=argmax_dim)
 
 
 def select_topk(prob_tensor: Tensor, topk: int = 1, dim: int = 1) -> Tensor:

@@ -112,6 +112,12 @@ def run(domains=None, verbose=True, modular=True):
 
     globset.update({k: v for k, v in vars(F).items() if not k.startswith("_")})
     globset.update({k: v for k, v in vars(M).items() if not k.startswith("_")})
+    import metrics_amd.utilities.compute as UC
+    import metrics_amd.utilities.data as UD
+    import metrics_amd.utilities.checks as UK
+
+    for mod in (UC, UD, UK):
+        globset.update({k: v for k, v in vars(mod).items() if not k.startswith("_") and callable(v)})
 
     paths = []
     for d in domains or DOMAINS:
@@ -120,7 +126,14 @@ def run(domains=None, verbose=True, modular=True):
             paths += sorted(glob.glob(f"/root/reference/src/torchmetrics/{d}/*.py"))
     if modular and (domains is None or "core" in (domains or [])):
         paths += sorted(glob.glob("/root/reference/src/torchmetrics/wrappers/*.py"))
-        paths += ["/root/reference/src/torchmetrics/aggregation.py"]
+        paths += [
+            "/root/reference/src/torchmetrics/aggregation.py",
+            "/root/reference/src/torchmetrics/collections.py",
+            "/root/reference/src/torchmetrics/metric.py",
+            "/root/reference/src/torchmetrics/utilities/checks.py",
+            "/root/reference/src/torchmetrics/utilities/compute.py",
+            "/root/reference/src/torchmetrics/utilities/data.py",
+        ]
 
     for path in paths:
         rel = path.split("torchmetrics/")[-1]
