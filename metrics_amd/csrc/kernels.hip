@@ -856,22 +856,41 @@ __global__ void k_curve_suffix_tiled(
         if (zero_hist) hist[(c0 + lc) * (ll)(T + 1) * 2 + rem] = 0;
     }
     __syncthreads();
-    // per-class suffix scan (one thread per class, O(T))
-    if (threadIdx.x < ctile) {
-        const int lc = threadIdx.x;
-        unsigned long long* h = stage + lc * (T + 1) * 2;
-        unsigned long long* sf = suf + lc * (T + 1) * 2;
-        unsigned long long tp = 0, fp = 0;
-        for (int j = T; j >= 0; j--) {
-            sf[j * 2 + 0] = fp;  // suffix EXCLUDING j ... shifted below
-            sf[j * 2 + 1] = tp;
-            fp += h[j * 2 + 0];
-            tp += h[j * 2 + 1];
+    // wave-parallel per-class suffix scan: lane partials over T/64 chunks,
+    // cross-lane exclusive suffix via shfl, then a short serial tail per
+    // chunk — serial depth drops from O(T) to O(T/64)+log2(64)
+    {
+        const int lane = threadIdx.x & (WAVE - 1);
+        const int wv = threadIdx.x / WAVE;
+        const int nwaves = blockDim.x / WAVE;
+        const int chunk = (T + 1 + WAVE - 1) / WAVE;
+        for (int lc = wv; lc < ctile; lc += nwaves) {
+            unsigned long long* h = stage + lc * (T + 1) * 2;
+            unsigned long long* sf = suf + lc * (T + 1) * 2;
+            const int j0 = lane * chunk;
+            const int j1 = min(j0 + chunk, T + 1);
+            unsigned long long pf = 0, pt = 0;
+            for (int j = j0; j < j1; j++) { pf += h[j * 2 + 0]; pt += h[j * 2 + 1]; }
+            unsigned long long sfp = pf, stp = pt;  // inclusive suffix over lanes >= l
+            for (int off = 1; off < WAVE; off <<= 1) {
+                unsigned long long of = __shfl_down(sfp, off);
+                unsigned long long ot = __shfl_down(stp, off);
+                if (lane + off < WAVE) { sfp += of; stp += ot; }
+            }
+            unsigned long long run_f = sfp - pf;  // exclusive: sum over lanes > l
+            unsigned long long run_t = stp - pt;
+            for (int j = j1 - 1; j >= j0; j--) {
+                sf[j * 2 + 0] = run_f;  // sum over (j, T]
+                sf[j * 2 + 1] = run_t;
+                run_f += h[j * 2 + 0];
+                run_t += h[j * 2 + 1];
+            }
+            // lane 0's inclusive suffix is the grand total; slot 0 (never read
+            // as a suffix) stashes neg/pos totals — t=0 is reconstructed below
+            const unsigned long long tot_f = __shfl(sfp, 0);
+            const unsigned long long tot_t = __shfl(stp, 0);
+            if (lane == 0) { sf[0] = tot_f; sf[1] = tot_t; }
         }
-        // store totals at slot 0's running end: reuse registers via last iter:
-        // after the loop, fp/tp hold the FULL totals; stash them in sf[...]
-        sf[0] = fp;  // slot (j=0, which is never read as suffix) = neg_total
-        sf[1] = tp;  //                                            pos_total
     }
     __syncthreads();
     // wait: suffix for threshold t is sum over j > t, i.e. sf at j=t+1..T —
@@ -1461,7 +1480,7 @@ int ma_curve_suffix(uintptr_t stream, uintptr_t hist, ll outer, int T, int trans
         static int ctile_sel = 0;
         if (ctile_sel == 0) {
             const char* e = getenv("MA_SUFFIX_CTILE");
-            ctile_sel = e ? atoi(e) : 2;
+            ctile_sel = e ? atoi(e) : 4;  // one wave per class in the scan phase
         }
         const int CT = ctile_sel;
         size_t shmem_t = (size_t)2 * CT * (T + 1) * 2 * sizeof(unsigned long long);
