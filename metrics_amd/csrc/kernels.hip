@@ -655,7 +655,18 @@ __global__ void __launch_bounds__(256) k_multiclass_curve_hist(
     __syncthreads();
     const int lane = threadIdx.x & (WAVE - 1);
     const int wave = threadIdx.x / WAVE;
-    const ll c_lo = (ll)blockIdx.x * c_chunk;
+    // XCD-aware chunk mapping: consecutive blockIdx.x round-robin over the 8
+    // XCDs (each with its own L2), but adjacent class-chunks read 8B slices
+    // of the SAME cache lines — give each XCD a contiguous band of chunks so
+    // line sharing stays within one L2 instead of fetching the line 8x.
+    ll bx = blockIdx.x;
+    {
+        const ll nx = gridDim.x;
+        const ll band = nx / 8;
+        // bijection on the first 8*band blocks; the tail keeps identity
+        if (band > 1 && bx < 8 * band) bx = (bx % 8) * band + bx / 8;
+    }
+    const ll c_lo = bx * c_chunk;
     const ll c_hi = min(c_lo + (ll)c_chunk, C);
     const ll row = (ll)blockIdx.y * 256 + wave * WAVE + lane;
     bool valid = row < B;
