@@ -117,3 +117,56 @@ def test_running_in_collection_context():
         running.append(r.compute().item())
     assert outs == [1.0, 5.0, 9.0]
     assert running == [1.0, 3.0, 7.0]
+
+
+def test_metric_tracker_best_and_compute_all():
+    import metrics_amd as ma
+
+    tracker = ma.MetricTracker(ma.MeanSquaredError(), maximize=False)
+    for err in (1.0, 0.5, 2.0):
+        tracker.increment()
+        tracker.update(torch.full((4,), err), torch.zeros(4))
+    allv = tracker.compute_all()
+    assert allv.shape[0] == 3
+    best, idx = tracker.best_metric(return_step=True)
+    assert best == pytest.approx(0.25)
+    assert idx == 1
+
+
+def test_bootstrapper_mean_close_to_point_estimate():
+    import metrics_amd as ma
+
+    torch.manual_seed(7)
+    base = ma.BinaryAccuracy()
+    boot = ma.BootStrapper(ma.BinaryAccuracy(), num_bootstraps=20, mean=True, std=True)
+    preds, target = torch.rand(500), torch.randint(0, 2, (500,))
+    base.update(preds, target)
+    boot.update(preds, target)
+    out = boot.compute()
+    assert abs(out["mean"] - base.compute()) < 0.05
+    assert out["std"] >= 0
+
+
+def test_multitask_wrapper_routes_inputs():
+    import metrics_amd as ma
+
+    mt = ma.MultitaskWrapper({
+        "cls": ma.BinaryAccuracy(),
+        "reg": ma.MeanSquaredError(),
+    })
+    preds = {"cls": torch.tensor([1, 0, 1]).float(), "reg": torch.tensor([1.0, 2.0, 3.0])}
+    tgts = {"cls": torch.tensor([1, 1, 1]), "reg": torch.tensor([1.0, 2.0, 2.0])}
+    mt.update(preds, tgts)
+    out = mt.compute()
+    assert out["cls"] == pytest.approx(2 / 3)
+    assert out["reg"] == pytest.approx(1 / 3)
+
+
+def test_classwise_wrapper_labels():
+    import metrics_amd as ma
+
+    cw = ma.ClasswiseWrapper(ma.MulticlassAccuracy(num_classes=3, average=None), labels=["a", "b", "c"])
+    cw.update(torch.tensor([0, 1, 2, 2]), torch.tensor([0, 1, 1, 2]))
+    out = cw.compute()
+    assert set(out) == {"multiclassaccuracy_a", "multiclassaccuracy_b", "multiclassaccuracy_c"}
+    assert out["multiclassaccuracy_a"] == pytest.approx(1.0)
