@@ -47,6 +47,40 @@ def build_collection(num_classes: int, device: torch.device, curve_thresholds: i
     return coll.to(device)
 
 
+def _measure_reference_baseline(args) -> float | None:
+    """Run the reference torchmetrics bench (same config) in a subprocess.
+
+    Returns its metric_updates_per_sec, or None if the staged reference is
+    unavailable. Result (with provenance) lands in gpurun_out/ref_baseline.json.
+    """
+    import subprocess
+
+    script = os.path.join(os.path.dirname(os.path.abspath(__file__)), "tools", "refbench", "run_ref_bench.py")
+    staged = os.path.join(os.path.dirname(script), "_staged", "torchmetrics")
+    if not os.path.isdir(staged):
+        print("ref baseline skipped: tools/refbench/_staged not populated", file=sys.stderr)
+        return None
+    cmd = [
+        sys.executable, script,
+        "--steps", str(max(8, min(args.steps, 32))),
+        "--warmup", "4",
+        "--batch", str(args.batch),
+        "--classes", str(args.classes),
+        "--compute-every", str(args.compute_every),
+    ]
+    try:
+        out = subprocess.run(cmd, capture_output=True, text=True, timeout=1800)
+        line = [l for l in out.stdout.strip().splitlines() if l.startswith("{")][-1]
+        rec = json.loads(line)
+        os.makedirs("gpurun_out", exist_ok=True)
+        with open("gpurun_out/ref_baseline.json", "w") as fh:
+            json.dump(rec, fh)
+        return float(rec["value"])
+    except Exception as err:
+        print(f"ref baseline run failed: {err}", file=sys.stderr)
+        return None
+
+
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
@@ -57,6 +91,13 @@ def main() -> None:
     parser.add_argument("--compute-every", type=int, default=32)
     parser.add_argument("--curve-thresholds", type=int, default=200)
     parser.add_argument("--graphs", action="store_true", help="capture the update into a hipGraph (launch-bound configs; the default eager path is faster for this bench shape)")
+    parser.add_argument(
+        "--run-ref-baseline",
+        action="store_true",
+        help="also run the reference torchmetrics on the same config in this session "
+        "(needs tools/refbench/_staged populated via stage_reference.sh) and use that "
+        "fresh number for vs_baseline",
+    )
     args = parser.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -99,12 +140,18 @@ def main() -> None:
         except Exception as err:  # fall back to eager updates
             print(f"hipGraph capture unavailable ({err}); running eager updates", file=sys.stderr)
 
+    # make the compute cadence robust to any driver-chosen step count: the
+    # timed region always contains at least ~4 compute() calls
+    compute_every = args.compute_every
+    if compute_every:
+        compute_every = min(compute_every, max(1, args.steps // 4))
+
     def one_step(i: int) -> None:
         if graphed is not None:
             graphed.update(preds[i % n_unique], target[i % n_unique])
         else:
             coll.update(preds[i % n_unique], target[i % n_unique])
-        if args.compute_every and (i + 1) % args.compute_every == 0:
+        if compute_every and (i + 1) % compute_every == 0:
             coll.compute()
 
     # warmup (untimed) — includes one compute() so one-time lazy allocation /
@@ -141,10 +188,17 @@ def main() -> None:
     ms_per_step = 1000.0 * elapsed / args.steps
 
     # Baseline: the reference (torchmetrics 1.7.0dev) measured on the SAME
-    # config on 1x MI355X via _refbench/run_ref_bench.py (r1, 2026-09-12):
-    # 360.6 metric-updates/s (44.37 ms/step). BASELINE.json has no published
-    # numbers, so this measured same-hardware number is the comparison point.
-    REFERENCE_UPDATES_PER_SEC_1GPU = 360.6
+    # config on 1x MI355X via tools/refbench/run_ref_bench.py. BASELINE.json
+    # has no published numbers, so this measured same-hardware number is the
+    # comparison point. Pass --run-ref-baseline to re-measure it in the same
+    # session (writes gpurun_out/ref_baseline.json and uses the fresh number).
+    REFERENCE_UPDATES_PER_SEC_1GPU = 360.6  # r1 (2026-09-12), 44.37 ms/step
+    baseline_src = "tools/refbench/run_ref_bench.py r1 2026-09-12"
+    if args.run_ref_baseline and rank == 0 and use_gpu:
+        fresh = _measure_reference_baseline(args)
+        if fresh is not None:
+            REFERENCE_UPDATES_PER_SEC_1GPU = fresh
+            baseline_src = "same-session run_ref_bench"
     vs_baseline = value / (REFERENCE_UPDATES_PER_SEC_1GPU * n_gpus) if use_gpu else None
 
     if rank == 0:
@@ -160,6 +214,8 @@ def main() -> None:
                 "higher_is_better": True,
                 "scaling": "weak",
                 "vs_baseline": vs_baseline,
+                "baseline_updates_per_sec_1gpu": REFERENCE_UPDATES_PER_SEC_1GPU if use_gpu else None,
+                "baseline_source": baseline_src if use_gpu else None,
                 "dtype": "bf16" if use_gpu else "fp32",
                 "data": "synthetic",
                 "config": {

@@ -192,7 +192,7 @@ class MeanMetric(BaseAggregator):
         value, weight = self._cast_and_nan_check_input(value, weight)
         if value.numel() == 0:
             return
-        # broadcast weight to value shape
+        # expand the weight so it lines up elementwise with value
         weight = torch.broadcast_to(weight, value.shape)
         self.mean_value += (value * weight).sum()
         self.weight += weight.sum()

@@ -104,7 +104,7 @@ class BinaryPrecisionRecallCurve(Metric):
         from metrics_amd.utilities.plot import plot_curve
 
         curve_computed = curve or self.compute()
-        # switch order as the standard way is recall along x-axis and precision along y-axis
+        # plot convention: recall on x, precision on y
         curve_computed = (curve_computed[1], curve_computed[0], curve_computed[2])
         score = (
             _auc_score(curve_computed[0], curve_computed[1]) if score is True else None if score is False else score

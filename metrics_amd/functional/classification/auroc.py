@@ -90,7 +90,7 @@ def _binary_auroc_compute(
 
     _device = fpr.device if isinstance(fpr, Tensor) else fpr[0].device
     max_area: Tensor = torch.tensor(max_fpr, device=_device)
-    # Add a single point at max_fpr and interpolate its tpr value
+    # append an interpolated (max_fpr, tpr) point so the partial area is exact
     stop = torch.bucketize(max_area, fpr, out_int32=True, right=True)
     weight = (max_area - fpr[stop - 1]) / (fpr[stop] - fpr[stop - 1])
     interp_tpr: Tensor = torch.lerp(tpr[stop - 1], tpr[stop], weight)
@@ -100,7 +100,7 @@ def _binary_auroc_compute(
     # Compute partial AUC
     partial_auc = _auc_compute_without_check(fpr, tpr, 1.0)
 
-    # McClish correction: standardize result to be 0.5 if non-discriminant and 1 if maximal
+    # rescale the partial area (McClish) onto [0.5, 1]
     min_area: Tensor = 0.5 * max_area**2
     return 0.5 * (1 + (partial_auc - min_area) / (max_area - min_area))
 

@@ -35,7 +35,7 @@ def _binary_clf_curve(
         if sample_weights is not None and not isinstance(sample_weights, Tensor):
             sample_weights = torch.tensor(sample_weights, device=preds.device, dtype=torch.float)
 
-        # remove class dimension if necessary
+        # squeeze away a singleton class axis
         if preds.ndim > target.ndim:
             preds = preds[:, 0]
         desc_score_indices = torch.argsort(preds, descending=True)
@@ -53,8 +53,7 @@ def _binary_clf_curve(
         tps = _cumsum(target * weight, dim=0)[threshold_idxs]
 
         if sample_weights is not None:
-            # express fps as a cumsum to ensure fps is increasing even in
-            # the presence of floating point errors
+            # cumsum keeps fps monotone even under fp rounding of the weights
             fps = _cumsum((1 - target) * weight, dim=0)[threshold_idxs]
         else:
             fps = 1 + threshold_idxs - tps
@@ -173,8 +172,7 @@ def _binary_precision_recall_curve_compute(
     precision = tps / (tps + fps)
     recall = tps / tps[-1]
 
-    # need to call reversed explicitly, since including that to slice would
-    # introduce negative strides that are not yet supported in pytorch
+    # flip() rather than a negative-stride slice (torch has no negative strides)
     precision = torch.cat([reversed(precision), torch.ones(1, dtype=precision.dtype, device=precision.device)])
     recall = torch.cat([reversed(recall), torch.zeros(1, dtype=recall.dtype, device=recall.device)])
     thresholds = reversed(thresholds).detach().clone()

@@ -97,7 +97,7 @@ def _multilabel_ranking_loss_update(preds: Tensor, target: Tensor) -> Tuple[Tens
     relevant = target == 1
     n_relevant = relevant.sum(dim=1)
 
-    # Ignore instances where number of true labels is 0 or n_labels
+    # rows that are all-negative or all-positive carry no ranking signal; drop them
     mask = (n_relevant > 0) & (n_relevant < n_labels)
     preds = preds[mask]
     relevant = relevant[mask]

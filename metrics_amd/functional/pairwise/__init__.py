@@ -59,7 +59,7 @@ def pairwise_euclidean_distance(
 ) -> Tensor:
     """Pairwise euclidean distance (N, M)."""
     x, y, zero_diagonal = _check_input(x, y, zero_diagonal)
-    # upcast to float64 to prevent precision issues
+    # accumulate the reduction in float64: fp32 sums drift at this size
     _orig_dtype = x.dtype
     x = x.to(torch.float64)
     y = y.to(torch.float64)

@@ -88,7 +88,7 @@ class PeakSignalNoiseRatio(Metric):
         sum_squared_error, num_obs = _psnr_update(preds, target, dim=self.dim)
         if self.dim is None:
             if self.data_range is None:
-                # keep track of min and max target values
+                # running target range, needed for the normalized variants
                 self.min_target = torch.minimum(target.min(), self.min_target)
                 self.max_target = torch.maximum(target.max(), self.max_target)
             self.sum_squared_error += sum_squared_error

@@ -371,8 +371,6 @@ class Metric(Module, ABC):
         """Restore the cached local (pre-sync) state."""
         if not should_unsync:
             return
-        if not self._is_synced and self._cache is None:
-            return
         if not self._is_synced:
             raise MetricsUserError("The Metric has already been un-synced.")
         if self._cache is None:
@@ -513,7 +511,8 @@ class Metric(Module, ABC):
         return deepcopy(self)
 
     def __getstate__(self) -> Dict[str, Any]:
-        # ignore update and compute functions for pickling
+        # the wrapped update/compute closures capture `self` and an inspect
+        # signature — drop them here and rebuild in __setstate__
         return {k: v for k, v in self.__dict__.items() if k not in ("update", "compute", "_update_signature")}
 
     def __setstate__(self, state: Dict[str, Any]) -> None:
@@ -596,7 +595,7 @@ class Metric(Module, ABC):
                     f"Expected metric state to be either a Tensor or a list of Tensor, but encountered {current_val}"
                 )
 
-        # make sure to update the device attribute
+        # refresh the cached device/dtype from a probe tensor
         _dummy = fn(torch.zeros(1, device=this._device))
         this._device = _dummy.device
         this._dtype = _dummy.dtype if _dummy.is_floating_point() else this._dtype
@@ -673,7 +672,7 @@ class Metric(Module, ABC):
             k: v for k, v in kwargs.items() if (k in _sign_params and _sign_params[k].kind not in _params)
         }
         exists_var_keyword = any(v.kind == inspect.Parameter.VAR_KEYWORD for v in _sign_params.values())
-        # if no kwargs filtered, return all kwargs as default
+        # nothing matched the signature: pass everything through unchanged
         if not filtered_kwargs and not exists_var_keyword:
             filtered_kwargs = kwargs
         if exists_var_keyword:
@@ -738,7 +737,8 @@ class Metric(Module, ABC):
         return CompositionalMetric(torch.add, other, self)
 
     def __rand__(self, other: Any) -> "CompositionalMetric":
-        # swap them since bitwise_and only supports that way and it's commutative
+        # & is commutative, so keep self first (torch.bitwise_and wants the
+        # tensor operand on the left when `other` is a plain int)
         return CompositionalMetric(torch.bitwise_and, self, other)
 
     def __rfloordiv__(self, other: Any) -> "CompositionalMetric":
@@ -839,7 +839,6 @@ class CompositionalMetric(Metric):
             self.metric_b.update(*args, **self.metric_b._filter_kwargs(**kwargs))
 
     def compute(self) -> Any:
-        # also some parsing for kwargs?
         val_a = self.metric_a.compute() if isinstance(self.metric_a, Metric) else self.metric_a
         val_b = self.metric_b.compute() if isinstance(self.metric_b, Metric) else self.metric_b
 

@@ -101,6 +101,13 @@ def _dtype_code(t: Tensor) -> int:
     raise TypeError(f"HIP metric kernels support float32/bfloat16 inputs, got {t.dtype}")
 
 
+def _to_supported(t: Tensor) -> Tensor:
+    """Kernels read fp32/bf16; cast anything else (fp16 under AMP, fp64) to fp32."""
+    if t.is_floating_point() and t.dtype not in (torch.float32, torch.bfloat16):
+        return t.float()
+    return t
+
+
 def mc_stat_logits(
     preds: Tensor, target: Tensor, ignore_index: Optional[int], want_confmat: bool,
     want_argmax: bool = False,
@@ -111,7 +118,7 @@ def mc_stat_logits(
     """
     lib = _lib()
     assert preds.ndim == 2 and target.ndim == 1 and preds.shape[0] == target.shape[0]
-    preds = preds.contiguous()
+    preds = _to_supported(preds).contiguous()
     target = target.contiguous().long()
     B, C = preds.shape
     dev = preds.device
@@ -195,13 +202,15 @@ def binary_stat(
 ) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
     """Fused binary tp/fp/tn/fn in one pass; picks raw vs sigmoid thresholding on-device."""
     lib = _lib()
-    preds = preds.contiguous()
+    preds = _to_supported(preds).contiguous()
     target = target.contiguous().long()
     N = preds.numel()
     dev = preds.device
     out = torch.zeros(2, 4, dtype=torch.long, device=dev)  # {raw,sig} x {tp,fp,tn,fn}
     flag = torch.zeros(1, dtype=torch.int32, device=dev)
-    thr = min(max(threshold, 1e-7), 1 - 1e-7)
+    # no clamp: at thr=0/1 the kernel's logit(thr) is -inf/+inf, which makes the
+    # sigmoid branch always/never positive — exactly the reference comparison
+    thr = float(threshold)
     rc = lib.ma_binary_stat(
         _stream(),
         preds.data_ptr(),
@@ -226,13 +235,13 @@ def multilabel_stat(
     """Fused per-label tp/fp/tn/fn over (N, L)."""
     lib = _lib()
     assert preds.ndim == 2
-    preds = preds.contiguous()
+    preds = _to_supported(preds).contiguous()
     target = target.contiguous().long()
     N, L = preds.shape
     dev = preds.device
     out = torch.zeros(2, L, 4, dtype=torch.long, device=dev)
     flag = torch.zeros(1, dtype=torch.int32, device=dev)
-    thr = min(max(threshold, 1e-7), 1 - 1e-7)
+    thr = float(threshold)
     rc = lib.ma_multilabel_stat(
         _stream(),
         preds.data_ptr(),
@@ -345,7 +354,7 @@ def binary_curve_confmat(
     ``preds`` must already be probabilities in [0,1].
     """
     lib = _lib()
-    preds = preds.contiguous()
+    preds = _to_supported(preds).contiguous()
     target = target.contiguous().long()
     thr = thresholds.contiguous().float()
     T = thr.numel()
@@ -391,7 +400,7 @@ def multiclass_curve_confmat(
     """(T,C,2,2) threshold confmats. mode 0: multiclass one-vs-rest (target (B,));
     mode 1: multilabel (target (B,C)). ``probs`` already normalized (B,C)."""
     lib = _lib()
-    probs = probs.contiguous()
+    probs = _to_supported(probs).contiguous()
     target = target.contiguous().long()
     thr = thresholds.contiguous().float()
     B, C = probs.shape
@@ -517,7 +526,7 @@ def curve_hist_into_confmat(
     uni, t0, inv_step = _uniform_params(thr)
     dev = probs.device
     if confmat_state.ndim == 3:  # binary (T,2,2)
-        preds = probs.contiguous().flatten()
+        preds = _to_supported(probs).contiguous().flatten()
         tgt = target.contiguous().long().flatten()
         hist = _pooled_hist(1, T, dev, owner)
         if norm == "sigmoid":
@@ -544,7 +553,7 @@ def curve_hist_into_confmat(
         _check(rc, "ma_binary_curve_hist")
         outer, transposed = 1, 0
     else:
-        probs = probs.contiguous()
+        probs = _to_supported(probs).contiguous()
         tgt = target.contiguous().long()
         B, C = probs.shape
         hist = _pooled_hist(C, T, dev, owner)
@@ -626,7 +635,7 @@ def mc_stat_into(
     s_fn = scratch[2 * C : 3 * C]
     s_valid_ptr = scratch.data_ptr() + 3 * C * 8
     if preds.ndim == 2 and preds.is_floating_point():
-        preds = preds.contiguous()
+        preds = _to_supported(preds).contiguous()
         target = target.contiguous().long()
         B, C2 = preds.shape
         rc = lib.ma_mc_stat_logits(
@@ -692,7 +701,7 @@ def mc_confmat_into(
     C = num_classes
     assert confmat_state.is_contiguous()
     if preds.ndim == 2 and preds.is_floating_point():
-        preds = preds.contiguous()
+        preds = _to_supported(preds).contiguous()
         target = target.contiguous().long()
         B, C2 = preds.shape
         rc = lib.ma_mc_stat_logits(
@@ -747,7 +756,7 @@ def mc_exact_into(
     C = num_classes
     s_valid_ptr = scratch.data_ptr() + 3 * C * 8
     if preds.ndim == 2 and preds.is_floating_point():
-        preds = preds.contiguous()
+        preds = _to_supported(preds).contiguous()
         target = target.contiguous().long()
         B, C2 = preds.shape
         rc = lib.ma_mc_stat_logits(
@@ -865,7 +874,7 @@ def mc_fused_collection_update(
     scratch, tp, fp, tn, fn = stat
     s_valid_ptr = scratch.data_ptr() + 3 * C * 8
     if preds.ndim == 2 and preds.is_floating_point():
-        preds = preds.contiguous()
+        preds = _to_supported(preds).contiguous()
         target = target.contiguous().long()
         B = preds.shape[0]
         rc = lib.ma_mc_stat_logits(

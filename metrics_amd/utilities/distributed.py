@@ -76,7 +76,7 @@ def gather_all_tensors(result: Tensor, group: Optional[Any] = None) -> List[Tens
     """
     if group is None:
         group = dist.group.WORLD
-    # convert tensors to contiguous format
+    # collectives need contiguous buffers
     result = result.contiguous()
     world_size = dist.get_world_size(group)
     dist.barrier(group=group)
@@ -240,7 +240,10 @@ def sync_states_fast(
         if tag == "list_cat":
             nonempty = [t for t in gath if t.numel() > 0]
             if not nonempty:
-                out[name] = []
+                # every rank was empty: the synced state is a zero-size tensor
+                # (reference behavior — dim_zero_cat of the placeholder gathers),
+                # so downstream compute sees an empty tensor, not an empty list
+                out[name] = torch.cat(gath, dim=0) if gath else []
             else:
                 out[name] = torch.cat(nonempty, dim=0)
             continue
