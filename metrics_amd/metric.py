@@ -135,6 +135,7 @@ class Metric(Module, ABC):
 
         self._is_synced = False
         self._cache: Optional[Dict[str, Union[List[Tensor], Tensor]]] = None
+        self._pending_sync_event: Optional[Any] = None
 
     # ------------------------------------------------------------------ props
     @property
@@ -332,11 +333,27 @@ class Metric(Module, ABC):
                 kinds[name] = "custom"
                 customs[name] = fn
         gather_fn = None if dist_sync_fn in (None, gather_all_tensors) else dist_sync_fn
-        synced = sync_states_fast(
+        synced, done_event = sync_states_fast(
             states, kinds, customs, group=process_group or self.process_group, gather_fn=gather_fn
         )
         for attr, val in synced.items():
             setattr(self, attr, val)
+        # collectives ran on the side HIP stream; whoever reads the synced
+        # states (compute) waits on this lazily — see _wait_pending_sync
+        self._pending_sync_event = done_event
+
+    def _wait_pending_sync(self) -> None:
+        """Fence the current stream behind an in-flight side-stream sync."""
+        ev = getattr(self, "_pending_sync_event", None)
+        if ev is not None:
+            torch.cuda.current_stream().wait_event(ev)
+            # pin the side-stream-allocated result buffers to this stream so
+            # the caching allocator cannot hand them out early
+            for attr in self._defaults:
+                val = getattr(self, attr)
+                if isinstance(val, Tensor) and val.is_cuda:
+                    val.record_stream(torch.cuda.current_stream())
+            self._pending_sync_event = None
 
     def sync(
         self,
@@ -380,6 +397,9 @@ class Metric(Module, ABC):
             setattr(self, attr, val)
         self._is_synced = False
         self._cache = None
+        # a still-in-flight side-stream sync only touches its own buffers,
+        # which we just dropped — nothing left to wait for
+        self._pending_sync_event = None
 
     @contextmanager
     def sync_context(
@@ -451,6 +471,7 @@ class Metric(Module, ABC):
                 should_sync=self._to_sync,
                 should_unsync=self._should_unsync,
             ), tracing.range(f"{self.__class__.__name__}.compute"):
+                self._wait_pending_sync()
                 value = _squeeze_if_scalar(compute(*args, **kwargs))
                 # clone so later in-place ops cannot alter the returned result
                 value = apply_to_collection(value, Tensor, lambda x: x.clone())
@@ -505,6 +526,7 @@ class Metric(Module, ABC):
         # reset internal sync states
         self._cache = None
         self._is_synced = False
+        self._pending_sync_event = None
 
     def clone(self) -> "Metric":
         """Return a deep copy of the metric."""
@@ -525,7 +547,16 @@ class Metric(Module, ABC):
     # walks parameter/buffer/module dicts per call (~1us each, dozens per
     # collection update) — bypass it for names that can never be any of those
     _FAST_ATTRS = frozenset(
-        {"_computed", "_update_count", "_forward_cache", "_to_sync", "_should_unsync", "_enable_grad", "_is_synced"}
+        {
+            "_computed",
+            "_update_count",
+            "_forward_cache",
+            "_to_sync",
+            "_should_unsync",
+            "_enable_grad",
+            "_is_synced",
+            "_pending_sync_event",
+        }
     )
 
     def __setattr__(self, name: str, value: Any) -> None:

@@ -115,6 +115,29 @@ def gather_all_tensors(result: Tensor, group: Optional[Any] = None) -> List[Tens
 
 _ALLREDUCE_OPS = {"sum": "sum", "mean": "sum", "max": "max", "min": "min"}
 
+# One dedicated HIP stream per device for state-sync collectives, so RCCL
+# kernels overlap with the next update()'s compute kernels on the default
+# stream (BASELINE north-star: side-stream overlapped sync).
+_SYNC_STREAMS: Dict[int, "torch.cuda.Stream"] = {}
+
+
+def _sync_stream(device: torch.device) -> "torch.cuda.Stream":
+    s = _SYNC_STREAMS.get(device.index)
+    if s is None:
+        s = torch.cuda.Stream(device=device)
+        _SYNC_STREAMS[device.index] = s
+    return s
+
+
+def _use_side_stream(group: Any) -> bool:
+    """Side-stream sync only makes sense for stream-ordered (RCCL) backends."""
+    if not torch.cuda.is_available():
+        return False
+    try:
+        return "nccl" in str(dist.get_backend(group)).lower()
+    except RuntimeError:
+        return False
+
 
 def _reduction_kind(reduce_fn: Union[str, Callable, None]) -> str:
     """Classify a dist_reduce_fx (already canonicalized to a string or callable)."""
@@ -131,7 +154,8 @@ def sync_states_fast(
     custom_fns: Dict[str, Callable],
     group: Optional[Any] = None,
     gather_fn: Optional[Callable] = None,
-) -> Dict[str, Union[Tensor, List[Tensor]]]:
+    overlap: bool = True,
+) -> Tuple[Dict[str, Union[Tensor, List[Tensor]]], Optional["torch.cuda.Event"]]:
     """Synchronize a metric's states across the process group.
 
     ``kinds[name]`` in {'sum','mean','max','min','cat','none','custom'};
@@ -140,9 +164,17 @@ def sync_states_fast(
     ``gather_fn``: if given (a user-supplied dist_sync_fn), EVERY state takes
     the gather path through it — reference semantics for custom sync fns.
 
-    Returns the dict of *synced* state values (same keys). List states come
-    back as a single concatenated tensor for 'cat' (reference behavior) and as
-    rank-order flattened lists otherwise.
+    Returns ``(synced, done_event)``. List states come back as a single
+    concatenated tensor for 'cat' (reference behavior) and as rank-order
+    flattened lists otherwise.
+
+    Overlap: with a RCCL backend the fused all-reduce path is issued from a
+    dedicated side HIP stream. Only the cheap state *read* (flat-buffer build)
+    is fenced back onto the caller's stream, so the next ``update()`` launches
+    immediately while the collectives run; ``done_event`` (non-None iff the
+    side stream was used) must be waited on by whatever stream *reads* the
+    synced values — :meth:`Metric._wait_pending_sync` does this lazily in
+    ``compute``.
     """
     if group is None:
         group = dist.group.WORLD
@@ -169,19 +201,75 @@ def sync_states_fast(
     works = []
     _gf = gather_fn if gather_fn is not None else gather_all_tensors
 
+    side = None
+    done_event = None
+    if fuse_buckets and overlap and _use_side_stream(group):
+        dev = next(iter(fuse_buckets))[1]
+        if dev.type == "cuda":
+            side = _sync_stream(dev)
+
     # --- fused all-reduce buckets ----------------------------------------
     red_op = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX, "min": dist.ReduceOp.MIN}
     bucket_views: List[Tuple[Tensor, List[Tuple[str, Tensor]]]] = []
-    for (dt, _dev, op), entries in fuse_buckets.items():
-        if len(entries) == 1:
-            name, val = entries[0]
-            buf = val.contiguous().to(dt) if val.dtype != dt else val.contiguous().clone()
-            works.append(dist.all_reduce(buf, op=red_op[op], group=group, async_op=True))
-            bucket_views.append((buf, entries))
-        else:
-            flat = torch.cat([v.reshape(-1).to(dt) for _, v in entries])
-            works.append(dist.all_reduce(flat, op=red_op[op], group=group, async_op=True))
-            bucket_views.append((flat, entries))
+
+    def _issue_allreduces() -> None:
+        for (dt, _dev, op), entries in fuse_buckets.items():
+            if len(entries) == 1:
+                name, val = entries[0]
+                buf = val.contiguous().to(dt) if val.dtype != dt else val.contiguous().clone()
+                works.append(dist.all_reduce(buf, op=red_op[op], group=group, async_op=True))
+                bucket_views.append((buf, entries))
+            else:
+                flat = torch.cat([v.reshape(-1).to(dt) for _, v in entries])
+                works.append(dist.all_reduce(flat, op=red_op[op], group=group, async_op=True))
+                bucket_views.append((flat, entries))
+
+    def _finish_allreduces() -> None:
+        for w in works:
+            if w is not None:
+                w.wait()
+        for buf, entries in bucket_views:
+            if len(entries) == 1:
+                name, val = entries[0]
+                res = buf.view(val.shape)
+                if kinds[name] == "mean":
+                    res = res / world_size
+                out[name] = res.to(val.dtype) if res.dtype != val.dtype and kinds[name] != "mean" else res
+            else:
+                offset = 0
+                for name, val in entries:
+                    n = val.numel()
+                    res = buf[offset : offset + n].view(val.shape)
+                    offset += n
+                    if kinds[name] == "mean":
+                        res = res / world_size
+                    elif res.dtype != val.dtype:
+                        res = res.to(val.dtype)
+                    out[name] = res
+
+    if side is not None:
+        main_stream = torch.cuda.current_stream()
+        produced = torch.cuda.Event()
+        produced.record(main_stream)
+        with torch.cuda.stream(side):
+            # order after the kernels that produced the current state values
+            side.wait_event(produced)
+            _issue_allreduces()
+            # the states have now been read into the flat buffers
+            read_done = torch.cuda.Event()
+            read_done.record(side)
+            _finish_allreduces()  # stream-level waits + scatter-back views
+            done_event = torch.cuda.Event()
+            done_event.record(side)
+        # later writers of the original state tensors (the next update) must
+        # not overwrite them before the side stream has copied them out —
+        # this only costs the D2D copy time, not the collective time
+        main_stream.wait_event(read_done)
+
+    if side is None:
+        # inline path: issue the async all-reduces first so their launch
+        # latencies overlap with the gather collectives below
+        _issue_allreduces()
 
     # --- gather path (cat / none / custom / autograd) ---------------------
     gathered: Dict[str, Any] = {}
@@ -204,28 +292,8 @@ def sync_states_fast(
         else:
             gathered[name] = ("tensor", _gf(val, group))
 
-    # --- wait for the fused all-reduces and scatter back -------------------
-    for w in works:
-        if w is not None:
-            w.wait()
-    for buf, entries in bucket_views:
-        if len(entries) == 1:
-            name, val = entries[0]
-            res = buf.view(val.shape)
-            if kinds[name] == "mean":
-                res = res / world_size
-            out[name] = res.to(val.dtype) if res.dtype != val.dtype and kinds[name] != "mean" else res
-        else:
-            offset = 0
-            for name, val in entries:
-                n = val.numel()
-                res = buf[offset : offset + n].view(val.shape)
-                offset += n
-                if kinds[name] == "mean":
-                    res = res / world_size
-                elif res.dtype != val.dtype:
-                    res = res.to(val.dtype)
-                out[name] = res
+    if side is None:
+        _finish_allreduces()
 
     # --- finalize gather states: apply the reference's post-gather reduction
     for name, (tag, gath) in gathered.items():
@@ -268,7 +336,7 @@ def sync_states_fast(
         else:  # custom callable
             stacked = torch.stack(gath, dim=0) if shapes_equal else gath
             out[name] = custom_fns[name](stacked)
-    return out
+    return out, done_event
 
 
 def _any_device(states: Dict[str, Any]) -> torch.device:

@@ -104,16 +104,22 @@ def run_functional_metric_test(
 
 
 # ---------------------------------------------------------------- distributed
-def _dist_worker(rank: int, world_size: int, port: int, fn: Callable, args: tuple) -> None:
+def _dist_worker(rank: int, world_size: int, port: int, fn: Callable, args: tuple, backend: str = "gloo") -> None:
     # Forked children inherit the parent's (possibly mid-operation) OpenMP pool
     # state; entering a parallel region then deadlocks on a stale futex. Keep
     # the child single-threaded so no parallel region is ever entered.
     torch.set_num_threads(1)
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
-    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    if backend == "nccl":
+        # single-box RCCL proof: every rank shares cuda:0
+        torch.cuda.set_device(0)
+    dist.init_process_group(backend, rank=rank, world_size=world_size)
     try:
         fn(rank, world_size, *args)
+        if backend == "nccl":
+            torch.cuda.synchronize()
+            dist.barrier()
     finally:
         dist.destroy_process_group()
 
@@ -128,11 +134,15 @@ def _free_port() -> int:
         return s.getsockname()[1]
 
 
-def run_distributed(fn: Callable, world_size: int = 2, args: tuple = ()) -> None:
-    """Run ``fn(rank, world_size, *args)`` in ``world_size`` gloo processes on localhost."""
+def run_distributed(fn: Callable, world_size: int = 2, args: tuple = (), backend: str = "gloo") -> None:
+    """Run ``fn(rank, world_size, *args)`` in ``world_size`` processes on localhost.
+
+    backend "gloo" = CPU cluster emulation (fork). backend "nccl" = real RCCL,
+    all ranks sharing cuda:0 (spawn — fork is unsafe after HIP init).
+    """
     port = _free_port()
     mp.start_processes(
-        partial(_dist_worker, world_size=world_size, port=port, fn=fn, args=args),
+        partial(_dist_worker, world_size=world_size, port=port, fn=fn, args=args, backend=backend),
         nprocs=world_size,
-        start_method="spawn" if os.environ.get("MA_DIST_SPAWN") else "fork",
+        start_method="spawn" if (backend == "nccl" or os.environ.get("MA_DIST_SPAWN")) else "fork",
     )
