@@ -111,13 +111,14 @@ def _dist_worker(rank: int, world_size: int, port: int, fn: Callable, args: tupl
     torch.set_num_threads(1)
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
-    if backend == "nccl":
-        # single-box RCCL proof: every rank shares cuda:0
+    on_gpu = backend in ("nccl", "gloo_cuda")
+    if on_gpu:
+        # single-box proof: every rank shares cuda:0
         torch.cuda.set_device(0)
-    dist.init_process_group(backend, rank=rank, world_size=world_size)
+    dist.init_process_group("gloo" if backend == "gloo_cuda" else backend, rank=rank, world_size=world_size)
     try:
         fn(rank, world_size, *args)
-        if backend == "nccl":
+        if on_gpu:
             torch.cuda.synchronize()
             dist.barrier()
     finally:
@@ -137,12 +138,15 @@ def _free_port() -> int:
 def run_distributed(fn: Callable, world_size: int = 2, args: tuple = (), backend: str = "gloo") -> None:
     """Run ``fn(rank, world_size, *args)`` in ``world_size`` processes on localhost.
 
-    backend "gloo" = CPU cluster emulation (fork). backend "nccl" = real RCCL,
-    all ranks sharing cuda:0 (spawn — fork is unsafe after HIP init).
+    backend "gloo" = CPU cluster emulation (fork). backend "nccl" = real RCCL.
+    backend "gloo_cuda" = gloo group with ranks pinned to cuda:0 (multi-rank
+    reductions of CUDA states on a 1-GPU box; gloo stages via host). GPU
+    backends use spawn — fork is unsafe after HIP init.
     """
     port = _free_port()
+    on_gpu = backend in ("nccl", "gloo_cuda")
     mp.start_processes(
         partial(_dist_worker, world_size=world_size, port=port, fn=fn, args=args, backend=backend),
         nprocs=world_size,
-        start_method="spawn" if (backend == "nccl" or os.environ.get("MA_DIST_SPAWN")) else "fork",
+        start_method="spawn" if (on_gpu or os.environ.get("MA_DIST_SPAWN")) else "fork",
     )

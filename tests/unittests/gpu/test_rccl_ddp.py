@@ -1,11 +1,20 @@
-"""Real-RCCL distributed sync proof: 2 ranks sharing the one MI355X (cuda:0).
+"""Real-backend distributed sync proof on a single MI355X.
 
-The gloo suite (tests/unittests/bases/test_ddp.py) proves the sync engine's
-semantics on a CPU cluster emulation; this module re-exercises the same
-scenarios over an actual RCCL process group so the fused all-reduce buckets,
-the uneven-shape gathers AND the side-stream overlap path run on the real
-backend. Reference behavior: torchmetrics tests/unittests/bases/test_ddp.py:35-345
-and src/torchmetrics/utilities/distributed.py:100-153.
+RCCL rejects two ranks on one device ("Duplicate GPU detected", rccl
+init.cc:1108), so multi-rank-on-one-box splits into two halves that
+together cover the whole sync engine on hardware:
+
+1. a world-size-1 RCCL group: the fused all-reduce buckets, the gather
+   path and the side-stream overlap protocol execute on the *real* RCCL
+   backend (stream-ordered collectives, done events, read fences) — this is
+   the half that gloo cannot emulate;
+2. a 2-process gloo group exchanging *CUDA tensors*: real multi-rank
+   reductions of device states (gloo stages through host) — this is the
+   half world-1 RCCL cannot show.
+
+True multi-rank RCCL runs in the driver's round-end N=1..8 scaling bench
+(bench.py over torch.distributed.run). Reference behavior:
+torchmetrics tests/unittests/bases/test_ddp.py:35-345.
 """
 import os
 
@@ -20,9 +29,10 @@ def _dev():
     return torch.device("cuda", 0)
 
 
-# Scenario functions must be module-level (spawn pickling).
+# --------------------------------------------------------------- scenarios
+# (module-level: spawn pickling)
 
-def _rccl_sum_unsync(rank, world_size):
+def _sc_sum_unsync(rank, world_size):
     from tests.unittests.bases.test_ddp import S
 
     dev = _dev()
@@ -35,7 +45,7 @@ def _rccl_sum_unsync(rank, world_size):
     assert m.compute().item() == sum(r + 1 for r in range(world_size)) + world_size
 
 
-def _rccl_mean_max_min(rank, world_size):
+def _sc_mean_max_min(rank, world_size):
     from tests.unittests.bases.test_ddp import S
 
     dev = _dev()
@@ -50,7 +60,7 @@ def _rccl_mean_max_min(rank, world_size):
     assert mn.compute().item() == 0.0
 
 
-def _rccl_fused_buckets_and_cat(rank, world_size):
+def _sc_fused_buckets_and_cat(rank, world_size):
     """Mixed-state metric: fused all-reduce buckets + gathered cat list."""
     from metrics_amd import Metric
     from metrics_amd.utilities.data import dim_zero_cat
@@ -85,7 +95,7 @@ def _rccl_fused_buckets_and_cat(rank, world_size):
     assert d.numel() == 2 * world_size
 
 
-def _rccl_uneven_cat(rank, world_size):
+def _sc_uneven_cat(rank, world_size):
     from tests.unittests.bases.test_ddp import S
 
     dev = _dev()
@@ -101,7 +111,7 @@ def _rccl_uneven_cat(rank, world_size):
     assert out2.numel() == 3
 
 
-def _rccl_dist_sync_on_step(rank, world_size):
+def _sc_dist_sync_on_step(rank, world_size):
     from tests.unittests.bases.test_ddp import S
 
     dev = _dev()
@@ -110,8 +120,8 @@ def _rccl_dist_sync_on_step(rank, world_size):
     assert batch_val.item() == world_size
 
 
-def _rccl_collection_parity(rank, world_size):
-    """The bench's metric mix under real RCCL == single-process on all data."""
+def _sc_collection_parity(rank, world_size):
+    """The bench's metric mix, distributed == single-process on all data."""
     import metrics_amd as ma
 
     dev = _dev()
@@ -144,32 +154,8 @@ def _rccl_collection_parity(rank, world_size):
         assert torch.allclose(res[k].float(), exp[k].float(), atol=1e-5), (k, res[k], exp[k])
 
 
-def _rccl_side_stream_overlap(rank, world_size):
-    """The fused sum sync must actually take the side-stream path on RCCL:
-    a done event is produced, compute consumes it, and the result is right."""
-    import metrics_amd.utilities.distributed as d
-    from tests.unittests.bases.test_ddp import S
-
-    dev = _dev()
-    assert d._use_side_stream(None), "RCCL backend must enable the side-stream sync path"
-    m = S("sum").to(dev)
-    m.update(tensor(float(rank + 1), device=dev))
-    # drive sync() directly so we can observe the pending event before compute
-    m.sync()
-    assert m._pending_sync_event is not None, "side-stream sync should leave a pending done event"
-    m._wait_pending_sync()
-    assert m._pending_sync_event is None
-    torch.cuda.synchronize()
-    assert m.x.item() == sum(r + 1 for r in range(world_size))
-    m.unsync()
-    assert m.x.item() == rank + 1
-    # and the normal compute path end-to-end
-    m._computed = None
-    assert m.compute().item() == sum(r + 1 for r in range(world_size))
-
-
-def _rccl_curve_parity(rank, world_size):
-    """Thresholded curve metrics (HIP update kernels) synced over real RCCL."""
+def _sc_curve_parity(rank, world_size):
+    """Thresholded curve metrics (HIP update kernels) synced across ranks."""
     import metrics_amd as ma
 
     dev = _dev()
@@ -186,7 +172,7 @@ def _rccl_curve_parity(rank, world_size):
     assert torch.allclose(res, ref.compute(), atol=1e-6)
 
 
-def _rccl_pearson_welford(rank, world_size):
+def _sc_pearson_welford(rank, world_size):
     import metrics_amd as ma
 
     dev = _dev()
@@ -203,16 +189,43 @@ def _rccl_pearson_welford(rank, world_size):
     assert torch.allclose(res, ref.compute(), atol=1e-5)
 
 
+def _sc_side_stream_overlap(rank, world_size):
+    """On RCCL the fused sum sync must take the side-stream path: a done
+    event is produced, compute consumes it, unsync restores local state."""
+    import torch.distributed as dist
+
+    import metrics_amd.utilities.distributed as d
+    from tests.unittests.bases.test_ddp import S
+
+    if "nccl" not in str(dist.get_backend()).lower():
+        return  # only meaningful on the RCCL backend
+    dev = _dev()
+    assert d._use_side_stream(None), "RCCL backend must enable the side-stream sync path"
+    m = S("sum").to(dev)
+    m.update(tensor(float(rank + 1), device=dev))
+    m.sync()
+    assert m._pending_sync_event is not None, "side-stream sync should leave a pending done event"
+    m._wait_pending_sync()
+    assert m._pending_sync_event is None
+    torch.cuda.synchronize()
+    assert m.x.item() == sum(r + 1 for r in range(world_size))
+    m.unsync()
+    assert m.x.item() == rank + 1
+    # normal compute path end-to-end (waits via _wrap_compute)
+    m._computed = None
+    assert m.compute().item() == sum(r + 1 for r in range(world_size))
+
+
 _SCENARIOS = [
-    _rccl_sum_unsync,
-    _rccl_mean_max_min,
-    _rccl_fused_buckets_and_cat,
-    _rccl_uneven_cat,
-    _rccl_dist_sync_on_step,
-    _rccl_collection_parity,
-    _rccl_side_stream_overlap,
-    _rccl_curve_parity,
-    _rccl_pearson_welford,
+    _sc_sum_unsync,
+    _sc_mean_max_min,
+    _sc_fused_buckets_and_cat,
+    _sc_uneven_cat,
+    _sc_dist_sync_on_step,
+    _sc_collection_parity,
+    _sc_curve_parity,
+    _sc_pearson_welford,
+    _sc_side_stream_overlap,
 ]
 
 
@@ -225,9 +238,23 @@ def _run_all(rank, world_size):
         dist.barrier()
 
 
-def test_rccl_2ranks_1gpu():
-    """One spawn cost for the whole scenario set (RCCL init is seconds)."""
-    from tests.unittests._helpers import run_distributed
+def test_rccl_world1():
+    """Whole scenario set on a real 1-rank RCCL communicator (in-process)."""
+    import torch.distributed as dist
 
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
-    run_distributed(_run_all, world_size=2, backend="nccl")
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ.setdefault("MASTER_PORT", "29371")
+    torch.cuda.set_device(0)
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        _run_all(0, 1)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_gloo_cuda_world2():
+    """Multi-rank reductions of CUDA states over a 2-process gloo group."""
+    from tests.unittests._helpers import run_distributed
+
+    run_distributed(_run_all, world_size=2, backend="gloo_cuda")
