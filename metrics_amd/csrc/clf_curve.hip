@@ -1,0 +1,159 @@
+// K2 — exact classification-curve core for gfx950 (CDNA4).
+//
+// Implements the reference's `_binary_clf_curve`
+// (torchmetrics functional/classification/precision_recall_curve.py:30-82):
+// descending sort of the scores, cumulative tp/fp at each DISTINCT score.
+// MI355X-native formulation: rocPRIM device radix sort (stable, fp32 keys)
+// + double-precision device scans + a fused gather/flag kernel and a fused
+// compaction kernel, all launched on the caller's torch HIP stream.
+// Counts are integers (< 2^53), so the fp64 scan is bit-exact for the
+// unweighted path and at least as accurate as the reference's fp32 cumsum
+// for the weighted one.
+//
+// Exposed C API (ctypes, see ops/_hip.py):
+//   ma_clf_curve_scratch_bytes(N, weighted, &bytes)
+//   ma_binary_clf_curve(stream, preds, target, weight|0, N, pos_label,
+//                       scratch, scratch_bytes, out_fps, out_tps,
+//                       out_thresh, out_count)
+
+#include <cstring>  // memset, needed by rocprim's texture_cache_iterator header
+
+#include <hip/hip_runtime.h>
+#include <rocprim/rocprim.hpp>
+
+#include <cstdint>
+
+using ll = long long;
+
+#define OK(x)                                                                  \
+    do {                                                                       \
+        hipError_t err_ = (x);                                                 \
+        if (err_ != hipSuccess) return (int)err_;                              \
+    } while (0)
+
+static constexpr size_t ALIGN = 256;
+
+static inline size_t align_up(size_t x) { return (x + ALIGN - 1) & ~(ALIGN - 1); }
+
+// ---------------------------------------------------------------- kernels
+
+__global__ void __launch_bounds__(256) k_iota(int* idx, ll n) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) idx[i] = (int)i;
+}
+
+// Gather target/weight through the sort permutation and mark distinct-score
+// boundaries in one pass.
+__global__ void __launch_bounds__(256) k_gather_flags(
+    const float* __restrict__ keys_sorted, const int* __restrict__ idx_sorted,
+    const ll* __restrict__ target, const float* __restrict__ weight, ll n,
+    ll pos_label, double* __restrict__ tvals, double* __restrict__ fvals,
+    int* __restrict__ flags) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    const int src = idx_sorted[i];
+    const double t = (target[src] == pos_label) ? 1.0 : 0.0;
+    const double w = weight ? (double)weight[src] : 1.0;
+    tvals[i] = t * w;
+    if (fvals) fvals[i] = (1.0 - t) * w;
+    flags[i] = (i == n - 1) || (keys_sorted[i] != keys_sorted[i + 1]);
+}
+
+// Compact (tps, fps, thresholds) at the flagged positions. pos_scan is the
+// INCLUSIVE scan of flags, so out position = pos_scan[i] - 1.
+__global__ void __launch_bounds__(256) k_compact(
+    const float* __restrict__ keys_sorted, const int* __restrict__ flags,
+    const int* __restrict__ pos_scan, const double* __restrict__ tps_full,
+    const double* __restrict__ fps_full /* null => unweighted */, ll n,
+    float* __restrict__ out_fps, float* __restrict__ out_tps,
+    float* __restrict__ out_thresh, ll* __restrict__ out_count) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    if (flags[i]) {
+        const int o = pos_scan[i] - 1;
+        const double tp = tps_full[i];
+        out_tps[o] = (float)tp;
+        out_fps[o] = fps_full ? (float)fps_full[i] : (float)((double)(i + 1) - tp);
+        out_thresh[o] = keys_sorted[i];
+    }
+    if (i == n - 1) *out_count = (ll)pos_scan[n - 1];
+}
+
+// ----------------------------------------------------------------- C API
+
+extern "C" int ma_clf_curve_scratch_bytes(ll n, int weighted, unsigned long long* out_bytes) {
+    size_t sort_tmp = 0, scan_tmp = 0, iscan_tmp = 0;
+    hipError_t e;
+    e = rocprim::radix_sort_pairs_desc((void*)nullptr, sort_tmp, (const float*)nullptr,
+                                       (float*)nullptr, (const int*)nullptr, (int*)nullptr,
+                                       (size_t)n);
+    if (e != hipSuccess) return (int)e;
+    e = rocprim::inclusive_scan((void*)nullptr, scan_tmp, (const double*)nullptr,
+                                (double*)nullptr, (size_t)n, rocprim::plus<double>());
+    if (e != hipSuccess) return (int)e;
+    e = rocprim::inclusive_scan((void*)nullptr, iscan_tmp, (const int*)nullptr,
+                                (int*)nullptr, (size_t)n, rocprim::plus<int>());
+    if (e != hipSuccess) return (int)e;
+    size_t tmp = sort_tmp;
+    if (scan_tmp > tmp) tmp = scan_tmp;
+    if (iscan_tmp > tmp) tmp = iscan_tmp;
+
+    size_t total = 0;
+    total += align_up((size_t)n * sizeof(float));   // keys_sorted
+    total += align_up((size_t)n * sizeof(int));     // idx (iota)
+    total += align_up((size_t)n * sizeof(int));     // idx_sorted
+    total += align_up((size_t)n * sizeof(double));  // tvals / tps_full (in-place scan)
+    if (weighted) total += align_up((size_t)n * sizeof(double));  // fvals / fps_full
+    total += align_up((size_t)n * sizeof(int));     // flags
+    total += align_up((size_t)n * sizeof(int));     // pos_scan
+    total += align_up(tmp);                          // rocprim temp
+    *out_bytes = (unsigned long long)total;
+    return 0;
+}
+
+extern "C" int ma_binary_clf_curve(
+    uint64_t stream_u, uint64_t preds_u, uint64_t target_u, uint64_t weight_u, ll n,
+    ll pos_label, uint64_t scratch_u, unsigned long long scratch_bytes,
+    uint64_t out_fps_u, uint64_t out_tps_u, uint64_t out_thresh_u, uint64_t out_count_u) {
+    hipStream_t stream = (hipStream_t)stream_u;
+    const float* preds = (const float*)preds_u;
+    const ll* target = (const ll*)target_u;
+    const float* weight = (const float*)weight_u;  // may be null
+
+    char* p = (char*)scratch_u;
+    float* keys_sorted = (float*)p;          p += align_up((size_t)n * sizeof(float));
+    int* idx = (int*)p;                      p += align_up((size_t)n * sizeof(int));
+    int* idx_sorted = (int*)p;               p += align_up((size_t)n * sizeof(int));
+    double* tvals = (double*)p;              p += align_up((size_t)n * sizeof(double));
+    double* fvals = nullptr;
+    if (weight) { fvals = (double*)p;        p += align_up((size_t)n * sizeof(double)); }
+    int* flags = (int*)p;                    p += align_up((size_t)n * sizeof(int));
+    int* pos_scan = (int*)p;                 p += align_up((size_t)n * sizeof(int));
+    void* rp_tmp = (void*)p;
+    size_t rp_bytes = (size_t)((char*)scratch_u + scratch_bytes - p);
+
+    const int B = 256;
+    const ll grid = (n + B - 1) / B;
+
+    hipLaunchKernelGGL(k_iota, dim3(grid), dim3(B), 0, stream, idx, n);
+    size_t tmp_bytes = rp_bytes;
+    OK(rocprim::radix_sort_pairs_desc(rp_tmp, tmp_bytes, preds, keys_sorted, idx, idx_sorted,
+                                      (size_t)n, 0, 32, stream));
+    hipLaunchKernelGGL(k_gather_flags, dim3(grid), dim3(B), 0, stream, keys_sorted, idx_sorted,
+                       target, weight, n, pos_label, tvals, fvals, flags);
+    tmp_bytes = rp_bytes;
+    OK(rocprim::inclusive_scan(rp_tmp, tmp_bytes, tvals, tvals, (size_t)n,
+                               rocprim::plus<double>(), stream));
+    if (fvals) {
+        tmp_bytes = rp_bytes;
+        OK(rocprim::inclusive_scan(rp_tmp, tmp_bytes, fvals, fvals, (size_t)n,
+                                   rocprim::plus<double>(), stream));
+    }
+    tmp_bytes = rp_bytes;
+    OK(rocprim::inclusive_scan(rp_tmp, tmp_bytes, flags, pos_scan, (size_t)n,
+                               rocprim::plus<int>(), stream));
+    hipLaunchKernelGGL(k_compact, dim3(grid), dim3(B), 0, stream, keys_sorted, flags, pos_scan,
+                       tvals, fvals, n, (float*)out_fps_u, (float*)out_tps_u,
+                       (float*)out_thresh_u, (ll*)out_count_u);
+    return (int)hipGetLastError();
+}

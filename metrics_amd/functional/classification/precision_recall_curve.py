@@ -38,6 +38,14 @@ def _binary_clf_curve(
         # squeeze away a singleton class axis
         if preds.ndim > target.ndim:
             preds = preds[:, 0]
+
+        if preds.is_cuda and preds.numel() > 0:
+            # K2 HIP path: rocPRIM radix sort + fused fp64 scans (csrc/clf_curve.hip)
+            fps, tps, thr = ops.hip_binary_clf_curve(
+                preds, target, sample_weights if isinstance(sample_weights, Tensor) else None, pos_label
+            )
+            return fps, tps, thr.to(preds.dtype)
+
         desc_score_indices = torch.argsort(preds, descending=True)
 
         preds = preds[desc_score_indices]

@@ -24,6 +24,13 @@ def hip_bincount(x: Tensor, minlength: int) -> Tensor:
     return _hip.bincount(x, minlength)
 
 
+def hip_binary_clf_curve(
+    preds: Tensor, target: Tensor, weights: Optional[Tensor] = None, pos_label: int = 1
+) -> Tuple[Tensor, Tensor, Tensor]:
+    """Exact clf-curve core (fps, tps, thresholds) — GPU only (K2 kernel)."""
+    return _hip.binary_clf_curve(preds, target, weights, pos_label)
+
+
 def box_iou_pairwise(boxes1: Tensor, boxes2: Tensor, variant: str = "iou") -> Tensor:
     """All-pairs box IoU with optional GIoU/DIoU/CIoU epilogues (GPU fused, CPU torch)."""
     if boxes1.is_cuda:

@@ -60,6 +60,8 @@ _SIGNATURES = {
     "ma_linear_stat_compute": [_U64, _U64, _U64, _U64, _U64, _LL, _F, _F, _F, _F, _F, _F, _F, _F, _I, _I, _F, _F, _F, _U64],
     "ma_err_reduce": [_U64, _U64, _U64, _I, _LL, _I, _D, _U64, _I, _I, _U64],
     "ma_box_iou": [_U64, _U64, _LL, _U64, _LL, _I, _U64],
+    "ma_clf_curve_scratch_bytes": [_LL, _I, ctypes.POINTER(ctypes.c_ulonglong)],
+    "ma_binary_clf_curve": [_U64, _U64, _U64, _U64, _LL, _LL, _U64, ctypes.c_ulonglong, _U64, _U64, _U64, _U64],
 }
 
 
@@ -447,6 +449,62 @@ def multiclass_curve_confmat(
     )
     _check(rc, "ma_curve_suffix")
     return confmat
+
+
+_CLF_SCRATCH_CACHE: dict = {}
+
+
+def binary_clf_curve(
+    preds: Tensor, target: Tensor, weights: Optional[Tensor] = None, pos_label: int = 1
+) -> Tuple[Tensor, Tensor, Tensor]:
+    """Exact (fps, tps, thresholds) at distinct descending scores.
+
+    rocPRIM radix sort (stable, fp32 keys) + fp64 device scans + fused
+    gather/flag and compaction kernels — the K2 path for thresholds=None
+    ROC/PR/AUROC. One D2H sync for the data-dependent output length (the
+    reference's torch path syncs there too).
+    """
+    lib = _lib()
+    preds = _to_supported(preds)
+    if preds.dtype != torch.float32:
+        preds = preds.float()
+    preds = preds.contiguous().flatten()
+    target = target.contiguous().long().flatten()
+    w = weights.contiguous().float().flatten() if weights is not None else None
+    N = preds.numel()
+    dev = preds.device
+    key = (N, w is not None)
+    nbytes = _CLF_SCRATCH_CACHE.get(key)
+    if nbytes is None:
+        out_b = ctypes.c_ulonglong(0)
+        rc = lib.ma_clf_curve_scratch_bytes(N, 1 if w is not None else 0, ctypes.byref(out_b))
+        _check(rc, "ma_clf_curve_scratch_bytes")
+        nbytes = out_b.value
+        if len(_CLF_SCRATCH_CACHE) > 256:
+            _CLF_SCRATCH_CACHE.clear()
+        _CLF_SCRATCH_CACHE[key] = nbytes
+    scratch = torch.empty(int(nbytes), dtype=torch.uint8, device=dev)
+    out_fps = torch.empty(N, dtype=torch.float32, device=dev)
+    out_tps = torch.empty(N, dtype=torch.float32, device=dev)
+    out_thr = torch.empty(N, dtype=torch.float32, device=dev)
+    out_cnt = torch.zeros(1, dtype=torch.long, device=dev)
+    rc = lib.ma_binary_clf_curve(
+        _stream(),
+        preds.data_ptr(),
+        target.data_ptr(),
+        w.data_ptr() if w is not None else 0,
+        N,
+        pos_label,
+        scratch.data_ptr(),
+        nbytes,
+        out_fps.data_ptr(),
+        out_tps.data_ptr(),
+        out_thr.data_ptr(),
+        out_cnt.data_ptr(),
+    )
+    _check(rc, "ma_binary_clf_curve")
+    k = int(out_cnt.item())
+    return out_fps[:k], out_tps[:k], out_thr[:k]
 
 
 _ERR_OPS = {"sq_err": (0, 1), "abs_err": (1, 1), "ape": (2, 1), "sq_log_err": (3, 1), "moments": (4, 6), "logcosh": (5, 1)}

@@ -789,3 +789,92 @@ def test_smallc_stat_kernel_gpu():
         cl = ma.MulticlassAccuracy(num_classes=C, average="micro")
         cl.update(am, target.cpu())
         assert torch.allclose(gl.compute().cpu(), cl.compute(), atol=1e-6), C
+
+
+# ----------------------------------------------------------- K2 exact curve
+def _torch_clf_curve(preds, target, weights=None, pos_label=1):
+    """The (CPU-identical) torch formulation, used as the oracle for the
+    rocPRIM sort+scan kernel path."""
+    desc = torch.argsort(preds, descending=True)
+    preds_s, target_s = preds[desc], target[desc]
+    weight = weights[desc] if weights is not None else 1.0
+    distinct = torch.where(preds_s[1:] - preds_s[:-1])[0]
+    thr_idx = torch.nn.functional.pad(distinct, [0, 1], value=target_s.size(0) - 1)
+    t = (target_s == pos_label).long()
+    tps = torch.cumsum(t * weight, dim=0)[thr_idx]
+    if weights is not None:
+        fps = torch.cumsum((1 - t) * weight, dim=0)[thr_idx]
+    else:
+        fps = 1 + thr_idx - tps
+    return fps, tps, preds_s[thr_idx]
+
+
+@pytest.mark.parametrize("n", [1, 37, 5000, 200_000])
+def test_hip_clf_curve_unweighted(n):
+    torch.manual_seed(3)
+    # heavy ties: scores quantized to 2 decimals
+    preds = (torch.rand(n, device="cuda") * 100).round() / 100
+    target = torch.randint(0, 2, (n,), device="cuda")
+    fps, tps, thr = ops.hip_binary_clf_curve(preds, target)
+    efps, etps, ethr = _torch_clf_curve(preds, target)
+    assert torch.equal(thr, ethr.float())
+    assert torch.equal(tps, etps.float())
+    assert torch.equal(fps, efps.float())
+
+
+def test_hip_clf_curve_weighted():
+    torch.manual_seed(4)
+    n = 10_000
+    preds = (torch.rand(n, device="cuda") * 50).round() / 50
+    target = torch.randint(0, 2, (n,), device="cuda")
+    w = torch.rand(n, device="cuda")
+    fps, tps, thr = ops.hip_binary_clf_curve(preds, target, w)
+    efps, etps, ethr = _torch_clf_curve(preds, target, w)
+    assert torch.equal(thr, ethr)
+    assert torch.allclose(tps, etps, atol=1e-3)
+    assert torch.allclose(fps, efps, atol=1e-3)
+
+
+def test_hip_clf_curve_bf16_and_poslabel():
+    torch.manual_seed(5)
+    n = 4096
+    preds = torch.rand(n, device="cuda", dtype=torch.bfloat16)
+    target = torch.randint(0, 2, (n,), device="cuda")
+    fps, tps, thr = ops.hip_binary_clf_curve(preds, target, pos_label=0)
+    efps, etps, ethr = _torch_clf_curve(preds.float(), target, pos_label=0)
+    assert torch.equal(tps, etps.float()) and torch.equal(fps, efps.float())
+    assert torch.equal(thr, ethr)
+
+
+def test_exact_roc_auroc_end_to_end_gpu():
+    """thresholds=None BinaryROC/AUROC/PrecisionRecallCurve run the HIP K2
+    path on GPU and must match the CPU (torch sort) result."""
+    torch.manual_seed(6)
+    n = 50_000
+    preds = torch.rand(n)
+    target = torch.randint(0, 2, (n,))
+    for cls in (ma.BinaryROC, ma.BinaryPrecisionRecallCurve):
+        mg = cls(thresholds=None).to("cuda")
+        mg.update(preds.cuda(), target.cuda())
+        res_g = mg.compute()
+        mc = cls(thresholds=None)
+        mc.update(preds, target)
+        res_c = mc.compute()
+        for a, b in zip(res_g, res_c):
+            assert torch.allclose(a.cpu(), b, atol=1e-6), cls.__name__
+    ag = ma.BinaryAUROC(thresholds=None).to("cuda")
+    ag.update(preds.cuda(), target.cuda())
+    ac = ma.BinaryAUROC(thresholds=None)
+    ac.update(preds, target)
+    assert torch.allclose(ag.compute().cpu(), ac.compute(), atol=1e-6)
+
+
+def test_exact_multiclass_curve_gpu():
+    torch.manual_seed(7)
+    preds = torch.randn(2000, 7).softmax(-1)
+    target = torch.randint(0, 7, (2000,))
+    mg = ma.MulticlassAveragePrecision(num_classes=7, thresholds=None, average="macro").to("cuda")
+    mg.update(preds.cuda(), target.cuda())
+    mc = ma.MulticlassAveragePrecision(num_classes=7, thresholds=None, average="macro")
+    mc.update(preds, target)
+    assert torch.allclose(mg.compute().cpu(), mc.compute(), atol=1e-6)
