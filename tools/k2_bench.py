@@ -6,7 +6,11 @@ Compares, on one MI355X:
 Writes one JSON line per timing.
 """
 import json
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
