@@ -618,16 +618,19 @@ class MeanAveragePrecision(Metric):
                 rc = tps / npig_safe[:, None, None]
                 pr = tps / (tps + fps + eps)
                 pr_env = np.maximum.accumulate(pr[..., ::-1], axis=-1)[..., ::-1]
-                # one searchsorted for ALL (A,T) rows: offset each row by 2*row
-                # so the (ascending, in [0,1]) rows concatenate globally ascending
+                # per-(area,iou) searchsorted in the ORIGINAL domain: an
+                # offset-batched variant quantizes away sub-ulp rc-vs-recThr
+                # differences at exact boundary hits (rc == 0.6 vs linspace
+                # 0.6), flipping the side-left result — 40 tiny calls are free
                 n_rows = A * T
-                row_off = 2.0 * np.arange(n_rows, dtype=np.float64)
-                big = (rc.reshape(n_rows, nc) + row_off[:, None]).ravel()
-                queries = (rec_thrs[None, :] + row_off[:, None]).ravel()
-                inds = np.searchsorted(big, queries, side="left")
-                local = inds - np.repeat(np.arange(n_rows) * nc, R)
+                rc2 = rc.reshape(n_rows, nc)
+                inds_rows = np.empty((n_rows, R), dtype=np.int64)
+                for row in range(n_rows):
+                    inds_rows[row] = np.searchsorted(rc2[row], rec_thrs, side="left")
+                local = inds_rows.ravel()
                 valid = local < nc
-                q = np.where(valid, pr_env.reshape(-1)[np.minimum(inds, n_rows * nc - 1)], 0.0)
+                pr2 = pr_env.reshape(n_rows, nc)
+                q = np.where(valid, np.take_along_axis(pr2, np.minimum(inds_rows, nc - 1), axis=1).ravel(), 0.0)
                 q3 = q.reshape(A, T, R)
                 rec3 = rc[..., -1]  # (A,T)
                 for ai in range(A):
