@@ -103,6 +103,52 @@ def run_functional_metric_test(
         _assert_allclose(res, ref, atol)
 
 
+def run_class_metric_ddp_test(
+    metric_class: Callable,
+    ref_fn: Callable,
+    preds: Tensor,
+    target: Tensor,
+    metric_args: Optional[Dict[str, Any]] = None,
+    atol: float = 1e-5,
+    world_size: int = 2,
+) -> None:
+    """DDP pool sweep (reference _class_test ddp mode): each rank consumes
+    interleaved batches; the synced compute must equal the oracle on ALL data."""
+
+    def worker(rank: int, ws: int) -> None:
+        m = metric_class(**(metric_args or {}))
+        for i in range(rank, preds.shape[0], ws):
+            m.update(preds[i], target[i])
+        val = m.compute()
+        flat_p = preds.reshape(-1, *preds.shape[2:])
+        flat_t = target.reshape(-1, *target.shape[2:])
+        _assert_allclose(val, ref_fn(flat_p, flat_t), atol)
+
+    run_distributed(worker, world_size=world_size)
+
+
+def run_dtype_test(
+    metric_class: Callable,
+    preds: Tensor,
+    target: Tensor,
+    metric_args: Optional[Dict[str, Any]] = None,
+    dtype: torch.dtype = torch.double,
+    atol: float = 1e-4,
+) -> None:
+    """Reference run_precision_test: the metric computed with states/inputs in
+    ``dtype`` must agree with the fp32 run."""
+    base = metric_class(**(metric_args or {}))
+    for i in range(preds.shape[0]):
+        base.update(preds[i], target[i])
+    expected = base.compute()
+
+    m = metric_class(**(metric_args or {})).set_dtype(dtype)
+    p = preds.to(dtype) if preds.is_floating_point() else preds
+    for i in range(p.shape[0]):
+        m.update(p[i], target[i])
+    _assert_allclose(m.compute(), expected, atol)
+
+
 # ---------------------------------------------------------------- distributed
 def _dist_worker(rank: int, world_size: int, port: int, fn: Callable, args: tuple, backend: str = "gloo") -> None:
     # Forked children inherit the parent's (possibly mid-operation) OpenMP pool
