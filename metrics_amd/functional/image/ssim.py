@@ -37,6 +37,17 @@ def _uniform_kernel_2d(
     return kernel.expand(channel, 1, kernel_size[0], kernel_size[1])
 
 
+def _gaussian_kernel_3d(
+    channel: int, kernel_size: Sequence[int], sigma: Sequence[float], dtype: torch.dtype, device: torch.device
+) -> Tensor:
+    gx = _gaussian(kernel_size[0], sigma[0], dtype, device)
+    gy = _gaussian(kernel_size[1], sigma[1], dtype, device)
+    gz = _gaussian(kernel_size[2], sigma[2], dtype, device)
+    kernel_xy = torch.matmul(gx.t(), gy)
+    kernel = kernel_xy.unsqueeze(-1) * gz.expand(kernel_size[0], kernel_size[1], kernel_size[2])
+    return kernel.expand(channel, 1, kernel_size[0], kernel_size[1], kernel_size[2])
+
+
 def _ssim_check_inputs(preds: Tensor, target: Tensor) -> Tuple[Tensor, Tensor]:
     if preds.dtype != target.dtype:
         target = target.to(preds.dtype)
@@ -59,13 +70,17 @@ def _ssim_compute(
     return_contrast_sensitivity: bool = False,
 ):
     is_3d = preds.ndim == 5
-    if is_3d:
-        raise NotImplementedError("3D SSIM not yet supported in this build (2D images only)")
+    dims = 3 if is_3d else 2
 
     if not isinstance(kernel_size, Sequence):
-        kernel_size = 2 * [kernel_size]
+        kernel_size = dims * [kernel_size]
     if not isinstance(sigma, Sequence):
-        sigma = 2 * [sigma]
+        sigma = dims * [sigma]
+    if len(kernel_size) != dims or len(sigma) != dims:
+        raise ValueError(
+            f"`kernel_size` and `sigma` must have {dims} elements for {dims}d input,"
+            f" got {kernel_size} and {sigma}"
+        )
 
     if any(x % 2 == 0 or x <= 0 for x in kernel_size):
         raise ValueError(f"Expected `kernel_size` to have odd positive number. Got {kernel_size}.")
@@ -93,17 +108,34 @@ def _ssim_compute(
     if gaussian_kernel:
         pad_h = (gauss_kernel_size[0] - 1) // 2
         pad_w = (gauss_kernel_size[1] - 1) // 2
-        kernel = _gaussian_kernel_2d(channel, gauss_kernel_size, sigma, dtype, device)
     else:
         pad_h = (kernel_size[0] - 1) // 2
         pad_w = (kernel_size[1] - 1) // 2
-        kernel = _uniform_kernel_2d(channel, kernel_size, dtype, device)
 
-    preds_p = F.pad(preds, (pad_w, pad_w, pad_h, pad_h), mode="reflect")
-    target_p = F.pad(target, (pad_w, pad_w, pad_h, pad_h), mode="reflect")
+    if is_3d:
+        # reference quirk kept: depth padding always follows kernel_size[2],
+        # even for gaussian windows (ref functional/image/ssim.py:136-141)
+        pad_d = (kernel_size[2] - 1) // 2
+        preds_p = F.pad(preds, (pad_d, pad_d, pad_w, pad_w, pad_h, pad_h), mode="reflect")
+        target_p = F.pad(target, (pad_d, pad_d, pad_w, pad_w, pad_h, pad_h), mode="reflect")
+        if gaussian_kernel:
+            kernel = _gaussian_kernel_3d(channel, gauss_kernel_size, sigma, dtype, device)
+        else:
+            kernel = torch.ones((channel, 1, *kernel_size), dtype=dtype, device=device) / torch.prod(
+                torch.tensor(kernel_size, dtype=dtype, device=device)
+            )
+    else:
+        preds_p = F.pad(preds, (pad_w, pad_w, pad_h, pad_h), mode="reflect")
+        target_p = F.pad(target, (pad_w, pad_w, pad_h, pad_h), mode="reflect")
+        if gaussian_kernel:
+            kernel = _gaussian_kernel_2d(channel, gauss_kernel_size, sigma, dtype, device)
+        else:
+            kernel = _uniform_kernel_2d(channel, kernel_size, dtype, device)
 
     input_list = torch.cat((preds_p, target_p, preds_p * preds_p, target_p * target_p, preds_p * target_p))
-    outputs = F.conv2d(input_list, kernel, groups=channel)
+    outputs = (
+        F.conv3d(input_list, kernel, groups=channel) if is_3d else F.conv2d(input_list, kernel, groups=channel)
+    )
     output_list = outputs.split(preds.shape[0])
 
     mu_pred_sq = output_list[0].pow(2)
@@ -123,7 +155,11 @@ def _ssim_compute(
     sim = ssim_idx_full_image.reshape(ssim_idx_full_image.shape[0], -1).mean(-1)
 
     if return_contrast_sensitivity:
-        contrast_sensitivity = (upper / lower)[..., pad_h:-pad_h, pad_w:-pad_w]
+        contrast_sensitivity = upper / lower
+        if is_3d:
+            contrast_sensitivity = contrast_sensitivity[..., pad_h:-pad_h, pad_w:-pad_w, pad_d:-pad_d]
+        else:
+            contrast_sensitivity = contrast_sensitivity[..., pad_h:-pad_h, pad_w:-pad_w]
         return sim, contrast_sensitivity.reshape(contrast_sensitivity.shape[0], -1).mean(-1)
     if return_full_image:
         return sim, ssim_idx_full_image
@@ -185,17 +221,28 @@ def multiscale_structural_similarity_index_measure(
     if normalize not in ("relu", "simple", None):
         raise ValueError("Argument `normalize` to be expected either `None` or one of 'relu' or 'simple'")
 
-    sizes = preds.shape[-2:]
-    if (sizes[0] // (2 ** (len(betas) - 1))) <= (kernel_size if isinstance(kernel_size, int) else kernel_size[0]) or (
-        sizes[1] // (2 ** (len(betas) - 1))
-    ) <= (kernel_size if isinstance(kernel_size, int) else kernel_size[-1]):
+    k0 = kernel_size if isinstance(kernel_size, int) else kernel_size[0]
+    k1_ = kernel_size if isinstance(kernel_size, int) else kernel_size[1]
+    if preds.size()[-1] < 2 ** len(betas) or preds.size()[-2] < 2 ** len(betas):
         raise ValueError(
-            f"For a given number of `betas` parameters {len(betas)}, the image height and width should be larger than"
-            f" {(kernel_size if isinstance(kernel_size, int) else kernel_size[0]) * (2 ** (len(betas) - 1))}"
+            f"For a given number of `betas` parameters {len(betas)}, the image height and width dimensions must be"
+            f" larger than or equal to {2 ** len(betas)}."
+        )
+    _betas_div = max(1, (len(betas) - 1)) ** 2
+    if preds.size()[-2] // _betas_div <= k0 - 1:
+        raise ValueError(
+            f"For a given number of `betas` parameters {len(betas)} and kernel size {k0},"
+            f" the image height must be larger than {(k0 - 1) * _betas_div}."
+        )
+    if preds.size()[-1] // _betas_div <= k1_ - 1:
+        raise ValueError(
+            f"For a given number of `betas` parameters {len(betas)} and kernel size {k1_},"
+            f" the image width must be larger than {(k1_ - 1) * _betas_div}."
         )
 
     mcs_list: List[Tensor] = []
     sim = None
+    is_3d = preds.ndim == 5
     for i in range(len(betas)):
         sim, contrast_sensitivity = _ssim_compute(
             preds, target, gaussian_kernel, sigma, kernel_size, data_range, k1, k2,
@@ -203,8 +250,12 @@ def multiscale_structural_similarity_index_measure(
         )
         mcs_list.append(contrast_sensitivity)
         if i < len(betas) - 1:
-            preds = F.avg_pool2d(preds, (2, 2))
-            target = F.avg_pool2d(target, (2, 2))
+            if is_3d:
+                preds = F.avg_pool3d(preds, (2, 2, 2))
+                target = F.avg_pool3d(target, (2, 2, 2))
+            else:
+                preds = F.avg_pool2d(preds, (2, 2))
+                target = F.avg_pool2d(target, (2, 2))
 
     mcs_list[-1] = sim
     mcs_stack = torch.stack(mcs_list)

@@ -239,3 +239,71 @@ def test_clip_iqa_and_infolm_raise_without_models():
         ma.multimodal.CLIPImageQualityAssessment()
     with pytest.raises(ModuleNotFoundError):
         ma.text.InfoLM()
+
+
+def _ref_image_functional(name):
+    """Import a reference functional (offline oracle); None if unavailable."""
+    import os
+    import sys
+
+    if not os.path.isdir("/root/reference/src"):
+        return None
+    sys.path.insert(0, "/root/repo/tools/refbench")
+    sys.path.insert(0, "/root/reference/src")
+    try:
+        import torchmetrics.functional.image as ref_img
+    except Exception:
+        return None
+    return getattr(ref_img, name)
+
+
+@pytest.mark.parametrize(
+    "kwargs",
+    [
+        {},
+        {"gaussian_kernel": False, "kernel_size": 5},
+        {"sigma": 1.0},
+        {"return_contrast_sensitivity": True},
+        {"reduction": "none"},
+        {"data_range": (0.1, 0.9)},
+    ],
+)
+def test_ssim_3d_vs_reference(kwargs):
+    ref = _ref_image_functional("structural_similarity_index_measure")
+    if ref is None:
+        pytest.skip("reference tree not available")
+    from metrics_amd.functional.image import structural_similarity_index_measure as ours
+
+    torch.manual_seed(0)
+    p = torch.rand(2, 1, 16, 16, 16)
+    t = torch.rand(2, 1, 16, 16, 16)
+    a, b = ours(p, t, **kwargs), ref(p, t, **kwargs)
+    if isinstance(a, tuple):
+        for x, y in zip(a, b):
+            assert torch.allclose(x, y, atol=1e-6)
+    else:
+        assert torch.allclose(a, b, atol=1e-6)
+
+
+def test_ms_ssim_3d_vs_reference():
+    ref = _ref_image_functional("multiscale_structural_similarity_index_measure")
+    if ref is None:
+        pytest.skip("reference tree not available")
+    from metrics_amd.functional.image import multiscale_structural_similarity_index_measure as ours
+
+    torch.manual_seed(1)
+    p = torch.rand(1, 1, 180, 180, 180)
+    t = torch.rand(1, 1, 180, 180, 180)
+    assert torch.allclose(ours(p, t), ref(p, t), atol=1e-6)
+
+
+def test_ssim_3d_identity_and_bounds():
+    """Self-contained (no reference needed): identical volumes score 1."""
+    from metrics_amd.functional.image import structural_similarity_index_measure as ssim
+
+    torch.manual_seed(2)
+    p = torch.rand(2, 2, 12, 12, 12)
+    assert torch.allclose(ssim(p, p.clone(), data_range=1.0), torch.tensor(1.0), atol=1e-5)
+    t = torch.rand(2, 2, 12, 12, 12)
+    v = ssim(p, t, data_range=1.0)
+    assert -1.0 <= float(v) <= 1.0
