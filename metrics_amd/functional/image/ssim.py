@@ -48,6 +48,70 @@ def _gaussian_kernel_3d(
     return kernel.expand(channel, 1, kernel_size[0], kernel_size[1], kernel_size[2])
 
 
+def _try_ssim_hip(
+    preds: Tensor,
+    target: Tensor,
+    gaussian_kernel: bool,
+    sigma: Sequence[float],
+    kernel_size: Sequence[int],
+    data_range_t,
+    k1: float,
+    k2: float,
+    is_3d: bool,
+    return_full_image: bool,
+    return_contrast_sensitivity: bool,
+):
+    """K8 fused one-kernel SSIM (csrc/ssim.hip) for the GPU 2D path.
+
+    Returns None when this shape/mode needs the torch formulation (3D,
+    full-image output, oversized windows, reflect-pad-invalid sizes).
+    """
+    if not preds.is_cuda or is_3d or return_full_image:
+        return None
+    gauss_kernel_size = [int(3.5 * s + 0.5) * 2 + 1 for s in sigma]
+    if gaussian_kernel:
+        pad_h = (gauss_kernel_size[0] - 1) // 2
+        pad_w = (gauss_kernel_size[1] - 1) // 2
+    else:
+        pad_h = (kernel_size[0] - 1) // 2
+        pad_w = (kernel_size[1] - 1) // 2
+    B, C, H, W = preds.shape
+    if pad_h >= H or pad_w >= W:  # reflect pad invalid: torch path raises
+        return None
+    if return_contrast_sensitivity and (H <= 2 * pad_h or W <= 2 * pad_w):
+        return None
+    dtype = preds.dtype
+    device = preds.device
+    if gaussian_kernel:
+        wh = _gaussian(gauss_kernel_size[0], sigma[0], torch.float32, device)[0]
+        ww = _gaussian(gauss_kernel_size[1], sigma[1], torch.float32, device)[0]
+    else:
+        wh = torch.full((kernel_size[0],), 1.0 / kernel_size[0], device=device)
+        ww = torch.full((kernel_size[1],), 1.0 / kernel_size[1], device=device)
+
+    if isinstance(data_range_t, Tensor):
+        dr = data_range_t.float()
+        c1 = c2 = 0.0
+    else:
+        dr = None
+        c1 = float(pow(k1 * data_range_t, 2))
+        c2 = float(pow(k2 * data_range_t, 2))
+    from metrics_amd.ops import _hip
+
+    try:
+        sum_sim, sum_cs = _hip.ssim2d_fused(
+            preds, target, wh, ww, c1, c2, dr, k1, k2,
+            return_contrast_sensitivity, pad_h, pad_w,
+        )
+    except _hip.SsimLdsOverflow:
+        return None
+    sim = (sum_sim / (C * H * W)).to(dtype)
+    if return_contrast_sensitivity:
+        cs = (sum_cs / (C * (H - 2 * pad_h) * (W - 2 * pad_w))).to(dtype)
+        return sim, cs
+    return sim
+
+
 def _ssim_check_inputs(preds: Tensor, target: Tensor) -> Tuple[Tensor, Tensor]:
     if preds.dtype != target.dtype:
         target = target.to(preds.dtype)
@@ -98,6 +162,13 @@ def _ssim_compute(
 
     c1 = pow(k1 * data_range_t, 2)
     c2 = pow(k2 * data_range_t, 2)
+
+    hip_out = _try_ssim_hip(
+        preds, target, gaussian_kernel, sigma, kernel_size, data_range_t, k1, k2,
+        is_3d, return_full_image, return_contrast_sensitivity,
+    )
+    if hip_out is not None:
+        return hip_out
 
     channel = preds.size(1)
     dtype = preds.dtype

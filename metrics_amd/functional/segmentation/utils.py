@@ -71,6 +71,14 @@ def binary_erosion(
     if origin is None:
         origin = structure.ndim * (1,)
 
+    if image.is_cuda and image.ndim == 4 and structure.ndim == 2:
+        # K8 windowed-min kernel (csrc/ssim.hip): no (N,C,H,W,k^2) unfold
+        from metrics_amd.ops import _hip
+
+        return _hip.binary_erosion2d(
+            image.to(torch.uint8), structure, (int(origin[0]), int(origin[1])), int(border_value)
+        )
+
     # pad so each output pixel sees its full neighborhood (origin-shifted)
     pad_spec = [x for i in range(len(origin)) for x in (origin[i], structure.shape[i] - origin[i] - 1)]
     image_pad = pad(image, pad_spec, mode="constant", value=border_value)

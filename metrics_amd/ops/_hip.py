@@ -62,6 +62,8 @@ _SIGNATURES = {
     "ma_box_iou": [_U64, _U64, _LL, _U64, _LL, _I, _U64],
     "ma_clf_curve_scratch_bytes": [_LL, _I, ctypes.POINTER(ctypes.c_ulonglong)],
     "ma_binary_clf_curve": [_U64, _U64, _U64, _U64, _LL, _LL, _U64, ctypes.c_ulonglong, _U64, _U64, _U64, _U64],
+    "ma_ssim2d_fused": [_U64, _U64, _U64, _I, _LL, _LL, _LL, _LL, _U64, _I, _U64, _I, _F, _F, _U64, _F, _F, _I, _I, _I, _U64, _U64],
+    "ma_binary_erosion2d": [_U64, _U64, _LL, _LL, _LL, _LL, _U64, _I, _I, _I, _I, _I, _U64],
 }
 
 
@@ -535,6 +537,86 @@ def err_reduce(x: Tensor, y: Tensor, op: str, eps: float = 1.17e-6) -> Tensor:
         out.data_ptr(),
     )
     _check(rc, "ma_err_reduce")
+    return out
+
+
+class SsimLdsOverflow(RuntimeError):
+    """Window too large for the one-kernel LDS tiling; caller falls back."""
+
+
+def ssim2d_fused(
+    preds: Tensor,
+    target: Tensor,
+    wh: Tensor,
+    ww: Tensor,
+    c1: float,
+    c2: float,
+    dr: Optional[Tensor],
+    k1: float,
+    k2: float,
+    want_cs: bool,
+    crop_h: int,
+    crop_w: int,
+) -> Tuple[Tensor, Optional[Tensor]]:
+    """Fused 2D SSIM: per-image Σssim (and optional cropped Σcs) in ONE kernel.
+
+    ``wh``/``ww`` are the separable window weight vectors (odd length);
+    ``dr`` an optional 0-dim float32 device tensor holding data_range (c1/c2
+    then computed in-kernel — no host sync for the data_range=None path).
+    """
+    lib = _lib()
+    preds = _to_supported(preds).contiguous()
+    target = target.to(preds.dtype).contiguous()
+    B, C, H, W = preds.shape
+    rh = (wh.numel() - 1) // 2
+    rw = (ww.numel() - 1) // 2
+    dev = preds.device
+    wh = wh.contiguous().float()
+    ww = ww.contiguous().float()
+    sum_sim = torch.zeros(B, dtype=torch.float64, device=dev)
+    sum_cs = torch.zeros(B, dtype=torch.float64, device=dev) if want_cs else None
+    rc = lib.ma_ssim2d_fused(
+        _stream(),
+        preds.data_ptr(),
+        target.data_ptr(),
+        _dtype_code(preds),
+        B, C, H, W,
+        wh.data_ptr(), rh,
+        ww.data_ptr(), rw,
+        c1, c2,
+        dr.data_ptr() if dr is not None else 0,
+        k1, k2,
+        1 if want_cs else 0,
+        crop_h, crop_w,
+        sum_sim.data_ptr(),
+        sum_cs.data_ptr() if sum_cs is not None else 0,
+    )
+    if rc == 9001:
+        raise SsimLdsOverflow(f"window radius ({rh},{rw}) exceeds the LDS tile budget")
+    _check(rc, "ma_ssim2d_fused")
+    return sum_sim, sum_cs
+
+
+def binary_erosion2d(img: Tensor, strel: Tensor, origin: Tuple[int, int], border_value: int) -> Tensor:
+    """Windowed erosion of a (N,C,H,W) binary uint8 image; returns uint8."""
+    lib = _lib()
+    img = img.contiguous()
+    assert img.dtype == torch.uint8 and img.ndim == 4
+    N, C, H, W = img.shape
+    strel = strel.contiguous().int()
+    kh, kw = strel.shape
+    out = torch.empty_like(img)
+    rc = lib.ma_binary_erosion2d(
+        _stream(),
+        img.data_ptr(),
+        N, C, H, W,
+        strel.data_ptr(),
+        kh, kw,
+        origin[0], origin[1],
+        int(border_value),
+        out.data_ptr(),
+    )
+    _check(rc, "ma_binary_erosion2d")
     return out
 
 

@@ -878,3 +878,96 @@ def test_exact_multiclass_curve_gpu():
     mc = ma.MulticlassAveragePrecision(num_classes=7, thresholds=None, average="macro")
     mc.update(preds, target)
     assert torch.allclose(mg.compute().cpu(), mc.compute(), atol=1e-6)
+
+
+# ----------------------------------------------------------- K8 fused SSIM
+@pytest.mark.parametrize("kwargs", [
+    {},
+    {"sigma": 1.0},
+    {"gaussian_kernel": False, "kernel_size": 7},
+    {"data_range": 2.5},
+    {"data_range": (0.1, 0.9)},
+    {"reduction": "none"},
+])
+def test_ssim_fused_vs_cpu(kwargs):
+    from metrics_amd.functional.image import structural_similarity_index_measure as ssim
+
+    torch.manual_seed(31)
+    p = torch.rand(3, 2, 57, 83)
+    t = torch.rand(3, 2, 57, 83)
+    got = ssim(p.cuda(), t.cuda(), **kwargs)
+    exp = ssim(p, t, **kwargs)
+    assert torch.allclose(got.cpu(), exp, atol=1e-5), (got, exp)
+
+
+def test_ssim_fused_contrast_sensitivity_and_ms():
+    from metrics_amd.functional.image import (
+        multiscale_structural_similarity_index_measure as ms_ssim,
+        structural_similarity_index_measure as ssim,
+    )
+
+    torch.manual_seed(32)
+    p = torch.rand(2, 3, 200, 200)
+    t = torch.rand(2, 3, 200, 200)
+    g_sim, g_cs = ssim(p.cuda(), t.cuda(), return_contrast_sensitivity=True)
+    c_sim, c_cs = ssim(p, t, return_contrast_sensitivity=True)
+    assert torch.allclose(g_sim.cpu(), c_sim, atol=1e-5)
+    assert torch.allclose(g_cs.cpu(), c_cs, atol=1e-5)
+    assert torch.allclose(ms_ssim(p.cuda(), t.cuda()).cpu(), ms_ssim(p, t), atol=1e-4)
+
+
+def test_ssim_fused_bf16():
+    from metrics_amd.functional.image import structural_similarity_index_measure as ssim
+
+    torch.manual_seed(33)
+    p = torch.rand(2, 1, 64, 64)
+    t = torch.rand(2, 1, 64, 64)
+    got = ssim(p.cuda().bfloat16(), t.cuda().bfloat16(), data_range=1.0)
+    exp = ssim(p, t, data_range=1.0)
+    assert torch.allclose(got.float().cpu(), exp, atol=2e-2)
+
+
+def test_ssim_metric_gpu_end_to_end():
+    torch.manual_seed(34)
+    p = torch.rand(4, 3, 96, 96)
+    t = torch.rand(4, 3, 96, 96)
+    mg = ma.StructuralSimilarityIndexMeasure(data_range=1.0).to("cuda")
+    mg.update(p.cuda(), t.cuda())
+    mc = ma.StructuralSimilarityIndexMeasure(data_range=1.0)
+    mc.update(p, t)
+    assert torch.allclose(mg.compute().cpu(), mc.compute(), atol=1e-5)
+
+
+# ------------------------------------------------------------- K8 erosion
+def test_binary_erosion_gpu_vs_cpu():
+    from metrics_amd.functional.segmentation.utils import binary_erosion, generate_binary_structure
+
+    torch.manual_seed(35)
+    img = (torch.rand(2, 3, 41, 53) > 0.4).int()
+    for structure, origin in [
+        (None, None),
+        (torch.ones(3, 3, dtype=torch.int32), None),
+        (torch.tensor([[0, 1, 0], [1, 1, 1], [0, 1, 0]], dtype=torch.int32), None),
+        (torch.ones(5, 3, dtype=torch.int32), (2, 1)),
+        (torch.tensor([[1, 1], [0, 1]], dtype=torch.int32), (0, 0)),
+    ]:
+        cpu = binary_erosion(img, structure, origin)
+        gpu = binary_erosion(
+            img.cuda(), structure.cuda() if structure is not None else None, origin
+        )
+        assert torch.equal(gpu.cpu(), cpu), (structure, origin)
+    # border_value=1
+    cpu = binary_erosion(img, border_value=1)
+    gpu = binary_erosion(img.cuda(), border_value=1)
+    assert torch.equal(gpu.cpu(), cpu)
+
+
+def test_hausdorff_gpu_vs_cpu():
+    torch.manual_seed(36)
+    p = torch.randint(0, 2, (2, 2, 32, 32))
+    t = torch.randint(0, 2, (2, 2, 32, 32))
+    mg = ma.segmentation.HausdorffDistance(num_classes=2).to("cuda")
+    mg.update(p.cuda(), t.cuda())
+    mc = ma.segmentation.HausdorffDistance(num_classes=2)
+    mc.update(p, t)
+    assert torch.allclose(mg.compute().cpu(), mc.compute(), atol=1e-5)
