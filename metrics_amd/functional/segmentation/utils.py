@@ -79,8 +79,13 @@ def binary_erosion(
             image.to(torch.uint8), structure, (int(origin[0]), int(origin[1])), int(border_value)
         )
 
-    # pad so each output pixel sees its full neighborhood (origin-shifted)
-    pad_spec = [x for i in range(len(origin)) for x in (origin[i], structure.shape[i] - origin[i] - 1)]
+    # pad so each output pixel sees its full neighborhood (origin-shifted).
+    # F.pad's spec is last-dim-first, so build it reversed — this makes
+    # non-square structuring elements work (the reference's un-reversed
+    # spec crashes on them; square elements are unaffected)
+    pad_spec = [
+        x for i in reversed(range(len(origin))) for x in (origin[i], structure.shape[i] - origin[i] - 1)
+    ]
     image_pad = pad(image, pad_spec, mode="constant", value=border_value)
 
     # windows: (N, C, *spatial, prod(kernel)) via unfold
