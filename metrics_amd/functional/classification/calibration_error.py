@@ -20,6 +20,16 @@ def _binning_bucketize(
     confidences: Tensor, accuracies: Tensor, bin_boundaries: Tensor
 ) -> Tuple[Tensor, Tensor, Tensor]:
     """Per-bin mean accuracy, mean confidence and bin proportion via bucketize + scatter."""
+    if confidences.is_cuda:
+        # K5: fused bucketize + triple histogram (csrc/kernels2.hip)
+        from metrics_amd.ops import _hip
+
+        acc_sum, conf_sum, count_bin = _hip.calib_bins(confidences, accuracies, bin_boundaries)
+        conf_bin = torch.nan_to_num(conf_sum / count_bin).to(confidences.dtype)
+        acc_bin = torch.nan_to_num(acc_sum / count_bin).to(confidences.dtype)
+        prop_bin = (count_bin / count_bin.sum()).to(confidences.dtype)
+        return acc_bin, conf_bin, prop_bin
+
     accuracies = accuracies.to(dtype=confidences.dtype)
     acc_bin = torch.zeros(len(bin_boundaries) - 1, device=confidences.device, dtype=confidences.dtype)
     conf_bin = torch.zeros(len(bin_boundaries) - 1, device=confidences.device, dtype=confidences.dtype)

@@ -304,6 +304,25 @@ def _multiclass_stat_scores_update(
             preds[idx] = num_classes
 
         if top_k > 1:
+            if (
+                preds.is_cuda
+                and multidim_average == "global"
+                and preds.is_floating_point()
+                and preds.ndim == 3
+                and preds.shape[2] == 1
+            ):
+                # K3: one-pass per-row top-k stat kernel (csrc/kernels2.hip).
+                # An out-of-range ignore_index was remapped to num_classes in
+                # the clone above — skip on the remapped value then.
+                from metrics_amd.ops import _hip
+
+                eff_ignore = ignore_index
+                if ignore_index is not None and not 0 <= ignore_index <= num_classes - 1:
+                    eff_ignore = num_classes
+                tp, fp, tn, fn = _hip.mc_topk_stat(
+                    preds[:, :, 0], target.flatten(), num_classes, top_k, eff_ignore
+                )
+                return tp, fp, tn, fn
             preds_oh = torch.movedim(select_topk(preds, topk=top_k, dim=1), 1, -1)
             preds_oh = _refine_preds_oh(preds, preds_oh, target, top_k)
         else:

@@ -64,6 +64,8 @@ _SIGNATURES = {
     "ma_binary_clf_curve": [_U64, _U64, _U64, _U64, _LL, _LL, _U64, ctypes.c_ulonglong, _U64, _U64, _U64, _U64],
     "ma_ssim2d_fused": [_U64, _U64, _U64, _I, _LL, _LL, _LL, _LL, _U64, _I, _U64, _I, _F, _F, _U64, _F, _F, _I, _I, _I, _U64, _U64],
     "ma_binary_erosion2d": [_U64, _U64, _LL, _LL, _LL, _LL, _U64, _I, _I, _I, _I, _I, _U64],
+    "ma_calib_bins": [_U64, _U64, _U64, _LL, _U64, _I, _I, _F, _F, _U64],
+    "ma_mc_topk_stat": [_U64, _U64, _I, _U64, _LL, _LL, _I, _LL, _I, _U64, _U64, _U64, _U64],
 }
 
 
@@ -538,6 +540,81 @@ def err_reduce(x: Tensor, y: Tensor, op: str, eps: float = 1.17e-6) -> Tensor:
     )
     _check(rc, "ma_err_reduce")
     return out
+
+
+_CALIB_SCALE = 8589934592.0  # 2^33, matches csrc/kernels2.hip
+
+
+def calib_bins(conf: Tensor, acc: Tensor, bounds: Tensor) -> Tuple[Tensor, Tensor, Tensor]:
+    """Fused bucketize + per-bin (count, Σconf, Σacc) — K5 kernel.
+
+    Integer fixed-point accumulation => deterministic. Returns fp32
+    (acc_bin_sum, conf_bin_sum, count_bin) of length ``len(bounds)-1``.
+    """
+    lib = _lib()
+    conf = _to_supported(conf).float().contiguous().flatten()
+    acc = acc.float().contiguous().flatten()
+    bounds = bounds.float().contiguous()
+    n_bins = bounds.numel() - 1
+    uni, b0, inv_step = _uniform_params(bounds)
+    out = torch.zeros(3, n_bins, dtype=torch.long, device=conf.device)
+    rc = lib.ma_calib_bins(
+        _stream(),
+        conf.data_ptr(),
+        acc.data_ptr(),
+        conf.numel(),
+        bounds.data_ptr(),
+        n_bins,
+        uni,
+        b0,
+        inv_step,
+        out.data_ptr(),
+    )
+    if rc == 9001:
+        raise RuntimeError("calibration bin count exceeds the LDS budget")
+    _check(rc, "ma_calib_bins")
+    count = out[0].float()
+    conf_sum = out[1].double().div_(_CALIB_SCALE).float()
+    acc_sum = out[2].double().div_(_CALIB_SCALE).float()
+    return acc_sum, conf_sum, count
+
+
+def mc_topk_stat(
+    preds: Tensor, target: Tensor, num_classes: int, k: int, ignore_index: Optional[int]
+) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
+    """Per-class (tp, fp, tn, fn) for top-k multiclass stat scores — K3 kernel.
+
+    torchmetrics semantics: the effective prediction is the target when it
+    ranks in the top-k, else the argmax (_refine_preds_oh).
+    """
+    lib = _lib()
+    assert preds.ndim == 2 and preds.is_floating_point()
+    preds = _to_supported(preds).contiguous()
+    target = target.contiguous().long()
+    B, C = preds.shape
+    dev = preds.device
+    tp = torch.zeros(C, dtype=torch.long, device=dev)
+    fp = torch.zeros(C, dtype=torch.long, device=dev)
+    fn = torch.zeros(C, dtype=torch.long, device=dev)
+    valid = torch.zeros(1, dtype=torch.long, device=dev)
+    rc = lib.ma_mc_topk_stat(
+        _stream(),
+        preds.data_ptr(),
+        _dtype_code(preds),
+        target.data_ptr(),
+        B,
+        C,
+        k,
+        ignore_index if ignore_index is not None else 0,
+        1 if ignore_index is not None else 0,
+        tp.data_ptr(),
+        fp.data_ptr(),
+        fn.data_ptr(),
+        valid.data_ptr(),
+    )
+    _check(rc, "ma_mc_topk_stat")
+    tn = valid - (tp + fp + fn)
+    return tp, fp, tn, fn
 
 
 class SsimLdsOverflow(RuntimeError):

@@ -971,3 +971,79 @@ def test_hausdorff_gpu_vs_cpu():
     mc = ma.segmentation.HausdorffDistance(num_classes=2)
     mc.update(p, t)
     assert torch.allclose(mg.compute().cpu(), mc.compute(), atol=1e-5)
+
+
+# ---------------------------------------------------------- K5 calibration
+@pytest.mark.parametrize("n_bins", [10, 15, 99])
+@pytest.mark.parametrize("norm", ["l1", "l2", "max"])
+def test_calibration_error_gpu_vs_cpu(n_bins, norm):
+    torch.manual_seed(41)
+    preds = torch.rand(5000)
+    target = torch.randint(0, 2, (5000,))
+    mg = ma.BinaryCalibrationError(n_bins=n_bins, norm=norm).to("cuda")
+    mg.update(preds.cuda(), target.cuda())
+    mc = ma.BinaryCalibrationError(n_bins=n_bins, norm=norm)
+    mc.update(preds, target)
+    assert torch.allclose(mg.compute().cpu(), mc.compute(), atol=1e-5)
+
+
+def test_multiclass_calibration_error_gpu():
+    torch.manual_seed(42)
+    preds = torch.randn(2000, 9).softmax(-1)
+    target = torch.randint(0, 9, (2000,))
+    mg = ma.MulticlassCalibrationError(num_classes=9).to("cuda")
+    mg.update(preds.cuda(), target.cuda())
+    mc = ma.MulticlassCalibrationError(num_classes=9)
+    mc.update(preds, target)
+    assert torch.allclose(mg.compute().cpu(), mc.compute(), atol=1e-5)
+
+
+def test_calib_bins_deterministic_repeat():
+    torch.manual_seed(43)
+    conf = torch.rand(200_000, device="cuda")
+    acc = (torch.rand(200_000, device="cuda") > 0.5).float()
+    bounds = torch.linspace(0, 1, 16, device="cuda")
+    from metrics_amd.ops import _hip
+
+    a1 = _hip.calib_bins(conf, acc, bounds)
+    a2 = _hip.calib_bins(conf, acc, bounds)
+    for x, y in zip(a1, a2):
+        assert torch.equal(x, y)
+
+
+# --------------------------------------------------------------- K3 top-k
+@pytest.mark.parametrize("k", [2, 3, 5])
+@pytest.mark.parametrize("ignore_index", [None, 2, -1])
+@pytest.mark.parametrize("average", ["micro", "macro"])
+def test_topk_stat_gpu_vs_cpu(k, ignore_index, average):
+    torch.manual_seed(44)
+    C = 11
+    preds = torch.randn(3000, C).softmax(-1)
+    target = torch.randint(0, C, (3000,))
+    if ignore_index is not None:
+        target[torch.rand(3000) < 0.1] = ignore_index
+    args = dict(num_classes=C, top_k=k, average=average, ignore_index=ignore_index, validate_args=False)
+    mg = ma.MulticlassAccuracy(**args).to("cuda")
+    mg.update(preds.cuda(), target.cuda())
+    mc = ma.MulticlassAccuracy(**args)
+    mc.update(preds, target)
+    assert torch.allclose(mg.compute().cpu(), mc.compute(), atol=1e-6)
+    # stat scores directly
+    sg = ma.MulticlassStatScores(**{**args, "average": None}).to("cuda")
+    sg.update(preds.cuda(), target.cuda())
+    sc = ma.MulticlassStatScores(**{**args, "average": None})
+    sc.update(preds, target)
+    assert torch.equal(sg.compute().cpu(), sc.compute())
+
+
+def test_topk_precision_recall_gpu():
+    torch.manual_seed(45)
+    C = 7
+    preds = torch.randn(1000, C).softmax(-1)
+    target = torch.randint(0, C, (1000,))
+    for cls in (ma.MulticlassPrecision, ma.MulticlassRecall, ma.MulticlassF1Score):
+        mg = cls(num_classes=C, top_k=3, average="macro").to("cuda")
+        mg.update(preds.cuda(), target.cuda())
+        mc = cls(num_classes=C, top_k=3, average="macro")
+        mc.update(preds, target)
+        assert torch.allclose(mg.compute().cpu(), mc.compute(), atol=1e-6), cls.__name__
