@@ -59,6 +59,50 @@ def _squeeze_if_scalar(data: Any) -> Any:
     return apply_to_collection(data, Tensor, lambda x: x.squeeze() if x.numel() == 1 else x)
 
 
+def _clone_result(value: Any) -> Any:
+    """Detach compute() results from the metric states.
+
+    Per-class curve results are lists of many small GPU tensors; cloning them
+    one by one costs one D2D launch each (~100 launches per compute on the
+    curve metrics — profiles/README.md prof15). Batch same-(device,dtype)
+    tensors through ONE cat and hand back views of the packed buffer: the
+    contract only requires the result not to alias metric state.
+    """
+    tensors: List[Tensor] = []
+
+    def _collect(t: Tensor) -> Tensor:
+        tensors.append(t)
+        return t
+
+    apply_to_collection(value, Tensor, _collect)
+    n_cuda = sum(1 for t in tensors if t.is_cuda)
+    if n_cuda < 4:
+        return apply_to_collection(value, Tensor, lambda x: x.clone())
+
+    groups: Dict[Any, List[Tensor]] = {}
+    for t in tensors:
+        groups.setdefault((t.device, t.dtype), []).append(t)
+    cloned: Dict[int, Tensor] = {}
+    for (dev, _dt), ts in groups.items():
+        if dev.type != "cuda" or len(ts) == 1:
+            for t in ts:
+                cloned.setdefault(id(t), t.clone())
+        else:
+            uniq = []
+            seen = set()
+            for t in ts:
+                if id(t) not in seen:
+                    seen.add(id(t))
+                    uniq.append(t)
+            flat = torch.cat([t.reshape(-1) for t in uniq])
+            off = 0
+            for t in uniq:
+                n = t.numel()
+                cloned[id(t)] = flat[off : off + n].view(t.shape)
+                off += n
+    return apply_to_collection(value, Tensor, lambda x: cloned[id(x)])
+
+
 class Metric(Module, ABC):
     """Base class for all metrics.
 
@@ -473,8 +517,9 @@ class Metric(Module, ABC):
             ), tracing.range(f"{self.__class__.__name__}.compute"):
                 self._wait_pending_sync()
                 value = _squeeze_if_scalar(compute(*args, **kwargs))
-                # clone so later in-place ops cannot alter the returned result
-                value = apply_to_collection(value, Tensor, lambda x: x.clone())
+                # detach from state so later in-place ops cannot alter the
+                # returned result (batched for many-tensor curve results)
+                value = _clone_result(value)
 
             if self.compute_with_cache:
                 self._computed = value

@@ -116,3 +116,29 @@ def test_clone_with_collection_groups():
     c2 = coll.clone(prefix="v_")
     r = c2.compute()
     assert set(r) == {"v_MulticlassPrecision", "v_MulticlassRecall"}
+
+
+def test_clone_result_no_alias():
+    """Batched result packing must not alias the inputs and must keep structure."""
+    from metrics_amd.metric import _clone_result
+
+    a = torch.arange(6.0)
+    b = torch.ones(2, 3)
+    c = torch.tensor(5.0)
+    out = _clone_result({"x": a, "y": [b, c], "z": (a,)})
+    assert torch.equal(out["x"], a) and out["x"].data_ptr() != a.data_ptr()
+    assert torch.equal(out["y"][0], b) and out["y"][0].data_ptr() != b.data_ptr()
+    assert torch.equal(out["y"][1], c)
+    out["x"] += 1  # mutating the result must not touch the original
+    assert torch.equal(a, torch.arange(6.0))
+
+
+def test_compute_result_does_not_alias_state():
+    import metrics_amd as ma
+
+    m = ma.MulticlassConfusionMatrix(num_classes=3)
+    m.update(torch.tensor([0, 1, 2]), torch.tensor([0, 1, 1]))
+    res = m.compute()
+    res += 100
+    m._computed = None
+    assert m.compute().max() < 100
