@@ -9,7 +9,8 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 
 import metrics_amd as ma
-import metrics_amd.functional.classification.stat_scores as ss
+import importlib
+ss = importlib.import_module("metrics_amd.functional.classification.stat_scores")
 from metrics_amd.utilities.data import select_topk
 
 

@@ -23,7 +23,7 @@ from metrics_amd import Metric
 class BigSum(Metric):
     full_state_update = False
 
-    def __init__(self, n=32_000_000):
+    def __init__(self, n=256_000_000):
         super().__init__()
         self.n = n
         self.add_state("x", torch.zeros(n), dist_reduce_fx="sum")
