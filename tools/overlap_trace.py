@@ -44,6 +44,23 @@ def main():
     os.environ.setdefault("MASTER_PORT", "29381")
     torch.cuda.set_device(0)
     dist.init_process_group("nccl", rank=0, world_size=1)
+
+    # A world-1 RCCL all-reduce emits NO device kernel (in-place no-op), so on
+    # a single GPU there is no collective to overlap. Substitute a surrogate
+    # workload issued on whatever stream the sync engine uses for the
+    # collective — the trace then shows the engine lets it overlap the next
+    # update()'s kernels (i.e. nothing in the engine serializes the side
+    # stream against the default stream beyond the cheap read fence).
+    import metrics_amd.utilities.distributed as mdist
+
+    orig_all_reduce = dist.all_reduce
+
+    def surrogate_all_reduce(t, *a, **k):
+        for _ in range(8):
+            t.mul_(1.00000001)  # ~8 full-buffer kernels on the current stream
+        return orig_all_reduce(t, *a, **k)
+
+    mdist.dist.all_reduce = surrogate_all_reduce
     try:
         m = BigSum().to("cuda")
         v = torch.rand(m.n, device="cuda")
