@@ -115,6 +115,42 @@ def _encode_masks_rle(masks: Tensor) -> Tensor:
     return torch.cat([header] + counts_per_mask)
 
 
+def _pack_rle_dicts(masks_seq) -> Tensor:
+    """Pack a sequence of pycocotools-style RLE dicts into the internal pack.
+
+    Accepts ``{"size": [h, w], "counts": <compressed str/bytes | list>}`` per
+    mask — the COCO run convention (column-major, background first) is the
+    same one the internal pack uses, so no dense decode happens.
+    """
+    counts_per = []
+    h = w = 0
+    for rle in masks_seq:
+        h, w = int(rle["size"][0]), int(rle["size"][1])
+        c = rle["counts"]
+        if isinstance(c, bytes):
+            c = c.decode()
+        if isinstance(c, str):
+            c = _coco_rle_str_decode(c)
+        counts_per.append(torch.as_tensor(list(c), dtype=torch.int64))
+    n = len(counts_per)
+    if n == 0:
+        return torch.tensor([h, w, 0], dtype=torch.int64)
+    header = torch.tensor([h, w, n] + [int(c.numel()) for c in counts_per], dtype=torch.int64)
+    return torch.cat([header] + counts_per)
+
+
+def _masks_to_pack(masks) -> Tensor:
+    """Dispatch tensor masks vs RLE-dict sequences to the internal pack."""
+    if isinstance(masks, Tensor):
+        return _encode_masks_rle(masks)
+    if isinstance(masks, (list, tuple)) and (len(masks) == 0 or isinstance(masks[0], dict)):
+        return _pack_rle_dicts(masks)
+    raise ValueError(
+        "Expected `masks` to be a (N, H, W) tensor or a sequence of pycocotools RLE dicts"
+        f" but got {type(masks)}"
+    )
+
+
 def _coco_rle_str_encode(counts) -> str:
     """COCO compressed-RLE string from run counts (5-bit varint, delta every 2nd)."""
     out = []
@@ -259,7 +295,7 @@ class MeanAveragePrecision(Metric):
             else:
                 self.detection_boxes.append(torch.zeros(n, 4, device=item["labels"].device))
             if use_masks:
-                self.detection_masks.append(_encode_masks_rle(item["masks"]))
+                self.detection_masks.append(_masks_to_pack(item["masks"]))
             self.detection_scores.append(item["scores"].float())
             self.detection_labels.append(item["labels"].long())
         for item in target:
@@ -271,7 +307,7 @@ class MeanAveragePrecision(Metric):
             else:
                 self.groundtruth_boxes.append(torch.zeros(n, 4, device=item["labels"].device))
             if use_masks:
-                self.groundtruth_masks.append(_encode_masks_rle(item["masks"]))
+                self.groundtruth_masks.append(_masks_to_pack(item["masks"]))
             self.groundtruth_labels.append(item["labels"].long())
             device = item["labels"].device
             crowds = item.get("iscrowd", torch.zeros(n, dtype=torch.long, device=device)).long()

@@ -152,3 +152,49 @@ def test_coco_rle_string_codec():
     for _ in range(20):
         counts = rng.integers(0, 10_000, size=rng.integers(1, 40)).tolist()
         assert _coco_rle_str_decode(_coco_rle_str_encode(counts)) == counts
+
+
+def test_segm_rle_dict_inputs_match_tensor_masks():
+    """update() accepts pycocotools-style RLE dicts (compressed string or
+    plain counts list) and scores identically to dense tensor masks."""
+    import numpy as np
+
+    from metrics_amd.detection.mean_ap import _coco_rle_str_encode
+
+    torch.manual_seed(77)
+
+    def to_rle_dicts(masks, compress):
+        out = []
+        for m in masks.numpy():
+            flat = np.asfortranarray(m).T.reshape(-1)  # column-major
+            change = np.flatnonzero(flat[1:] != flat[:-1]) + 1
+            idx = np.concatenate([[0], change, [flat.size]])
+            counts = (idx[1:] - idx[:-1]).tolist()
+            if flat[0]:
+                counts = [0] + counts
+            c = _coco_rle_str_encode(counts) if compress else counts
+            out.append({"size": [m.shape[0], m.shape[1]], "counts": c})
+        return out
+
+    H = W = 24
+    preds_t, tgts_t, preds_r, tgts_r = [], [], [], []
+    for i in range(3):
+        nd, ng = 4, 3
+        dmasks = (torch.rand(nd, H, W) > 0.6).to(torch.uint8)
+        gmasks = (torch.rand(ng, H, W) > 0.6).to(torch.uint8)
+        scores = torch.rand(nd)
+        dl = torch.randint(0, 2, (nd,))
+        gl = torch.randint(0, 2, (ng,))
+        preds_t.append({"masks": dmasks.bool(), "scores": scores, "labels": dl})
+        tgts_t.append({"masks": gmasks.bool(), "labels": gl})
+        preds_r.append({"masks": to_rle_dicts(dmasks, compress=i % 2 == 0), "scores": scores, "labels": dl})
+        tgts_r.append({"masks": to_rle_dicts(gmasks, compress=i % 2 == 1), "labels": gl})
+
+    mt = MeanAveragePrecision(iou_type="segm")
+    mt.update(preds_t, tgts_t)
+    rt = mt.compute()
+    mr = MeanAveragePrecision(iou_type="segm")
+    mr.update(preds_r, tgts_r)
+    rr = mr.compute()
+    for k in ("map", "map_50", "mar_100"):
+        assert torch.allclose(rt[k], rr[k], atol=1e-6), k
