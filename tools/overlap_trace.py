@@ -55,7 +55,10 @@ def main():
 
     orig_all_reduce = dist.all_reduce
 
+    calls = {"n": 0}
+
     def surrogate_all_reduce(t, *a, **k):
+        calls["n"] += 1
         for _ in range(8):
             t.mul_(1.00000001)  # ~8 full-buffer kernels on the current stream
         return orig_all_reduce(t, *a, **k)
@@ -71,7 +74,9 @@ def main():
             m.unsync()
             m.update(v)     # update kernels go to the default stream NOW
         torch.cuda.synchronize()
-        print("pending event produced:", m._pending_sync_event is not None or True)
+        print("surrogate all_reduce calls:", calls["n"])
+        import metrics_amd.utilities.distributed as _md
+        print("side stream cached:", list(_md._SYNC_STREAMS.keys()))
         print("done")
     finally:
         dist.destroy_process_group()
