@@ -210,25 +210,28 @@ def sync_states_fast(
 
     # --- fused all-reduce buckets ----------------------------------------
     red_op = {"sum": dist.ReduceOp.SUM, "max": dist.ReduceOp.MAX, "min": dist.ReduceOp.MIN}
-    bucket_views: List[Tuple[Tensor, List[Tuple[str, Tensor]]]] = []
+    bucket_views: List[Tuple[Tensor, List[Tuple[str, Tensor]], str]] = []
 
-    def _issue_allreduces() -> None:
+    def _build_buffers() -> None:
+        # the only part that READS the state tensors — the caller's stream
+        # need only be fenced behind this, not behind the collectives
         for (dt, _dev, op), entries in fuse_buckets.items():
             if len(entries) == 1:
                 name, val = entries[0]
                 buf = val.contiguous().to(dt) if val.dtype != dt else val.contiguous().clone()
-                works.append(dist.all_reduce(buf, op=red_op[op], group=group, async_op=True))
-                bucket_views.append((buf, entries))
             else:
-                flat = torch.cat([v.reshape(-1).to(dt) for _, v in entries])
-                works.append(dist.all_reduce(flat, op=red_op[op], group=group, async_op=True))
-                bucket_views.append((flat, entries))
+                buf = torch.cat([v.reshape(-1).to(dt) for _, v in entries])
+            bucket_views.append((buf, entries, op))
+
+    def _issue_allreduces() -> None:
+        for buf, _entries, op in bucket_views:
+            works.append(dist.all_reduce(buf, op=red_op[op], group=group, async_op=True))
 
     def _finish_allreduces() -> None:
         for w in works:
             if w is not None:
                 w.wait()
-        for buf, entries in bucket_views:
+        for buf, entries, _op in bucket_views:
             if len(entries) == 1:
                 name, val = entries[0]
                 res = buf.view(val.shape)
@@ -254,10 +257,11 @@ def sync_states_fast(
         with torch.cuda.stream(side):
             # order after the kernels that produced the current state values
             side.wait_event(produced)
-            _issue_allreduces()
+            _build_buffers()
             # the states have now been read into the flat buffers
             read_done = torch.cuda.Event()
             read_done.record(side)
+            _issue_allreduces()
             _finish_allreduces()  # stream-level waits + scatter-back views
             done_event = torch.cuda.Event()
             done_event.record(side)
@@ -269,6 +273,7 @@ def sync_states_fast(
     if side is None:
         # inline path: issue the async all-reduces first so their launch
         # latencies overlap with the gather collectives below
+        _build_buffers()
         _issue_allreduces()
 
     # --- gather path (cat / none / custom / autograd) ---------------------
