@@ -23,16 +23,18 @@ from metrics_amd import Metric
 class BigSum(Metric):
     full_state_update = False
 
-    def __init__(self, n=256_000_000):
+    def __init__(self, n=64_000_000):
         super().__init__()
         self.n = n
         self.add_state("x", torch.zeros(n), dist_reduce_fx="sum")
 
     def update(self, v):
-        # a few real kernels on the default stream
-        self.x = self.x + v
-        self.x = self.x * 1.0000001
-        self.x = self.x + 0.5 * v
+        # a few real kernels on the default stream; IN-PLACE so the trace has
+        # no 1-GB allocations (cross-stream block reuse makes the caching
+        # allocator synchronize, which would mask the overlap being measured)
+        self.x.add_(v)
+        self.x.mul_(1.0000001)
+        self.x.add_(v, alpha=0.5)
 
     def compute(self):
         return self.x.sum()
