@@ -9,7 +9,12 @@ from typing import List, Tuple, Union
 import torch
 from torch import Tensor, tensor
 
-from metrics_amd.functional.text.helper import _edit_distance, _edit_distance_counts
+from metrics_amd.functional.text.helper import (
+    _edit_distance,
+    _edit_distance_batch,
+    _edit_distance_counts,
+    _edit_distance_counts_batch,
+)
 
 
 def _norm_inputs(preds: Union[str, List[str]], target: Union[str, List[str]]) -> Tuple[List[str], List[str]]:
@@ -24,11 +29,10 @@ def _cer_update(preds, target) -> Tuple[Tensor, Tensor]:
     preds, target = _norm_inputs(preds, target)
     errors = tensor(0, dtype=torch.float)
     total = tensor(0, dtype=torch.float)
-    for p, t in zip(preds, target):
-        p_tokens = p
-        t_tokens = t
-        errors += _edit_distance(list(p_tokens), list(t_tokens))
-        total += len(t_tokens)
+    dists = _edit_distance_batch([(list(p), list(t)) for p, t in zip(preds, target)])
+    for d, t in zip(dists, target):
+        errors += d
+        total += len(t)
     return errors, total
 
 
@@ -42,10 +46,10 @@ def _wer_update(preds, target) -> Tuple[Tensor, Tensor]:
     preds, target = _norm_inputs(preds, target)
     errors = tensor(0, dtype=torch.float)
     total = tensor(0, dtype=torch.float)
-    for p, t in zip(preds, target):
-        p_tokens = p.split()
-        t_tokens = t.split()
-        errors += _edit_distance(p_tokens, t_tokens)
+    pairs = [(p.split(), t.split()) for p, t in zip(preds, target)]
+    dists = _edit_distance_batch(pairs)
+    for d, (_, t_tokens) in zip(dists, pairs):
+        errors += d
         total += len(t_tokens)
     return errors, total
 
@@ -63,10 +67,8 @@ def _mer_wil_wip_update(preds, target) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
     total = tensor(0, dtype=torch.float)
     target_total = tensor(0, dtype=torch.float)
     preds_total = tensor(0, dtype=torch.float)
-    for p, t in zip(preds, target):
-        p_tokens = p.split()
-        t_tokens = t.split()
-        subs, ins, dels, hits = _edit_distance_counts(p_tokens, t_tokens)
+    pairs = [(p.split(), t.split()) for p, t in zip(preds, target)]
+    for (p_tokens, t_tokens), (subs, ins, dels, hits) in zip(pairs, _edit_distance_counts_batch(pairs)):
         errors += subs + ins + dels
         total += hits + subs + ins + dels
         target_total += len(t_tokens)
@@ -91,10 +93,8 @@ def word_information_preserved(preds, target) -> Tensor:
     total_hits = tensor(0, dtype=torch.float)
     target_total = tensor(0, dtype=torch.float)
     preds_total = tensor(0, dtype=torch.float)
-    for p, t in zip(preds, target):
-        p_tokens = p.split()
-        t_tokens = t.split()
-        _, _, _, hits = _edit_distance_counts(p_tokens, t_tokens)
+    pairs = [(p.split(), t.split()) for p, t in zip(preds, target)]
+    for (p_tokens, t_tokens), (_, _, _, hits) in zip(pairs, _edit_distance_counts_batch(pairs)):
         total_hits += hits
         target_total += len(t_tokens)
         preds_total += len(p_tokens)
@@ -120,7 +120,7 @@ def edit_distance(preds, target, substitution_cost: int = 1, reduction: str = "m
 
         vals = [dist(p, t) for p, t in zip(preds, target)]
     else:
-        vals = [_edit_distance(list(p), list(t)) for p, t in zip(preds, target)]
+        vals = _edit_distance_batch([(list(p), list(t)) for p, t in zip(preds, target)])
     res = torch.tensor(vals, dtype=torch.int32)
     if reduction == "mean":
         return res.float().mean()

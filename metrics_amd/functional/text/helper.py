@@ -8,6 +8,78 @@ import re
 from typing import List, Sequence, Tuple, Union
 
 
+def _intern_pairs(pairs):
+    """Map token sequences to int32 id arrays + offsets for the native batch."""
+    import numpy as np
+
+    ids: dict = {}
+
+    def _to_ids(seq):
+        out = np.empty(len(seq), dtype=np.int32)
+        for i, tok in enumerate(seq):
+            out[i] = ids.setdefault(tok, len(ids))
+        return out
+
+    a_arrs = [_to_ids(p) for p, _ in pairs]
+    b_arrs = [_to_ids(t) for _, t in pairs]
+    off_a = np.zeros(len(pairs) + 1, dtype=np.int64)
+    off_b = np.zeros(len(pairs) + 1, dtype=np.int64)
+    off_a[1:] = np.cumsum([len(x) for x in a_arrs])
+    off_b[1:] = np.cumsum([len(x) for x in b_arrs])
+    cat = lambda arrs: np.concatenate(arrs) if arrs and off_a is not None else np.zeros(0, np.int32)
+    tok_a = np.concatenate(a_arrs) if a_arrs else np.zeros(0, np.int32)
+    tok_b = np.concatenate(b_arrs) if b_arrs else np.zeros(0, np.int32)
+    return tok_a.astype(np.int32), off_a, tok_b.astype(np.int32), off_b
+
+
+def _edit_distance_batch(pairs) -> list:
+    """Levenshtein distances for a batch of (pred_tokens, ref_tokens) pairs.
+
+    Uses the OpenMP native kernel (csrc/edit_distance.cpp) when the CPU
+    library is built; falls back to the Python DP otherwise.
+    """
+    from metrics_amd.ops import _coco
+
+    if not pairs:
+        return []
+    lib = _coco._load()
+    if lib is None:
+        return [_edit_distance(list(p), list(t)) for p, t in pairs]
+    import ctypes
+
+    import numpy as np
+
+    tok_a, off_a, tok_b, off_b = _intern_pairs(pairs)
+    out = np.zeros(len(pairs), dtype=np.int64)
+    lib.ma_edit_distance_batch(
+        _coco._ptr(tok_a), _coco._ptr(off_a), _coco._ptr(tok_b), _coco._ptr(off_b),
+        ctypes.c_int64(len(pairs)), ctypes.c_int(0), _coco._ptr(out),
+    )
+    return out.tolist()
+
+
+def _edit_distance_counts_batch(pairs) -> list:
+    """(subs, ins, dels, hits) per pair, native when available."""
+    from metrics_amd.ops import _coco
+
+    if not pairs:
+        return []
+    lib = _coco._load()
+    if lib is None:
+        return [_edit_distance_counts(p, t) for p, t in pairs]
+    import ctypes
+
+    import numpy as np
+
+    tok_a, off_a, tok_b, off_b = _intern_pairs(pairs)
+    out = np.zeros(4 * len(pairs), dtype=np.int64)
+    lib.ma_edit_distance_batch(
+        _coco._ptr(tok_a), _coco._ptr(off_a), _coco._ptr(tok_b), _coco._ptr(off_b),
+        ctypes.c_int64(len(pairs)), ctypes.c_int(1), _coco._ptr(out),
+    )
+    return [tuple(out[4 * i : 4 * i + 4].tolist()) for i in range(len(pairs))]
+
+
 def _edit_distance(prediction_tokens: Sequence, reference_tokens: Sequence) -> int:
     """Levenshtein distance between two token sequences (O(nm) DP, two rows)."""
     n, m = len(prediction_tokens), len(reference_tokens)

@@ -79,7 +79,12 @@ def gather_all_tensors(result: Tensor, group: Optional[Any] = None) -> List[Tens
     # collectives need contiguous buffers
     result = result.contiguous()
     world_size = dist.get_world_size(group)
-    dist.barrier(group=group)
+    if not _use_side_stream(group):
+        # the reference barriers before every gather; on RCCL the collectives
+        # are stream-ordered and all_gather is itself a synchronization point,
+        # so the barrier is a pure extra collective — keep it only for
+        # host-blocking backends (gloo), where it preserves reference timing
+        dist.barrier(group=group)
 
     if result.ndim == 0:
         return _simple_gather_all_tensors(result, group, world_size)
