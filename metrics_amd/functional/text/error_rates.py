@@ -29,7 +29,7 @@ def _cer_update(preds, target) -> Tuple[Tensor, Tensor]:
     preds, target = _norm_inputs(preds, target)
     errors = tensor(0, dtype=torch.float)
     total = tensor(0, dtype=torch.float)
-    dists = _edit_distance_batch([(list(p), list(t)) for p, t in zip(preds, target)])
+    dists = _edit_distance_batch(list(zip(preds, target)))  # str pairs: utf-32 fast path
     for d, t in zip(dists, target):
         errors += d
         total += len(t)
@@ -120,7 +120,7 @@ def edit_distance(preds, target, substitution_cost: int = 1, reduction: str = "m
 
         vals = [dist(p, t) for p, t in zip(preds, target)]
     else:
-        vals = _edit_distance_batch([(list(p), list(t)) for p, t in zip(preds, target)])
+        vals = _edit_distance_batch(list(zip(preds, target)))  # str pairs: utf-32 fast path
     res = torch.tensor(vals, dtype=torch.int32)
     if reduction == "mean":
         return res.float().mean()

@@ -9,12 +9,18 @@ from typing import List, Sequence, Tuple, Union
 
 
 def _intern_pairs(pairs):
-    """Map token sequences to int32 id arrays + offsets for the native batch."""
+    """Map token sequences to int32 id arrays + offsets for the native batch.
+
+    Plain strings (char-level CER/EditDistance) skip interning: their
+    utf-32-le byte view IS the int32 codepoint array.
+    """
     import numpy as np
 
     ids: dict = {}
 
     def _to_ids(seq):
+        if isinstance(seq, str):
+            return np.frombuffer(seq.encode("utf-32-le"), dtype=np.int32)
         out = np.empty(len(seq), dtype=np.int32)
         for i, tok in enumerate(seq):
             out[i] = ids.setdefault(tok, len(ids))
