@@ -38,7 +38,9 @@ def _normalized_root_mean_squared_error_compute(
     sum_squared_error: Tensor, num_obs: Union[int, Tensor], denom: Tensor
 ) -> Tensor:
     rmse = torch.sqrt(sum_squared_error / num_obs)
-    return rmse / torch.abs(denom)
+    # the reference divides by the RAW statistic: a negative target mean
+    # legitimately yields a negative score
+    return rmse / denom
 
 
 def normalized_root_mean_squared_error(

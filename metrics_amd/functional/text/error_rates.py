@@ -61,16 +61,21 @@ def word_error_rate(preds, target) -> Tensor:
 
 
 def _mer_wil_wip_update(preds, target) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
-    """Return (errors, total_for_mer, target_total, preds_total) using hit counts."""
+    """Return (Σ edit distance, Σ max(len_t, len_p), Σ len_t, Σ len_p).
+
+    Reference formulation (functional/text/{mer,wip}.py): the denominator is
+    the max sentence length, NOT backtrace hit counts — the two differ when a
+    pair needs both insertions and deletions.
+    """
     preds, target = _norm_inputs(preds, target)
     errors = tensor(0, dtype=torch.float)
     total = tensor(0, dtype=torch.float)
     target_total = tensor(0, dtype=torch.float)
     preds_total = tensor(0, dtype=torch.float)
     pairs = [(p.split(), t.split()) for p, t in zip(preds, target)]
-    for (p_tokens, t_tokens), (subs, ins, dels, hits) in zip(pairs, _edit_distance_counts_batch(pairs)):
-        errors += subs + ins + dels
-        total += hits + subs + ins + dels
+    for (p_tokens, t_tokens), d in zip(pairs, _edit_distance_batch(pairs)):
+        errors += d
+        total += max(len(t_tokens), len(p_tokens))
         target_total += len(t_tokens)
         preds_total += len(p_tokens)
     return errors, total, target_total, preds_total
@@ -88,17 +93,10 @@ def word_information_lost(preds, target) -> Tensor:
 
 
 def word_information_preserved(preds, target) -> Tensor:
-    """Word information preserved = hits/len(t) * hits/len(p)."""
-    preds, target = _norm_inputs(preds, target)
-    total_hits = tensor(0, dtype=torch.float)
-    target_total = tensor(0, dtype=torch.float)
-    preds_total = tensor(0, dtype=torch.float)
-    pairs = [(p.split(), t.split()) for p, t in zip(preds, target)]
-    for (p_tokens, t_tokens), (_, _, _, hits) in zip(pairs, _edit_distance_counts_batch(pairs)):
-        total_hits += hits
-        target_total += len(t_tokens)
-        preds_total += len(p_tokens)
-    return total_hits / target_total * (total_hits / preds_total)
+    """Word information preserved = ((total - errors)/len_t) * ((total - errors)/len_p)."""
+    errors, total, target_total, preds_total = _mer_wil_wip_update(preds, target)
+    kept = total - errors
+    return kept / target_total * (kept / preds_total)
 
 
 def edit_distance(preds, target, substitution_cost: int = 1, reduction: str = "mean") -> Tensor:
