@@ -215,35 +215,62 @@ def _chrf_f_score(tp: Tensor, fp: Tensor, fn: Tensor, beta: float) -> Tensor:
 
 
 # ------------------------------------------------------------------------- TER
-def _ter_shifts(pred: List[str], ref: List[str]) -> int:
-    """tercom-style greedy shifts + edit distance."""
-    edits = 0
-    pred = list(pred)
-    while True:
-        base = _edit_distance(pred, ref)
-        best_delta = 0
-        best_state = None
-        # try shifting each matching sub-span of pred to a new position
-        n = len(pred)
-        for start in range(n):
-            for length in range(1, min(n - start, 10) + 1):
-                span = pred[start : start + length]
-                # candidate target positions: where span occurs in ref
-                for rpos in range(len(ref) - length + 1):
-                    if ref[rpos : rpos + length] != span:
-                        continue
-                    rest = pred[:start] + pred[start + length :]
-                    ins = min(rpos, len(rest))
-                    cand = rest[:ins] + span + rest[ins:]
-                    d = _edit_distance(cand, ref)
-                    delta = base - d - 1  # shift costs 1 edit
-                    if delta > best_delta:
-                        best_delta = delta
-                        best_state = cand
-        if best_state is None:
-            return edits + base
-        edits += 1
-        pred = best_state
+def _tercom_normalize(sentence: str, asian_support: bool) -> str:
+    """Tercom's general/western normalization (Normalizer.java rules)."""
+    import re as _re
+
+    sentence = f" {sentence} "
+    rules = [
+        (r"\n-", ""),
+        (r"\n", " "),
+        (r"&quot;", '"'),
+        (r"&amp;", "&"),
+        (r"&lt;", "<"),
+        (r"&gt;", ">"),
+        (r"([{-~[-` -&(-+:-@/])", r" \1 "),
+        (r"'s ", r" 's "),
+        (r"'s$", r" 's"),
+        (r"([^0-9])([\.,])", r"\1 \2 "),
+        (r"([\.,])([^0-9])", r" \1 \2"),
+        (r"([0-9])(-)", r"\1 \2 "),
+    ]
+    for pat, repl in rules:
+        sentence = _re.sub(pat, repl, sentence)
+    if asian_support:
+        for pat in (
+            r"([一-鿿㐀-䶿])",
+            r"([㇀-㇯⺀-⻿])",
+            r"([㌀-㏿豈-﫿︰-﹏])",
+            r"([㈀-㼢])",
+            _ASIAN_PUNCT,
+            _FULLWIDTH_PUNCT,
+        ):
+            sentence = _re.sub(pat, r" \1 ", sentence)
+    return sentence
+
+
+_ASIAN_PUNCT = r"([、。〈-】〔-〟｡-･・])"
+_FULLWIDTH_PUNCT = r"([．，？：；！＂（）])"
+
+
+def _tercom_tokenize(
+    sentence: str, normalize: bool, no_punctuation: bool, lowercase: bool, asian_support: bool
+) -> List[str]:
+    import re as _re
+
+    sentence = sentence.rstrip()
+    if not sentence:
+        return []
+    if lowercase:
+        sentence = sentence.lower()
+    if normalize:
+        sentence = _tercom_normalize(sentence, asian_support)
+    if no_punctuation:
+        sentence = _re.sub(r"[\.,\?:;!\"\(\)]", "", sentence)
+        if asian_support:
+            sentence = _re.sub(_ASIAN_PUNCT, "", sentence)
+            sentence = _re.sub(_FULLWIDTH_PUNCT, "", sentence)
+    return " ".join(sentence.split()).split()
 
 
 def translation_edit_rate(
@@ -255,37 +282,31 @@ def translation_edit_rate(
     asian_support: bool = False,
     return_sentence_level_score: bool = False,
 ):
-    """TER: shifts + edits over average reference length."""
-    import re as _re
+    """TER: tercom shifts + edits over average reference length (ter_core)."""
+    from metrics_amd.functional.text.ter_core import sentence_ter
 
     preds_ = [preds] if isinstance(preds, str) else preds
     target_ = [[t] if isinstance(t, str) else t for t in target]
 
-    def _norm(s: str) -> List[str]:
-        if lowercase:
-            s = s.lower()
-        if no_punctuation:
-            s = _re.sub(r"[\.,\?:;!\"\(\)]", "", s)
-        if normalize:
-            s = _re.sub(r"([\.,\?:;!\"\(\)])", r" \1 ", s)
-        return s.split()
+    def _tok(s: str) -> List[str]:
+        return _tercom_tokenize(s, normalize, no_punctuation, lowercase, asian_support)
 
-    total_edits = tensor(0.0)
-    total_len = tensor(0.0)
+    def _score(edits: float, avg_len: float) -> Tensor:
+        if avg_len > 0 and edits > 0:
+            return tensor(edits / avg_len)
+        if avg_len == 0 and edits > 0:
+            return tensor(1.0)
+        return tensor(0.0)
+
+    total_edits = 0.0
+    total_len = 0.0
     sent_scores = []
     for p, refs in zip(preds_, target_):
-        p_tok = _norm(p)
-        best = None
-        avg_len = sum(len(_norm(r)) for r in refs) / len(refs)
-        for r in refs:
-            r_tok = _norm(r)
-            e = _ter_shifts(p_tok, r_tok)
-            if best is None or e < best:
-                best = e
+        best, avg_len = sentence_ter(_tok(p), [_tok(r) for r in refs])
         total_edits += best
         total_len += avg_len
-        sent_scores.append(tensor(best / avg_len if avg_len > 0 else (0.0 if best == 0 else 1.0)))
-    score = total_edits / total_len if total_len > 0 else tensor(0.0)
+        sent_scores.append(_score(best, avg_len))
+    score = _score(total_edits, total_len)
     if return_sentence_level_score:
         return score, torch.stack(sent_scores)
     return score

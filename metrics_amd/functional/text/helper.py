@@ -143,38 +143,97 @@ _13A_RE_NUM2 = re.compile(r"([\.,])([0-9])")
 _13A_RE_DASH = re.compile(r"([0-9])(-)")
 
 
+# sacrebleu's mteval-equivalent post-processing rules
+_SB_REGEX = (
+    (re.compile(r"([\{-\~\[-\` -\&\(-\+\:-\@\/])"), r" \1 "),   # symbols/punct blocks
+    (re.compile(r"([^0-9])([\.,])"), r"\1 \2 "),                 # . , not after digit
+    (re.compile(r"([\.,])([^0-9])"), r" \1 \2"),                 # . , not before digit
+    (re.compile(r"([0-9])(-)"), r"\1 \2 "),                      # dash after digit
+)
+
+# CJK blocks the zh tokenizer pads (sacrebleu's published range table)
+_CJK_RANGES = (
+    ("㐀", "䶵"), ("一", "龥"), ("龦", "龻"),
+    ("豈", "鶴"), ("侮", "頻"), ("並", "龎"),
+    ("\U00020000", "\U0002a6d6"), ("\U0002f800", "\U0002fa1d"),
+    ("＀", "￯"), ("⺀", "⻿"), ("　", "〿"),
+    ("㇀", "㇯"), ("⼀", "⿟"), ("⿰", "⿿"),
+    ("㄀", "ㄯ"), ("ㆠ", "ㆿ"), ("︐", "︟"),
+    ("︰", "﹏"), ("☀", "⛿"), ("✀", "➿"),
+    ("㈀", "㋿"), ("㌀", "㏿"),
+)
+
+
+def _sb_regex_post(line: str) -> str:
+    for rx, repl in _SB_REGEX:
+        line = rx.sub(repl, line)
+    return " ".join(line.split())
+
+
+def _finish(line: str, lowercase: bool) -> List[str]:
+    # lowercase applies AFTER tokenization (reference __call__ order)
+    return (line.lower() if lowercase else line).split()
+
+
 def _tokenize_13a(line: str, lowercase: bool = False) -> List[str]:
     """mteval-v13a tokenizer (sacrebleu default)."""
     line = line.replace("<skipped>", "")
     line = line.replace("-\n", "")
     line = line.replace("\n", " ")
-    line = line.replace("&quot;", '"').replace("&amp;", "&").replace("&lt;", "<").replace("&gt;", ">")
-    line = f" {line} "
-    line = _13A_RE_NUM.sub(r"\1 \2 ", line)
-    line = _13A_RE_NUM2.sub(r" \1 \2", line)
-    line = _13A_RE_DASH.sub(r"\1 \2 ", line)
-    line = _13A_RE1.sub(r" \1 ", line)
-    if lowercase:
-        line = line.lower()
-    return line.split()
+    if "&" in line:
+        line = line.replace("&quot;", '"').replace("&amp;", "&").replace("&lt;", "<").replace("&gt;", ">")
+    return _finish(_sb_regex_post(f" {line} "), lowercase)
+
+
+def _tokenize_intl(line: str, lowercase: bool = False) -> List[str]:
+    """International tokenizer: unicode punctuation/symbol splitting (regex module)."""
+    import regex
+
+    rules = getattr(_tokenize_intl, "_rules", None)
+    if rules is None:
+        rules = (
+            (regex.compile(r"(\P{N})(\p{P})"), r"\1 \2 "),
+            (regex.compile(r"(\p{P})(\P{N})"), r" \1 \2"),
+            (regex.compile(r"(\p{S})"), r" \1 "),
+        )
+        _tokenize_intl._rules = rules
+    for rx, repl in rules:
+        line = rx.sub(repl, line)
+    return _finish(line, lowercase)
 
 
 def _tokenize_char(line: str, lowercase: bool = False) -> List[str]:
-    if lowercase:
-        line = line.lower()
-    return list(line.strip())
+    # every non-whitespace character is a token (spaces become separators)
+    return _finish(" ".join(ch for ch in line), lowercase)
+
+
+def _tokenize_zh(line: str, lowercase: bool = False) -> List[str]:
+    def is_cjk(ch: str) -> bool:
+        return any(lo <= ch <= hi for lo, hi in _CJK_RANGES)
+
+    line = line.strip()
+    padded = "".join(f" {ch} " if is_cjk(ch) else ch for ch in line)
+    return _finish(_sb_regex_post(padded), lowercase)
 
 
 def _tokenize_none(line: str, lowercase: bool = False) -> List[str]:
-    if lowercase:
-        line = line.lower()
-    return line.strip().split()
+    return _finish(line, lowercase)
 
 
-_TOKENIZERS = {"13a": _tokenize_13a, "char": _tokenize_char, "none": _tokenize_none, "intl": _tokenize_13a}
+_TOKENIZERS = {
+    "13a": _tokenize_13a,
+    "char": _tokenize_char,
+    "none": _tokenize_none,
+    "intl": _tokenize_intl,
+    "zh": _tokenize_zh,
+}
 
 
 def get_tokenizer(name: str):
+    if name in ("ja-mecab", "ko-mecab", "flores101", "flores200"):
+        raise ModuleNotFoundError(
+            f"Tokenizer `{name}` needs an external model/dependency that is not available offline"
+        )
     if name not in _TOKENIZERS:
         raise ValueError(f"Unsupported tokenizer {name}; expected one of {list(_TOKENIZERS)}")
     return _TOKENIZERS[name]

@@ -289,8 +289,7 @@ def _resolve(pkg_name, name):
     return None
 
 
-@pytest.mark.parametrize(("name", "gen", "kwargs", "atol"), CASES, ids=[c[0] + str(i) for i, c in enumerate(CASES)])
-def test_ref_differential(name, gen, kwargs, atol):
+def _run_case(name, gen, kwargs, atol):
     _ref_functional()
     ours = _resolve("metrics_amd.functional", name)
     ref = _resolve("torchmetrics.functional", name)
@@ -304,3 +303,81 @@ def test_ref_differential(name, gen, kwargs, atol):
             pytest.skip(f"reference needs optional dep: {err}")
         got = ours(*args, **kwargs)
         _cmp(got, expected, atol)
+
+
+@pytest.mark.parametrize(("name", "gen", "kwargs", "atol"), CASES, ids=[c[0] + str(i) for i, c in enumerate(CASES)])
+def test_ref_differential(name, gen, kwargs, atol):
+    _run_case(name, gen, kwargs, atol)
+
+
+# ---------------------------------------------------------------- wave 2
+def _panoptic(seed):
+    g = _g(seed)
+    # (B, H, W, 2): category id + instance id
+    cats = torch.tensor([0, 1, 2])[torch.randint(0, 3, (2, 12, 12), generator=g)]
+    inst = torch.randint(0, 3, (2, 12, 12), generator=g)
+    p = torch.stack([cats, inst], dim=-1)
+    cats2 = torch.tensor([0, 1, 2])[torch.randint(0, 3, (2, 12, 12), generator=g)]
+    inst2 = torch.randint(0, 3, (2, 12, 12), generator=g)
+    t = torch.stack([cats2, inst2], dim=-1)
+    return p, t
+
+
+CASES2 = [
+    # at-fixed / logauc family (built on the K2/bucketized curve cores)
+    ("binary_recall_at_fixed_precision", _bin_probs, {"min_precision": 0.5, "thresholds": None}, 1e-6),
+    ("binary_recall_at_fixed_precision", _bin_probs, {"min_precision": 0.5, "thresholds": 50}, 1e-6),
+    ("binary_precision_at_fixed_recall", _bin_probs, {"min_recall": 0.5, "thresholds": None}, 1e-6),
+    ("binary_sensitivity_at_specificity", _bin_probs, {"min_specificity": 0.6, "thresholds": None}, 1e-6),
+    ("binary_specificity_at_sensitivity", _bin_probs, {"min_sensitivity": 0.6, "thresholds": None}, 1e-6),
+    ("binary_logauc", _bin_probs, {}, 1e-5),
+    # curve outputs (tuple/list results)
+    ("multiclass_roc", lambda s: (_mc_logits(s)[0].softmax(-1), _mc_logits(s)[1]), {"num_classes": 7, "thresholds": None}, 1e-6),
+    ("multiclass_roc", lambda s: (_mc_logits(s)[0].softmax(-1), _mc_logits(s)[1]), {"num_classes": 7, "thresholds": 25}, 1e-6),
+    ("multiclass_precision_recall_curve", lambda s: (_mc_logits(s)[0].softmax(-1), _mc_logits(s)[1]), {"num_classes": 7, "thresholds": 25}, 1e-6),
+    ("multilabel_roc", _ml_probs, {"num_labels": 4, "thresholds": 25}, 1e-6),
+    ("multiclass_auroc", lambda s: (_mc_logits(s)[0].softmax(-1), _mc_logits(s)[1]), {"num_classes": 7, "average": "weighted", "thresholds": None}, 1e-5),
+    ("multilabel_auroc", _ml_probs, {"num_labels": 4, "average": "macro", "thresholds": None}, 1e-5),
+    # average variants on prf
+    ("multiclass_f1_score", _mc_logits, {"num_classes": 7, "average": "weighted"}, 1e-6),
+    ("multiclass_precision", _mc_logits, {"num_classes": 7, "average": None}, 1e-6),
+    ("multiclass_recall", _mc_logits, {"num_classes": 7, "top_k": 2, "average": "macro"}, 1e-6),
+    ("multiclass_stat_scores", _mc_logits, {"num_classes": 7, "average": None}, 1e-6),
+    ("multiclass_stat_scores", lambda s: _mc_multidim(s), {"num_classes": 7, "average": None, "multidim_average": "samplewise"}, 1e-6),
+    # fairness
+    ("demographic_parity", lambda s: (_bin_probs(s)[0], torch.randint(0, 2, (B * 4,), generator=_g(s + 9))), {}, 1e-6),
+    ("equal_opportunity", lambda s: (*_bin_probs(s), torch.randint(0, 2, (B * 4,), generator=_g(s + 9))), {}, 1e-6),
+    # segmentation / detection extras
+    ("panoptic_quality", _panoptic, {"things": {0, 1}, "stuffs": {2}}, 1e-5),
+    ("modified_panoptic_quality", _panoptic, {"things": {0, 1}, "stuffs": {2}}, 1e-5),
+    # text n-gram / structured
+    ("sacre_bleu_score", _text_pair, {}, 1e-6),
+    ("sacre_bleu_score", _text_pair, {"tokenize": "char", "lowercase": True}, 1e-6),
+    ("chrf_score", _text_pair, {}, 1e-5),
+    ("chrf_score", _text_pair, {"n_word_order": 2, "return_sentence_level_score": True}, 1e-5),
+    ("translation_edit_rate", _text_pair, {}, 1e-5),
+    ("translation_edit_rate", _text_pair, {"normalize": True, "lowercase": False}, 1e-5),
+    ("extended_edit_distance", _text_pair, {}, 1e-5),
+    ("bleu_score", _text_pair, {"n_gram": 2, "smooth": True}, 1e-6),
+    ("word_error_rate", lambda s: (_text_pair(s)[0], [t[0] for t in _text_pair(s)[1]]), {}, 1e-6),
+    # squad-format dict inputs
+    (
+        "squad",
+        lambda s: (
+            [{"prediction_text": "the cat sat", "id": "q1"}, {"prediction_text": "blue sky", "id": "q2"}],
+            [
+                {"answers": {"answer_start": [0], "text": ["the cat sat on the mat"]}, "id": "q1"},
+                {"answers": {"answer_start": [0], "text": ["blue sky"]}, "id": "q2"},
+            ],
+        ),
+        {},
+        1e-6,
+    ),
+]
+
+
+@pytest.mark.parametrize(
+    ("name", "gen", "kwargs", "atol"), CASES2, ids=[c[0] + "_w2_" + str(i) for i, c in enumerate(CASES2)]
+)
+def test_ref_differential_wave2(name, gen, kwargs, atol):
+    _run_case(name, gen, kwargs, atol)
