@@ -57,7 +57,9 @@ def universal_image_quality_index(
     eps = torch.finfo(preds.dtype).eps
     uqi_idx = ((2 * mu_pred_target) * upper) / ((mu_pred_sq + mu_target_sq) * lower + eps)
     uqi_idx = uqi_idx[..., pad_h:-pad_h, pad_w:-pad_w]
-    return reduce(uqi_idx.reshape(uqi_idx.shape[0], -1).mean(-1), reduction or "none")
+    # the reduction applies to the raw quality MAP (reference uqi.py):
+    # 'none' returns the (B,C,H',W') map, 'sum' sums every element
+    return reduce(uqi_idx, reduction or "none")
 
 
 def spectral_angle_mapper(

@@ -228,9 +228,11 @@ class UniversalImageQualityIndex(_ScoreAverageMetric):
         self.reduction = reduction
 
     def update(self, preds: Tensor, target: Tensor) -> None:
-        """Accumulate per-image UQI."""
-        score = universal_image_quality_index(preds, target, self.kernel_size, self.sigma, None)
-        self._accumulate(score, preds.shape[0])
+        """Accumulate the summed quality map + element count (matches the
+        reference's sum_uqi/numel states now that the functional's 'none'
+        reduction returns the raw map)."""
+        score_map = universal_image_quality_index(preds, target, self.kernel_size, self.sigma, None)
+        self._accumulate(score_map.sum().unsqueeze(0), score_map.numel())
 
 
 class SpectralAngleMapper(_ScoreAverageMetric):

@@ -381,3 +381,59 @@ CASES2 = [
 )
 def test_ref_differential_wave2(name, gen, kwargs, atol):
     _run_case(name, gen, kwargs, atol)
+
+
+# ---------------------------------------------------------------- wave 3
+def _pit_inputs(seed):
+    g = _g(seed)
+    return torch.randn(3, 2, 1000, generator=g), torch.randn(3, 2, 1000, generator=g)
+
+
+def _pansharpen(seed):
+    g = _g(seed)
+    return torch.rand(2, 3, 32, 32, generator=g), torch.rand(2, 3, 16, 16, generator=g)
+
+
+CASES3 = [
+    ("mean_squared_error", lambda s: _reg_pair(s, (128, 3)), {"num_outputs": 3}, 1e-5),
+    ("pearson_corrcoef", lambda s: _reg_pair(s, (128, 3)), {}, 1e-5),
+    ("spearman_corrcoef", lambda s: _reg_pair(s, (128, 3)), {}, 1e-5),
+    ("r2_score", lambda s: _reg_pair(s, (128, 3)), {"multioutput": "variance_weighted"}, 1e-5),
+    ("explained_variance", lambda s: _reg_pair(s, (128, 3)), {"multioutput": "raw_values"}, 1e-5),
+    ("binary_stat_scores", lambda s: (torch.rand(4, 2, 32, generator=_g(s)), torch.randint(0, 2, (4, 2, 32), generator=_g(s + 1))), {"multidim_average": "samplewise"}, 1e-6),
+    ("multilabel_stat_scores", _ml_probs, {"num_labels": 4, "average": None}, 1e-6),
+    ("multilabel_exact_match", _ml_probs, {"num_labels": 4}, 1e-6),
+    ("peak_signal_noise_ratio", _img_pair, {"data_range": (0.1, 0.8)}, 1e-4),
+    ("peak_signal_noise_ratio", _img_pair, {"reduction": "none", "dim": (1, 2, 3), "data_range": 1.0}, 1e-4),
+    ("spectral_distortion_index", _pansharpen, {}, 1e-4),
+    ("universal_image_quality_index", _img_pair, {"reduction": "sum"}, 1e-4),
+    ("spectral_angle_mapper", _img_pair, {"reduction": "none"}, 1e-5),
+    ("permutation_invariant_training", _pit_inputs, {"metric_func": None}, 1e-4),
+    ("signal_noise_ratio", _audio_pair, {"zero_mean": True}, 1e-4),
+    ("scale_invariant_signal_distortion_ratio", _audio_pair, {"zero_mean": True}, 1e-4),
+    ("retrieval_precision_recall_curve", lambda s: _retrieval(s)[:2], {"max_k": 10}, 1e-6),
+    ("chrf_score", _text_pair, {"n_char_order": 4, "beta": 1.0}, 1e-5),
+    ("bleu_score", _text_pair, {"n_gram": 3}, 1e-6),
+]
+
+
+@pytest.mark.parametrize(
+    ("name", "gen", "kwargs", "atol"), CASES3, ids=[c[0] + "_w3_" + str(i) for i, c in enumerate(CASES3)]
+)
+def test_ref_differential_wave3(name, gen, kwargs, atol):
+    if name == "permutation_invariant_training":
+        # needs a callable metric_func on both sides
+        from metrics_amd.functional.audio import scale_invariant_signal_noise_ratio as our_m
+        import torchmetrics.functional.audio as ref_a
+
+        import metrics_amd.functional.audio as our_a
+
+        p, t = gen(3)
+        got = our_a.permutation_invariant_training(p, t, our_m, "speaker-wise", "max")
+        exp = ref_a.permutation_invariant_training(
+            p, t, ref_a.scale_invariant_signal_noise_ratio, "speaker-wise", "max"
+        )
+        _cmp(got[0], exp[0], atol)
+        _cmp(got[1], exp[1], atol)
+        return
+    _run_case(name, gen, kwargs, atol)
