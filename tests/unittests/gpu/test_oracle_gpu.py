@@ -135,9 +135,10 @@ def test_gpu_calibration_vs_manual_numpy():
     target = torch.randint(0, 2, (B,), generator=g)
     m = ma.BinaryCalibrationError(n_bins=15, norm="l1").to("cuda")
     m.update(conf.cuda(), target.cuda())
-    # independent numpy ECE
-    c = torch.where(target == 1, conf, 1 - conf).numpy()
-    acc = ((conf.numpy() > 0.5).astype(int) == target.numpy()).astype(float)
+    # independent numpy ECE (reference convention: confidences are the raw
+    # class-1 probabilities, accuracies the 0/1 targets)
+    c = conf.numpy()
+    acc = target.numpy().astype(float)
     edges = np.linspace(0, 1, 16)
     idx = np.clip(np.searchsorted(edges, c, side="right") - 1, 0, 14)
     ece = 0.0
