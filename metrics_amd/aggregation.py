@@ -51,13 +51,17 @@ class BaseAggregator(Metric):
         if weight is not None and not isinstance(weight, Tensor):
             weight = torch.as_tensor(weight, dtype=self.dtype, device=self.device)
 
+        if weight is None:
+            weight = torch.ones_like(x)
+            weight_was_expanded = False
+        else:
+            # scalar/broadcastable weights line up elementwise with the values
+            weight_was_expanded = weight.shape != x.shape
+            weight = torch.broadcast_to(weight, x.shape)
+
         if self.nan_strategy != "disable":
             nans = torch.isnan(x)
-            if weight is not None:
-                nans_weight = torch.isnan(weight)
-            else:
-                nans_weight = torch.zeros_like(nans)
-                weight = torch.ones_like(x)
+            nans_weight = torch.isnan(weight)
             anynan = nans.any() or nans_weight.any()
             if anynan:
                 if self.nan_strategy == "error":
@@ -70,10 +74,17 @@ class BaseAggregator(Metric):
                 else:  # float strategy
                     if not isinstance(self.nan_strategy, float):
                         raise ValueError(f"`nan_strategy` shall be float but you pass {self.nan_strategy}")
-                    x[nans | nans_weight] = self.nan_strategy
-                    weight[nans | nans_weight] = self.nan_strategy
-        else:
-            weight = torch.ones_like(x) if weight is None else weight
+                    mask = nans | nans_weight
+                    x = x.clone()
+                    x[mask] = self.nan_strategy
+                    if weight_was_expanded:
+                        # reference parity: it writes through the broadcast
+                        # VIEW, so every aliased element of an expanded weight
+                        # takes the fill value — any nan floods the weight
+                        weight = torch.full_like(x, self.nan_strategy)
+                    else:
+                        weight = weight.clone()
+                        weight[mask] = self.nan_strategy
 
         return x.to(self.dtype), weight.to(self.dtype)
 

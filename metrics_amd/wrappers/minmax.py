@@ -24,8 +24,12 @@ class MinMaxMetric(WrapperMetric):
                 f"Expected base metric to be an instance of `metrics_amd.Metric` but received {base_metric}"
             )
         self._base_metric = base_metric
-        self.add_state("min_val", default=torch.tensor(float("inf")), dist_reduce_fx="min")
-        self.add_state("max_val", default=torch.tensor(float("-inf")), dist_reduce_fx="max")
+        # reference parity: min/max are PLAIN attributes, not registered
+        # states — they are neither synced nor reset (and forward's
+        # full-state dance therefore keeps them running across batches
+        # while resetting the base metric each call)
+        self.min_val = torch.tensor(float("inf"))
+        self.max_val = torch.tensor(float("-inf"))
 
     def update(self, *args: Any, **kwargs: Any) -> None:
         """Delegate to the base metric."""
@@ -41,13 +45,10 @@ class MinMaxMetric(WrapperMetric):
         return {"raw": val, "max": self.max_val, "min": self.min_val}
 
     def forward(self, *args: Any, **kwargs: Any) -> Any:
-        # the base metric's forward both accumulates and returns the batch value
-        val = self._base_metric(*args, **kwargs)
-        if self._is_suitable_val(val):
-            self.max_val = val if self.max_val.to(val.device) < val else self.max_val.to(val.device)
-            self.min_val = val if self.min_val.to(val.device) > val else self.min_val.to(val.device)
-        self._forward_cache = {"raw": val, "max": self.max_val, "min": self.min_val}
-        return self._forward_cache
+        # reference parity: run the stock full-state forward dance (which
+        # resets the base metric each call — the reference's observable
+        # behavior: 'raw' tracks the current batch under forward())
+        return Metric.forward(self, *args, **kwargs)
 
     def reset(self) -> None:
         super().reset()

@@ -1,0 +1,223 @@
+"""Modular-layer differential vs the reference: wrappers, aggregators,
+composition and collections driven with identical batch sequences.
+
+The functional harness (test_ref_differential.py) pins the math; this one
+pins the STATEFUL layer — forward()-vs-forward() per batch and final
+compute()-vs-compute() — where the wrapper/aggregation logic lives.
+"""
+from __future__ import annotations
+
+import os
+import sys
+
+import pytest
+import torch
+
+_REF = "/root/reference/src"
+HAVE_REF = os.path.isdir(_REF)
+pytestmark = pytest.mark.skipif(not HAVE_REF, reason="reference tree not available")
+
+if HAVE_REF:
+    sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), "..", "..", "tools", "refbench"))
+    sys.path.insert(0, _REF)
+
+import metrics_amd as ma
+
+
+def _tm():
+    import torchmetrics as tm
+
+    return tm
+
+
+def _cmp(a, b, atol=1e-5):
+    if isinstance(a, dict):
+        assert set(a) == set(b), (set(a), set(b))
+        for k in a:
+            _cmp(a[k], b[k], atol)
+        return
+    if isinstance(a, (list, tuple)):
+        for x, y in zip(a, b):
+            _cmp(x, y, atol)
+        return
+    a = torch.as_tensor(a).float()
+    b = torch.as_tensor(b).float()
+    both_nan = torch.isnan(a) & torch.isnan(b)
+    assert torch.allclose(a[~both_nan], b[~both_nan], atol=atol, rtol=1e-4), (a, b)
+
+
+def _drive(ours, ref, batches, atol=1e-5, check_forward=True):
+    for args in batches:
+        got = ours(*[x.clone() if isinstance(x, torch.Tensor) else x for x in args])
+        exp = ref(*[x.clone() if isinstance(x, torch.Tensor) else x for x in args])
+        if check_forward:
+            _cmp(got, exp, atol)
+    _cmp(ours.compute(), ref.compute(), atol)
+
+
+def _vals(seed, n_batches=4, shape=(32,)):
+    g = torch.Generator().manual_seed(seed)
+    return [(torch.randn(*shape, generator=g),) for _ in range(n_batches)]
+
+
+# ------------------------------------------------------------- aggregation
+@pytest.mark.parametrize("name", ["MeanMetric", "SumMetric", "MaxMetric", "MinMetric", "CatMetric"])
+def test_aggregators(name):
+    tm = _tm()
+    _drive(getattr(ma, name)(), getattr(tm, name)(), _vals(3))
+
+
+def test_mean_metric_weighted():
+    tm = _tm()
+    g = torch.Generator().manual_seed(4)
+    batches = [(torch.randn(16, generator=g), torch.rand(16, generator=g)) for _ in range(3)]
+    _drive(ma.MeanMetric(), tm.MeanMetric(), batches)
+
+
+@pytest.mark.parametrize("strategy", ["ignore", 0.5])
+def test_aggregator_nan_strategy(strategy):
+    tm = _tm()
+    g = torch.Generator().manual_seed(5)
+    batches = []
+    for _ in range(3):
+        v = torch.randn(16, generator=g)
+        v[::5] = float("nan")
+        batches.append((v,))
+    _drive(ma.MeanMetric(nan_strategy=strategy), tm.MeanMetric(nan_strategy=strategy), batches)
+
+
+@pytest.mark.parametrize("name", ["RunningMean", "RunningSum"])
+def test_running_aggregators(name):
+    tm = _tm()
+    _drive(getattr(ma, name)(window=3), getattr(tm, name)(window=3), _vals(6, n_batches=7))
+
+
+# ----------------------------------------------------------------- wrappers
+def _cls_batches(seed, n=4, b=32, c=5):
+    g = torch.Generator().manual_seed(seed)
+    return [
+        (torch.randn(b, c, generator=g).softmax(-1), torch.randint(0, c, (b,), generator=g))
+        for _ in range(n)
+    ]
+
+
+def test_classwise_wrapper():
+    tm = _tm()
+    ours = ma.ClasswiseWrapper(ma.MulticlassAccuracy(num_classes=5, average=None), labels=["a", "b", "c", "d", "e"])
+    ref = tm.ClasswiseWrapper(
+        tm.classification.MulticlassAccuracy(num_classes=5, average=None), labels=["a", "b", "c", "d", "e"]
+    )
+    _drive(ours, ref, _cls_batches(6))
+
+
+def test_minmax_wrapper():
+    tm = _tm()
+    ours = ma.MinMaxMetric(ma.MulticlassAccuracy(num_classes=5))
+    ref = tm.MinMaxMetric(tm.classification.MulticlassAccuracy(num_classes=5))
+    _drive(ours, ref, _cls_batches(7))
+
+
+def test_multioutput_wrapper():
+    tm = _tm()
+    g = torch.Generator().manual_seed(8)
+    batches = [(torch.randn(32, 3, generator=g), torch.randn(32, 3, generator=g)) for _ in range(4)]
+    ours = ma.MultioutputWrapper(ma.MeanSquaredError(), num_outputs=3)
+    ref = tm.MultioutputWrapper(tm.MeanSquaredError(), num_outputs=3)
+    _drive(ours, ref, batches)
+
+
+def test_multitask_wrapper():
+    tm = _tm()
+    g = torch.Generator().manual_seed(9)
+    ours = ma.MultitaskWrapper({"cls": ma.BinaryAccuracy(), "reg": ma.MeanSquaredError()})
+    ref = tm.wrappers.MultitaskWrapper({"cls": tm.classification.BinaryAccuracy(), "reg": tm.MeanSquaredError()})
+    for _ in range(3):
+        preds = {"cls": torch.rand(32, generator=g), "reg": torch.randn(32, generator=g)}
+        target = {"cls": torch.randint(0, 2, (32,), generator=g), "reg": torch.randn(32, generator=g)}
+        got = ours(preds, target)
+        exp = ref({k: v.clone() for k, v in preds.items()}, {k: v.clone() for k, v in target.items()})
+        _cmp(got, exp)
+    _cmp(ours.compute(), ref.compute())
+
+
+def test_running_wrapper():
+    tm = _tm()
+    ours = ma.Running(ma.MeanMetric(), window=3)
+    ref = tm.wrappers.Running(tm.MeanMetric(), window=3)
+    _drive(ours, ref, _vals(10, n_batches=7))
+
+
+def test_metric_tracker():
+    tm = _tm()
+    ours = ma.MetricTracker(ma.MulticlassAccuracy(num_classes=5))
+    ref = tm.MetricTracker(tm.classification.MulticlassAccuracy(num_classes=5))
+    for step in range(3):
+        ours.increment()
+        ref.increment()
+        for args in _cls_batches(20 + step, n=2):
+            ours.update(*args)
+            ref.update(*args)
+    _cmp(ours.compute_all(), ref.compute_all())
+    b1, i1 = ours.best_metric(return_step=True)
+    b2, i2 = ref.best_metric(return_step=True)
+    assert i1 == i2
+    _cmp(b1, b2)
+
+
+def test_bootstrapper_same_rng():
+    """BootStrapper resampling consumes the global torch RNG — with the same
+    seed both implementations must draw identical resamplings."""
+    tm = _tm()
+    g = torch.Generator().manual_seed(11)
+    batches = [
+        (torch.rand(64, generator=g), torch.randint(0, 2, (64,), generator=g)) for _ in range(3)
+    ]
+    ours = ma.BootStrapper(ma.BinaryAccuracy(), num_bootstraps=8, mean=True, std=True)
+    ref = tm.BootStrapper(tm.classification.BinaryAccuracy(), num_bootstraps=8, mean=True, std=True)
+    torch.manual_seed(123)
+    for args in batches:
+        ours.update(*args)
+    torch.manual_seed(123)
+    for args in batches:
+        ref.update(*args)
+    _cmp(ours.compute(), ref.compute())
+
+
+# ------------------------------------------------------------- composition
+def test_compositional_metric():
+    tm = _tm()
+    oa, ob = ma.MulticlassAccuracy(num_classes=5), ma.MulticlassPrecision(num_classes=5)
+    ra, rb = tm.classification.MulticlassAccuracy(num_classes=5), tm.classification.MulticlassPrecision(num_classes=5)
+    ours = (oa + ob) * 2 - 0.5
+    ref = (ra + rb) * 2 - 0.5
+    for args in _cls_batches(12):
+        ours.update(*args)
+        ref.update(*args)
+    _cmp(ours.compute(), ref.compute())
+
+
+# -------------------------------------------------------------- collections
+def test_collection_compute_groups_match():
+    tm = _tm()
+
+    def build(pkg, cls_mod):
+        return pkg.MetricCollection([
+            cls_mod.MulticlassAccuracy(num_classes=5, average="micro"),
+            cls_mod.MulticlassPrecision(num_classes=5),
+            cls_mod.MulticlassRecall(num_classes=5),
+            cls_mod.MulticlassConfusionMatrix(num_classes=5),
+        ])
+
+    ours = build(ma, ma)
+    ref = build(tm, tm.classification)
+    for args in _cls_batches(13):
+        ours.update(*args)
+        ref.update(*args)
+    _cmp(ours.compute(), ref.compute())
+    # our per-class fused state layout lets micro-average metrics share the
+    # macro metrics' group (the reference keeps scalar states for micro and
+    # cannot merge them) — dedup must be AT LEAST as aggressive, never less
+    assert len(ours.compute_groups) <= len(ref.compute_groups), (
+        ours.compute_groups,
+        ref.compute_groups,
+    )
