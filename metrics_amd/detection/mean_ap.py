@@ -570,6 +570,59 @@ class MeanAveragePrecision(Metric):
 
         area_rngs = np.array(self._AREA_RANGES, dtype=np.float32)
 
+        # torch accumulate stage: GPU when available, else threaded CPU torch
+        acc_dev = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+        rec_thrs_t = torch.from_numpy(rec_thrs.astype(np.float64, copy=False)).to(acc_dev)
+
+        def _accumulate_np(ki, order, scores_sorted, rank_sorted, dtm_a, dti_a, npig_a, valid_ai):
+            """CPU accumulate (numpy): torch-CPU op dispatch costs more than it
+            wins at per-class sizes; the torch path handles the GPU case."""
+            eps = np.finfo(np.float64).eps
+            m4_all = np.stack([dtm_a[ai][:, order] for ai in range(A)])  # (A,T,n)
+            i4_all = np.stack([dti_a[ai][:, order] for ai in range(A)])
+            npig_safe = np.where(valid_ai, npig_a, 1).astype(np.float64)
+            for mi, max_det in enumerate(max_dets):
+                if max_det >= max_det_top:
+                    m4, i4, sc = m4_all, i4_all, scores_sorted
+                else:
+                    keep = rank_sorted < max_det
+                    m4, i4, sc = m4_all[..., keep], i4_all[..., keep], scores_sorted[keep]
+                nc = m4.shape[-1]
+                if nc == 0:
+                    for ai in range(A):
+                        if valid_ai[ai]:
+                            recall[:, ki, ai, mi] = 0.0
+                            precision[:, :, ki, ai, mi] = 0.0
+                    continue
+                scored = ~i4
+                tps = np.cumsum(m4 & scored, axis=-1, dtype=np.int32).astype(np.float64)
+                fps = np.cumsum(~m4 & scored, axis=-1, dtype=np.int32).astype(np.float64)
+                rc = tps / npig_safe[:, None, None]
+                pr = tps / (tps + fps + eps)
+                pr_env = np.maximum.accumulate(pr[..., ::-1], axis=-1)[..., ::-1]
+                n_rows = A * T
+                rc2 = rc.reshape(n_rows, nc)
+                inds_rows = np.empty((n_rows, R), dtype=np.int64)
+                for row in range(n_rows):
+                    inds_rows[row] = np.searchsorted(rc2[row], rec_thrs, side="left")
+                local = inds_rows.ravel()
+                valid = local < nc
+                pr2 = pr_env.reshape(n_rows, nc)
+                q = np.where(valid, np.take_along_axis(pr2, np.minimum(inds_rows, nc - 1), axis=1).ravel(), 0.0)
+                q3 = q.reshape(A, T, R)
+                rec3 = rc[..., -1]
+                for ai in range(A):
+                    if not valid_ai[ai]:
+                        continue
+                    precision[:, :, ki, ai, mi] = q3[ai]
+                    recall[:, ki, ai, mi] = rec3[ai]
+                if self.extended_summary:
+                    ss = np.where(valid, sc[np.minimum(local, nc - 1)], 0.0)
+                    ss3 = ss.reshape(A, T, R)
+                    for ai in range(A):
+                        if valid_ai[ai]:
+                            scores_out[:, :, ki, ai, mi] = ss3[ai]
+
         def _process_class(ki: int) -> None:
             dlo, dhi = int(d_cls_off[ki]), int(d_cls_off[ki + 1])
             glo, ghi = int(g_cls_off[ki]), int(g_cls_off[ki + 1])
@@ -615,14 +668,19 @@ class MeanAveragePrecision(Metric):
                     dt_off, gt_off, area_rngs, iou_thrs,
                 )
 
-            # fully batched accumulation over (area-range, max-det): one score
-            # sort per class; smaller max_det caps are masks on the sorted
-            # arrays — a masked det freezes the cumsums, which plateaus rc/pr
-            # and leaves the interpolated precision unchanged (COCO-equivalent)
+            # fully batched accumulation over (area-range, max-det) in torch
+            # (GPU when available — the reference leaves the GPU idle through
+            # this stage): one score sort per class; smaller max_det caps are
+            # masks on the sorted arrays — a masked det freezes the cumsums,
+            # which plateaus rc/pr and leaves the interpolated precision
+            # unchanged (COCO-equivalent). int32 cumsum is exact for counts;
+            # rc/pr promote to float64 where the searchsorted boundary
+            # compares happen (side-left at exact rc == recThr hits). Batched
+            # torch.searchsorted keeps each row in its own domain, so no
+            # offset quantization.
             order = np.argsort(-scores_k, kind="stable")
             scores_sorted = scores_k[order]
             rank_sorted = rank_k[order]
-            eps = np.finfo(np.float64).eps
             n_cols = order.shape[0]
             valid_ai = npig_a > 0
             if n_cols == 0 or not valid_ai.any():
@@ -631,16 +689,23 @@ class MeanAveragePrecision(Metric):
                         recall[:, ki, ai, :] = 0.0
                         precision[:, :, ki, ai, :] = 0.0
                 return
-            m4_all = np.stack([dtm_a[ai][:, order] for ai in range(A)])  # (A,T,n)
-            i4_all = np.stack([dti_a[ai][:, order] for ai in range(A)])
-            npig_safe = np.where(valid_ai, npig_a, 1).astype(np.float64)
+            eps = float(np.finfo(np.float64).eps)
+            if acc_dev.type != "cuda":
+                _accumulate_np(ki, order, scores_sorted, rank_sorted, dtm_a, dti_a, npig_a, valid_ai)
+                return
+            m4_all = torch.from_numpy(np.stack([dtm_a[ai][:, order] for ai in range(A)])).to(acc_dev)
+            i4_all = torch.from_numpy(np.stack([dti_a[ai][:, order] for ai in range(A)])).to(acc_dev)
+            sc_all = torch.from_numpy(scores_sorted.astype(np.float64, copy=False)).to(acc_dev)
+            rank_t = torch.from_numpy(np.ascontiguousarray(rank_sorted)).to(acc_dev)
+            npig_safe = torch.from_numpy(np.where(valid_ai, npig_a, 1).astype(np.float64)).to(acc_dev)
+            n_rows = A * T
             for mi, max_det in enumerate(max_dets):
                 # compress to the capped dets (maxdet=1 keeps ~n_imgs entries)
                 if max_det >= max_det_top:
-                    m4, i4, sc = m4_all, i4_all, scores_sorted
+                    m4, i4, sc = m4_all, i4_all, sc_all
                 else:
-                    keep = rank_sorted < max_det
-                    m4, i4, sc = m4_all[..., keep], i4_all[..., keep], scores_sorted[keep]
+                    keep = rank_t < max_det
+                    m4, i4, sc = m4_all[..., keep], i4_all[..., keep], sc_all[keep]
                 nc = m4.shape[-1]
                 if nc == 0:
                     for ai in range(A):
@@ -649,40 +714,39 @@ class MeanAveragePrecision(Metric):
                             precision[:, :, ki, ai, mi] = 0.0
                     continue
                 scored = ~i4
-                tps = np.cumsum(m4 & scored, axis=-1, dtype=np.float64)  # (A,T,nc)
-                fps = np.cumsum(~m4 & scored, axis=-1, dtype=np.float64)
+                tps = torch.cumsum(m4 & scored, dim=-1, dtype=torch.int32).double()
+                fps = torch.cumsum(~m4 & scored, dim=-1, dtype=torch.int32).double()
                 rc = tps / npig_safe[:, None, None]
                 pr = tps / (tps + fps + eps)
-                pr_env = np.maximum.accumulate(pr[..., ::-1], axis=-1)[..., ::-1]
-                # per-(area,iou) searchsorted in the ORIGINAL domain: an
-                # offset-batched variant quantizes away sub-ulp rc-vs-recThr
-                # differences at exact boundary hits (rc == 0.6 vs linspace
-                # 0.6), flipping the side-left result — 40 tiny calls are free
-                n_rows = A * T
+                pr_env = pr.flip(-1).cummax(-1).values.flip(-1)
                 rc2 = rc.reshape(n_rows, nc)
-                inds_rows = np.empty((n_rows, R), dtype=np.int64)
-                for row in range(n_rows):
-                    inds_rows[row] = np.searchsorted(rc2[row], rec_thrs, side="left")
-                local = inds_rows.ravel()
-                valid = local < nc
-                pr2 = pr_env.reshape(n_rows, nc)
-                q = np.where(valid, np.take_along_axis(pr2, np.minimum(inds_rows, nc - 1), axis=1).ravel(), 0.0)
-                q3 = q.reshape(A, T, R)
-                rec3 = rc[..., -1]  # (A,T)
+                inds = torch.searchsorted(rc2, rec_thrs_t.expand(n_rows, R).contiguous())
+                valid = inds < nc
+                gathered = torch.take_along_dim(pr_env.reshape(n_rows, nc), inds.clamp(max=nc - 1), dim=1)
+                q3 = torch.where(valid, gathered, torch.zeros((), dtype=torch.float64, device=acc_dev))
+                q3 = q3.reshape(A, T, R).cpu().numpy()
+                rec3 = rc[..., -1].cpu().numpy()  # (A,T)
                 for ai in range(A):
                     if not valid_ai[ai]:
                         continue
                     precision[:, :, ki, ai, mi] = q3[ai]
                     recall[:, ki, ai, mi] = rec3[ai]
                 if self.extended_summary:
-                    ss = np.where(valid, sc[np.minimum(local, nc - 1)], 0.0)
-                    ss3 = ss.reshape(A, T, R)
+                    ss = torch.where(valid, sc[inds.clamp(max=nc - 1)], torch.zeros((), dtype=torch.float64, device=acc_dev))
+                    ss3 = ss.reshape(A, T, R).cpu().numpy()
                     for ai in range(A):
                         if valid_ai[ai]:
                             scores_out[:, :, ki, ai, mi] = ss3[ai]
 
-        for ki in range(K):
-            _process_class(ki)
+        # classes are independent and the heavy numpy ops release the GIL
+        if K > 1:
+            from concurrent.futures import ThreadPoolExecutor
+
+            with ThreadPoolExecutor(max_workers=min(8, K)) as pool:
+                list(pool.map(_process_class, range(K)))
+        else:
+            for ki in range(K):
+                _process_class(ki)
 
         def _summarize(ap: bool, iou_thr: Optional[float] = None, area: int = 0, max_det_idx: int = -1) -> Tensor:
             if ap:
