@@ -570,9 +570,15 @@ class MeanAveragePrecision(Metric):
 
         area_rngs = np.array(self._AREA_RANGES, dtype=np.float32)
 
-        # torch accumulate stage: GPU when available, else threaded CPU torch
+        # torch accumulate stage: GPU when available, else numpy on CPU
         acc_dev = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
         rec_thrs_t = torch.from_numpy(rec_thrs.astype(np.float64, copy=False)).to(acc_dev)
+        if acc_dev.type == "cuda":
+            # device-resident result tensors: ONE host transfer after the
+            # class loop instead of a sync per (class, max_det)
+            precision_t = torch.full((T, R, K, A, M), -1.0, dtype=torch.float64, device=acc_dev)
+            recall_t = torch.full((T, K, A, M), -1.0, dtype=torch.float64, device=acc_dev)
+            scores_t = torch.full((T, R, K, A, M), -1.0, dtype=torch.float64, device=acc_dev)
 
         def _accumulate_np(ki, order, scores_sorted, rank_sorted, dtm_a, dti_a, npig_a, valid_ai):
             """CPU accumulate (numpy): torch-CPU op dispatch costs more than it
@@ -686,8 +692,12 @@ class MeanAveragePrecision(Metric):
             if n_cols == 0 or not valid_ai.any():
                 for ai in range(A):
                     if npig_a[ai] > 0:
-                        recall[:, ki, ai, :] = 0.0
-                        precision[:, :, ki, ai, :] = 0.0
+                        if acc_dev.type == "cuda":
+                            recall_t[:, ki, ai, :] = 0.0
+                            precision_t[:, :, ki, ai, :] = 0.0
+                        else:
+                            recall[:, ki, ai, :] = 0.0
+                            precision[:, :, ki, ai, :] = 0.0
                 return
             eps = float(np.finfo(np.float64).eps)
             if acc_dev.type != "cuda":
@@ -710,8 +720,8 @@ class MeanAveragePrecision(Metric):
                 if nc == 0:
                     for ai in range(A):
                         if valid_ai[ai]:
-                            recall[:, ki, ai, mi] = 0.0
-                            precision[:, :, ki, ai, mi] = 0.0
+                            recall_t[:, ki, ai, mi] = 0.0
+                            precision_t[:, :, ki, ai, mi] = 0.0
                     continue
                 scored = ~i4
                 tps = torch.cumsum(m4 & scored, dim=-1, dtype=torch.int32).double()
@@ -724,29 +734,35 @@ class MeanAveragePrecision(Metric):
                 valid = inds < nc
                 gathered = torch.take_along_dim(pr_env.reshape(n_rows, nc), inds.clamp(max=nc - 1), dim=1)
                 q3 = torch.where(valid, gathered, torch.zeros((), dtype=torch.float64, device=acc_dev))
-                q3 = q3.reshape(A, T, R).cpu().numpy()
-                rec3 = rc[..., -1].cpu().numpy()  # (A,T)
+                q3 = q3.reshape(A, T, R)
+                rec3 = rc[..., -1]  # (A,T)
                 for ai in range(A):
                     if not valid_ai[ai]:
                         continue
-                    precision[:, :, ki, ai, mi] = q3[ai]
-                    recall[:, ki, ai, mi] = rec3[ai]
+                    precision_t[:, :, ki, ai, mi] = q3[ai]
+                    recall_t[:, ki, ai, mi] = rec3[ai]
                 if self.extended_summary:
                     ss = torch.where(valid, sc[inds.clamp(max=nc - 1)], torch.zeros((), dtype=torch.float64, device=acc_dev))
-                    ss3 = ss.reshape(A, T, R).cpu().numpy()
+                    ss3 = ss.reshape(A, T, R)
                     for ai in range(A):
                         if valid_ai[ai]:
-                            scores_out[:, :, ki, ai, mi] = ss3[ai]
+                            scores_t[:, :, ki, ai, mi] = ss3[ai]
 
-        # classes are independent and the heavy numpy ops release the GIL
-        if K > 1:
+        # classes are independent; on CPU the heavy numpy ops release the GIL,
+        # on GPU launches are async so a plain loop pipelines fine
+        if acc_dev.type == "cuda" or K <= 1:
+            for ki in range(K):
+                _process_class(ki)
+        else:
             from concurrent.futures import ThreadPoolExecutor
 
             with ThreadPoolExecutor(max_workers=min(8, K)) as pool:
                 list(pool.map(_process_class, range(K)))
-        else:
-            for ki in range(K):
-                _process_class(ki)
+        if acc_dev.type == "cuda" and K > 0:
+            precision = precision_t.cpu().numpy()
+            recall = recall_t.cpu().numpy()
+            if self.extended_summary:
+                scores_out = scores_t.cpu().numpy()
 
         def _summarize(ap: bool, iou_thr: Optional[float] = None, area: int = 0, max_det_idx: int = -1) -> Tensor:
             if ap:
