@@ -32,7 +32,6 @@ def _intern_pairs(pairs):
     off_b = np.zeros(len(pairs) + 1, dtype=np.int64)
     off_a[1:] = np.cumsum([len(x) for x in a_arrs])
     off_b[1:] = np.cumsum([len(x) for x in b_arrs])
-    cat = lambda arrs: np.concatenate(arrs) if arrs and off_a is not None else np.zeros(0, np.int32)
     tok_a = np.concatenate(a_arrs) if a_arrs else np.zeros(0, np.int32)
     tok_b = np.concatenate(b_arrs) if b_arrs else np.zeros(0, np.int32)
     return tok_a.astype(np.int32), off_a, tok_b.astype(np.int32), off_b

@@ -99,23 +99,21 @@ class WordInfoLost(Metric):
 
     def __init__(self, **kwargs: Any) -> None:
         super().__init__(**kwargs)
-        self.add_state("hits", tensor(0.0), dist_reduce_fx="sum")
+        # reference state layout (text/wip.py): errors = Σ(dist - max_len),
+        # a non-positive "kept words" count — NOT backtrace hit counts
+        self.add_state("errors", tensor(0.0), dist_reduce_fx="sum")
         self.add_state("target_total", tensor(0.0), dist_reduce_fx="sum")
         self.add_state("preds_total", tensor(0.0), dist_reduce_fx="sum")
 
     def update(self, preds, target) -> None:
-        """Accumulate hit counts."""
-        from metrics_amd.functional.text.helper import _edit_distance_counts
-
-        preds, target = _norm_inputs(preds, target)
-        for p, t in zip(preds, target):
-            _, _, _, hits = _edit_distance_counts(p.split(), t.split())
-            self.hits += hits
-            self.target_total += len(t.split())
-            self.preds_total += len(p.split())
+        """Accumulate edit/length statistics."""
+        errors, total, target_total, preds_total = _mer_wil_wip_update(preds, target)
+        self.errors += errors - total
+        self.target_total += target_total
+        self.preds_total += preds_total
 
     def compute(self) -> Tensor:
-        return 1 - self.hits / self.target_total * (self.hits / self.preds_total)
+        return 1 - (self.errors / self.target_total) * (self.errors / self.preds_total)
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)
@@ -127,7 +125,7 @@ class WordInfoPreserved(WordInfoLost):
     higher_is_better = True
 
     def compute(self) -> Tensor:
-        return self.hits / self.target_total * (self.hits / self.preds_total)
+        return (self.errors / self.target_total) * (self.errors / self.preds_total)
 
 
 class EditDistance(Metric):

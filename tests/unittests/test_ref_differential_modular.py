@@ -221,3 +221,30 @@ def test_collection_compute_groups_match():
         ours.compute_groups,
         ref.compute_groups,
     )
+
+
+# ------------------------------------------------------------------ text
+def test_text_error_rate_modules():
+    tm = _tm()
+    import random
+
+    rnd = random.Random(41)
+    words = ["the", "cat", "sat", "mat", "dog", "ran"]
+
+    def sent():
+        return " ".join(rnd.choice(words) for _ in range(rnd.randint(3, 9)))
+
+    batches = [([sent() for _ in range(8)], [sent() for _ in range(8)]) for _ in range(3)]
+    for our_cls, ref_cls in [
+        (ma.text.CharErrorRate, tm.text.CharErrorRate),
+        (ma.text.WordErrorRate, tm.text.WordErrorRate),
+        (ma.text.MatchErrorRate, tm.text.MatchErrorRate),
+        (ma.text.WordInfoLost, tm.text.WordInfoLost),
+        (ma.text.WordInfoPreserved, tm.text.WordInfoPreserved),
+        (ma.text.TranslationEditRate, tm.text.TranslationEditRate),
+    ]:
+        ours, ref = our_cls(), ref_cls()
+        for p, t in batches:
+            got, exp = ours(p, t), ref(p, t)
+            _cmp(got, exp)
+        _cmp(ours.compute(), ref.compute())
