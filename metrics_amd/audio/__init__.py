@@ -30,20 +30,22 @@ class _AvgAudioMetric(Metric):
     is_differentiable = True
     higher_is_better = True
 
-    sum_value: Tensor
     total: Tensor
+
+    # per-subclass so checkpoint layouts match the reference (sum_snr, ...)
+    _SUM_STATE = "sum_value"
 
     def __init__(self, **kwargs: Any) -> None:
         super().__init__(**kwargs)
-        self.add_state("sum_value", default=torch.tensor(0.0), dist_reduce_fx="sum")
+        self.add_state(self._SUM_STATE, default=torch.tensor(0.0), dist_reduce_fx="sum")
         self.add_state("total", default=torch.tensor(0), dist_reduce_fx="sum")
 
     def _acc(self, val: Tensor) -> None:
-        self.sum_value += val.sum()
+        setattr(self, self._SUM_STATE, getattr(self, self._SUM_STATE) + val.sum())
         self.total += val.numel()
 
     def compute(self) -> Tensor:
-        return self.sum_value / self.total
+        return getattr(self, self._SUM_STATE) / self.total
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)
@@ -51,6 +53,8 @@ class _AvgAudioMetric(Metric):
 
 class SignalNoiseRatio(_AvgAudioMetric):
     """SNR (stateful)."""
+
+    _SUM_STATE = "sum_snr"
 
     def __init__(self, zero_mean: bool = False, **kwargs: Any) -> None:
         super().__init__(**kwargs)
@@ -64,6 +68,8 @@ class SignalNoiseRatio(_AvgAudioMetric):
 class ScaleInvariantSignalNoiseRatio(_AvgAudioMetric):
     """SI-SNR (stateful)."""
 
+    _SUM_STATE = "sum_si_snr"
+
     def update(self, preds: Tensor, target: Tensor) -> None:
         """Accumulate per-sample SI-SNR."""
         self._acc(scale_invariant_signal_noise_ratio(preds, target))
@@ -71,6 +77,8 @@ class ScaleInvariantSignalNoiseRatio(_AvgAudioMetric):
 
 class SignalDistortionRatio(_AvgAudioMetric):
     """SDR (stateful)."""
+
+    _SUM_STATE = "sum_sdr"
 
     def __init__(
         self,
@@ -95,6 +103,8 @@ class SignalDistortionRatio(_AvgAudioMetric):
 
 class ScaleInvariantSignalDistortionRatio(_AvgAudioMetric):
     """SI-SDR (stateful)."""
+
+    _SUM_STATE = "sum_si_sdr"
 
     def __init__(self, zero_mean: bool = False, **kwargs: Any) -> None:
         super().__init__(**kwargs)

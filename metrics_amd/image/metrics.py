@@ -113,26 +113,33 @@ class PeakSignalNoiseRatio(Metric):
 
 
 class _ScoreAverageMetric(Metric):
-    """Base: accumulate per-image scores as (sum, count)."""
+    """Base: accumulate per-image scores as (sum, count).
+
+    State names are per-subclass so checkpoint layouts match the reference
+    (e.g. SSIM -> similarity/total, SAM -> sum_sam/numel).
+    """
 
     is_differentiable = True
     higher_is_better = True
     full_state_update = False
 
-    score_sum: Tensor
-    total: Tensor
+    _SCORE_STATE = "score_sum"
+    _TOTAL_STATE = "total"
+    _TOTAL_FLOAT = False  # reference SSIM/MS-SSIM keep a float total
 
     def __init__(self, **kwargs: Any) -> None:
         super().__init__(**kwargs)
-        self.add_state("score_sum", torch.tensor(0.0), dist_reduce_fx="sum")
-        self.add_state("total", torch.tensor(0), dist_reduce_fx="sum")
+        self.add_state(self._SCORE_STATE, torch.tensor(0.0), dist_reduce_fx="sum")
+        total_default = torch.tensor(0.0) if self._TOTAL_FLOAT else torch.tensor(0)
+        self.add_state(self._TOTAL_STATE, total_default, dist_reduce_fx="sum")
 
     def _accumulate(self, scores: Tensor, n: int) -> None:
-        self.score_sum += scores.sum() if scores.ndim else scores * n
-        self.total += n
+        cur = getattr(self, self._SCORE_STATE)
+        setattr(self, self._SCORE_STATE, cur + (scores.sum() if scores.ndim else scores * n))
+        setattr(self, self._TOTAL_STATE, getattr(self, self._TOTAL_STATE) + n)
 
     def compute(self) -> Tensor:
-        return self.score_sum / self.total
+        return getattr(self, self._SCORE_STATE) / getattr(self, self._TOTAL_STATE)
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)
@@ -140,6 +147,9 @@ class _ScoreAverageMetric(Metric):
 
 class StructuralSimilarityIndexMeasure(_ScoreAverageMetric):
     """SSIM (stateful)."""
+
+    _SCORE_STATE = "similarity"
+    _TOTAL_FLOAT = True
 
     plot_lower_bound: float = 0.0
     plot_upper_bound: float = 1.0
@@ -175,6 +185,9 @@ class StructuralSimilarityIndexMeasure(_ScoreAverageMetric):
 
 class MultiScaleStructuralSimilarityIndexMeasure(_ScoreAverageMetric):
     """MS-SSIM (stateful)."""
+
+    _SCORE_STATE = "similarity"
+    _TOTAL_FLOAT = True
 
     plot_lower_bound: float = 0.0
     plot_upper_bound: float = 1.0
@@ -215,6 +228,9 @@ class MultiScaleStructuralSimilarityIndexMeasure(_ScoreAverageMetric):
 class UniversalImageQualityIndex(_ScoreAverageMetric):
     """UQI (stateful)."""
 
+    _SCORE_STATE = "sum_uqi"
+    _TOTAL_STATE = "numel"
+
     plot_lower_bound: float = 0.0
     plot_upper_bound: float = 1.0
 
@@ -237,6 +253,9 @@ class UniversalImageQualityIndex(_ScoreAverageMetric):
 
 class SpectralAngleMapper(_ScoreAverageMetric):
     """SAM (stateful, radians; lower is better)."""
+
+    _SCORE_STATE = "sum_sam"
+    _TOTAL_STATE = "numel"
 
     higher_is_better = False
     plot_lower_bound: float = 0.0
@@ -276,7 +295,7 @@ class TotalVariation(Metric):
     full_state_update = False
     plot_lower_bound: float = 0.0
 
-    score_sum: Tensor
+    score: Tensor
     num_elements: Tensor
 
     def __init__(self, reduction: str = "sum", **kwargs: Any) -> None:
@@ -284,10 +303,10 @@ class TotalVariation(Metric):
         if reduction is not None and reduction not in ("sum", "mean", "none"):
             raise ValueError("Expected argument `reduction` to either be 'sum', 'mean', 'none' or None")
         self.reduction = reduction
-        if self.reduction is None or self.reduction == "none":
-            self.add_state("score_list", default=[], dist_reduce_fx="cat")
-        else:
-            self.add_state("score_sum", default=torch.tensor(0.0), dist_reduce_fx="sum")
+        # reference state layout (image/tv.py): both the list and the scalar
+        # states always exist; the reduction picks which one accumulates
+        self.add_state("score_list", default=[], dist_reduce_fx="cat")
+        self.add_state("score", default=torch.tensor(0.0), dist_reduce_fx="sum")
         self.add_state("num_elements", default=torch.tensor(0, dtype=torch.int), dist_reduce_fx="sum")
 
     def update(self, img: Tensor) -> None:
@@ -296,15 +315,15 @@ class TotalVariation(Metric):
         if self.reduction is None or self.reduction == "none":
             self.score_list.append(score)
         else:
-            self.score_sum += score.sum()
+            self.score += score.sum()
         self.num_elements += score.numel()
 
     def compute(self) -> Tensor:
         if self.reduction is None or self.reduction == "none":
             return dim_zero_cat(self.score_list)
         if self.reduction == "mean":
-            return self.score_sum / self.num_elements
-        return self.score_sum
+            return self.score / self.num_elements
+        return self.score
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)
@@ -312,6 +331,9 @@ class TotalVariation(Metric):
 
 class RootMeanSquaredErrorUsingSlidingWindow(_ScoreAverageMetric):
     """RMSE over sliding windows (stateful; lower is better)."""
+
+    _SCORE_STATE = "rmse_val_sum"
+    _TOTAL_STATE = "total_images"
 
     higher_is_better = False
     plot_lower_bound: float = 0.0

@@ -124,14 +124,14 @@ class FleissKappa(Metric):
         if mode not in ("counts", "probs"):
             raise ValueError("Argument `mode` should be one of 'counts' or 'probs'")
         self.mode = mode
-        self.add_state("ratings", default=[], dist_reduce_fx="cat")
+        self.add_state("counts", default=[], dist_reduce_fx="cat")
 
     def update(self, ratings: Tensor) -> None:
         """Append a ratings matrix."""
-        self.ratings.append(ratings)
+        self.counts.append(ratings)
 
     def compute(self) -> Tensor:
-        return fleiss_kappa(dim_zero_cat(self.ratings), self.mode)
+        return fleiss_kappa(dim_zero_cat(self.counts), self.mode)
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)

@@ -27,7 +27,7 @@ class LogCoshError(Metric):
             raise ValueError(f"Expected num_outputs to be a positive integer but got {num_outputs}")
         self.num_outputs = num_outputs
         self.add_state("sum_log_cosh_error", default=torch.zeros(num_outputs), dist_reduce_fx="sum")
-        self.add_state("total", default=torch.tensor(0.0), dist_reduce_fx="sum")
+        self.add_state("total", default=torch.tensor(0), dist_reduce_fx="sum")
 
     def update(self, preds: Tensor, target: Tensor) -> None:
         """Accumulate logcosh errors."""

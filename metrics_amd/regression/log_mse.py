@@ -24,7 +24,7 @@ class MeanSquaredLogError(Metric):
     def __init__(self, **kwargs: Any) -> None:
         super().__init__(**kwargs)
         self.add_state("sum_squared_log_error", default=torch.tensor(0.0), dist_reduce_fx="sum")
-        self.add_state("total", default=torch.tensor(0.0), dist_reduce_fx="sum")
+        self.add_state("total", default=torch.tensor(0), dist_reduce_fx="sum")
 
     def update(self, preds: Tensor, target: Tensor) -> None:
         """Accumulate squared log errors."""

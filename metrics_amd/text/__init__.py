@@ -280,12 +280,15 @@ class CHRFScore(Metric):
         self.whitespace = whitespace
         self.return_sentence_level_score = return_sentence_level_score
 
-        total = n_char_order + n_word_order
-        self.add_state("total_tp", torch.zeros(total), dist_reduce_fx="sum")
-        self.add_state("total_fp", torch.zeros(total), dist_reduce_fx="sum")
-        self.add_state("total_fn", torch.zeros(total), dist_reduce_fx="sum")
+        # reference state layout (text/chrf.py): one scalar per
+        # (matching|preds|target) x (char 1..n_char, word 1..n_word)
+        for kind in ("matching", "preds", "target"):
+            for n in range(1, n_char_order + 1):
+                self.add_state(f"total_{kind}_char_{n}_grams", tensor(0.0), dist_reduce_fx="sum")
+            for n in range(1, n_word_order + 1):
+                self.add_state(f"total_{kind}_word_{n}_grams", tensor(0.0), dist_reduce_fx="sum")
         if return_sentence_level_score:
-            self.add_state("sentence_chrf", [], dist_reduce_fx="cat")
+            self.add_state("sentence_chrf_score", [], dist_reduce_fx="cat")
 
     def update(self, preds: Sequence[str], target: Sequence[Sequence[str]]) -> None:
         """Accumulate n-gram statistics (best reference per sentence)."""
@@ -335,19 +338,30 @@ class CHRFScore(Metric):
                 if best_stats is None or f >= best_f:
                     best_f = f
                     best_stats = (tp, fp, fn)
-            self.total_tp += best_stats[0]
-            self.total_fp += best_stats[1]
-            self.total_fn += best_stats[2]
+            tp_b, fp_b, fn_b = best_stats
+            for i, name in enumerate(self._order_names()):
+                setattr(self, f"total_matching_{name}", getattr(self, f"total_matching_{name}") + tp_b[i])
+                setattr(self, f"total_preds_{name}", getattr(self, f"total_preds_{name}") + tp_b[i] + fp_b[i])
+                setattr(self, f"total_target_{name}", getattr(self, f"total_target_{name}") + tp_b[i] + fn_b[i])
             if self.return_sentence_level_score:
-                self.sentence_chrf.append(best_f.reshape(1))
+                self.sentence_chrf_score.append(best_f.reshape(1))
+
+    def _order_names(self):
+        return [f"char_{n}_grams" for n in range(1, self.n_char_order + 1)] + [
+            f"word_{n}_grams" for n in range(1, self.n_word_order + 1)
+        ]
 
     def compute(self):
-        """Corpus chrF."""
+        """Corpus chrF from the per-order count states."""
         from metrics_amd.functional.text.bleu import _chrf_f_score
 
-        score = _chrf_f_score(self.total_tp, self.total_fp, self.total_fn, self.beta)
+        names = self._order_names()
+        tp = torch.stack([getattr(self, f"total_matching_{n}") for n in names])
+        fp = torch.stack([getattr(self, f"total_preds_{n}") for n in names]) - tp
+        fn = torch.stack([getattr(self, f"total_target_{n}") for n in names]) - tp
+        score = _chrf_f_score(tp, fp, fn, self.beta)
         if self.return_sentence_level_score:
-            return score, dim_zero_cat(self.sentence_chrf)
+            return score, dim_zero_cat(self.sentence_chrf_score)
         return score
 
     def plot(self, val=None, ax=None):
@@ -378,7 +392,7 @@ class TranslationEditRate(Metric):
         self.asian_support = asian_support
         self.return_sentence_level_score = return_sentence_level_score
         self.add_state("total_num_edits", tensor(0.0), dist_reduce_fx="sum")
-        self.add_state("total_tgt_length", tensor(0.0), dist_reduce_fx="sum")
+        self.add_state("total_tgt_len", tensor(0.0), dist_reduce_fx="sum")
         if return_sentence_level_score:
             self.add_state("sentence_ter", [], dist_reduce_fx="cat")
 
@@ -406,13 +420,13 @@ class TranslationEditRate(Metric):
         for p, refs, s_ter in zip(preds_, target_, sent):
             avg_len = sum(len(_norm(r)) for r in refs) / len(refs)
             self.total_num_edits += s_ter * avg_len
-            self.total_tgt_length += avg_len
+            self.total_tgt_len += avg_len
             if self.return_sentence_level_score:
                 self.sentence_ter.append(s_ter.reshape(1))
 
     def compute(self):
         """Corpus TER."""
-        score = self.total_num_edits / self.total_tgt_length
+        score = self.total_num_edits / self.total_tgt_len
         if self.return_sentence_level_score:
             return score, dim_zero_cat(self.sentence_ter)
         return score
@@ -547,7 +561,7 @@ class SQuAD(Metric):
         super().__init__(**kwargs)
         self.add_state("f1_score", tensor(0.0), dist_reduce_fx="sum")
         self.add_state("exact_match", tensor(0.0), dist_reduce_fx="sum")
-        self.add_state("total", tensor(0), dist_reduce_fx="sum")
+        self.add_state("total", tensor(0, dtype=torch.int32), dist_reduce_fx="sum")
 
     def update(self, preds, target) -> None:
         """Accumulate EM/F1 sums."""
