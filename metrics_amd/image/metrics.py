@@ -334,6 +334,7 @@ class RootMeanSquaredErrorUsingSlidingWindow(_ScoreAverageMetric):
 
     _SCORE_STATE = "rmse_val_sum"
     _TOTAL_STATE = "total_images"
+    _TOTAL_FLOAT = True  # reference keeps a float image count
 
     higher_is_better = False
     plot_lower_bound: float = 0.0
