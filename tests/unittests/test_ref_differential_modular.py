@@ -248,3 +248,79 @@ def test_text_error_rate_modules():
             got, exp = ours(p, t), ref(p, t)
             _cmp(got, exp)
         _cmp(ours.compute(), ref.compute())
+
+
+# ------------------------------------------------------------- retrieval
+@pytest.mark.parametrize(
+    ("name", "kwargs"),
+    [
+        ("RetrievalMAP", {}),
+        ("RetrievalMRR", {}),
+        ("RetrievalPrecision", {"top_k": 3}),
+        ("RetrievalRecall", {"top_k": 3}),
+        ("RetrievalHitRate", {"top_k": 3}),
+        ("RetrievalFallOut", {"top_k": 3}),
+        ("RetrievalNormalizedDCG", {}),
+        ("RetrievalRPrecision", {}),
+        ("RetrievalMAP", {"empty_target_action": "skip"}),
+        ("RetrievalPrecision", {"empty_target_action": "pos", "top_k": 2}),
+    ],
+)
+def test_retrieval_modules(name, kwargs):
+    tm = _tm()
+    g = torch.Generator().manual_seed(51)
+    batches = [
+        (
+            torch.rand(120, generator=g),
+            torch.randint(0, 2, (120,), generator=g),
+            torch.randint(0, 12, (120,), generator=g),
+        )
+        for _ in range(3)
+    ]
+    ours = getattr(ma.retrieval, name)(**kwargs)
+    ref = getattr(tm.retrieval, name)(**kwargs)
+    for p, t, idx in batches:
+        ours.update(p, t, indexes=idx)
+        ref.update(p, t, indexes=idx)
+    _cmp(ours.compute(), ref.compute())
+
+
+def test_detection_iou_module():
+    tm = _tm()
+    try:
+        ref = tm.detection.IntersectionOverUnion()
+    except Exception as err:
+        pytest.skip(f"reference IoU module unavailable offline: {err}")
+    g = torch.Generator().manual_seed(52)
+
+    def boxes(n):
+        xy = torch.rand(n, 2, generator=g) * 60
+        wh = torch.rand(n, 2, generator=g) * 25 + 4
+        return torch.cat([xy, xy + wh], 1)
+
+    ours = ma.detection.IntersectionOverUnion()
+    for _ in range(2):
+        p = [{"boxes": boxes(6), "scores": torch.rand(6, generator=g), "labels": torch.randint(0, 2, (6,), generator=g)}]
+        t = [{"boxes": boxes(4), "labels": torch.randint(0, 2, (4,), generator=g)}]
+        ours.update(p, t)
+        ref.update(p, t)
+    _cmp(ours.compute(), ref.compute())
+
+
+def test_classification_curve_modules():
+    tm = _tm()
+    g = torch.Generator().manual_seed(53)
+    batches = [(torch.rand(96, generator=g), torch.randint(0, 2, (96,), generator=g)) for _ in range(3)]
+    for our_cls, ref_cls, kw in [
+        (ma.BinaryROC, tm.classification.BinaryROC, {"thresholds": None}),
+        (ma.BinaryROC, tm.classification.BinaryROC, {"thresholds": 15}),
+        (ma.BinaryPrecisionRecallCurve, tm.classification.BinaryPrecisionRecallCurve, {"thresholds": None}),
+        (ma.BinaryAUROC, tm.classification.BinaryAUROC, {"thresholds": None}),
+        (ma.BinaryAveragePrecision, tm.classification.BinaryAveragePrecision, {"thresholds": None}),
+        (ma.BinaryCalibrationError, tm.classification.BinaryCalibrationError, {"n_bins": 10}),
+    ]:
+        ours, ref = our_cls(**kw), ref_cls(**kw)
+        for p, t in batches:
+            ours.update(p, t)
+            ref.update(p, t)
+        _cmp(ours.compute(), ref.compute())
