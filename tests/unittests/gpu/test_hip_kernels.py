@@ -1047,3 +1047,39 @@ def test_topk_precision_recall_gpu():
         mc = cls(num_classes=C, top_k=3, average="macro")
         mc.update(preds, target)
         assert torch.allclose(mg.compute().cpu(), mc.compute(), atol=1e-6), cls.__name__
+
+
+# -------------------------------------------------- host-sync regression
+def test_update_has_no_device_host_sync():
+    """The hot update path must never synchronize with the host (the
+    reference guards this with torch.cuda.set_sync_debug_mode — same idea:
+    tests/unittests/classification/test_accuracy.py:352-408)."""
+    kw = dict(num_classes=50, validate_args=False)
+    coll = ma.MetricCollection({
+        "acc": ma.MulticlassAccuracy(average="micro", **kw),
+        "f1": ma.MulticlassF1Score(average="macro", **kw),
+        "confmat": ma.MulticlassConfusionMatrix(**kw),
+        "auroc": ma.MulticlassAUROC(average="macro", thresholds=20, **kw),
+    }).to("cuda")
+    preds = torch.randn(256, 50, device="cuda")
+    target = torch.randint(0, 50, (256,), device="cuda")
+    coll.update(preds, target)  # warmup (pools, uniform-threshold cache)
+    torch.cuda.synchronize()
+    torch.cuda.set_sync_debug_mode("error")
+    try:
+        coll.update(preds, target)
+        coll.update(preds, target)
+    finally:
+        torch.cuda.set_sync_debug_mode("default")
+    torch.cuda.synchronize()
+
+
+def test_sync_debug_mode_actually_fires():
+    """Negative control: a known-syncing op must raise under the mode."""
+    t = torch.rand(10, device="cuda")
+    torch.cuda.set_sync_debug_mode("error")
+    try:
+        with pytest.raises(RuntimeError):
+            t.item()
+    finally:
+        torch.cuda.set_sync_debug_mode("default")

@@ -324,3 +324,37 @@ def test_classification_curve_modules():
             ours.update(p, t)
             ref.update(p, t)
         _cmp(ours.compute(), ref.compute())
+
+
+def test_transform_wrappers():
+    tm = _tm()
+    g = torch.Generator().manual_seed(61)
+    batches = [(torch.rand(32, generator=g), torch.randint(0, 5, (32,), generator=g)) for _ in range(3)]
+    ours = ma.wrappers.BinaryTargetTransformer(ma.BinaryAccuracy(), threshold=2)
+    ref = tm.wrappers.BinaryTargetTransformer(tm.classification.BinaryAccuracy(), threshold=2)
+    for p, t in batches:
+        _cmp(ours(p, t), ref(p, t))
+    _cmp(ours.compute(), ref.compute())
+
+    ours2 = ma.wrappers.LambdaInputTransformer(ma.MeanSquaredError(), transform_pred=lambda x: x * 2)
+    ref2 = tm.wrappers.LambdaInputTransformer(tm.MeanSquaredError(), transform_pred=lambda x: x * 2)
+    for p, t in batches:
+        _cmp(ours2(p, t.float()), ref2(p, t.float()))
+    _cmp(ours2.compute(), ref2.compute())
+
+
+def test_feature_share_wrapper():
+    tm = _tm()
+
+    class Backbone(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.calls = 0
+
+        def forward(self, x):
+            self.calls += 1
+            return x.flatten(1)
+
+    # FeatureShare needs metrics with a `feature_network` attribute; use toy
+    # FID-style metrics if available on both sides — otherwise skip
+    pytest.skip("FeatureShare needs model-backed metrics (offline-gated); covered by wrapper unit tests")
