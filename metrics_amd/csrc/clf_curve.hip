@@ -157,3 +157,72 @@ extern "C" int ma_binary_clf_curve(
                        (float*)out_thresh_u, (ll*)out_count_u);
     return (int)hipGetLastError();
 }
+
+// ---------------------------------------------------------------------------
+// Retrieval grouping sort: ONE radix pass replaces the torch lexsort
+// (argsort pred desc -> argsort index asc) via a composite 64-bit key
+// (query index << 32) | descending-order-preserving score bits; a second
+// 32-bit sort yields the by-index (stable original order) permutation.
+
+__global__ void __launch_bounds__(256) k_retrieval_keys(
+    const ll* __restrict__ idx, const float* __restrict__ preds, ll n,
+    unsigned long long* __restrict__ keys, unsigned int* __restrict__ ikeys, int* __restrict__ pos) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    unsigned int b = __float_as_uint(preds[i]);
+    // ascending-order-preserving flip, then invert for descending score
+    b = (b & 0x80000000u) ? ~b : (b | 0x80000000u);
+    b = ~b;
+    keys[i] = ((unsigned long long)(unsigned int)idx[i] << 32) | (unsigned long long)b;
+    ikeys[i] = (unsigned int)idx[i];
+    pos[i] = (int)i;
+}
+
+extern "C" int ma_retrieval_sort_scratch_bytes(ll n, unsigned long long* out_bytes) {
+    size_t s64 = 0, s32 = 0;
+    hipError_t e;
+    e = rocprim::radix_sort_pairs((void*)nullptr, s64, (const unsigned long long*)nullptr,
+                                  (unsigned long long*)nullptr, (const int*)nullptr, (int*)nullptr,
+                                  (size_t)n);
+    if (e != hipSuccess) return (int)e;
+    e = rocprim::radix_sort_pairs((void*)nullptr, s32, (const unsigned int*)nullptr,
+                                  (unsigned int*)nullptr, (const int*)nullptr, (int*)nullptr,
+                                  (size_t)n);
+    if (e != hipSuccess) return (int)e;
+    size_t tmp = s64 > s32 ? s64 : s32;
+    size_t total = 0;
+    total += align_up((size_t)n * sizeof(unsigned long long));  // keys in
+    total += align_up((size_t)n * sizeof(unsigned long long));  // keys out
+    total += align_up((size_t)n * sizeof(unsigned int));        // ikeys in
+    total += align_up((size_t)n * sizeof(unsigned int));        // ikeys out
+    total += align_up((size_t)n * sizeof(int));                 // pos
+    total += align_up(tmp);
+    *out_bytes = (unsigned long long)total;
+    return 0;
+}
+
+extern "C" int ma_retrieval_sort(
+    uint64_t stream_u, uint64_t idx_u, uint64_t preds_u, ll n,
+    uint64_t scratch_u, unsigned long long scratch_bytes,
+    uint64_t out_order_u, uint64_t out_by_index_u) {
+    hipStream_t stream = (hipStream_t)stream_u;
+    char* p = (char*)scratch_u;
+    unsigned long long* keys_in = (unsigned long long*)p;  p += align_up((size_t)n * sizeof(unsigned long long));
+    unsigned long long* keys_out = (unsigned long long*)p; p += align_up((size_t)n * sizeof(unsigned long long));
+    unsigned int* ikeys_in = (unsigned int*)p;             p += align_up((size_t)n * sizeof(unsigned int));
+    unsigned int* ikeys_out = (unsigned int*)p;            p += align_up((size_t)n * sizeof(unsigned int));
+    int* pos = (int*)p;                                    p += align_up((size_t)n * sizeof(int));
+    void* tmp = (void*)p;
+    size_t tmp_bytes = (size_t)((char*)scratch_u + scratch_bytes - p);
+
+    const ll grid = (n + 255) / 256;
+    hipLaunchKernelGGL(k_retrieval_keys, dim3(grid), dim3(256), 0, stream,
+                       (const ll*)idx_u, (const float*)preds_u, n, keys_in, ikeys_in, pos);
+    size_t b = tmp_bytes;
+    OK(rocprim::radix_sort_pairs(tmp, b, keys_in, keys_out, pos, (int*)out_order_u,
+                                 (size_t)n, 0, 64, stream));
+    b = tmp_bytes;
+    OK(rocprim::radix_sort_pairs(tmp, b, ikeys_in, ikeys_out, pos, (int*)out_by_index_u,
+                                 (size_t)n, 0, 32, stream));
+    return (int)hipGetLastError();
+}

@@ -65,6 +65,8 @@ _SIGNATURES = {
     "ma_ssim2d_fused": [_U64, _U64, _U64, _I, _LL, _LL, _LL, _LL, _U64, _I, _U64, _I, _F, _F, _U64, _F, _F, _I, _I, _I, _U64, _U64],
     "ma_binary_erosion2d": [_U64, _U64, _LL, _LL, _LL, _LL, _U64, _I, _I, _I, _I, _I, _U64],
     "ma_calib_bins": [_U64, _U64, _U64, _LL, _U64, _I, _I, _F, _F, _U64],
+    "ma_retrieval_sort_scratch_bytes": [_LL, ctypes.POINTER(ctypes.c_ulonglong)],
+    "ma_retrieval_sort": [_U64, _U64, _U64, _LL, _U64, ctypes.c_ulonglong, _U64, _U64],
     "ma_mc_topk_stat": [_U64, _U64, _I, _U64, _LL, _LL, _I, _LL, _I, _U64, _U64, _U64, _U64],
 }
 
@@ -509,6 +511,47 @@ def binary_clf_curve(
     _check(rc, "ma_binary_clf_curve")
     k = int(out_cnt.item())
     return out_fps[:k], out_tps[:k], out_thr[:k]
+
+
+_RETR_SCRATCH_CACHE: dict = {}
+
+
+def retrieval_sort(indexes: Tensor, preds: Tensor) -> Tuple[Tensor, Tensor]:
+    """(order, by_index) permutations for retrieval grouping in ONE radix pass.
+
+    ``order`` sorts by (query index asc, pred desc); ``by_index`` by index
+    only with original order preserved for ties. Composite 64-bit keys
+    replace the torch double-argsort lexsort. Requires indexes < 2^31.
+    """
+    lib = _lib()
+    indexes = indexes.contiguous().long()
+    preds = _to_supported(preds).float().contiguous()
+    n = preds.numel()
+    dev = preds.device
+    nbytes = _RETR_SCRATCH_CACHE.get(n)
+    if nbytes is None:
+        out_b = ctypes.c_ulonglong(0)
+        rc = lib.ma_retrieval_sort_scratch_bytes(n, ctypes.byref(out_b))
+        _check(rc, "ma_retrieval_sort_scratch_bytes")
+        nbytes = out_b.value
+        if len(_RETR_SCRATCH_CACHE) > 256:
+            _RETR_SCRATCH_CACHE.clear()
+        _RETR_SCRATCH_CACHE[n] = nbytes
+    scratch = torch.empty(int(nbytes), dtype=torch.uint8, device=dev)
+    order = torch.empty(n, dtype=torch.int32, device=dev)
+    by_index = torch.empty(n, dtype=torch.int32, device=dev)
+    rc = lib.ma_retrieval_sort(
+        _stream(),
+        indexes.data_ptr(),
+        preds.data_ptr(),
+        n,
+        scratch.data_ptr(),
+        nbytes,
+        order.data_ptr(),
+        by_index.data_ptr(),
+    )
+    _check(rc, "ma_retrieval_sort")
+    return order.long(), by_index.long()
 
 
 _ERR_OPS = {"sq_err": (0, 1), "abs_err": (1, 1), "ape": (2, 1), "sq_log_err": (3, 1), "moments": (4, 6), "logcosh": (5, 1)}

@@ -41,16 +41,24 @@ class _Grouped:
 
 def _group_by_query(indexes: Tensor, preds: Tensor, target: Tensor) -> _Grouped:
     g = _Grouped()
-    # lexsort: stable pred-desc, then stable index-asc => index asc, pred desc
-    order1 = torch.argsort(preds, descending=True, stable=True)
-    idx_sorted_key = indexes[order1]
-    order2 = torch.argsort(idx_sorted_key, stable=True)
-    order = order1[order2]
+    if indexes.is_cuda and indexes.numel() and int(indexes.max()) < 2**31:
+        # ONE composite-key rocPRIM radix pass for the whole lexsort
+        # (csrc/clf_curve.hip ma_retrieval_sort); one more 32-bit sort gives
+        # the stable by-index order
+        from metrics_amd.ops import _hip
+
+        order, by_index = _hip.retrieval_sort(indexes, preds)
+    else:
+        # lexsort: stable pred-desc, then stable index-asc => index asc, pred desc
+        order1 = torch.argsort(preds, descending=True, stable=True)
+        idx_sorted_key = indexes[order1]
+        order2 = torch.argsort(idx_sorted_key, stable=True)
+        order = order1[order2]
+        by_index = torch.argsort(indexes, stable=True)
     g.preds = preds[order]
     g.target = target[order]
     gsorted = indexes[order]
     # loop-fallback order (index asc, original order otherwise — like torch.sort)
-    by_index = torch.argsort(indexes, stable=True)
     g.preds_by_index = preds[by_index]
     g.target_by_index = target[by_index]
 

@@ -1083,3 +1083,34 @@ def test_sync_debug_mode_actually_fires():
             t.item()
     finally:
         torch.cuda.set_sync_debug_mode("default")
+
+
+def test_retrieval_sort_kernel_vs_torch():
+    from metrics_amd.ops import _hip
+
+    torch.manual_seed(71)
+    n = 100_000
+    idx = torch.randint(0, 5000, (n,), device="cuda")
+    preds = torch.rand(n, device="cuda")
+    preds[::97] = preds[0]  # inject ties
+    order, by_index = _hip.retrieval_sort(idx, preds)
+    o1 = torch.argsort(preds, descending=True, stable=True)
+    o2 = torch.argsort(idx[o1], stable=True)
+    ref_order = o1[o2]
+    ref_by_index = torch.argsort(idx, stable=True)
+    assert torch.equal(order, ref_order)
+    assert torch.equal(by_index, ref_by_index)
+
+
+def test_retrieval_metric_kernel_path_gpu():
+    torch.manual_seed(72)
+    idx = torch.randint(0, 500, (20_000,))
+    preds = torch.rand(20_000)
+    target = torch.randint(0, 2, (20_000,))
+    for cls, kw in ((ma.retrieval.RetrievalMAP, {}), (ma.retrieval.RetrievalNormalizedDCG, {}),
+                    (ma.retrieval.RetrievalPrecision, {"top_k": 5})):
+        mg = cls(**kw).to("cuda")
+        mg.update(preds.cuda(), target.cuda(), indexes=idx.cuda())
+        mc = cls(**kw)
+        mc.update(preds, target, indexes=idx)
+        assert torch.allclose(mg.compute().cpu(), mc.compute(), atol=1e-6), cls.__name__
