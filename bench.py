@@ -225,7 +225,7 @@ def main() -> None:
                     "parallelism": f"dp{n_gpus}",
                     "num_classes": args.classes,
                     "batch_per_gpu": args.batch,
-                    "compute_every": args.compute_every,
+                    "compute_every": compute_every,
                     "curve_thresholds": args.curve_thresholds,
                     "validate_args": False,
                 },

@@ -341,3 +341,34 @@ def _test_retrieval_batched_ddp(rank, world_size):
 )
 def test_ddp_metric_parity(fn):
     run_distributed(fn, world_size=2)
+
+
+def _test_bench_collection_world8(rank, world_size):
+    """The exact bench.py metric mix at world_size 8 (tiny shapes): the
+    driver's 8-GPU scale run must be correct by construction."""
+    import bench as bench_mod
+
+    torch.manual_seed(100 + rank)
+    coll = bench_mod.build_collection(13, torch.device("cpu"), curve_thresholds=10)
+    preds = torch.randn(3, 16, 13)
+    target = torch.randint(0, 13, (3, 16))
+    for i in range(3):
+        coll.update(preds[i], target[i])
+    res = coll.compute()  # sync_on_compute over the world-8 group
+    assert set(res) == {
+        "acc_micro", "acc_macro", "precision", "recall", "f1", "fbeta2", "specificity",
+        "npv", "hamming", "jaccard", "exact_match", "cohen_kappa", "mcc", "confmat",
+        "auroc", "avg_precision",
+    }
+    for k, v in res.items():
+        assert torch.isfinite(v.float()).all(), k
+    # a second round: unsync restored local states, accumulation continues
+    coll.update(preds[0], target[0])
+    for m in coll.values(copy_state=False):
+        m._computed = None
+    res2 = coll.compute()
+    assert torch.isfinite(res2["acc_micro"].float()).all()
+
+
+def test_bench_collection_world8():
+    run_distributed(_test_bench_collection_world8, world_size=8)
