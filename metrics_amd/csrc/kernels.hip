@@ -997,6 +997,45 @@ __global__ void k_exact_apply(unsigned long long* __restrict__ scratch, ll C, ll
     }
 }
 
+// fused epilogue for the collection path: stat-delta apply + exact-match
+// accumulation in ONE single-block launch (replaces the back-to-back
+// k_exact_apply + k_apply_stat_deltas pair; saves a launch per step).
+__global__ void k_apply_stat_exact(
+    unsigned long long* __restrict__ scratch /* 3*C + 1 */, ll C, ll B,
+    ll* __restrict__ tp, ll* __restrict__ fp, ll* __restrict__ tn, ll* __restrict__ fn,
+    ll* __restrict__ correct, ll* __restrict__ total) {
+    __shared__ unsigned long long part[256];
+    __shared__ unsigned long long valid_s;
+    if (threadIdx.x == 0) valid_s = scratch[3 * C];
+    __syncthreads();
+    const ll valid = (ll)valid_s;
+    if (threadIdx.x == 0) scratch[3 * C] = 0;
+    unsigned long long acc = 0;
+    for (ll i = threadIdx.x; i < C; i += blockDim.x) {
+        const ll dtp = (ll)scratch[i];
+        const ll dfp = (ll)scratch[C + i];
+        const ll dfn = (ll)scratch[2 * C + i];
+        scratch[i] = 0;
+        scratch[C + i] = 0;
+        scratch[2 * C + i] = 0;
+        acc += (unsigned long long)dtp;
+        tp[i] += dtp;
+        fp[i] += dfp;
+        fn[i] += dfn;
+        tn[i] += valid - dtp - dfp - dfn;
+    }
+    part[threadIdx.x] = acc;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if (threadIdx.x < off) part[threadIdx.x] += part[threadIdx.x + off];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        correct[0] += (ll)part[0] + (B - valid);
+        total[0] += B;
+    }
+}
+
 // one-launch stat-score compute: every precision/recall/accuracy/F-beta/
 // specificity/NPV/hamming reduction is post(safe_div(n.s, d.s)) with linear
 // coefficients over (tp,fp,tn,fn), then micro / macro / weighted averaging —
@@ -1542,6 +1581,15 @@ int ma_curve_auc_from_confmat(uintptr_t stream, uintptr_t confmat, int T, ll C, 
     hipStream_t s = (hipStream_t)stream;
     k_curve_auc_from_confmat<<<(unsigned)C, 256, 0, s>>>((const ll*)confmat, T, C, mode,
                                                          (float*)out, (float*)weights);
+    return (int)hipGetLastError();
+}
+
+int ma_apply_stat_exact(uintptr_t stream, uintptr_t scratch, ll C, ll B, uintptr_t tp,
+                        uintptr_t fp, uintptr_t tn, uintptr_t fn, uintptr_t correct,
+                        uintptr_t total) {
+    hipStream_t s = (hipStream_t)stream;
+    k_apply_stat_exact<<<1, 256, 0, s>>>((unsigned long long*)scratch, C, B, (ll*)tp, (ll*)fp,
+                                         (ll*)tn, (ll*)fn, (ll*)correct, (ll*)total);
     return (int)hipGetLastError();
 }
 

@@ -56,6 +56,7 @@ _SIGNATURES = {
     "ma_curve_suffix": [_U64, _U64, _LL, _I, _I, _I, _U64, _U64],
     "ma_apply_stat_deltas": [_U64, _U64, _LL, _U64, _U64, _U64, _U64],
     "ma_exact_apply": [_U64, _U64, _LL, _LL, _I, _U64, _U64],
+    "ma_apply_stat_exact": [_U64, _U64, _LL, _LL, _U64, _U64, _U64, _U64, _U64, _U64],
     "ma_curve_auc_from_confmat": [_U64, _U64, _I, _LL, _I, _U64, _U64],
     "ma_linear_stat_compute": [_U64, _U64, _U64, _U64, _U64, _LL, _F, _F, _F, _F, _F, _F, _F, _F, _I, _I, _F, _F, _F, _U64],
     "ma_err_reduce": [_U64, _U64, _U64, _I, _LL, _I, _D, _U64, _I, _I, _U64],
@@ -1164,13 +1165,17 @@ def mc_fused_collection_update(
         )
         _check(rc, "ma_mc_stat_labels")
     if exact is not None:
+        # fused epilogue: stat-delta apply + exact-match in one launch
         correct, total = exact
-        rc = lib.ma_exact_apply(
-            _stream(), scratch.data_ptr(), C, B, 0, correct.data_ptr(), total.data_ptr()
+        rc = lib.ma_apply_stat_exact(
+            _stream(), scratch.data_ptr(), C, B,
+            tp.data_ptr(), fp.data_ptr(), tn.data_ptr(), fn.data_ptr(),
+            correct.data_ptr(), total.data_ptr(),
         )
-        _check(rc, "ma_exact_apply")
-    rc = lib.ma_apply_stat_deltas(
-        _stream(), scratch.data_ptr(), C,
-        tp.data_ptr(), fp.data_ptr(), tn.data_ptr(), fn.data_ptr(),
-    )
-    _check(rc, "ma_apply_stat_deltas")
+        _check(rc, "ma_apply_stat_exact")
+    else:
+        rc = lib.ma_apply_stat_deltas(
+            _stream(), scratch.data_ptr(), C,
+            tp.data_ptr(), fp.data_ptr(), tn.data_ptr(), fn.data_ptr(),
+        )
+        _check(rc, "ma_apply_stat_deltas")
