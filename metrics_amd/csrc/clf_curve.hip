@@ -226,3 +226,144 @@ extern "C" int ma_retrieval_sort(
                                  (size_t)n, 0, 32, stream));
     return (int)hipGetLastError();
 }
+
+// ---------------------------------------------------------------------------
+// Batched multiclass/multilabel exact curves: ONE composite-key sort for all
+// classes ((class << 32) | descending-score bits) replaces C per-class sorts,
+// and ONE (C,) count transfer replaces C data-dependent-size syncs. Each
+// class's segment is exactly B elements, so segment bases are implicit.
+
+__global__ void __launch_bounds__(256) k_mc_keys(
+    const float* __restrict__ probs /* (B, C) */, ll B, ll C,
+    unsigned long long* __restrict__ keys, int* __restrict__ pos) {
+    ll e = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    if (e >= B * C) return;
+    const ll c = e % C;  // element e = (row, c) row-major
+    unsigned int b = __float_as_uint(probs[e]);
+    b = (b & 0x80000000u) ? ~b : (b | 0x80000000u);
+    b = ~b;  // descending within class
+    keys[e] = ((unsigned long long)c << 32) | (unsigned long long)b;
+    pos[e] = (int)e;
+}
+
+__global__ void __launch_bounds__(256) k_mc_gather_flags(
+    const unsigned long long* __restrict__ keys_sorted, const int* __restrict__ pos_sorted,
+    const float* __restrict__ probs, const ll* __restrict__ target, ll B, ll C, ll n,
+    double* __restrict__ tvals, int* __restrict__ flags) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    const ll src = pos_sorted[i];
+    const ll row = src / C;
+    const ll cls = (ll)(keys_sorted[i] >> 32);
+    tvals[i] = (target[row] == cls) ? 1.0 : 0.0;
+    flags[i] = (i == n - 1) || (keys_sorted[i] != keys_sorted[i + 1]);
+}
+
+// multilabel variant: target is (B, C) 0/1
+__global__ void __launch_bounds__(256) k_ml_gather_flags(
+    const unsigned long long* __restrict__ keys_sorted, const int* __restrict__ pos_sorted,
+    const ll* __restrict__ target, ll n,
+    double* __restrict__ tvals, int* __restrict__ flags) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    tvals[i] = (double)target[pos_sorted[i]];
+    flags[i] = (i == n - 1) || (keys_sorted[i] != keys_sorted[i + 1]);
+}
+
+__global__ void __launch_bounds__(256) k_mc_compact(
+    const unsigned long long* __restrict__ keys_sorted, const int* __restrict__ pos_sorted,
+    const float* __restrict__ probs, const int* __restrict__ flags,
+    const int* __restrict__ pos_scan, const double* __restrict__ tps_full, ll B, ll C, ll n,
+    float* __restrict__ out_fps, float* __restrict__ out_tps, float* __restrict__ out_thr) {
+    ll i = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    if (!flags[i]) return;
+    const ll cls = (ll)(keys_sorted[i] >> 32);
+    const ll seg0 = cls * B;
+    const double base = (seg0 > 0) ? tps_full[seg0 - 1] : 0.0;
+    const double tp = tps_full[i] - base;
+    const int o = pos_scan[i] - 1;
+    out_tps[o] = (float)tp;
+    out_fps[o] = (float)((double)(i - seg0 + 1) - tp);
+    out_thr[o] = probs[pos_sorted[i]];
+}
+
+__global__ void __launch_bounds__(256) k_mc_counts(
+    const int* __restrict__ pos_scan, ll B, ll C, ll* __restrict__ counts) {
+    ll c = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    const int hi = pos_scan[(c + 1) * B - 1];
+    const int lo = (c > 0) ? pos_scan[c * B - 1] : 0;
+    counts[c] = hi - lo;
+}
+
+extern "C" int ma_mc_clf_curve_scratch_bytes(ll B, ll C, unsigned long long* out_bytes) {
+    const ll n = B * C;
+    size_t sort_tmp = 0, scan_tmp = 0, iscan_tmp = 0;
+    hipError_t e;
+    e = rocprim::radix_sort_pairs((void*)nullptr, sort_tmp, (const unsigned long long*)nullptr,
+                                  (unsigned long long*)nullptr, (const int*)nullptr, (int*)nullptr,
+                                  (size_t)n);
+    if (e != hipSuccess) return (int)e;
+    e = rocprim::inclusive_scan((void*)nullptr, scan_tmp, (const double*)nullptr,
+                                (double*)nullptr, (size_t)n, rocprim::plus<double>());
+    if (e != hipSuccess) return (int)e;
+    e = rocprim::inclusive_scan((void*)nullptr, iscan_tmp, (const int*)nullptr,
+                                (int*)nullptr, (size_t)n, rocprim::plus<int>());
+    if (e != hipSuccess) return (int)e;
+    size_t tmp = sort_tmp;
+    if (scan_tmp > tmp) tmp = scan_tmp;
+    if (iscan_tmp > tmp) tmp = iscan_tmp;
+    size_t total = 0;
+    total += align_up((size_t)n * sizeof(unsigned long long));  // keys in
+    total += align_up((size_t)n * sizeof(unsigned long long));  // keys out
+    total += align_up((size_t)n * sizeof(int));                 // pos in
+    total += align_up((size_t)n * sizeof(int));                 // pos out
+    total += align_up((size_t)n * sizeof(double));              // tvals/scan
+    total += align_up((size_t)n * sizeof(int));                 // flags
+    total += align_up((size_t)n * sizeof(int));                 // pos_scan
+    total += align_up(tmp);
+    *out_bytes = (unsigned long long)total;
+    return 0;
+}
+
+extern "C" int ma_mc_clf_curve(
+    uint64_t stream_u, uint64_t probs_u, uint64_t target_u, ll B, ll C, int multilabel,
+    uint64_t scratch_u, unsigned long long scratch_bytes,
+    uint64_t out_fps_u, uint64_t out_tps_u, uint64_t out_thr_u, uint64_t out_counts_u) {
+    hipStream_t stream = (hipStream_t)stream_u;
+    const ll n = B * C;
+    char* p = (char*)scratch_u;
+    unsigned long long* keys_in = (unsigned long long*)p;  p += align_up((size_t)n * sizeof(unsigned long long));
+    unsigned long long* keys_out = (unsigned long long*)p; p += align_up((size_t)n * sizeof(unsigned long long));
+    int* pos_in = (int*)p;                                 p += align_up((size_t)n * sizeof(int));
+    int* pos_out = (int*)p;                                p += align_up((size_t)n * sizeof(int));
+    double* tvals = (double*)p;                            p += align_up((size_t)n * sizeof(double));
+    int* flags = (int*)p;                                  p += align_up((size_t)n * sizeof(int));
+    int* pos_scan = (int*)p;                               p += align_up((size_t)n * sizeof(int));
+    void* tmp = (void*)p;
+    size_t tmp_bytes = (size_t)((char*)scratch_u + scratch_bytes - p);
+
+    const ll grid = (n + 255) / 256;
+    hipLaunchKernelGGL(k_mc_keys, dim3(grid), dim3(256), 0, stream,
+                       (const float*)probs_u, B, C, keys_in, pos_in);
+    size_t b = tmp_bytes;
+    OK(rocprim::radix_sort_pairs(tmp, b, keys_in, keys_out, pos_in, pos_out, (size_t)n, 0, 64, stream));
+    if (multilabel)
+        hipLaunchKernelGGL(k_ml_gather_flags, dim3(grid), dim3(256), 0, stream,
+                           keys_out, pos_out, (const ll*)target_u, n, tvals, flags);
+    else
+        hipLaunchKernelGGL(k_mc_gather_flags, dim3(grid), dim3(256), 0, stream,
+                           keys_out, pos_out, (const float*)probs_u, (const ll*)target_u, B, C, n,
+                           tvals, flags);
+    b = tmp_bytes;
+    OK(rocprim::inclusive_scan(tmp, b, tvals, tvals, (size_t)n, rocprim::plus<double>(), stream));
+    b = tmp_bytes;
+    OK(rocprim::inclusive_scan(tmp, b, flags, pos_scan, (size_t)n, rocprim::plus<int>(), stream));
+    hipLaunchKernelGGL(k_mc_compact, dim3(grid), dim3(256), 0, stream,
+                       keys_out, pos_out, (const float*)probs_u, flags, pos_scan, tvals, B, C, n,
+                       (float*)out_fps_u, (float*)out_tps_u, (float*)out_thr_u);
+    hipLaunchKernelGGL(k_mc_counts, dim3((C + 255) / 256), dim3(256), 0, stream,
+                       pos_scan, B, C, (ll*)out_counts_u);
+    return (int)hipGetLastError();
+}

@@ -165,6 +165,7 @@ def _binary_precision_recall_curve_compute(
     state: Union[Tensor, Tuple[Tensor, Tensor]],
     thresholds: Optional[Tensor],
     pos_label: int = 1,
+    precomputed: Optional[Tuple[Tensor, Tensor, Tensor]] = None,
 ) -> Tuple[Tensor, Tensor, Tensor]:
     if isinstance(state, Tensor) and thresholds is not None:
         tps = state[:, 1, 1]
@@ -176,7 +177,10 @@ def _binary_precision_recall_curve_compute(
         recall = torch.cat([recall, torch.zeros(1, dtype=recall.dtype, device=recall.device)])
         return precision, recall, thresholds
 
-    fps, tps, thresholds = _binary_clf_curve(state[0], state[1], pos_label=pos_label)
+    if precomputed is not None:
+        fps, tps, thresholds = precomputed
+    else:
+        fps, tps, thresholds = _binary_clf_curve(state[0], state[1], pos_label=pos_label)
     precision = tps / (tps + fps)
     recall = tps / tps[-1]
 
@@ -319,8 +323,16 @@ def _multiclass_precision_recall_curve_compute(
         tensor_state = True
     else:
         precision_list, recall_list, thres_list = [], [], []
+        # GPU: ONE composite-key sort yields every class's curve (K2 batched)
+        curves = (
+            ops.hip_mc_clf_curve(state[0], state[1])
+            if state[0].is_cuda and state[0].numel()
+            else [None] * num_classes
+        )
         for i in range(num_classes):
-            res = _binary_precision_recall_curve_compute((state[0][:, i], state[1]), thresholds=None, pos_label=i)
+            res = _binary_precision_recall_curve_compute(
+                (state[0][:, i], state[1]), thresholds=None, pos_label=i, precomputed=curves[i]
+            )
             precision_list.append(res[0])
             recall_list.append(res[1])
             thres_list.append(res[2])
@@ -454,6 +466,11 @@ def _multilabel_precision_recall_curve_compute(
         return precision.T, recall.T, thresholds
 
     precision_list, recall_list, thres_list = [], [], []
+    curves = (
+        ops.hip_mc_clf_curve(state[0], state[1], multilabel=True)
+        if state[0].is_cuda and state[0].numel() and ignore_index is None
+        else [None] * num_labels
+    )
     for i in range(num_labels):
         preds = state[0][:, i]
         target = state[1][:, i]
@@ -461,7 +478,9 @@ def _multilabel_precision_recall_curve_compute(
             idx = target != ignore_index
             preds = preds[idx]
             target = target[idx]
-        res = _binary_precision_recall_curve_compute((preds, target), thresholds=None, pos_label=1)
+        res = _binary_precision_recall_curve_compute(
+            (preds, target), thresholds=None, pos_label=1, precomputed=curves[i]
+        )
         precision_list.append(res[0])
         recall_list.append(res[1])
         thres_list.append(res[2])

@@ -30,6 +30,7 @@ def _binary_roc_compute(
     state: Union[Tensor, Tuple[Tensor, Tensor]],
     thresholds: Optional[Tensor],
     pos_label: int = 1,
+    precomputed=None,
 ) -> Tuple[Tensor, Tensor, Tensor]:
     if isinstance(state, Tensor) and thresholds is not None:
         tps = state[:, 1, 1]
@@ -40,7 +41,10 @@ def _binary_roc_compute(
         fpr = _safe_divide(fps, fps + tns).flip(0)
         thres = thresholds.flip(0)
     else:
-        fps, tps, thres = _binary_clf_curve(preds=state[0], target=state[1], pos_label=pos_label)
+        if precomputed is not None:
+            fps, tps, thres = precomputed
+        else:
+            fps, tps, thres = _binary_clf_curve(preds=state[0], target=state[1], pos_label=pos_label)
         # add extra threshold position so that the curve starts at (0, 0)
         tps = torch.cat([torch.zeros(1, dtype=tps.dtype, device=tps.device), tps])
         fps = torch.cat([torch.zeros(1, dtype=fps.dtype, device=fps.device), fps])
@@ -110,8 +114,17 @@ def _multiclass_roc_compute(
         tensor_state = True
     else:
         fpr_list, tpr_list, thres_list = [], [], []
+        from metrics_amd import ops as _ops
+
+        curves = (
+            _ops.hip_mc_clf_curve(state[0], state[1])
+            if state[0].is_cuda and state[0].numel()
+            else [None] * num_classes
+        )
         for i in range(num_classes):
-            res = _binary_roc_compute((state[0][:, i], state[1]), thresholds=None, pos_label=i)
+            res = _binary_roc_compute(
+                (state[0][:, i], state[1]), thresholds=None, pos_label=i, precomputed=curves[i]
+            )
             fpr_list.append(res[0])
             tpr_list.append(res[1])
             thres_list.append(res[2])
@@ -175,6 +188,13 @@ def _multilabel_roc_compute(
         return fpr, tpr, thres
 
     fpr_list, tpr_list, thres_list = [], [], []
+    from metrics_amd import ops as _ops
+
+    curves = (
+        _ops.hip_mc_clf_curve(state[0], state[1], multilabel=True)
+        if state[0].is_cuda and state[0].numel() and ignore_index is None
+        else [None] * num_labels
+    )
     for i in range(num_labels):
         preds = state[0][:, i]
         target = state[1][:, i]
@@ -182,7 +202,7 @@ def _multilabel_roc_compute(
             idx = target != ignore_index
             preds = preds[idx]
             target = target[idx]
-        res = _binary_roc_compute((preds, target), thresholds=None, pos_label=1)
+        res = _binary_roc_compute((preds, target), thresholds=None, pos_label=1, precomputed=curves[i])
         fpr_list.append(res[0])
         tpr_list.append(res[1])
         thres_list.append(res[2])

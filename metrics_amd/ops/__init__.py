@@ -24,6 +24,14 @@ def hip_bincount(x: Tensor, minlength: int) -> Tensor:
     return _hip.bincount(x, minlength)
 
 
+def hip_mc_clf_curve(probs: Tensor, target: Tensor, multilabel: bool = False):
+    """Batched per-class exact curves (one composite-key sort) — GPU only."""
+    out = _hip.mc_clf_curve(probs, target, multilabel)
+    if probs.dtype != torch.float32:
+        out = [(f, t, thr.to(probs.dtype)) for f, t, thr in out]
+    return out
+
+
 def hip_binary_clf_curve(
     preds: Tensor, target: Tensor, weights: Optional[Tensor] = None, pos_label: int = 1
 ) -> Tuple[Tensor, Tensor, Tensor]:

@@ -66,6 +66,8 @@ _SIGNATURES = {
     "ma_ssim2d_fused": [_U64, _U64, _U64, _I, _LL, _LL, _LL, _LL, _U64, _I, _U64, _I, _F, _F, _U64, _F, _F, _I, _I, _I, _U64, _U64],
     "ma_binary_erosion2d": [_U64, _U64, _LL, _LL, _LL, _LL, _U64, _I, _I, _I, _I, _I, _U64],
     "ma_calib_bins": [_U64, _U64, _U64, _LL, _U64, _I, _I, _F, _F, _U64],
+    "ma_mc_clf_curve_scratch_bytes": [_LL, _LL, ctypes.POINTER(ctypes.c_ulonglong)],
+    "ma_mc_clf_curve": [_U64, _U64, _U64, _LL, _LL, _I, _U64, ctypes.c_ulonglong, _U64, _U64, _U64, _U64],
     "ma_retrieval_sort_scratch_bytes": [_LL, ctypes.POINTER(ctypes.c_ulonglong)],
     "ma_retrieval_sort": [_U64, _U64, _U64, _LL, _U64, ctypes.c_ulonglong, _U64, _U64],
     "ma_mc_topk_stat": [_U64, _U64, _I, _U64, _LL, _LL, _I, _LL, _I, _U64, _U64, _U64, _U64],
@@ -512,6 +514,64 @@ def binary_clf_curve(
     _check(rc, "ma_binary_clf_curve")
     k = int(out_cnt.item())
     return out_fps[:k], out_tps[:k], out_thr[:k]
+
+
+_MC_CURVE_SCRATCH_CACHE: dict = {}
+
+
+def mc_clf_curve(probs: Tensor, target: Tensor, multilabel: bool = False):
+    """Per-class exact curves for ALL classes in one composite-key sort.
+
+    ``probs`` (B, C); ``target`` (B,) class ids (multiclass) or (B, C) 0/1
+    (multilabel). Returns a list of C ``(fps, tps, thresholds)`` tuples —
+    one device sort + ONE host count transfer instead of C sorts + C syncs.
+    """
+    lib = _lib()
+    probs = _to_supported(probs)
+    if probs.dtype != torch.float32:
+        probs = probs.float()
+    probs = probs.contiguous()
+    target = target.contiguous().long()
+    B, C = probs.shape
+    n = B * C
+    dev = probs.device
+    nbytes = _MC_CURVE_SCRATCH_CACHE.get((B, C))
+    if nbytes is None:
+        out_b = ctypes.c_ulonglong(0)
+        rc = lib.ma_mc_clf_curve_scratch_bytes(B, C, ctypes.byref(out_b))
+        _check(rc, "ma_mc_clf_curve_scratch_bytes")
+        nbytes = out_b.value
+        if len(_MC_CURVE_SCRATCH_CACHE) > 256:
+            _MC_CURVE_SCRATCH_CACHE.clear()
+        _MC_CURVE_SCRATCH_CACHE[(B, C)] = nbytes
+    scratch = torch.empty(int(nbytes), dtype=torch.uint8, device=dev)
+    out_fps = torch.empty(n, dtype=torch.float32, device=dev)
+    out_tps = torch.empty(n, dtype=torch.float32, device=dev)
+    out_thr = torch.empty(n, dtype=torch.float32, device=dev)
+    out_counts = torch.zeros(C, dtype=torch.long, device=dev)
+    rc = lib.ma_mc_clf_curve(
+        _stream(),
+        probs.data_ptr(),
+        target.data_ptr(),
+        B,
+        C,
+        1 if multilabel else 0,
+        scratch.data_ptr(),
+        nbytes,
+        out_fps.data_ptr(),
+        out_tps.data_ptr(),
+        out_thr.data_ptr(),
+        out_counts.data_ptr(),
+    )
+    _check(rc, "ma_mc_clf_curve")
+    counts = out_counts.cpu().tolist()  # the single host transfer
+    res = []
+    off = 0
+    for c in range(C):
+        k = counts[c]
+        res.append((out_fps[off : off + k], out_tps[off : off + k], out_thr[off : off + k]))
+        off += k
+    return res
 
 
 _RETR_SCRATCH_CACHE: dict = {}

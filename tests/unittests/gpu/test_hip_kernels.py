@@ -1114,3 +1114,52 @@ def test_retrieval_metric_kernel_path_gpu():
         mc = cls(**kw)
         mc.update(preds, target, indexes=idx)
         assert torch.allclose(mg.compute().cpu(), mc.compute(), atol=1e-6), cls.__name__
+
+
+def test_mc_clf_curve_batched_vs_per_class():
+    """One composite-key sort must reproduce the per-class K2 curves exactly
+    (multiclass AND multilabel), including heavy score ties."""
+    from metrics_amd.ops import _hip
+
+    torch.manual_seed(81)
+    B, C = 3000, 9
+    probs = ((torch.rand(B, C, device="cuda") * 50).round() / 50)
+    target = torch.randint(0, C, (B,), device="cuda")
+    batched = _hip.mc_clf_curve(probs, target, multilabel=False)
+    for c in range(C):
+        f, t, th = _hip.binary_clf_curve(probs[:, c].contiguous(), target, pos_label=c)
+        bf, bt, bth = batched[c]
+        assert torch.equal(bf, f) and torch.equal(bt, t) and torch.equal(bth, th), c
+    # multilabel
+    ml_t = torch.randint(0, 2, (B, C), device="cuda")
+    batched_ml = _hip.mc_clf_curve(probs, ml_t, multilabel=True)
+    for c in range(C):
+        f, t, th = _hip.binary_clf_curve(probs[:, c].contiguous(), ml_t[:, c].contiguous(), pos_label=1)
+        bf, bt, bth = batched_ml[c]
+        assert torch.equal(bf, f) and torch.equal(bt, t) and torch.equal(bth, th), c
+
+
+def test_exact_multiclass_roc_prc_gpu_batched():
+    torch.manual_seed(82)
+    preds = torch.randn(1500, 6).softmax(-1)
+    target = torch.randint(0, 6, (1500,))
+    for cls, kw in [
+        (ma.MulticlassROC, {"num_classes": 6, "thresholds": None}),
+        (ma.MulticlassPrecisionRecallCurve, {"num_classes": 6, "thresholds": None}),
+        (ma.MultilabelROC, {"num_labels": 6, "thresholds": None}),
+    ]:
+        if "num_labels" in kw:
+            tgt = torch.randint(0, 2, (1500, 6))
+        else:
+            tgt = target
+        mg = cls(**kw).to("cuda")
+        mg.update(preds.cuda(), tgt.cuda())
+        mc = cls(**kw)
+        mc.update(preds, tgt)
+        rg, rc_ = mg.compute(), mc.compute()
+        for a, b in zip(rg, rc_):
+            if isinstance(a, list):
+                for x, y in zip(a, b):
+                    assert torch.allclose(x.cpu(), y, atol=1e-6), cls.__name__
+            else:
+                assert torch.allclose(a.cpu(), b, atol=1e-6), cls.__name__
