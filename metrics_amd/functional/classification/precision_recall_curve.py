@@ -178,7 +178,7 @@ def _binary_precision_recall_curve_compute(
         return precision, recall, thresholds
 
     if precomputed is not None:
-        fps, tps, thresholds = precomputed
+        fps, tps, thresholds = precomputed[:3]
     else:
         fps, tps, thresholds = _binary_clf_curve(state[0], state[1], pos_label=pos_label)
     precision = tps / (tps + fps)

@@ -41,8 +41,9 @@ def _binary_roc_compute(
         fpr = _safe_divide(fps, fps + tns).flip(0)
         thres = thresholds.flip(0)
     else:
+        known_pos = known_neg = None
         if precomputed is not None:
-            fps, tps, thres = precomputed
+            fps, tps, thres, known_pos, known_neg = precomputed
         else:
             fps, tps, thres = _binary_clf_curve(preds=state[0], target=state[1], pos_label=pos_label)
         # add extra threshold position so that the curve starts at (0, 0)
@@ -50,7 +51,9 @@ def _binary_roc_compute(
         fps = torch.cat([torch.zeros(1, dtype=fps.dtype, device=fps.device), fps])
         thres = torch.cat([torch.ones(1, dtype=thres.dtype, device=thres.device), thres])
 
-        if fps[-1] <= 0:
+        # fps[-1] == #negatives and tps[-1] == #positives: with the batched
+        # curve path those counts are already on the host — no device sync
+        if (known_neg <= 0) if known_neg is not None else (fps[-1] <= 0):
             rank_zero_warn(
                 "No negative samples in targets, false positive value should be meaningless."
                 " Returning zero tensor in false positive score",
@@ -60,7 +63,7 @@ def _binary_roc_compute(
         else:
             fpr = fps / fps[-1]
 
-        if tps[-1] <= 0:
+        if (known_pos <= 0) if known_pos is not None else (tps[-1] <= 0):
             rank_zero_warn(
                 "No positive samples in targets, true positive value should be meaningless."
                 " Returning zero tensor in true positive score",

@@ -28,7 +28,7 @@ def hip_mc_clf_curve(probs: Tensor, target: Tensor, multilabel: bool = False):
     """Batched per-class exact curves (one composite-key sort) — GPU only."""
     out = _hip.mc_clf_curve(probs, target, multilabel)
     if probs.dtype != torch.float32:
-        out = [(f, t, thr.to(probs.dtype)) for f, t, thr in out]
+        out = [(f, t, thr.to(probs.dtype), p, n) for f, t, thr, p, n in out]
     return out
 
 

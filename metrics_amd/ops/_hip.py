@@ -564,12 +564,24 @@ def mc_clf_curve(probs: Tensor, target: Tensor, multilabel: bool = False):
         out_counts.data_ptr(),
     )
     _check(rc, "ma_mc_clf_curve")
-    counts = out_counts.cpu().tolist()  # the single host transfer
+    # per-class positive counts ride the same single host transfer: the
+    # callers' "no positive / no negative samples" branches then need no
+    # further syncs (neg_c = B - pos_c)
+    if multilabel:
+        pos = target.sum(0)
+    else:
+        pos = torch.bincount(target.clamp(min=0), minlength=C)[:C]
+    packed = torch.cat([out_counts, pos]).cpu()
+    counts = packed[:C].tolist()
+    pos_counts = packed[C:].tolist()
     res = []
     off = 0
     for c in range(C):
         k = counts[c]
-        res.append((out_fps[off : off + k], out_tps[off : off + k], out_thr[off : off + k]))
+        res.append(
+            (out_fps[off : off + k], out_tps[off : off + k], out_thr[off : off + k],
+             int(pos_counts[c]), B - int(pos_counts[c]))
+        )
         off += k
     return res
 

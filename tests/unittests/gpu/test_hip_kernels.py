@@ -1128,15 +1128,17 @@ def test_mc_clf_curve_batched_vs_per_class():
     batched = _hip.mc_clf_curve(probs, target, multilabel=False)
     for c in range(C):
         f, t, th = _hip.binary_clf_curve(probs[:, c].contiguous(), target, pos_label=c)
-        bf, bt, bth = batched[c]
+        bf, bt, bth, n_pos, n_neg = batched[c]
         assert torch.equal(bf, f) and torch.equal(bt, t) and torch.equal(bth, th), c
+        assert n_pos == int((target == c).sum()) and n_pos + n_neg == B
     # multilabel
     ml_t = torch.randint(0, 2, (B, C), device="cuda")
     batched_ml = _hip.mc_clf_curve(probs, ml_t, multilabel=True)
     for c in range(C):
         f, t, th = _hip.binary_clf_curve(probs[:, c].contiguous(), ml_t[:, c].contiguous(), pos_label=1)
-        bf, bt, bth = batched_ml[c]
+        bf, bt, bth, n_pos, n_neg = batched_ml[c]
         assert torch.equal(bf, f) and torch.equal(bt, t) and torch.equal(bth, th), c
+        assert n_pos == int(ml_t[:, c].sum()) and n_pos + n_neg == B
 
 
 def test_exact_multiclass_roc_prc_gpu_batched():
