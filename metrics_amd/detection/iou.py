@@ -61,42 +61,61 @@ class IntersectionOverUnion(Metric):
             raise ValueError("Expected argument `respect_labels` to be a boolean")
         self.respect_labels = respect_labels
 
+        self.add_state("groundtruth_labels", default=[], dist_reduce_fx=None)
         self.add_state("iou_matrix", default=[], dist_reduce_fx=None)
-        self.add_state("iou_label", default=[], dist_reduce_fx=None)
 
     def update(self, preds: List[Dict[str, Tensor]], target: List[Dict[str, Tensor]]) -> None:
-        """Accumulate per-pair IoU for matched (or all) box pairs."""
+        """Accumulate the label-masked all-pairs IoU matrices (reference
+        semantics: compute() means over every pair whose entry survived the
+        label/threshold masking — NOT a best-match assignment)."""
         _input_validator_iou(preds, target, ignore_score=True)
         for p, t in zip(preds, target):
             p_boxes, p_labels = p["boxes"].float(), p["labels"]
             t_boxes, t_labels = t["boxes"].float(), t["labels"]
-            if p_boxes.numel() == 0 or t_boxes.numel() == 0:
-                continue
-            iou = ops.box_iou_pairwise(p_boxes, t_boxes, self._iou_variant)
-            if self.respect_labels:
-                label_eq = p_labels.unsqueeze(1) == t_labels.unsqueeze(0)
-                iou[~label_eq] = self._invalid_val
+            self.groundtruth_labels.append(t_labels)
+            # reference empty-input quirks (functional/detection/iou.py:35-38):
+            # no preds -> zeros(M, M); no gts -> zeros(N, N)
+            if p_boxes.numel() == 0:
+                iou = torch.zeros(t_boxes.shape[0], t_boxes.shape[0], device=t_boxes.device)
+            elif t_boxes.numel() == 0:
+                iou = torch.zeros(p_boxes.shape[0], p_boxes.shape[0], device=p_boxes.device)
+            else:
+                iou = ops.box_iou_pairwise(p_boxes, t_boxes, self._iou_variant)
             if self.iou_threshold is not None:
                 iou[iou < self.iou_threshold] = self._invalid_val
-            valid = iou > self._invalid_val
-            # per prediction: best gt
-            best, best_idx = iou.max(dim=1)
-            keep = valid.any(dim=1)
-            self.iou_matrix.append(best[keep])
-            self.iou_label.append(t_labels[best_idx[keep]])
+            if self.respect_labels:
+                if p_boxes.numel() > 0 and t_boxes.numel() > 0:
+                    label_eq = p_labels.unsqueeze(1) == t_labels.unsqueeze(0)
+                else:
+                    label_eq = torch.eye(iou.shape[0], dtype=torch.bool, device=iou.device)
+                iou[~label_eq] = self._invalid_val
+            self.iou_matrix.append(iou)
 
     def compute(self) -> Dict[str, Tensor]:
-        """Mean (per-class) IoU."""
+        """Mean (per-class) IoU over all surviving pairs."""
         name = self._iou_variant
         if not self.iou_matrix:
-            out = {name: torch.tensor(0.0)}
-            return out
-        scores = torch.cat([x for x in self.iou_matrix]) if isinstance(self.iou_matrix, list) else self.iou_matrix
-        labels = torch.cat([x for x in self.iou_label]) if isinstance(self.iou_label, list) else self.iou_label
-        result = {name: scores.mean() if scores.numel() else torch.tensor(0.0)}
+            return {name: torch.tensor(0.0)}
+        valid = [m[m != self._invalid_val] for m in self.iou_matrix]
+        score = torch.cat(valid, 0).mean() if valid else torch.tensor(0.0)
+        result = {name: score}
+        if torch.isnan(score):
+            result[name] = torch.tensor(0.0, device=score.device)
         if self.class_metrics:
-            for c in labels.unique().tolist():
-                result[f"{name}/cl_{c}"] = scores[labels == c].mean()
+            from metrics_amd.utilities.data import dim_zero_cat
+
+            gt_labels = dim_zero_cat(self.groundtruth_labels)
+            classes = gt_labels.unique().tolist() if len(gt_labels) > 0 else []
+            for cl in classes:
+                masked = torch.zeros_like(score)
+                observed = torch.zeros_like(score)
+                for mat, gt_lab in zip(self.iou_matrix, self.groundtruth_labels):
+                    if mat.shape[-1] != gt_lab.shape[0]:
+                        continue  # empty-input square quirk: columns are not gts
+                    scores_cl = mat[:, gt_lab == cl]
+                    masked += scores_cl[scores_cl != self._invalid_val].sum()
+                    observed += (scores_cl != self._invalid_val).sum()
+                result[f"{name}/cl_{cl}"] = masked / observed
         return result
 
     def plot(self, val=None, ax=None):
@@ -107,7 +126,7 @@ class GeneralizedIntersectionOverUnion(IntersectionOverUnion):
     """Mean GIoU between matched boxes."""
 
     _iou_variant = "giou"
-    _invalid_val = -2.0
+    _invalid_val = -1.0
     plot_lower_bound: float = -1.0
 
 
@@ -115,7 +134,7 @@ class DistanceIntersectionOverUnion(IntersectionOverUnion):
     """Mean DIoU between matched boxes."""
 
     _iou_variant = "diou"
-    _invalid_val = -2.0
+    _invalid_val = -1.0
     plot_lower_bound: float = -1.0
 
 

@@ -52,6 +52,36 @@ def _box_convert(boxes, in_fmt, out_fmt):
     raise ValueError(in_fmt)
 
 
+def _box_iou_variant(a, b, variant):
+    iou = _box_iou(a, b)
+    lt_c = torch.min(a[:, None, :2], b[None, :, :2])
+    rb_c = torch.max(a[:, None, 2:], b[None, :, 2:])
+    whc = rb_c - lt_c
+    area1, area2 = _box_area(a), _box_area(b)
+    lt = torch.max(a[:, None, :2], b[None, :, :2])
+    rb = torch.min(a[:, None, 2:], b[None, :, 2:])
+    wh = (rb - lt).clamp(min=0)
+    union = area1[:, None] + area2[None, :] - wh[..., 0] * wh[..., 1]
+    if variant == "giou":
+        carea = whc[..., 0] * whc[..., 1]
+        return torch.where(carea > 0, iou - (carea - union) / carea, iou)
+    cdiag = whc[..., 0] ** 2 + whc[..., 1] ** 2 + 1e-7
+    c1 = (a[:, :2] + a[:, 2:]) / 2
+    c2 = (b[:, :2] + b[:, 2:]) / 2
+    dist = ((c1[:, None, :] - c2[None, :, :]) ** 2).sum(-1)
+    if variant == "diou":
+        return iou - dist / cdiag
+    w1, h1 = a[:, 2] - a[:, 0], a[:, 3] - a[:, 1]
+    w2, h2 = b[:, 2] - b[:, 0], b[:, 3] - b[:, 1]
+    import math
+
+    v = (4 / math.pi**2) * (
+        torch.atan(w2[None, :] / (h2[None, :] + 1e-7)) - torch.atan(w1[:, None] / (h1[:, None] + 1e-7))
+    ) ** 2
+    alpha = v / (1 - iou + v + 1e-7)
+    return iou - dist / cdiag - alpha * v
+
+
 def _stub(name: str) -> types.ModuleType:
     mod = types.ModuleType(name)
     mod.__spec__ = importlib.machinery.ModuleSpec(name, None)
@@ -73,6 +103,12 @@ def load_legacy_map():
         ops.box_area = _box_area
         ops.box_iou = _box_iou
         ops.box_convert = _box_convert
+        # giou/diou/ciou variants (standard formulas; used by the module-level
+        # IoU differential, which tests AGGREGATION semantics — the box math
+        # itself is covered by hand-computed cases in the detection tests)
+        ops.generalized_box_iou = lambda a, b: _box_iou_variant(a, b, "giou")
+        ops.distance_box_iou = lambda a, b: _box_iou_variant(a, b, "diou")
+        ops.complete_box_iou = lambda a, b: _box_iou_variant(a, b, "ciou")
         tv.ops = ops
     if "pycocotools" not in sys.modules:
         pc = _stub("pycocotools")

@@ -287,8 +287,13 @@ def test_retrieval_modules(name, kwargs):
 
 def test_detection_iou_module():
     tm = _tm()
+    from tests.unittests.detection._ref_oracle import load_legacy_map
+
+    load_legacy_map()  # installs the torchvision stub so the ref module runs
+    import importlib
+
     try:
-        ref = tm.detection.IntersectionOverUnion()
+        ref = importlib.import_module("torchmetrics.detection.iou").IntersectionOverUnion()
     except Exception as err:
         pytest.skip(f"reference IoU module unavailable offline: {err}")
     g = torch.Generator().manual_seed(52)
@@ -358,3 +363,38 @@ def test_feature_share_wrapper():
     # FeatureShare needs metrics with a `feature_network` attribute; use toy
     # FID-style metrics if available on both sides — otherwise skip
     pytest.skip("FeatureShare needs model-backed metrics (offline-gated); covered by wrapper unit tests")
+
+
+@pytest.mark.parametrize("name", ["IntersectionOverUnion", "GeneralizedIntersectionOverUnion", "DistanceIntersectionOverUnion", "CompleteIntersectionOverUnion"])
+@pytest.mark.parametrize("kwargs", [{}, {"respect_labels": False}, {"class_metrics": True}, {"iou_threshold": 0.3}])
+def test_detection_iou_family(name, kwargs):
+    tm = _tm()
+    from tests.unittests.detection._ref_oracle import load_legacy_map
+
+    load_legacy_map()
+    # torchmetrics.detection.__init__ gates exports on torchvision at its
+    # first import (before our stub existed) — import the submodule directly
+    import importlib
+
+    mod_name = {"IntersectionOverUnion": "iou", "GeneralizedIntersectionOverUnion": "giou",
+                "DistanceIntersectionOverUnion": "diou", "CompleteIntersectionOverUnion": "ciou"}[name]
+    try:
+        ref_mod = importlib.import_module(f"torchmetrics.detection.{mod_name}")
+        ref = getattr(ref_mod, name)(**kwargs)
+    except Exception as err:
+        pytest.skip(f"reference module unavailable offline: {err}")
+    g = torch.Generator().manual_seed(63)
+
+    def boxes(n):
+        xy = torch.rand(n, 2, generator=g) * 60
+        wh = torch.rand(n, 2, generator=g) * 25 + 4
+        return torch.cat([xy, xy + wh], 1)
+
+    ours = getattr(ma.detection, name)(**kwargs)
+    for i in range(3):
+        nd, ng = (6, 4) if i != 1 else (0, 3)  # include an empty-preds image
+        p = [{"boxes": boxes(nd), "scores": torch.rand(nd, generator=g), "labels": torch.randint(0, 2, (nd,), generator=g)}]
+        t = [{"boxes": boxes(ng), "labels": torch.randint(0, 2, (ng,), generator=g)}]
+        ours.update(p, t)
+        ref.update(p, t)
+    _cmp(ours.compute(), ref.compute())
