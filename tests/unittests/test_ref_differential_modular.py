@@ -398,3 +398,127 @@ def test_detection_iou_family(name, kwargs):
         ours.update(p, t)
         ref.update(p, t)
     _cmp(ours.compute(), ref.compute())
+
+
+# ------------------------------------------- broad accumulation differential
+def _img(seed, s=32):
+    g = torch.Generator().manual_seed(seed)
+    return torch.rand(2, 3, s, s, generator=g), torch.rand(2, 3, s, s, generator=g)
+
+
+def _aud(seed):
+    g = torch.Generator().manual_seed(seed)
+    t = torch.randn(2, 4000, generator=g)
+    return t + 0.4 * torch.randn(2, 4000, generator=g), t
+
+
+def _reg2(seed):
+    g = torch.Generator().manual_seed(seed)
+    p = torch.randn(64, generator=g)
+    return p, 0.6 * p + 0.5 * torch.randn(64, generator=g)
+
+
+def _lab(seed, k=5):
+    g = torch.Generator().manual_seed(seed)
+    return torch.randint(0, k, (80,), generator=g), torch.randint(0, k, (80,), generator=g)
+
+
+def _seg1h(seed, c=3, s=12):
+    g = torch.Generator().manual_seed(seed)
+    p = torch.nn.functional.one_hot(torch.randint(0, c, (2, s, s), generator=g), c).movedim(-1, 1)
+    t = torch.nn.functional.one_hot(torch.randint(0, c, (2, s, s), generator=g), c).movedim(-1, 1)
+    return p, t
+
+
+_ACC_CASES = [
+    # (path, kwargs, input_gen, atol)
+    ("PeakSignalNoiseRatio", {}, _img, 1e-4),
+    ("PeakSignalNoiseRatio", {"data_range": 1.0}, _img, 1e-4),
+    ("StructuralSimilarityIndexMeasure", {"data_range": 1.0}, _img, 1e-5),
+    ("MultiScaleStructuralSimilarityIndexMeasure", {"data_range": 1.0}, lambda s: _img(s, 192), 1e-4),
+    ("UniversalImageQualityIndex", {}, _img, 1e-5),
+    ("SpectralAngleMapper", {}, _img, 1e-5),
+    ("ErrorRelativeGlobalDimensionlessSynthesis", {}, _img, 1e-3),
+    ("RelativeAverageSpectralError", {}, _img, 1e-3),
+    ("RootMeanSquaredErrorUsingSlidingWindow", {}, _img, 1e-4),
+    ("TotalVariation", {}, lambda s: (_img(s)[0],), 1e-3),
+    ("TotalVariation", {"reduction": "mean"}, lambda s: (_img(s)[0],), 1e-4),
+    ("SignalNoiseRatio", {}, _aud, 1e-4),
+    ("ScaleInvariantSignalNoiseRatio", {}, _aud, 1e-4),
+    ("SignalDistortionRatio", {}, _aud, 1e-2),
+    ("ScaleInvariantSignalDistortionRatio", {}, _aud, 1e-4),
+    ("MeanSquaredError", {}, _reg2, 1e-5),
+    ("MeanAbsoluteError", {}, _reg2, 1e-5),
+    ("MeanSquaredLogError", {}, lambda s: tuple(x.abs() for x in _reg2(s)), 1e-5),
+    ("LogCoshError", {}, _reg2, 1e-5),
+    ("MinkowskiDistance", {"p": 3}, _reg2, 1e-4),
+    ("TweedieDevianceScore", {"power": 1.5}, lambda s: tuple(x.abs() + 0.1 for x in _reg2(s)), 1e-4),
+    ("KLDivergence", {}, lambda s: (torch.rand(8, 5, generator=torch.Generator().manual_seed(s)).softmax(-1), torch.rand(8, 5, generator=torch.Generator().manual_seed(s + 1)).softmax(-1)), 1e-5),
+    ("CosineSimilarity", {}, lambda s: (torch.randn(16, 8, generator=torch.Generator().manual_seed(s)), torch.randn(16, 8, generator=torch.Generator().manual_seed(s + 1))), 1e-5),
+    ("ExplainedVariance", {}, _reg2, 1e-5),
+    ("R2Score", {}, _reg2, 1e-5),
+    ("PearsonCorrCoef", {}, _reg2, 1e-5),
+    ("SpearmanCorrCoef", {}, _reg2, 1e-5),
+    ("KendallRankCorrCoef", {}, _reg2, 1e-5),
+    ("ConcordanceCorrCoef", {}, _reg2, 1e-5),
+    ("CriticalSuccessIndex", {"threshold": 0.5}, lambda s: (torch.rand(100, generator=torch.Generator().manual_seed(s)), torch.rand(100, generator=torch.Generator().manual_seed(s + 1))), 1e-5),
+    ("CohenKappa", {"task": "binary"}, lambda s: (torch.rand(80, generator=torch.Generator().manual_seed(s)), torch.randint(0, 2, (80,), generator=torch.Generator().manual_seed(s + 1))), 1e-5),
+]
+
+_ACC_NS = [
+    ("image", "SpatialCorrelationCoefficient", {}, _img, 1e-4),
+    ("image", "VisualInformationFidelity", {}, lambda s: _img(s, 64), 1e-4),
+    ("image", "PeakSignalNoiseRatioWithBlockedEffect", {}, lambda s: tuple(x[:, :1] for x in _img(s)), 1e-4),
+    ("clustering", "MutualInfoScore", {}, _lab, 1e-5),
+    ("clustering", "AdjustedRandScore", {}, _lab, 1e-5),
+    ("clustering", "NormalizedMutualInfoScore", {}, _lab, 1e-5),
+    ("clustering", "FowlkesMallowsIndex", {}, _lab, 1e-5),
+    ("clustering", "HomogeneityScore", {}, _lab, 1e-5),
+    ("clustering", "CompletenessScore", {}, _lab, 1e-5),
+    ("clustering", "VMeasureScore", {}, _lab, 1e-5),
+    ("nominal", "CramersV", {"num_classes": 5}, _lab, 1e-4),
+    ("nominal", "PearsonsContingencyCoefficient", {"num_classes": 5}, _lab, 1e-4),
+    ("nominal", "TschuprowsT", {"num_classes": 5}, _lab, 1e-4),
+    ("nominal", "TheilsU", {"num_classes": 5}, _lab, 1e-4),
+    ("segmentation", "MeanIoU", {"num_classes": 3}, _seg1h, 1e-5),
+    ("segmentation", "DiceScore", {"num_classes": 3}, _seg1h, 1e-5),
+    ("segmentation", "GeneralizedDiceScore", {"num_classes": 3}, _seg1h, 1e-5),
+    ("segmentation", "HausdorffDistance", {"num_classes": 3}, _seg1h, 1e-4),
+]
+
+
+def _acc_check(our_cls, ref_cls, kwargs, gen, atol):
+    import warnings
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours, ref = our_cls(**kwargs), ref_cls(**kwargs)
+        for b in range(3):
+            args = gen(90 + b)
+            ours.update(*args)
+            ref.update(*args)
+        _cmp(ours.compute(), ref.compute(), atol)
+
+
+@pytest.mark.parametrize(
+    ("name", "kwargs", "gen", "atol"), _ACC_CASES, ids=[f"{c[0]}_{i}" for i, c in enumerate(_ACC_CASES)]
+)
+def test_modular_accumulation(name, kwargs, gen, atol):
+    tm = _tm()
+    _acc_check(getattr(ma, name), getattr(tm, name), kwargs, gen, atol)
+
+
+@pytest.mark.parametrize(
+    ("ns", "name", "kwargs", "gen", "atol"), _ACC_NS, ids=[f"{c[1]}_{i}" for i, c in enumerate(_ACC_NS)]
+)
+def test_modular_accumulation_ns(ns, name, kwargs, gen, atol):
+    import importlib
+
+    _tm()
+    our_ns = importlib.import_module(f"metrics_amd.{ns}")
+    ref_ns = importlib.import_module(f"torchmetrics.{ns}")
+    our_cls = getattr(our_ns, name, None)
+    ref_cls = getattr(ref_ns, name, None)
+    if our_cls is None or ref_cls is None:
+        pytest.fail(f"{ns}.{name} missing: ours={our_cls is not None} ref={ref_cls is not None}")
+    _acc_check(our_cls, ref_cls, kwargs, gen, atol)
