@@ -37,25 +37,18 @@ def _validate_inputs(preds: Tensor, target: Tensor) -> None:
         )
 
 
-def _segments(pan: Tensor, things: Set[int], stuffs: Set[int]) -> Dict[Tuple[int, int], Tensor]:
-    """Map (category, instance) -> boolean mask. Stuff collapses instance ids to 0."""
-    cats = pan[..., 0]
-    insts = pan[..., 1]
-    segs: Dict[Tuple[int, int], Tensor] = {}
-    pairs = torch.stack([cats.flatten(), insts.flatten()], dim=1)
-    uniq = torch.unique(pairs, dim=0)
-    for cat, inst in uniq.tolist():
-        if cat in stuffs:
-            key = (cat, 0)
-            mask = cats == cat
-        elif cat in things:
-            key = (cat, inst)
-            mask = (cats == cat) & (insts == inst)
-        else:
-            continue  # void
-        if key not in segs:
-            segs[key] = mask
-    return segs
+def _flat_colors(pan: Tensor, things: Set[int], stuffs: Set[int], void_cat: int) -> Tuple[Tensor, Tensor]:
+    """Flatten to per-pixel (category, instance); stuff instances collapse to 0,
+    unknown categories map to the void color ``(void_cat, 0)``."""
+    cats = pan[..., 0].flatten().clone()
+    insts = pan[..., 1].flatten().clone()
+    dev = pan.device
+    is_stuff = torch.isin(cats, torch.tensor(sorted(stuffs), device=dev))
+    insts[is_stuff] = 0
+    unknown = ~torch.isin(cats, torch.tensor(sorted(things | stuffs), device=dev))
+    cats[unknown] = void_cat
+    insts[unknown] = 0
+    return cats, insts
 
 
 def _panoptic_quality_update_sample(
@@ -65,12 +58,36 @@ def _panoptic_quality_update_sample(
     stuffs: Set[int],
     modified_metric_stuffs: Optional[Set[int]] = None,
 ) -> Tuple[Dict[int, float], Dict[int, int], Dict[int, int], Dict[int, int]]:
-    """One image: per-category (iou_sum, tp, fp, fn)."""
-    void_pred = ~torch.isin(pred[..., 0], torch.tensor(sorted(things | stuffs), device=pred.device))
-    void_target = ~torch.isin(target[..., 0], torch.tensor(sorted(things | stuffs), device=target.device))
+    """One image: per-category (iou_sum, tp, fp, fn).
 
-    pred_segs = _segments(pred, things, stuffs)
-    target_segs = _segments(target, things, stuffs)
+    Matching follows the published PQ algorithm exactly as the reference
+    implements it (functional/detection/_panoptic_quality_common.py:312):
+    every same-category (pred segment, target segment) intersection with
+    IoU > 0.5 is a TP (the gate makes matches exclusive); the union subtracts
+    the pred segment's overlap with void target AND the target segment's
+    overlap with void pred; unmatched segments that are majority-void in the
+    other map are ignored rather than counted as FP/FN. For the modified PQ
+    variant, stuff classes accumulate IoU at any overlap > 0, count one TP
+    per target segment present, and contribute no FP/FN.
+    """
+    mod = modified_metric_stuffs or set()
+    void_cat = max(things | stuffs) + 1
+    VOID = (void_cat, 0)
+    pc, pi = _flat_colors(pred, things, stuffs, void_cat)
+    tc, ti = _flat_colors(target, things, stuffs, void_cat)
+
+    # one unique pass over per-pixel (pred color, target color) rows gives the
+    # pairwise intersection areas; segment areas are its row/col marginals
+    rows = torch.stack([pc, pi, tc, ti], dim=1)
+    uniq, counts = torch.unique(rows, dim=0, return_counts=True)
+    pair_area: Dict[Tuple[Tuple[int, int], Tuple[int, int]], int] = {}
+    pred_area: Dict[Tuple[int, int], int] = {}
+    target_area: Dict[Tuple[int, int], int] = {}
+    for (a, b, c, d), n in zip(uniq.tolist(), counts.tolist()):
+        pcol, tcol = (a, b), (c, d)
+        pair_area[(pcol, tcol)] = pair_area.get((pcol, tcol), 0) + n
+        pred_area[pcol] = pred_area.get(pcol, 0) + n
+        target_area[tcol] = target_area.get(tcol, 0) + n
 
     iou_sum: Dict[int, float] = {}
     tp: Dict[int, int] = {}
@@ -79,61 +96,42 @@ def _panoptic_quality_update_sample(
     matched_pred: Set[Tuple[int, int]] = set()
     matched_target: Set[Tuple[int, int]] = set()
 
-    for tkey, tmask in target_segs.items():
-        cat = tkey[0]
-        if modified_metric_stuffs is not None and cat in modified_metric_stuffs:
-            continue  # handled separately below
-        best_iou, best_pkey = 0.0, None
-        for pkey, pmask in pred_segs.items():
-            if pkey[0] != cat or pkey in matched_pred:
-                continue
-            inter = (tmask & pmask).sum().item()
-            if inter == 0:
-                continue
-            union = (tmask | pmask).sum().item() - (pmask & void_target).sum().item()
-            iou = inter / union if union > 0 else 0.0
-            if iou > best_iou:
-                best_iou, best_pkey = iou, pkey
-        if best_iou > 0.5:
-            matched_pred.add(best_pkey)
-            matched_target.add(tkey)
-            iou_sum[cat] = iou_sum.get(cat, 0.0) + best_iou
-            tp[cat] = tp.get(cat, 0) + 1
-
-    # modified PQ: stuff categories scored by pixel-level IoU without the 0.5 gate
-    if modified_metric_stuffs:
-        for tkey, tmask in target_segs.items():
-            cat = tkey[0]
-            if cat not in modified_metric_stuffs:
-                continue
-            pmask = None
-            for pkey, pm in pred_segs.items():
-                if pkey[0] == cat:
-                    pmask = pm if pmask is None else (pmask | pm)
-                    matched_pred.add(pkey)
-            if pmask is None:
-                fn[cat] = fn.get(cat, 0) + 1
-                continue
-            inter = (tmask & pmask).sum().item()
-            union = (tmask | pmask).sum().item()
-            iou = inter / union if union > 0 else 0.0
+    for (pcol, tcol), inter in pair_area.items():
+        if tcol == VOID or pcol == VOID or pcol[0] != tcol[0]:
+            continue
+        cat = tcol[0]
+        union = (
+            pred_area[pcol]
+            - pair_area.get((pcol, VOID), 0)
+            + target_area[tcol]
+            - pair_area.get((VOID, tcol), 0)
+            - inter
+        )
+        iou = inter / union if union > 0 else 0.0
+        if cat not in mod and iou > 0.5:
+            matched_pred.add(pcol)
+            matched_target.add(tcol)
             iou_sum[cat] = iou_sum.get(cat, 0.0) + iou
             tp[cat] = tp.get(cat, 0) + 1
-            matched_target.add(tkey)
+        elif cat in mod and iou > 0:
+            iou_sum[cat] = iou_sum.get(cat, 0.0) + iou
 
-    for tkey, tmask in target_segs.items():
-        if tkey not in matched_target and not (
-            modified_metric_stuffs and tkey[0] in modified_metric_stuffs
-        ):
-            fn[tkey[0]] = fn.get(tkey[0], 0) + 1
-    for pkey, pmask in pred_segs.items():
-        if pkey in matched_pred:
+    for tcol in set(target_area) - matched_target:
+        if tcol == VOID or tcol[0] in mod:
             continue
-        # FPs with >50% void overlap in the target are ignored (COCO PQ rule)
-        void_frac = (pmask & void_target).sum().item() / max(pmask.sum().item(), 1)
-        if void_frac > 0.5:
+        if pair_area.get((VOID, tcol), 0) / target_area[tcol] <= 0.5:
+            fn[tcol[0]] = fn.get(tcol[0], 0) + 1
+
+    for pcol in set(pred_area) - matched_pred:
+        if pcol == VOID or pcol[0] in mod:
             continue
-        fp[pkey[0]] = fp.get(pkey[0], 0) + 1
+        if pair_area.get((pcol, VOID), 0) / pred_area[pcol] <= 0.5:
+            fp[pcol[0]] = fp.get(pcol[0], 0) + 1
+
+    for tcol in target_area:
+        if tcol != VOID and tcol[0] in mod:
+            tp[tcol[0]] = tp.get(tcol[0], 0) + 1
+
     return iou_sum, tp, fp, fn
 
 

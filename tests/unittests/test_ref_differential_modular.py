@@ -522,3 +522,97 @@ def test_modular_accumulation_ns(ns, name, kwargs, gen, atol):
     if our_cls is None or ref_cls is None:
         pytest.fail(f"{ns}.{name} missing: ours={our_cls is not None} ref={ref_cls is not None}")
     _acc_check(our_cls, ref_cls, kwargs, gen, atol)
+
+
+def _panoptic(seed):
+    # preds = target with 15% label noise so IoU>0.5 matches actually occur
+    # (fully random maps give PQ == 0 on both sides, a trivial comparison)
+    g = torch.Generator().manual_seed(seed)
+    cat = torch.tensor([0, 1, 2, 7])[torch.randint(0, 4, (2, 16, 16), generator=g)]
+    inst = torch.randint(0, 3, (2, 16, 16), generator=g)
+    pt = torch.stack([cat, inst], dim=-1)
+    pp = pt.clone()
+    noise = torch.rand(pt.shape[:-1], generator=g) < 0.15
+    pp[..., 0][noise] = torch.tensor([0, 1, 2, 7])[torch.randint(0, 4, (int(noise.sum()),), generator=g)]
+    return pp, pt
+
+
+def _multisrc(seed):
+    g = torch.Generator().manual_seed(seed)
+    t = torch.randn(2, 3, 1500, generator=g)
+    return t + 0.3 * torch.randn(2, 3, 1500, generator=g), t
+
+
+def _texts(seed):
+    corpus = [
+        "the cat sat on the mat",
+        "a quick brown fox jumps over the lazy dog",
+        "hello world this is a test",
+        "metrics are computed on device",
+        "the rain in spain stays mainly on the plain",
+        "pack my box with five dozen liquor jugs",
+    ]
+    g = torch.Generator().manual_seed(seed)
+    idx = torch.randint(0, len(corpus), (4,), generator=g).tolist()
+    jdx = torch.randint(0, len(corpus), (4,), generator=g).tolist()
+    preds = [corpus[i] for i in idx]
+    target = [[corpus[j], corpus[(j + 1) % len(corpus)]] for j in jdx]
+    return preds, target
+
+
+def _texts_flat(seed):
+    p, t = _texts(seed)
+    return p, [x[0] for x in t]
+
+
+_ACC_NS2 = [
+    ("detection", "PanopticQuality", {"things": {1, 2}, "stuffs": {7}, "allow_unknown_preds_category": True}, _panoptic, 1e-5),
+    ("detection", "ModifiedPanopticQuality", {"things": {1, 2}, "stuffs": {7}, "allow_unknown_preds_category": True}, _panoptic, 1e-5),
+    ("audio", "ComplexScaleInvariantSignalNoiseRatio", {}, lambda s: (torch.randn(2, 40, 10, 2, generator=torch.Generator().manual_seed(s)), torch.randn(2, 40, 10, 2, generator=torch.Generator().manual_seed(s + 1))), 1e-4),
+    ("audio", "SourceAggregatedSignalDistortionRatio", {}, _multisrc, 1e-4),
+    ("audio", "PermutationInvariantTraining", {"metric_func": None}, _multisrc, 1e-4),
+    ("text", "BLEUScore", {}, _texts, 1e-5),
+    ("text", "BLEUScore", {"n_gram": 2, "smooth": True}, _texts, 1e-5),
+    ("text", "SacreBLEUScore", {}, _texts, 1e-5),
+    ("text", "CHRFScore", {}, _texts, 1e-5),
+    ("text", "CHRFScore", {"return_sentence_level_score": True}, _texts, 1e-5),
+    ("text", "TranslationEditRate", {}, _texts, 1e-5),
+    ("text", "CharErrorRate", {}, _texts_flat, 1e-5),
+    ("text", "WordErrorRate", {}, _texts_flat, 1e-5),
+    ("text", "MatchErrorRate", {}, _texts_flat, 1e-5),
+    ("text", "WordInfoLost", {}, _texts_flat, 1e-5),
+    ("text", "WordInfoPreserved", {}, _texts_flat, 1e-5),
+    ("text", "ExtendedEditDistance", {}, _texts_flat, 1e-5),
+    ("text", "EditDistance", {}, _texts_flat, 1e-5),
+    ("text", "EditDistance", {"reduction": "sum"}, _texts_flat, 1e-5),
+]
+
+
+@pytest.mark.parametrize(
+    ("ns", "name", "kwargs", "gen", "atol"), _ACC_NS2, ids=[f"{c[1]}_{i}" for i, c in enumerate(_ACC_NS2)]
+)
+def test_modular_accumulation_ns2(ns, name, kwargs, gen, atol):
+    import importlib
+
+    _tm()
+    if name == "PermutationInvariantTraining":
+        import torchmetrics.functional.audio as ref_fa
+
+        import metrics_amd.functional.audio as our_fa
+
+        our_kw = {"metric_func": our_fa.scale_invariant_signal_noise_ratio}
+        ref_kw = {"metric_func": ref_fa.scale_invariant_signal_noise_ratio}
+    else:
+        our_kw = ref_kw = kwargs
+    our_ns = importlib.import_module(f"metrics_amd.{ns}")
+    ref_ns = importlib.import_module(f"torchmetrics.{ns}")
+    import warnings
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours, ref = getattr(our_ns, name)(**our_kw), getattr(ref_ns, name)(**ref_kw)
+        for b in range(3):
+            args = gen(90 + b)
+            ours.update(*args)
+            ref.update(*args)
+        _cmp(ours.compute(), ref.compute(), atol)
