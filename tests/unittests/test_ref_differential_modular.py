@@ -616,3 +616,54 @@ def test_modular_accumulation_ns2(ns, name, kwargs, gen, atol):
             ours.update(*args)
             ref.update(*args)
         _cmp(ours.compute(), ref.compute(), atol)
+
+
+def _dlambda(seed):
+    g = torch.Generator().manual_seed(seed)
+    return torch.rand(2, 4, 16, 16, generator=g), torch.rand(2, 4, 16, 16, generator=g)
+
+
+_ACC_NS3 = [
+    ("regression", "MeanAbsolutePercentageError", {}, lambda s: tuple(x.abs() + 0.1 for x in _reg2(s)), 1e-5),
+    ("regression", "SymmetricMeanAbsolutePercentageError", {}, lambda s: tuple(x.abs() + 0.1 for x in _reg2(s)), 1e-5),
+    ("regression", "WeightedMeanAbsolutePercentageError", {}, lambda s: tuple(x.abs() + 0.1 for x in _reg2(s)), 1e-5),
+    ("regression", "RelativeSquaredError", {}, _reg2, 1e-5),
+    ("regression", "RelativeSquaredError", {"squared": False}, _reg2, 1e-5),
+    ("regression", "NormalizedRootMeanSquaredError", {"normalization": "mean"}, lambda s: tuple(x.abs() + 0.1 for x in _reg2(s)), 1e-5),
+    ("regression", "NormalizedRootMeanSquaredError", {"normalization": "range"}, _reg2, 1e-5),
+    ("regression", "NormalizedRootMeanSquaredError", {"normalization": "std"}, _reg2, 1e-5),
+    ("image", "SpectralDistortionIndex", {}, _dlambda, 1e-5),
+    ("image", "QualityWithNoReference", {}, lambda s: (
+        torch.rand(2, 3, 32, 32, generator=torch.Generator().manual_seed(s)),
+        {
+            "ms": torch.rand(2, 3, 8, 8, generator=torch.Generator().manual_seed(s + 1)),
+            "pan": torch.rand(2, 3, 32, 32, generator=torch.Generator().manual_seed(s + 2)),
+            "pan_lr": torch.rand(2, 3, 8, 8, generator=torch.Generator().manual_seed(s + 3)),
+        },
+    ), 1e-4),
+]
+
+
+@pytest.mark.parametrize(
+    ("ns", "name", "kwargs", "gen", "atol"), _ACC_NS3, ids=[f"{c[1]}_{i}" for i, c in enumerate(_ACC_NS3)]
+)
+def test_modular_accumulation_ns3(ns, name, kwargs, gen, atol):
+    import importlib
+    import warnings
+
+    _tm()
+    our_ns = importlib.import_module(f"metrics_amd.{ns}")
+    ref_ns = importlib.import_module(f"torchmetrics.{ns}")
+    our_cls = getattr(our_ns, name, None)
+    ref_cls = getattr(ref_ns, name, None)
+    if ref_cls is None:
+        pytest.skip(f"{ns}.{name} absent in reference")
+    assert our_cls is not None, f"{ns}.{name} missing in metrics_amd"
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours, ref = our_cls(**kwargs), ref_cls(**kwargs)
+        for b in range(3):
+            args = gen(90 + b)
+            ours.update(*args)
+            ref.update(*args)
+        _cmp(ours.compute(), ref.compute(), atol)
