@@ -67,8 +67,10 @@ class PeakSignalNoiseRatio(Metric):
             if dim is not None:
                 raise ValueError("The `data_range` must be given when `dim` is not None.")
             self.data_range = None
-            self.add_state("min_target", default=torch.tensor(float("inf")), dist_reduce_fx=torch.min)
-            self.add_state("max_target", default=torch.tensor(-float("inf")), dist_reduce_fx=torch.max)
+            # the reference seeds the running range at 0.0 (image/psnr.py:100),
+            # so the inferred data_range always includes 0 — replicated as-is
+            self.add_state("min_target", default=torch.tensor(0.0), dist_reduce_fx=torch.min)
+            self.add_state("max_target", default=torch.tensor(0.0), dist_reduce_fx=torch.max)
             self.clamping_fn = None
         elif isinstance(data_range, tuple):
             self.add_state("data_range", default=torch.tensor(data_range[1] - data_range[0]), dist_reduce_fx="mean")

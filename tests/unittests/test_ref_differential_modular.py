@@ -667,3 +667,27 @@ def test_modular_accumulation_ns3(ns, name, kwargs, gen, atol):
             ours.update(*args)
             ref.update(*args)
         _cmp(ours.compute(), ref.compute(), atol)
+
+
+_FWD_CASES = [c for c in _ACC_CASES if c[0] not in {"SpearmanCorrCoef", "KendallRankCorrCoef"}]
+
+
+@pytest.mark.parametrize(
+    ("name", "kwargs", "gen", "atol"), _FWD_CASES, ids=[f"{c[0]}_{i}" for i, c in enumerate(_FWD_CASES)]
+)
+def test_modular_forward(name, kwargs, gen, atol):
+    """forward() parity: per-batch return values AND the accumulated compute.
+
+    Exercises the full-state/reduced-state forward machinery, which the
+    update()-based accumulation differential does not touch.
+    """
+    import warnings
+
+    tm = _tm()
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours, ref = getattr(ma, name)(**kwargs), getattr(tm, name)(**kwargs)
+        for b in range(3):
+            args = gen(70 + b)
+            _cmp(ours(*args), ref(*args), atol)
+        _cmp(ours.compute(), ref.compute(), atol)
