@@ -51,7 +51,9 @@ def _bleu_score_compute(
     smooth: bool = False,
 ) -> Tensor:
     device = numerator.device
-    if min(numerator) == 0.0 and not smooth:
+    # reference quirk (functional/text/bleu.py:132): any zero n-gram numerator
+    # short-circuits to 0 even when smoothing is requested
+    if min(numerator) == 0.0:
         return tensor(0.0, device=device)
 
     if smooth:

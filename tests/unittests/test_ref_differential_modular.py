@@ -691,3 +691,30 @@ def test_modular_forward(name, kwargs, gen, atol):
             args = gen(70 + b)
             _cmp(ours(*args), ref(*args), atol)
         _cmp(ours.compute(), ref.compute(), atol)
+
+
+_FWD_NS = [c for c in _ACC_NS + _ACC_NS2 + _ACC_NS3 if c[1] not in {"PermutationInvariantTraining"}]
+
+
+@pytest.mark.parametrize(
+    ("ns", "name", "kwargs", "gen", "atol"), _FWD_NS, ids=[f"{c[1]}_{i}" for i, c in enumerate(_FWD_NS)]
+)
+def test_modular_forward_ns(ns, name, kwargs, gen, atol):
+    import importlib
+    import warnings
+
+    _tm()
+    our_ns = importlib.import_module(f"metrics_amd.{ns}")
+    ref_ns = importlib.import_module(f"torchmetrics.{ns}")
+    our_cls = getattr(our_ns, name, None)
+    ref_cls = getattr(ref_ns, name, None)
+    if ref_cls is None:
+        pytest.skip(f"{ns}.{name} absent in reference")
+    assert our_cls is not None, f"{ns}.{name} missing in metrics_amd"
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours, ref = our_cls(**kwargs), ref_cls(**kwargs)
+        for b in range(3):
+            args = gen(70 + b)
+            _cmp(ours(*args), ref(*args), atol)
+        _cmp(ours.compute(), ref.compute(), atol)
