@@ -718,3 +718,135 @@ def test_modular_forward_ns(ns, name, kwargs, gen, atol):
             args = gen(70 + b)
             _cmp(ours(*args), ref(*args), atol)
         _cmp(ours.compute(), ref.compute(), atol)
+
+
+def _bin_pt(seed):
+    g = torch.Generator().manual_seed(seed)
+    return torch.rand(120, generator=g), torch.randint(0, 2, (120,), generator=g)
+
+
+def _mc_pt(seed, c=5):
+    g = torch.Generator().manual_seed(seed)
+    return torch.randn(120, c, generator=g).softmax(-1), torch.randint(0, c, (120,), generator=g)
+
+
+def _ml_pt(seed, l=4):
+    g = torch.Generator().manual_seed(seed)
+    return torch.rand(120, l, generator=g), torch.randint(0, 2, (120, l), generator=g)
+
+
+_CLS_FWD = [
+    ("BinaryAccuracy", {}, _bin_pt),
+    ("BinaryAccuracy", {"ignore_index": 0}, _bin_pt),
+    ("BinaryPrecision", {}, _bin_pt),
+    ("BinaryRecall", {"threshold": 0.3}, _bin_pt),
+    ("BinaryF1Score", {}, _bin_pt),
+    ("BinaryFBetaScore", {"beta": 0.5}, _bin_pt),
+    ("BinarySpecificity", {}, _bin_pt),
+    ("BinaryCohenKappa", {}, _bin_pt),
+    ("BinaryMatthewsCorrCoef", {}, _bin_pt),
+    ("BinaryHammingDistance", {}, _bin_pt),
+    ("BinaryJaccardIndex", {}, _bin_pt),
+    ("BinaryAUROC", {}, _bin_pt),
+    ("BinaryAUROC", {"thresholds": 25}, _bin_pt),
+    ("BinaryAveragePrecision", {}, _bin_pt),
+    ("BinaryCalibrationError", {}, _bin_pt),
+    ("BinaryStatScores", {}, _bin_pt),
+    ("BinaryConfusionMatrix", {}, _bin_pt),
+    ("BinaryNegativePredictiveValue", {}, _bin_pt),
+    ("BinaryHingeLoss", {}, _bin_pt),
+    ("MulticlassAccuracy", {"num_classes": 5}, _mc_pt),
+    ("MulticlassAccuracy", {"num_classes": 5, "average": "micro"}, _mc_pt),
+    ("MulticlassAccuracy", {"num_classes": 5, "average": "weighted"}, _mc_pt),
+    ("MulticlassAccuracy", {"num_classes": 5, "average": None}, _mc_pt),
+    ("MulticlassAccuracy", {"num_classes": 5, "top_k": 2}, _mc_pt),
+    ("MulticlassAccuracy", {"num_classes": 5, "ignore_index": 2}, _mc_pt),
+    ("MulticlassPrecision", {"num_classes": 5}, _mc_pt),
+    ("MulticlassRecall", {"num_classes": 5, "average": "weighted"}, _mc_pt),
+    ("MulticlassF1Score", {"num_classes": 5}, _mc_pt),
+    ("MulticlassFBetaScore", {"num_classes": 5, "beta": 2.0}, _mc_pt),
+    ("MulticlassSpecificity", {"num_classes": 5}, _mc_pt),
+    ("MulticlassCohenKappa", {"num_classes": 5}, _mc_pt),
+    ("MulticlassCohenKappa", {"num_classes": 5, "weights": "linear"}, _mc_pt),
+    ("MulticlassMatthewsCorrCoef", {"num_classes": 5}, _mc_pt),
+    ("MulticlassJaccardIndex", {"num_classes": 5}, _mc_pt),
+    ("MulticlassAUROC", {"num_classes": 5}, _mc_pt),
+    ("MulticlassAUROC", {"num_classes": 5, "thresholds": 25}, _mc_pt),
+    ("MulticlassAveragePrecision", {"num_classes": 5}, _mc_pt),
+    ("MulticlassCalibrationError", {"num_classes": 5}, _mc_pt),
+    ("MulticlassExactMatch", {"num_classes": 5}, lambda s: (_mc_pt(s)[0].reshape(24, 5, 5).argmax(1), _mc_pt(s)[1].reshape(24, 5))),
+    ("MulticlassConfusionMatrix", {"num_classes": 5}, _mc_pt),
+    ("MulticlassHingeLoss", {"num_classes": 5}, _mc_pt),
+    ("MultilabelAccuracy", {"num_labels": 4}, _ml_pt),
+    ("MultilabelAccuracy", {"num_labels": 4, "average": "micro"}, _ml_pt),
+    ("MultilabelPrecision", {"num_labels": 4}, _ml_pt),
+    ("MultilabelRecall", {"num_labels": 4}, _ml_pt),
+    ("MultilabelF1Score", {"num_labels": 4}, _ml_pt),
+    ("MultilabelSpecificity", {"num_labels": 4}, _ml_pt),
+    ("MultilabelHammingDistance", {"num_labels": 4}, _ml_pt),
+    ("MultilabelJaccardIndex", {"num_labels": 4}, _ml_pt),
+    ("MultilabelAUROC", {"num_labels": 4}, _ml_pt),
+    ("MultilabelAveragePrecision", {"num_labels": 4}, _ml_pt),
+    ("MultilabelExactMatch", {"num_labels": 4}, _ml_pt),
+    ("MultilabelRankingAveragePrecision", {"num_labels": 4}, _ml_pt),
+    ("MultilabelRankingLoss", {"num_labels": 4}, _ml_pt),
+    ("MultilabelCoverageError", {"num_labels": 4}, _ml_pt),
+    ("MultilabelConfusionMatrix", {"num_labels": 4}, _ml_pt),
+]
+
+
+@pytest.mark.parametrize(
+    ("name", "kwargs", "gen"), _CLS_FWD, ids=[f"{c[0]}_{i}" for i, c in enumerate(_CLS_FWD)]
+)
+def test_classification_forward(name, kwargs, gen):
+    import warnings
+
+    tm = _tm()
+    our_cls = getattr(ma.classification, name)
+    ref_cls = getattr(tm.classification, name)
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours, ref = our_cls(**kwargs), ref_cls(**kwargs)
+        for b in range(3):
+            args = gen(60 + b)
+            _cmp(ours(*args), ref(*args), 1e-5)
+        _cmp(ours.compute(), ref.compute(), 1e-5)
+
+
+def _ret_pt(seed):
+    g = torch.Generator().manual_seed(seed)
+    n = 150
+    return (
+        torch.rand(n, generator=g),
+        torch.randint(0, 2, (n,), generator=g),
+        torch.randint(0, 12, (n,), generator=g),
+    )
+
+
+_RET_FWD = [
+    ("RetrievalMAP", {}),
+    ("RetrievalMRR", {}),
+    ("RetrievalNormalizedDCG", {}),
+    ("RetrievalNormalizedDCG", {"top_k": 5}),
+    ("RetrievalPrecision", {"top_k": 3}),
+    ("RetrievalRecall", {"top_k": 3}),
+    ("RetrievalFallOut", {"top_k": 3}),
+    ("RetrievalHitRate", {"top_k": 3}),
+    ("RetrievalRPrecision", {}),
+    ("RetrievalAUROC", {}),
+]
+
+
+@pytest.mark.parametrize(("name", "kwargs"), _RET_FWD, ids=[f"{c[0]}_{i}" for i, c in enumerate(_RET_FWD)])
+def test_retrieval_forward(name, kwargs):
+    import warnings
+
+    tm = _tm()
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours = getattr(ma.retrieval, name)(**kwargs)
+        ref = getattr(tm.retrieval, name)(**kwargs)
+        for b in range(3):
+            p, t, idx = _ret_pt(60 + b)
+            _cmp(ours(p, t, indexes=idx), ref(p, t, indexes=idx), 1e-5)
+        _cmp(ours.compute(), ref.compute(), 1e-5)
