@@ -850,3 +850,79 @@ def test_retrieval_forward(name, kwargs):
             p, t, idx = _ret_pt(60 + b)
             _cmp(ours(p, t, indexes=idx), ref(p, t, indexes=idx), 1e-5)
         _cmp(ours.compute(), ref.compute(), 1e-5)
+
+
+def test_wrappers_forward_differential():
+    """forward() parity for the wrapper classes around a live base metric."""
+    import warnings
+
+    tm = _tm()
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        pairs = [
+            (
+                ma.MultioutputWrapper(ma.MeanSquaredError(), num_outputs=3),
+                tm.MultioutputWrapper(tm.MeanSquaredError(), num_outputs=3),
+                lambda s: (
+                    torch.randn(32, 3, generator=torch.Generator().manual_seed(s)),
+                    torch.randn(32, 3, generator=torch.Generator().manual_seed(s + 1)),
+                ),
+            ),
+            (
+                ma.wrappers.Running(ma.MeanSquaredError(), window=2),
+                tm.wrappers.Running(tm.MeanSquaredError(), window=2),
+                lambda s: (
+                    torch.randn(32, generator=torch.Generator().manual_seed(s)),
+                    torch.randn(32, generator=torch.Generator().manual_seed(s + 1)),
+                ),
+            ),
+            (
+                ma.ClasswiseWrapper(ma.MulticlassF1Score(num_classes=4, average=None)),
+                tm.ClasswiseWrapper(tm.classification.MulticlassF1Score(num_classes=4, average=None)),
+                lambda s: (
+                    torch.randn(64, 4, generator=torch.Generator().manual_seed(s)).softmax(-1),
+                    torch.randint(0, 4, (64,), generator=torch.Generator().manual_seed(s + 1)),
+                ),
+            ),
+        ]
+        for ours, ref, gen in pairs:
+            for b in range(4):
+                args = gen(50 + b)
+                _cmp(ours(*args), ref(*args), 1e-5)
+            _cmp(ours.compute(), ref.compute(), 1e-5)
+
+
+def test_collection_forward_differential():
+    """MetricCollection forward parity (compute groups on) across mixed tasks."""
+    import warnings
+
+    tm = _tm()
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours = ma.MetricCollection(
+            {
+                "acc": ma.MulticlassAccuracy(num_classes=5),
+                "prec": ma.MulticlassPrecision(num_classes=5),
+                "auroc": ma.MulticlassAUROC(num_classes=5),
+                "cal": ma.MulticlassCalibrationError(num_classes=5),
+            }
+        )
+        ref = tm.MetricCollection(
+            {
+                "acc": tm.classification.MulticlassAccuracy(num_classes=5),
+                "prec": tm.classification.MulticlassPrecision(num_classes=5),
+                "auroc": tm.classification.MulticlassAUROC(num_classes=5),
+                "cal": tm.classification.MulticlassCalibrationError(num_classes=5),
+            }
+        )
+        for b in range(3):
+            g = torch.Generator().manual_seed(40 + b)
+            p = torch.randn(96, 5, generator=g).softmax(-1)
+            t = torch.randint(0, 5, (96,), generator=g)
+            vo, vr = ours(p, t), ref(p, t)
+            assert set(vo) == set(vr)
+            for k in vr:
+                _cmp(vo[k], vr[k], 1e-5)
+        vo, vr = ours.compute(), ref.compute()
+        for k in vr:
+            _cmp(vo[k], vr[k], 1e-5)
