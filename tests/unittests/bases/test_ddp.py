@@ -372,3 +372,44 @@ def _test_bench_collection_world8(rank, world_size):
 
 def test_bench_collection_world8():
     run_distributed(_test_bench_collection_world8, world_size=8)
+
+
+def _test_pooled_map_states_ddp(rank, world_size):
+    # RASE keeps lazily-shaped pooled maps (scalar default -> (C,H,W) on first
+    # update); PQ keeps per-category count vectors. Both must DDP-reduce to the
+    # single-process result when every rank has updated at least once.
+    import metrics_amd as ma
+
+    torch.manual_seed(7)
+    preds = torch.rand(4, 3, 16, 16)
+    target = torch.rand(4, 3, 16, 16)
+    m = ma.image.RelativeAverageSpectralError()
+    for i in range(rank, 4, world_size):
+        m.update(preds[i : i + 1], target[i : i + 1])
+    out = m.compute()
+    ref = ma.image.RelativeAverageSpectralError()
+    ref.sync_on_compute = False
+    ref._to_sync = False
+    for i in range(4):
+        ref.update(preds[i : i + 1], target[i : i + 1])
+    assert torch.allclose(out, ref.compute(), atol=1e-5), (out, ref.compute())
+
+    cats = torch.randint(0, 3, (4, 12, 12))
+    inst = torch.randint(0, 2, (4, 12, 12))
+    pan_t = torch.stack([cats, inst], dim=-1)
+    pan_p = pan_t.clone()
+    pan_p[..., 0] = (pan_p[..., 0] + (torch.rand(4, 12, 12) < 0.2).long()) % 3
+    pq = ma.detection.PanopticQuality(things={0, 1}, stuffs={2})
+    for i in range(rank, 4, world_size):
+        pq.update(pan_p[i : i + 1], pan_t[i : i + 1])
+    out = pq.compute()
+    pq_ref = ma.detection.PanopticQuality(things={0, 1}, stuffs={2})
+    pq_ref.sync_on_compute = False
+    pq_ref._to_sync = False
+    for i in range(4):
+        pq_ref.update(pan_p[i : i + 1], pan_t[i : i + 1])
+    assert torch.allclose(out, pq_ref.compute(), atol=1e-6), (out, pq_ref.compute())
+
+
+def test_pooled_map_states_ddp():
+    run_distributed(_test_pooled_map_states_ddp, world_size=2)
