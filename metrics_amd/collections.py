@@ -212,6 +212,11 @@ class MetricCollection(ModuleDict):
         for m in self.values(copy_state=False):
             m.update(*args, **m._filter_kwargs(**kwargs))
         if self._enable_compute_groups:
+            # group detection compares state VALUES: lazily-accumulated curve
+            # histograms would leave every curve confmat at zero and merge
+            # unrelated metrics — materialize first
+            for m in self.values(copy_state=False):
+                m._maybe_flush_lazy()
             self._merge_compute_groups()
             self._compute_groups_create_state_ref()
             self._groups_checked = True
@@ -221,6 +226,11 @@ class MetricCollection(ModuleDict):
         return self._compute_and_reduce("compute")
 
     def _compute_and_reduce(self, method_name: str, *args: Any, **kwargs: Any) -> Dict[str, Any]:
+        # lazy curve histograms live on group LEADERS; members alias the
+        # leader's states and may compute first — materialize before iterating
+        if self._groups_checked:
+            for members in self._groups.values():
+                getattr(self, members[0])._maybe_flush_lazy()
         result = {}
         for k, m in self.items(keep_base=True, copy_state=False):
             if method_name == "compute":
@@ -302,6 +312,9 @@ class MetricCollection(ModuleDict):
 
     def _compute_groups_create_state_ref(self, copy: bool = False) -> None:
         """Alias (or deep-copy) the leader's states onto every group member."""
+        if copy:
+            for members in self._groups.values():
+                getattr(self, members[0])._maybe_flush_lazy()
         if not self._state_is_copy:
             for members in self._groups.values():
                 leader = getattr(self, members[0])

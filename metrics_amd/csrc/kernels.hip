@@ -937,6 +937,11 @@ __global__ void k_curve_suffix_tiled(
     if (E && blockIdx.x == 0 && threadIdx.x == 0) atomicAdd(&E[1], 1u);
 }
 
+// single-thread epoch close for the lazy-confmat path (see ma_curve_epoch_bump)
+__global__ void k_epoch_bump(unsigned int* __restrict__ E) {
+    if (threadIdx.x == 0) E[1] += 1u;
+}
+
 // ---------------------------------------------------------------------------
 // fused stat-delta apply: given the kernel scratch [tp|fp|fn|valid] produced
 // by k_mc_stat_*, add the deltas into the four metric state tensors
@@ -1553,6 +1558,15 @@ int ma_curve_suffix(uintptr_t stream, uintptr_t hist, ll outer, int T, int trans
     k_curve_suffix<<<(int)outer, 256, shmem, s>>>((unsigned long long*)hist, T, outer, transposed,
                                                   zero_hist, (unsigned int*)epoch_buf,
                                                   (ll*)confmat);
+    return (int)hipGetLastError();
+}
+
+// close a curve update's device epoch WITHOUT the suffix pass: used by the
+// lazy-confmat path, where histograms accumulate across updates and the
+// suffix/confmat materialization is deferred to compute()/state access.
+int ma_curve_epoch_bump(uintptr_t stream, uintptr_t epoch_buf) {
+    hipStream_t s = (hipStream_t)stream;
+    k_epoch_bump<<<1, 1, 0, s>>>((unsigned int*)epoch_buf);
     return (int)hipGetLastError();
 }
 

@@ -238,8 +238,25 @@ class Metric(Module, ABC):
         self._reductions[name] = dist_reduce_fx
 
     # ---------------------------------------------------------------- forward
+    def _lazy_flush(self) -> None:
+        """Materialize deferred device-side state (lazy curve histograms).
+
+        Metrics on the lazy GPU curve path accumulate raw histograms per
+        update and defer the suffix-sum/confmat materialization to the first
+        state read (compute / sync / state_dict / forward / device move).
+        """
+        from metrics_amd.ops import _hip
+
+        _hip.flush_curve_hist(self)
+
+    def _maybe_flush_lazy(self) -> None:
+        if self.__dict__.get("_lazy_dirty"):
+            self.__dict__["_lazy_dirty"] = False
+            self._lazy_flush()
+
     def forward(self, *args: Any, **kwargs: Any) -> Any:
         """Accumulate the batch into the global state AND return the batch value."""
+        self._maybe_flush_lazy()
         if self.full_state_update or self.full_state_update is None or self.dist_sync_on_step:
             self._forward_cache = self._forward_full_state_update(*args, **kwargs)
         else:
@@ -407,6 +424,7 @@ class Metric(Module, ABC):
         distributed_available: Optional[Callable] = None,
     ) -> None:
         """Synchronize metric states across processes (caches local state for :meth:`unsync`)."""
+        self._maybe_flush_lazy()
         if self._is_synced and should_sync:
             raise MetricsUserError("The Metric has already been synced.")
 
@@ -499,6 +517,7 @@ class Metric(Module, ABC):
     def _wrap_compute(self, compute: Callable) -> Callable:
         @functools.wraps(compute)
         def wrapped_func(*args: Any, **kwargs: Any) -> Any:
+            self._maybe_flush_lazy()
             if self._update_count == 0:
                 rank_zero_warn(
                     f"The ``compute`` method of metric {self.__class__.__name__}"
@@ -557,6 +576,11 @@ class Metric(Module, ABC):
 
     def reset(self) -> None:
         """Reset all metric states to their defaults."""
+        if self.__dict__.get("_lazy_dirty"):
+            self.__dict__["_lazy_dirty"] = False
+            buf = self.__dict__.get("_hip_hist_buf")
+            if buf is not None:
+                buf.zero_()
         self._update_count = 0
         self._forward_cache = None
         self._computed = None
@@ -643,6 +667,7 @@ class Metric(Module, ABC):
 
     def _apply(self, fn: Callable, exclude_state: Sequence[str] = ()) -> Module:
         """Extend nn.Module._apply to also move/cast the registered metric states."""
+        self._maybe_flush_lazy()
         this = super()._apply(fn)
         fs = str(fn)
         is_dtype_fn = any(
@@ -695,6 +720,7 @@ class Metric(Module, ABC):
         prefix: str = "",
         keep_vars: bool = False,
     ) -> Dict[str, Any]:
+        self._maybe_flush_lazy()
         destination = super().state_dict(destination=destination, prefix=prefix, keep_vars=keep_vars)
         # register metric states under `prefix + state_name` (byte-compatible layout)
         for key in self._defaults:
@@ -719,6 +745,11 @@ class Metric(Module, ABC):
         unexpected_keys: List[str],
         error_msgs: List[str],
     ) -> None:
+        if self.__dict__.get("_lazy_dirty"):
+            self.__dict__["_lazy_dirty"] = False
+            buf = self.__dict__.get("_hip_hist_buf")
+            if buf is not None:
+                buf.zero_()
         for key in self._defaults:
             name = prefix + key
             if name in state_dict:
@@ -729,6 +760,7 @@ class Metric(Module, ABC):
 
     def _copy_state_dict(self) -> Dict[str, Union[Tensor, List[Any]]]:
         """Detached deep copy of the current metric states."""
+        self._maybe_flush_lazy()
         cache: Dict[str, Union[Tensor, List[Any]]] = {}
         for attr in self._defaults:
             current_value = getattr(self, attr)

@@ -1165,3 +1165,112 @@ def test_exact_multiclass_roc_prc_gpu_batched():
                     assert torch.allclose(x.cpu(), y, atol=1e-6), cls.__name__
             else:
                 assert torch.allclose(a.cpu(), b, atol=1e-6), cls.__name__
+
+
+def test_lazy_curve_confmat_accumulation():
+    """Lazy curve path: histograms accumulate across updates; the suffix/
+    confmat materialization defers to the first state read and matches the
+    per-update (eager) result exactly."""
+    import metrics_amd as ma
+
+    torch.manual_seed(11)
+    batches = [
+        (torch.randn(256, 10, device="cuda"), torch.randint(0, 10, (256,), device="cuda"))
+        for _ in range(4)
+    ]
+    lazy = ma.MulticlassAUROC(num_classes=10, thresholds=50).to("cuda")
+    for p, t in batches:
+        lazy.update(p, t)
+    # histogram accumulated, confmat still all-zero until a state read
+    assert lazy.__dict__.get("_lazy_dirty") is True
+    assert int(lazy.confmat.sum().item()) == 0
+    out_lazy = lazy.compute()
+    assert lazy.__dict__.get("_lazy_dirty") is False
+
+    # eager reference: flush after every update
+    eager = ma.MulticlassAUROC(num_classes=10, thresholds=50).to("cuda")
+    for p, t in batches:
+        eager.update(p, t)
+        eager._maybe_flush_lazy()
+    assert torch.equal(lazy.confmat, eager.confmat)
+    assert torch.allclose(out_lazy, eager.compute())
+
+    # CPU oracle
+    cpu = ma.MulticlassAUROC(num_classes=10, thresholds=50)
+    for p, t in batches:
+        cpu.update(p.cpu().float(), t.cpu())
+    assert torch.allclose(out_lazy.cpu(), cpu.compute(), atol=1e-5)
+
+
+def test_lazy_curve_state_dict_and_reset():
+    import metrics_amd as ma
+
+    torch.manual_seed(12)
+    p = torch.randn(128, 5, device="cuda")
+    t = torch.randint(0, 5, (128,), device="cuda")
+    m = ma.MulticlassPrecisionRecallCurve(num_classes=5, thresholds=20).to("cuda")
+    m.update(p, t)
+    sd = m.state_dict()  # state access must flush
+    assert int(sd["confmat"].sum().item()) > 0
+    m2 = ma.MulticlassPrecisionRecallCurve(num_classes=5, thresholds=20).to("cuda")
+    m2.update(p, t)  # pending lazy counts, then load over them
+    m2.load_state_dict(sd)
+    pr1 = m.compute()
+    pr2 = m2.compute()
+    for a, b in zip(pr1, pr2):
+        assert torch.equal(a, b)
+    # reset drops pending histogram counts
+    m3 = ma.MulticlassPrecisionRecallCurve(num_classes=5, thresholds=20).to("cuda")
+    m3.update(p, t)
+    m3.reset()
+    m3.update(p, t)
+    pr3 = m3.compute()
+    for a, b in zip(pr1, pr3):
+        assert torch.equal(a, b)
+
+
+def test_lazy_curve_forward_matches_reference_semantics():
+    import metrics_amd as ma
+
+    torch.manual_seed(13)
+    m = ma.MulticlassAUROC(num_classes=5, thresholds=25).to("cuda")
+    cpu = ma.MulticlassAUROC(num_classes=5, thresholds=25)
+    for i in range(3):
+        p = torch.randn(64, 5, device="cuda")
+        t = torch.randint(0, 5, (64,), device="cuda")
+        v_gpu = m(p, t)
+        v_cpu = cpu(p.cpu().float(), t.cpu())
+        assert torch.allclose(v_gpu.cpu(), v_cpu, atol=1e-5), i
+    assert torch.allclose(m.compute().cpu(), cpu.compute(), atol=1e-5)
+
+
+def test_lazy_curve_collection_fused_and_groups():
+    """The fused collection path and compute groups stay correct with lazy
+    curve state (members alias the leader's confmat; flush happens before
+    any member computes)."""
+    import metrics_amd as ma
+
+    torch.manual_seed(14)
+    coll = ma.MetricCollection(
+        {
+            "acc": ma.MulticlassAccuracy(num_classes=8, average="micro"),
+            "auroc": ma.MulticlassAUROC(num_classes=8, thresholds=40),
+            "ap": ma.MulticlassAveragePrecision(num_classes=8, thresholds=40),
+        }
+    ).to("cuda")
+    cpu = ma.MetricCollection(
+        {
+            "acc": ma.MulticlassAccuracy(num_classes=8, average="micro"),
+            "auroc": ma.MulticlassAUROC(num_classes=8, thresholds=40),
+            "ap": ma.MulticlassAveragePrecision(num_classes=8, thresholds=40),
+        }
+    )
+    for i in range(3):
+        p = torch.randn(256, 8, device="cuda")
+        t = torch.randint(0, 8, (256,), device="cuda")
+        coll.update(p, t)
+        cpu.update(p.cpu().float(), t.cpu())
+    out = coll.compute()
+    ref = cpu.compute()
+    for k in ref:
+        assert torch.allclose(out[k].cpu(), ref[k], atol=1e-5), k
