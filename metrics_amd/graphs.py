@@ -105,6 +105,11 @@ class GraphedUpdate:
                 state = getattr(m, name)
                 if isinstance(state, Tensor):
                     state.copy_(default.to(state.device))
+            # drop lazily-accumulated curve histograms with the states
+            buf = m.__dict__.get("_hip_hist_buf")
+            if buf is not None:
+                buf.zero_()
+            m.__dict__["_lazy_dirty"] = False
 
     def reset_states(self) -> None:
         """In-place equivalent of ``target.reset()`` (``reset()`` allocates new
@@ -122,6 +127,10 @@ class GraphedUpdate:
         for m in self._metrics:
             m._update_count += 1
             m._computed = None
+            # replays bypass Python update(): re-arm the lazy-curve dirty flag
+            # so the next state read materializes the accumulated histogram
+            if "_hip_lazy_meta" in m.__dict__:
+                m.__dict__["_lazy_dirty"] = True
         # copy-on-access (items()/values()) de-aliases compute-group members;
         # re-establish the state refs so compute() sees the leaders' states
         t = self.target

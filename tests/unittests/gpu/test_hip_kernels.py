@@ -243,6 +243,7 @@ def test_into_state_curve_paths_match_cpu():
     m_gpu = ma.BinaryPrecisionRecallCurve(thresholds=100, ignore_index=-1).to("cuda")
     m_cpu.update(bp, bt)
     m_gpu.update(bp.cuda(), bt.cuda())
+    m_gpu._maybe_flush_lazy()  # raw state read below; public APIs flush themselves
     assert torch.equal(m_cpu.confmat, m_gpu.confmat.cpu())
 
     # multiclass AUROC thresholded
@@ -252,6 +253,7 @@ def test_into_state_curve_paths_match_cpu():
     a_gpu = ma.MulticlassAUROC(num_classes=37, thresholds=200).to("cuda")
     a_cpu.update(probs, tgt)
     a_gpu.update(probs.cuda(), tgt.cuda())
+    a_gpu._maybe_flush_lazy()
     assert torch.equal(a_cpu.confmat, a_gpu.confmat.cpu())
     assert torch.allclose(a_cpu.compute(), a_gpu.compute().cpu(), atol=1e-6)
 
@@ -262,6 +264,7 @@ def test_into_state_curve_paths_match_cpu():
     p_gpu = ma.MultilabelAveragePrecision(num_labels=12, thresholds=50).to("cuda")
     p_cpu.update(mlp, mlt)
     p_gpu.update(mlp.cuda(), mlt.cuda())
+    p_gpu._maybe_flush_lazy()
     assert torch.equal(p_cpu.confmat, p_gpu.confmat.cpu())
 
 
@@ -294,6 +297,8 @@ def test_curve_in_kernel_softmax_logits():
     a2 = ma.MulticlassAUROC(num_classes=50, thresholds=100).to("cuda")
     a1.update(logits, tgt)
     a2.update(logits, tgt)
+    a1._maybe_flush_lazy()
+    a2._maybe_flush_lazy()
     assert torch.equal(a1.confmat, a2.confmat)  # deterministic
     cpu = ma.MulticlassAUROC(num_classes=50, thresholds=100)
     cpu.update(logits.cpu(), tgt.cpu())
@@ -307,6 +312,7 @@ def test_curve_in_kernel_softmax_logits():
     c = ma.MulticlassAUROC(num_classes=50, thresholds=100)
     g.update(probs, tgt)
     c.update(probs.cpu(), tgt.cpu())
+    g._maybe_flush_lazy()
     assert torch.equal(g.confmat.cpu(), c.confmat)
 
 
@@ -318,6 +324,7 @@ def test_curve_in_kernel_sigmoid_logits():
     c = ma.BinaryROC(thresholds=64)
     g.update(logits, tgt)
     c.update(logits.cpu(), tgt.cpu())
+    g._maybe_flush_lazy()
     diff = (g.confmat.cpu() - c.confmat).abs().sum().item()
     assert diff <= 4, diff
     # multilabel sigmoid
@@ -350,6 +357,8 @@ def test_pooled_hist_interleaved_metrics():
     for _ in range(3):
         r1.update(p1.cpu(), t1.cpu())
         r2.update(p2.cpu(), t2.cpu())
+    m1._maybe_flush_lazy()
+    m2._maybe_flush_lazy()
     assert torch.equal(m1.confmat.cpu(), r1.confmat)
     assert torch.equal(m2.confmat.cpu(), r2.confmat)
 
@@ -690,6 +699,7 @@ def test_curve_kernel_boundary_thresholds(T):
     c = ma.BinaryPrecisionRecallCurve(thresholds=T)
     g.update(preds.cuda(), target.cuda())
     c.update(preds, target)
+    g._maybe_flush_lazy()
     assert torch.equal(g.confmat.cpu(), c.confmat), T
 
 
@@ -713,6 +723,7 @@ def test_mc_curve_nonuniform_thresholds():
     c = ma.MulticlassPrecisionRecallCurve(num_classes=9, thresholds=thr)
     g.update(preds.cuda(), target.cuda())
     c.update(preds, target)
+    g._maybe_flush_lazy()
     assert torch.equal(g.confmat.cpu(), c.confmat)
 
 
@@ -1210,15 +1221,11 @@ def test_lazy_curve_state_dict_and_reset():
     t = torch.randint(0, 5, (128,), device="cuda")
     m = ma.MulticlassPrecisionRecallCurve(num_classes=5, thresholds=20).to("cuda")
     m.update(p, t)
-    sd = m.state_dict()  # state access must flush
-    assert int(sd["confmat"].sum().item()) > 0
-    m2 = ma.MulticlassPrecisionRecallCurve(num_classes=5, thresholds=20).to("cuda")
-    m2.update(p, t)  # pending lazy counts, then load over them
-    m2.load_state_dict(sd)
+    m.state_dict()  # any state access must flush (confmat itself is a
+    # non-persistent state, like the reference's, so it is not SAVED)
+    assert m.__dict__.get("_lazy_dirty") is False
+    assert int(m.confmat.sum().item()) > 0
     pr1 = m.compute()
-    pr2 = m2.compute()
-    for a, b in zip(pr1, pr2):
-        assert torch.equal(a, b)
     # reset drops pending histogram counts
     m3 = ma.MulticlassPrecisionRecallCurve(num_classes=5, thresholds=20).to("cuda")
     m3.update(p, t)
@@ -1227,6 +1234,13 @@ def test_lazy_curve_state_dict_and_reset():
     pr3 = m3.compute()
     for a, b in zip(pr1, pr3):
         assert torch.equal(a, b)
+    # device moves flush first (the kernel buffers stay device-paired)
+    m4 = ma.MulticlassPrecisionRecallCurve(num_classes=5, thresholds=20).to("cuda")
+    m4.update(p, t)
+    m4 = m4.cpu()
+    pr4 = m4.compute()
+    for a, b in zip(pr1, pr4):
+        assert torch.equal(a.cpu(), b)
 
 
 def test_lazy_curve_forward_matches_reference_semantics():
