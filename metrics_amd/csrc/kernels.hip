@@ -948,7 +948,9 @@ __global__ void k_epoch_bump(unsigned int* __restrict__ E) {
 // (tn += valid - tp - fp - fn). ONE launch replaces ~8 small torch kernels.
 __global__ void k_apply_stat_deltas(
     unsigned long long* __restrict__ scratch /* 3*C + 1 */, ll C,
-    ll* __restrict__ tp, ll* __restrict__ fp, ll* __restrict__ tn, ll* __restrict__ fn) {
+    ll* __restrict__ tp, ll* __restrict__ fp, ll* __restrict__ tn, ll* __restrict__ fn,
+    unsigned int* __restrict__ E /* nullable: close the curve epoch here */) {
+    if (E && threadIdx.x == 0) E[1] += 1u;
     // SINGLE block: every thread can read the valid slot before thread 0
     // zeroes it (post-sync), so the whole scratch is consumed and re-zeroed
     // in one launch with no cross-block race and no host-side epochs —
@@ -1008,7 +1010,9 @@ __global__ void k_exact_apply(unsigned long long* __restrict__ scratch, ll C, ll
 __global__ void k_apply_stat_exact(
     unsigned long long* __restrict__ scratch /* 3*C + 1 */, ll C, ll B,
     ll* __restrict__ tp, ll* __restrict__ fp, ll* __restrict__ tn, ll* __restrict__ fn,
-    ll* __restrict__ correct, ll* __restrict__ total) {
+    ll* __restrict__ correct, ll* __restrict__ total,
+    unsigned int* __restrict__ E /* nullable: close the curve epoch here */) {
+    if (E && threadIdx.x == 0) E[1] += 1u;
     __shared__ unsigned long long part[256];
     __shared__ unsigned long long valid_s;
     if (threadIdx.x == 0) valid_s = scratch[3 * C];
@@ -1571,10 +1575,11 @@ int ma_curve_epoch_bump(uintptr_t stream, uintptr_t epoch_buf) {
 }
 
 int ma_apply_stat_deltas(uintptr_t stream, uintptr_t scratch, ll C, uintptr_t tp,
-                         uintptr_t fp, uintptr_t tn, uintptr_t fn) {
+                         uintptr_t fp, uintptr_t tn, uintptr_t fn, uintptr_t epoch_buf) {
     hipStream_t s = (hipStream_t)stream;
     k_apply_stat_deltas<<<1, 256, 0, s>>>(
-        (unsigned long long*)scratch, C, (ll*)tp, (ll*)fp, (ll*)tn, (ll*)fn);
+        (unsigned long long*)scratch, C, (ll*)tp, (ll*)fp, (ll*)tn, (ll*)fn,
+        (unsigned int*)epoch_buf);
     return (int)hipGetLastError();
 }
 
@@ -1600,10 +1605,11 @@ int ma_curve_auc_from_confmat(uintptr_t stream, uintptr_t confmat, int T, ll C, 
 
 int ma_apply_stat_exact(uintptr_t stream, uintptr_t scratch, ll C, ll B, uintptr_t tp,
                         uintptr_t fp, uintptr_t tn, uintptr_t fn, uintptr_t correct,
-                        uintptr_t total) {
+                        uintptr_t total, uintptr_t epoch_buf) {
     hipStream_t s = (hipStream_t)stream;
     k_apply_stat_exact<<<1, 256, 0, s>>>((unsigned long long*)scratch, C, B, (ll*)tp, (ll*)fp,
-                                         (ll*)tn, (ll*)fn, (ll*)correct, (ll*)total);
+                                         (ll*)tn, (ll*)fn, (ll*)correct, (ll*)total,
+                                         (unsigned int*)epoch_buf);
     return (int)hipGetLastError();
 }
 

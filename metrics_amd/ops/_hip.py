@@ -57,7 +57,7 @@ _SIGNATURES = {
     "ma_curve_epoch_bump": [_U64, _U64],
     "ma_apply_stat_deltas": [_U64, _U64, _LL, _U64, _U64, _U64, _U64],
     "ma_exact_apply": [_U64, _U64, _LL, _LL, _I, _U64, _U64],
-    "ma_apply_stat_exact": [_U64, _U64, _LL, _LL, _U64, _U64, _U64, _U64, _U64, _U64],
+    "ma_apply_stat_exact": [_U64, _U64, _LL, _LL, _U64, _U64, _U64, _U64, _U64, _U64, _U64],
     "ma_curve_auc_from_confmat": [_U64, _U64, _I, _LL, _I, _U64, _U64],
     "ma_linear_stat_compute": [_U64, _U64, _U64, _U64, _U64, _LL, _F, _F, _F, _F, _F, _F, _F, _F, _I, _I, _F, _F, _F, _U64],
     "ma_err_reduce": [_U64, _U64, _U64, _I, _LL, _I, _D, _U64, _I, _I, _U64],
@@ -841,6 +841,7 @@ def curve_hist_into_confmat(
     probs: Tensor, target: Tensor, thresholds: Tensor, ignore_index: Optional[int],
     confmat_state: Tensor, mode: int, norm: Optional[str] = None, owner=None,
     stats_ready: bool = False, lazy: Optional[bool] = None,
+    skip_epoch_bump: bool = False,
 ) -> None:
     """Bucketized histogram + transposed suffix-sum accumulated DIRECTLY into the
     metric's confmat state ((T,2,2) binary / (T,C,2,2) multiclass|multilabel) —
@@ -947,7 +948,7 @@ def curve_hist_into_confmat(
     if lazy is None:
         lazy = owner is not None
     if lazy and owner is not None:
-        if norm is not None:
+        if norm is not None and not skip_epoch_bump:
             rc = lib.ma_curve_epoch_bump(_stream(), _epoch_buf(dev, owner).data_ptr())
             _check(rc, "ma_curve_epoch_bump")
         owner.__dict__["_hip_lazy_meta"] = (outer, T, transposed)
@@ -1047,6 +1048,7 @@ def mc_stat_into(
         fp.data_ptr(),
         tn.data_ptr(),
         fn.data_ptr(),
+        0,
     )
     _check(rc, "ma_apply_stat_deltas")
 
@@ -1221,8 +1223,9 @@ def curve_auc_from_confmat(confmat_state: Tensor, mode: int) -> Tuple[Tensor, Te
 
 def mc_fused_collection_update(
     preds: Tensor, target: Tensor, num_classes: int, ignore_index: Optional[int],
-    stat=None, confmat=None, exact=None, rowstats=None,
-) -> None:
+    stat=None, confmat=None, exact=None, rowstats=None, defer_apply: bool = False,
+    bump_epoch_ptr: int = 0,
+):
     """Collection-level fused update: ONE pass over the (B, C) logits feeds the
     stat-scores, confusion-matrix and exact-match leaders at once.
 
@@ -1268,18 +1271,27 @@ def mc_fused_collection_update(
             s_valid_ptr,
         )
         _check(rc, "ma_mc_stat_labels")
-    if exact is not None:
-        # fused epilogue: stat-delta apply + exact-match in one launch
-        correct, total = exact
-        rc = lib.ma_apply_stat_exact(
-            _stream(), scratch.data_ptr(), C, B,
-            tp.data_ptr(), fp.data_ptr(), tn.data_ptr(), fn.data_ptr(),
-            correct.data_ptr(), total.data_ptr(),
-        )
-        _check(rc, "ma_apply_stat_exact")
-    else:
-        rc = lib.ma_apply_stat_deltas(
-            _stream(), scratch.data_ptr(), C,
-            tp.data_ptr(), fp.data_ptr(), tn.data_ptr(), fn.data_ptr(),
-        )
-        _check(rc, "ma_apply_stat_deltas")
+    def _apply_pass() -> None:
+        # fused epilogue: stat-delta apply (+ exact-match) in ONE launch; when
+        # the lazy curve hist rides the same stream the caller defers this so
+        # it lands AFTER the hist kernel and closes the device epoch for free
+        # (a separate 1-thread bump kernel costs a full ~4us dispatch)
+        if exact is not None:
+            correct, total = exact
+            rc2 = lib.ma_apply_stat_exact(
+                _stream(), scratch.data_ptr(), C, B,
+                tp.data_ptr(), fp.data_ptr(), tn.data_ptr(), fn.data_ptr(),
+                correct.data_ptr(), total.data_ptr(), bump_epoch_ptr,
+            )
+            _check(rc2, "ma_apply_stat_exact")
+        else:
+            rc2 = lib.ma_apply_stat_deltas(
+                _stream(), scratch.data_ptr(), C,
+                tp.data_ptr(), fp.data_ptr(), tn.data_ptr(), fn.data_ptr(), bump_epoch_ptr,
+            )
+            _check(rc2, "ma_apply_stat_deltas")
+
+    if defer_apply:
+        return _apply_pass
+    _apply_pass()
+    return None
