@@ -55,7 +55,7 @@ _SIGNATURES = {
     "ma_multiclass_curve_hist": [_U64, _U64, _I, _U64, _LL, _LL, _U64, _I, _LL, _I, _I, _I, _F, _F, _I, _U64, _U64, _U64, _I, _I, _I, _U64],
     "ma_curve_suffix": [_U64, _U64, _LL, _I, _I, _I, _U64, _U64],
     "ma_curve_epoch_bump": [_U64, _U64],
-    "ma_apply_stat_deltas": [_U64, _U64, _LL, _U64, _U64, _U64, _U64],
+    "ma_apply_stat_deltas": [_U64, _U64, _LL, _U64, _U64, _U64, _U64, _U64],
     "ma_exact_apply": [_U64, _U64, _LL, _LL, _I, _U64, _U64],
     "ma_apply_stat_exact": [_U64, _U64, _LL, _LL, _U64, _U64, _U64, _U64, _U64, _U64, _U64],
     "ma_curve_auc_from_confmat": [_U64, _U64, _I, _LL, _I, _U64, _U64],
