@@ -937,6 +937,83 @@ __global__ void k_curve_suffix_tiled(
     if (E && blockIdx.x == 0 && threadIdx.x == 0) atomicAdd(&E[1], 1u);
 }
 
+// ---------------------------------------------------------------------------
+// fused (C,C) confusion-matrix scalar computes: MCC + unweighted Cohen kappa +
+// macro Jaccard in TWO launches. The eager torch chains emit ~25 small
+// kernels (row/col sums, traces, dot products, guarded divides) per compute;
+// at ~4us dispatch each that is pure launch overhead. Phase A tiles columns
+// (coalesced) and emits row sums (wave-reduced, one atomic per row-tile),
+// column sums and the diagonal; phase B (one block) folds the C-sized
+// vectors into the three scalars and re-zeroes the row-sum scratch in-flight.
+__global__ void k_confmat_moments(
+    const ll* __restrict__ cm, ll C,
+    unsigned long long* __restrict__ tk /* (C), pre-zeroed */,
+    unsigned long long* __restrict__ pk, unsigned long long* __restrict__ diag) {
+    const ll c0 = (ll)blockIdx.x * blockDim.x;
+    const ll c = c0 + threadIdx.x;
+    const int lane = threadIdx.x & (WAVE - 1);
+    unsigned long long col_acc = 0;
+    for (ll r = 0; r < C; r++) {
+        unsigned long long v = (c < C) ? (unsigned long long)cm[r * C + c] : 0ULL;
+        col_acc += v;
+        if (c == r) diag[r] = v;
+        // wave-reduce this tile's contribution to row r: one atomic per wave
+        unsigned long long rv = v;
+        for (int off = WAVE / 2; off > 0; off >>= 1) rv += __shfl_down(rv, off);
+        if (lane == 0 && rv) atomicAdd(&tk[r], rv);
+    }
+    if (c < C) pk[c] = col_acc;
+}
+
+__global__ void k_confmat_scalars(
+    const unsigned long long* __restrict__ pk, const unsigned long long* __restrict__ diag,
+    unsigned long long* __restrict__ tk /* consumed + re-zeroed */, ll C,
+    float zero_division, float* __restrict__ out /* [mcc, kappa, jaccard_macro] */) {
+    __shared__ double sh[6][256];
+    double s_tot = 0, s_tr = 0, s_tkpk = 0, s_tk2 = 0, s_pk2 = 0;
+    double j_sum = 0, j_cnt = 0;
+    for (ll i = threadIdx.x; i < C; i += blockDim.x) {
+        const double t = (double)tk[i];
+        const double p = (double)pk[i];
+        const double d = (double)diag[i];
+        tk[i] = 0;  // scratch consumed: zero for the next call
+        s_tot += t;
+        s_tr += d;
+        s_tkpk += t * p;
+        s_tk2 += t * t;
+        s_pk2 += p * p;
+        const double den = t + p - d;
+        if (t + p > 0) {
+            j_cnt += 1.0;
+            j_sum += den > 0 ? d / den : (double)zero_division;
+        }
+    }
+    sh[0][threadIdx.x] = s_tot; sh[1][threadIdx.x] = s_tr; sh[2][threadIdx.x] = s_tkpk;
+    sh[3][threadIdx.x] = s_tk2; sh[4][threadIdx.x] = s_pk2; sh[5][threadIdx.x] = j_sum;
+    __shared__ double shc[256];
+    shc[threadIdx.x] = j_cnt;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+        if (threadIdx.x < off) {
+            for (int q = 0; q < 6; q++) sh[q][threadIdx.x] += sh[q][threadIdx.x + off];
+            shc[threadIdx.x] += shc[threadIdx.x + off];
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        // float32 to match the torch reference chains bit-closely
+        const float s = (float)sh[0][0], c = (float)sh[1][0];
+        const float tkpk = (float)sh[2][0], tk2 = (float)sh[3][0], pk2 = (float)sh[4][0];
+        const float cov = c * s - tkpk;
+        const float den = (s * s - pk2) * (s * s - tk2);
+        out[0] = den == 0.0f ? 0.0f : cov / sqrtf(fmaxf(den, 1.1754944e-38f));
+        const float po_n = s - c;            // sum(w*cm),  w = 1 - I
+        const float pe_n = s - tkpk / s;     // sum(w*E)
+        out[1] = 1.0f - po_n / pe_n;         // 0/0 -> nan, x/0 -> inf: torch parity
+        out[2] = shc[0] > 0 ? (float)(sh[5][0] / shc[0]) : zero_division;
+    }
+}
+
 // single-thread epoch close for the lazy-confmat path (see ma_curve_epoch_bump)
 __global__ void k_epoch_bump(unsigned int* __restrict__ E) {
     if (threadIdx.x == 0) E[1] += 1u;
@@ -1568,6 +1645,18 @@ int ma_curve_suffix(uintptr_t stream, uintptr_t hist, ll outer, int T, int trans
 // close a curve update's device epoch WITHOUT the suffix pass: used by the
 // lazy-confmat path, where histograms accumulate across updates and the
 // suffix/confmat materialization is deferred to compute()/state access.
+int ma_confmat_scalars(uintptr_t stream, uintptr_t cm, ll C, uintptr_t scratch /* 3C u64 */,
+                       float zero_division, uintptr_t out /* 3 f32 */) {
+    hipStream_t s = (hipStream_t)stream;
+    unsigned long long* tk = (unsigned long long*)scratch;
+    unsigned long long* pk = tk + C;
+    unsigned long long* diag = pk + C;
+    const int grid = (int)((C + 255) / 256);
+    k_confmat_moments<<<grid, 256, 0, s>>>((const ll*)cm, C, tk, pk, diag);
+    k_confmat_scalars<<<1, 256, 0, s>>>(pk, diag, tk, C, zero_division, (float*)out);
+    return (int)hipGetLastError();
+}
+
 int ma_curve_epoch_bump(uintptr_t stream, uintptr_t epoch_buf) {
     hipStream_t s = (hipStream_t)stream;
     k_epoch_bump<<<1, 1, 0, s>>>((unsigned int*)epoch_buf);

@@ -21,6 +21,11 @@ from metrics_amd.functional.classification.confusion_matrix import (
 
 def _cohen_kappa_reduce(confmat: Tensor, weights: Optional[str] = None) -> Tensor:
     """Kappa from a confusion matrix with optional linear/quadratic disagreement weighting."""
+    if weights is None and confmat.is_cuda and confmat.ndim == 2:
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            return _hip.confmat_scalars(confmat)[1].clone()
     confmat = confmat.float()
     num_classes = confmat.shape[0]
     sum0 = confmat.sum(dim=0, keepdim=True)

@@ -40,6 +40,17 @@ def _jaccard_index_reduce(
 
     ignore_index_cond = ignore_index is not None and 0 <= ignore_index < confmat.shape[0]
     multilabel = confmat.ndim == 3
+    if (
+        average == "macro"
+        and not multilabel
+        and not ignore_index_cond
+        and confmat.is_cuda
+        and zero_division == 0.0
+    ):
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            return _hip.confmat_scalars(confmat)[2].clone()
     if multilabel:
         num = confmat[:, 1, 1]
         denom = confmat[:, 1, 1] + confmat[:, 0, 1] + confmat[:, 1, 0]

@@ -55,6 +55,7 @@ _SIGNATURES = {
     "ma_multiclass_curve_hist": [_U64, _U64, _I, _U64, _LL, _LL, _U64, _I, _LL, _I, _I, _I, _F, _F, _I, _U64, _U64, _U64, _I, _I, _I, _U64],
     "ma_curve_suffix": [_U64, _U64, _LL, _I, _I, _I, _U64, _U64],
     "ma_curve_epoch_bump": [_U64, _U64],
+    "ma_confmat_scalars": [_U64, _U64, _LL, _U64, ctypes.c_float, _U64],
     "ma_apply_stat_deltas": [_U64, _U64, _LL, _U64, _U64, _U64, _U64, _U64],
     "ma_exact_apply": [_U64, _U64, _LL, _LL, _I, _U64, _U64],
     "ma_apply_stat_exact": [_U64, _U64, _LL, _LL, _U64, _U64, _U64, _U64, _U64, _U64, _U64],
@@ -967,6 +968,48 @@ def curve_hist_into_confmat(
     _check(rc, "ma_curve_suffix")
 
 
+import weakref as _weakref
+
+_CONFMAT_MUTATIONS: "_weakref.WeakKeyDictionary" = _weakref.WeakKeyDictionary()
+
+
+def _mark_kernel_mutated(t: Tensor) -> None:
+    """Raw-pointer kernel writes bypass torch's _version counter; note them so
+    version-keyed caches (confmat_scalars) invalidate correctly."""
+    _CONFMAT_MUTATIONS[t] = _CONFMAT_MUTATIONS.get(t, 0) + 1
+
+
+_CONFMAT_SCALARS_CACHE: "_weakref.WeakKeyDictionary" = _weakref.WeakKeyDictionary()
+_CONFMAT_SCRATCH: dict = {}
+
+
+def confmat_scalars(confmat: Tensor, zero_division: float = 0.0) -> Tensor:
+    """Fused (C,C) confusion-matrix scalars: (mcc, unweighted kappa, macro
+    jaccard) in two launches, cached per (tensor, version) so the metrics of a
+    compute group (which alias one confmat state) pay for it once."""
+    ver = (confmat._version, _CONFMAT_MUTATIONS.get(confmat, 0))
+    hit = _CONFMAT_SCALARS_CACHE.get(confmat)
+    if hit is not None and hit[0] == ver and hit[2] == zero_division:
+        return hit[1]
+    lib = _lib()
+    C = confmat.shape[0]
+    dev = confmat.device
+    key = (C, dev.index)
+    scratch = _CONFMAT_SCRATCH.get(key)
+    if scratch is None:
+        scratch = torch.zeros(3 * C, dtype=torch.long, device=dev)
+        if len(_CONFMAT_SCRATCH) > 16:
+            _CONFMAT_SCRATCH.clear()
+        _CONFMAT_SCRATCH[key] = scratch
+    out = torch.empty(3, dtype=torch.float32, device=dev)
+    cm = confmat if confmat.is_contiguous() else confmat.contiguous()
+    rc = lib.ma_confmat_scalars(_stream(), cm.data_ptr(), C, scratch.data_ptr(),
+                                float(zero_division), out.data_ptr())
+    _check(rc, "ma_confmat_scalars")
+    _CONFMAT_SCALARS_CACHE[confmat] = (ver, out, zero_division)
+    return out
+
+
 def flush_curve_hist(owner) -> None:
     """Materialize an owner's lazily-accumulated curve histogram into its
     ``confmat`` state (suffix-sum; re-zeroes the histogram in-flight)."""
@@ -1107,6 +1150,7 @@ def mc_confmat_into(
             dummy[3 * C :].data_ptr(),
         )
         _check(rc, "ma_mc_stat_labels")
+    _mark_kernel_mutated(confmat_state)
 
 
 def mc_exact_into(
@@ -1291,6 +1335,8 @@ def mc_fused_collection_update(
             )
             _check(rc2, "ma_apply_stat_deltas")
 
+    if confmat is not None:
+        _mark_kernel_mutated(confmat)
     if defer_apply:
         return _apply_pass
     _apply_pass()

@@ -1288,3 +1288,70 @@ def test_lazy_curve_collection_fused_and_groups():
     ref = cpu.compute()
     for k in ref:
         assert torch.allclose(out[k].cpu(), ref[k], atol=1e-5), k
+
+
+def test_confmat_scalars_fused_vs_cpu():
+    """Fused MCC/kappa/jaccard scalars match the CPU torch chains, and the
+    version cache invalidates across kernel-side confmat updates."""
+    import metrics_amd as ma
+
+    torch.manual_seed(21)
+    mcc_g = ma.MulticlassMatthewsCorrCoef(num_classes=37).to("cuda")
+    kap_g = ma.MulticlassCohenKappa(num_classes=37).to("cuda")
+    jac_g = ma.MulticlassJaccardIndex(num_classes=37, average="macro").to("cuda")
+    mcc_c = ma.MulticlassMatthewsCorrCoef(num_classes=37)
+    kap_c = ma.MulticlassCohenKappa(num_classes=37)
+    jac_c = ma.MulticlassJaccardIndex(num_classes=37, average="macro")
+    for step in range(3):
+        p = torch.randn(512, 37, device="cuda").softmax(-1)
+        t = torch.randint(0, 37, (512,), device="cuda")
+        for m in (mcc_g, kap_g, jac_g):
+            m.update(p, t)
+        for m in (mcc_c, kap_c, jac_c):
+            m.update(p.cpu(), t.cpu())
+        # compute EVERY step: a stale cache would freeze the value at step 0
+        assert torch.allclose(mcc_g.compute().cpu(), mcc_c.compute(), atol=1e-5), step
+        assert torch.allclose(kap_g.compute().cpu(), kap_c.compute(), atol=1e-5), step
+        assert torch.allclose(jac_g.compute().cpu(), jac_c.compute(), atol=1e-5), step
+
+    # skewed degenerate case: single predicted class
+    m1 = ma.MulticlassMatthewsCorrCoef(num_classes=5).to("cuda")
+    m2 = ma.MulticlassMatthewsCorrCoef(num_classes=5)
+    p = torch.zeros(64, 5, device="cuda")
+    p[:, 2] = 10.0
+    t = torch.randint(0, 5, (64,), device="cuda")
+    m1.update(p, t)
+    m2.update(p.cpu(), t.cpu())
+    assert torch.allclose(m1.compute().cpu(), m2.compute(), atol=1e-6)
+
+
+def test_confmat_scalars_in_collection():
+    """The three confmat-derived metrics share one compute-group confmat: the
+    fused scalars are computed once per generation and stay correct."""
+    import metrics_amd as ma
+
+    torch.manual_seed(22)
+    coll = ma.MetricCollection(
+        {
+            "mcc": ma.MulticlassMatthewsCorrCoef(num_classes=19),
+            "kappa": ma.MulticlassCohenKappa(num_classes=19),
+            "jaccard": ma.MulticlassJaccardIndex(num_classes=19),
+            "confmat": ma.MulticlassConfusionMatrix(num_classes=19),
+        }
+    ).to("cuda")
+    ref = ma.MetricCollection(
+        {
+            "mcc": ma.MulticlassMatthewsCorrCoef(num_classes=19),
+            "kappa": ma.MulticlassCohenKappa(num_classes=19),
+            "jaccard": ma.MulticlassJaccardIndex(num_classes=19),
+            "confmat": ma.MulticlassConfusionMatrix(num_classes=19),
+        }
+    )
+    for _ in range(2):
+        p = torch.randn(300, 19, device="cuda").softmax(-1)
+        t = torch.randint(0, 19, (300,), device="cuda")
+        coll.update(p, t)
+        ref.update(p.cpu(), t.cpu())
+        out, exp = coll.compute(), ref.compute()
+        for k in exp:
+            assert torch.allclose(out[k].float().cpu(), exp[k].float(), atol=1e-5), k
