@@ -970,25 +970,43 @@ def curve_hist_into_confmat(
 
 import weakref as _weakref
 
-_CONFMAT_MUTATIONS: "_weakref.WeakKeyDictionary" = _weakref.WeakKeyDictionary()
+# Tensors cannot key a WeakKeyDictionary (their __eq__ is elementwise), so
+# these caches key on id(tensor) and evict via a weakref finalizer; entries
+# double-check identity through the stored weakref to survive id reuse.
+_CONFMAT_MUTATIONS: dict = {}
+_CONFMAT_SCALARS_CACHE: dict = {}
+_CONFMAT_SCRATCH: dict = {}
+
+
+def _tensor_cache_get(cache: dict, t: Tensor):
+    ent = cache.get(id(t))
+    if ent is not None and ent[0]() is t:
+        return ent[1]
+    return None
+
+
+def _tensor_cache_put(cache: dict, t: Tensor, value) -> None:
+    key = id(t)
+
+    def _evict(_ref, _key=key, _cache=cache):
+        _cache.pop(_key, None)
+
+    cache[key] = (_weakref.ref(t, _evict), value)
 
 
 def _mark_kernel_mutated(t: Tensor) -> None:
     """Raw-pointer kernel writes bypass torch's _version counter; note them so
     version-keyed caches (confmat_scalars) invalidate correctly."""
-    _CONFMAT_MUTATIONS[t] = _CONFMAT_MUTATIONS.get(t, 0) + 1
-
-
-_CONFMAT_SCALARS_CACHE: "_weakref.WeakKeyDictionary" = _weakref.WeakKeyDictionary()
-_CONFMAT_SCRATCH: dict = {}
+    cur = _tensor_cache_get(_CONFMAT_MUTATIONS, t) or 0
+    _tensor_cache_put(_CONFMAT_MUTATIONS, t, cur + 1)
 
 
 def confmat_scalars(confmat: Tensor, zero_division: float = 0.0) -> Tensor:
     """Fused (C,C) confusion-matrix scalars: (mcc, unweighted kappa, macro
     jaccard) in two launches, cached per (tensor, version) so the metrics of a
     compute group (which alias one confmat state) pay for it once."""
-    ver = (confmat._version, _CONFMAT_MUTATIONS.get(confmat, 0))
-    hit = _CONFMAT_SCALARS_CACHE.get(confmat)
+    ver = (confmat._version, _tensor_cache_get(_CONFMAT_MUTATIONS, confmat) or 0)
+    hit = _tensor_cache_get(_CONFMAT_SCALARS_CACHE, confmat)
     if hit is not None and hit[0] == ver and hit[2] == zero_division:
         return hit[1]
     lib = _lib()
@@ -1006,7 +1024,7 @@ def confmat_scalars(confmat: Tensor, zero_division: float = 0.0) -> Tensor:
     rc = lib.ma_confmat_scalars(_stream(), cm.data_ptr(), C, scratch.data_ptr(),
                                 float(zero_division), out.data_ptr())
     _check(rc, "ma_confmat_scalars")
-    _CONFMAT_SCALARS_CACHE[confmat] = (ver, out, zero_division)
+    _tensor_cache_put(_CONFMAT_SCALARS_CACHE, confmat, (ver, out, zero_division))
     return out
 
 
