@@ -21,7 +21,7 @@ from metrics_amd.functional.classification.confusion_matrix import (
 
 def _cohen_kappa_reduce(confmat: Tensor, weights: Optional[str] = None) -> Tensor:
     """Kappa from a confusion matrix with optional linear/quadratic disagreement weighting."""
-    if weights is None and confmat.is_cuda and confmat.ndim == 2:
+    if weights is None and confmat.is_cuda and confmat.ndim == 2 and confmat.dtype == torch.long:
         from metrics_amd.ops import _hip
 
         if _hip.hip_available():

@@ -34,23 +34,26 @@ def _jaccard_index_reduce(
     allowed_average = ["binary", "micro", "macro", "weighted", "none", None]
     if average not in allowed_average:
         raise ValueError(f"The `average` has to be one of {allowed_average}, got {average}.")
+    if (
+        average == "macro"
+        and confmat.ndim == 2
+        and confmat.is_cuda
+        and confmat.dtype == torch.long
+        and zero_division == 0.0
+        and not (ignore_index is not None and 0 <= ignore_index < confmat.shape[0])
+    ):
+        from metrics_amd.ops import _hip
+
+        if _hip.hip_available():
+            # fused two-launch scalar compute on the raw int64 state (MUST run
+            # before the float() cast: the kernel reads 8-byte counts)
+            return _hip.confmat_scalars(confmat)[2].clone()
     confmat = confmat.float()
     if average == "binary":
         return _safe_divide(confmat[1, 1], (confmat[0, 1] + confmat[1, 0] + confmat[1, 1]), zero_division=zero_division)
 
     ignore_index_cond = ignore_index is not None and 0 <= ignore_index < confmat.shape[0]
     multilabel = confmat.ndim == 3
-    if (
-        average == "macro"
-        and not multilabel
-        and not ignore_index_cond
-        and confmat.is_cuda
-        and zero_division == 0.0
-    ):
-        from metrics_amd.ops import _hip
-
-        if _hip.hip_available():
-            return _hip.confmat_scalars(confmat)[2].clone()
     if multilabel:
         num = confmat[:, 1, 1]
         denom = confmat[:, 1, 1] + confmat[:, 0, 1] + confmat[:, 1, 0]

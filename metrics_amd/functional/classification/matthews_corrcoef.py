@@ -27,7 +27,7 @@ def _matthews_corrcoef_reduce(confmat: Tensor) -> Tensor:
     """MCC from a confusion matrix (multilabel (L,2,2) is summed to one 2x2)."""
     if confmat.ndim == 3:  # multilabel
         confmat = confmat.sum(0)
-    if confmat.is_cuda and confmat.numel() != 4:
+    if confmat.is_cuda and confmat.numel() != 4 and confmat.dtype == torch.long and confmat.ndim == 2:
         from metrics_amd.ops import _hip
 
         if _hip.hip_available():
