@@ -996,16 +996,21 @@ def _tensor_cache_put(cache: dict, t: Tensor, value) -> None:
 
 def _mark_kernel_mutated(t: Tensor) -> None:
     """Raw-pointer kernel writes bypass torch's _version counter; note them so
-    version-keyed caches (confmat_scalars) invalidate correctly."""
-    cur = _tensor_cache_get(_CONFMAT_MUTATIONS, t) or 0
-    _tensor_cache_put(_CONFMAT_MUTATIONS, t, cur + 1)
+    version-keyed caches (confmat_scalars) invalidate correctly. Steady-state
+    cost is one dict hit + in-place increment (this runs every update)."""
+    ent = _CONFMAT_MUTATIONS.get(id(t))
+    if ent is not None and ent[0]() is t:
+        ent[1][0] += 1
+        return
+    _tensor_cache_put(_CONFMAT_MUTATIONS, t, [1])
 
 
 def confmat_scalars(confmat: Tensor, zero_division: float = 0.0) -> Tensor:
     """Fused (C,C) confusion-matrix scalars: (mcc, unweighted kappa, macro
     jaccard) in two launches, cached per (tensor, version) so the metrics of a
     compute group (which alias one confmat state) pay for it once."""
-    ver = (confmat._version, _tensor_cache_get(_CONFMAT_MUTATIONS, confmat) or 0)
+    muts = _tensor_cache_get(_CONFMAT_MUTATIONS, confmat)
+    ver = (confmat._version, muts[0] if muts is not None else 0)
     hit = _tensor_cache_get(_CONFMAT_SCALARS_CACHE, confmat)
     if hit is not None and hit[0] == ver and hit[2] == zero_division:
         return hit[1]
