@@ -946,14 +946,19 @@ __global__ void k_curve_suffix_tiled(
 // column sums and the diagonal; phase B (one block) folds the C-sized
 // vectors into the three scalars and re-zeroes the row-sum scratch in-flight.
 __global__ void k_confmat_moments(
-    const ll* __restrict__ cm, ll C,
+    const ll* __restrict__ cm, ll C, ll rows_per_block,
     unsigned long long* __restrict__ tk /* (C), pre-zeroed */,
-    unsigned long long* __restrict__ pk, unsigned long long* __restrict__ diag) {
-    const ll c0 = (ll)blockIdx.x * blockDim.x;
-    const ll c = c0 + threadIdx.x;
+    unsigned long long* __restrict__ pk /* (C), pre-zeroed */,
+    unsigned long long* __restrict__ diag) {
+    // 2D grid: x tiles 256 columns (coalesced), y tiles rows — a column-only
+    // grid is 4 blocks at C=1000 (0.4% occupancy, measured 575us); row tiling
+    // brings it to the ~10us memory floor at the cost of pk atomics
+    const ll c = (ll)blockIdx.x * blockDim.x + threadIdx.x;
+    const ll r_lo = (ll)blockIdx.y * rows_per_block;
+    const ll r_hi = min(r_lo + rows_per_block, C);
     const int lane = threadIdx.x & (WAVE - 1);
     unsigned long long col_acc = 0;
-    for (ll r = 0; r < C; r++) {
+    for (ll r = r_lo; r < r_hi; r++) {
         unsigned long long v = (c < C) ? (unsigned long long)cm[r * C + c] : 0ULL;
         col_acc += v;
         if (c == r) diag[r] = v;
@@ -962,11 +967,11 @@ __global__ void k_confmat_moments(
         for (int off = WAVE / 2; off > 0; off >>= 1) rv += __shfl_down(rv, off);
         if (lane == 0 && rv) atomicAdd(&tk[r], rv);
     }
-    if (c < C) pk[c] = col_acc;
+    if (c < C && col_acc) atomicAdd(&pk[c], col_acc);
 }
 
 __global__ void k_confmat_scalars(
-    const unsigned long long* __restrict__ pk, const unsigned long long* __restrict__ diag,
+    unsigned long long* __restrict__ pk, const unsigned long long* __restrict__ diag,
     unsigned long long* __restrict__ tk /* consumed + re-zeroed */, ll C,
     float zero_division, float* __restrict__ out /* [mcc, kappa, jaccard_macro] */) {
     __shared__ double sh[6][256];
@@ -977,6 +982,7 @@ __global__ void k_confmat_scalars(
         const double p = (double)pk[i];
         const double d = (double)diag[i];
         tk[i] = 0;  // scratch consumed: zero for the next call
+        pk[i] = 0;
         s_tot += t;
         s_tr += d;
         s_tkpk += t * p;
@@ -1651,8 +1657,14 @@ int ma_confmat_scalars(uintptr_t stream, uintptr_t cm, ll C, uintptr_t scratch /
     unsigned long long* tk = (unsigned long long*)scratch;
     unsigned long long* pk = tk + C;
     unsigned long long* diag = pk + C;
-    const int grid = (int)((C + 255) / 256);
-    k_confmat_moments<<<grid, 256, 0, s>>>((const ll*)cm, C, tk, pk, diag);
+    const int gx = (int)((C + 255) / 256);
+    // enough row tiles to fill the chip (>=1024 blocks), bounded by C
+    ll row_tiles = (1024 + gx - 1) / gx;
+    if (row_tiles > C) row_tiles = C;
+    const ll rows_per_block = (C + row_tiles - 1) / row_tiles;
+    row_tiles = (C + rows_per_block - 1) / rows_per_block;
+    dim3 grid((unsigned)gx, (unsigned)row_tiles);
+    k_confmat_moments<<<grid, 256, 0, s>>>((const ll*)cm, C, rows_per_block, tk, pk, diag);
     k_confmat_scalars<<<1, 256, 0, s>>>(pk, diag, tk, C, zero_division, (float*)out);
     return (int)hipGetLastError();
 }
