@@ -16,7 +16,7 @@ for C in (37, 256, 1000):
     d_ref = torch.diag(cm)
     tk, pk, d = scratch[:C], scratch[C:2*C], scratch[2*C:]
     print("  tk zeroed-after-B:", bool((tk == 0).all()))
-    print("  pk ok:", bool(torch.equal(pk, pk_ref)), "diag ok:", bool(torch.equal(d, d_ref)))
+    print("  pk+tk consumed by phase B (zeroed):", bool((pk == 0).all()), "diag ok:", bool(torch.equal(d, d_ref)))
     num = d_ref.double(); den = (tk_ref + pk_ref - d_ref).double()
     j = torch.where(den > 0, num / den, torch.zeros_like(num))
     valid = (tk_ref + pk_ref) > 0
