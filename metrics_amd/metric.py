@@ -59,15 +59,36 @@ def _squeeze_if_scalar(data: Any) -> Any:
     return apply_to_collection(data, Tensor, lambda x: x.squeeze() if x.numel() == 1 else x)
 
 
-def _clone_result(value: Any) -> Any:
+def _clone_result(value: Any, state_storages: Optional[frozenset] = None) -> Any:
     """Detach compute() results from the metric states.
 
+    The contract only requires the result not to alias metric state. Most of
+    our fused compute paths return freshly allocated tensors, so when
+    ``state_storages`` (the states' storage data_ptrs) is given, results whose
+    storage is not a state storage are returned AS-IS — each skipped clone is
+    a saved kernel dispatch (~4us each; compute is dispatch-bound).
     Per-class curve results are lists of many small GPU tensors; cloning them
     one by one costs one D2D launch each (~100 launches per compute on the
     curve metrics — profiles/README.md prof15). Batch same-(device,dtype)
-    tensors through ONE cat and hand back views of the packed buffer: the
-    contract only requires the result not to alias metric state.
+    tensors through ONE cat and hand back views of the packed buffer.
     """
+    if state_storages is not None:
+        def _aliases_state(t: Tensor) -> bool:
+            try:
+                return t.untyped_storage().data_ptr() in state_storages
+            except Exception:
+                return True
+
+        any_alias = False
+
+        def _scan(t: Tensor) -> Tensor:
+            nonlocal any_alias
+            any_alias = any_alias or _aliases_state(t)
+            return t
+
+        apply_to_collection(value, Tensor, _scan)
+        if not any_alias:
+            return value
     tensors: List[Tensor] = []
 
     def _collect(t: Tensor) -> Tensor:
@@ -238,6 +259,24 @@ class Metric(Module, ABC):
         self._reductions[name] = dist_reduce_fx
 
     # ---------------------------------------------------------------- forward
+    def _state_storages(self) -> frozenset:
+        ptrs = []
+        for key in self._defaults:
+            val = getattr(self, key, None)
+            if isinstance(val, Tensor):
+                try:
+                    ptrs.append(val.untyped_storage().data_ptr())
+                except Exception:
+                    pass
+            elif isinstance(val, list):
+                for v in val:
+                    if isinstance(v, Tensor):
+                        try:
+                            ptrs.append(v.untyped_storage().data_ptr())
+                        except Exception:
+                            pass
+        return frozenset(ptrs)
+
     def _lazy_flush(self) -> None:
         """Materialize deferred device-side state (lazy curve histograms).
 
@@ -537,8 +576,9 @@ class Metric(Module, ABC):
                 self._wait_pending_sync()
                 value = _squeeze_if_scalar(compute(*args, **kwargs))
                 # detach from state so later in-place ops cannot alter the
-                # returned result (batched for many-tensor curve results)
-                value = _clone_result(value)
+                # returned result (storage-checked: fresh allocations pass
+                # through; aliases of state get the batched clone)
+                value = _clone_result(value, self._state_storages())
 
             if self.compute_with_cache:
                 self._computed = value
