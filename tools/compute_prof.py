@@ -38,6 +38,6 @@ for _ in range(n):
     bust()
     coll.compute()
 pr.disable()
-st = pstats.Stats(pr); st.sort_stats("tottime")
-buf = io.StringIO(); st.stream = buf; st.print_stats(18)
+st = pstats.Stats(pr); st.sort_stats("cumulative")
+buf = io.StringIO(); st.stream = buf; st.print_stats(34)
 print("\n".join(buf.getvalue().splitlines()[4:30]))
