@@ -72,9 +72,16 @@ def main():
 
     coll = build_collection(args.classes, device)
 
+    # same effective-cadence rule as bench.py: the timed region always
+    # contains ~4 compute() calls regardless of the chosen step count, so the
+    # reference and metrics_amd numbers stay apples-to-apples at any --steps
+    compute_every = args.compute_every
+    if compute_every:
+        compute_every = min(compute_every, max(1, args.steps // 4))
+
     def one_step(i):
         coll.update(preds[i % n_unique], target[i % n_unique])
-        if args.compute_every and (i + 1) % args.compute_every == 0:
+        if compute_every and (i + 1) % compute_every == 0:
             coll.compute()
 
     for i in range(args.warmup):
@@ -95,7 +102,7 @@ def main():
         "metric": "metric_updates_per_sec",
         "value": 16 * args.steps / elapsed,
         "ms_per_step": 1000 * elapsed / args.steps,
-        "config": vars(args),
+        "config": {**vars(args), "compute_every": compute_every},
     }))
 
 
