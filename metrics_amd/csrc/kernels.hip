@@ -1134,7 +1134,7 @@ __global__ void k_apply_stat_exact(
 // ONE kernel replaces the ~15-launch torch chain per metric at compute time.
 // avg_mode: 0 macro, 1 weighted, 2 micro. Matches _adjust_weights_safe_divide
 // bit-for-bit: per-class (w*s)/W summed (not sum(w*s)/W).
-__global__ void k_linear_stat_compute(
+__device__ void _linear_stat_eval(
     const ll* __restrict__ tp, const ll* __restrict__ fp, const ll* __restrict__ tn,
     const ll* __restrict__ fn, ll C, float n0, float n1, float n2, float n3, float d0, float d1,
     float d2, float d3, int avg_mode, int zero_w_topk, float zero_division, float post_a,
@@ -1205,6 +1205,28 @@ __global__ void k_linear_stat_compute(
     } else if (threadIdx.x == 0) {
         out[0] = (float)red[0];
     }
+}
+
+__global__ void k_linear_stat_compute(
+    const ll* __restrict__ tp, const ll* __restrict__ fp, const ll* __restrict__ tn,
+    const ll* __restrict__ fn, ll C, float n0, float n1, float n2, float n3, float d0, float d1,
+    float d2, float d3, int avg_mode, int zero_w_topk, float zero_division, float post_a,
+    float post_b, float* __restrict__ out) {
+    _linear_stat_eval(tp, fp, tn, fn, C, n0, n1, n2, n3, d0, d1, d2, d3, avg_mode, zero_w_topk,
+                      zero_division, post_a, post_b, out);
+}
+
+// batched variant: one block per formula over the SAME shared (C,) stat
+// states — the 9 linear stat metrics of a compute group pay ONE dispatch per
+// compute generation instead of nine. params: 13 floats per formula
+// (n0-3, d0-3, avg_mode, zero_w_topk, zero_division, post_a, post_b).
+__global__ void k_linear_stat_multi(
+    const ll* __restrict__ tp, const ll* __restrict__ fp, const ll* __restrict__ tn,
+    const ll* __restrict__ fn, ll C, const float* __restrict__ params,
+    float* __restrict__ outs) {
+    const float* P = params + (ll)blockIdx.x * 13;
+    _linear_stat_eval(tp, fp, tn, fn, C, P[0], P[1], P[2], P[3], P[4], P[5], P[6], P[7],
+                      (int)P[8], (int)P[9], P[10], P[11], P[12], outs + blockIdx.x);
 }
 
 // per-class AUROC / AveragePrecision straight from the thresholded curve
@@ -1651,6 +1673,15 @@ int ma_curve_suffix(uintptr_t stream, uintptr_t hist, ll outer, int T, int trans
 // close a curve update's device epoch WITHOUT the suffix pass: used by the
 // lazy-confmat path, where histograms accumulate across updates and the
 // suffix/confmat materialization is deferred to compute()/state access.
+int ma_linear_stat_multi(uintptr_t stream, uintptr_t tp, uintptr_t fp, uintptr_t tn,
+                         uintptr_t fn, ll C, int n_formulas, uintptr_t params, uintptr_t outs) {
+    hipStream_t s = (hipStream_t)stream;
+    k_linear_stat_multi<<<n_formulas, 256, 0, s>>>((const ll*)tp, (const ll*)fp, (const ll*)tn,
+                                                   (const ll*)fn, C, (const float*)params,
+                                                   (float*)outs);
+    return (int)hipGetLastError();
+}
+
 int ma_confmat_scalars(uintptr_t stream, uintptr_t cm, ll C, uintptr_t scratch /* 3C u64 */,
                        float zero_division, uintptr_t out /* 3 f32 */) {
     hipStream_t s = (hipStream_t)stream;

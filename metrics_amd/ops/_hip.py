@@ -56,6 +56,7 @@ _SIGNATURES = {
     "ma_curve_suffix": [_U64, _U64, _LL, _I, _I, _I, _U64, _U64],
     "ma_curve_epoch_bump": [_U64, _U64],
     "ma_confmat_scalars": [_U64, _U64, _LL, _U64, ctypes.c_float, _U64],
+    "ma_linear_stat_multi": [_U64, _U64, _U64, _U64, _U64, _LL, _I, _U64, _U64],
     "ma_apply_stat_deltas": [_U64, _U64, _LL, _U64, _U64, _U64, _U64, _U64],
     "ma_exact_apply": [_U64, _U64, _LL, _LL, _I, _U64, _U64],
     "ma_apply_stat_exact": [_U64, _U64, _LL, _LL, _U64, _U64, _U64, _U64, _U64, _U64, _U64],
@@ -1129,6 +1130,7 @@ def mc_stat_into(
         0,
     )
     _check(rc, "ma_apply_stat_deltas")
+    _mark_kernel_mutated(tp)
 
 
 def mc_confmat_into(
@@ -1253,6 +1255,13 @@ def mc_exact_into(
     _check(rc, "ma_exact_apply")
 
 
+# Batched linear-stat plans: the stat metrics of a compute group share one
+# (tp,fp,tn,fn) state; each generation the FIRST caller batch-evaluates every
+# formula seen so far in ONE kernel and later callers hit the cache. Keyed by
+# id(tp) with weakref eviction; generation = (tp._version, kernel mutations).
+_LINEAR_PLANS: dict = {}
+
+
 def linear_stat_compute(
     tp: Tensor, fp: Tensor, tn: Tensor, fn: Tensor,
     num_coefs, den_coefs, average: str, zero_w_topk: bool = False,
@@ -1263,26 +1272,49 @@ def linear_stat_compute(
     score_c = post(safe_div(num_coefs . stats_c, den_coefs . stats_c)), then
     micro / macro / weighted averaging exactly like
     ``_adjust_weights_safe_divide``. Returns a 0-dim float32 tensor.
+    Formulas over the same state tensor are batched per compute generation:
+    one grid launch evaluates all of them (see ``k_linear_stat_multi``).
     """
     lib = _lib()
     C = tp.numel()
-    out = torch.empty((), dtype=torch.float32, device=tp.device)
     avg_mode = {"macro": 0, "weighted": 1, "micro": 2}[average]
-    rc = lib.ma_linear_stat_compute(
-        _stream(),
-        tp.data_ptr(), fp.data_ptr(), tn.data_ptr(), fn.data_ptr(),
-        C,
+    key = (
         float(num_coefs[0]), float(num_coefs[1]), float(num_coefs[2]), float(num_coefs[3]),
         float(den_coefs[0]), float(den_coefs[1]), float(den_coefs[2]), float(den_coefs[3]),
-        avg_mode,
-        1 if zero_w_topk else 0,
-        float(zero_division),
-        float(post_a),
-        float(post_b),
-        out.data_ptr(),
+        float(avg_mode), float(1 if zero_w_topk else 0),
+        float(zero_division), float(post_a), float(post_b),
     )
-    _check(rc, "ma_linear_stat_compute")
-    return out
+    muts = _tensor_cache_get(_CONFMAT_MUTATIONS, tp)
+    gen = (tp._version, muts[0] if muts is not None else 0)
+    plan = _tensor_cache_get(_LINEAR_PLANS, tp)
+    if plan is None:
+        # formulas: key -> slot; params_dev rebuilt when the set grows
+        plan = {"formulas": {}, "params_dev": None, "gen": None, "outs": None}
+        _tensor_cache_put(_LINEAR_PLANS, tp, plan)
+    slot = plan["formulas"].get(key)
+    if slot is None:
+        slot = len(plan["formulas"])
+        plan["formulas"][key] = slot
+        plan["params_dev"] = None  # set changed: re-upload next batch eval
+        plan["gen"] = None
+    if plan["gen"] == gen and plan["outs"] is not None:
+        return plan["outs"][slot].reshape(())
+    n = len(plan["formulas"])
+    if plan["params_dev"] is None:
+        flat = [0.0] * (13 * n)
+        for k2, s2 in plan["formulas"].items():
+            flat[13 * s2 : 13 * s2 + 13] = list(k2)
+        plan["params_dev"] = torch.tensor(flat, dtype=torch.float32, device=tp.device)
+    # fresh outs per generation: returned views must never be recycled
+    outs = torch.empty(n, dtype=torch.float32, device=tp.device)
+    rc = lib.ma_linear_stat_multi(
+        _stream(), tp.data_ptr(), fp.data_ptr(), tn.data_ptr(), fn.data_ptr(),
+        C, n, plan["params_dev"].data_ptr(), outs.data_ptr(),
+    )
+    _check(rc, "ma_linear_stat_multi")
+    plan["gen"] = gen
+    plan["outs"] = outs
+    return outs[slot].reshape(())
 
 
 def curve_auc_from_confmat(confmat_state: Tensor, mode: int) -> Tuple[Tensor, Tensor]:
@@ -1372,6 +1404,7 @@ def mc_fused_collection_update(
 
     if confmat is not None:
         _mark_kernel_mutated(confmat)
+    _mark_kernel_mutated(tp)  # generation key for the batched linear-stat plan
     if defer_apply:
         return _apply_pass
     _apply_pass()
