@@ -258,3 +258,39 @@ def test_gloo_cuda_world2():
     from tests.unittests._helpers import run_distributed
 
     run_distributed(_run_all, world_size=2, backend="gloo_cuda")
+
+
+def test_rccl_world1_lazy_curve_sync():
+    """sync() on a lazily-accumulated curve metric must materialize the
+    confmat BEFORE the collective reads it (world-1 RCCL exercises the real
+    sync path end to end)."""
+    import os
+
+    import torch.distributed as dist
+
+    import metrics_amd as ma
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29771")
+    os.environ["RANK"] = "0"
+    os.environ["WORLD_SIZE"] = "1"
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        m = ma.MulticlassAUROC(num_classes=7, thresholds=30).to("cuda")
+        p = torch.randn(256, 7, device="cuda").softmax(-1)
+        t = torch.randint(0, 7, (256,), device="cuda")
+        m.update(p, t)
+        assert m.__dict__.get("_lazy_dirty") is True
+        m.sync()
+        assert m.__dict__.get("_lazy_dirty") is False
+        assert int(m.confmat.sum().item()) > 0
+        m.unsync()
+        out = m.compute()
+        ref = ma.MulticlassAUROC(num_classes=7, thresholds=30)
+        ref.sync_on_compute = False
+        ref.update(p.cpu().float(), t.cpu())
+        assert torch.allclose(out.cpu(), ref.compute(), atol=1e-5)
+    finally:
+        dist.destroy_process_group()
