@@ -926,3 +926,53 @@ def test_collection_forward_differential():
         vo, vr = ours.compute(), ref.compute()
         for k in vr:
             _cmp(vo[k], vr[k], 1e-5)
+
+
+def test_rouge_perplexity_accumulation():
+    """ROUGE (dict-valued) and Perplexity accumulation parity."""
+    import warnings
+
+    tm = _tm()
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        # rougeLsum needs nltk in the reference (not installable offline)
+        keys = ("rouge1", "rouge2", "rougeL")
+        ours = ma.text.ROUGEScore(rouge_keys=keys)
+        ref = tm.text.ROUGEScore(rouge_keys=keys)
+        for b in range(3):
+            p, t = _texts_flat(80 + b)
+            ours.update(p, t)
+            ref.update(p, t)
+        vo, vr = ours.compute(), ref.compute()
+        assert set(vo) == set(vr)
+        for k in vr:
+            _cmp(vo[k], vr[k], 1e-5)
+
+        op = ma.text.Perplexity(ignore_index=-100)
+        rp = tm.text.Perplexity(ignore_index=-100)
+        for b in range(3):
+            g = torch.Generator().manual_seed(80 + b)
+            logits = torch.randn(2, 8, 20, generator=g)
+            target = torch.randint(0, 20, (2, 8), generator=g)
+            target[0, :2] = -100
+            op.update(logits, target)
+            rp.update(logits, target)
+        _cmp(op.compute(), rp.compute(), 1e-4)
+
+
+def test_rouge_forward_differential():
+    import warnings
+
+    tm = _tm()
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours = ma.text.ROUGEScore(use_stemmer=False, rouge_keys=("rouge1", "rouge2", "rougeL"))
+        ref = tm.text.ROUGEScore(use_stemmer=False, rouge_keys=("rouge1", "rouge2", "rougeL"))
+        for b in range(3):
+            p, t = _texts_flat(85 + b)
+            vo, vr = ours(p, t), ref(p, t)
+            for k in vr:
+                _cmp(vo[k], vr[k], 1e-5)
+        vo, vr = ours.compute(), ref.compute()
+        for k in vr:
+            _cmp(vo[k], vr[k], 1e-5)
