@@ -1355,3 +1355,21 @@ def test_confmat_scalars_in_collection():
         out, exp = coll.compute(), ref.compute()
         for k in exp:
             assert torch.allclose(out[k].float().cpu(), exp[k].float(), atol=1e-5), k
+
+
+@pytest.mark.parametrize("average", ["macro", "weighted", "none"])
+@pytest.mark.parametrize("cls_name", ["MulticlassAUROC", "MulticlassAveragePrecision"])
+def test_curve_auc_average_variants_gpu(cls_name, average):
+    """All fused-path average modes of the curve-area metrics match CPU."""
+    import metrics_amd as ma
+
+    torch.manual_seed(31)
+    cls = getattr(ma, cls_name)
+    g = cls(num_classes=13, thresholds=60, average=average).to("cuda")
+    c = cls(num_classes=13, thresholds=60, average=average)
+    for _ in range(2):
+        p = torch.randn(300, 13, device="cuda").softmax(-1)
+        t = torch.randint(0, 13, (300,), device="cuda")
+        g.update(p, t)
+        c.update(p.cpu(), t.cpu())
+    assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-5)
