@@ -976,3 +976,52 @@ def test_rouge_forward_differential():
         vo, vr = ours.compute(), ref.compute()
         for k in vr:
             _cmp(vo[k], vr[k], 1e-5)
+
+
+def test_aggregation_differential():
+    """SumMetric/MeanMetric/MaxMetric/MinMetric/CatMetric + Running wrappers
+    under each nan strategy, with scalar / tensor / weighted updates."""
+    import warnings
+
+    tm = _tm()
+    g = torch.Generator().manual_seed(5)
+    batches = [torch.randn(6, generator=g) for _ in range(3)]
+    batches[1][2] = float("nan")
+    weights = [torch.rand(6, generator=g) + 0.1 for _ in range(3)]
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        for name in ("SumMetric", "MaxMetric", "MinMetric", "CatMetric"):
+            for nan_strategy in ("warn", "ignore", 0.5):
+                if name in ("MaxMetric", "MinMetric") and nan_strategy == "ignore":
+                    pass
+                ours = getattr(ma, name)(nan_strategy=nan_strategy)
+                ref = getattr(tm, name)(nan_strategy=nan_strategy)
+                for b in batches:
+                    ours.update(b)
+                    ref.update(b)
+                _cmp(ours.compute(), ref.compute(), 1e-6)
+
+        for nan_strategy in ("warn", "ignore", 0.25):
+            ours = ma.MeanMetric(nan_strategy=nan_strategy)
+            ref = tm.MeanMetric(nan_strategy=nan_strategy)
+            for b, w in zip(batches, weights):
+                ours.update(b, w)
+                ref.update(b, w)
+            _cmp(ours.compute(), ref.compute(), 1e-6)
+
+        # scalar updates + default weight path
+        ours = ma.MeanMetric()
+        ref = tm.MeanMetric()
+        for v in (1.5, 2.0, -0.5):
+            ours.update(v)
+            ref.update(v)
+        _cmp(ours.compute(), ref.compute(), 1e-6)
+
+        for cls_name in ("RunningMean", "RunningSum"):
+            ours = getattr(ma, cls_name)(window=2)
+            ref = getattr(tm, cls_name)(window=2)
+            for b in batches:
+                vo, vr = ours(b.nan_to_num()), ref(b.nan_to_num())
+                _cmp(vo, vr, 1e-6)
+            _cmp(ours.compute(), ref.compute(), 1e-6)
