@@ -1025,3 +1025,32 @@ def test_aggregation_differential():
                 vo, vr = ours(b.nan_to_num()), ref(b.nan_to_num())
                 _cmp(vo, vr, 1e-6)
             _cmp(ours.compute(), ref.compute(), 1e-6)
+
+
+_DTYPE_CASES = [
+    ("MeanSquaredError", {}, _reg2),
+    ("PearsonCorrCoef", {}, _reg2),
+    ("PeakSignalNoiseRatio", {}, _img),
+    ("StructuralSimilarityIndexMeasure", {"data_range": 1.0}, _img),
+    ("UniversalImageQualityIndex", {}, _img),
+    ("ScaleInvariantSignalNoiseRatio", {}, _aud),
+    ("R2Score", {}, _reg2),
+]
+
+
+@pytest.mark.parametrize(("name", "kwargs", "gen"), _DTYPE_CASES, ids=[c[0] for c in _DTYPE_CASES])
+@pytest.mark.parametrize("dtype", [torch.float64, torch.float16])
+def test_modular_dtype_parity(name, kwargs, gen, dtype):
+    """fp64/fp16 inputs accumulate to the same value as the reference."""
+    import warnings
+
+    tm = _tm()
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours, ref = getattr(ma, name)(**kwargs), getattr(tm, name)(**kwargs)
+        for b in range(2):
+            args = tuple(x.to(dtype) for x in gen(95 + b))
+            ours.update(*args)
+            ref.update(*args)
+        atol = 1e-6 if dtype == torch.float64 else 1e-2
+        _cmp(ours.compute().float(), ref.compute().float(), atol)
