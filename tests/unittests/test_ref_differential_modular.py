@@ -1054,3 +1054,49 @@ def test_modular_dtype_parity(name, kwargs, gen, dtype):
             ref.update(*args)
         atol = 1e-6 if dtype == torch.float64 else 1e-2
         _cmp(ours.compute().float(), ref.compute().float(), atol)
+
+
+def _emb(seed):
+    g = torch.Generator().manual_seed(seed)
+    return torch.randn(60, 4, generator=g), torch.randint(0, 3, (60,), generator=g)
+
+
+_ACC_NS4 = [
+    ("classification", "BinaryPrecisionAtFixedRecall", {"min_recall": 0.5}, _bin_pt, 1e-5),
+    ("classification", "BinaryRecallAtFixedPrecision", {"min_precision": 0.5}, _bin_pt, 1e-5),
+    ("classification", "BinarySpecificityAtSensitivity", {"min_sensitivity": 0.5}, _bin_pt, 1e-5),
+    ("classification", "BinarySensitivityAtSpecificity", {"min_specificity": 0.5}, _bin_pt, 1e-5),
+    ("classification", "BinaryLogAUC", {}, _bin_pt, 1e-4),
+    ("classification", "MulticlassPrecisionAtFixedRecall", {"num_classes": 5, "min_recall": 0.5}, _mc_pt, 1e-5),
+    ("classification", "MulticlassRecallAtFixedPrecision", {"num_classes": 5, "min_precision": 0.5}, _mc_pt, 1e-5),
+    ("clustering", "CalinskiHarabaszScore", {}, _emb, 1e-4),
+    ("clustering", "DaviesBouldinScore", {}, _emb, 1e-4),
+    ("clustering", "DunnIndex", {}, _emb, 1e-4),
+    ("clustering", "RandScore", {}, _lab, 1e-5),
+    ("clustering", "AdjustedMutualInfoScore", {}, _lab, 1e-5),
+]
+
+
+@pytest.mark.parametrize(
+    ("ns", "name", "kwargs", "gen", "atol"), _ACC_NS4, ids=[f"{c[1]}_{i}" for i, c in enumerate(_ACC_NS4)]
+)
+def test_modular_accumulation_ns4(ns, name, kwargs, gen, atol):
+    import importlib
+    import warnings
+
+    _tm()
+    our_ns = importlib.import_module(f"metrics_amd.{ns}")
+    ref_ns = importlib.import_module(f"torchmetrics.{ns}")
+    our_cls = getattr(our_ns, name, None)
+    ref_cls = getattr(ref_ns, name, None)
+    if ref_cls is None:
+        pytest.skip(f"{ns}.{name} absent in reference")
+    assert our_cls is not None, f"{ns}.{name} missing in metrics_amd"
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours, ref = our_cls(**kwargs), ref_cls(**kwargs)
+        for b in range(3):
+            args = gen(90 + b)
+            ours.update(*args)
+            ref.update(*args)
+        _cmp(ours.compute(), ref.compute(), atol)
