@@ -1142,3 +1142,36 @@ def test_modular_accumulation_ns5(ns, name, kwargs, gen, atol):
             ours.update(*args)
             ref.update(*args)
         _cmp(ours.compute(), ref.compute(), atol)
+
+
+def test_compositional_differential():
+    """Metric arithmetic (CompositionalMetric) parity with the reference."""
+    import warnings
+
+    tm = _tm()
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+
+        def make(mod):
+            a = mod.classification.BinaryPrecision()
+            b = mod.classification.BinaryRecall()
+            return a, b
+
+        for expr in (
+            lambda a, b: a + b,
+            lambda a, b: 2 * a * b / (a + b + 1e-8),  # F1 from parts
+            lambda a, b: a - b,
+            lambda a, b: a**2,
+            lambda a, b: abs(a - b),
+            lambda a, b: a == b,
+        ):
+            oa, ob = make(ma)
+            ra, rb = make(tm)
+            oc, rc = expr(oa, ob), expr(ra, rb)
+            for s in range(3):
+                g = torch.Generator().manual_seed(30 + s)
+                p = torch.rand(80, generator=g)
+                t = torch.randint(0, 2, (80,), generator=g)
+                oc.update(p, t)
+                rc.update(p, t)
+            _cmp(oc.compute().float(), rc.compute().float(), 1e-6)
