@@ -1175,3 +1175,27 @@ def test_compositional_differential():
                 oc.update(p, t)
                 rc.update(p, t)
             _cmp(oc.compute().float(), rc.compute().float(), 1e-6)
+
+
+def test_tracker_multi_epoch_differential():
+    import warnings
+
+    tm = _tm()
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours = ma.MetricTracker(ma.MulticlassAccuracy(num_classes=4), maximize=True)
+        ref = tm.MetricTracker(tm.classification.MulticlassAccuracy(num_classes=4), maximize=True)
+        for epoch in range(3):
+            ours.increment()
+            ref.increment()
+            for b in range(2):
+                g = torch.Generator().manual_seed(epoch * 10 + b)
+                p = torch.randn(50, 4, generator=g).softmax(-1)
+                t = torch.randint(0, 4, (50,), generator=g)
+                ours.update(p, t)
+                ref.update(p, t)
+        _cmp(torch.stack(ours.compute_all() if isinstance(ours.compute_all(), list) else [ours.compute_all()]).flatten(),
+             torch.stack(ref.compute_all() if isinstance(ref.compute_all(), list) else [ref.compute_all()]).flatten(), 1e-6)
+        ob, oi = ours.best_metric(return_step=True)
+        rb, ri = ref.best_metric(return_step=True)
+        assert abs(float(ob) - float(rb)) < 1e-6 and int(oi) == int(ri)
