@@ -47,7 +47,7 @@ def _run_pair(our_cls, kwargs, gen, atol):
         flat_b = b_ if isinstance(b_, torch.Tensor) else torch.cat([x.flatten().float().cpu() for x in b_])
         if isinstance(flat_b, torch.Tensor):
             flat_b = flat_b.cpu()
-        assert torch.allclose(flat_a.float(), flat_b.float(), atol=max(atol, 1e-4), rtol=1e-3), (a, b_)
+        assert torch.allclose(flat_a.float(), flat_b.float(), atol=max(atol, 1e-4), rtol=1e-3, equal_nan=True), (a, b_)
 
 
 @pytest.mark.parametrize(
