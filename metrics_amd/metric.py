@@ -391,6 +391,9 @@ class Metric(Module, ABC):
 
     def merge_state(self, incoming_state: Union[Dict[str, Any], "Metric"]) -> None:
         """Merge an external metric state (another instance or a state dict) into this one."""
+        self._maybe_flush_lazy()
+        if isinstance(incoming_state, Metric):
+            incoming_state._maybe_flush_lazy()
         if not isinstance(incoming_state, (dict, Metric)):
             raise ValueError(
                 f"Expected incoming state to be a dict or an instance of Metric but got {type(incoming_state)}"
