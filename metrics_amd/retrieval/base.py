@@ -20,6 +20,22 @@ from metrics_amd.utilities.checks import _check_retrieval_inputs
 from metrics_amd.utilities.data import _flexible_bincount, dim_zero_cat
 
 
+def _retrieval_aggregate(values: Tensor, aggregation="mean", dim: Optional[int] = None) -> Tensor:
+    """Aggregate per-query retrieval scores: mean/median/min/max or a callable.
+
+    Parity: reference retrieval/base.py:26 (_retrieval_aggregate).
+    """
+    if aggregation == "mean":
+        return values.mean() if dim is None else values.mean(dim=dim)
+    if aggregation == "median":
+        return values.median() if dim is None else values.median(dim=dim).values
+    if aggregation == "min":
+        return values.min() if dim is None else values.min(dim=dim).values
+    if aggregation == "max":
+        return values.max() if dim is None else values.max(dim=dim).values
+    return aggregation(values, dim=dim)
+
+
 class _Grouped:
     """All-queries grouping for the batched retrieval path.
 
@@ -95,10 +111,17 @@ class RetrievalMetric(Metric, ABC):
         self,
         empty_target_action: str = "neg",
         ignore_index: Optional[int] = None,
+        aggregation="mean",
         **kwargs: Any,
     ) -> None:
         super().__init__(**kwargs)
         self.allow_non_binary_target = False
+        if not (aggregation in ("mean", "median", "min", "max") or callable(aggregation)):
+            raise ValueError(
+                "Argument `aggregation` must be one of `mean`, `median`, `min`, `max` or a custom callable function"
+                f"which takes tensor of values, but got {aggregation}."
+            )
+        self.aggregation = aggregation
 
         empty_target_action_options = ("error", "skip", "neg", "pos")
         if empty_target_action not in empty_target_action_options:
@@ -152,7 +175,9 @@ class RetrievalMetric(Metric, ABC):
                 else:
                     fill = 1.0 if self.empty_target_action == "pos" else 0.0
                     scores = torch.where(empty, torch.tensor(fill, device=scores.device), scores)
-            return scores.mean().to(preds.dtype) if scores.numel() else torch.tensor(0.0).to(preds)
+            if not scores.numel():
+                return torch.tensor(0.0).to(preds)
+            return _retrieval_aggregate(scores, self.aggregation).to(preds.dtype)
 
         # generic fallback: per-query loop over sorted slices
         split_sizes = grouped.counts.detach().cpu().tolist()
@@ -173,7 +198,9 @@ class RetrievalMetric(Metric, ABC):
             else:
                 res.append(self._metric(mini_preds, mini_target))
 
-        return torch.stack([x.to(preds) for x in res]).mean() if res else torch.tensor(0.0).to(preds)
+        if not res:
+            return torch.tensor(0.0).to(preds)
+        return _retrieval_aggregate(torch.stack([x.to(preds) for x in res]), self.aggregation)
 
     def _batched_scores(self, grouped: "_Grouped") -> Optional[Tensor]:
         """Vectorized per-query scores (G,), or None to use the per-query loop."""

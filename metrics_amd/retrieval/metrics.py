@@ -187,8 +187,10 @@ class RetrievalPrecisionRecallCurve(RetrievalMetric):
     """Averaged precision/recall at k = 1..max_k."""
 
     def __init__(self, max_k: Optional[int] = None, adaptive_k: bool = False,
-                 empty_target_action: str = "neg", ignore_index: Optional[int] = None, **kwargs: Any) -> None:
-        super().__init__(empty_target_action=empty_target_action, ignore_index=ignore_index, **kwargs)
+                 empty_target_action: str = "neg", ignore_index: Optional[int] = None,
+                 aggregation="mean", **kwargs: Any) -> None:
+        super().__init__(empty_target_action=empty_target_action, ignore_index=ignore_index,
+                         aggregation=aggregation, **kwargs)
         if max_k is not None and not (isinstance(max_k, int) and max_k > 0):
             raise ValueError("`max_k` has to be a positive integer or None")
         self.max_k = max_k
@@ -229,7 +231,13 @@ class RetrievalPrecisionRecallCurve(RetrievalMetric):
         top_k = torch.arange(1, max_k + 1, device=preds.device)
         if not precisions:
             return torch.zeros(max_k, device=preds.device), torch.zeros(max_k, device=preds.device), top_k
-        return torch.stack(precisions).mean(0), torch.stack(recalls).mean(0), top_k
+        from metrics_amd.retrieval.base import _retrieval_aggregate
+
+        return (
+            _retrieval_aggregate(torch.stack(precisions), self.aggregation, dim=0),
+            _retrieval_aggregate(torch.stack(recalls), self.aggregation, dim=0),
+            top_k,
+        )
 
 
 class RetrievalRecallAtFixedPrecision(RetrievalMetric):

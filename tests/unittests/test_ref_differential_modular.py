@@ -1199,3 +1199,23 @@ def test_tracker_multi_epoch_differential():
         ob, oi = ours.best_metric(return_step=True)
         rb, ri = ref.best_metric(return_step=True)
         assert abs(float(ob) - float(rb)) < 1e-6 and int(oi) == int(ri)
+
+
+@pytest.mark.parametrize("aggregation", ["mean", "median", "min", "max"])
+def test_retrieval_aggregation_differential(aggregation):
+    import warnings
+
+    tm = _tm()
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        for name in ("RetrievalMAP", "RetrievalMRR", "RetrievalPrecision"):
+            kw = {"aggregation": aggregation}
+            if name == "RetrievalPrecision":
+                kw["top_k"] = 3
+            ours = getattr(ma.retrieval, name)(**kw)
+            ref = getattr(tm.retrieval, name)(**kw)
+            for b in range(2):
+                p, t, idx = _ret_pt(60 + b)
+                ours.update(p, t, indexes=idx)
+                ref.update(p, t, indexes=idx)
+            _cmp(ours.compute(), ref.compute(), 1e-6)
