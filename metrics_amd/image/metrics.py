@@ -147,12 +147,18 @@ class _ScoreAverageMetric(Metric):
         return self._plot(val, ax)
 
 
-class StructuralSimilarityIndexMeasure(_ScoreAverageMetric):
-    """SSIM (stateful)."""
+class StructuralSimilarityIndexMeasure(Metric):
+    """SSIM (stateful).
 
-    _SCORE_STATE = "similarity"
-    _TOTAL_FLOAT = True
+    Full reference semantics (image/ssim.py:99): reduction
+    elementwise_mean/sum keep scalar sum states, none/None keeps a cat list;
+    ``return_full_image`` / ``return_contrast_sensitivity`` accumulate the
+    per-image map / contrast term in a cat state returned by compute.
+    """
 
+    is_differentiable = True
+    higher_is_better = True
+    full_state_update = False
     plot_lower_bound: float = 0.0
     plot_upper_bound: float = 1.0
 
@@ -165,9 +171,23 @@ class StructuralSimilarityIndexMeasure(_ScoreAverageMetric):
         data_range: Optional[Union[float, Tuple[float, float]]] = None,
         k1: float = 0.01,
         k2: float = 0.03,
+        return_full_image: bool = False,
+        return_contrast_sensitivity: bool = False,
         **kwargs: Any,
     ) -> None:
         super().__init__(**kwargs)
+        valid_reduction = ("elementwise_mean", "sum", "none", None)
+        if reduction not in valid_reduction:
+            raise ValueError(f"Argument `reduction` must be one of {valid_reduction}, but got {reduction}")
+        if reduction in ("elementwise_mean", "sum"):
+            self.add_state("similarity", default=torch.tensor(0.0), dist_reduce_fx="sum")
+        else:
+            self.add_state("similarity", default=[], dist_reduce_fx="cat")
+        self.add_state("total", default=torch.tensor(0.0), dist_reduce_fx="sum")
+        if return_contrast_sensitivity and return_full_image:
+            raise ValueError("Arguments `return_full_image` and `return_contrast_sensitivity` are mutually exclusive")
+        if return_contrast_sensitivity or return_full_image:
+            self.add_state("image_return", default=[], dist_reduce_fx="cat")
         self.gaussian_kernel = gaussian_kernel
         self.sigma = sigma
         self.kernel_size = kernel_size
@@ -175,14 +195,40 @@ class StructuralSimilarityIndexMeasure(_ScoreAverageMetric):
         self.data_range = data_range
         self.k1 = k1
         self.k2 = k2
+        self.return_full_image = return_full_image
+        self.return_contrast_sensitivity = return_contrast_sensitivity
 
     def update(self, preds: Tensor, target: Tensor) -> None:
-        """Accumulate per-image SSIM."""
-        sim = structural_similarity_index_measure(
+        """Accumulate per-image SSIM (and optionally the full map / cs term)."""
+        pack = structural_similarity_index_measure(
             preds, target, self.gaussian_kernel, self.sigma, self.kernel_size, None,
             self.data_range, self.k1, self.k2,
+            self.return_full_image, self.return_contrast_sensitivity,
         )
-        self._accumulate(sim, preds.shape[0])
+        if isinstance(pack, tuple):
+            sim, image = pack
+            self.image_return.append(image)
+        else:
+            sim = pack
+        if self.reduction in ("elementwise_mean", "sum"):
+            self.similarity = self.similarity + sim.sum()
+            self.total = self.total + preds.shape[0]
+        else:
+            self.similarity.append(sim)
+
+    def compute(self):
+        if self.reduction == "elementwise_mean":
+            similarity = self.similarity / self.total
+        elif self.reduction == "sum":
+            similarity = self.similarity
+        else:
+            similarity = dim_zero_cat(self.similarity)
+        if self.return_contrast_sensitivity or self.return_full_image:
+            return similarity, dim_zero_cat(self.image_return)
+        return similarity
+
+    def plot(self, val=None, ax=None):
+        return self._plot(val, ax)
 
 
 class MultiScaleStructuralSimilarityIndexMeasure(_ScoreAverageMetric):

@@ -1219,3 +1219,31 @@ def test_retrieval_aggregation_differential(aggregation):
                 ours.update(p, t, indexes=idx)
                 ref.update(p, t, indexes=idx)
             _cmp(ours.compute(), ref.compute(), 1e-6)
+
+
+@pytest.mark.parametrize("opt", ["sum", "none", "full_image", "contrast"])
+def test_ssim_option_differential(opt):
+    import warnings
+
+    tm = _tm()
+    kw = {"data_range": 1.0}
+    if opt in ("sum", "none"):
+        kw["reduction"] = opt
+    elif opt == "full_image":
+        kw["return_full_image"] = True
+    else:
+        kw["return_contrast_sensitivity"] = True
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        ours = ma.StructuralSimilarityIndexMeasure(**kw)
+        ref = tm.StructuralSimilarityIndexMeasure(**kw)
+        for b in range(2):
+            p, t = _img(88 + b)
+            ours.update(p, t)
+            ref.update(p, t)
+        vo, vr = ours.compute(), ref.compute()
+        if isinstance(vr, tuple):
+            for a, b_ in zip(vo, vr):
+                _cmp(a, b_, 1e-5)
+        else:
+            _cmp(vo, vr, 1e-5)
