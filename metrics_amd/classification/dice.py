@@ -29,10 +29,25 @@ class Dice(MulticlassStatScores):
         num_classes: Optional[int] = None,
         threshold: float = 0.5,
         average: Optional[str] = "micro",
+        mdmc_average: Optional[str] = "global",
         ignore_index: Optional[int] = None,
         top_k: Optional[int] = None,
+        multiclass: Optional[bool] = None,
         **kwargs: Any,
     ) -> None:
+        # legacy reference knobs (classification/dice.py:154): the default
+        # values reproduce the implemented behavior; the removed-in-reference
+        # samplewise/multiclass-coercion paths raise explicitly
+        if mdmc_average not in ("global", None):
+            raise NotImplementedError(
+                "Dice(mdmc_average='samplewise') belongs to the reference's deprecated legacy input"
+                " machinery (removed in its v1.7); use segmentation.DiceScore(aggregation_level='samplewise')."
+            )
+        if multiclass is not None:
+            raise NotImplementedError(
+                "Dice(multiclass=...) input coercion belongs to the reference's deprecated legacy"
+                " machinery (removed in its v1.7); pass explicit multiclass/binary shaped inputs instead."
+            )
         allowed_average = ("micro", "macro", "weighted", "samples", "none", None)
         if average not in allowed_average:
             raise ValueError(f"The `average` has to be one of {allowed_average}, got {average}.")

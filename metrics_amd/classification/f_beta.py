@@ -28,8 +28,10 @@ class BinaryFBetaScore(BinaryStatScores):
         multidim_average: str = "global",
         ignore_index: Optional[int] = None,
         validate_args: bool = True,
+        zero_division: float = 0,
         **kwargs: Any,
     ) -> None:
+        kwargs["zero_division"] = zero_division
         super().__init__(
             threshold=threshold,
             multidim_average=multidim_average,
@@ -72,8 +74,10 @@ class MulticlassFBetaScore(MulticlassStatScores):
         multidim_average: str = "global",
         ignore_index: Optional[int] = None,
         validate_args: bool = True,
+        zero_division: float = 0,
         **kwargs: Any,
     ) -> None:
+        kwargs["zero_division"] = zero_division
         super().__init__(
             num_classes=num_classes,
             top_k=top_k,
@@ -118,8 +122,10 @@ class MultilabelFBetaScore(MultilabelStatScores):
         multidim_average: str = "global",
         ignore_index: Optional[int] = None,
         validate_args: bool = True,
+        zero_division: float = 0,
         **kwargs: Any,
     ) -> None:
+        kwargs["zero_division"] = zero_division
         super().__init__(
             num_labels=num_labels,
             threshold=threshold,
@@ -154,8 +160,10 @@ class BinaryF1Score(BinaryFBetaScore):
         multidim_average: str = "global",
         ignore_index: Optional[int] = None,
         validate_args: bool = True,
+        zero_division: float = 0,
         **kwargs: Any,
     ) -> None:
+        kwargs["zero_division"] = zero_division
         super().__init__(
             beta=1.0,
             threshold=threshold,
@@ -177,8 +185,10 @@ class MulticlassF1Score(MulticlassFBetaScore):
         multidim_average: str = "global",
         ignore_index: Optional[int] = None,
         validate_args: bool = True,
+        zero_division: float = 0,
         **kwargs: Any,
     ) -> None:
+        kwargs["zero_division"] = zero_division
         super().__init__(
             beta=1.0,
             num_classes=num_classes,
@@ -202,8 +212,10 @@ class MultilabelF1Score(MultilabelFBetaScore):
         multidim_average: str = "global",
         ignore_index: Optional[int] = None,
         validate_args: bool = True,
+        zero_division: float = 0,
         **kwargs: Any,
     ) -> None:
+        kwargs["zero_division"] = zero_division
         super().__init__(
             beta=1.0,
             num_labels=num_labels,

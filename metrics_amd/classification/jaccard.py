@@ -41,9 +41,11 @@ class BinaryJaccardIndex(Metric):
         threshold: float = 0.5,
         ignore_index: Optional[int] = None,
         validate_args: bool = True,
+        zero_division: float = 0.0,
         **kwargs: Any,
     ) -> None:
         super().__init__(**kwargs)
+        self.zero_division = zero_division
         if validate_args:
             _binary_confusion_matrix_arg_validation(threshold, ignore_index, normalize=None)
         self.threshold = threshold
@@ -60,7 +62,7 @@ class BinaryJaccardIndex(Metric):
         self.confmat += _binary_confusion_matrix_update(preds, target)
 
     def compute(self) -> Tensor:
-        return _jaccard_index_reduce(self.confmat, average="binary")
+        return _jaccard_index_reduce(self.confmat, average="binary", zero_division=self.zero_division)
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)
@@ -84,9 +86,11 @@ class MulticlassJaccardIndex(Metric):
         average: Optional[str] = "macro",
         ignore_index: Optional[int] = None,
         validate_args: bool = True,
+        zero_division: float = 0.0,
         **kwargs: Any,
     ) -> None:
         super().__init__(**kwargs)
+        self.zero_division = zero_division
         if validate_args:
             _multiclass_confusion_matrix_arg_validation(num_classes, ignore_index, normalize=None)
         self.num_classes = num_classes
@@ -108,7 +112,7 @@ class MulticlassJaccardIndex(Metric):
         self.confmat += _multiclass_confusion_matrix_update(preds, target, self.num_classes)
 
     def compute(self) -> Tensor:
-        return _jaccard_index_reduce(self.confmat, self.average, ignore_index=self.ignore_index)
+        return _jaccard_index_reduce(self.confmat, self.average, ignore_index=self.ignore_index, zero_division=self.zero_division)
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)
@@ -132,9 +136,11 @@ class MultilabelJaccardIndex(Metric):
         average: Optional[str] = "macro",
         ignore_index: Optional[int] = None,
         validate_args: bool = True,
+        zero_division: float = 0.0,
         **kwargs: Any,
     ) -> None:
         super().__init__(**kwargs)
+        self.zero_division = zero_division
         if validate_args:
             _multilabel_confusion_matrix_arg_validation(num_labels, threshold, ignore_index, normalize=None)
         self.num_labels = num_labels
@@ -155,7 +161,7 @@ class MultilabelJaccardIndex(Metric):
         self.confmat += _multilabel_confusion_matrix_update(preds, target, self.num_labels)
 
     def compute(self) -> Tensor:
-        return _jaccard_index_reduce(self.confmat, self.average)
+        return _jaccard_index_reduce(self.confmat, self.average, zero_division=self.zero_division)
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)

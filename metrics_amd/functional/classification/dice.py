@@ -24,12 +24,24 @@ def dice(
     target: Tensor,
     zero_division: int = 0,
     average: Optional[str] = "micro",
+    mdmc_average: Optional[str] = "global",
     threshold: float = 0.5,
     top_k: Optional[int] = None,
     num_classes: Optional[int] = None,
+    multiclass: Optional[bool] = None,
     ignore_index: Optional[int] = None,
 ) -> Tensor:
     """Dice = 2*TP / (2*TP + FP + FN) with micro/macro/weighted/samples-free averaging."""
+    if mdmc_average not in ("global", None):
+        raise NotImplementedError(
+            "dice(mdmc_average='samplewise') belongs to the reference's deprecated legacy input machinery"
+            " (removed in its v1.7); use functional.segmentation.dice_score."
+        )
+    if multiclass is not None:
+        raise NotImplementedError(
+            "dice(multiclass=...) input coercion belongs to the reference's deprecated legacy machinery"
+            " (removed in its v1.7); pass explicitly shaped inputs instead."
+        )
     if average not in ("micro", "macro", "weighted", "none", None):
         raise ValueError(f"The `average` has to be one of 'micro'/'macro'/'weighted'/'none', got {average}.")
     if num_classes is None:

@@ -28,6 +28,7 @@ def _negative_predictive_value_reduce(
     average: Optional[str],
     multidim_average: str = "global",
     multilabel: bool = False,
+    zero_division: float = 0,
 ) -> Tensor:
     if (
         tp.is_cuda and tp.ndim == 1 and multidim_average == "global" and not multilabel
@@ -36,14 +37,16 @@ def _negative_predictive_value_reduce(
         from metrics_amd.ops import _hip
 
         if _hip.hip_available():
-            return _hip.linear_stat_compute(tp, fp, tn, fn, (0, 0, 1, 0), (0, 0, 1, 1), average)
+            return _hip.linear_stat_compute(
+                tp, fp, tn, fn, (0, 0, 1, 0), (0, 0, 1, 1), average, zero_division=zero_division
+            )
     if average == "binary":
-        return _safe_divide(tn, tn + fn)
+        return _safe_divide(tn, tn + fn, zero_division)
     if average == "micro":
         tn = tn.sum(dim=0 if multidim_average == "global" else 1)
         fn = fn.sum(dim=0 if multidim_average == "global" else 1)
-        return _safe_divide(tn, tn + fn)
-    score = _safe_divide(tn, tn + fn)
+        return _safe_divide(tn, tn + fn, zero_division)
+    score = _safe_divide(tn, tn + fn, zero_division)
     return _adjust_weights_safe_divide(score, average, multilabel, tp, fp, fn)
 
 
@@ -54,13 +57,16 @@ def binary_negative_predictive_value(
     multidim_average: str = "global",
     ignore_index: Optional[int] = None,
     validate_args: bool = True,
+    zero_division: float = 0,
 ) -> Tensor:
     """NPV for binary tasks."""
     if validate_args:
         _binary_stat_scores_arg_validation(threshold, multidim_average, ignore_index)
         _binary_stat_scores_tensor_validation(preds, target, multidim_average, ignore_index)
     tp, fp, tn, fn = _binary_stat_scores_pipeline(preds, target, threshold, multidim_average, ignore_index)
-    return _negative_predictive_value_reduce(tp, fp, tn, fn, average="binary", multidim_average=multidim_average)
+    return _negative_predictive_value_reduce(
+        tp, fp, tn, fn, average="binary", multidim_average=multidim_average, zero_division=zero_division
+    )
 
 
 def multiclass_negative_predictive_value(
@@ -72,6 +78,7 @@ def multiclass_negative_predictive_value(
     multidim_average: str = "global",
     ignore_index: Optional[int] = None,
     validate_args: bool = True,
+    zero_division: float = 0,
 ) -> Tensor:
     """NPV for multiclass tasks."""
     if validate_args:
@@ -80,7 +87,9 @@ def multiclass_negative_predictive_value(
     tp, fp, tn, fn = _multiclass_stat_scores_pipeline(
         preds, target, num_classes, top_k, average, multidim_average, ignore_index
     )
-    return _negative_predictive_value_reduce(tp, fp, tn, fn, average=average, multidim_average=multidim_average)
+    return _negative_predictive_value_reduce(
+        tp, fp, tn, fn, average=average, multidim_average=multidim_average, zero_division=zero_division
+    )
 
 
 def multilabel_negative_predictive_value(
@@ -92,6 +101,7 @@ def multilabel_negative_predictive_value(
     multidim_average: str = "global",
     ignore_index: Optional[int] = None,
     validate_args: bool = True,
+    zero_division: float = 0,
 ) -> Tensor:
     """NPV for multilabel tasks."""
     if validate_args:
@@ -101,7 +111,8 @@ def multilabel_negative_predictive_value(
         preds, target, num_labels, threshold, multidim_average, ignore_index
     )
     return _negative_predictive_value_reduce(
-        tp, fp, tn, fn, average=average, multidim_average=multidim_average, multilabel=True
+        tp, fp, tn, fn, average=average, multidim_average=multidim_average, multilabel=True,
+        zero_division=zero_division,
     )
 
 
@@ -117,23 +128,26 @@ def negative_predictive_value(
     top_k: int = 1,
     ignore_index: Optional[int] = None,
     validate_args: bool = True,
+    zero_division: float = 0,
 ) -> Tensor:
     """Task-dispatching negative predictive value."""
     task = ClassificationTask.from_str(task)
     if task == ClassificationTask.BINARY:
         return binary_negative_predictive_value(
-            preds, target, threshold, multidim_average, ignore_index, validate_args
+            preds, target, threshold, multidim_average, ignore_index, validate_args, zero_division
         )
     if task == ClassificationTask.MULTICLASS:
         if not isinstance(num_classes, int):
             raise ValueError(f"`num_classes` is expected to be `int` but `{type(num_classes)} was passed.`")
         return multiclass_negative_predictive_value(
-            preds, target, num_classes, average, top_k, multidim_average, ignore_index, validate_args
+            preds, target, num_classes, average, top_k, multidim_average, ignore_index, validate_args,
+            zero_division,
         )
     if task == ClassificationTask.MULTILABEL:
         if not isinstance(num_labels, int):
             raise ValueError(f"`num_labels` is expected to be `int` but `{type(num_labels)} was passed.`")
         return multilabel_negative_predictive_value(
-            preds, target, num_labels, threshold, average, multidim_average, ignore_index, validate_args
+            preds, target, num_labels, threshold, average, multidim_average, ignore_index, validate_args,
+            zero_division,
         )
     raise ValueError(f"Not handled value: {task}")

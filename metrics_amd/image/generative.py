@@ -197,6 +197,7 @@ class MemorizationInformedFrechetInceptionDistance(Metric):
     def __init__(
         self,
         feature: Any = 2048,
+        reset_real_features: bool = True,
         cosine_distance_eps: float = 0.1,
         normalize: bool = False,
         feature_extractor: Optional[Module] = None,
@@ -206,6 +207,9 @@ class MemorizationInformedFrechetInceptionDistance(Metric):
         if feature_extractor is None and isinstance(feature, Module):
             feature_extractor = feature
         self.inception = _feature_net_or_raise("MemorizationInformedFrechetInceptionDistance", feature_extractor)
+        if not isinstance(reset_real_features, bool):
+            raise ValueError("Argument `reset_real_features` expected to be a bool")
+        self.reset_real_features = reset_real_features
         if not (isinstance(cosine_distance_eps, float) and 1 > cosine_distance_eps > 0):
             raise ValueError("Argument `cosine_distance_eps` expected to be a float greater than 0 and less than 1")
         self.cosine_distance_eps = cosine_distance_eps
@@ -237,6 +241,15 @@ class MemorizationInformedFrechetInceptionDistance(Metric):
         mean_min_d = d.min(dim=1).values.mean()
         m_dist = mean_min_d if mean_min_d < self.cosine_distance_eps else torch.ones_like(mean_min_d)
         return (fid / (m_dist + 1e-15)).float()
+
+    def reset(self) -> None:
+        """Reset states; cached real features survive when reset_real_features=False."""
+        if not self.reset_real_features:
+            value = self._defaults.pop("real_features")
+            super().reset()
+            self._defaults["real_features"] = value
+        else:
+            super().reset()
 
     def plot(self, val=None, ax=None):
         return self._plot(val, ax)

@@ -405,9 +405,9 @@ class SpatialCorrelationCoefficient(_ScoreAverageMetric):
     plot_lower_bound: float = -1.0
     plot_upper_bound: float = 1.0
 
-    def __init__(self, hp_filter: Optional[Tensor] = None, window_size: int = 8, **kwargs: Any) -> None:
+    def __init__(self, high_pass_filter: Optional[Tensor] = None, window_size: int = 8, **kwargs: Any) -> None:
         super().__init__(**kwargs)
-        self.hp_filter = hp_filter
+        self.hp_filter = high_pass_filter
         self.window_size = window_size
 
     def update(self, preds: Tensor, target: Tensor) -> None:

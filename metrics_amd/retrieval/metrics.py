@@ -26,8 +26,9 @@ class _TopKRetrievalMetric(RetrievalMetric):
     """Shared top_k handling."""
 
     def __init__(self, empty_target_action: str = "neg", ignore_index: Optional[int] = None,
-                 top_k: Optional[int] = None, **kwargs: Any) -> None:
-        super().__init__(empty_target_action=empty_target_action, ignore_index=ignore_index, **kwargs)
+                 top_k: Optional[int] = None, aggregation="mean", **kwargs: Any) -> None:
+        super().__init__(empty_target_action=empty_target_action, ignore_index=ignore_index,
+                         aggregation=aggregation, **kwargs)
         if top_k is not None and not (isinstance(top_k, int) and top_k > 0):
             raise ValueError("`top_k` has to be a positive integer or None")
         self.top_k = top_k
@@ -78,8 +79,8 @@ class RetrievalPrecision(_TopKRetrievalMetric):
     """Precision@k."""
 
     def __init__(self, empty_target_action: str = "neg", ignore_index: Optional[int] = None,
-                 top_k: Optional[int] = None, adaptive_k: bool = False, **kwargs: Any) -> None:
-        super().__init__(empty_target_action, ignore_index, top_k, **kwargs)
+                 top_k: Optional[int] = None, adaptive_k: bool = False, aggregation="mean", **kwargs: Any) -> None:
+        super().__init__(empty_target_action, ignore_index, top_k, aggregation=aggregation, **kwargs)
         if not isinstance(adaptive_k, bool):
             raise ValueError("`adaptive_k` has to be a boolean")
         self.adaptive_k = adaptive_k
@@ -139,8 +140,8 @@ class RetrievalNormalizedDCG(_TopKRetrievalMetric):
     """Normalized DCG (graded relevance allowed)."""
 
     def __init__(self, empty_target_action: str = "neg", ignore_index: Optional[int] = None,
-                 top_k: Optional[int] = None, **kwargs: Any) -> None:
-        super().__init__(empty_target_action, ignore_index, top_k, **kwargs)
+                 top_k: Optional[int] = None, aggregation="mean", **kwargs: Any) -> None:
+        super().__init__(empty_target_action, ignore_index, top_k, aggregation=aggregation, **kwargs)
         self.allow_non_binary_target = True
 
     def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
@@ -175,8 +176,8 @@ class RetrievalAUROC(_TopKRetrievalMetric):
     """Per-query AUROC, averaged."""
 
     def __init__(self, empty_target_action: str = "neg", ignore_index: Optional[int] = None,
-                 top_k: Optional[int] = None, max_fpr: Optional[float] = None, **kwargs: Any) -> None:
-        super().__init__(empty_target_action, ignore_index, top_k, **kwargs)
+                 top_k: Optional[int] = None, max_fpr: Optional[float] = None, aggregation="mean", **kwargs: Any) -> None:
+        super().__init__(empty_target_action, ignore_index, top_k, aggregation=aggregation, **kwargs)
         self.max_fpr = max_fpr
 
     def _metric(self, preds: Tensor, target: Tensor) -> Tensor:

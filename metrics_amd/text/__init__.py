@@ -642,8 +642,19 @@ class BERTScore(Metric):
         all_layers: bool = False,
         model=None,
         user_tokenizer=None,
+        user_forward_fn=None,
+        verbose: bool = False,
+        idf: bool = False,
+        device=None,
         max_length: int = 512,
         batch_size: int = 64,
+        num_threads: int = 0,
+        return_hash: bool = False,
+        lang: str = "en",
+        rescale_with_baseline: bool = False,
+        baseline_path: Optional[str] = None,
+        baseline_url: Optional[str] = None,
+        truncation: bool = False,
         **kwargs: Any,
     ) -> None:
         super().__init__(**kwargs)
@@ -652,10 +663,32 @@ class BERTScore(Metric):
                 "BERTScore needs a local transformer model + tokenizer: pass `model=` and `user_tokenizer=`"
                 " (weights cannot be downloaded in this offline environment)."
             )
+        # reference-parity knobs that require online artifacts are rejected
+        # loudly instead of silently ignored
+        if idf:
+            raise NotImplementedError(
+                "BERTScore(idf=True) is not supported offline (cannot be validated without real model weights)."
+            )
+        if rescale_with_baseline:
+            raise NotImplementedError(
+                "BERTScore(rescale_with_baseline=True) needs baseline files that cannot be fetched offline."
+            )
+        if return_hash:
+            raise NotImplementedError("BERTScore(return_hash=True) refers to a downloaded-model hash (offline N/A).")
         self.model = model
         self.tokenizer = user_tokenizer
+        self.user_forward_fn = user_forward_fn
+        self.verbose = verbose
+        self.lang = lang
+        self.truncation = truncation
         self.max_length = max_length
         self.batch_size = batch_size
+        if num_threads:
+            import torch as _torch
+
+            _torch.set_num_threads(num_threads)
+        if device is not None:
+            self.model = self.model.to(device)
         self.model.eval()
         self.add_state("precision_scores", [], dist_reduce_fx="cat")
         self.add_state("recall_scores", [], dist_reduce_fx="cat")
@@ -663,8 +696,14 @@ class BERTScore(Metric):
 
     @torch.no_grad()
     def _embed(self, texts: List[str]):
-        enc = self.tokenizer(texts, return_tensors="pt", padding=True, truncation=True, max_length=self.max_length)
-        out = self.model(**enc)
+        enc = self.tokenizer(
+            texts, return_tensors="pt", padding=True, truncation=self.truncation or True,
+            max_length=self.max_length,
+        )
+        if self.user_forward_fn is not None:
+            out = self.user_forward_fn(self.model, enc)
+        else:
+            out = self.model(**enc)
         emb = out.last_hidden_state if hasattr(out, "last_hidden_state") else out[0]
         mask = enc["attention_mask"].bool()
         emb = emb / emb.norm(dim=-1, keepdim=True).clamp(min=1e-12)
@@ -713,6 +752,14 @@ class InfoLM(Metric):
         temperature: float = 0.25,
         information_measure: str = "kl_divergence",
         idf: bool = False,
+        alpha: float = 0.5,
+        beta: float = 0.5,
+        device=None,
+        max_length: Optional[int] = None,
+        batch_size: int = 64,
+        num_threads: int = 0,
+        verbose: bool = True,
+        return_sentence_level_score: bool = False,
         model=None,
         user_tokenizer=None,
         **kwargs: Any,
@@ -732,6 +779,18 @@ class InfoLM(Metric):
         self.temperature = temperature
         self.information_measure = information_measure
         self.idf = idf
+        self.alpha = alpha
+        self.beta = beta
+        self.max_length = max_length
+        self.batch_size = batch_size
+        self.verbose = verbose
+        self.return_sentence_level_score = return_sentence_level_score
+        if num_threads:
+            import torch as _torch
+
+            _torch.set_num_threads(num_threads)
+        if device is not None:
+            self.model = self.model.to(device)
         self.model.eval()
         self.add_state("scores", [], dist_reduce_fx="cat")
 
@@ -762,12 +821,35 @@ class InfoLM(Metric):
             score = (t - p).abs().max(-1).values
         elif self.information_measure == "fisher_rao_distance":
             score = 2 * torch.acos(((t * p).sqrt().sum(-1)).clamp(0, 1))
+        elif self.information_measure == "alpha_divergence":
+            a = self.alpha
+            score = (1.0 / (a * (a - 1.0))) * ((t.pow(a) * p.pow(1 - a)).sum(-1) - 1.0)
+        elif self.information_measure == "beta_divergence":
+            b = self.beta
+            score = (
+                (t.pow(b + 1).sum(-1) / (b * (b + 1)))
+                + (p.pow(b + 1).sum(-1) / (b + 1))
+                - ((t * p.pow(b)).sum(-1) / b)
+            )
+        elif self.information_measure == "ab_divergence":
+            a, b = self.alpha, self.beta
+            score = (
+                (t.pow(a + b).sum(-1) / (b * (a + b)))
+                + (p.pow(a + b).sum(-1) / (a * (a + b)))
+                - ((t.pow(a) * p.pow(b)).sum(-1) / (a * b))
+            )
+        elif self.information_measure == "renyi_divergence":
+            a = self.alpha
+            score = ((t.pow(a) * p.pow(1 - a)).sum(-1)).log() / (a - 1.0)
         else:
             score = (t * (t / p).log()).sum(-1)
         self.scores.append(score)
 
-    def compute(self) -> Tensor:
-        return dim_zero_cat(self.scores).mean()
+    def compute(self):
+        scores = dim_zero_cat(self.scores)
+        if self.return_sentence_level_score:
+            return scores.mean(), scores
+        return scores.mean()
 
 
 __all__ = [

@@ -1247,3 +1247,61 @@ def test_ssim_option_differential(opt):
                 _cmp(a, b_, 1e-5)
         else:
             _cmp(vo, vr, 1e-5)
+
+
+def test_constructor_signature_parity():
+    """Every shared public callable accepts at least the reference's named
+    parameters (introspection parity: help()/IDE/keyword-call compatible)."""
+    import importlib
+    import inspect
+    import warnings
+
+    tm = _tm()
+    namespaces = ["", "classification", "regression", "retrieval", "text", "audio", "image",
+                  "detection", "segmentation", "clustering", "nominal", "aggregation",
+                  "wrappers", "shape", "pairwise", "functional"]
+    mismatches = []
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        for ns in namespaces:
+            try:
+                our_ns = importlib.import_module(f"metrics_amd.{ns}") if ns else ma
+                ref_ns = importlib.import_module(f"torchmetrics.{ns}") if ns else tm
+            except Exception:
+                continue
+            for n in getattr(ref_ns, "__all__", []):
+                rc, oc = getattr(ref_ns, n, None), getattr(our_ns, n, None)
+                if rc is None or oc is None or not (inspect.isclass(rc) or callable(rc)):
+                    continue
+                try:
+                    rsig = inspect.signature(rc.__init__ if inspect.isclass(rc) else rc)
+                    osig = inspect.signature(oc.__init__ if inspect.isclass(oc) else oc)
+                except (ValueError, TypeError):
+                    continue
+                miss = [p for p in rsig.parameters
+                        if p not in ("self", "kwargs", "args") and p not in osig.parameters]
+                if miss:
+                    mismatches.append((ns or "top", n, miss))
+    assert not mismatches, mismatches
+
+
+@pytest.mark.parametrize("zero_division", [0.0, 1.0])
+def test_zero_division_differential(zero_division):
+    """zero_division flows to the same values as the reference (degenerate
+    all-one-class inputs make divisions by zero actually happen)."""
+    import warnings
+
+    tm = _tm()
+    p = torch.zeros(40)  # all predicted negative -> tp + fp = 0
+    t = torch.ones(40, dtype=torch.long)
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        for name in ("BinaryF1Score", "BinaryFBetaScore", "BinaryJaccardIndex"):
+            kw = {"zero_division": zero_division}
+            if "FBeta" in name:
+                kw["beta"] = 0.5
+            ours = getattr(ma, name)(**kw)
+            ref = getattr(tm.classification, name)(**kw)
+            ours.update(p, t)
+            ref.update(p, t)
+            _cmp(ours.compute(), ref.compute(), 1e-6)
