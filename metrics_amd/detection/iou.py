@@ -45,12 +45,16 @@ class IntersectionOverUnion(Metric):
 
     def __init__(
         self,
+        box_format: str = "xyxy",
         iou_threshold: Optional[float] = None,
         class_metrics: bool = False,
         respect_labels: bool = True,
         **kwargs: Any,
     ) -> None:
         super().__init__(**kwargs)
+        if box_format not in ("xyxy", "xywh", "cxcywh"):
+            raise ValueError(f"Expected argument `box_format` to be one of ('xyxy', 'xywh', 'cxcywh') but got {box_format}")
+        self.box_format = box_format
         if iou_threshold is not None and not isinstance(iou_threshold, float):
             raise ValueError(f"Expected argument `iou_threshold` to be a float or None, but got {iou_threshold}")
         self.iou_threshold = iou_threshold
@@ -69,9 +73,16 @@ class IntersectionOverUnion(Metric):
         semantics: compute() means over every pair whose entry survived the
         label/threshold masking — NOT a best-match assignment)."""
         _input_validator_iou(preds, target, ignore_score=True)
+        from metrics_amd.detection.mean_ap import box_convert
+
         for p, t in zip(preds, target):
             p_boxes, p_labels = p["boxes"].float(), p["labels"]
             t_boxes, t_labels = t["boxes"].float(), t["labels"]
+            if self.box_format != "xyxy":
+                if p_boxes.numel():
+                    p_boxes = box_convert(p_boxes, in_fmt=self.box_format, out_fmt="xyxy")
+                if t_boxes.numel():
+                    t_boxes = box_convert(t_boxes, in_fmt=self.box_format, out_fmt="xyxy")
             self.groundtruth_labels.append(t_labels)
             # reference empty-input quirks (functional/detection/iou.py:35-38):
             # no preds -> zeros(M, M); no gts -> zeros(N, N)
