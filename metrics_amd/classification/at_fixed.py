@@ -434,7 +434,7 @@ class BinaryLogAUC(BinaryPrecisionRecallCurve):
     plot_upper_bound: float = 1.0
 
     def __init__(self, fpr_range: Tuple[float, float] = (0.001, 0.1), thresholds=None, ignore_index=None,
-                 validate_args: bool = True, **kwargs: Any) -> None:
+                 validate_args: bool = False, **kwargs: Any) -> None:
         super().__init__(thresholds, ignore_index, validate_args=False, **kwargs)
         self.fpr_range = tuple(fpr_range)
         self.validate_args = validate_args
@@ -452,7 +452,7 @@ class MulticlassLogAUC(MulticlassPrecisionRecallCurve):
     higher_is_better = True
     full_state_update = False
 
-    def __init__(self, num_classes: int, fpr_range: Tuple[float, float] = (0.001, 0.1), average: Optional[str] = "macro",
+    def __init__(self, num_classes: int, fpr_range: Tuple[float, float] = (0.001, 0.1), average: Optional[str] = None,
                  thresholds=None, ignore_index=None, validate_args: bool = True, **kwargs: Any) -> None:
         super().__init__(num_classes, thresholds, ignore_index=ignore_index, validate_args=False, **kwargs)
         self.fpr_range = tuple(fpr_range)
@@ -475,7 +475,7 @@ class MultilabelLogAUC(MultilabelPrecisionRecallCurve):
     higher_is_better = True
     full_state_update = False
 
-    def __init__(self, num_labels: int, fpr_range: Tuple[float, float] = (0.001, 0.1), average: Optional[str] = "macro",
+    def __init__(self, num_labels: int, fpr_range: Tuple[float, float] = (0.001, 0.1), average: Optional[str] = None,
                  thresholds=None, ignore_index=None, validate_args: bool = True, **kwargs: Any) -> None:
         super().__init__(num_labels, thresholds, ignore_index=ignore_index, validate_args=False, **kwargs)
         self.fpr_range = tuple(fpr_range)

@@ -133,7 +133,7 @@ class MulticlassStatScores(_AbstractStatScores):
 
     def __init__(
         self,
-        num_classes: int,
+        num_classes: Optional[int] = None,
         top_k: int = 1,
         average: Optional[str] = "macro",
         multidim_average: str = "global",
@@ -143,7 +143,12 @@ class MulticlassStatScores(_AbstractStatScores):
     ) -> None:
         zero_division = kwargs.pop("zero_division", 0)
         super(_AbstractStatScores, self).__init__(**kwargs)
-        if validate_args:
+        if num_classes is None and average != "micro":
+            # reference semantics (functional/classification/stat_scores.py:238)
+            raise ValueError(
+                f"Argument `num_classes` can only be `None` for `average='micro'`, but got `average={average}`."
+            )
+        if validate_args and num_classes is not None:
             _multiclass_stat_scores_arg_validation(
                 num_classes, top_k, average, multidim_average, ignore_index, zero_division
             )
@@ -157,12 +162,30 @@ class MulticlassStatScores(_AbstractStatScores):
 
         # per-class states regardless of averaging: micro reduces at compute
         # time, so micro- and macro-averaged metrics land in the SAME compute
-        # group (one fused kernel launch per collection update)
-        self._create_state(size=num_classes, multidim_average=multidim_average)
+        # group (one fused kernel launch per collection update). With
+        # num_classes=None (micro-only) the class count is inferred per batch
+        # and the totals land in size-1 states.
+        self._create_state(size=num_classes if num_classes is not None else 1,
+                           multidim_average=multidim_average)
 
     def update(self, preds: Tensor, target: Tensor) -> None:
         """Accumulate batch statistics (GPU: fused argmax+count HIP kernel
         accumulating DIRECTLY into the tp/fp/tn/fn states — 3 launches)."""
+        if self.num_classes is None:
+            # micro-only convenience path: infer the class count per batch
+            # and accumulate global totals into the size-1 states
+            if preds.is_floating_point() and preds.ndim == target.ndim + 1:
+                n_cls = preds.shape[1]
+            else:
+                n_cls = int(torch.maximum(preds.max(), target.max()).item()) + 1
+            tp, fp, tn, fn = _multiclass_stat_scores_pipeline(
+                preds, target, n_cls, self.top_k, "micro", self.multidim_average, self.ignore_index
+            )
+            self.tp = self.tp + tp.sum()
+            self.fp = self.fp + fp.sum()
+            self.tn = self.tn + tn.sum()
+            self.fn = self.fn + fn.sum()
+            return
         if self.validate_args:
             _multiclass_stat_scores_tensor_validation(
                 preds, target, self.num_classes, self.multidim_average, self.ignore_index

@@ -239,7 +239,7 @@ class MeanAveragePrecision(Metric):
         class_metrics: bool = False,
         extended_summary: bool = False,
         average: str = "macro",
-        backend: str = "native",
+        backend: str = "pycocotools",
         **kwargs: Any,
     ) -> None:
         super().__init__(**kwargs)
@@ -261,6 +261,12 @@ class MeanAveragePrecision(Metric):
         if average not in ("macro", "micro"):
             raise ValueError(f"Expected argument `average` to be one of ('macro', 'micro') but got {average}")
         self.average = average
+        # both reference backend names run THIS engine (HIP IoU + OpenMP
+        # matcher + numpy accumulate — pycocotools-faithful semantics, fuzz
+        # tested against the reference's COCOeval); the arg is kept for
+        # reference signature compatibility
+        if backend not in ("pycocotools", "faster_coco_eval", "native"):
+            raise ValueError(f"Expected argument `backend` to be one of ('pycocotools', 'faster_coco_eval') but got {backend}")
         self.backend = backend
 
         self.add_state("detection_boxes", default=[], dist_reduce_fx=None)

@@ -321,17 +321,19 @@ class PerceptualPathLength(Metric):
         self,
         num_samples: int = 10_000,
         conditional: bool = False,
-        batch_size: int = 64,
+        batch_size: int = 128,
         interpolation_method: str = "lerp",
         epsilon: float = 1e-4,
         resize: Optional[int] = 64,
         lower_discard: Optional[float] = 0.01,
         upper_discard: Optional[float] = 0.99,
-        sim_net: Optional[Callable] = None,
+        sim_net: Any = "vgg",
         **kwargs: Any,
     ) -> None:
         super().__init__(**kwargs)
-        if sim_net is None:
+        if sim_net is None or isinstance(sim_net, str):
+            # the reference's default 'vgg' downloads LPIPS weights; offline
+            # that path cannot work — require a user-supplied network
             raise ModuleNotFoundError(
                 "PerceptualPathLength needs a similarity network (LPIPS-style): pass `sim_net=`."
             )

@@ -51,7 +51,7 @@ class _NominalBase(Metric):
 class CramersV(_NominalBase):
     """Cramer's V (stateful)."""
 
-    def __init__(self, num_classes: Optional[int] = None, bias_correction: bool = True,
+    def __init__(self, num_classes: int, bias_correction: bool = True,
                  nan_strategy: str = "replace", nan_replace_value: Optional[float] = 0.0, **kwargs: Any) -> None:
         super().__init__(nan_strategy, nan_replace_value, **kwargs)
         self.num_classes = num_classes
@@ -67,7 +67,7 @@ class CramersV(_NominalBase):
 class PearsonsContingencyCoefficient(_NominalBase):
     """Pearson's contingency coefficient (stateful)."""
 
-    def __init__(self, num_classes: Optional[int] = None, nan_strategy: str = "replace",
+    def __init__(self, num_classes: int, nan_strategy: str = "replace",
                  nan_replace_value: Optional[float] = 0.0, **kwargs: Any) -> None:
         super().__init__(nan_strategy, nan_replace_value, **kwargs)
         self.num_classes = num_classes
@@ -81,7 +81,7 @@ class PearsonsContingencyCoefficient(_NominalBase):
 class TschuprowsT(_NominalBase):
     """Tschuprow's T (stateful)."""
 
-    def __init__(self, num_classes: Optional[int] = None, bias_correction: bool = True,
+    def __init__(self, num_classes: int, bias_correction: bool = True,
                  nan_strategy: str = "replace", nan_replace_value: Optional[float] = 0.0, **kwargs: Any) -> None:
         super().__init__(nan_strategy, nan_replace_value, **kwargs)
         self.num_classes = num_classes
@@ -97,7 +97,7 @@ class TschuprowsT(_NominalBase):
 class TheilsU(_NominalBase):
     """Theil's U (stateful)."""
 
-    def __init__(self, num_classes: Optional[int] = None, nan_strategy: str = "replace",
+    def __init__(self, num_classes: int, nan_strategy: str = "replace",
                  nan_replace_value: Optional[float] = 0.0, **kwargs: Any) -> None:
         super().__init__(nan_strategy, nan_replace_value, **kwargs)
         self.num_classes = num_classes

@@ -128,6 +128,12 @@ class RetrievalFallOut(_TopKRetrievalMetric):
     higher_is_better = False
     _empty_on_negatives = True
 
+    def __init__(self, empty_target_action: str = "pos", ignore_index: Optional[int] = None,
+                 top_k: Optional[int] = None, aggregation="mean", **kwargs: Any) -> None:
+        # reference default differs from the family: a query with no negatives
+        # is a PERFECT fall-out, so the fill default is "pos" (retrieval/fall_out.py)
+        super().__init__(empty_target_action, ignore_index, top_k, aggregation=aggregation, **kwargs)
+
     def _metric(self, preds: Tensor, target: Tensor) -> Tensor:
         return retrieval_fall_out(preds, target, top_k=self.top_k)
 

@@ -32,7 +32,7 @@ class MeanIoU(Metric):
 
     def __init__(
         self,
-        num_classes: Optional[int] = None,
+        num_classes: int,
         include_background: bool = True,
         per_class: bool = False,
         input_format: str = "one-hot",

@@ -748,12 +748,12 @@ class InfoLM(Metric):
 
     def __init__(
         self,
-        model_name_or_path: Optional[str] = None,
+        model_name_or_path: str = "bert-base-uncased",
         temperature: float = 0.25,
         information_measure: str = "kl_divergence",
-        idf: bool = False,
-        alpha: float = 0.5,
-        beta: float = 0.5,
+        idf: bool = True,
+        alpha: Optional[float] = None,
+        beta: Optional[float] = None,
         device=None,
         max_length: Optional[int] = None,
         batch_size: int = 64,
@@ -774,6 +774,10 @@ class InfoLM(Metric):
                    "renyi_divergence", "l1_distance", "l2_distance", "l_infinity_distance", "fisher_rao_distance")
         if information_measure not in allowed:
             raise ValueError(f"Argument `information_measure` expected one of {allowed}")
+        if information_measure in ("alpha_divergence", "ab_divergence", "renyi_divergence") and alpha is None:
+            raise ValueError(f"Argument `alpha` is required for information_measure={information_measure}")
+        if information_measure in ("beta_divergence", "ab_divergence") and beta is None:
+            raise ValueError(f"Argument `beta` is required for information_measure={information_measure}")
         self.model = model
         self.tokenizer = user_tokenizer
         self.temperature = temperature

@@ -70,7 +70,7 @@ def test_multiclass_at_fixed_and_logauc_classes():
         m.update(p, t)
         v, thr = m.compute()
         assert v.shape == (5,)
-    la = ma.MulticlassLogAUC(num_classes=5, thresholds=50)
+    la = ma.MulticlassLogAUC(average="macro", num_classes=5, thresholds=50)
     la.update(p, t)
     assert la.compute().ndim == 0
     # task-dispatch wrappers construct the right subclass
@@ -177,7 +177,7 @@ def test_multilabel_at_fixed_and_logauc_classes():
         m.update(MLP, MLT)
         v, thr = m.compute()
         assert v.shape == (L,), cls.__name__
-    la = ma.MultilabelLogAUC(num_labels=L, thresholds=25)
+    la = ma.MultilabelLogAUC(average="macro", num_labels=L, thresholds=25)
     la.update(MLP, MLT)
     assert la.compute().ndim == 0
     # abstract bases are exported for subclassing (reference parity)

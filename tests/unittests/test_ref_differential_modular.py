@@ -1282,6 +1282,17 @@ def test_constructor_signature_parity():
                         if p not in ("self", "kwargs", "args") and p not in osig.parameters]
                 if miss:
                     mismatches.append((ns or "top", n, miss))
+                for pname, rp in rsig.parameters.items():
+                    if pname in ("self", "kwargs", "args") or pname not in osig.parameters:
+                        continue
+                    rd, od = rp.default, osig.parameters[pname].default
+                    if rd is inspect.Parameter.empty or od is inspect.Parameter.empty:
+                        if (rd is inspect.Parameter.empty) != (od is inspect.Parameter.empty):
+                            mismatches.append((ns or "top", n, pname, "required-ness", repr(rd), repr(od)))
+                        continue
+                    eq = rd == od
+                    if isinstance(eq, bool) and not eq and not (rd != rd and od != od):
+                        mismatches.append((ns or "top", n, pname, "default", repr(rd), repr(od)))
     assert not mismatches, mismatches
 
 
