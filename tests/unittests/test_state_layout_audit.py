@@ -56,6 +56,10 @@ _KNOWN_DEVIATIONS = {
     # simple RCCL-reducible denominator stats (sum/min/max) instead of the
     # reference's running mean/var with dist_reduce_fx=None merge
     "NormalizedRootMeanSquaredError",
+    # nominal association metrics: O(1) (C,C) contingency counts instead of
+    # the reference's unbounded preds/target cat lists (num_classes is a
+    # required argument in both; value parity in test_ref_differential_modular)
+    "CramersV", "PearsonsContingencyCoefficient", "TheilsU", "TschuprowsT",
 }
 
 
