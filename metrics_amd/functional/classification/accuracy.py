@@ -3,6 +3,7 @@ from __future__ import annotations
 
 from typing import Optional
 
+import torch
 from torch import Tensor
 
 from metrics_amd.utilities.compute import _adjust_weights_safe_divide, _safe_divide
@@ -75,7 +76,7 @@ def binary_accuracy(
 def multiclass_accuracy(
     preds: Tensor,
     target: Tensor,
-    num_classes: int,
+    num_classes: Optional[int] = None,
     average: Optional[str] = "macro",
     top_k: int = 1,
     multidim_average: str = "global",
@@ -85,7 +86,14 @@ def multiclass_accuracy(
     """Accuracy for multiclass tasks (GPU: fused argmax+count kernel)."""
     if validate_args:
         _multiclass_stat_scores_arg_validation(num_classes, top_k, average, multidim_average, ignore_index)
-        _multiclass_stat_scores_tensor_validation(preds, target, num_classes, multidim_average, ignore_index)
+        if num_classes is not None:
+            _multiclass_stat_scores_tensor_validation(preds, target, num_classes, multidim_average, ignore_index)
+    if num_classes is None:
+        # micro-only convenience (reference accuracy.py:169): infer the count
+        if preds.is_floating_point() and preds.ndim == target.ndim + 1:
+            num_classes = preds.shape[1]
+        else:
+            num_classes = int(torch.maximum(preds.max(), target.max()).item()) + 1
     tp, fp, tn, fn = _multiclass_stat_scores_pipeline(
         preds, target, num_classes, top_k, average, multidim_average, ignore_index
     )

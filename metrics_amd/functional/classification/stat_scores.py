@@ -150,13 +150,20 @@ def binary_stat_scores(
 
 
 def _multiclass_stat_scores_arg_validation(
-    num_classes: int,
+    num_classes: Optional[int],
     top_k: int = 1,
     average: Optional[str] = "macro",
     multidim_average: str = "global",
     ignore_index: Optional[int] = None,
     zero_division: float = 0,
 ) -> None:
+    if num_classes is None:
+        # reference semantics: only micro averaging works without a class count
+        if average != "micro":
+            raise ValueError(
+                f"Argument `num_classes` can only be `None` for `average='micro'`, but got `average={average}`."
+            )
+        num_classes = 2  # satisfy the downstream top_k bound check
     if not isinstance(num_classes, int) or num_classes < 2:
         raise ValueError(f"Expected argument `num_classes` to be an integer larger than 1, but got {num_classes}")
     if not isinstance(top_k, int) and top_k < 1:

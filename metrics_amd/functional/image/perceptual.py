@@ -44,7 +44,7 @@ def perceptual_path_length(
     resize: Optional[int] = 64,
     lower_discard: Optional[float] = 0.01,
     upper_discard: Optional[float] = 0.99,
-    sim_net: Optional[Callable] = None,
+    sim_net: Any = "vgg",
     device: Union[str, torch.device] = "cpu",
 ) -> Tuple[Tensor, Tensor, Tensor]:
     """Perceptual path length of a generator; returns (mean, std, distances)."""

@@ -10,7 +10,7 @@ from torch import Tensor
 def clip_score(
     source: Union[Tensor, List[Tensor], List[str], str],
     target: Union[Tensor, List[Tensor], List[str], str],
-    model_name_or_path: Optional[str] = None,
+    model_name_or_path: str = "openai/clip-vit-large-patch14",
     model: Any = None,
     processor: Any = None,
 ) -> Tensor:
@@ -27,7 +27,7 @@ def clip_score(
 
 def clip_image_quality_assessment(
     images: Tensor,
-    model_name_or_path: Optional[str] = None,
+    model_name_or_path: str = "clip_iqa",
     data_range: float = 1.0,
     prompts: tuple = ("quality",),
     model: Any = None,

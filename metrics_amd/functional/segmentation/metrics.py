@@ -36,7 +36,7 @@ def _format_inputs(preds: Tensor, target: Tensor, num_classes: int, input_format
 def _mean_iou_update(
     preds: Tensor,
     target: Tensor,
-    num_classes: Optional[int] = None,
+    num_classes: int,
     include_background: bool = False,
     input_format: str = "one-hot",
 ) -> Tuple[Tensor, Tensor]:
