@@ -1259,7 +1259,12 @@ def test_constructor_signature_parity():
     tm = _tm()
     namespaces = ["", "classification", "regression", "retrieval", "text", "audio", "image",
                   "detection", "segmentation", "clustering", "nominal", "aggregation",
-                  "wrappers", "shape", "pairwise", "functional"]
+                  "wrappers", "shape", "pairwise", "functional",
+                  "functional.classification", "functional.regression", "functional.retrieval",
+                  "functional.text", "functional.audio", "functional.image",
+                  "functional.detection", "functional.segmentation", "functional.clustering",
+                  "functional.nominal", "functional.pairwise", "functional.shape",
+                  "functional.multimodal"]
     mismatches = []
     with warnings.catch_warnings():
         warnings.simplefilter("ignore")
