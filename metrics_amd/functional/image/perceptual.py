@@ -38,7 +38,7 @@ def perceptual_path_length(
     generator,
     num_samples: int = 10_000,
     conditional: bool = False,
-    batch_size: int = 128,
+    batch_size: int = 64,
     interpolation_method: str = "lerp",
     epsilon: float = 1e-4,
     resize: Optional[int] = 64,

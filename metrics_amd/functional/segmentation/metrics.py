@@ -68,7 +68,7 @@ def _mean_iou_compute(intersection: Tensor, union: Tensor, zero_division: Union[
 def mean_iou(
     preds: Tensor,
     target: Tensor,
-    num_classes: Optional[int] = None,
+    num_classes: int,
     include_background: bool = True,
     per_class: bool = False,
     input_format: str = "one-hot",
