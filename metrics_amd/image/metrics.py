@@ -11,7 +11,6 @@ from metrics_amd.utilities.data import dim_zero_cat
 from metrics_amd.functional.image.psnr import (
     _psnr_compute,
     _psnr_update,
-    peak_signal_noise_ratio_with_blocked_effect,
 )
 from metrics_amd.functional.image.ssim import (
     multiscale_structural_similarity_index_measure,
@@ -19,7 +18,6 @@ from metrics_amd.functional.image.ssim import (
 )
 from metrics_amd.functional.image.misc import (
     error_relative_global_dimensionless_synthesis,
-    relative_average_spectral_error,
     root_mean_squared_error_using_sliding_window,
     spatial_correlation_coefficient,
     spectral_angle_mapper,
