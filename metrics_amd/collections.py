@@ -24,6 +24,7 @@ from torch import Tensor
 from torch.nn import ModuleDict
 
 from metrics_amd.metric import Metric
+from metrics_amd.utilities import tracing
 from metrics_amd.utilities.data import _flatten_dict, allclose
 from metrics_amd.utilities.prints import rank_zero_warn
 
@@ -198,7 +199,8 @@ class MetricCollection(ModuleDict):
                 self._fused_plan = _FusedMulticlassUpdatePlan.build(self)
             fused_done: tuple = ()
             if self._fused_plan is not None:
-                fused_done = self._fused_plan.try_run(*args, **kwargs)
+                with tracing.range("MetricCollection.fused_update"):
+                    fused_done = self._fused_plan.try_run(*args, **kwargs)
             # run only each group's leader
             for members in self._groups.values():
                 leader = getattr(self, members[0])

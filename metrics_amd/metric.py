@@ -571,14 +571,15 @@ class Metric(Module, ABC):
             if self._computed is not None:
                 return self._computed
 
-            # fast path: nothing to sync (single process / sync disabled) and
-            # no pending side-stream event — skip the sync/unsync context
-            # machinery entirely (two generator contexts per metric per
-            # compute are pure dispatch overhead in the hot loop)
+            # fast path: nothing to sync (single process / sync disabled),
+            # no pending side-stream event, tracing off — skip the sync/unsync
+            # context machinery entirely (two generator contexts per metric
+            # per compute are pure dispatch overhead in the hot loop)
             if (
                 not self._is_synced
                 and self.__dict__.get("_pending_sync_event") is None
                 and not (self._to_sync and self.distributed_available_fn())
+                and not tracing.is_enabled()
             ):
                 value = _squeeze_if_scalar(compute(*args, **kwargs))
                 value = _clone_result(value, self._state_storages())
