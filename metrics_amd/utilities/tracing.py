@@ -23,7 +23,14 @@ def _load() -> Optional[ctypes.CDLL]:
     global _LIB
     if _LIB is not None:
         return _LIB
-    for cand in ("/opt/rocm/lib/libroctx64.so", "libroctx64.so"):
+    # the rocprofiler-SDK roctx is what rocprofv3 --marker-trace intercepts;
+    # legacy libroctx64 ranges are invisible to it (verified on ROCm 7.x)
+    for cand in (
+        "/opt/rocm/lib/librocprofiler-sdk-roctx.so",
+        "librocprofiler-sdk-roctx.so",
+        "/opt/rocm/lib/libroctx64.so",
+        "libroctx64.so",
+    ):
         try:
             _LIB = ctypes.CDLL(cand)
             _LIB.roctxRangePushA.argtypes = [ctypes.c_char_p]
