@@ -1373,3 +1373,20 @@ def test_curve_auc_average_variants_gpu(cls_name, average):
         g.update(p, t)
         c.update(p.cpu(), t.cpu())
     assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-5)
+
+
+@pytest.mark.parametrize("aggregation", ["mean", "median", "min", "max"])
+def test_retrieval_aggregation_gpu(aggregation):
+    """Aggregation modes ride the batched GPU scoring path (composite-key
+    rocPRIM sort) and match CPU."""
+    import metrics_amd as ma
+
+    torch.manual_seed(41)
+    p = torch.rand(2000, device="cuda")
+    t = torch.randint(0, 2, (2000,), device="cuda")
+    idx = torch.randint(0, 50, (2000,), device="cuda")
+    g = ma.retrieval.RetrievalMAP(aggregation=aggregation).to("cuda")
+    c = ma.retrieval.RetrievalMAP(aggregation=aggregation)
+    g.update(p, t, indexes=idx)
+    c.update(p.cpu(), t.cpu(), indexes=idx.cpu())
+    assert torch.allclose(g.compute().cpu(), c.compute(), atol=1e-6)
