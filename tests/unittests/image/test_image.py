@@ -307,3 +307,43 @@ def test_ssim_3d_identity_and_bounds():
     t = torch.rand(2, 2, 12, 12, 12)
     v = ssim(p, t, data_range=1.0)
     assert -1.0 <= float(v) <= 1.0
+
+
+def test_clip_score_math_with_toy_model():
+    """CLIPScore math (normalized feature cosine * 100, clamp-at-0 mean)
+    verified with a deterministic toy model + processor."""
+    import torch
+
+    import metrics_amd as ma
+
+    class ToyModel:
+        def eval(self):
+            return self
+
+        def get_image_features(self, pixel_values):
+            return pixel_values.reshape(pixel_values.shape[0], -1)[:, :4].float()
+
+        def get_text_features(self, input_ids, attention_mask):
+            return input_ids[:, :4].float()
+
+    class ToyProcessor:
+        def __call__(self, text, images, return_tensors, padding):
+            pix = torch.stack([i.reshape(-1)[:12].reshape(3, 2, 2).float() for i in images])
+            ids = torch.stack([torch.arange(1, 5) * (len(t) % 7 + 1) for t in text])
+            return {"pixel_values": pix, "input_ids": ids, "attention_mask": torch.ones_like(ids)}
+
+    m = ma.multimodal.CLIPScore(model=ToyModel(), processor=ToyProcessor())
+    g = torch.Generator().manual_seed(3)
+    imgs = [torch.randint(0, 255, (3, 8, 8), generator=g).byte() for _ in range(3)]
+    texts = ["a cat", "a dog on grass", "tree"]
+    m.update(imgs, texts)
+    out = m.compute()
+
+    # hand-computed expectation
+    proc = ToyProcessor()(texts, imgs, "pt", True)
+    fi = ToyModel().get_image_features(proc["pixel_values"])
+    fi = fi / fi.norm(p=2, dim=-1, keepdim=True)
+    ft = ToyModel().get_text_features(proc["input_ids"], proc["attention_mask"])
+    ft = ft / ft.norm(p=2, dim=-1, keepdim=True)
+    exp = torch.clamp((100 * (fi * ft).sum(-1)).mean(), min=0)
+    assert torch.allclose(out, exp, atol=1e-5)
